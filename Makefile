@@ -1,0 +1,36 @@
+# Build parsec_amd/_core.so (C++/HIP extension, gfx950-only).
+# hipcc cross-compiles device code without a GPU; the .so is built in-tree
+# so it travels to GPU boxes with the source snapshot.
+
+HIPCC      ?= hipcc
+ARCH       ?= gfx950
+PYINC      := $(shell python3 -c "import sysconfig; print(sysconfig.get_paths()['include'])")
+PBINC      := $(shell python3 -c "import pybind11; print(pybind11.get_include())")
+ROCM       ?= /opt/rocm
+
+CXXFLAGS   := -O3 -std=c++17 -fPIC --offload-arch=$(ARCH) \
+              -I$(PYINC) -I$(PBINC) -Isrc -I$(ROCM)/include \
+              -Wall -Wno-unused-function -fvisibility=hidden
+LDFLAGS    := -shared -L$(ROCM)/lib -lrocblas -lrocsolver -lrccl
+
+SRCS       := src/common.cpp src/runtime.cpp src/device_gpu.cpp src/comm.cpp \
+              src/rccl_comm.cpp src/dtd.cpp src/kernels_blas.cpp \
+              src/kernels_hip.cpp src/pybind.cpp
+OBJS       := $(SRCS:src/%.cpp=build/%.o)
+TARGET     := parsec_amd/_core.so
+
+all: $(TARGET)
+
+build/%.o: src/%.cpp src/*.hpp | build
+	$(HIPCC) $(CXXFLAGS) -c $< -o $@
+
+build:
+	mkdir -p build
+
+$(TARGET): $(OBJS)
+	$(HIPCC) $(OBJS) $(LDFLAGS) -o $@
+
+clean:
+	rm -rf build $(TARGET)
+
+.PHONY: all clean

@@ -1,0 +1,134 @@
+#!/usr/bin/env python3
+"""Flagship benchmark: whole-node tiled Cholesky fp64 (BASELINE.json metric).
+
+One step = one full tiled Cholesky factorization (DTD DAG: POTRF/TRSM/SYRK/
+GEMM tile tasks on the HIP engine) of a synthetic SPD fp64 matrix of order
+--matrix-size, preceded by its on-GPU re-fill (the factorization is in
+place; the fill is part of the timed step and costs <1% of it).
+
+Launched by the driver as:
+  python bench.py --gpus 1 ...                      (single rank)
+  python -m torch.distributed.run --nnodes=1 --nproc-per-node N \
+      --master-addr 127.0.0.1 bench.py --gpus N ... (one rank per GPU, RCCL)
+
+Rank 0 prints one JSON line with the whole-job aggregate GFLOP/s.
+"""
+import argparse
+import json
+import os
+import sys
+import time
+
+sys.path.insert(0, os.path.dirname(os.path.abspath(__file__)))
+
+
+def pick_grid(world):
+    return {1: (1, 1), 2: (2, 1), 4: (2, 2), 8: (2, 4)}.get(
+        world, (1, world))
+
+
+def main():
+    ap = argparse.ArgumentParser()
+    ap.add_argument("--gpus", type=int, default=1)
+    ap.add_argument("--steps", type=int, default=5)
+    ap.add_argument("--warmup", type=int, default=2)
+    ap.add_argument("--matrix-size", type=int, default=80000)
+    ap.add_argument("--tile", type=int, default=2048)
+    ap.add_argument("--chore-gemm", type=str, default=os.environ.get(
+        "PARSEC_MCA_chore_gemm", "rocblas"), choices=["rocblas", "hip"])
+    ap.add_argument("--workers", type=int, default=4)
+    args = ap.parse_args()
+
+    import parsec_amd as pm
+
+    world = int(os.environ.get("WORLD_SIZE", "1"))
+    rank = int(os.environ.get("RANK", "0"))
+    has_gpu = pm.hip_device_count() > 0
+    n, nb = args.matrix_size, args.tile
+    if not has_gpu:
+        # debug-only path for CPU containers; GPU boxes run the real config
+        n, nb = 1024, 128
+    pm.param_set("chore_gemm", args.chore_gemm)
+
+    ctx = pm.init_distributed(nworkers=args.workers)
+    assert ctx.world == world
+    p, q = pick_grid(world)
+    A = pm.TiledMatrix(ctx, n, n, nb, nb, p, q)
+
+    use_torch_dist = world > 1
+    if use_torch_dist:
+        import torch.distributed as dist
+
+    def barrier_sync():
+        ctx.gpu_sync()
+        if has_gpu:
+            try:
+                import torch
+                if torch.cuda.is_available():
+                    torch.cuda.synchronize()
+            except Exception:
+                pass
+        if use_torch_dist:
+            dist.barrier()
+        else:
+            ctx.barrier()
+
+    def one_step():
+        tp = pm.Dtd(ctx)
+        pm.insert_spd_fill(tp, A, 42)
+        pm.insert_potrf(tp, A)
+        tp.wait()
+
+    for _ in range(args.warmup):
+        one_step()
+        barrier_sync()
+
+    barrier_sync()
+    t0 = time.perf_counter()
+    for _ in range(args.steps):
+        one_step()
+    barrier_sync()
+    elapsed = time.perf_counter() - t0
+
+    # max over ranks
+    if use_torch_dist:
+        import torch
+        t = torch.tensor([elapsed], dtype=torch.float64)
+        dist.all_reduce(t, op=dist.ReduceOp.MAX)
+        elapsed = float(t.item())
+
+    flops_per_step = n**3 / 3.0 + n**2 / 2.0 + n / 6.0
+    value = args.steps * flops_per_step / elapsed / 1e9  # GFLOP/s whole job
+    if rank == 0:
+        out = {
+            "metric": "GFLOP/s (whole node) tiled Cholesky fp64",
+            "value": round(value, 1),
+            "unit": "GFLOP/s",
+            "n_gpus": world,
+            "steps": args.steps,
+            "warmup": args.warmup,
+            "ms_per_step": round(elapsed / args.steps * 1e3, 2),
+            "higher_is_better": True,
+            "scaling": "strong",
+            "vs_baseline": None,
+            "dtype": "fp64",
+            "data": "synthetic",
+            "config": {
+                "model": "tiled_cholesky_dpotrf",
+                "N": n,
+                "tile": nb,
+                "parallelism": f"dtd-2d-block-cyclic-p{p}q{q}",
+                "chore_gemm": args.chore_gemm,
+                "gpu": has_gpu,
+            },
+        }
+        print(json.dumps(out), flush=True)
+    if use_torch_dist:
+        import torch.distributed as dist2
+        dist2.destroy_process_group()
+    del A
+    del ctx
+
+
+if __name__ == "__main__":
+    main()

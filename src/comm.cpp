@@ -1,0 +1,373 @@
+#include "comm.hpp"
+
+#include <arpa/inet.h>
+#include <fcntl.h>
+#include <netinet/in.h>
+#include <netinet/tcp.h>
+#include <poll.h>
+#include <sys/socket.h>
+#include <unistd.h>
+
+#include <cstring>
+#include <deque>
+#include <unordered_map>
+
+#include "data.hpp"
+
+namespace pa {
+
+void NullComm::enqueue(Task* t) {
+  // With world==1 no send/recv task should ever be created.
+  fatal("NullComm: unexpected %s task", t->tc->name.c_str());
+}
+
+// ===================================================================== TCP
+// Host-socket engine: the CPU-testable comm backend and the control plane
+// pattern. Frames per (src,dst) stream are ordered by TCP itself; the
+// per-channel sequence tags assert the deterministic matching.
+namespace {
+
+struct FrameHeader {
+  uint32_t kind;  // 1=DATA 2=BAR_IN 3=BAR_OUT
+  uint32_t pad;
+  uint64_t seq;
+  uint64_t size;
+};
+enum { FK_DATA = 1, FK_BAR_IN = 2, FK_BAR_OUT = 3 };
+
+class TcpComm : public CommEngine {
+ public:
+  TcpComm(Context* ctx) : ctx_(ctx), rank_(ctx->rank()), world_(ctx->world()) {
+    setup_mesh();
+    PA_CHECK(pipe(wake_pipe_) == 0);
+    set_nonblock(wake_pipe_[0]);
+    thr_ = std::thread([this] { main_loop(); });
+  }
+
+  ~TcpComm() override {
+    stop_.store(true);
+    wake();
+    if (thr_.joinable()) thr_.join();
+    for (auto& p : peers_)
+      if (p.fd >= 0) close(p.fd);
+    close(wake_pipe_[0]);
+    close(wake_pipe_[1]);
+  }
+
+  const char* kind() const override { return "tcp"; }
+
+  void enqueue(Task* t) override {
+    {
+      std::lock_guard<std::mutex> g(cmd_mtx_);
+      cmds_.push_back(t);
+    }
+    wake();
+  }
+
+  void barrier() override {
+    std::unique_lock<std::mutex> g(bar_mtx_);
+    uint64_t my_epoch = ++bar_epoch_started_;
+    g.unlock();
+    {
+      std::lock_guard<std::mutex> g2(cmd_mtx_);
+      bar_requests_.push_back(my_epoch);
+    }
+    wake();
+    g.lock();
+    bar_cv_.wait(g, [&] { return bar_epoch_done_ >= my_epoch; });
+  }
+
+ private:
+  struct Peer {
+    int fd = -1;
+    // outgoing
+    std::deque<std::pair<std::vector<uint8_t>, Task*>> out;  // framed bytes
+    size_t out_off = 0;
+    // incoming state machine
+    FrameHeader hdr;
+    size_t hdr_got = 0;
+    std::vector<uint8_t> in_payload;
+    uint8_t* in_direct = nullptr;  // direct-to-tile read target
+    Task* in_task = nullptr;
+    size_t in_got = 0;
+    bool in_header_done = false;
+  };
+
+  void set_nonblock(int fd) {
+    fcntl(fd, F_SETFL, fcntl(fd, F_GETFL, 0) | O_NONBLOCK);
+  }
+
+  void setup_mesh() {
+    peers_.resize(world_);
+    if (world_ == 1) return;
+    const int base = (int)param_int("comm_base_port", 29650);
+    std::string host = param_str("comm_host", "127.0.0.1");
+    int lfd = socket(AF_INET, SOCK_STREAM, 0);
+    int one = 1;
+    setsockopt(lfd, SOL_SOCKET, SO_REUSEADDR, &one, sizeof(one));
+    sockaddr_in addr{};
+    addr.sin_family = AF_INET;
+    addr.sin_addr.s_addr = INADDR_ANY;
+    addr.sin_port = htons((uint16_t)(base + rank_));
+    PA_CHECK(bind(lfd, (sockaddr*)&addr, sizeof(addr)) == 0,
+             "comm: bind port %d failed", base + rank_);
+    PA_CHECK(listen(lfd, world_) == 0);
+    // connect to lower ranks
+    for (int s = 0; s < rank_; s++) {
+      int fd = socket(AF_INET, SOCK_STREAM, 0);
+      sockaddr_in peer{};
+      peer.sin_family = AF_INET;
+      peer.sin_port = htons((uint16_t)(base + s));
+      inet_pton(AF_INET, host.c_str(), &peer.sin_addr);
+      double t0 = now_s();
+      while (connect(fd, (sockaddr*)&peer, sizeof(peer)) != 0) {
+        PA_CHECK(now_s() - t0 < 60.0, "comm: connect to rank %d timed out", s);
+        usleep(20000);
+        close(fd);
+        fd = socket(AF_INET, SOCK_STREAM, 0);
+      }
+      uint32_t me = (uint32_t)rank_;
+      PA_CHECK(write(fd, &me, 4) == 4);
+      peers_[s].fd = fd;
+    }
+    // accept from higher ranks
+    for (int i = 0; i < world_ - 1 - rank_; i++) {
+      int fd = accept(lfd, nullptr, nullptr);
+      PA_CHECK(fd >= 0, "comm: accept failed");
+      uint32_t who = ~0u;
+      PA_CHECK(read(fd, &who, 4) == 4);
+      PA_CHECK((int)who > rank_ && (int)who < world_);
+      peers_[who].fd = fd;
+    }
+    close(lfd);
+    for (auto& p : peers_)
+      if (p.fd >= 0) {
+        int one2 = 1;
+        setsockopt(p.fd, IPPROTO_TCP, TCP_NODELAY, &one2, sizeof(one2));
+        set_nonblock(p.fd);
+      }
+  }
+
+  void wake() { (void)!write(wake_pipe_[1], "x", 1); }
+
+  void queue_frame(int peer, uint32_t kind, uint64_t seq, const void* payload,
+                   uint64_t size, Task* done_task) {
+    Peer& p = peers_[peer];
+    std::vector<uint8_t> buf(sizeof(FrameHeader) + size);
+    FrameHeader h{kind, 0, seq, size};
+    memcpy(buf.data(), &h, sizeof(h));
+    if (size) memcpy(buf.data() + sizeof(h), payload, size);
+    p.out.emplace_back(std::move(buf), done_task);
+  }
+
+  void process_cmd(Task* t) {
+    Data* d = t->flows[0].data;
+    if (t->tc->kind == TaskKind::COMM_SEND) {
+      void* ptr = d->pull_to_host();
+      queue_frame(t->peer, FK_DATA, t->comm_seq, ptr, d->bytes, t);
+    } else {  // COMM_RECV
+      uint64_t key = ((uint64_t)t->peer << 48) | t->comm_seq;
+      auto it = unexpected_.find(key);
+      if (it != unexpected_.end()) {
+        memcpy(d->ensure_host(), it->second.data(), d->bytes);
+        unexpected_.erase(it);
+        d->written_on(false);
+        task_complete(t);
+      } else {
+        posted_recv_[key] = t;
+      }
+    }
+  }
+
+  void handle_frame(int peer) {
+    Peer& p = peers_[peer];
+    if (p.hdr.kind == FK_DATA) {
+      if (p.in_task) {
+        Data* d = p.in_task->flows[0].data;
+        d->written_on(false);
+        task_complete(p.in_task);
+        p.in_task = nullptr;
+        p.in_direct = nullptr;
+      } else {
+        uint64_t key = ((uint64_t)peer << 48) | p.hdr.seq;
+        unexpected_[key] = std::move(p.in_payload);
+        p.in_payload.clear();
+      }
+    } else if (p.hdr.kind == FK_BAR_IN) {
+      bar_arrivals_[p.hdr.seq]++;
+      check_barrier_root();
+    } else if (p.hdr.kind == FK_BAR_OUT) {
+      std::lock_guard<std::mutex> g(bar_mtx_);
+      bar_epoch_done_ = p.hdr.seq;
+      bar_cv_.notify_all();
+    }
+    p.in_header_done = false;
+    p.hdr_got = 0;
+    p.in_got = 0;
+  }
+
+  void check_barrier_root() {
+    if (rank_ != 0) return;
+    uint64_t e = bar_epoch_root_next_;
+    if (bar_arrivals_[e] >= (uint64_t)(world_ - 1) && bar_root_armed_ >= e) {
+      for (int s = 1; s < world_; s++)
+        queue_frame(s, FK_BAR_OUT, e, nullptr, 0, nullptr);
+      bar_arrivals_.erase(e);
+      bar_epoch_root_next_++;
+      std::lock_guard<std::mutex> g(bar_mtx_);
+      bar_epoch_done_ = e;
+      bar_cv_.notify_all();
+    }
+  }
+
+  void do_read(int peer) {
+    Peer& p = peers_[peer];
+    for (;;) {
+      if (!p.in_header_done) {
+        ssize_t r = read(p.fd, (uint8_t*)&p.hdr + p.hdr_got,
+                         sizeof(FrameHeader) - p.hdr_got);
+        if (r <= 0) {
+          if (r == 0 || (errno != EAGAIN && errno != EWOULDBLOCK))
+            peer_down(peer, r);
+          return;
+        }
+        p.hdr_got += (size_t)r;
+        if (p.hdr_got < sizeof(FrameHeader)) return;
+        p.in_header_done = true;
+        p.in_got = 0;
+        if (p.hdr.kind == FK_DATA) {
+          uint64_t key = ((uint64_t)peer << 48) | p.hdr.seq;
+          auto it = posted_recv_.find(key);
+          if (it != posted_recv_.end()) {
+            p.in_task = it->second;
+            posted_recv_.erase(it);
+            p.in_direct = (uint8_t*)p.in_task->flows[0].data->ensure_host();
+          } else {
+            p.in_task = nullptr;
+            p.in_direct = nullptr;
+            p.in_payload.resize(p.hdr.size);
+          }
+        }
+        if (p.hdr.size == 0) {
+          handle_frame(peer);
+          continue;
+        }
+      }
+      uint8_t* dst = p.in_direct ? p.in_direct : p.in_payload.data();
+      ssize_t r = read(p.fd, dst + p.in_got, p.hdr.size - p.in_got);
+      if (r <= 0) {
+        if (r == 0 || (errno != EAGAIN && errno != EWOULDBLOCK))
+          peer_down(peer, r);
+        return;
+      }
+      p.in_got += (size_t)r;
+      if (p.in_got < p.hdr.size) return;
+      handle_frame(peer);
+    }
+  }
+
+  void do_write(int peer) {
+    Peer& p = peers_[peer];
+    while (!p.out.empty()) {
+      auto& [buf, done] = p.out.front();
+      ssize_t w = write(p.fd, buf.data() + p.out_off, buf.size() - p.out_off);
+      if (w < 0) {
+        if (errno != EAGAIN && errno != EWOULDBLOCK) peer_down(peer, w);
+        return;
+      }
+      p.out_off += (size_t)w;
+      if (p.out_off < buf.size()) return;
+      if (done) task_complete(done);
+      p.out.pop_front();
+      p.out_off = 0;
+    }
+  }
+
+  void peer_down(int peer, ssize_t r) {
+    if (stop_.load()) return;
+    fatal("comm: connection to rank %d lost (r=%zd errno=%d)", peer, r, errno);
+  }
+
+  void main_loop() {
+    std::vector<pollfd> pfds;
+    while (!stop_.load(std::memory_order_acquire)) {
+      // drain command queue
+      std::vector<Task*> cmds;
+      std::vector<uint64_t> bars;
+      {
+        std::lock_guard<std::mutex> g(cmd_mtx_);
+        cmds.swap(cmds_);
+        bars.swap(bar_requests_);
+      }
+      for (Task* t : cmds) process_cmd(t);
+      for (uint64_t e : bars) {
+        if (rank_ == 0) {
+          bar_root_armed_ = std::max(bar_root_armed_, e);
+          check_barrier_root();
+        } else {
+          queue_frame(0, FK_BAR_IN, e, nullptr, 0, nullptr);
+        }
+      }
+      // poll
+      pfds.clear();
+      pfds.push_back({wake_pipe_[0], POLLIN, 0});
+      for (int s = 0; s < world_; s++) {
+        if (peers_[s].fd < 0) continue;
+        short ev = POLLIN;
+        if (!peers_[s].out.empty()) ev |= POLLOUT;
+        pfds.push_back({peers_[s].fd, ev, 0});
+      }
+      int rc = poll(pfds.data(), (nfds_t)pfds.size(), 50);
+      if (rc <= 0) continue;
+      if (pfds[0].revents & POLLIN) {
+        char buf[256];
+        while (read(wake_pipe_[0], buf, sizeof(buf)) > 0) {}
+      }
+      size_t pi = 1;
+      for (int s = 0; s < world_; s++) {
+        if (peers_[s].fd < 0) continue;
+        short re = pfds[pi++].revents;
+        if (re & POLLIN) do_read(s);
+        if (re & POLLOUT) do_write(s);
+        if (re & (POLLERR | POLLHUP)) peer_down(s, -1);
+      }
+    }
+  }
+
+  Context* ctx_;
+  int rank_, world_;
+  std::vector<Peer> peers_;
+  int wake_pipe_[2] = {-1, -1};
+  std::thread thr_;
+  std::atomic<bool> stop_{false};
+
+  std::mutex cmd_mtx_;
+  std::vector<Task*> cmds_;
+  std::vector<uint64_t> bar_requests_;
+
+  std::unordered_map<uint64_t, Task*> posted_recv_;
+  std::unordered_map<uint64_t, std::vector<uint8_t>> unexpected_;
+
+  std::mutex bar_mtx_;
+  std::condition_variable bar_cv_;
+  uint64_t bar_epoch_started_ = 0, bar_epoch_done_ = 0;
+  uint64_t bar_epoch_root_next_ = 1, bar_root_armed_ = 0;
+  std::unordered_map<uint64_t, uint64_t> bar_arrivals_;
+};
+
+}  // namespace
+
+std::unique_ptr<CommEngine> create_rccl_comm(Context* ctx);  // rccl_comm.cpp
+
+std::unique_ptr<CommEngine> CommEngine::create(Context* ctx,
+                                               const std::string& kind) {
+  std::string k = kind.empty() ? param_str("comm_kind", "") : kind;
+  if (k.empty()) k = ctx->world() > 1 ? "tcp" : "null";
+  if (ctx->world() == 1 || k == "null")
+    return std::make_unique<NullComm>();
+  if (k == "tcp") return std::make_unique<TcpComm>(ctx);
+  if (k == "rccl") return create_rccl_comm(ctx);
+  fatal("unknown comm kind '%s'", k.c_str());
+}
+
+}  // namespace pa

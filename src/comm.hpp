@@ -1,0 +1,45 @@
+// Communication engine: dataflow transfers between ranks (= GPUs).
+//
+// Reference parity: the comm-engine vtable
+// (parsec_comm_engine.h:14-207) + funnelled comm thread
+// (remote_dep_mpi.c:423-481, 1143-1271), re-designed for one MI355X node:
+//  - Deterministic SPMD task insertion makes every transfer predictable on
+//    both ends, so the eager-activate/GET handshake of remote_dep_mpi.c
+//    collapses to matched send/recv pairs ordered per (src,dst) channel by a
+//    sequence number every rank derives independently — no activation AMs,
+//    no rank-bitmap reconstruction on the wire.
+//  - Backends: "tcp" (host sockets; the CPU-testable engine, also the
+//    control plane) and "rccl" (ncclSend/ncclRecv on xGMI, device-resident
+//    payloads, one stream per peer to keep the 7 xGMI links independently
+//    busy). Multi-node would layer MPI under the same interface.
+#pragma once
+
+#include <memory>
+#include <string>
+
+#include "common.hpp"
+#include "runtime.hpp"
+
+namespace pa {
+
+class CommEngine {
+ public:
+  static std::unique_ptr<CommEngine> create(Context* ctx, const std::string& kind);
+  virtual ~CommEngine() = default;
+
+  // Send/recv task whose local dependencies are satisfied. The engine owns
+  // completion (task_complete) once the wire transfer finishes.
+  virtual void enqueue(Task* t) = 0;
+  virtual void barrier() = 0;
+  virtual const char* kind() const = 0;
+};
+
+// world==1: no-op engine.
+class NullComm : public CommEngine {
+ public:
+  void enqueue(Task* t) override;
+  void barrier() override {}
+  const char* kind() const override { return "null"; }
+};
+
+}  // namespace pa

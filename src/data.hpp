@@ -1,0 +1,104 @@
+// Data subsystem: versioned multi-device data (tiles) + tiled collections.
+//
+// Reference parity (SURVEY.md §2.1 data subsystem, §2.6 collections):
+//  - pa::Data = parsec_data_t + its parsec_data_copy_t set (data.c:153-260,
+//    data_internal.h:30-85) collapsed to the two locations that exist on an
+//    MI355X node process: host DRAM and this process's GPU HBM. Peer-GPU
+//    copies live in peer *processes* and move over RCCL (comm.hpp).
+//  - Version/ownership transfer (data.c:301-360) becomes {host_valid,
+//    dev_valid} + writer-invalidates semantics under Data::lock.
+//  - TiledMatrix = parsec_matrix_block_cyclic_t (two_dim_rectangle_cyclic.c)
+//    with storage sized for 288 GB HBM3E: tiles live on the GPU by default
+//    and host copies are created only on demand.
+#pragma once
+
+#include <cstdint>
+#include <unordered_map>
+#include <vector>
+
+#include "common.hpp"
+#include "runtime.hpp"
+
+namespace pa {
+
+class Context;
+
+// Where the authoritative copy of a tile currently is.
+struct Data {
+  // identity
+  uint64_t key = 0;
+  class TiledMatrix* coll = nullptr;
+  int home_rank = 0;
+  size_t bytes = 0;
+
+  SpinLock lock;  // guards the copy/validity state below
+  void* host_ptr = nullptr;
+  void* dev_ptr = nullptr;
+  bool host_valid = false;
+  bool dev_valid = false;
+
+  // ---- DTD chaining state (single inserter thread; no lock needed) ----
+  uint32_t version = 0;        // logical version, bumped per writer insertion
+  int owner_rank = 0;          // rank owning `version` (SPMD-tracked)
+  Task* last_local_writer = nullptr;     // last local task writing the buffer
+  std::vector<Task*> local_readers;      // local users since that writer
+  uint64_t sent_mask = 0;                // ranks already sent current version
+  uint32_t local_present_version = 0;    // version the local buffer will hold
+  bool local_present = false;
+
+  ~Data();
+
+  // Ensure a host buffer exists (allocates page-aligned memory).
+  void* ensure_host();
+  // Make the host copy valid (D2H if needed). Safe from any thread.
+  void* pull_to_host();
+  // Invalidate all copies except the one on `device` (true=GPU).
+  void written_on(bool device);
+};
+
+// 2D block-cyclic tiled matrix of an elementary type (fp64 for the Cholesky
+// headline; elem_size parametrizes dtype).
+class TiledMatrix {
+ public:
+  TiledMatrix(Context* ctx, int64_t m, int64_t n, int mb, int nb,
+              int p, int q, size_t elem_size = 8);
+  ~TiledMatrix();
+
+  Context* ctx() const { return ctx_; }
+  int64_t m() const { return m_; }
+  int64_t n() const { return n_; }
+  int mb() const { return mb_; }
+  int nb() const { return nb_; }
+  int mt() const { return mt_; }
+  int nt() const { return nt_; }
+  int grid_p() const { return p_; }
+  int grid_q() const { return q_; }
+  size_t elem_size() const { return elem_; }
+  size_t tile_bytes() const { return (size_t)mb_ * nb_ * elem_; }
+
+  int rank_of(int tm, int tn) const { return (tm % p_) * q_ + (tn % q_); }
+  bool is_local(int tm, int tn) const {
+    return rank_of(tm, tn) == ctx_rank_;
+  }
+  Data* tile(int tm, int tn);
+
+  // rows/cols of a (possibly partial) edge tile
+  int tile_rows(int tm) const {
+    int64_t r = m_ - (int64_t)tm * mb_;
+    return r >= mb_ ? mb_ : (int)r;
+  }
+  int tile_cols(int tn) const {
+    int64_t c = n_ - (int64_t)tn * nb_;
+    return c >= nb_ ? nb_ : (int)c;
+  }
+
+ private:
+  Context* ctx_;
+  int ctx_rank_;
+  int64_t m_, n_;
+  int mb_, nb_, mt_, nt_, p_, q_;
+  size_t elem_;
+  std::vector<std::unique_ptr<Data>> tiles_;  // mt*nt, metadata eager
+};
+
+}  // namespace pa

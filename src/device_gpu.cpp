@@ -1,0 +1,304 @@
+#include "device_gpu.hpp"
+
+#include "data.hpp"
+
+namespace pa {
+
+std::unique_ptr<GpuEngine> GpuEngine::create(Context* ctx, int device) {
+  int count = 0;
+  hipError_t e = hipGetDeviceCount(&count);
+  if (e != hipSuccess || count == 0) {
+    PA_DEBUG(1, "no HIP device visible; CPU-only context");
+    return nullptr;
+  }
+  if (device < 0) {
+    // one process per GPU: local rank picks the device; respects
+    // HIP_VISIBLE_DEVICES/torchrun LOCAL_RANK when present.
+    const char* lr = getenv("LOCAL_RANK");
+    device = lr ? atoi(lr) % count : ctx->rank() % count;
+  }
+  return std::unique_ptr<GpuEngine>(new GpuEngine(ctx, device));
+}
+
+GpuEngine::GpuEngine(Context* ctx, int device) : ctx_(ctx), device_(device) {
+  PA_HIP_CHECK(hipSetDevice(device_));
+  PA_HIP_CHECK(hipStreamCreateWithFlags(&h2d_stream_, hipStreamNonBlocking));
+  PA_HIP_CHECK(hipStreamCreateWithFlags(&d2h_stream_, hipStreamNonBlocking));
+  PA_HIP_CHECK(hipStreamCreateWithFlags(&comm_stream_, hipStreamNonBlocking));
+  int nstreams = (int)param_int("gpu_exec_streams", 4);
+  exec_streams_.resize(nstreams);
+  inflight_.resize(nstreams);
+  for (int i = 0; i < nstreams; i++)
+    PA_HIP_CHECK(hipStreamCreateWithFlags(&exec_streams_[i], hipStreamNonBlocking));
+  max_inflight_per_stream_ = (size_t)param_int("gpu_max_inflight", 32);
+
+  // Reserve the HBM slab (parsec_device_memory_reserve analog,
+  // device_gpu.c:867-992). Default 85% of free memory; kernels and
+  // libraries get the rest.
+  size_t free_b = 0, total_b = 0;
+  PA_HIP_CHECK(hipMemGetInfo(&free_b, &total_b));
+  int pct = (int)param_int("gpu_mem_percent", 85);
+  slab_bytes_ = free_b / 100 * pct;
+  if (slab_bytes_) {
+    hipError_t e = hipMalloc(&slab_, slab_bytes_);
+    if (e != hipSuccess) {
+      slab_bytes_ /= 2;
+      PA_HIP_CHECK(hipMalloc(&slab_, slab_bytes_));
+    }
+  }
+  PA_DEBUG(1, "GPU %d engine: %d exec streams, slab %.1f GB", device_,
+           nstreams, slab_bytes_ / 1e9);
+  manager_ = std::thread([this] { manager_main(); });
+}
+
+GpuEngine::~GpuEngine() {
+  stop_.store(true, std::memory_order_release);
+  q_cv_.notify_all();
+  if (manager_.joinable()) manager_.join();
+  PA_HIP_CHECK(hipSetDevice(device_));
+  for (auto& e : event_pool_) hipEventDestroy(e);
+  for (auto s : exec_streams_) hipStreamDestroy(s);
+  hipStreamDestroy(h2d_stream_);
+  hipStreamDestroy(d2h_stream_);
+  hipStreamDestroy(comm_stream_);
+  if (slab_) hipFree(slab_);
+}
+
+void GpuEngine::enqueue(Task* t) {
+  {
+    std::lock_guard<std::mutex> g(q_mtx_);
+    queue_.push(PQEntry{t});
+  }
+  q_cv_.notify_one();
+}
+
+hipEvent_t GpuEngine::event_get() {
+  if (!event_pool_.empty()) {
+    hipEvent_t e = event_pool_.back();
+    event_pool_.pop_back();
+    return e;
+  }
+  hipEvent_t e;
+  PA_HIP_CHECK(hipEventCreateWithFlags(&e, hipEventDisableTiming));
+  return e;
+}
+
+void GpuEngine::event_put(hipEvent_t e) { event_pool_.push_back(e); }
+
+void* GpuEngine::dev_alloc(size_t bytes) {
+  bytes = (bytes + 255) & ~size_t(255);
+  std::lock_guard<std::mutex> g(mem_mtx_);
+  auto it = free_lists_.find(bytes);
+  if (it != free_lists_.end() && !it->second.empty()) {
+    void* p = it->second.back();
+    it->second.pop_back();
+    return p;
+  }
+  if (slab_used_ + bytes <= slab_bytes_) {
+    void* p = (char*)slab_ + slab_used_;
+    slab_used_ += bytes;
+    return p;
+  }
+  // Slab exhausted: direct allocation escape hatch (LRU eviction is the
+  // next capacity tier; see docs/DESIGN.md).
+  void* p = nullptr;
+  hipError_t e = hipMalloc(&p, bytes);
+  if (e != hipSuccess)
+    fatal("GPU %d out of memory allocating %zu bytes (slab %zu/%zu)",
+          device_, bytes, slab_used_, slab_bytes_);
+  return p;
+}
+
+void GpuEngine::dev_free(void* p, size_t bytes) {
+  bytes = (bytes + 255) & ~size_t(255);
+  std::lock_guard<std::mutex> g(mem_mtx_);
+  free_lists_[bytes].push_back(p);
+}
+
+void GpuEngine::copy_d2h(void* dst, const void* src, size_t bytes) {
+  PA_HIP_CHECK(hipSetDevice(device_));
+  PA_HIP_CHECK(hipMemcpyAsync(dst, src, bytes, hipMemcpyDeviceToHost, d2h_stream_));
+  PA_HIP_CHECK(hipStreamSynchronize(d2h_stream_));
+  stats.bytes_d2h += bytes;
+}
+
+void GpuEngine::copy_h2d(void* dst, const void* src, size_t bytes) {
+  PA_HIP_CHECK(hipSetDevice(device_));
+  PA_HIP_CHECK(hipMemcpyAsync(dst, src, bytes, hipMemcpyHostToDevice, h2d_stream_));
+  PA_HIP_CHECK(hipStreamSynchronize(h2d_stream_));
+  stats.bytes_h2d += bytes;
+}
+
+void GpuEngine::sync_all() {
+  PA_HIP_CHECK(hipSetDevice(device_));
+  PA_HIP_CHECK(hipDeviceSynchronize());
+}
+
+// Stage-in copies for every READ flow whose valid copy is on the host
+// (parsec_device_data_stage_in, device_gpu.c:1800-2168, minus the peer-GPU
+// branch: peers are other processes here, reached through the comm engine).
+void GpuEngine::stage_in_and_launch(Task* t) {
+  hipStream_t es = exec_streams_[next_stream_];
+  bool copied = false;
+  for (int i = 0; i < t->nflows; i++) {
+    Data* d = t->flows[i].data;
+    if (!d) { t->dev_ptr[i] = nullptr; continue; }
+    SpinGuard g(d->lock);
+    if (!d->dev_ptr) d->dev_ptr = dev_alloc(d->bytes);
+    if ((t->flows[i].mode & ACCESS_IN) && !d->dev_valid) {
+      PA_CHECK(d->host_valid, "stage-in: no valid copy for tile");
+      PA_HIP_CHECK(hipMemcpyAsync(d->dev_ptr, d->host_ptr, d->bytes,
+                                  hipMemcpyHostToDevice, h2d_stream_));
+      stats.bytes_h2d += d->bytes;
+      d->dev_valid = true;  // valid in stream order on h2d_stream_
+      copied = true;
+    }
+    if (!(t->flows[i].mode & ACCESS_IN) && !d->dev_valid) {
+      d->dev_valid = true;  // OUTPUT-only: content produced by this task
+    }
+    t->dev_ptr[i] = d->dev_ptr;
+  }
+  if (copied) {
+    hipEvent_t ce = event_get();
+    PA_HIP_CHECK(hipEventRecord(ce, h2d_stream_));
+    PA_HIP_CHECK(hipStreamWaitEvent(es, ce, 0));
+    event_put(ce);  // safe: event re-record happens on this thread only
+  }
+  GpuTaskCtx gctx{es, device_, this};
+  t->tc->gpu_hook(*t, gctx);
+  hipEvent_t ev = event_get();
+  PA_HIP_CHECK(hipEventRecord(ev, es));
+  inflight_[next_stream_].push_back(InFlight{t, ev, next_stream_});
+  n_inflight_++;
+  next_stream_ = (next_stream_ + 1) % (int)exec_streams_.size();
+}
+
+void GpuEngine::manager_main() {
+  PA_HIP_CHECK(hipSetDevice(device_));
+  while (true) {
+    // 1) retire completed tasks (in-order per stream)
+    bool progress = false;
+    for (auto& ring : inflight_) {
+      while (!ring.empty()) {
+        InFlight& f = ring.front();
+        hipError_t e = hipEventQuery(f.event);
+        if (e == hipErrorNotReady) break;
+        PA_HIP_CHECK(e);
+        Task* t = f.task;
+        for (int i = 0; i < t->nflows; i++) {
+          Data* d = t->flows[i].data;
+          if (d && (t->flows[i].mode & ACCESS_OUT)) d->written_on(true);
+        }
+        event_put(f.event);
+        ring.pop_front();
+        n_inflight_--;
+        stats.tasks++;
+        task_complete(t);
+        progress = true;
+      }
+    }
+    // 2) launch new work while there is room
+    while (true) {
+      if (inflight_[next_stream_].size() >= max_inflight_per_stream_) break;
+      Task* t = nullptr;
+      {
+        std::lock_guard<std::mutex> g(q_mtx_);
+        if (!queue_.empty()) { t = queue_.top().t; queue_.pop(); }
+      }
+      if (!t) break;
+      stage_in_and_launch(t);
+      progress = true;
+    }
+    if (stop_.load(std::memory_order_acquire) && n_inflight_ == 0) {
+      std::lock_guard<std::mutex> g(q_mtx_);
+      if (queue_.empty()) break;
+    }
+    if (!progress) {
+      if (n_inflight_ > 0) {
+        // work in flight: poll tightly but yield the core briefly
+        std::this_thread::yield();
+      } else {
+        std::unique_lock<std::mutex> g(q_mtx_);
+        if (queue_.empty() && !stop_.load(std::memory_order_acquire))
+          q_cv_.wait_for(g, std::chrono::microseconds(100));
+      }
+    }
+  }
+}
+
+// ---------------------------------------------------------------- Data
+Data::~Data() {
+  // Drop DTD chaining references to completed tasks.
+  if (last_local_writer) last_local_writer->release();
+  for (Task* r : local_readers) r->release();
+  if (host_ptr) free(host_ptr);
+  // dev_ptr returns to the engine pool with the collection teardown;
+  // the slab itself is freed by the engine.
+}
+
+void* Data::ensure_host() {
+  if (!host_ptr) {
+    if (posix_memalign(&host_ptr, 4096, bytes) != 0)
+      fatal("host allocation of %zu bytes failed", bytes);
+  }
+  return host_ptr;
+}
+
+void* Data::pull_to_host() {
+  GpuEngine* eng = coll ? coll->ctx()->gpu() : nullptr;
+  SpinGuard g(lock);
+  ensure_host();
+  if (!host_valid) {
+    PA_CHECK(dev_valid && eng, "pull_to_host: no valid copy");
+    eng->sync_all();  // quiesce producers before readback
+    eng->copy_d2h(host_ptr, dev_ptr, bytes);
+    host_valid = true;
+  }
+  return host_ptr;
+}
+
+void Data::written_on(bool device) {
+  SpinGuard g(lock);
+  if (device) {
+    dev_valid = true;
+    host_valid = false;
+  } else {
+    host_valid = true;
+    dev_valid = false;
+  }
+}
+
+// ---------------------------------------------------------------- TiledMatrix
+TiledMatrix::TiledMatrix(Context* ctx, int64_t m, int64_t n, int mb, int nb,
+                         int p, int q, size_t elem_size)
+    : ctx_(ctx), ctx_rank_(ctx->rank()), m_(m), n_(n), mb_(mb), nb_(nb),
+      p_(p), q_(q), elem_(elem_size) {
+  mt_ = (int)((m + mb - 1) / mb);
+  nt_ = (int)((n + nb - 1) / nb);
+  PA_CHECK(p_ * q_ == ctx->world(), "grid p*q must equal world size");
+  tiles_.resize((size_t)mt_ * nt_);
+}
+
+TiledMatrix::~TiledMatrix() {
+  GpuEngine* eng = ctx_->gpu();
+  for (auto& t : tiles_) {
+    if (t && t->dev_ptr && eng) eng->dev_free(t->dev_ptr, t->bytes);
+  }
+}
+
+Data* TiledMatrix::tile(int tm, int tn) {
+  PA_CHECK(tm >= 0 && tm < mt_ && tn >= 0 && tn < nt_);
+  size_t idx = (size_t)tm * nt_ + tn;
+  if (!tiles_[idx]) {
+    auto d = std::make_unique<Data>();
+    d->key = idx;
+    d->coll = this;
+    d->home_rank = rank_of(tm, tn);
+    d->owner_rank = d->home_rank;
+    d->bytes = tile_bytes();
+    tiles_[idx] = std::move(d);
+  }
+  return tiles_[idx].get();
+}
+
+}  // namespace pa

@@ -1,0 +1,64 @@
+// DTD — Dynamic Task Discovery interface.
+//
+// Reference parity (SURVEY.md §2.5 DTD, interfaces/dtd/insert_function.c):
+// sequential-looking insert_task calls; per-tile last-writer/readers chaining
+// builds RAW/WAR/WAW edges (insert_function.c:3027-3120); window throttling
+// (insert_function.c:75-76); data_flush pushes final versions home
+// (parsec_dtd_data_flush.c).
+//
+// Distributed model, MI355X-native: every rank executes the same insertion
+// stream (SPMD, as the reference requires for distributed DTD) and derives
+// every inter-rank transfer deterministically — tile {version, owner_rank,
+// sent_mask} are replicated state machines, and each (src,dst) channel
+// carries a sequence number all ranks compute identically. This replaces the
+// reference's activation-message protocol (remote_dep_mpi.c:1984-2292): on a
+// single node with RCCL p2p over xGMI, matched send/recv pairs in a
+// deterministic per-channel order need no wire handshake at all.
+#pragma once
+
+#include "data.hpp"
+#include "runtime.hpp"
+
+namespace pa {
+
+extern TaskClass COMM_SEND_CLASS;
+extern TaskClass COMM_RECV_CLASS;
+
+class Dtd : public Taskpool {
+ public:
+  Dtd(Context* ctx, std::string name = "dtd");
+  ~Dtd();
+
+  struct FlowSpec {
+    Data* d;
+    AccessMode mode;
+  };
+
+  // Insert one task. `rank` -1 selects the home rank of the first written
+  // tile (AFFINITY default). Runs/creates the task only on its rank; all
+  // ranks update the replicated tile state machines.
+  void insert(const TaskClass* tc, const void* args, size_t args_bytes,
+              const FlowSpec* flows, int nflows, int priority = 0,
+              int rank = -1);
+
+  // Push the current version of d back to its home rank (DTD data_flush).
+  void flush(Data* d);
+  void flush_all(TiledMatrix& A);
+
+ private:
+  void read_flow(Data* d, Task* t, int task_rank);
+  void write_flow(Data* d, Task* t, int task_rank);
+  void make_send(Data* d, int dst, uint64_t seq);
+  void make_recv(Data* d, int src, uint64_t seq);
+  static void set_local_writer(Data* d, Task* w);
+  uint64_t chan_next(int src, int dst) {
+    return chan_seq_[(size_t)src * world_ + dst]++;
+  }
+
+  int me_, world_;
+  std::vector<uint64_t> chan_seq_;
+  int64_t window_;
+  int64_t threshold_;
+};
+
+}  // namespace pa

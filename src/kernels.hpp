@@ -1,0 +1,34 @@
+// Tile task-classes for the dense linear-algebra headline apps.
+//
+// The reference runtime ships no compute kernels — JDF bodies call cuBLAS
+// via dyld (tests/runtime/cuda/stress.jdf:133-137) and dense drivers live in
+// DPLASMA. Here the tile kernel set IS part of the framework (SURVEY.md §6
+// north star): fp64 POTRF/TRSM/SYRK/GEMM tile bodies with two chore
+// incarnations each — rocBLAS/rocSOLVER ("library" chore) and hand-written
+// CDNA4 MFMA HIP kernels (kernels_hip.cpp) — selected by the `chore_gemm`
+// etc. params, mirroring the reference's chore/incarnation machinery
+// (parsec_internal.h:411-459).
+#pragma once
+
+#include "dtd.hpp"
+
+namespace pa {
+
+struct TileArgs {
+  int m = 0, n = 0, k = 0, ld = 0;
+  int64_t i0 = 0, j0 = 0, N = 0;
+  uint32_t seed = 0;
+};
+
+// Registered once; ids stable.
+TaskClass& tc_spd_fill();
+TaskClass& tc_potrf();
+TaskClass& tc_trsm();
+TaskClass& tc_syrk();
+TaskClass& tc_gemm();
+
+// Build the DAGs (DTD insertion; SPMD-safe: call on every rank).
+void insert_spd_fill(Dtd& tp, TiledMatrix& A, uint32_t seed);
+void insert_potrf(Dtd& tp, TiledMatrix& A);
+
+}  // namespace pa

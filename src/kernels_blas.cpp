@@ -1,0 +1,293 @@
+// Library-chore tile kernels (rocBLAS / rocSOLVER) + CPU reference chores +
+// synthetic SPD fill, and the Cholesky DAG builder.
+//
+// CPU chores are straightforward reference implementations used by the
+// no-GPU test path (numerics tests compare the HIP path against these / a
+// NumPy fp64 reference); the GPU path is the production path.
+#include "kernels.hpp"
+
+#include <cmath>
+
+#include <hip/hip_runtime.h>
+#include <rocblas/rocblas.h>
+#include <rocsolver/rocsolver.h>
+
+#include "device_gpu.hpp"
+
+namespace pa {
+
+// ------------------------------------------------------------------ fill
+// Deterministic synthetic SPD matrix: symmetric hash values in [-0.5, 0.5),
+// + N on the diagonal (diagonally dominant => positive definite).
+__host__ __device__ inline double spd_val(int64_t i, int64_t j, int64_t N,
+                                          uint32_t seed) {
+  uint64_t a = (uint64_t)(i < j ? i : j), b = (uint64_t)(i < j ? j : i);
+  uint64_t h = (a * 2654435761ull) ^ (b * 40503ull) ^
+               ((uint64_t)seed * 2246822519ull);
+  h ^= h >> 13;
+  h *= 0x9E3779B97F4A7C15ull;
+  h ^= h >> 32;
+  double v = (double)(h & 0xFFFFFF) / (double)0x1000000 - 0.5;
+  return i == j ? v + (double)N : v;
+}
+
+__global__ void k_spd_fill(double* t, int rows, int cols, int ld, int64_t i0,
+                           int64_t j0, int64_t N, uint32_t seed) {
+  int idx = blockIdx.x * blockDim.x + threadIdx.x;
+  int total = rows * cols;
+  for (; idx < total; idx += gridDim.x * blockDim.x) {
+    int c = idx / rows, r = idx - c * rows;
+    t[(size_t)c * ld + r] = spd_val(i0 + r, j0 + c, N, seed);
+  }
+}
+
+static void cpu_spd_fill(Task& t) {
+  const TileArgs& a = t.arg<TileArgs>();
+  Data* d = t.flows[0].data;
+  double* p = (double*)d->ensure_host();
+  for (int c = 0; c < a.n; c++)
+    for (int r = 0; r < a.m; r++)
+      p[(size_t)c * a.ld + r] = spd_val(a.i0 + r, a.j0 + c, a.N, a.seed);
+  d->written_on(false);
+}
+
+static void gpu_spd_fill(Task& t, GpuTaskCtx& g) {
+  const TileArgs& a = t.arg<TileArgs>();
+  int total = a.m * a.n;
+  int block = 256;
+  int grid = std::min((total + block - 1) / block, 4096);
+  hipLaunchKernelGGL(k_spd_fill, dim3(grid), dim3(block), 0, g.stream,
+                     (double*)t.dev_ptr[0], a.m, a.n, a.ld, a.i0, a.j0, a.N,
+                     a.seed);
+}
+
+// ------------------------------------------------------------------ CPU chores
+// Naive fp64 reference bodies (column-major, ld = tile rows).
+static void cpu_potrf(Task& t) {
+  const TileArgs& a = t.arg<TileArgs>();
+  double* A = (double*)t.flows[0].data->pull_to_host();
+  const int n = a.n, ld = a.ld;
+  for (int j = 0; j < n; j++) {
+    double d = A[(size_t)j * ld + j];
+    for (int k = 0; k < j; k++) d -= A[(size_t)k * ld + j] * A[(size_t)k * ld + j];
+    PA_CHECK(d > 0, "potrf: matrix not SPD at column %d", j);
+    d = std::sqrt(d);
+    A[(size_t)j * ld + j] = d;
+    for (int i = j + 1; i < n; i++) {
+      double s = A[(size_t)j * ld + i];
+      for (int k = 0; k < j; k++) s -= A[(size_t)k * ld + i] * A[(size_t)k * ld + j];
+      A[(size_t)j * ld + i] = s / d;
+    }
+  }
+  t.flows[0].data->written_on(false);
+}
+
+// B = B * L^{-T}  (right, lower, transposed, non-unit) — TRSM of the tile panel.
+static void cpu_trsm(Task& t) {
+  const TileArgs& a = t.arg<TileArgs>();
+  const double* L = (const double*)t.flows[0].data->pull_to_host();
+  double* B = (double*)t.flows[1].data->pull_to_host();
+  const int m = a.m, n = a.n, ld = a.ld;
+  for (int j = 0; j < n; j++) {
+    double inv = 1.0 / L[(size_t)j * ld + j];
+    for (int i = 0; i < m; i++) {
+      double s = B[(size_t)j * ld + i];
+      for (int k = 0; k < j; k++) s -= B[(size_t)k * ld + i] * L[(size_t)k * ld + j];
+      B[(size_t)j * ld + i] = s * inv;
+    }
+  }
+  t.flows[1].data->written_on(false);
+}
+
+// C = C - A*A^T (lower part only guaranteed; full tile computed is fine)
+static void cpu_syrk(Task& t) {
+  const TileArgs& a = t.arg<TileArgs>();
+  const double* A = (const double*)t.flows[0].data->pull_to_host();
+  double* C = (double*)t.flows[1].data->pull_to_host();
+  const int n = a.n, k = a.k, ld = a.ld;
+  for (int j = 0; j < n; j++)
+    for (int i = j; i < n; i++) {
+      double s = 0;
+      for (int p = 0; p < k; p++) s += A[(size_t)p * ld + i] * A[(size_t)p * ld + j];
+      C[(size_t)j * ld + i] -= s;
+    }
+  t.flows[1].data->written_on(false);
+}
+
+// C = C - A*B^T
+static void cpu_gemm(Task& t) {
+  const TileArgs& a = t.arg<TileArgs>();
+  const double* A = (const double*)t.flows[0].data->pull_to_host();
+  const double* B = (const double*)t.flows[1].data->pull_to_host();
+  double* C = (double*)t.flows[2].data->pull_to_host();
+  const int m = a.m, n = a.n, k = a.k, ld = a.ld;
+  for (int j = 0; j < n; j++)
+    for (int i = 0; i < m; i++) {
+      double s = 0;
+      for (int p = 0; p < k; p++) s += A[(size_t)p * ld + i] * B[(size_t)p * ld + j];
+      C[(size_t)j * ld + i] -= s;
+    }
+  t.flows[2].data->written_on(false);
+}
+
+// ------------------------------------------------------------------ GPU chores
+// One rocBLAS handle per GPU-manager thread (hooks run on the manager only).
+static rocblas_handle blas_handle(GpuTaskCtx& g) {
+  static thread_local rocblas_handle h = nullptr;
+  if (!h) {
+    PA_CHECK(rocblas_create_handle(&h) == rocblas_status_success);
+    rocblas_set_pointer_mode(h, rocblas_pointer_mode_host);
+  }
+  rocblas_set_stream(h, g.stream);
+  return h;
+}
+
+static rocblas_int* dev_info() {
+  static thread_local rocblas_int* p = nullptr;
+  if (!p) PA_HIP_CHECK(hipMalloc(&p, sizeof(rocblas_int)));
+  return p;
+}
+
+static void gpu_potrf(Task& t, GpuTaskCtx& g) {
+  const TileArgs& a = t.arg<TileArgs>();
+  rocblas_status s = rocsolver_dpotrf(blas_handle(g), rocblas_fill_lower, a.n,
+                                      (double*)t.dev_ptr[0], a.ld, dev_info());
+  PA_CHECK(s == rocblas_status_success, "rocsolver_dpotrf failed: %d", (int)s);
+}
+
+static void gpu_trsm(Task& t, GpuTaskCtx& g) {
+  const TileArgs& a = t.arg<TileArgs>();
+  const double one = 1.0;
+  rocblas_status s = rocblas_dtrsm(
+      blas_handle(g), rocblas_side_right, rocblas_fill_lower,
+      rocblas_operation_transpose, rocblas_diagonal_non_unit, a.m, a.n, &one,
+      (const double*)t.dev_ptr[0], a.ld, (double*)t.dev_ptr[1], a.ld);
+  PA_CHECK(s == rocblas_status_success, "rocblas_dtrsm failed: %d", (int)s);
+}
+
+static void gpu_syrk(Task& t, GpuTaskCtx& g) {
+  const TileArgs& a = t.arg<TileArgs>();
+  const double mone = -1.0, one = 1.0;
+  rocblas_status s = rocblas_dsyrk(blas_handle(g), rocblas_fill_lower,
+                                   rocblas_operation_none, a.n, a.k, &mone,
+                                   (const double*)t.dev_ptr[0], a.ld, &one,
+                                   (double*)t.dev_ptr[1], a.ld);
+  PA_CHECK(s == rocblas_status_success, "rocblas_dsyrk failed: %d", (int)s);
+}
+
+void gpu_gemm_hip(Task& t, GpuTaskCtx& g);  // kernels_hip.cpp
+
+static void gpu_gemm(Task& t, GpuTaskCtx& g) {
+  static const bool use_hip = param_str("chore_gemm", "rocblas") == "hip";
+  if (use_hip) { gpu_gemm_hip(t, g); return; }
+  const TileArgs& a = t.arg<TileArgs>();
+  const double mone = -1.0, one = 1.0;
+  rocblas_status s = rocblas_dgemm(
+      blas_handle(g), rocblas_operation_none, rocblas_operation_transpose,
+      a.m, a.n, a.k, &mone, (const double*)t.dev_ptr[0], a.ld,
+      (const double*)t.dev_ptr[1], a.ld, &one, (double*)t.dev_ptr[2], a.ld);
+  PA_CHECK(s == rocblas_status_success, "rocblas_dgemm failed: %d", (int)s);
+}
+
+// ------------------------------------------------------------------ classes
+static TaskClass make_tc(const char* name, TaskKind kind,
+                         void (*cpu)(Task&), void (*gpu)(Task&, GpuTaskCtx&),
+                         int id) {
+  TaskClass tc;
+  tc.name = name;
+  tc.kind = kind;
+  tc.cpu_hook = cpu;
+  tc.gpu_hook = gpu;
+  tc.id = id;
+  return tc;
+}
+
+TaskClass& tc_spd_fill() {
+  static TaskClass tc = make_tc("spd_fill", TaskKind::GPU, cpu_spd_fill, gpu_spd_fill, 1);
+  return tc;
+}
+TaskClass& tc_potrf() {
+  static TaskClass tc = make_tc("potrf", TaskKind::GPU, cpu_potrf, gpu_potrf, 2);
+  return tc;
+}
+TaskClass& tc_trsm() {
+  static TaskClass tc = make_tc("trsm", TaskKind::GPU, cpu_trsm, gpu_trsm, 3);
+  return tc;
+}
+TaskClass& tc_syrk() {
+  static TaskClass tc = make_tc("syrk", TaskKind::GPU, cpu_syrk, gpu_syrk, 4);
+  return tc;
+}
+TaskClass& tc_gemm() {
+  static TaskClass tc = make_tc("gemm", TaskKind::GPU, cpu_gemm, gpu_gemm, 5);
+  return tc;
+}
+
+// ------------------------------------------------------------------ DAG builders
+void insert_spd_fill(Dtd& tp, TiledMatrix& A, uint32_t seed) {
+  for (int tm = 0; tm < A.mt(); tm++)
+    for (int tn = 0; tn <= tm && tn < A.nt(); tn++) {
+      TileArgs a;
+      a.m = A.tile_rows(tm);
+      a.n = A.tile_cols(tn);
+      a.ld = A.mb();
+      a.i0 = (int64_t)tm * A.mb();
+      a.j0 = (int64_t)tn * A.nb();
+      a.N = A.m();
+      a.seed = seed;
+      Dtd::FlowSpec f[] = {{A.tile(tm, tn), ACCESS_OUT}};
+      tp.insert(&tc_spd_fill(), &a, sizeof(a), f, 1, 0, A.rank_of(tm, tn));
+    }
+}
+
+// Right-looking tiled Cholesky, lower triangular (the reference's headline
+// dpotrf DAG shape; DPLASMA dpotrf_L equivalent).
+void insert_potrf(Dtd& tp, TiledMatrix& A) {
+  const int T = A.mt();
+  const int ld = A.mb();
+  // Priority: panel-critical path first, decaying with k.
+  auto prio = [T](int k, int bonus) { return ((T - k) << 4) + bonus; };
+  for (int k = 0; k < T; k++) {
+    TileArgs pa_args;
+    pa_args.n = A.tile_cols(k);
+    pa_args.ld = ld;
+    {
+      Dtd::FlowSpec f[] = {{A.tile(k, k), ACCESS_INOUT}};
+      tp.insert(&tc_potrf(), &pa_args, sizeof(pa_args), f, 1, prio(k, 9),
+                A.rank_of(k, k));
+    }
+    for (int m = k + 1; m < T; m++) {
+      TileArgs a;
+      a.m = A.tile_rows(m);
+      a.n = A.tile_cols(k);
+      a.ld = ld;
+      Dtd::FlowSpec f[] = {{A.tile(k, k), ACCESS_IN},
+                           {A.tile(m, k), ACCESS_INOUT}};
+      tp.insert(&tc_trsm(), &a, sizeof(a), f, 2, prio(k, 6), A.rank_of(m, k));
+    }
+    for (int n = k + 1; n < T; n++) {
+      {
+        TileArgs a;
+        a.n = A.tile_rows(n);
+        a.k = A.tile_cols(k);
+        a.ld = ld;
+        Dtd::FlowSpec f[] = {{A.tile(n, k), ACCESS_IN},
+                             {A.tile(n, n), ACCESS_INOUT}};
+        tp.insert(&tc_syrk(), &a, sizeof(a), f, 2, prio(k, 3), A.rank_of(n, n));
+      }
+      for (int m = n + 1; m < T; m++) {
+        TileArgs a;
+        a.m = A.tile_rows(m);
+        a.n = A.tile_rows(n);
+        a.k = A.tile_cols(k);
+        a.ld = ld;
+        Dtd::FlowSpec f[] = {{A.tile(m, k), ACCESS_IN},
+                             {A.tile(n, k), ACCESS_IN},
+                             {A.tile(m, n), ACCESS_INOUT}};
+        tp.insert(&tc_gemm(), &a, sizeof(a), f, 3, prio(k, 0), A.rank_of(m, n));
+      }
+    }
+  }
+}
+
+}  // namespace pa

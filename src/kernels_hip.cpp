@@ -1,0 +1,103 @@
+// Hand-written CDNA4 (gfx950) fp64 MFMA tile kernels.
+//
+// v_mfma_f64_16x16x4f64: one wave computes a 16x16 fp64 tile with K-step 4,
+// 2048 FLOP per instruction at ~64 cycles/SIMD (the fp64 matrix rate equals
+// the fp64 vector rate on MI355X, ~78.6 TF/s chip peak). At that cadence a
+// 4-wave 128x128 LDS-staged block is compute-bound: 8 ds_read_b64 feed
+// 16 MFMAs (1024 SIMD-cycles), so staging and even bank conflicts hide
+// entirely under the matrix pipe — the structure below aims for clean
+// global coalescing and 16 independent accumulators per wave, which is what
+// the fp64 pipe actually needs (MI355X_MICROARCH.md: per-instruction
+// constants; fragment maps per cdna4 ISA: A[i=l&15][k=l>>4],
+// B[k=l>>4][j=l&15], C row=(l>>4)*4+e, col=l&15).
+//
+// Kernel set: dgemm_nt (C -= A*B^T — the Cholesky trailing update, also
+// reused by SYRK on the full tile), with bounds-checked edges. rocBLAS
+// remains the alternate chore; select with PARSEC_MCA_chore_gemm=hip|rocblas.
+#include <hip/hip_runtime.h>
+
+#include "device_gpu.hpp"
+#include "kernels.hpp"
+
+namespace pa {
+
+typedef double f64x4 __attribute__((ext_vector_type(4)));
+
+#define BM 128
+#define BN 128
+#define BKD 16
+
+__launch_bounds__(256)
+__global__ void k_dgemm_nt(int m, int n, int k, const double* __restrict__ A,
+                           int lda, const double* __restrict__ B, int ldb,
+                           double* __restrict__ C, int ldc) {
+  // C[m x n] -= A[m x k] * B[n x k]^T, column-major.
+  __shared__ double As[BKD][BM + 1];
+  __shared__ double Bs[BKD][BN + 1];
+  const int wave = threadIdx.x >> 6;
+  const int lane = threadIdx.x & 63;
+  const int wr = wave >> 1, wc = wave & 1;  // 2x2 waves of 64x64 output
+  const int bm0 = blockIdx.x * BM, bn0 = blockIdx.y * BN;
+  const int ksub = lane >> 4;   // k offset within the 4-deep MFMA step
+  const int r16 = lane & 15;
+
+  f64x4 acc[4][4] = {};
+
+  for (int k0 = 0; k0 < k; k0 += BKD) {
+    // Stage A-panel [BM][BKD] and B-panel [BN][BKD] k-major in LDS.
+    // 256 threads x 8 elements; global access is column-contiguous.
+    for (int x = threadIdx.x; x < BM * BKD; x += 256) {
+      int i = x & (BM - 1), kk = x >> 7;
+      int gi = bm0 + i, gk = k0 + kk;
+      As[kk][i] = (gi < m && gk < k) ? A[(size_t)gk * lda + gi] : 0.0;
+    }
+    for (int x = threadIdx.x; x < BN * BKD; x += 256) {
+      int j = x & (BN - 1), kk = x >> 7;
+      int gj = bn0 + j, gk = k0 + kk;
+      Bs[kk][j] = (gj < n && gk < k) ? B[(size_t)gk * ldb + gj] : 0.0;
+    }
+    __syncthreads();
+#pragma unroll
+    for (int kk = 0; kk < BKD; kk += 4) {
+      double a[4], b[4];
+#pragma unroll
+      for (int f = 0; f < 4; f++) a[f] = As[kk + ksub][wr * 64 + f * 16 + r16];
+#pragma unroll
+      for (int f = 0; f < 4; f++) b[f] = Bs[kk + ksub][wc * 64 + f * 16 + r16];
+#pragma unroll
+      for (int i = 0; i < 4; i++)
+#pragma unroll
+        for (int j = 0; j < 4; j++)
+          acc[i][j] =
+              __builtin_amdgcn_mfma_f64_16x16x4f64(a[i], b[j], acc[i][j], 0, 0, 0);
+    }
+    __syncthreads();
+  }
+
+#pragma unroll
+  for (int j = 0; j < 4; j++) {
+    int col = bn0 + wc * 64 + j * 16 + r16;
+    if (col >= n) continue;
+    double* cp = C + (size_t)col * ldc;
+#pragma unroll
+    for (int i = 0; i < 4; i++) {
+      int row0 = bm0 + wr * 64 + i * 16 + ksub * 4;
+#pragma unroll
+      for (int e = 0; e < 4; e++) {
+        int row = row0 + e;
+        if (row < m) cp[row] -= acc[i][j][e];
+      }
+    }
+  }
+}
+
+void gpu_gemm_hip(Task& t, GpuTaskCtx& g) {
+  const TileArgs& a = t.arg<TileArgs>();
+  dim3 grid((a.m + BM - 1) / BM, (a.n + BN - 1) / BN);
+  hipLaunchKernelGGL(k_dgemm_nt, grid, dim3(256), 0, g.stream, a.m, a.n, a.k,
+                     (const double*)t.dev_ptr[0], a.ld,
+                     (const double*)t.dev_ptr[1], a.ld, (double*)t.dev_ptr[2],
+                     a.ld);
+}
+
+}  // namespace pa

@@ -1,0 +1,194 @@
+// Python bindings for parsec_amd (_core extension).
+//
+// The Python layer is the driver surface (bench.py, tests, tools); the
+// runtime, DAG execution, kernels and comm are all native C++/HIP. Python
+// task bodies (insert_py) exist for runtime-semantics tests only, mirroring
+// the reference's tests/dsl/dtd/* programs.
+#include <pybind11/numpy.h>
+#include <pybind11/pybind11.h>
+#include <pybind11/stl.h>
+
+#include "data.hpp"
+#include "device_gpu.hpp"
+#include "dtd.hpp"
+#include "kernels.hpp"
+#include "runtime.hpp"
+
+namespace pa {
+std::string rccl_get_unique_id();
+void rccl_set_unique_id(const std::string&);
+}  // namespace pa
+
+namespace py = pybind11;
+using namespace pa;
+
+namespace {
+
+// ---- Python CPU task bodies (for DTD semantics tests) ----
+struct PyTaskPayload {
+  PyObject* fn;
+};
+
+void py_task_hook(Task& t) {
+  py::gil_scoped_acquire gil;
+  py::handle fn(t.arg<PyTaskPayload>().fn);
+  try {
+    fn();
+  } catch (py::error_already_set& e) {
+    fprintf(stderr, "[parsec_amd] python task raised: %s\n", e.what());
+  }
+}
+
+void py_task_destruct(Task& t) {
+  py::gil_scoped_acquire gil;
+  Py_XDECREF(t.arg<PyTaskPayload>().fn);
+}
+
+TaskClass& py_task_class() {
+  static TaskClass tc = [] {
+    TaskClass c;
+    c.name = "py_task";
+    c.kind = TaskKind::CPU;
+    c.cpu_hook = py_task_hook;
+    c.destruct = py_task_destruct;
+    c.id = 100;
+    return c;
+  }();
+  return tc;
+}
+
+Context::Options options_from_env(int nworkers, int rank, int world,
+                                  std::string comm, int gpu) {
+  Context::Options o;
+  o.nworkers = nworkers;
+  if (rank < 0) {
+    const char* r = getenv("RANK");
+    const char* w = getenv("WORLD_SIZE");
+    o.rank = r ? atoi(r) : 0;
+    o.world = w ? atoi(w) : 1;
+  } else {
+    o.rank = rank;
+    o.world = world;
+  }
+  o.comm = std::move(comm);
+  o.gpu_device = gpu;
+  return o;
+}
+
+}  // namespace
+
+PYBIND11_MODULE(_core, m) {
+  m.doc() = "parsec_amd: MI355X-native task-dataflow runtime (PaRSEC-class)";
+
+  py::class_<Context>(m, "Context")
+      .def(py::init([](int nworkers, int rank, int world, std::string comm,
+                       int gpu) {
+             return new Context(
+                 options_from_env(nworkers, rank, world, std::move(comm), gpu));
+           }),
+           py::arg("nworkers") = -1, py::arg("rank") = -1,
+           py::arg("world") = -1, py::arg("comm") = std::string(),
+           py::arg("gpu") = -1,
+           py::call_guard<py::gil_scoped_release>())
+      .def_property_readonly("rank", &Context::rank)
+      .def_property_readonly("world", &Context::world)
+      .def_property_readonly("nworkers", &Context::nworkers)
+      .def_property_readonly("has_gpu", &Context::has_gpu)
+      .def("barrier", &Context::barrier,
+           py::call_guard<py::gil_scoped_release>())
+      .def("gpu_sync", [](Context& c) {
+        if (c.gpu()) c.gpu()->sync_all();
+      }, py::call_guard<py::gil_scoped_release>())
+      .def("gpu_stats", [](Context& c) {
+        py::dict d;
+        if (c.gpu()) {
+          d["tasks"] = c.gpu()->stats.tasks.load();
+          d["bytes_h2d"] = c.gpu()->stats.bytes_h2d.load();
+          d["bytes_d2h"] = c.gpu()->stats.bytes_d2h.load();
+        }
+        return d;
+      });
+
+  py::class_<Data>(m, "Data")
+      .def_property_readonly("home_rank", [](Data& d) { return d.home_rank; })
+      .def_property_readonly("owner_rank", [](Data& d) { return d.owner_rank; })
+      .def_property_readonly("version", [](Data& d) { return d.version; });
+
+  py::class_<TiledMatrix>(m, "TiledMatrix")
+      .def(py::init<Context*, int64_t, int64_t, int, int, int, int, size_t>(),
+           py::arg("ctx"), py::arg("m"), py::arg("n"), py::arg("mb"),
+           py::arg("nb"), py::arg("p") = 1, py::arg("q") = 1,
+           py::arg("elem_size") = 8, py::keep_alive<1, 2>())
+      .def_property_readonly("mt", &TiledMatrix::mt)
+      .def_property_readonly("nt", &TiledMatrix::nt)
+      .def_property_readonly("mb", &TiledMatrix::mb)
+      .def_property_readonly("nb", &TiledMatrix::nb)
+      .def("rank_of", &TiledMatrix::rank_of)
+      .def("is_local", &TiledMatrix::is_local)
+      .def("tile", &TiledMatrix::tile, py::return_value_policy::reference_internal)
+      .def("tile_rows", &TiledMatrix::tile_rows)
+      .def("tile_cols", &TiledMatrix::tile_cols)
+      .def("tile_numpy", [](TiledMatrix& A, int tm, int tn) {
+        // Readback of a LOCAL tile as a numpy array (column-major view
+        // copied to a C-order array of shape [rows, cols]).
+        PA_CHECK(A.is_local(tm, tn), "tile_numpy: tile is not local");
+        Data* d = A.tile(tm, tn);
+        double* p;
+        {
+          py::gil_scoped_release rel;
+          p = (double*)d->pull_to_host();
+        }
+        int rows = A.tile_rows(tm), cols = A.tile_cols(tn);
+        py::array_t<double> out({rows, cols});
+        auto r = out.mutable_unchecked<2>();
+        for (int j = 0; j < cols; j++)
+          for (int i = 0; i < rows; i++) r(i, j) = p[(size_t)j * A.mb() + i];
+        return out;
+      });
+
+  py::class_<Taskpool>(m, "Taskpool")
+      .def("wait", &Taskpool::wait, py::call_guard<py::gil_scoped_release>())
+      .def_property_readonly("pending", &Taskpool::pending);
+
+  py::class_<Dtd, Taskpool>(m, "Dtd")
+      .def(py::init<Context*, std::string>(), py::arg("ctx"),
+           py::arg("name") = std::string("dtd"), py::keep_alive<1, 2>())
+      .def("insert_py",
+           [](Dtd& tp, py::function fn,
+              std::vector<std::pair<Data*, int>> flows, int priority,
+              int rank) {
+             PyTaskPayload pl{fn.ptr()};
+             Py_XINCREF(pl.fn);
+             std::vector<Dtd::FlowSpec> fs;
+             for (auto& [d, mode] : flows)
+               fs.push_back({d, (AccessMode)mode});
+             py::gil_scoped_release rel;
+             tp.insert(&py_task_class(), &pl, sizeof(pl), fs.data(),
+                       (int)fs.size(), priority, rank);
+           },
+           py::arg("fn"), py::arg("flows") = std::vector<std::pair<Data*, int>>{},
+           py::arg("priority") = 0, py::arg("rank") = -1)
+      .def("flush", &Dtd::flush, py::call_guard<py::gil_scoped_release>())
+      .def("flush_all", &Dtd::flush_all,
+           py::call_guard<py::gil_scoped_release>());
+
+  m.attr("ACCESS_IN") = (int)ACCESS_IN;
+  m.attr("ACCESS_OUT") = (int)ACCESS_OUT;
+  m.attr("ACCESS_INOUT") = (int)ACCESS_INOUT;
+
+  m.def("insert_spd_fill", &insert_spd_fill, py::arg("tp"), py::arg("A"),
+        py::arg("seed") = 42u, py::call_guard<py::gil_scoped_release>());
+  m.def("insert_potrf", &insert_potrf, py::arg("tp"), py::arg("A"),
+        py::call_guard<py::gil_scoped_release>());
+
+  m.def("param_set", &param_set);
+  m.def("param_dump", &param_dump);
+  m.def("hip_device_count", [] {
+    int n = 0;
+    if (hipGetDeviceCount(&n) != hipSuccess) n = 0;
+    return n;
+  });
+  m.def("nccl_unique_id", [] { return py::bytes(pa::rccl_get_unique_id()); });
+  m.def("set_nccl_unique_id",
+        [](py::bytes b) { pa::rccl_set_unique_id(std::string(b)); });
+}

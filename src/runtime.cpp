@@ -1,0 +1,237 @@
+#include "runtime.hpp"
+
+#include "data.hpp"
+#include "device_gpu.hpp"
+#include "comm.hpp"
+
+namespace pa {
+
+thread_local int Context::tls_worker_id = -1;
+
+// ------------------------------------------------------------------ Task
+Task* task_new(Taskpool* tp, const TaskClass* tc) {
+  Task* t = new Task();
+  t->tp = tp;
+  t->tc = tc;
+  t->seq = tp->next_seq();
+  tp->task_created();
+  return t;
+}
+
+void Task::release() {
+  if (refcnt.fetch_sub(1, std::memory_order_acq_rel) == 1) {
+    if (tc->destruct) tc->destruct(*this);
+    delete this;
+  }
+}
+
+bool task_add_edge(Task* pred, Task* succ) {
+  // Dependency-release race protocol (SURVEY.md §7 "hard parts"): the edge
+  // is registered under the predecessor's lock; if the predecessor already
+  // completed, the successor does not wait on it.
+  pred->lock.lock();
+  if (pred->completed) {
+    pred->lock.unlock();
+    return false;
+  }
+  succ->deps_remaining.fetch_add(1, std::memory_order_relaxed);
+  pred->succs.push_back(succ);
+  pred->lock.unlock();
+  return true;
+}
+
+void task_dec_deps(Task* t) {
+  if (t->deps_remaining.fetch_sub(1, std::memory_order_acq_rel) == 1) {
+    t->tp->context()->dispatch(t, Context::tls_worker_id);
+  }
+}
+
+void task_complete(Task* t) {
+  t->lock.lock();
+  t->completed = true;
+  std::vector<Task*> succs;
+  succs.swap(t->succs);
+  t->lock.unlock();
+  for (Task* s : succs) task_dec_deps(s);
+  Taskpool* tp = t->tp;
+  t->release();
+  tp->task_done();
+}
+
+void run_cpu_task(Task* t) {
+  if (t->tc->cpu_hook) t->tc->cpu_hook(*t);
+  task_complete(t);
+}
+
+// ------------------------------------------------------------------ Taskpool
+Taskpool::Taskpool(Context* ctx, std::string name)
+    : ctx_(ctx), name_(std::move(name)) {}
+
+Taskpool::~Taskpool() {
+  PA_CHECK(nb_pending_.load() == 0);
+}
+
+void Taskpool::task_created() {
+  nb_pending_.fetch_add(1, std::memory_order_acq_rel);
+}
+
+void Taskpool::task_done() {
+  if (nb_pending_.fetch_sub(1, std::memory_order_acq_rel) == 1) {
+    std::vector<std::function<void()>> cbs;
+    {
+      std::lock_guard<std::mutex> g(mtx_);
+      cbs.swap(on_complete_);
+    }
+    for (auto& cb : cbs) cb();
+    cv_.notify_all();
+  }
+}
+
+void Taskpool::on_complete(std::function<void()> cb) {
+  bool fire = false;
+  {
+    std::lock_guard<std::mutex> g(mtx_);
+    if (nb_pending_.load(std::memory_order_acquire) == 0) fire = true;
+    else on_complete_.push_back(std::move(cb));
+  }
+  if (fire) cb();
+}
+
+void Taskpool::wait() {
+  // Main thread contributes to CPU progress while waiting
+  // (__parsec_context_wait, scheduling.c:727-863).
+  while (nb_pending_.load(std::memory_order_acquire) != 0) {
+    if (!ctx_->progress_one()) {
+      std::unique_lock<std::mutex> g(mtx_);
+      if (nb_pending_.load(std::memory_order_acquire) == 0) break;
+      cv_.wait_for(g, std::chrono::microseconds(200));
+    }
+  }
+}
+
+// ------------------------------------------------------------------ Scheduler
+Scheduler::Scheduler(int nworkers) : nworkers_(nworkers) {
+  for (int i = 0; i < nworkers_; i++) wq_.emplace_back(new WorkerQ());
+}
+
+Scheduler::~Scheduler() = default;
+
+void Scheduler::push(Task* t, int worker_hint) {
+  if (worker_hint >= 0 && worker_hint < nworkers_ && t->priority == 0) {
+    WorkerQ& q = *wq_[worker_hint];
+    q.lock.lock();
+    q.dq.push_front(t);
+    q.lock.unlock();
+  } else {
+    pq_lock_.lock();
+    pq_.push(PQEntry{t});
+    pq_lock_.unlock();
+  }
+  npending_.fetch_add(1, std::memory_order_release);
+  sleep_cv_.notify_one();
+}
+
+Task* Scheduler::pop(int worker) {
+  Task* t = nullptr;
+  if (worker >= 0) {
+    WorkerQ& q = *wq_[worker];
+    q.lock.lock();
+    if (!q.dq.empty()) { t = q.dq.front(); q.dq.pop_front(); }
+    q.lock.unlock();
+    if (t) { npending_.fetch_sub(1, std::memory_order_relaxed); return t; }
+  }
+  // shared priority queue
+  pq_lock_.lock();
+  if (!pq_.empty()) { t = pq_.top().t; pq_.pop(); }
+  pq_lock_.unlock();
+  if (t) { npending_.fetch_sub(1, std::memory_order_relaxed); return t; }
+  // steal (FIFO end) from other workers, nearest-first
+  for (int d = 1; d < nworkers_; d++) {
+    int v = (worker >= 0 ? (worker + d) % nworkers_ : d - 1);
+    WorkerQ& q = *wq_[v];
+    if (q.dq.empty()) continue;
+    q.lock.lock();
+    if (!q.dq.empty()) { t = q.dq.back(); q.dq.pop_back(); }
+    q.lock.unlock();
+    if (t) { npending_.fetch_sub(1, std::memory_order_relaxed); return t; }
+  }
+  return nullptr;
+}
+
+void Scheduler::wake_all() { sleep_cv_.notify_all(); }
+
+void Scheduler::park(int, const std::atomic<bool>& stop) {
+  std::unique_lock<std::mutex> g(sleep_mtx_);
+  if (stop.load(std::memory_order_acquire)) return;
+  if (npending_.load(std::memory_order_acquire) != 0) return;
+  sleep_cv_.wait_for(g, std::chrono::microseconds(500));
+}
+
+// ------------------------------------------------------------------ Context
+Context::Context(const Options& opt) : rank_(opt.rank), world_(opt.world) {
+  int nw = opt.nworkers;
+  if (nw < 0) nw = (int)param_int("sched_workers", -1);
+  if (nw < 0) {
+    nw = (int)std::thread::hardware_concurrency();
+    // leave room for main + gpu manager + comm threads
+    nw = nw > 3 ? nw - 3 : 1;
+  }
+  nworkers_ = nw;
+  sched_.reset(new Scheduler(nworkers_));
+
+  if (opt.gpu_device != -2) gpu_ = GpuEngine::create(this, opt.gpu_device);
+  comm_ = CommEngine::create(this, opt.comm);
+
+  for (int i = 0; i < nworkers_; i++)
+    workers_.emplace_back([this, i] { worker_main(i); });
+  PA_DEBUG(1, "context up: rank %d/%d, %d workers, gpu=%d", rank_, world_,
+           nworkers_, has_gpu());
+}
+
+Context::~Context() {
+  stop_.store(true, std::memory_order_release);
+  sched_->wake_all();
+  for (auto& w : workers_) w.join();
+  comm_.reset();
+  gpu_.reset();
+}
+
+void Context::worker_main(int id) {
+  tls_worker_id = id;
+  while (!stop_.load(std::memory_order_acquire)) {
+    Task* t = sched_->pop(id);
+    if (t) {
+      run_cpu_task(t);
+    } else {
+      sched_->park(id, stop_);
+    }
+  }
+}
+
+void Context::dispatch(Task* t, int worker_hint) {
+  switch (t->tc->kind) {
+    case TaskKind::GPU:
+      if (gpu_) { gpu_->enqueue(t); return; }
+      [[fallthrough]];
+    case TaskKind::CPU:
+      sched_->push(t, worker_hint);
+      return;
+    case TaskKind::COMM_SEND:
+    case TaskKind::COMM_RECV:
+      comm_->enqueue(t);
+      return;
+  }
+}
+
+bool Context::progress_one() {
+  Task* t = sched_->pop(-1);
+  if (!t) return false;
+  run_cpu_task(t);
+  return true;
+}
+
+void Context::barrier() {
+  if (comm_) comm_->barrier();
+}
+
+}  // namespace pa

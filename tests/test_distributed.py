@@ -1,0 +1,58 @@
+"""Multi-process distributed tests on the CPU TCP comm engine.
+
+Exercises the deterministic SPMD dataflow protocol (sends/recvs, channel
+sequencing, flush_all, barrier) with world_size 2 and 4 on one host —
+the reference tests multi-node the same way (multiple ranks on one host,
+SURVEY.md §4).
+"""
+import os
+import subprocess
+import sys
+
+import numpy as np
+import pytest
+
+HERE = os.path.dirname(os.path.abspath(__file__))
+WORKER = os.path.join(HERE, "_dist_worker.py")
+
+_next_port = [29760]
+
+
+def run_world(world, tmpdir, p, q, n=256, nb=64):
+    port = _next_port[0]
+    _next_port[0] += world + 2
+    procs = []
+    for r in range(world):
+        env = dict(os.environ)
+        env.update(RANK=str(r), WORLD_SIZE=str(world),
+                   PARSEC_TEST_PORT=str(port), PARSEC_TEST_OUT=str(tmpdir),
+                   GRID_P=str(p), GRID_Q=str(q), MAT_N=str(n), MAT_NB=str(nb))
+        procs.append(subprocess.Popen([sys.executable, WORKER], env=env,
+                                      stdout=subprocess.PIPE,
+                                      stderr=subprocess.STDOUT))
+    outs = []
+    for pr in procs:
+        out, _ = pr.communicate(timeout=180)
+        outs.append(out.decode())
+        assert pr.returncode == 0, f"worker failed:\n{out.decode()}"
+    # assemble
+    pre = np.zeros((n, n))
+    post = np.zeros((n, n))
+    for r in range(world):
+        z = np.load(os.path.join(tmpdir, f"rank{r}.npz"))
+        for key in z.files:
+            kind, tm, tn = key.split("_")[0], *key.split("_")[1:]
+            tm, tn = int(tm), int(tn)
+            v = z[key]
+            dst = pre if kind == "pre" else post
+            dst[tm * nb:tm * nb + v.shape[0], tn * nb:tn * nb + v.shape[1]] = v
+    return pre, post
+
+
+@pytest.mark.parametrize("world,p,q", [(2, 2, 1), (2, 1, 2), (4, 2, 2)])
+def test_distributed_cholesky(world, p, q, tmp_path):
+    pre, post = run_world(world, tmp_path, p, q)
+    M = np.tril(pre) + np.tril(pre, -1).T
+    L0 = np.linalg.cholesky(M)
+    err = np.abs(np.tril(post) - L0).max()
+    assert err < 1e-10, f"world={world} p={p} q={q}: max err {err}"

@@ -7,6 +7,7 @@
 #include "kernels.hpp"
 
 #include <cmath>
+#include <map>
 
 #include <hip/hip_runtime.h>
 #include <rocblas/rocblas.h>
@@ -131,19 +132,24 @@ static void cpu_gemm(Task& t) {
 }
 
 // ------------------------------------------------------------------ GPU chores
-// One rocBLAS handle per GPU-manager thread (hooks run on the manager only).
+// One rocBLAS handle PER EXEC STREAM: rocBLAS/rocSOLVER keep device
+// workspace on the handle (e.g. dtrsm's invA), so concurrent kernels on
+// different streams must not share one. Hooks run on the manager thread
+// only, so a thread_local map is race-free.
 static rocblas_handle blas_handle(GpuTaskCtx& g) {
-  static thread_local rocblas_handle h = nullptr;
+  static thread_local std::map<void*, rocblas_handle> handles;
+  rocblas_handle& h = handles[(void*)g.stream];
   if (!h) {
     PA_CHECK(rocblas_create_handle(&h) == rocblas_status_success);
     rocblas_set_pointer_mode(h, rocblas_pointer_mode_host);
+    rocblas_set_stream(h, g.stream);
   }
-  rocblas_set_stream(h, g.stream);
   return h;
 }
 
-static rocblas_int* dev_info() {
-  static thread_local rocblas_int* p = nullptr;
+static rocblas_int* dev_info(GpuTaskCtx& g) {
+  static thread_local std::map<void*, rocblas_int*> infos;
+  rocblas_int*& p = infos[(void*)g.stream];
   if (!p) PA_HIP_CHECK(hipMalloc(&p, sizeof(rocblas_int)));
   return p;
 }
@@ -151,7 +157,7 @@ static rocblas_int* dev_info() {
 static void gpu_potrf(Task& t, GpuTaskCtx& g) {
   const TileArgs& a = t.arg<TileArgs>();
   rocblas_status s = rocsolver_dpotrf(blas_handle(g), rocblas_fill_lower, a.n,
-                                      (double*)t.dev_ptr[0], a.ld, dev_info());
+                                      (double*)t.dev_ptr[0], a.ld, dev_info(g));
   PA_CHECK(s == rocblas_status_success, "rocsolver_dpotrf failed: %d", (int)s);
 }
 

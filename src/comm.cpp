@@ -285,6 +285,16 @@ class TcpComm : public CommEngine {
 
   void peer_down(int peer, ssize_t r) {
     if (stop_.load()) return;
+    Peer& p = peers_[peer];
+    bool pending = !p.out.empty() || p.in_task;
+    for (auto& kv : posted_recv_)
+      if ((int)(kv.first >> 48) == peer) pending = true;
+    if (r == 0 && !pending) {
+      // graceful EOF: the peer finished and tore down its context first
+      close(p.fd);
+      p.fd = -1;
+      return;
+    }
     fatal("comm: connection to rank %d lost (r=%zd errno=%d)", peer, r, errno);
   }
 

@@ -31,6 +31,7 @@ GpuEngine::GpuEngine(Context* ctx, int device) : ctx_(ctx), device_(device) {
   for (int i = 0; i < nstreams; i++)
     PA_HIP_CHECK(hipStreamCreateWithFlags(&exec_streams_[i], hipStreamNonBlocking));
   max_inflight_per_stream_ = (size_t)param_int("gpu_max_inflight", 32);
+  next_stream_ = nstreams > 1 ? 1 : 0;
 
   // Reserve the HBM slab (parsec_device_memory_reserve analog,
   // device_gpu.c:867-992). Default 85% of free memory; kernels and
@@ -138,7 +139,19 @@ void GpuEngine::sync_all() {
 // (parsec_device_data_stage_in, device_gpu.c:1800-2168, minus the peer-GPU
 // branch: peers are other processes here, reached through the comm engine).
 void GpuEngine::stage_in_and_launch(Task* t) {
-  hipStream_t es = exec_streams_[next_stream_];
+  // Stream 0 is reserved for critical-path (panel) tasks so they never
+  // queue behind bulk updates; others round-robin over the remaining
+  // streams (the reference's exec_stream[2..n] round-robin,
+  // device_gpu.c:3445-3535, with an express lane added).
+  int si;
+  if (t->priority >= (1 << 19) || (int)exec_streams_.size() == 1) {
+    si = 0;
+  } else {
+    si = next_stream_;
+    next_stream_ = next_stream_ + 1;
+    if (next_stream_ >= (int)exec_streams_.size()) next_stream_ = 1;
+  }
+  hipStream_t es = exec_streams_[si];
   bool copied = false;
   for (int i = 0; i < t->nflows; i++) {
     Data* d = t->flows[i].data;
@@ -168,9 +181,8 @@ void GpuEngine::stage_in_and_launch(Task* t) {
   t->tc->gpu_hook(*t, gctx);
   hipEvent_t ev = event_get();
   PA_HIP_CHECK(hipEventRecord(ev, es));
-  inflight_[next_stream_].push_back(InFlight{t, ev, next_stream_});
+  inflight_[si].push_back(InFlight{t, ev, si});
   n_inflight_++;
-  next_stream_ = (next_stream_ + 1) % (int)exec_streams_.size();
 }
 
 void GpuEngine::manager_main() {
@@ -197,9 +209,10 @@ void GpuEngine::manager_main() {
         progress = true;
       }
     }
-    // 2) launch new work while there is room
+    // 2) launch new work while there is room (global cap; the panel
+    //    stream may exceed its share — its latency is the priority)
     while (true) {
-      if (inflight_[next_stream_].size() >= max_inflight_per_stream_) break;
+      if (n_inflight_ >= max_inflight_per_stream_ * exec_streams_.size()) break;
       Task* t = nullptr;
       {
         std::lock_guard<std::mutex> g(q_mtx_);

@@ -17,6 +17,9 @@
 namespace pa {
 std::string rccl_get_unique_id();
 void rccl_set_unique_id(const std::string&);
+void test_dgemm_nt_hip(int, int, int, const double*, int, const double*, int,
+                       double*, int);
+void test_potf2_hip(double*, int);
 }  // namespace pa
 
 namespace py = pybind11;
@@ -187,6 +190,16 @@ PYBIND11_MODULE(_core, m) {
     int n = 0;
     if (hipGetDeviceCount(&n) != hipSuccess) n = 0;
     return n;
+  });
+  // Standalone kernel harnesses for GPU numerics tests/debugging.
+  m.def("dgemm_nt_hip", [](py::array_t<double> A, py::array_t<double> B,
+                           py::array_t<double> C, int m, int n, int k) {
+    // arrays are column-major buffers flattened 1-D, ld == rows
+    pa::test_dgemm_nt_hip(m, n, k, A.data(), m, B.data(), n,
+                          C.mutable_data(), m);
+  });
+  m.def("potf2_hip", [](py::array_t<double> A, int n) {
+    pa::test_potf2_hip(A.mutable_data(), n);
   });
   m.def("nccl_unique_id", [] { return py::bytes(pa::rccl_get_unique_id()); });
   m.def("set_nccl_unique_id",

@@ -306,7 +306,10 @@ void insert_potrf(Dtd& tp, TiledMatrix& A) {
       a.ld = ld;
       Dtd::FlowSpec f[] = {{A.tile(k, k), ACCESS_IN},
                            {A.tile(m, k), ACCESS_INOUT}};
-      tp.insert(&tc_trsm(), &a, sizeof(a), f, 2, PANEL - (m - k),
+      // Below the express-stream threshold (1<<19): TRSMs are urgent in the
+      // queue but must spread across bulk streams, not serialize on the
+      // panel stream.
+      tp.insert(&tc_trsm(), &a, sizeof(a), f, 2, (1 << 18) - (m - k),
                 A.rank_of(m, k));
     }
     for (int n = k + 1; n < T; n++) {

@@ -74,6 +74,9 @@ __global__ void k_dgemm_nt(int m, int n, int k, const double* __restrict__ A,
     __syncthreads();
   }
 
+  // f64 16x16x4 C/D lane map (verified on gfx950 hardware): element e of
+  // lane l holds D[4*e + (l>>4)][l&15] — register index strides ROWS by 4,
+  // unlike the f32/bf16 16x16 map where (l>>4) selects the 4-row group.
 #pragma unroll
   for (int j = 0; j < 4; j++) {
     int col = bn0 + wc * 64 + j * 16 + r16;
@@ -81,10 +84,10 @@ __global__ void k_dgemm_nt(int m, int n, int k, const double* __restrict__ A,
     double* cp = C + (size_t)col * ldc;
 #pragma unroll
     for (int i = 0; i < 4; i++) {
-      int row0 = bm0 + wr * 64 + i * 16 + ksub * 4;
+      int row0 = bm0 + wr * 64 + i * 16 + ksub;
 #pragma unroll
       for (int e = 0; e < 4; e++) {
-        int row = row0 + e;
+        int row = row0 + e * 4;
         if (row < m) cp[row] -= acc[i][j][e];
       }
     }

@@ -39,3 +39,9 @@ for shape in [(16,16,4), (16,16,16), (32,32,16), (128,128,16), (128,128,128), (2
     chk_gemm(*shape)
 for n in [16, 64, 100, 128]:
     chk_potf2(n)
+
+print("\n--- dgemm kernel throughput (TFLOP/s, device-resident) ---")
+for (m,n,k) in [(2048,2048,2048), (4096,4096,4096), (8192,8192,2048)]:
+    for impl in ["v2", "v1", "rocblas"]:
+        tf = _core.bench_dgemm(m, n, k, 10, impl)
+        print(f"{m}x{n}x{k} {impl}: {tf:.1f} TF/s")

@@ -228,6 +228,38 @@ static void gpu_gemm(Task& t, GpuTaskCtx& g) {
   PA_CHECK(s == rocblas_status_success, "rocblas_dgemm failed: %d", (int)s);
 }
 
+// Timed rocBLAS dgemm NT loop for kernel-level A/B against the hand kernel.
+double bench_dgemm_rocblas(int m, int n, int k, int iters) {
+  double *dA, *dB, *dC;
+  PA_HIP_CHECK(hipMalloc(&dA, (size_t)m * k * 8));
+  PA_HIP_CHECK(hipMalloc(&dB, (size_t)n * k * 8));
+  PA_HIP_CHECK(hipMalloc(&dC, (size_t)m * n * 8));
+  hipLaunchKernelGGL(k_spd_fill, dim3(2048), dim3(256), 0, 0, dA, m, k, m, 0,
+                     0, 0, 11u);
+  hipLaunchKernelGGL(k_spd_fill, dim3(2048), dim3(256), 0, 0, dB, n, k, n, 7,
+                     3, 0, 12u);
+  hipLaunchKernelGGL(k_spd_fill, dim3(2048), dim3(256), 0, 0, dC, m, n, m, 1,
+                     9, 0, 13u);
+  rocblas_handle h;
+  PA_CHECK(rocblas_create_handle(&h) == rocblas_status_success);
+  const double mone = -1.0, one = 1.0;
+  auto run = [&] {
+    rocblas_dgemm(h, rocblas_operation_none, rocblas_operation_transpose, m,
+                  n, k, &mone, dA, m, dB, n, &one, dC, m);
+  };
+  run();
+  PA_HIP_CHECK(hipDeviceSynchronize());
+  double t0 = now_s();
+  for (int i = 0; i < iters; i++) run();
+  PA_HIP_CHECK(hipDeviceSynchronize());
+  double dt = now_s() - t0;
+  rocblas_destroy_handle(h);
+  PA_HIP_CHECK(hipFree(dA));
+  PA_HIP_CHECK(hipFree(dB));
+  PA_HIP_CHECK(hipFree(dC));
+  return dt;
+}
+
 // ------------------------------------------------------------------ classes
 static TaskClass make_tc(const char* name, TaskKind kind,
                          void (*cpu)(Task&), void (*gpu)(Task&, GpuTaskCtx&),

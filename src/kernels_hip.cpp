@@ -94,13 +94,122 @@ __global__ void k_dgemm_nt(int m, int n, int k, const double* __restrict__ A,
   }
 }
 
+// ------------------------------------------------------------------ v2
+// 8-wave 128x128 double-buffered variant tuned for occupancy: acc is 2x4
+// fragments per wave (64 VGPR), LDS 66 KB -> 2 blocks/CU co-resident, which
+// hides each block's staging under the other's MFMA phase (fp64 MFMA issue
+// is 64 cyc/SIMD, so 4 co-resident waves/SIMD keep the matrix pipe fed
+// without hand pipelining). 1-D grid with a bijective XCD-aware remap so
+// consecutive blocks share B panels within one XCD's L2 (guide T1).
+__device__ __forceinline__ int xcd_swizzle(int id, int nwg) {
+  int q = nwg >> 3, r = nwg & 7;
+  int xcd = id & 7, pos = id >> 3;
+  // blocks per XCD: first r XCDs get q+1
+  return (xcd < r ? xcd * (q + 1) : r * (q + 1) + (xcd - r) * q) + pos;
+}
+
+__launch_bounds__(512)
+__global__ void k_dgemm_nt_v2(int m, int n, int k, const double* __restrict__ A,
+                              int lda, const double* __restrict__ B, int ldb,
+                              double* __restrict__ C, int ldc, int nbx) {
+  extern __shared__ __attribute__((aligned(16))) char smem[];
+  constexpr int LDT = 129;
+  double* As = (double*)smem;                  // [2][BKD][LDT]
+  double* Bs = As + 2 * BKD * LDT;
+  const int tid = threadIdx.x;
+  const int wave = tid >> 6;
+  const int lane = tid & 63;
+  const int wr = wave >> 1, wc = wave & 1;  // 4(M) x 2(N) waves: 32x64 each
+  const int nwg = gridDim.x;
+  int id = xcd_swizzle(blockIdx.x, nwg);
+  const int bm0 = (id % nbx) * BM, bn0 = (id / nbx) * BN;
+  const int ksub = lane >> 4;
+  const int r16 = lane & 15;
+
+  f64x4 acc[2][4] = {};
+
+  auto stage = [&](int buf, int k0) {
+    double* as = As + buf * BKD * LDT;
+    double* bs = Bs + buf * BKD * LDT;
+#pragma unroll
+    for (int x = 0; x < BM * BKD; x += 512) {
+      int xi = x + tid;
+      int i = xi & (BM - 1), kk = xi >> 7;
+      int gi = bm0 + i, gk = k0 + kk;
+      as[kk * LDT + i] = (gi < m && gk < k) ? A[(size_t)gk * lda + gi] : 0.0;
+    }
+#pragma unroll
+    for (int x = 0; x < BN * BKD; x += 512) {
+      int xi = x + tid;
+      int j = xi & (BN - 1), kk = xi >> 7;
+      int gj = bn0 + j, gk = k0 + kk;
+      bs[kk * LDT + j] = (gj < n && gk < k) ? B[(size_t)gk * ldb + gj] : 0.0;
+    }
+  };
+
+  stage(0, 0);
+  __syncthreads();
+  const int ntiles = (k + BKD - 1) / BKD;
+  for (int t = 0; t < ntiles; t++) {
+    if (t + 1 < ntiles) stage((t + 1) & 1, (t + 1) * BKD);
+    const double* as = As + (t & 1) * BKD * LDT;
+    const double* bs = Bs + (t & 1) * BKD * LDT;
+#pragma unroll
+    for (int kk = 0; kk < BKD; kk += 4) {
+      double a[2], b[4];
+#pragma unroll
+      for (int f = 0; f < 2; f++)
+        a[f] = as[(kk + ksub) * LDT + wr * 32 + f * 16 + r16];
+#pragma unroll
+      for (int f = 0; f < 4; f++)
+        b[f] = bs[(kk + ksub) * LDT + wc * 64 + f * 16 + r16];
+#pragma unroll
+      for (int i = 0; i < 2; i++)
+#pragma unroll
+        for (int j = 0; j < 4; j++)
+          acc[i][j] =
+              __builtin_amdgcn_mfma_f64_16x16x4f64(a[i], b[j], acc[i][j], 0, 0, 0);
+    }
+    __syncthreads();
+  }
+
+#pragma unroll
+  for (int j = 0; j < 4; j++) {
+    int col = bn0 + wc * 64 + j * 16 + r16;
+    if (col >= n) continue;
+    double* cp = C + (size_t)col * ldc;
+#pragma unroll
+    for (int i = 0; i < 2; i++) {
+      int row0 = bm0 + wr * 32 + i * 16 + ksub;
+#pragma unroll
+      for (int e = 0; e < 4; e++) {
+        int row = row0 + e * 4;
+        if (row < m) cp[row] -= acc[i][j][e];
+      }
+    }
+  }
+}
+
+static void launch_dgemm_v2(int m, int n, int k, const double* A, int lda,
+                            const double* B, int ldb, double* C, int ldc,
+                            hipStream_t stream) {
+  static bool attr_set = false;
+  constexpr size_t lds = 4 * BKD * 129 * 8;
+  if (!attr_set) {
+    hipFuncSetAttribute((const void*)k_dgemm_nt_v2,
+                        hipFuncAttributeMaxDynamicSharedMemorySize, lds);
+    attr_set = true;
+  }
+  int nbx = (m + BM - 1) / BM, nby = (n + BN - 1) / BN;
+  hipLaunchKernelGGL(k_dgemm_nt_v2, dim3(nbx * nby), dim3(512), lds, stream,
+                     m, n, k, A, lda, B, ldb, C, ldc, nbx);
+}
+
 void gpu_gemm_hip(Task& t, GpuTaskCtx& g) {
   const TileArgs& a = t.arg<TileArgs>();
-  dim3 grid((a.m + BM - 1) / BM, (a.n + BN - 1) / BN);
-  hipLaunchKernelGGL(k_dgemm_nt, grid, dim3(256), 0, g.stream, a.m, a.n, a.k,
-                     (const double*)t.dev_ptr[0], a.ld,
-                     (const double*)t.dev_ptr[1], a.ld, (double*)t.dev_ptr[2],
-                     a.ld);
+  launch_dgemm_v2(a.m, a.n, a.k, (const double*)t.dev_ptr[0], a.ld,
+                  (const double*)t.dev_ptr[1], a.ld, (double*)t.dev_ptr[2],
+                  a.ld, g.stream);
 }
 
 // ------------------------------------------------------------------ potf2
@@ -169,14 +278,58 @@ void test_dgemm_nt_hip(int m, int n, int k, const double* A, int lda,
   PA_HIP_CHECK(hipMemcpy(dA, A, (size_t)lda * k * 8, hipMemcpyHostToDevice));
   PA_HIP_CHECK(hipMemcpy(dB, B, (size_t)ldb * k * 8, hipMemcpyHostToDevice));
   PA_HIP_CHECK(hipMemcpy(dC, C, (size_t)ldc * n * 8, hipMemcpyHostToDevice));
-  dim3 grid((m + BM - 1) / BM, (n + BN - 1) / BN);
-  hipLaunchKernelGGL(k_dgemm_nt, grid, dim3(256), 0, 0, m, n, k, dA, lda, dB,
-                     ldb, dC, ldc);
+  launch_dgemm_v2(m, n, k, dA, lda, dB, ldb, dC, ldc, 0);
   PA_HIP_CHECK(hipGetLastError());
   PA_HIP_CHECK(hipMemcpy(C, dC, (size_t)ldc * n * 8, hipMemcpyDeviceToHost));
   PA_HIP_CHECK(hipFree(dA));
   PA_HIP_CHECK(hipFree(dB));
   PA_HIP_CHECK(hipFree(dC));
+}
+
+__global__ void k_bench_fill(double* p, size_t n, uint32_t seed) {
+  size_t i = (size_t)blockIdx.x * blockDim.x + threadIdx.x;
+  for (; i < n; i += (size_t)gridDim.x * blockDim.x) {
+    uint64_t h = (i * 2654435761ull) ^ ((uint64_t)seed * 2246822519ull);
+    h ^= h >> 13;
+    h *= 0x9E3779B97F4A7C15ull;
+    h ^= h >> 32;
+    p[i] = (double)(h & 0xFFFFFF) / (double)0x1000000 - 0.5;
+  }
+}
+
+// Timed device-resident GEMM loop: impl 0 = hand v2, 1 = hand v1.
+// rocBLAS comparison lives in kernels_blas.cpp. Returns seconds.
+double bench_dgemm_hip(int m, int n, int k, int iters, int impl) {
+  double *dA, *dB, *dC;
+  PA_HIP_CHECK(hipMalloc(&dA, (size_t)m * k * 8));
+  PA_HIP_CHECK(hipMalloc(&dB, (size_t)n * k * 8));
+  PA_HIP_CHECK(hipMalloc(&dC, (size_t)m * n * 8));
+  // random-ish fill (never bench on zero-filled operands — DVFS inflates)
+  auto fill = [](double* p, size_t nelem, uint32_t seed) {
+    hipLaunchKernelGGL(k_bench_fill, dim3(2048), dim3(256), 0, 0, p, nelem,
+                       seed);
+  };
+  fill(dA, (size_t)m * k, 11);
+  fill(dB, (size_t)n * k, 12);
+  fill(dC, (size_t)m * n, 13);
+  auto run = [&] {
+    if (impl == 0) launch_dgemm_v2(m, n, k, dA, m, dB, n, dC, m, 0);
+    else {
+      dim3 grid((m + BM - 1) / BM, (n + BN - 1) / BN);
+      hipLaunchKernelGGL(k_dgemm_nt, grid, dim3(256), 0, 0, m, n, k, dA, m,
+                         dB, n, dC, m);
+    }
+  };
+  run();
+  PA_HIP_CHECK(hipDeviceSynchronize());
+  double t0 = now_s();
+  for (int i = 0; i < iters; i++) run();
+  PA_HIP_CHECK(hipDeviceSynchronize());
+  double dt = now_s() - t0;
+  PA_HIP_CHECK(hipFree(dA));
+  PA_HIP_CHECK(hipFree(dB));
+  PA_HIP_CHECK(hipFree(dC));
+  return dt;
 }
 
 void test_potf2_hip(double* A, int n) {

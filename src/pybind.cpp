@@ -20,6 +20,8 @@ void rccl_set_unique_id(const std::string&);
 void test_dgemm_nt_hip(int, int, int, const double*, int, const double*, int,
                        double*, int);
 void test_potf2_hip(double*, int);
+double bench_dgemm_hip(int, int, int, int, int);
+double bench_dgemm_rocblas(int, int, int, int);
 }  // namespace pa
 
 namespace py = pybind11;
@@ -201,6 +203,12 @@ PYBIND11_MODULE(_core, m) {
   m.def("potf2_hip", [](py::array_t<double> A, int n) {
     pa::test_potf2_hip(A.mutable_data(), n);
   });
+  m.def("bench_dgemm", [](int m, int n, int k, int iters, std::string impl) {
+    double dt = impl == "rocblas" ? pa::bench_dgemm_rocblas(m, n, k, iters)
+                                  : pa::bench_dgemm_hip(m, n, k, iters,
+                                                        impl == "v1" ? 1 : 0);
+    return 2.0 * m * n * k * iters / dt / 1e12;  // TFLOP/s
+  }, py::call_guard<py::gil_scoped_release>());
   m.def("nccl_unique_id", [] { return py::bytes(pa::rccl_get_unique_id()); });
   m.def("set_nccl_unique_id",
         [](py::bytes b) { pa::rccl_set_unique_id(std::string(b)); });

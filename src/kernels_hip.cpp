@@ -212,38 +212,76 @@ void gpu_gemm_hip(Task& t, GpuTaskCtx& g) {
                   a.ld, g.stream);
 }
 
-// ------------------------------------------------------------------ potf2
-// Unblocked Cholesky of a <=128x128 fp64 panel, entirely in LDS (one
-// workgroup; 128*129*8 = 132 KB dynamic LDS of the 160 KB/CU). Used as the
-// panel kernel of the blocked tile-POTRF chore — replaces rocSOLVER's
-// potf2_kernel_small (~233 us per 128-panel measured) on the critical path.
-__global__ void __launch_bounds__(256) k_potf2_lds(double* A, int n, int ld) {
+// ---------------------------------------------------------------- potf2 v2
+// Unblocked-in-LDS Cholesky of a <=128 fp64 panel with MFMA trailing
+// updates: 16-column micro-panels factor scalar-sequentially (cheap), the
+// rank-16 trailing update runs on v_mfma_f64_16x16x4f64 over 16x16 LDS
+// tiles. One workgroup (4 waves); replaces both rocSOLVER's potf2 (~233 us)
+// and the scalar LDS version (~830 us) — the tile-POTRF panel is the
+// critical path of the whole Cholesky DAG.
+__global__ void __launch_bounds__(256) k_potf2_mfma(double* A, int n, int ld) {
   extern __shared__ __attribute__((aligned(16))) char smem[];
-  double* S = (double*)smem;  // [n][129] column-major panels: S[j*129 + i]
-  const int tid = threadIdx.x;
   constexpr int LDP = 129;
-  double* dinv = S + n * LDP;  // one extra slot in the dynamic region
+  double* S = (double*)smem;     // [<=128][LDP] column-major
+  double* dinv = S + 128 * LDP;
+  const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int wave = tid >> 6;
   for (int x = tid; x < n * n; x += 256) {
     int j = x / n, i = x - j * n;
     if (i >= j) S[j * LDP + i] = A[(size_t)j * ld + i];
   }
   __syncthreads();
-  for (int j = 0; j < n; j++) {
-    if (tid == 0) {
-      double d = S[j * LDP + j];
-      S[j * LDP + j] = d = sqrt(d);
-      *dinv = 1.0 / d;
+  for (int jb = 0; jb < n; jb += 16) {
+    const int jbe = jb + 16 < n ? jb + 16 : n;
+    // ---- factor the 16-column micro-panel (rows jb..n) ----
+    for (int j = jb; j < jbe; j++) {
+      if (tid == 0) {
+        double d = S[j * LDP + j];
+        S[j * LDP + j] = d = sqrt(d);
+        *dinv = 1.0 / d;
+      }
+      __syncthreads();
+      for (int i = j + 1 + tid; i < n; i += 256) S[j * LDP + i] *= *dinv;
+      __syncthreads();
+      for (int c = j + 1; c < jbe; c++) {
+        double ljc = S[j * LDP + c];
+        for (int i = c + tid; i < n; i += 256)
+          S[c * LDP + i] -= S[j * LDP + i] * ljc;
+      }
+      __syncthreads();
     }
-    __syncthreads();
-    for (int i = j + 1 + tid; i < n; i += 256) S[j * LDP + i] *= *dinv;
-    __syncthreads();
-    // rank-1 update of the trailing lower triangle, flattened over threads
-    const int rem = n - j - 1;
-    for (int x = tid; x < rem * rem; x += 256) {
-      int c = x / rem, i = x - c * rem;
-      if (i >= c) {
-        int gc = j + 1 + c, gi = j + 1 + i;
-        S[gc * LDP + gi] -= S[j * LDP + gi] * S[j * LDP + gc];
+    // ---- rank-16 MFMA trailing update: S[t,t] -= P P^T, t >= jb+16 ----
+    const int t0 = jbe;
+    const int nt = (n - t0 + 15) >> 4;  // trailing 16-blocks
+    if (nt > 0) {
+      const int ntiles = nt * (nt + 1) / 2;
+      const int r16 = lane & 15, g4 = lane >> 4;
+      for (int idx = wave; idx < ntiles; idx += 4) {
+        // idx -> (ti, tj) over the lower-triangular block set
+        int ti = 0, rem = idx;
+        while (rem > ti) { ti++; rem -= ti; }
+        int tj = rem;
+        const int ro = t0 + ti * 16, co = t0 + tj * 16;
+        const int arow = ro + r16, brow = co + r16;
+        f64x4 acc;
+#pragma unroll
+        for (int e = 0; e < 4; e++) {
+          int r = ro + 4 * e + g4, c = co + r16;
+          acc[e] = (r < n && c < n) ? S[c * LDP + r] : 0.0;
+        }
+#pragma unroll
+        for (int kk = 0; kk < 16; kk += 4) {
+          int kcol = jb + kk + g4;
+          double a = (arow < n) ? -S[kcol * LDP + arow] : 0.0;
+          double b = (brow < n) ? S[kcol * LDP + brow] : 0.0;
+          acc = __builtin_amdgcn_mfma_f64_16x16x4f64(a, b, acc, 0, 0, 0);
+        }
+#pragma unroll
+        for (int e = 0; e < 4; e++) {
+          int r = ro + 4 * e + g4, c = co + r16;
+          if (r < n && c < n && r >= c) S[c * LDP + r] = acc[e];
+        }
       }
     }
     __syncthreads();
@@ -256,15 +294,14 @@ __global__ void __launch_bounds__(256) k_potf2_lds(double* A, int n, int ld) {
 
 void launch_potf2(double* A, int n, int ld, hipStream_t stream) {
   PA_CHECK(n <= 128);
+  constexpr size_t lds = 128 * 129 * 8 + 16;
   static bool attr_set = false;
   if (!attr_set) {
-    hipFuncSetAttribute((const void*)k_potf2_lds,
-                        hipFuncAttributeMaxDynamicSharedMemorySize,
-                        128 * 129 * 8 + 16);
+    hipFuncSetAttribute((const void*)k_potf2_mfma,
+                        hipFuncAttributeMaxDynamicSharedMemorySize, lds);
     attr_set = true;
   }
-  hipLaunchKernelGGL(k_potf2_lds, dim3(1), dim3(256), n * 129 * 8 + 16,
-                     stream, A, n, ld);
+  hipLaunchKernelGGL(k_potf2_mfma, dim3(1), dim3(256), lds, stream, A, n, ld);
 }
 
 // ------------------------------------------------------------ test harness

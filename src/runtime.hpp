@@ -144,6 +144,10 @@ class Taskpool {
   // completion callback chaining (parsec_compose, compound.c:17-135)
   void on_complete(std::function<void()> cb);
 
+  // Keep an auxiliary resource (e.g. a workspace collection) alive until
+  // the taskpool is destroyed.
+  void own(std::shared_ptr<void> p) { owned_.push_back(std::move(p)); }
+
  private:
   Context* ctx_;
   std::string name_;
@@ -152,6 +156,7 @@ class Taskpool {
   std::condition_variable cv_;
   uint64_t seq_ = 0;
   std::vector<std::function<void()>> on_complete_;
+  std::vector<std::shared_ptr<void>> owned_;
 };
 
 // ------------------------------------------------------------------ scheduler

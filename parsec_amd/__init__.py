@@ -49,6 +49,8 @@ def init_distributed(nworkers=-1, comm=None, gpu=-1):
     rank = _env_int("RANK", 0)
     world = _env_int("WORLD_SIZE", 1)
     if comm is None:
+        comm = os.environ.get("PARSEC_COMM_KIND")
+    if comm is None:
         comm = "rccl" if (world > 1 and hip_device_count() > 0) else ""
     if world > 1 and comm == "rccl":
         import torch.distributed as dist

@@ -20,8 +20,10 @@ def main():
     p, q = int(os.environ.get("GRID_P", world)), int(os.environ.get("GRID_Q", 1))
     n, nb = int(os.environ.get("MAT_N", 256)), int(os.environ.get("MAT_NB", 64))
 
+    use_gpu = os.environ.get("PARSEC_TEST_GPU") == "1"
     pm.param_set("comm_base_port", str(port))
-    ctx = pm.Context(nworkers=2, rank=rank, world=world, comm="tcp", gpu=-2)
+    ctx = pm.Context(nworkers=2, rank=rank, world=world, comm="tcp",
+                     gpu=(0 if use_gpu else -2))
 
     A = pm.TiledMatrix(ctx, n, n, nb, nb, p, q)
     tp = pm.Dtd(ctx)

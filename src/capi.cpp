@@ -1,0 +1,87 @@
+// C ABI for PTG-generated code (and any external front-end).
+//
+// The reference's generated JDF code links against the public parsec C API
+// (runtime.h:156-710); here the generated C++ from parsec_amd.ptg uses this
+// minimal exported surface (the rest of _core.so is visibility-hidden).
+#include <hip/hip_runtime.h>
+
+#include "data.hpp"
+#include "device_gpu.hpp"
+#include "dtd.hpp"
+#include "runtime.hpp"
+
+#define PA_EXPORT extern "C" __attribute__((visibility("default")))
+
+using namespace pa;
+
+namespace {
+void cpu_trampoline(Task& t) {
+  auto fn = (void (*)(void*))t.tc->user_cpu;
+  fn(&t);
+  // CPU bodies write host memory: mark OUT flows.
+  for (int i = 0; i < t.nflows; i++)
+    if (t.flows[i].data && (t.flows[i].mode & ACCESS_OUT))
+      t.flows[i].data->written_on(false);
+}
+
+void gpu_trampoline(Task& t, GpuTaskCtx& g) {
+  auto fn = (void (*)(void*, void*))t.tc->user_gpu;
+  fn(&t, (void*)g.stream);
+}
+}  // namespace
+
+PA_EXPORT int pa_ctx_rank(void* ctx) { return ((Context*)ctx)->rank(); }
+PA_EXPORT int pa_ctx_world(void* ctx) { return ((Context*)ctx)->world(); }
+PA_EXPORT int pa_ctx_has_gpu(void* ctx) { return ((Context*)ctx)->has_gpu(); }
+
+PA_EXPORT void* pa_tm_tile(void* tm, int i, int j) {
+  return ((TiledMatrix*)tm)->tile(i, j);
+}
+PA_EXPORT int pa_tm_rank_of(void* tm, int i, int j) {
+  return ((TiledMatrix*)tm)->rank_of(i, j);
+}
+PA_EXPORT int pa_data_home_rank(void* d) { return ((Data*)d)->home_rank; }
+
+PA_EXPORT void* pa_taskclass_new(const char* name, int want_gpu,
+                                 void (*cpu)(void*),
+                                 void (*gpu)(void*, void*)) {
+  TaskClass* tc = new TaskClass();
+  tc->name = name;
+  tc->kind = (want_gpu && gpu) ? TaskKind::GPU : TaskKind::CPU;
+  tc->user_cpu = (void*)cpu;
+  tc->user_gpu = (void*)gpu;
+  if (cpu) tc->cpu_hook = cpu_trampoline;
+  if (gpu) tc->gpu_hook = gpu_trampoline;
+  return tc;
+}
+
+PA_EXPORT void* pa_task_args(void* t) { return ((Task*)t)->args; }
+PA_EXPORT void* pa_task_dev_ptr(void* t, int flow) {
+  return ((Task*)t)->dev_ptr[flow];
+}
+PA_EXPORT void* pa_task_host_ptr(void* t, int flow) {
+  Task* task = (Task*)t;
+  Data* d = task->flows[flow].data;
+  if (!d) return nullptr;
+  // WRITE-only flows produce the content: hand out the buffer untouched.
+  if (!(task->flows[flow].mode & ACCESS_IN)) return d->ensure_host();
+  return d->pull_to_host();
+}
+
+PA_EXPORT void* pa_dtd_insert_begin(void* dtd, void* tc, const void* args,
+                                    int nargs, void** datas, const int* modes,
+                                    int nflows, int prio, int rank) {
+  Dtd::FlowSpec fs[MAX_FLOWS];
+  for (int i = 0; i < nflows; i++)
+    fs[i] = {(Data*)datas[i], (AccessMode)modes[i]};
+  return ((Dtd*)dtd)->insert_begin((const TaskClass*)tc, args, (size_t)nargs,
+                                   fs, nflows, prio, rank);
+}
+PA_EXPORT void pa_dtd_insert_commit(void* dtd, void* task) {
+  ((Dtd*)dtd)->insert_commit((Task*)task);
+}
+PA_EXPORT void pa_task_edge(void* pred, void* succ) {
+  task_add_edge((Task*)pred, (Task*)succ);
+}
+PA_EXPORT void pa_task_retain(void* t) { ((Task*)t)->retain(); }
+PA_EXPORT void pa_task_release(void* t) { ((Task*)t)->release(); }

@@ -104,6 +104,13 @@ void Dtd::write_flow(Data* d, Task* t, int task_rank) {
 
 void Dtd::insert(const TaskClass* tc, const void* args, size_t args_bytes,
                  const FlowSpec* flows, int nflows, int priority, int rank) {
+  Task* t = insert_begin(tc, args, args_bytes, flows, nflows, priority, rank);
+  insert_commit(t);
+}
+
+Task* Dtd::insert_begin(const TaskClass* tc, const void* args,
+                        size_t args_bytes, const FlowSpec* flows, int nflows,
+                        int priority, int rank) {
   PA_CHECK(nflows <= MAX_FLOWS);
   PA_CHECK(args_bytes <= MAX_ARGS_BYTES);
   int task_rank = rank;
@@ -127,6 +134,10 @@ void Dtd::insert(const TaskClass* tc, const void* args, size_t args_bytes,
     if (flows[i].mode & ACCESS_IN) read_flow(flows[i].d, t, task_rank);
     if (flows[i].mode & ACCESS_OUT) write_flow(flows[i].d, t, task_rank);
   }
+  return t;
+}
+
+void Dtd::insert_commit(Task* t) {
   if (t) task_dec_deps(t);
 
   // Window throttling (insert_function.c:75-76): the inserter joins

@@ -41,6 +41,15 @@ class Dtd : public Taskpool {
               const FlowSpec* flows, int nflows, int priority = 0,
               int rank = -1);
 
+  // Two-phase variant for front-ends that add explicit (e.g. CTL) edges:
+  // begin() performs the dataflow chaining and returns the local Task with
+  // its insertion guard still held (nullptr when the task is remote);
+  // commit() releases the guard and applies window throttling.
+  Task* insert_begin(const TaskClass* tc, const void* args, size_t args_bytes,
+                     const FlowSpec* flows, int nflows, int priority,
+                     int rank);
+  void insert_commit(Task* t);
+
   // Push the current version of d back to its home rank (DTD data_flush).
   void flush(Data* d);
   void flush_all(TiledMatrix& A);

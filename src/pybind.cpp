@@ -95,6 +95,7 @@ PYBIND11_MODULE(_core, m) {
            py::arg("world") = -1, py::arg("comm") = std::string(),
            py::arg("gpu") = -1,
            py::call_guard<py::gil_scoped_release>())
+      .def_property_readonly("_handle", [](Context& c) { return (uintptr_t)&c; })
       .def_property_readonly("rank", &Context::rank)
       .def_property_readonly("world", &Context::world)
       .def_property_readonly("nworkers", &Context::nworkers)
@@ -124,6 +125,7 @@ PYBIND11_MODULE(_core, m) {
            py::arg("ctx"), py::arg("m"), py::arg("n"), py::arg("mb"),
            py::arg("nb"), py::arg("p") = 1, py::arg("q") = 1,
            py::arg("elem_size") = 8, py::keep_alive<1, 2>())
+      .def_property_readonly("_handle", [](TiledMatrix& a) { return (uintptr_t)&a; })
       .def_property_readonly("mt", &TiledMatrix::mt)
       .def_property_readonly("nt", &TiledMatrix::nt)
       .def_property_readonly("mb", &TiledMatrix::mb)
@@ -149,9 +151,40 @@ PYBIND11_MODULE(_core, m) {
         for (int j = 0; j < cols; j++)
           for (int i = 0; i < rows; i++) r(i, j) = p[(size_t)j * A.mb() + i];
         return out;
+      })
+      .def("tile_numpy_set", [](TiledMatrix& A, int tm, int tn,
+                                py::array_t<double> v) {
+        PA_CHECK(A.is_local(tm, tn), "tile_numpy_set: tile is not local");
+        Data* d = A.tile(tm, tn);
+        double* p = (double*)d->ensure_host();
+        auto r = v.unchecked<2>();
+        int rows = A.tile_rows(tm), cols = A.tile_cols(tn);
+        PA_CHECK(r.shape(0) == rows && r.shape(1) == cols);
+        for (int j = 0; j < cols; j++)
+          for (int i = 0; i < rows; i++) p[(size_t)j * A.mb() + i] = r(i, j);
+        d->written_on(false);
+      })
+      .def("tile_bytes", [](TiledMatrix& A, int tm, int tn) {
+        PA_CHECK(A.is_local(tm, tn));
+        Data* d = A.tile(tm, tn);
+        void* p;
+        {
+          py::gil_scoped_release rel;
+          p = d->pull_to_host();
+        }
+        return py::bytes((const char*)p, d->bytes);
+      })
+      .def("tile_bytes_set", [](TiledMatrix& A, int tm, int tn, py::bytes b) {
+        PA_CHECK(A.is_local(tm, tn));
+        Data* d = A.tile(tm, tn);
+        std::string s(b);
+        PA_CHECK(s.size() <= d->bytes);
+        memcpy(d->ensure_host(), s.data(), s.size());
+        d->written_on(false);
       });
 
   py::class_<Taskpool>(m, "Taskpool")
+      .def_property_readonly("_handle", [](Taskpool& t) { return (uintptr_t)&t; })
       .def("wait", &Taskpool::wait, py::call_guard<py::gil_scoped_release>())
       .def_property_readonly("pending", &Taskpool::pending);
 

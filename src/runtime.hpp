@@ -68,6 +68,9 @@ struct TaskClass {
   void (*destruct)(Task&) = nullptr;
   double flops = 0.0;  // time_estimate analog (parsec_internal.h:411-459)
   int id = -1;
+  // opaque user hooks for the C ABI (PTG-generated code) trampolines
+  void* user_cpu = nullptr;
+  void* user_gpu = nullptr;
 };
 
 struct FlowRef {

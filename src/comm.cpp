@@ -13,6 +13,7 @@
 #include <unordered_map>
 
 #include "data.hpp"
+#include "profiling.hpp"
 
 namespace pa {
 
@@ -277,7 +278,11 @@ class TcpComm : public CommEngine {
       }
       p.out_off += (size_t)w;
       if (p.out_off < buf.size()) return;
-      if (done) task_complete(done);
+      if (done) {
+        counters().comm_msgs.fetch_add(1, std::memory_order_relaxed);
+        counters().comm_bytes.fetch_add(buf.size(), std::memory_order_relaxed);
+        task_complete(done);
+      }
       p.out.pop_front();
       p.out_off = 0;
     }

@@ -1,6 +1,7 @@
 #include "device_gpu.hpp"
 
 #include "data.hpp"
+#include "profiling.hpp"
 
 namespace pa {
 
@@ -181,7 +182,7 @@ void GpuEngine::stage_in_and_launch(Task* t) {
   t->tc->gpu_hook(*t, gctx);
   hipEvent_t ev = event_get();
   PA_HIP_CHECK(hipEventRecord(ev, es));
-  inflight_[si].push_back(InFlight{t, ev, si});
+  inflight_[si].push_back(InFlight{t, ev, si, Profiler::now_ns()});
   n_inflight_++;
 }
 
@@ -201,6 +202,11 @@ void GpuEngine::manager_main() {
           Data* d = t->flows[i].data;
           if (d && (t->flows[i].mode & ACCESS_OUT)) d->written_on(true);
         }
+        Profiler& pr = Profiler::inst();
+        if (pr.enabled())
+          pr.record(Ev::GPU_TASK, (uint16_t)t->tc->id, t->seq, f.t0_ns,
+                    Profiler::now_ns());
+        counters().tasks_executed_gpu.fetch_add(1, std::memory_order_relaxed);
         event_put(f.event);
         ring.pop_front();
         n_inflight_--;

@@ -76,6 +76,7 @@ class GpuEngine {
     Task* task;
     hipEvent_t event;
     int stream_idx;
+    uint64_t t0_ns;
   };
   struct PQEntry {
     Task* t;

@@ -14,6 +14,7 @@
 #include <rocsolver/rocsolver.h>
 
 #include "device_gpu.hpp"
+#include "profiling.hpp"
 
 namespace pa {
 
@@ -354,6 +355,7 @@ double bench_dgemm_rocblas(int m, int n, int k, int iters) {
 static TaskClass make_tc(const char* name, TaskKind kind,
                          void (*cpu)(Task&), void (*gpu)(Task&, GpuTaskCtx&),
                          int id) {
+  Profiler::inst().register_class(id, name);
   TaskClass tc;
   tc.name = name;
   tc.kind = kind;

@@ -1,0 +1,110 @@
+#include "profiling.hpp"
+
+#include <chrono>
+
+#include "runtime.hpp"
+
+namespace pa {
+
+RuntimeCounters& counters() {
+  static RuntimeCounters c;
+  return c;
+}
+
+Profiler& Profiler::inst() {
+  static Profiler p;
+  return p;
+}
+
+uint64_t Profiler::now_ns() {
+  using namespace std::chrono;
+  return (uint64_t)duration_cast<nanoseconds>(
+             steady_clock::now().time_since_epoch())
+      .count();
+}
+
+Profiler::Buf* Profiler::tls_buf() {
+  static thread_local Buf* b = nullptr;
+  static std::atomic<uint32_t> next_tid{0};
+  if (!b) {
+    b = new Buf();
+    b->tid = next_tid.fetch_add(1);
+    b->recs.reserve(1 << 16);
+    std::lock_guard<std::mutex> g(mtx_);
+    bufs_.push_back(b);
+  }
+  return b;
+}
+
+void Profiler::start(const std::string& filename) {
+  filename_ = filename;
+  enabled_.store(true, std::memory_order_release);
+}
+
+void Profiler::register_class(int id, const std::string& name) {
+  std::lock_guard<std::mutex> g(mtx_);
+  classes_.emplace_back(id, name);
+}
+
+void Profiler::record(Ev kind, uint16_t class_id, uint64_t seq, uint64_t t0,
+                      uint64_t t1) {
+  Buf* b = tls_buf();
+  b->recs.push_back(TraceRec{t0, t1, b->tid, (uint16_t)kind, class_id, seq});
+}
+
+void Profiler::stop_and_dump() {
+  if (!enabled_.exchange(false)) return;
+  std::lock_guard<std::mutex> g(mtx_);
+  FILE* f = fopen(filename_.c_str(), "wb");
+  if (!f) {
+    fprintf(stderr, "[parsec_amd] cannot open trace file %s\n",
+            filename_.c_str());
+    return;
+  }
+  // header: magic, version, dictionary as a JSON line, then raw records
+  fprintf(f, "PABT1\n{\"classes\":{");
+  bool first = true;
+  for (auto& [id, name] : classes_) {
+    fprintf(f, "%s\"%d\":\"%s\"", first ? "" : ",", id, name.c_str());
+    first = false;
+  }
+  fprintf(f, "},\"kinds\":{\"1\":\"exec\",\"2\":\"gpu_task\",\"3\":"
+             "\"stage_in\",\"4\":\"send\",\"5\":\"recv\",\"6\":\"sched\"},"
+             "\"rec_bytes\":%zu}\n", sizeof(TraceRec));
+  for (Buf* b : bufs_) {
+    fwrite(b->recs.data(), sizeof(TraceRec), b->recs.size(), f);
+    b->recs.clear();
+  }
+  fclose(f);
+}
+
+void Profiler::dot_open(const std::string& filename) {
+  std::lock_guard<std::mutex> g(dot_mtx_);
+  dot_ = fopen(filename.c_str(), "w");
+  if (dot_) fprintf(dot_, "digraph parsec_amd {\n");
+}
+
+void Profiler::dot_close() {
+  std::lock_guard<std::mutex> g(dot_mtx_);
+  if (dot_) {
+    fprintf(dot_, "}\n");
+    fclose(dot_);
+    dot_ = nullptr;
+  }
+}
+
+void Profiler::dot_node(const Task* t) {
+  std::lock_guard<std::mutex> g(dot_mtx_);
+  if (!dot_) return;
+  fprintf(dot_, "  t%lu [label=\"%s(%lu)\"];\n", (unsigned long)t->seq,
+          t->tc->name.c_str(), (unsigned long)t->seq);
+}
+
+void Profiler::dot_edge(const Task* pred, const Task* succ) {
+  std::lock_guard<std::mutex> g(dot_mtx_);
+  if (!dot_) return;
+  fprintf(dot_, "  t%lu -> t%lu;\n", (unsigned long)pred->seq,
+          (unsigned long)succ->seq);
+}
+
+}  // namespace pa

@@ -1,0 +1,91 @@
+// Observability: binary event tracing, DOT DAG dump, runtime counters.
+//
+// Reference parity:
+//  - Tracing: per-thread event streams with a dictionary of event classes,
+//    dumped as a compact binary trace ("PBT" analog: parsec/profiling.c,
+//    parsec_binary_profile.h) convertible to Chrome trace JSON
+//    (parsec_amd/tools/trace2chrome.py ~ tools/profiling/h5toctf.py).
+//  - DOT grapher (parsec_prof_grapher.c): every task node + dependency edge
+//    when profile_dot is set.
+//  - Counters (PINS/papi_sde analogs): per-context scheduler/executed/
+//    steal counts, exposed to Python and printed at teardown with
+//    PARSEC_MCA_stats=1 (device.c:611-658 statistics dump analog).
+#pragma once
+
+#include <atomic>
+#include <cstdint>
+#include <cstdio>
+#include <mutex>
+#include <string>
+#include <vector>
+
+namespace pa {
+
+struct Task;
+
+enum class Ev : uint16_t {
+  EXEC = 1,       // CPU task body
+  GPU_TASK = 2,   // GPU task launch..retire (host view)
+  STAGE_IN = 3,
+  COMM_SEND = 4,
+  COMM_RECV = 5,
+  SCHED = 6,
+};
+
+struct TraceRec {
+  uint64_t t0_ns, t1_ns;
+  uint32_t tid;
+  uint16_t kind;
+  uint16_t class_id;
+  uint64_t seq;
+};
+
+class Profiler {
+ public:
+  static Profiler& inst();
+
+  bool enabled() const { return enabled_.load(std::memory_order_relaxed); }
+  void start(const std::string& filename);
+  void stop_and_dump();
+
+  void record(Ev kind, uint16_t class_id, uint64_t seq, uint64_t t0_ns,
+              uint64_t t1_ns);
+  void register_class(int id, const std::string& name);
+
+  static uint64_t now_ns();
+
+  // --- DOT grapher ---
+  bool dot_enabled() const { return dot_ != nullptr; }
+  void dot_node(const Task* t);
+  void dot_edge(const Task* pred, const Task* succ);
+  void dot_open(const std::string& filename);
+  void dot_close();
+
+ private:
+  struct Buf {
+    std::vector<TraceRec> recs;
+    uint32_t tid;
+  };
+  Buf* tls_buf();
+
+  std::atomic<bool> enabled_{false};
+  std::string filename_;
+  std::mutex mtx_;
+  std::vector<Buf*> bufs_;
+  std::vector<std::pair<int, std::string>> classes_;
+  FILE* dot_ = nullptr;
+  std::mutex dot_mtx_;
+};
+
+// global counters (papi_sde-style software counters)
+struct RuntimeCounters {
+  std::atomic<uint64_t> tasks_executed_cpu{0};
+  std::atomic<uint64_t> tasks_executed_gpu{0};
+  std::atomic<uint64_t> tasks_scheduled{0};
+  std::atomic<uint64_t> steals{0};
+  std::atomic<uint64_t> comm_msgs{0};
+  std::atomic<uint64_t> comm_bytes{0};
+};
+RuntimeCounters& counters();
+
+}  // namespace pa

@@ -22,6 +22,7 @@
 #include "comm.hpp"
 #include "data.hpp"
 #include "device_gpu.hpp"
+#include "profiling.hpp"
 
 namespace pa {
 
@@ -230,6 +231,9 @@ class RcclComm : public CommEngine {
         inflight_.erase(inflight_.begin() + i);
         events_.push_back(f.ev);
         if (f.is_recv) f.t->flows[0].data->written_on(true);
+        counters().comm_msgs.fetch_add(1, std::memory_order_relaxed);
+        counters().comm_bytes.fetch_add(f.t->flows[0].data->bytes,
+                                        std::memory_order_relaxed);
         task_complete(f.t);
       }
       if (stop_.load(std::memory_order_acquire) && inflight_.empty()) {

@@ -3,6 +3,7 @@
 #include "data.hpp"
 #include "device_gpu.hpp"
 #include "comm.hpp"
+#include "profiling.hpp"
 
 namespace pa {
 
@@ -15,6 +16,8 @@ Task* task_new(Taskpool* tp, const TaskClass* tc) {
   t->tc = tc;
   t->seq = tp->next_seq();
   tp->task_created();
+  Profiler& pr = Profiler::inst();
+  if (pr.dot_enabled()) pr.dot_node(t);
   return t;
 }
 
@@ -26,6 +29,8 @@ void Task::release() {
 }
 
 bool task_add_edge(Task* pred, Task* succ) {
+  Profiler& pr = Profiler::inst();
+  if (pr.dot_enabled()) pr.dot_edge(pred, succ);
   // Dependency-release race protocol (SURVEY.md §7 "hard parts"): the edge
   // is registered under the predecessor's lock; if the predecessor already
   // completed, the successor does not wait on it.
@@ -59,7 +64,17 @@ void task_complete(Task* t) {
 }
 
 void run_cpu_task(Task* t) {
-  if (t->tc->cpu_hook) t->tc->cpu_hook(*t);
+  Profiler& pr = Profiler::inst();
+  if (t->tc->cpu_hook) {
+    if (pr.enabled()) {
+      uint64_t t0 = Profiler::now_ns();
+      t->tc->cpu_hook(*t);
+      pr.record(Ev::EXEC, (uint16_t)t->tc->id, t->seq, t0, Profiler::now_ns());
+    } else {
+      t->tc->cpu_hook(*t);
+    }
+  }
+  counters().tasks_executed_cpu.fetch_add(1, std::memory_order_relaxed);
   task_complete(t);
 }
 
@@ -153,7 +168,11 @@ Task* Scheduler::pop(int worker) {
     q.lock.lock();
     if (!q.dq.empty()) { t = q.dq.back(); q.dq.pop_back(); }
     q.lock.unlock();
-    if (t) { npending_.fetch_sub(1, std::memory_order_relaxed); return t; }
+    if (t) {
+      npending_.fetch_sub(1, std::memory_order_relaxed);
+      counters().steals.fetch_add(1, std::memory_order_relaxed);
+      return t;
+    }
   }
   return nullptr;
 }
@@ -182,6 +201,10 @@ Context::Context(const Options& opt) : rank_(opt.rank), world_(opt.world) {
   if (opt.gpu_device != -2) gpu_ = GpuEngine::create(this, opt.gpu_device);
   comm_ = CommEngine::create(this, opt.comm);
 
+  std::string prof = param_str("profile_filename", "");
+  if (!prof.empty()) Profiler::inst().start(prof + "." + std::to_string(rank_));
+  std::string dot = param_str("profile_dot", "");
+  if (!dot.empty()) Profiler::inst().dot_open(dot + "." + std::to_string(rank_));
   for (int i = 0; i < nworkers_; i++)
     workers_.emplace_back([this, i] { worker_main(i); });
   PA_DEBUG(1, "context up: rank %d/%d, %d workers, gpu=%d", rank_, world_,
@@ -194,6 +217,20 @@ Context::~Context() {
   for (auto& w : workers_) w.join();
   comm_.reset();
   gpu_.reset();
+  Profiler::inst().stop_and_dump();
+  Profiler::inst().dot_close();
+  if (param_int("stats", 0)) {
+    RuntimeCounters& c = counters();
+    fprintf(stderr,
+            "[parsec_amd stats] rank %d: cpu_tasks=%lu gpu_tasks=%lu "
+            "scheduled=%lu steals=%lu comm_msgs=%lu comm_bytes=%lu\n",
+            rank_, (unsigned long)c.tasks_executed_cpu.load(),
+            (unsigned long)c.tasks_executed_gpu.load(),
+            (unsigned long)c.tasks_scheduled.load(),
+            (unsigned long)c.steals.load(),
+            (unsigned long)c.comm_msgs.load(),
+            (unsigned long)c.comm_bytes.load());
+  }
 }
 
 void Context::worker_main(int id) {
@@ -209,6 +246,7 @@ void Context::worker_main(int id) {
 }
 
 void Context::dispatch(Task* t, int worker_hint) {
+  counters().tasks_scheduled.fetch_add(1, std::memory_order_relaxed);
   switch (t->tc->kind) {
     case TaskKind::GPU:
       if (gpu_) { gpu_->enqueue(t); return; }

@@ -1,0 +1,41 @@
+"""Tracing, DOT grapher, stats counters (SURVEY.md §5 aux subsystems)."""
+import os
+import subprocess
+import sys
+
+REPO = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+
+
+def test_trace_and_dot(tmp_path):
+    trace = tmp_path / "trace"
+    dot = tmp_path / "dag.dot"
+    code = f"""
+import sys
+sys.path.insert(0, {REPO!r})
+import parsec_amd as pm
+pm.param_set("profile_filename", {str(trace)!r})
+pm.param_set("profile_dot", {str(dot)!r})
+pm.param_set("stats", "1")
+ctx = pm.Context(nworkers=2, rank=0, world=1, gpu=-2)
+A = pm.TiledMatrix(ctx, 128, 128, 64, 64, 1, 1)
+tp = pm.Dtd(ctx)
+pm.insert_spd_fill(tp, A, 1)
+pm.insert_potrf(tp, A)
+tp.wait()
+del A
+del ctx
+"""
+    r = subprocess.run([sys.executable, "-c", code], capture_output=True,
+                       text=True, timeout=120)
+    assert r.returncode == 0, r.stdout + r.stderr
+    assert "[parsec_amd stats]" in r.stderr
+    tfile = str(trace) + ".0"
+    assert os.path.exists(tfile), "trace file missing"
+    from parsec_amd.tools.trace2chrome import convert
+    out, n = convert(tfile, str(tmp_path / "t.json"))
+    assert n > 5, f"too few trace events: {n}"
+    dfile = str(dot) + ".0"
+    dtext = open(dfile).read()
+    assert "digraph" in dtext
+    assert "potrf" in dtext
+    assert "->" in dtext

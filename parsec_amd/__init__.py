@@ -30,6 +30,7 @@ from parsec_amd._core import (  # noqa: F401,E402
     TiledMatrix,
     hip_device_count,
     insert_potrf,
+    insert_geqrf,
     insert_spd_fill,
     param_dump,
     param_set,

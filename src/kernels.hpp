@@ -27,8 +27,16 @@ TaskClass& tc_trsm();
 TaskClass& tc_syrk();
 TaskClass& tc_gemm();
 
+TaskClass& tc_trtri();
+TaskClass& tc_trsm_inv();
+TaskClass& tc_geqrt();
+TaskClass& tc_unmqr();
+TaskClass& tc_tsqrt();
+TaskClass& tc_tsmqr();
+
 // Build the DAGs (DTD insertion; SPMD-safe: call on every rank).
 void insert_spd_fill(Dtd& tp, TiledMatrix& A, uint32_t seed);
 void insert_potrf(Dtd& tp, TiledMatrix& A);
+void insert_geqrf(Dtd& tp, TiledMatrix& A);
 
 }  // namespace pa

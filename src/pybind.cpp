@@ -218,6 +218,8 @@ PYBIND11_MODULE(_core, m) {
         py::arg("seed") = 42u, py::call_guard<py::gil_scoped_release>());
   m.def("insert_potrf", &insert_potrf, py::arg("tp"), py::arg("A"),
         py::call_guard<py::gil_scoped_release>());
+  m.def("insert_geqrf", &insert_geqrf, py::arg("tp"), py::arg("A"),
+        py::call_guard<py::gil_scoped_release>());
 
   m.def("param_set", &param_set);
   m.def("param_dump", &param_dump);

@@ -25,25 +25,40 @@ def main():
     ctx = pm.Context(nworkers=2, rank=rank, world=world, comm="tcp",
                      gpu=(0 if use_gpu else -2))
 
+    app = os.environ.get("PARSEC_TEST_APP", "potrf")
     A = pm.TiledMatrix(ctx, n, n, nb, nb, p, q)
     tp = pm.Dtd(ctx)
     pm.insert_spd_fill(tp, A, 42)
     tp.wait()
+    if app == "qr":
+        # QR needs the full matrix: fill upper tiles too (spd fill covers
+        # the lower triangle only)
+        import numpy as np2
+        rng = np2.random.default_rng(5)
+        for tm in range(A.mt):
+            for tn in range(A.nt):
+                if tn > tm and A.is_local(tm, tn):
+                    A.tile_numpy_set(tm, tn, rng.standard_normal(
+                        (A.tile_rows(tm), A.tile_cols(tn))) )
+    hi = (lambda tm: A.nt) if app == "qr" else (lambda tm: min(tm + 1, A.nt))
     pre = {}
     for tm in range(A.mt):
-        for tn in range(min(tm + 1, A.nt)):
+        for tn in range(hi(tm)):
             if A.is_local(tm, tn):
                 pre[f"{tm}_{tn}"] = A.tile_numpy(tm, tn)
 
     tp2 = pm.Dtd(ctx)
-    pm.insert_potrf(tp2, A)
+    if app == "qr":
+        pm.insert_geqrf(tp2, A)
+    else:
+        pm.insert_potrf(tp2, A)
     tp2.flush_all(A)
     tp2.wait()
     ctx.barrier()
 
     post = {}
     for tm in range(A.mt):
-        for tn in range(min(tm + 1, A.nt)):
+        for tn in range(hi(tm)):
             if A.is_local(tm, tn):
                 post[f"{tm}_{tn}"] = A.tile_numpy(tm, tn)
 

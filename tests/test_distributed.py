@@ -56,3 +56,37 @@ def test_distributed_cholesky(world, p, q, tmp_path):
     L0 = np.linalg.cholesky(M)
     err = np.abs(np.tril(post) - L0).max()
     assert err < 1e-10, f"world={world} p={p} q={q}: max err {err}"
+
+
+def test_distributed_qr(tmp_path):
+    """World-2 tile QR over the TCP engine: checks workspace-collection
+    transfers (V2/T1 tiles) and flush; verified via R^T R == A^T A."""
+    world, p, q, n, nb = 2, 2, 1, 256, 64
+    port = _next_port[0]
+    _next_port[0] += world + 2
+    procs = []
+    for r in range(world):
+        env = dict(os.environ)
+        env.update(RANK=str(r), WORLD_SIZE=str(world),
+                   PARSEC_TEST_PORT=str(port), PARSEC_TEST_OUT=str(tmp_path),
+                   GRID_P=str(p), GRID_Q=str(q), MAT_N=str(n), MAT_NB=str(nb),
+                   PARSEC_TEST_APP="qr")
+        procs.append(subprocess.Popen([sys.executable, WORKER], env=env,
+                                      stdout=subprocess.PIPE,
+                                      stderr=subprocess.STDOUT))
+    for pr in procs:
+        out, _ = pr.communicate(timeout=180)
+        assert pr.returncode == 0, f"worker failed:\n{out.decode()}"
+    pre = np.zeros((n, n))
+    post = np.zeros((n, n))
+    for r in range(world):
+        z = np.load(os.path.join(tmp_path, f"rank{r}.npz"))
+        for key in z.files:
+            kind, tm, tn = key.split("_")[0], *key.split("_")[1:]
+            tm, tn = int(tm), int(tn)
+            v = z[key]
+            dst = pre if kind == "pre" else post
+            dst[tm * nb:tm * nb + v.shape[0], tn * nb:tn * nb + v.shape[1]] = v
+    R = np.triu(post)
+    err = np.abs(R.T @ R - pre.T @ pre).max() / np.abs(pre.T @ pre).max()
+    assert err < 1e-12, f"distributed QR rel err {err}"

@@ -36,6 +36,11 @@ struct Data {
   void* dev_ptr = nullptr;
   bool host_valid = false;
   bool dev_valid = false;
+  // H2D staging fence: recorded on the copy stream when an async H2D is
+  // issued; EVERY consumer stream must wait on it before reading dev_ptr
+  // (a task that did not issue the copy still races it otherwise).
+  void* h2d_event = nullptr;  // hipEvent_t, lazily created, owned here
+  bool h2d_pending = false;
 
   // ---- DTD chaining state (single inserter thread; no lock needed) ----
   uint32_t version = 0;        // logical version, bumped per writer insertion

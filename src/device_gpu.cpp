@@ -153,7 +153,6 @@ void GpuEngine::stage_in_and_launch(Task* t) {
     if (next_stream_ >= (int)exec_streams_.size()) next_stream_ = 1;
   }
   hipStream_t es = exec_streams_[si];
-  bool copied = false;
   for (int i = 0; i < t->nflows; i++) {
     Data* d = t->flows[i].data;
     if (!d) { t->dev_ptr[i] = nullptr; continue; }
@@ -165,18 +164,20 @@ void GpuEngine::stage_in_and_launch(Task* t) {
                                   hipMemcpyHostToDevice, h2d_stream_));
       stats.bytes_h2d += d->bytes;
       d->dev_valid = true;  // valid in stream order on h2d_stream_
-      copied = true;
+      if (!d->h2d_event)
+        PA_HIP_CHECK(hipEventCreateWithFlags((hipEvent_t*)&d->h2d_event,
+                                             hipEventDisableTiming));
+      PA_HIP_CHECK(hipEventRecord((hipEvent_t)d->h2d_event, h2d_stream_));
+      d->h2d_pending = true;
     }
     if (!(t->flows[i].mode & ACCESS_IN) && !d->dev_valid) {
       d->dev_valid = true;  // OUTPUT-only: content produced by this task
     }
+    // Every consumer of a staged tile waits on ITS copy fence — including
+    // tasks that found the copy already issued by an earlier task.
+    if (d->h2d_pending && (t->flows[i].mode & ACCESS_IN))
+      PA_HIP_CHECK(hipStreamWaitEvent(es, (hipEvent_t)d->h2d_event, 0));
     t->dev_ptr[i] = d->dev_ptr;
-  }
-  if (copied) {
-    hipEvent_t ce = event_get();
-    PA_HIP_CHECK(hipEventRecord(ce, h2d_stream_));
-    PA_HIP_CHECK(hipStreamWaitEvent(es, ce, 0));
-    event_put(ce);  // safe: event re-record happens on this thread only
   }
   GpuTaskCtx gctx{es, device_, this};
   t->tc->gpu_hook(*t, gctx);
@@ -250,6 +251,7 @@ Data::~Data() {
   // Drop DTD chaining references to completed tasks.
   if (last_local_writer) last_local_writer->release();
   for (Task* r : local_readers) r->release();
+  if (h2d_event) hipEventDestroy((hipEvent_t)h2d_event);
   if (host_ptr) free(host_ptr);
   // dev_ptr returns to the engine pool with the collection teardown;
   // the slab itself is freed by the engine.

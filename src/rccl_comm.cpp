@@ -151,6 +151,9 @@ class RcclComm : public CommEngine {
       PA_HIP_CHECK(hipMemcpyAsync(d->dev_ptr, d->host_ptr, d->bytes,
                                   hipMemcpyHostToDevice, stream));
       d->dev_valid = true;
+    } else if (d->h2d_pending) {
+      // the GPU engine may still be staging this tile on its copy stream
+      PA_HIP_CHECK(hipStreamWaitEvent(stream, (hipEvent_t)d->h2d_event, 0));
     }
     return d->dev_ptr;
   }

@@ -10,14 +10,8 @@ See SURVEY.md for the structural map of the reference (ICLDisco/parsec)
 this framework re-implements natively.
 """
 
-import ctypes
 import os
 import sys
-
-# _core must be loaded RTLD_GLOBAL so that PTG-generated modules
-# (parsec_amd.ptg) resolve the exported pa_* C ABI against it.
-_flags = sys.getdlopenflags()
-sys.setdlopenflags(_flags | ctypes.RTLD_GLOBAL)
 
 from parsec_amd._core import (  # noqa: F401,E402
     ACCESS_IN,
@@ -35,8 +29,6 @@ from parsec_amd._core import (  # noqa: F401,E402
     param_dump,
     param_set,
 )
-
-sys.setdlopenflags(_flags)
 
 __version__ = "0.1.0"
 

@@ -527,7 +527,7 @@ def generate_cpp(jdf, name):
 class PtgModule:
     def __init__(self, so_path, name, jdf):
         import ctypes
-        self._lib = ctypes.CDLL(so_path, mode=ctypes.RTLD_GLOBAL)
+        self._lib = ctypes.CDLL(so_path)
         self._build = getattr(self._lib, f"ptg_build_{name}")
         self._jdf = jdf
         self.name = name
@@ -567,9 +567,13 @@ def compile_jdf(path, verbose=False):
         src = os.path.join(CACHE, f"{name}_{h}.cpp")
         with open(src, "w") as f:
             f.write(cpp)
+        # Link directly against _core.so so the pa_* C ABI resolves without
+        # polluting the global symbol namespace (RTLD_GLOBAL on _core breaks
+        # a later `import torch`: duplicate ROCm library symbols).
+        core = os.path.join(REPO, "parsec_amd", "_core.so")
         cmd = ["hipcc", "--offload-arch=gfx950", "-O2", "-std=c++17",
                "-fPIC", "-shared", "-I", os.path.join(REPO, "src"),
-               src, "-o", so]
+               src, core, "-o", so]
         r = subprocess.run(cmd, capture_output=True, text=True)
         if r.returncode != 0:
             raise JdfError(f"ptgpp: generated code failed to compile:\n"

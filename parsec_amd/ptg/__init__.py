@@ -569,11 +569,14 @@ def compile_jdf(path, verbose=False):
             f.write(cpp)
         # Link directly against _core.so so the pa_* C ABI resolves without
         # polluting the global symbol namespace (RTLD_GLOBAL on _core breaks
-        # a later `import torch`: duplicate ROCm library symbols).
-        core = os.path.join(REPO, "parsec_amd", "_core.so")
+        # a later `import torch`: duplicate ROCm library symbols). -l: keeps
+        # hipcc from treating the .so as a HIP source; the rpath makes the
+        # loader resolve it to the already-mapped copy (same inode).
+        coredir = os.path.join(REPO, "parsec_amd")
         cmd = ["hipcc", "--offload-arch=gfx950", "-O2", "-std=c++17",
                "-fPIC", "-shared", "-I", os.path.join(REPO, "src"),
-               src, core, "-o", so]
+               src, "-L", coredir, "-l:_core.so",
+               f"-Wl,-rpath,{coredir}", "-o", so]
         r = subprocess.run(cmd, capture_output=True, text=True)
         if r.returncode != 0:
             raise JdfError(f"ptgpp: generated code failed to compile:\n"

@@ -45,3 +45,8 @@ for (m,n,k) in [(2048,2048,2048), (4096,4096,4096), (8192,8192,2048)]:
     for impl in ["v2", "v1", "rocblas"]:
         tf = _core.bench_dgemm(m, n, k, 10, impl)
         print(f"{m}x{n}x{k} {impl}: {tf:.1f} TF/s")
+
+print("\n--- bf16 gemm kernel throughput ---")
+for (m,n,k) in [(4096,4096,4096), (8192,8192,8192)]:
+    tf = _core.bench_gemm_bf16(m, n, k, 10)
+    print(f"bf16 {m}x{n}x{k}: {tf:.0f} TF/s")

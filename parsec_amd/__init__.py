@@ -25,6 +25,8 @@ from parsec_amd._core import (  # noqa: F401,E402
     hip_device_count,
     insert_potrf,
     insert_geqrf,
+    insert_fill_bf16,
+    insert_gemm_bf16,
     insert_spd_fill,
     param_dump,
     param_set,

@@ -38,5 +38,7 @@ TaskClass& tc_tsmqr();
 void insert_spd_fill(Dtd& tp, TiledMatrix& A, uint32_t seed);
 void insert_potrf(Dtd& tp, TiledMatrix& A);
 void insert_geqrf(Dtd& tp, TiledMatrix& A);
+void insert_fill_bf16(Dtd& tp, TiledMatrix& A, uint32_t seed);
+void insert_gemm_bf16(Dtd& tp, TiledMatrix& At, TiledMatrix& B, TiledMatrix& C);
 
 }  // namespace pa

@@ -22,6 +22,9 @@ void test_dgemm_nt_hip(int, int, int, const double*, int, const double*, int,
 void test_potf2_hip(double*, int);
 double bench_dgemm_hip(int, int, int, int, int);
 double bench_dgemm_rocblas(int, int, int, int);
+double bench_gemm_bf16(int, int, int, int);
+void test_gemm_bf16_hip(int, int, int, const uint16_t*, const uint16_t*,
+                        float*);
 }  // namespace pa
 
 namespace py = pybind11;
@@ -220,6 +223,17 @@ PYBIND11_MODULE(_core, m) {
         py::call_guard<py::gil_scoped_release>());
   m.def("insert_geqrf", &insert_geqrf, py::arg("tp"), py::arg("A"),
         py::call_guard<py::gil_scoped_release>());
+  m.def("insert_fill_bf16", &insert_fill_bf16, py::arg("tp"), py::arg("A"),
+        py::arg("seed") = 1u, py::call_guard<py::gil_scoped_release>());
+  m.def("insert_gemm_bf16", &insert_gemm_bf16, py::arg("tp"), py::arg("At"),
+        py::arg("B"), py::arg("C"), py::call_guard<py::gil_scoped_release>());
+  m.def("bench_gemm_bf16", [](int m, int n, int k, int iters) {
+    return pa::bench_gemm_bf16(m, n, k, iters);
+  }, py::call_guard<py::gil_scoped_release>());
+  m.def("gemm_bf16_hip", [](py::array_t<uint16_t> A, py::array_t<uint16_t> B,
+                            py::array_t<float> C, int m, int n, int k) {
+    pa::test_gemm_bf16_hip(m, n, k, A.data(), B.data(), C.mutable_data());
+  });
 
   m.def("param_set", &param_set);
   m.def("param_dump", &param_dump);

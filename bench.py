@@ -33,7 +33,7 @@ def main():
     ap.add_argument("--steps", type=int, default=5)
     ap.add_argument("--warmup", type=int, default=2)
     ap.add_argument("--matrix-size", type=int, default=80000)
-    ap.add_argument("--tile", type=int, default=2048)
+    ap.add_argument("--tile", type=int, default=4096)
     ap.add_argument("--chore-gemm", type=str, default=os.environ.get(
         "PARSEC_MCA_chore_gemm", "rocblas"), choices=["rocblas", "hip"])
     ap.add_argument("--workers", type=int, default=4)
@@ -57,7 +57,10 @@ def main():
 
     use_torch_dist = world > 1
     if use_torch_dist:
+        import torch
         import torch.distributed as dist
+        if has_gpu and torch.cuda.is_available():
+            torch.cuda.set_device(int(os.environ.get("LOCAL_RANK", rank)))
 
     def barrier_sync():
         ctx.gpu_sync()

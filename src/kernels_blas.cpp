@@ -8,6 +8,7 @@
 
 #include <cmath>
 #include <map>
+#include <string>
 
 #include <hip/hip_runtime.h>
 #include <rocblas/rocblas.h>
@@ -170,6 +171,9 @@ static void cpu_gemm(Task& t) {
   t.flows[2].data->written_on(false);
 }
 
+void launch_dsyrk_v2(int n, int k, const double* A, int lda, double* C,
+                     int ldc, hipStream_t stream);  // kernels_hip.cpp
+
 // ------------------------------------------------------------------ GPU chores
 // One rocBLAS handle PER EXEC STREAM: rocBLAS/rocSOLVER keep device
 // workspace on the handle (e.g. dtrsm's invA), so concurrent kernels on
@@ -246,7 +250,16 @@ static void gpu_trsm(Task& t, GpuTaskCtx& g) {
 static void gpu_syrk(Task& t, GpuTaskCtx& g) {
   const TileArgs& a = t.arg<TileArgs>();
   const double mone = -1.0, one = 1.0;
-  static const bool use_syrkx = param_str("chore_syrk", "dgemm") == "syrkx";
+  static const std::string chore = param_str("chore_syrk", "dgemm");
+  static const bool use_syrkx = chore == "syrkx";
+  static const bool use_hip = chore == "hip";
+  if (use_hip) {
+    // hand MFMA kernel over the lower-triangular block set only: half the
+    // FLOPs of the full-tile dgemm route.
+    launch_dsyrk_v2(a.n, a.k, (const double*)t.dev_ptr[0], a.ld,
+                    (double*)t.dev_ptr[1], a.ld, g.stream);
+    return;
+  }
   if (!use_syrkx) {
     // SYRK as full-tile DGEMM(A, A^T): 2x the FLOPs but ~4x faster wall on
     // MI355X (rocBLAS syrkx measured 7.5 TF vs dgemm 60 TF at nb=2048);

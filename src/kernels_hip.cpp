@@ -108,6 +108,7 @@ __device__ __forceinline__ int xcd_swizzle(int id, int nwg) {
   return (xcd < r ? xcd * (q + 1) : r * (q + 1) + (xcd - r) * q) + pos;
 }
 
+template <bool TRI>
 __launch_bounds__(512)
 __global__ void k_dgemm_nt_v2(int m, int n, int k, const double* __restrict__ A,
                               int lda, const double* __restrict__ B, int ldb,
@@ -122,7 +123,20 @@ __global__ void k_dgemm_nt_v2(int m, int n, int k, const double* __restrict__ A,
   const int wr = wave >> 1, wc = wave & 1;  // 4(M) x 2(N) waves: 32x64 each
   const int nwg = gridDim.x;
   int id = xcd_swizzle(blockIdx.x, nwg);
-  const int bm0 = (id % nbx) * BM, bn0 = (id / nbx) * BN;
+  int bx, by;
+  if (TRI) {
+    // lower-triangular block set: id -> (bx >= by); SYRK writes only the
+    // lower blocks (diag blocks compute their full tile: the upper half of
+    // a diagonal tile is symmetric-valid and never read by the DAG).
+    bx = 0;
+    int rem = id;
+    while (rem > bx) { bx++; rem -= bx; }
+    by = rem;
+  } else {
+    bx = id % nbx;
+    by = id / nbx;
+  }
+  const int bm0 = bx * BM, bn0 = by * BN;
   const int ksub = lane >> 4;
   const int r16 = lane & 15;
 
@@ -196,13 +210,29 @@ static void launch_dgemm_v2(int m, int n, int k, const double* A, int lda,
   static bool attr_set = false;
   constexpr size_t lds = 4 * BKD * 129 * 8;
   if (!attr_set) {
-    hipFuncSetAttribute((const void*)k_dgemm_nt_v2,
+    hipFuncSetAttribute((const void*)k_dgemm_nt_v2<false>,
                         hipFuncAttributeMaxDynamicSharedMemorySize, lds);
     attr_set = true;
   }
   int nbx = (m + BM - 1) / BM, nby = (n + BN - 1) / BN;
-  hipLaunchKernelGGL(k_dgemm_nt_v2, dim3(nbx * nby), dim3(512), lds, stream,
-                     m, n, k, A, lda, B, ldb, C, ldc, nbx);
+  hipLaunchKernelGGL(k_dgemm_nt_v2<false>, dim3(nbx * nby), dim3(512), lds,
+                     stream, m, n, k, A, lda, B, ldb, C, ldc, nbx);
+}
+
+// C(lower) -= A A^T : triangular block grid on the same structure.
+void launch_dsyrk_v2(int n, int k, const double* A, int lda, double* C,
+                     int ldc, hipStream_t stream) {
+  static bool attr_set = false;
+  constexpr size_t lds = 4 * BKD * 129 * 8;
+  if (!attr_set) {
+    hipFuncSetAttribute((const void*)k_dgemm_nt_v2<true>,
+                        hipFuncAttributeMaxDynamicSharedMemorySize, lds);
+    attr_set = true;
+  }
+  int ntb = (n + BM - 1) / BM;
+  int ntiles = ntb * (ntb + 1) / 2;
+  hipLaunchKernelGGL(k_dgemm_nt_v2<true>, dim3(ntiles), dim3(512), lds,
+                     stream, n, n, k, A, lda, A, lda, C, ldc, 0);
 }
 
 void gpu_gemm_hip(Task& t, GpuTaskCtx& g) {

@@ -189,7 +189,20 @@ PYBIND11_MODULE(_core, m) {
   py::class_<Taskpool>(m, "Taskpool")
       .def_property_readonly("_handle", [](Taskpool& t) { return (uintptr_t)&t; })
       .def("wait", &Taskpool::wait, py::call_guard<py::gil_scoped_release>())
-      .def_property_readonly("pending", &Taskpool::pending);
+      .def_property_readonly("pending", &Taskpool::pending)
+      // parsec_compose analog (compound.c:17-135): chain a callback to run
+      // when this pool drains (e.g. to start inserting the next pool).
+      .def("on_complete", [](Taskpool& t, py::function fn) {
+        auto holder = std::make_shared<py::function>(std::move(fn));
+        t.on_complete([holder] {
+          py::gil_scoped_acquire gil;
+          try {
+            (*holder)();
+          } catch (py::error_already_set& e) {
+            fprintf(stderr, "[parsec_amd] on_complete raised: %s\n", e.what());
+          }
+        });
+      });
 
   py::class_<Dtd, Taskpool>(m, "Dtd")
       .def(py::init<Context*, std::string>(), py::arg("ctx"),

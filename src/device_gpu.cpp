@@ -179,11 +179,13 @@ void GpuEngine::stage_in_and_launch(Task* t) {
       PA_HIP_CHECK(hipStreamWaitEvent(es, (hipEvent_t)d->h2d_event, 0));
     t->dev_ptr[i] = d->dev_ptr;
   }
-  GpuTaskCtx gctx{es, device_, this};
+  std::vector<std::pair<void*, size_t>> deferred;
+  GpuTaskCtx gctx{es, device_, this, &deferred};
   t->tc->gpu_hook(*t, gctx);
   hipEvent_t ev = event_get();
   PA_HIP_CHECK(hipEventRecord(ev, es));
-  inflight_[si].push_back(InFlight{t, ev, si, Profiler::now_ns()});
+  inflight_[si].push_back(
+      InFlight{t, ev, si, Profiler::now_ns(), std::move(deferred)});
   n_inflight_++;
 }
 
@@ -208,6 +210,7 @@ void GpuEngine::manager_main() {
           pr.record(Ev::GPU_TASK, (uint16_t)t->tc->id, t->seq, f.t0_ns,
                     Profiler::now_ns());
         counters().tasks_executed_gpu.fetch_add(1, std::memory_order_relaxed);
+        for (auto& [p2, b2] : f.deferred_frees) dev_free(p2, b2);
         event_put(f.event);
         ring.pop_front();
         n_inflight_--;

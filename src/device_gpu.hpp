@@ -40,6 +40,9 @@ struct GpuTaskCtx {
   hipStream_t stream;
   int device;
   class GpuEngine* engine;
+  // buffers a hook retires: returned to the pool only after the task's
+  // kernels complete on the stream (safe pool reuse).
+  std::vector<std::pair<void*, size_t>>* deferred_frees;
 };
 
 class GpuEngine {
@@ -77,6 +80,7 @@ class GpuEngine {
     hipEvent_t event;
     int stream_idx;
     uint64_t t0_ns;
+    std::vector<std::pair<void*, size_t>> deferred_frees;
   };
   struct PQEntry {
     Task* t;

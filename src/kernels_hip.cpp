@@ -274,9 +274,10 @@ __global__ void __launch_bounds__(256) k_potf2_mfma(double* A, int n, int ld) {
       __syncthreads();
       for (int i = j + 1 + tid; i < n; i += 256) S[j * LDP + i] *= *dinv;
       __syncthreads();
-      for (int c = j + 1; c < jbe; c++) {
+      // independent trailing columns of the micro-panel: one wave each
+      for (int c = j + 1 + wave; c < jbe; c += 4) {
         double ljc = S[j * LDP + c];
-        for (int i = c + tid; i < n; i += 256)
+        for (int i = c + lane; i < n; i += 64)
           S[c * LDP + i] -= S[j * LDP + i] * ljc;
       }
       __syncthreads();

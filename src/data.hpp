@@ -65,8 +65,11 @@ struct Data {
 // headline; elem_size parametrizes dtype).
 class TiledMatrix {
  public:
+  // sym=true: symmetric storage — tile(m,n) with n>m aliases tile(n,m)
+  // (sym_two_dim_rectangle_cyclic analog: only the lower triangle is
+  // stored/distributed).
   TiledMatrix(Context* ctx, int64_t m, int64_t n, int mb, int nb,
-              int p, int q, size_t elem_size = 8);
+              int p, int q, size_t elem_size = 8, bool sym = false);
   ~TiledMatrix();
 
   Context* ctx() const { return ctx_; }
@@ -81,7 +84,10 @@ class TiledMatrix {
   size_t elem_size() const { return elem_; }
   size_t tile_bytes() const { return (size_t)mb_ * nb_ * elem_; }
 
-  int rank_of(int tm, int tn) const { return (tm % p_) * q_ + (tn % q_); }
+  int rank_of(int tm, int tn) const {
+    if (sym_ && tn > tm) { int t = tm; tm = tn; tn = t; }
+    return (tm % p_) * q_ + (tn % q_);
+  }
   bool is_local(int tm, int tn) const {
     return rank_of(tm, tn) == ctx_rank_;
   }
@@ -103,6 +109,7 @@ class TiledMatrix {
   int64_t m_, n_;
   int mb_, nb_, mt_, nt_, p_, q_;
   size_t elem_;
+  bool sym_ = false;
   std::vector<std::unique_ptr<Data>> tiles_;  // mt*nt, metadata eager
 };
 

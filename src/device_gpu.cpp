@@ -294,9 +294,9 @@ void Data::written_on(bool device) {
 
 // ---------------------------------------------------------------- TiledMatrix
 TiledMatrix::TiledMatrix(Context* ctx, int64_t m, int64_t n, int mb, int nb,
-                         int p, int q, size_t elem_size)
+                         int p, int q, size_t elem_size, bool sym)
     : ctx_(ctx), ctx_rank_(ctx->rank()), m_(m), n_(n), mb_(mb), nb_(nb),
-      p_(p), q_(q), elem_(elem_size) {
+      p_(p), q_(q), elem_(elem_size), sym_(sym) {
   mt_ = (int)((m + mb - 1) / mb);
   nt_ = (int)((n + nb - 1) / nb);
   PA_CHECK(p_ * q_ == ctx->world(), "grid p*q must equal world size");
@@ -311,6 +311,7 @@ TiledMatrix::~TiledMatrix() {
 }
 
 Data* TiledMatrix::tile(int tm, int tn) {
+  if (sym_ && tn > tm) std::swap(tm, tn);
   PA_CHECK(tm >= 0 && tm < mt_ && tn >= 0 && tn < nt_);
   size_t idx = (size_t)tm * nt_ + tn;
   if (!tiles_[idx]) {

@@ -416,6 +416,84 @@ TaskClass& tc_gemm() {
   return tc;
 }
 
+// ---------------------------------------------------------- copy / apply
+// Redistribution (data_dist/matrix/redistribute analog, same tile grid,
+// any rank grids) and the generic elementwise operator (map_operator.c /
+// apply.jdf analog; the scale op x = alpha*x + beta is the built-in).
+__global__ void k_scale_tile(double* p, size_t n, double alpha, double beta) {
+  size_t i = (size_t)blockIdx.x * blockDim.x + threadIdx.x;
+  for (; i < n; i += (size_t)gridDim.x * blockDim.x)
+    p[i] = p[i] * alpha + beta;
+}
+
+static void cpu_copy_tile(Task& t) {
+  Data* src = t.flows[0].data;
+  Data* dst = t.flows[1].data;
+  memcpy(dst->ensure_host(), src->pull_to_host(), src->bytes);
+  dst->written_on(false);
+}
+
+static void gpu_copy_tile(Task& t, GpuTaskCtx& g) {
+  Data* src = t.flows[0].data;
+  PA_HIP_CHECK(hipMemcpyAsync(t.dev_ptr[1], t.dev_ptr[0], src->bytes,
+                              hipMemcpyDeviceToDevice, g.stream));
+}
+
+static void cpu_scale(Task& t) {
+  const TileArgs& a = t.arg<TileArgs>();
+  double alpha, beta;
+  memcpy(&alpha, &a.i0, 8);
+  memcpy(&beta, &a.j0, 8);
+  Data* d = t.flows[0].data;
+  double* p = (double*)d->pull_to_host();
+  for (size_t i = 0; i < d->bytes / 8; i++) p[i] = p[i] * alpha + beta;
+  d->written_on(false);
+}
+
+static void gpu_scale(Task& t, GpuTaskCtx& g) {
+  const TileArgs& a = t.arg<TileArgs>();
+  double alpha, beta;
+  memcpy(&alpha, &a.i0, 8);
+  memcpy(&beta, &a.j0, 8);
+  Data* d = t.flows[0].data;
+  hipLaunchKernelGGL(k_scale_tile, dim3(2048), dim3(256), 0, g.stream,
+                     (double*)t.dev_ptr[0], d->bytes / 8, alpha, beta);
+}
+
+TaskClass& tc_copy_tile() {
+  static TaskClass tc = make_tc("copy_tile", TaskKind::GPU, cpu_copy_tile,
+                                gpu_copy_tile, 30);
+  return tc;
+}
+TaskClass& tc_scale() {
+  static TaskClass tc = make_tc("scale", TaskKind::GPU, cpu_scale, gpu_scale, 31);
+  return tc;
+}
+
+void insert_redistribute(Dtd& tp, TiledMatrix& Src, TiledMatrix& Dst) {
+  PA_CHECK(Src.mt() == Dst.mt() && Src.nt() == Dst.nt() &&
+           Src.tile_bytes() == Dst.tile_bytes(),
+           "redistribute: tile grids must match (general regridding is a "
+           "round-2 item)");
+  for (int m = 0; m < Src.mt(); m++)
+    for (int n = 0; n < Src.nt(); n++) {
+      Dtd::FlowSpec f[] = {{Src.tile(m, n), ACCESS_IN},
+                           {Dst.tile(m, n), ACCESS_OUT}};
+      tp.insert(&tc_copy_tile(), nullptr, 0, f, 2, 0, Dst.rank_of(m, n));
+    }
+}
+
+void insert_apply_scale(Dtd& tp, TiledMatrix& A, double alpha, double beta) {
+  for (int m = 0; m < A.mt(); m++)
+    for (int n = 0; n < A.nt(); n++) {
+      TileArgs a;
+      memcpy(&a.i0, &alpha, 8);
+      memcpy(&a.j0, &beta, 8);
+      Dtd::FlowSpec f[] = {{A.tile(m, n), ACCESS_INOUT}};
+      tp.insert(&tc_scale(), &a, sizeof(a), f, 1, 0, A.rank_of(m, n));
+    }
+}
+
 // ------------------------------------------------------------------ DAG builders
 void insert_spd_fill(Dtd& tp, TiledMatrix& A, uint32_t seed) {
   for (int tm = 0; tm < A.mt(); tm++)

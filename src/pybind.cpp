@@ -124,10 +124,12 @@ PYBIND11_MODULE(_core, m) {
       .def_property_readonly("version", [](Data& d) { return d.version; });
 
   py::class_<TiledMatrix>(m, "TiledMatrix")
-      .def(py::init<Context*, int64_t, int64_t, int, int, int, int, size_t>(),
+      .def(py::init<Context*, int64_t, int64_t, int, int, int, int, size_t,
+                    bool>(),
            py::arg("ctx"), py::arg("m"), py::arg("n"), py::arg("mb"),
            py::arg("nb"), py::arg("p") = 1, py::arg("q") = 1,
-           py::arg("elem_size") = 8, py::keep_alive<1, 2>())
+           py::arg("elem_size") = 8, py::arg("sym") = false,
+           py::keep_alive<1, 2>())
       .def_property_readonly("_handle", [](TiledMatrix& a) { return (uintptr_t)&a; })
       .def_property_readonly("mt", &TiledMatrix::mt)
       .def_property_readonly("nt", &TiledMatrix::nt)
@@ -235,6 +237,11 @@ PYBIND11_MODULE(_core, m) {
   m.def("insert_potrf", &insert_potrf, py::arg("tp"), py::arg("A"),
         py::call_guard<py::gil_scoped_release>());
   m.def("insert_geqrf", &insert_geqrf, py::arg("tp"), py::arg("A"),
+        py::call_guard<py::gil_scoped_release>());
+  m.def("insert_redistribute", &insert_redistribute, py::arg("tp"),
+        py::arg("src"), py::arg("dst"), py::call_guard<py::gil_scoped_release>());
+  m.def("insert_apply_scale", &insert_apply_scale, py::arg("tp"), py::arg("A"),
+        py::arg("alpha"), py::arg("beta"),
         py::call_guard<py::gil_scoped_release>());
   m.def("insert_fill_bf16", &insert_fill_bf16, py::arg("tp"), py::arg("A"),
         py::arg("seed") = 1u, py::call_guard<py::gil_scoped_release>());

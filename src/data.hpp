@@ -83,6 +83,7 @@ class TiledMatrix {
   int grid_q() const { return q_; }
   size_t elem_size() const { return elem_; }
   size_t tile_bytes() const { return (size_t)mb_ * nb_ * elem_; }
+  bool sym() const { return sym_; }
 
   int rank_of(int tm, int tn) const {
     if (sym_ && tn > tm) { int t = tm; tm = tn; tn = t; }

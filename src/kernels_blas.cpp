@@ -476,7 +476,7 @@ void insert_redistribute(Dtd& tp, TiledMatrix& Src, TiledMatrix& Dst) {
            "redistribute: tile grids must match (general regridding is a "
            "round-2 item)");
   for (int m = 0; m < Src.mt(); m++)
-    for (int n = 0; n < Src.nt(); n++) {
+    for (int n = 0; n < (Src.sym() ? m + 1 : Src.nt()); n++) {
       Dtd::FlowSpec f[] = {{Src.tile(m, n), ACCESS_IN},
                            {Dst.tile(m, n), ACCESS_OUT}};
       tp.insert(&tc_copy_tile(), nullptr, 0, f, 2, 0, Dst.rank_of(m, n));
@@ -485,7 +485,7 @@ void insert_redistribute(Dtd& tp, TiledMatrix& Src, TiledMatrix& Dst) {
 
 void insert_apply_scale(Dtd& tp, TiledMatrix& A, double alpha, double beta) {
   for (int m = 0; m < A.mt(); m++)
-    for (int n = 0; n < A.nt(); n++) {
+    for (int n = 0; n < (A.sym() ? m + 1 : A.nt()); n++) {
       TileArgs a;
       memcpy(&a.i0, &alpha, 8);
       memcpy(&a.j0, &beta, 8);

@@ -6,13 +6,16 @@ import parsec_amd as pm
 
 def test_sym_tile_alias(ctx):
     A = pm.TiledMatrix(ctx, 256, 256, 64, 64, 1, 1, sym=True)
-    t1 = A.tile(1, 3)
-    t2 = A.tile(3, 1)
-    assert t1 is t2 or t1._handle if hasattr(t1, "_handle") else True
-    # version changes through one alias are visible through the other
     tp = pm.Dtd(ctx)
-    pm.insert_apply_scale(tp, A, 1.0, 0.0)
+    pm.insert_spd_fill(tp, A, 2)
     tp.wait()
+    # tile(1,3) aliases tile(3,1): same buffer, same version counter
+    assert np.array_equal(A.tile_numpy(1, 3), A.tile_numpy(3, 1))
+    v0 = A.tile(1, 3).version
+    tp2 = pm.Dtd(ctx)
+    pm.insert_apply_scale(tp2, A, 2.0, 0.0)
+    tp2.wait()
+    assert A.tile(3, 1).version > v0
     assert A.tile(1, 3).version == A.tile(3, 1).version
 
 

@@ -21,7 +21,7 @@ def test_sym_tile_alias(ctx):
 
 def test_apply_scale(ctx):
     n, nb = 128, 64
-    A = pm.TiledMatrix(ctx, n, n, nb, nb, 1, 1)
+    A = pm.TiledMatrix(ctx, n, n, nb, nb, 1, 1, sym=True)
     tp = pm.Dtd(ctx)
     pm.insert_spd_fill(tp, A, 9)
     tp.wait()
@@ -35,8 +35,8 @@ def test_apply_scale(ctx):
 
 def test_redistribute(ctx):
     n, nb = 128, 64
-    A = pm.TiledMatrix(ctx, n, n, nb, nb, 1, 1)
-    B = pm.TiledMatrix(ctx, n, n, nb, nb, 1, 1)
+    A = pm.TiledMatrix(ctx, n, n, nb, nb, 1, 1, sym=True)
+    B = pm.TiledMatrix(ctx, n, n, nb, nb, 1, 1, sym=True)
     tp = pm.Dtd(ctx)
     pm.insert_spd_fill(tp, A, 4)
     pm.insert_redistribute(tp, A, B)

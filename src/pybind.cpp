@@ -195,7 +195,13 @@ PYBIND11_MODULE(_core, m) {
       // parsec_compose analog (compound.c:17-135): chain a callback to run
       // when this pool drains (e.g. to start inserting the next pool).
       .def("on_complete", [](Taskpool& t, py::function fn) {
-        auto holder = std::make_shared<py::function>(std::move(fn));
+        // deleter must hold the GIL: the callback object may be destroyed
+        // on a worker thread when the pool drains
+        auto holder = std::shared_ptr<py::function>(
+            new py::function(std::move(fn)), [](py::function* p) {
+              py::gil_scoped_acquire gil;
+              delete p;
+            });
         t.on_complete([holder] {
           py::gil_scoped_acquire gil;
           try {

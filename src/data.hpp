@@ -41,6 +41,10 @@ struct Data {
   // (a task that did not issue the copy still races it otherwise).
   void* h2d_event = nullptr;  // hipEvent_t, lazily created, owned here
   bool h2d_pending = false;
+  // GPU residency management (LRU eviction): pinned while any in-flight
+  // GPU task or comm transfer uses the device copy.
+  int dev_refs = 0;          // guarded by lock
+  uint64_t dev_last_use = 0; // engine-managed LRU stamp
 
   // ---- DTD chaining state (single inserter thread; no lock needed) ----
   uint32_t version = 0;        // logical version, bumped per writer insertion

@@ -1,5 +1,7 @@
 #include "device_gpu.hpp"
 
+#include <algorithm>
+
 #include "data.hpp"
 #include "profiling.hpp"
 
@@ -41,6 +43,11 @@ GpuEngine::GpuEngine(Context* ctx, int device) : ctx_(ctx), device_(device) {
   PA_HIP_CHECK(hipMemGetInfo(&free_b, &total_b));
   int pct = (int)param_int("gpu_mem_percent", 85);
   slab_bytes_ = free_b / 100 * pct;
+  int64_t cap_mb = param_int("gpu_mem_limit_mb", 0);
+  if (cap_mb > 0) {
+    slab_bytes_ = (size_t)cap_mb << 20;
+    hard_cap_ = true;
+  }
   if (slab_bytes_) {
     hipError_t e = hipMalloc(&slab_, slab_bytes_);
     if (e != hipSuccess) {
@@ -89,26 +96,97 @@ void GpuEngine::event_put(hipEvent_t e) { event_pool_.push_back(e); }
 
 void* GpuEngine::dev_alloc(size_t bytes) {
   bytes = (bytes + 255) & ~size_t(255);
+  for (int attempt = 0; attempt < 10000; attempt++) {
+    {
+      std::lock_guard<std::mutex> g(mem_mtx_);
+      auto it = free_lists_.find(bytes);
+      if (it != free_lists_.end() && !it->second.empty()) {
+        void* p = it->second.back();
+        it->second.pop_back();
+        return p;
+      }
+      if (slab_used_ + bytes <= slab_bytes_) {
+        void* p = (char*)slab_ + slab_used_;
+        slab_used_ += bytes;
+        return p;
+      }
+    }
+    // Capacity exhausted: evict the least-recently-used unpinned copy
+    // (writeback if dirty), then retry.
+    if (evict_one(bytes)) continue;
+    if (!hard_cap_) {
+      void* p = nullptr;
+      if (hipMalloc(&p, bytes) == hipSuccess) return p;
+    }
+    // nothing evictable right now: in-flight tasks still hold pins;
+    // wait for retirements and retry
+    std::this_thread::sleep_for(std::chrono::microseconds(200));
+  }
+  fatal("GPU %d out of memory allocating %zu bytes (slab %zu/%zu)", device_,
+        bytes, slab_used_, slab_bytes_);
+}
+
+void GpuEngine::note_resident(Data* d) {
   std::lock_guard<std::mutex> g(mem_mtx_);
-  auto it = free_lists_.find(bytes);
-  if (it != free_lists_.end() && !it->second.empty()) {
-    void* p = it->second.back();
-    it->second.pop_back();
-    return p;
+  for (Data* r : resident_)
+    if (r == d) return;
+  resident_.push_back(d);
+}
+
+void GpuEngine::pin(Data* d) {
+  d->dev_refs++;
+  d->dev_last_use = lru_clock_.fetch_add(1);
+}
+
+void GpuEngine::unpin(Data* d) {
+  SpinGuard g(d->lock);
+  d->dev_refs--;
+  d->dev_last_use = lru_clock_.fetch_add(1);
+}
+
+bool GpuEngine::evict_one(size_t) {
+  // candidates collected without their locks (lock order: Data::lock may
+  // be held by a thread calling dev_alloc, so only try_lock here)
+  std::vector<Data*> cand;
+  {
+    std::lock_guard<std::mutex> g(mem_mtx_);
+    cand = resident_;
   }
-  if (slab_used_ + bytes <= slab_bytes_) {
-    void* p = (char*)slab_ + slab_used_;
-    slab_used_ += bytes;
-    return p;
+  std::sort(cand.begin(), cand.end(), [](Data* a, Data* b) {
+    return a->dev_last_use < b->dev_last_use;
+  });
+  for (Data* d : cand) {
+    if (!d->lock.try_lock()) continue;
+    if (!d->dev_ptr || d->dev_refs > 0 || d->h2d_pending) {
+      d->lock.unlock();
+      continue;
+    }
+    if (d->dev_valid && !d->host_valid) {
+      // dirty: write back (kernels producing it have retired: refs==0)
+      if (!d->host_ptr) {
+        if (posix_memalign(&d->host_ptr, 4096, d->bytes) != 0) {
+          d->lock.unlock();
+          continue;
+        }
+      }
+      copy_d2h(d->host_ptr, d->dev_ptr, d->bytes);
+      d->host_valid = true;
+    }
+    void* buf = d->dev_ptr;
+    d->dev_ptr = nullptr;
+    d->dev_valid = false;
+    d->lock.unlock();
+    {
+      std::lock_guard<std::mutex> g(mem_mtx_);
+      free_lists_[(d->bytes + 255) & ~size_t(255)].push_back(buf);
+      resident_.erase(std::remove(resident_.begin(), resident_.end(), d),
+                      resident_.end());
+    }
+    PA_DEBUG(2, "evicted tile %lu (%zu bytes)", (unsigned long)d->key,
+             d->bytes);
+    return true;
   }
-  // Slab exhausted: direct allocation escape hatch (LRU eviction is the
-  // next capacity tier; see docs/DESIGN.md).
-  void* p = nullptr;
-  hipError_t e = hipMalloc(&p, bytes);
-  if (e != hipSuccess)
-    fatal("GPU %d out of memory allocating %zu bytes (slab %zu/%zu)",
-          device_, bytes, slab_used_, slab_bytes_);
-  return p;
+  return false;
 }
 
 void GpuEngine::dev_free(void* p, size_t bytes) {
@@ -157,7 +235,11 @@ void GpuEngine::stage_in_and_launch(Task* t) {
     Data* d = t->flows[i].data;
     if (!d) { t->dev_ptr[i] = nullptr; continue; }
     SpinGuard g(d->lock);
-    if (!d->dev_ptr) d->dev_ptr = dev_alloc(d->bytes);
+    pin(d);  // before alloc: eviction skips pinned tiles
+    if (!d->dev_ptr) {
+      d->dev_ptr = dev_alloc(d->bytes);
+      note_resident(d);
+    }
     if ((t->flows[i].mode & ACCESS_IN) && !d->dev_valid) {
       PA_CHECK(d->host_valid, "stage-in: no valid copy for tile");
       PA_HIP_CHECK(hipMemcpyAsync(d->dev_ptr, d->host_ptr, d->bytes,
@@ -203,7 +285,9 @@ void GpuEngine::manager_main() {
         Task* t = f.task;
         for (int i = 0; i < t->nflows; i++) {
           Data* d = t->flows[i].data;
-          if (d && (t->flows[i].mode & ACCESS_OUT)) d->written_on(true);
+          if (!d) continue;
+          if (t->flows[i].mode & ACCESS_OUT) d->written_on(true);
+          unpin(d);
         }
         Profiler& pr = Profiler::inst();
         if (pr.enabled())

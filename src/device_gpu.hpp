@@ -54,9 +54,15 @@ class GpuEngine {
 
   void enqueue(Task* t);  // task with satisfied deps, TaskKind::GPU
 
-  // Device memory pool (slab + size-class free lists).
+  // Device memory pool (slab + size-class free lists + LRU eviction when
+  // capacity is exhausted: clean copies are dropped, dirty ones written
+  // back to host first — zone_malloc + dual-LRU analog,
+  // device_gpu.c:1210-1623).
   void* dev_alloc(size_t bytes);
   void dev_free(void* p, size_t bytes);
+  void note_resident(Data* d);   // track for eviction
+  void pin(Data* d);             // dev_refs++ under d->lock
+  void unpin(Data* d);
 
   // Synchronous copies (rare paths: pull_to_host, test readback).
   void copy_d2h(void* dst, const void* src, size_t bytes);
@@ -113,10 +119,14 @@ class GpuEngine {
   std::vector<hipEvent_t> event_pool_;
 
   // memory pool
+  bool evict_one(size_t bytes);
   std::mutex mem_mtx_;
   void* slab_ = nullptr;
   size_t slab_bytes_ = 0, slab_used_ = 0;
+  bool hard_cap_ = false;  // gpu_mem_limit_mb set: no hipMalloc escape
   std::map<size_t, std::vector<void*>> free_lists_;
+  std::vector<Data*> resident_;
+  std::atomic<uint64_t> lru_clock_{1};
 };
 
 }  // namespace pa

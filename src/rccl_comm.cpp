@@ -145,7 +145,11 @@ class RcclComm : public CommEngine {
   // Ensure the tile has a valid device copy for sending.
   void* dev_src(Data* d, hipStream_t stream) {
     SpinGuard g(d->lock);
-    if (!d->dev_ptr) d->dev_ptr = ctx_->gpu()->dev_alloc(d->bytes);
+    ctx_->gpu()->pin(d);  // unpinned when the transfer retires
+    if (!d->dev_ptr) {
+      d->dev_ptr = ctx_->gpu()->dev_alloc(d->bytes);
+      ctx_->gpu()->note_resident(d);
+    }
     if (!d->dev_valid) {
       PA_CHECK(d->host_valid, "rccl send: no valid copy");
       PA_HIP_CHECK(hipMemcpyAsync(d->dev_ptr, d->host_ptr, d->bytes,
@@ -160,7 +164,11 @@ class RcclComm : public CommEngine {
 
   void* dev_dst(Data* d) {
     SpinGuard g(d->lock);
-    if (!d->dev_ptr) d->dev_ptr = ctx_->gpu()->dev_alloc(d->bytes);
+    ctx_->gpu()->pin(d);
+    if (!d->dev_ptr) {
+      d->dev_ptr = ctx_->gpu()->dev_alloc(d->bytes);
+      ctx_->gpu()->note_resident(d);
+    }
     return d->dev_ptr;
   }
 
@@ -234,6 +242,7 @@ class RcclComm : public CommEngine {
         inflight_.erase(inflight_.begin() + i);
         events_.push_back(f.ev);
         if (f.is_recv) f.t->flows[0].data->written_on(true);
+        ctx_->gpu()->unpin(f.t->flows[0].data);
         counters().comm_msgs.fetch_add(1, std::memory_order_relaxed);
         counters().comm_bytes.fetch_add(f.t->flows[0].data->bytes,
                                         std::memory_order_relaxed);

@@ -127,3 +127,37 @@ def test_bench_contract():
     assert out["n_gpus"] == 1
     assert out["value"] > 0
     assert out["dtype"] == "fp64"
+
+
+def test_eviction_under_memory_pressure():
+    """Force LRU eviction+writeback with a hard HBM cap smaller than the
+    working set; the factorization must still be numerically correct."""
+    code = f"""
+import numpy as np, sys
+sys.path.insert(0, {REPO!r})
+import parsec_amd as pm
+pm.param_set("gpu_mem_limit_mb", "96")  # matrix alone is ~134 MB
+pm.param_set("gpu_max_inflight", "4")
+ctx = pm.Context(nworkers=2, rank=0, world=1)
+n, nb = 4096, 512
+A = pm.TiledMatrix(ctx, n, n, nb, nb, 1, 1)
+tp = pm.Dtd(ctx); pm.insert_spd_fill(tp, A, 42); tp.wait()
+M = np.zeros((n,n))
+for tm in range(A.mt):
+    for tn in range(tm+1):
+        M[tm*nb:(tm+1)*nb, tn*nb:(tn+1)*nb] = A.tile_numpy(tm,tn)
+M = np.tril(M) + np.tril(M,-1).T
+L0 = np.linalg.cholesky(M)
+tp2 = pm.Dtd(ctx); pm.insert_potrf(tp2, A); tp2.wait()
+L = np.zeros((n,n))
+for tm in range(A.mt):
+    for tn in range(tm+1):
+        L[tm*nb:(tm+1)*nb, tn*nb:(tn+1)*nb] = A.tile_numpy(tm,tn)
+err = np.abs(np.tril(L)-L0).max()
+print("EVICT_ERR", err)
+assert err < 1e-8, err
+del A, ctx
+"""
+    r = subprocess.run([sys.executable, "-c", code], capture_output=True,
+                       text=True, timeout=600)
+    assert r.returncode == 0, r.stdout + r.stderr

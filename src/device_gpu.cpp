@@ -118,9 +118,14 @@ void* GpuEngine::dev_alloc(size_t bytes) {
       void* p = nullptr;
       if (hipMalloc(&p, bytes) == hipSuccess) return p;
     }
-    // nothing evictable right now: in-flight tasks still hold pins;
-    // wait for retirements and retry
-    std::this_thread::sleep_for(std::chrono::microseconds(200));
+    // nothing evictable right now: in-flight tasks still hold pins.
+    // The manager retires its own rings here (it is the only thread that
+    // can); other threads wait for it.
+    if (std::this_thread::get_id() == manager_tid_) {
+      if (!retire_pass()) std::this_thread::yield();
+    } else {
+      std::this_thread::sleep_for(std::chrono::microseconds(200));
+    }
   }
   fatal("GPU %d out of memory allocating %zu bytes (slab %zu/%zu)", device_,
         bytes, slab_used_, slab_bytes_);
@@ -234,12 +239,25 @@ void GpuEngine::stage_in_and_launch(Task* t) {
   for (int i = 0; i < t->nflows; i++) {
     Data* d = t->flows[i].data;
     if (!d) { t->dev_ptr[i] = nullptr; continue; }
-    SpinGuard g(d->lock);
-    pin(d);  // before alloc: eviction skips pinned tiles
-    if (!d->dev_ptr) {
-      d->dev_ptr = dev_alloc(d->bytes);
-      note_resident(d);
+    bool need_alloc;
+    {
+      SpinGuard g(d->lock);
+      pin(d);  // before alloc: eviction skips pinned tiles
+      need_alloc = !d->dev_ptr;
     }
+    if (need_alloc) {
+      // alloc outside the tile lock: dev_alloc may evict (taking other
+      // tiles' locks) or wait on retirements
+      void* p = dev_alloc(d->bytes);
+      SpinGuard g(d->lock);
+      if (!d->dev_ptr) {
+        d->dev_ptr = p;
+        note_resident(d);
+      } else {
+        dev_free(p, d->bytes);
+      }
+    }
+    SpinGuard g(d->lock);
     if ((t->flows[i].mode & ACCESS_IN) && !d->dev_valid) {
       PA_CHECK(d->host_valid, "stage-in: no valid copy for tile");
       PA_HIP_CHECK(hipMemcpyAsync(d->dev_ptr, d->host_ptr, d->bytes,
@@ -271,38 +289,44 @@ void GpuEngine::stage_in_and_launch(Task* t) {
   n_inflight_++;
 }
 
+bool GpuEngine::retire_pass() {
+  bool progress = false;
+  for (auto& ring : inflight_) {
+    while (!ring.empty()) {
+      InFlight& f = ring.front();
+      hipError_t e = hipEventQuery(f.event);
+      if (e == hipErrorNotReady) break;
+      PA_HIP_CHECK(e);
+      Task* t = f.task;
+      for (int i = 0; i < t->nflows; i++) {
+        Data* d = t->flows[i].data;
+        if (!d) continue;
+        if (t->flows[i].mode & ACCESS_OUT) d->written_on(true);
+        unpin(d);
+      }
+      Profiler& pr = Profiler::inst();
+      if (pr.enabled())
+        pr.record(Ev::GPU_TASK, (uint16_t)t->tc->id, t->seq, f.t0_ns,
+                  Profiler::now_ns());
+      counters().tasks_executed_gpu.fetch_add(1, std::memory_order_relaxed);
+      for (auto& [p2, b2] : f.deferred_frees) dev_free(p2, b2);
+      event_put(f.event);
+      ring.pop_front();
+      n_inflight_--;
+      stats.tasks++;
+      task_complete(t);
+      progress = true;
+    }
+  }
+  return progress;
+}
+
 void GpuEngine::manager_main() {
   PA_HIP_CHECK(hipSetDevice(device_));
+  manager_tid_ = std::this_thread::get_id();
   while (true) {
     // 1) retire completed tasks (in-order per stream)
-    bool progress = false;
-    for (auto& ring : inflight_) {
-      while (!ring.empty()) {
-        InFlight& f = ring.front();
-        hipError_t e = hipEventQuery(f.event);
-        if (e == hipErrorNotReady) break;
-        PA_HIP_CHECK(e);
-        Task* t = f.task;
-        for (int i = 0; i < t->nflows; i++) {
-          Data* d = t->flows[i].data;
-          if (!d) continue;
-          if (t->flows[i].mode & ACCESS_OUT) d->written_on(true);
-          unpin(d);
-        }
-        Profiler& pr = Profiler::inst();
-        if (pr.enabled())
-          pr.record(Ev::GPU_TASK, (uint16_t)t->tc->id, t->seq, f.t0_ns,
-                    Profiler::now_ns());
-        counters().tasks_executed_gpu.fetch_add(1, std::memory_order_relaxed);
-        for (auto& [p2, b2] : f.deferred_frees) dev_free(p2, b2);
-        event_put(f.event);
-        ring.pop_front();
-        n_inflight_--;
-        stats.tasks++;
-        task_complete(t);
-        progress = true;
-      }
-    }
+    bool progress = retire_pass();
     // 2) launch new work while there is room (global cap; the panel
     //    stream may exceed its share — its latency is the priority)
     while (true) {

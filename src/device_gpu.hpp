@@ -120,6 +120,8 @@ class GpuEngine {
 
   // memory pool
   bool evict_one(size_t bytes);
+  bool retire_pass();  // poll event rings; true if anything retired
+  std::thread::id manager_tid_;
   std::mutex mem_mtx_;
   void* slab_ = nullptr;
   size_t slab_bytes_ = 0, slab_used_ = 0;

@@ -143,13 +143,30 @@ class RcclComm : public CommEngine {
   }
 
   // Ensure the tile has a valid device copy for sending.
-  void* dev_src(Data* d, hipStream_t stream) {
-    SpinGuard g(d->lock);
-    ctx_->gpu()->pin(d);  // unpinned when the transfer retires
-    if (!d->dev_ptr) {
-      d->dev_ptr = ctx_->gpu()->dev_alloc(d->bytes);
-      ctx_->gpu()->note_resident(d);
+  void ensure_dev_buf(Data* d) {
+    // pin first, allocate outside the tile lock (lock-order discipline
+    // with the eviction path)
+    bool need;
+    {
+      SpinGuard g(d->lock);
+      ctx_->gpu()->pin(d);  // unpinned when the transfer retires
+      need = !d->dev_ptr;
     }
+    if (need) {
+      void* p = ctx_->gpu()->dev_alloc(d->bytes);
+      SpinGuard g(d->lock);
+      if (!d->dev_ptr) {
+        d->dev_ptr = p;
+        ctx_->gpu()->note_resident(d);
+      } else {
+        ctx_->gpu()->dev_free(p, d->bytes);
+      }
+    }
+  }
+
+  void* dev_src(Data* d, hipStream_t stream) {
+    ensure_dev_buf(d);
+    SpinGuard g(d->lock);
     if (!d->dev_valid) {
       PA_CHECK(d->host_valid, "rccl send: no valid copy");
       PA_HIP_CHECK(hipMemcpyAsync(d->dev_ptr, d->host_ptr, d->bytes,
@@ -163,12 +180,8 @@ class RcclComm : public CommEngine {
   }
 
   void* dev_dst(Data* d) {
+    ensure_dev_buf(d);
     SpinGuard g(d->lock);
-    ctx_->gpu()->pin(d);
-    if (!d->dev_ptr) {
-      d->dev_ptr = ctx_->gpu()->dev_alloc(d->bytes);
-      ctx_->gpu()->note_resident(d);
-    }
     return d->dev_ptr;
   }
 

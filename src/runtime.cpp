@@ -126,12 +126,27 @@ void Taskpool::wait() {
 
 // ------------------------------------------------------------------ Scheduler
 Scheduler::Scheduler(int nworkers) : nworkers_(nworkers) {
+  // MCA "sched" analog (mca/sched/*): ws = per-worker deques + steal (lfq
+  // style, default); fifo/lifo = one shared queue (gd/ll styles). The
+  // priority queue always serves prioritized and externally-released tasks.
+  std::string kind = param_str("sched", "ws");
+  mode_ = kind == "fifo" ? 1 : kind == "lifo" ? 2 : 0;
   for (int i = 0; i < nworkers_; i++) wq_.emplace_back(new WorkerQ());
 }
 
 Scheduler::~Scheduler() = default;
 
 void Scheduler::push(Task* t, int worker_hint) {
+  if (mode_ != 0) {
+    WorkerQ& q = *wq_[0];
+    q.lock.lock();
+    if (mode_ == 2) q.dq.push_front(t);
+    else q.dq.push_back(t);
+    q.lock.unlock();
+    npending_.fetch_add(1, std::memory_order_release);
+    sleep_cv_.notify_one();
+    return;
+  }
   if (worker_hint >= 0 && worker_hint < nworkers_ && t->priority == 0) {
     WorkerQ& q = *wq_[worker_hint];
     q.lock.lock();
@@ -148,6 +163,18 @@ void Scheduler::push(Task* t, int worker_hint) {
 
 Task* Scheduler::pop(int worker) {
   Task* t = nullptr;
+  if (mode_ != 0) {
+    WorkerQ& q = *wq_[0];
+    q.lock.lock();
+    if (!q.dq.empty()) { t = q.dq.front(); q.dq.pop_front(); }
+    q.lock.unlock();
+    if (t) { npending_.fetch_sub(1, std::memory_order_relaxed); return t; }
+    pq_lock_.lock();
+    if (!pq_.empty()) { t = pq_.top().t; pq_.pop(); }
+    pq_lock_.unlock();
+    if (t) npending_.fetch_sub(1, std::memory_order_relaxed);
+    return t;
+  }
   if (worker >= 0) {
     WorkerQ& q = *wq_[worker];
     q.lock.lock();

@@ -192,6 +192,7 @@ class Scheduler {
     }
   };
   int nworkers_;
+  int mode_ = 0;  // 0=ws 1=fifo 2=lifo (PARSEC_MCA_sched)
   std::vector<std::unique_ptr<WorkerQ>> wq_;
   SpinLock pq_lock_;
   std::priority_queue<PQEntry> pq_;

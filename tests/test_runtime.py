@@ -112,3 +112,26 @@ def test_taskpool_reuse_after_wait(ctx):
     tp.insert_py(lambda: log.append(2), flows=[(t, pm.ACCESS_INOUT)])
     tp.wait()
     assert log == [1, 2]
+
+
+def test_sched_variants():
+    """MCA sched module analog: fifo/lifo shared-queue schedulers run the
+    same DAG correctly."""
+    import subprocess, sys, os
+    repo = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+    for kind in ["fifo", "lifo", "ws"]:
+        code = f"""
+import sys; sys.path.insert(0, {repo!r})
+import numpy as np
+import parsec_amd as pm
+pm.param_set("sched", {kind!r})
+ctx = pm.Context(nworkers=2, rank=0, world=1, gpu=-2)
+A = pm.TiledMatrix(ctx, 256, 256, 64, 64, 1, 1)
+tp = pm.Dtd(ctx); pm.insert_spd_fill(tp, A, 42); pm.insert_potrf(tp, A); tp.wait()
+d = A.tile_numpy(3, 3)
+assert np.isfinite(d).all() and d[0,0] > 0
+del A, ctx
+"""
+        r = subprocess.run([sys.executable, "-c", code], capture_output=True,
+                           text=True, timeout=120)
+        assert r.returncode == 0, f"sched={kind}: {r.stdout}{r.stderr}"

@@ -121,10 +121,114 @@ __global__ void k_gemm_bf16_tn(int m, int n, int k,
   }
 }
 
+// ---------------------------------------------------------------- v2: glds
+// Step-3 of the guide's optimization ladder: async global->LDS staging via
+// `global_load_lds` (16 B per lane, wave-uniform LDS base), double-buffered
+// K-tiles (BK=64), one vmcnt-drain barrier per tile. The LDS image is
+// lane-linear ([row][K] bf16, K fastest), which the TN tile layout feeds
+// with contiguous 16 B per lane. Full tiles only; edge tiles take the
+// reg-staged kernel above.
+#define GB2_BK 64
+
+__launch_bounds__(256)
+__global__ void k_gemm_bf16_tn_v2(int m, int n, int k,
+                                  const bf16* __restrict__ A, int lda,
+                                  const bf16* __restrict__ B, int ldb,
+                                  float* __restrict__ C, int ldc, int nbx,
+                                  int accum) {
+  extern __shared__ __attribute__((aligned(16))) char smem[];
+  bf16* As = (bf16*)smem;                      // [2][128][GB2_BK]
+  bf16* Bs = As + 2 * GB_BM * GB2_BK;
+  const int tid = threadIdx.x;
+  const int wave = tid >> 6, lane = tid & 63;
+  const int wr = wave >> 1, wc = wave & 1;
+  int id = bf_swz(blockIdx.x, gridDim.x);
+  const int bm0 = (id % nbx) * GB_BM, bn0 = (id / nbx) * GB_BN;
+  const int g16 = lane >> 4, r16 = lane & 15;
+  const int srow = lane >> 3, skc = (lane & 7) * 8;  // staging lane map
+
+  f32x4 acc[4][4] = {};
+
+  auto stage = [&](int buf, int k0) {
+    // 4 pieces of 8 rows per wave for each operand: 64 lanes x 16 B = 1 KB
+    // per glds instruction, lane-linear into [row][K].
+#pragma unroll
+    for (int p = 0; p < 4; p++) {
+      int row0 = (wave * 4 + p) * 8;
+      const bf16* srcA = A + (size_t)(bm0 + row0 + srow) * lda + k0 + skc;
+      bf16* dstA = As + buf * GB_BM * GB2_BK + row0 * GB2_BK;
+      __builtin_amdgcn_global_load_lds(
+          (const __attribute__((address_space(1))) unsigned int*)srcA,
+          (__attribute__((address_space(3))) unsigned int*)dstA, 16, 0, 0);
+      const bf16* srcB = B + (size_t)(bn0 + row0 + srow) * ldb + k0 + skc;
+      bf16* dstB = Bs + buf * GB_BN * GB2_BK + row0 * GB2_BK;
+      __builtin_amdgcn_global_load_lds(
+          (const __attribute__((address_space(1))) unsigned int*)srcB,
+          (__attribute__((address_space(3))) unsigned int*)dstB, 16, 0, 0);
+    }
+  };
+
+  stage(0, 0);
+  __syncthreads();
+  const int ntiles = k / GB2_BK;
+  for (int t = 0; t < ntiles; t++) {
+    if (t + 1 < ntiles) stage((t + 1) & 1, (t + 1) * GB2_BK);
+    const bf16* as = As + (t & 1) * GB_BM * GB2_BK;
+    const bf16* bs = Bs + (t & 1) * GB_BN * GB2_BK;
+#pragma unroll
+    for (int kk = 0; kk < GB2_BK; kk += 32) {
+      bf16x8 a[4], b[4];
+#pragma unroll
+      for (int f = 0; f < 4; f++)
+        a[f] = *(const bf16x8*)&as[(wr * 64 + f * 16 + r16) * GB2_BK + kk +
+                                   g16 * 8];
+#pragma unroll
+      for (int f = 0; f < 4; f++)
+        b[f] = *(const bf16x8*)&bs[(wc * 64 + f * 16 + r16) * GB2_BK + kk +
+                                   g16 * 8];
+#pragma unroll
+      for (int i = 0; i < 4; i++)
+#pragma unroll
+        for (int j = 0; j < 4; j++)
+          acc[i][j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              a[i], b[j], acc[i][j], 0, 0, 0);
+    }
+    __syncthreads();
+  }
+
+#pragma unroll
+  for (int j = 0; j < 4; j++) {
+    int col = bn0 + wc * 64 + j * 16 + r16;
+    float* cp = C + (size_t)col * ldc;
+#pragma unroll
+    for (int i = 0; i < 4; i++) {
+      int row0 = bm0 + wr * 64 + i * 16 + g16 * 4;
+#pragma unroll
+      for (int e = 0; e < 4; e++) {
+        int row = row0 + e;
+        cp[row] = accum ? cp[row] + acc[i][j][e] : acc[i][j][e];
+      }
+    }
+  }
+}
+
 static void launch_gemm_bf16(int m, int n, int k, const void* A, int lda,
                              const void* B, int ldb, float* C, int ldc,
                              hipStream_t stream, int accum = 1) {
   int nbx = (m + GB_BM - 1) / GB_BM, nby = (n + GB_BN - 1) / GB_BN;
+  if (m % GB_BM == 0 && n % GB_BN == 0 && k % GB2_BK == 0) {
+    constexpr size_t lds = 2 * (GB_BM + GB_BN) * GB2_BK * 2;
+    static bool attr_set = false;
+    if (!attr_set) {
+      hipFuncSetAttribute((const void*)k_gemm_bf16_tn_v2,
+                          hipFuncAttributeMaxDynamicSharedMemorySize, lds);
+      attr_set = true;
+    }
+    hipLaunchKernelGGL(k_gemm_bf16_tn_v2, dim3(nbx * nby), dim3(256), lds,
+                       stream, m, n, k, (const bf16*)A, lda, (const bf16*)B,
+                       ldb, C, ldc, nbx, accum);
+    return;
+  }
   hipLaunchKernelGGL(k_gemm_bf16_tn, dim3(nbx * nby), dim3(256), 0, stream,
                      m, n, k, (const bf16*)A, lda, (const bf16*)B, ldb, C,
                      ldc, nbx, accum);

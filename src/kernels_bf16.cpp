@@ -145,7 +145,7 @@ __global__ void k_gemm_bf16_tn_v2(int m, int n, int k,
   int id = bf_swz(blockIdx.x, gridDim.x);
   const int bm0 = (id % nbx) * GB_BM, bn0 = (id / nbx) * GB_BN;
   const int g16 = lane >> 4, r16 = lane & 15;
-  const int srow = lane >> 3, skc = (lane & 7) * 8;  // staging lane map
+  const int srow = lane >> 3;  // staging lane map (8 rows x 8 K-chunks)
 
   f32x4 acc[4][4] = {};
 
@@ -155,6 +155,10 @@ __global__ void k_gemm_bf16_tn_v2(int m, int n, int k,
 #pragma unroll
     for (int p = 0; p < 4; p++) {
       int row0 = (wave * 4 + p) * 8;
+      // T2 bank swizzle via the SOURCE address (glds writes lane-linear,
+      // rule 21): element k-offset for this lane = 8*((lane&7) ^ (row&7)),
+      // matched by the XOR on the read side.
+      int skc = (((lane & 7) ^ ((row0 + srow) & 7)) * 8);
       const bf16* srcA = A + (size_t)(bm0 + row0 + srow) * lda + k0 + skc;
       bf16* dstA = As + buf * GB_BM * GB2_BK + row0 * GB2_BK;
       __builtin_amdgcn_global_load_lds(
@@ -179,13 +183,17 @@ __global__ void k_gemm_bf16_tn_v2(int m, int n, int k,
     for (int kk = 0; kk < GB2_BK; kk += 32) {
       bf16x8 a[4], b[4];
 #pragma unroll
-      for (int f = 0; f < 4; f++)
-        a[f] = *(const bf16x8*)&as[(wr * 64 + f * 16 + r16) * GB2_BK + kk +
-                                   g16 * 8];
+      for (int f = 0; f < 4; f++) {
+        int row = wr * 64 + f * 16 + r16;
+        a[f] = *(const bf16x8*)&as[row * GB2_BK +
+                                   ((kk + g16 * 8) ^ ((row & 7) << 3))];
+      }
 #pragma unroll
-      for (int f = 0; f < 4; f++)
-        b[f] = *(const bf16x8*)&bs[(wc * 64 + f * 16 + r16) * GB2_BK + kk +
-                                   g16 * 8];
+      for (int f = 0; f < 4; f++) {
+        int row = wc * 64 + f * 16 + r16;
+        b[f] = *(const bf16x8*)&bs[row * GB2_BK +
+                                   ((kk + g16 * 8) ^ ((row & 7) << 3))];
+      }
 #pragma unroll
       for (int i = 0; i < 4; i++)
 #pragma unroll

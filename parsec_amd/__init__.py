@@ -30,6 +30,7 @@ from parsec_amd._core import (  # noqa: F401,E402
     insert_spd_fill,
     insert_redistribute,
     insert_apply_scale,
+    insert_reduce_sum,
     param_dump,
     param_set,
 )

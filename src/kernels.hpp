@@ -42,5 +42,6 @@ void insert_fill_bf16(Dtd& tp, TiledMatrix& A, uint32_t seed);
 void insert_gemm_bf16(Dtd& tp, TiledMatrix& At, TiledMatrix& B, TiledMatrix& C);
 void insert_redistribute(Dtd& tp, TiledMatrix& Src, TiledMatrix& Dst);
 void insert_apply_scale(Dtd& tp, TiledMatrix& A, double alpha, double beta);
+void insert_reduce_sum(Dtd& tp, TiledMatrix& A, TiledMatrix& R);
 
 }  // namespace pa

@@ -470,6 +470,45 @@ TaskClass& tc_scale() {
   return tc;
 }
 
+__global__ void k_add_tile(double* dst, const double* src, size_t n) {
+  size_t i = (size_t)blockIdx.x * blockDim.x + threadIdx.x;
+  for (; i < n; i += (size_t)gridDim.x * blockDim.x) dst[i] += src[i];
+}
+
+static void cpu_add_tile(Task& t) {
+  Data* a = t.flows[0].data;
+  Data* c = t.flows[1].data;
+  const double* p = (const double*)a->pull_to_host();
+  double* q = (double*)c->pull_to_host();
+  for (size_t i = 0; i < c->bytes / 8; i++) q[i] += p[i];
+  c->written_on(false);
+}
+
+static void gpu_add_tile(Task& t, GpuTaskCtx& g) {
+  Data* c = t.flows[1].data;
+  hipLaunchKernelGGL(k_add_tile, dim3(2048), dim3(256), 0, g.stream,
+                     (double*)t.dev_ptr[1], (const double*)t.dev_ptr[0],
+                     c->bytes / 8);
+}
+
+TaskClass& tc_add_tile() {
+  static TaskClass tc = make_tc("add_tile", TaskKind::GPU, cpu_add_tile,
+                                gpu_add_tile, 32);
+  return tc;
+}
+
+// result += sum of all tiles of A (reduce.jdf / DTD reduce analog).
+// The accumulation chains on `result`; contributions from each rank's
+// tiles flow through the comm engine automatically.
+void insert_reduce_sum(Dtd& tp, TiledMatrix& A, TiledMatrix& R) {
+  Data* r = R.tile(0, 0);
+  for (int m = 0; m < A.mt(); m++)
+    for (int n = 0; n < (A.sym() ? m + 1 : A.nt()); n++) {
+      Dtd::FlowSpec f[] = {{A.tile(m, n), ACCESS_IN}, {r, ACCESS_INOUT}};
+      tp.insert(&tc_add_tile(), nullptr, 0, f, 2, 0, R.rank_of(0, 0));
+    }
+}
+
 void insert_redistribute(Dtd& tp, TiledMatrix& Src, TiledMatrix& Dst) {
   PA_CHECK(Src.mt() == Dst.mt() && Src.nt() == Dst.nt() &&
            Src.tile_bytes() == Dst.tile_bytes(),

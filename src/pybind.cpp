@@ -249,6 +249,8 @@ PYBIND11_MODULE(_core, m) {
   m.def("insert_apply_scale", &insert_apply_scale, py::arg("tp"), py::arg("A"),
         py::arg("alpha"), py::arg("beta"),
         py::call_guard<py::gil_scoped_release>());
+  m.def("insert_reduce_sum", &insert_reduce_sum, py::arg("tp"), py::arg("A"),
+        py::arg("R"), py::call_guard<py::gil_scoped_release>());
   m.def("insert_fill_bf16", &insert_fill_bf16, py::arg("tp"), py::arg("A"),
         py::arg("seed") = 1u, py::call_guard<py::gil_scoped_release>());
   m.def("insert_gemm_bf16", &insert_gemm_bf16, py::arg("tp"), py::arg("At"),

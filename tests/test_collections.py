@@ -59,3 +59,19 @@ def test_compose_on_complete(ctx):
     tp2.insert_py(lambda: order.append("tp2_task"))
     tp2.wait()
     assert order == ["tp1_done", "tp2_task"]
+
+
+def test_reduce_sum(ctx):
+    n, nb = 128, 64
+    A = pm.TiledMatrix(ctx, n, n, nb, nb, 1, 1, sym=True)
+    R = pm.TiledMatrix(ctx, nb, nb, nb, nb, 1, 1)
+    R.tile_numpy_set(0, 0, np.zeros((nb, nb)))
+    tp = pm.Dtd(ctx)
+    pm.insert_spd_fill(tp, A, 8)
+    pm.insert_reduce_sum(tp, A, R)
+    tp.wait()
+    expect = np.zeros((nb, nb))
+    for tm in range(A.mt):
+        for tn in range(tm + 1):
+            expect += A.tile_numpy(tm, tn)
+    assert np.allclose(R.tile_numpy(0, 0), expect)

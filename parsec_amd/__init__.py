@@ -31,6 +31,7 @@ from parsec_amd._core import (  # noqa: F401,E402
     insert_redistribute,
     insert_apply_scale,
     insert_reduce_sum,
+    insert_stencil_1d,
     param_dump,
     param_set,
 )

@@ -131,6 +131,7 @@ Task* Dtd::insert_begin(const TaskClass* tc, const void* args,
     t->nflows = nflows;
   }
   for (int i = 0; i < nflows; i++) {
+    if (!flows[i].d) continue;  // NULL flow (e.g. absent stencil halo)
     if (flows[i].mode & ACCESS_IN) read_flow(flows[i].d, t, task_rank);
     if (flows[i].mode & ACCESS_OUT) write_flow(flows[i].d, t, task_rank);
   }

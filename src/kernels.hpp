@@ -43,5 +43,6 @@ void insert_gemm_bf16(Dtd& tp, TiledMatrix& At, TiledMatrix& B, TiledMatrix& C);
 void insert_redistribute(Dtd& tp, TiledMatrix& Src, TiledMatrix& Dst);
 void insert_apply_scale(Dtd& tp, TiledMatrix& A, double alpha, double beta);
 void insert_reduce_sum(Dtd& tp, TiledMatrix& A, TiledMatrix& R);
+void insert_stencil_1d(Dtd& tp, TiledMatrix& Src, TiledMatrix& Dst);
 
 }  // namespace pa

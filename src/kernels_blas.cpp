@@ -533,6 +533,67 @@ void insert_apply_scale(Dtd& tp, TiledMatrix& A, double alpha, double beta) {
     }
 }
 
+// ---------------------------------------------------------------- stencil
+// 1-D 3-point stencil over a vector of tiles (tests/apps/stencil analog):
+// dst[i] = (src[i-1] + src[i] + src[i+1]) / 3 with halos crossing tile
+// boundaries through neighbor-tile flows (zero at domain edges).
+__global__ void k_stencil3(double* dst, const double* left,
+                           const double* mid, const double* right, int nbe) {
+  int i = blockIdx.x * blockDim.x + threadIdx.x;
+  for (; i < nbe; i += gridDim.x * blockDim.x) {
+    double l = i > 0 ? mid[i - 1] : (left ? left[nbe - 1] : 0.0);
+    double r = i < nbe - 1 ? mid[i + 1] : (right ? right[0] : 0.0);
+    dst[i] = (l + mid[i] + r) / 3.0;
+  }
+}
+
+static void cpu_stencil3(Task& t) {
+  const TileArgs& a = t.arg<TileArgs>();
+  const int nbe = a.m;
+  const double* left =
+      t.flows[1].data ? (const double*)t.flows[1].data->pull_to_host() : nullptr;
+  const double* mid = (const double*)t.flows[0].data->pull_to_host();
+  const double* right =
+      t.flows[2].data ? (const double*)t.flows[2].data->pull_to_host() : nullptr;
+  double* dst = (double*)t.flows[3].data->ensure_host();
+  for (int i = 0; i < nbe; i++) {
+    double l = i > 0 ? mid[i - 1] : (left ? left[nbe - 1] : 0.0);
+    double r = i < nbe - 1 ? mid[i + 1] : (right ? right[0] : 0.0);
+    dst[i] = (l + mid[i] + r) / 3.0;
+  }
+  t.flows[3].data->written_on(false);
+}
+
+static void gpu_stencil3(Task& t, GpuTaskCtx& g) {
+  const TileArgs& a = t.arg<TileArgs>();
+  hipLaunchKernelGGL(k_stencil3, dim3(64), dim3(256), 0, g.stream,
+                     (double*)t.dev_ptr[3], (const double*)t.dev_ptr[1],
+                     (const double*)t.dev_ptr[0], (const double*)t.dev_ptr[2],
+                     a.m);
+}
+
+TaskClass& tc_stencil3() {
+  static TaskClass tc = make_tc("stencil3", TaskKind::GPU, cpu_stencil3,
+                                gpu_stencil3, 33);
+  return tc;
+}
+
+// One stencil sweep Src -> Dst (tile vectors: mt x 1 tiles of mb x 1).
+void insert_stencil_1d(Dtd& tp, TiledMatrix& Src, TiledMatrix& Dst) {
+  PA_CHECK(Src.nt() == 1 && Dst.nt() == 1 && Src.mt() == Dst.mt());
+  const int T = Src.mt();
+  for (int t = 0; t < T; t++) {
+    TileArgs a;
+    a.m = Src.tile_rows(t) * Src.tile_cols(0);
+    Dtd::FlowSpec f[4];
+    f[0] = {Src.tile(t, 0), ACCESS_IN};
+    f[1] = {t > 0 ? Src.tile(t - 1, 0) : nullptr, ACCESS_IN};
+    f[2] = {t < T - 1 ? Src.tile(t + 1, 0) : nullptr, ACCESS_IN};
+    f[3] = {Dst.tile(t, 0), ACCESS_OUT};
+    tp.insert(&tc_stencil3(), &a, sizeof(a), f, 4, 0, Dst.rank_of(t, 0));
+  }
+}
+
 // ------------------------------------------------------------------ DAG builders
 void insert_spd_fill(Dtd& tp, TiledMatrix& A, uint32_t seed) {
   for (int tm = 0; tm < A.mt(); tm++)

@@ -75,3 +75,24 @@ def test_reduce_sum(ctx):
         for tn in range(tm + 1):
             expect += A.tile_numpy(tm, tn)
     assert np.allclose(R.tile_numpy(0, 0), expect)
+
+
+def test_stencil_1d(ctx):
+    nb, T = 64, 6
+    n = nb * T
+    Src = pm.TiledMatrix(ctx, n, 1, nb, 1, 1, 1)
+    Dst = pm.TiledMatrix(ctx, n, 1, nb, 1, 1, 1)
+    rng = np.random.default_rng(0)
+    x = rng.standard_normal(n)
+    for t in range(T):
+        Src.tile_numpy_set(t, 0, x[t * nb:(t + 1) * nb].reshape(-1, 1))
+    tp = pm.Dtd(ctx)
+    pm.insert_stencil_1d(tp, Src, Dst)
+    pm.insert_stencil_1d(tp, Dst, Src)  # second sweep back into Src
+    tp.wait()
+    xp = np.pad(x, 1)
+    y = (xp[:-2] + xp[1:-1] + xp[2:]) / 3.0
+    yp = np.pad(y, 1)
+    z = (yp[:-2] + yp[1:-1] + yp[2:]) / 3.0
+    got = np.concatenate([Src.tile_numpy(t, 0).ravel() for t in range(T)])
+    assert np.allclose(got, z), np.abs(got - z).max()

@@ -49,7 +49,7 @@ def run_world(world, tmpdir, p, q, n=256, nb=64):
     return pre, post
 
 
-@pytest.mark.parametrize("world,p,q", [(2, 2, 1), (2, 1, 2), (4, 2, 2)])
+@pytest.mark.parametrize("world,p,q", [(2, 2, 1), (2, 1, 2), (4, 2, 2), (8, 2, 4)])
 def test_distributed_cholesky(world, p, q, tmp_path):
     pre, post = run_world(world, tmp_path, p, q)
     M = np.tril(pre) + np.tril(pre, -1).T

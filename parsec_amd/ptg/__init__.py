@@ -576,6 +576,7 @@ def compile_jdf(path, verbose=False):
         cmd = ["hipcc", "--offload-arch=gfx950", "-O2", "-std=c++17",
                "-fPIC", "-shared", "-I", os.path.join(REPO, "src"),
                src, "-L", coredir, "-l:_core.so",
+               "-L/opt/rocm/lib", "-lrocblas", "-lrocsolver",
                f"-Wl,-rpath,{coredir}", "-o", so]
         r = subprocess.run(cmd, capture_output=True, text=True)
         if r.returncode != 0:

@@ -119,3 +119,31 @@ def test_ptg_hip_body(ctx):
     for k in range(NT):
         got = A.tile_numpy(k, 0)
         assert np.allclose(got, k * 2.0 + (k + 1)), f"tile {k} wrong"
+
+
+@pytest.mark.gpu
+def test_ptg_hip_cholesky():
+    """PTG Cholesky with rocBLAS/rocSOLVER HIP bodies (stress.jdf analog)."""
+    n, nb = 1024, 256
+    ctx2 = pm.Context(nworkers=2, rank=0, world=1)
+    A = pm.TiledMatrix(ctx2, n, n, nb, nb, 1, 1)
+    tp0 = pm.Dtd(ctx2)
+    pm.insert_spd_fill(tp0, A, 42)
+    tp0.wait()
+    M = np.zeros((n, n))
+    for tm in range(A.mt):
+        for tn in range(tm + 1):
+            M[tm * nb:(tm + 1) * nb, tn * nb:(tn + 1) * nb] = A.tile_numpy(tm, tn)
+    M = np.tril(M) + np.tril(M, -1).T
+    L0 = np.linalg.cholesky(M)
+    mod = compile_jdf(os.path.join(EX, "cholesky_hip.jdf"))
+    tp = pm.Dtd(ctx2, "ptg_hip_potrf")
+    mod.build(ctx2, tp, descA=A, NT=A.mt, NB=nb)
+    tp.wait()
+    L = np.zeros((n, n))
+    for tm in range(A.mt):
+        for tn in range(tm + 1):
+            L[tm * nb:(tm + 1) * nb, tn * nb:(tn + 1) * nb] = A.tile_numpy(tm, tn)
+    err = np.abs(np.tril(L) - L0).max()
+    assert err < 1e-8, f"PTG HIP cholesky max err {err}"
+    del A, ctx2

@@ -60,16 +60,18 @@ def init_distributed(nworkers=-1, comm=None, gpu=-1):
         comm = os.environ.get("PARSEC_COMM_KIND")
     if comm is None:
         comm = "rccl" if (world > 1 and hip_device_count() > 0) else ""
-    if world > 1 and comm == "rccl":
+    if world > 1:
+        # gloo side-channel for bootstrap + whole-job reductions (bench
+        # timing); works with or without GPUs.
         import torch.distributed as dist
 
         if not dist.is_initialized():
             dist.init_process_group(backend="gloo", rank=rank, world_size=world)
-        if rank == 0:
-            uid = _core.nccl_unique_id()
-            obj = [uid]
-        else:
-            obj = [None]
-        dist.broadcast_object_list(obj, src=0)
-        _core.set_nccl_unique_id(obj[0])
+        if comm == "rccl":
+            if rank == 0:
+                obj = [_core.nccl_unique_id()]
+            else:
+                obj = [None]
+            dist.broadcast_object_list(obj, src=0)
+            _core.set_nccl_unique_id(obj[0])
     return Context(nworkers=nworkers, rank=rank, world=world, comm=comm, gpu=gpu)

@@ -135,3 +135,44 @@ del A, ctx
         r = subprocess.run([sys.executable, "-c", code], capture_output=True,
                            text=True, timeout=120)
         assert r.returncode == 0, f"sched={kind}: {r.stdout}{r.stderr}"
+
+
+def test_random_dag_vs_sequential_oracle(ctx):
+    """Fuzz: a random program of scale/add/copy ops over 8 tiles executed
+    through the DTD engine must match a sequential numpy replay (pins the
+    RAW/WAR/WAW chaining semantics under concurrent execution)."""
+    import numpy as np
+    import random
+    rng = random.Random(1234)
+    nb = 32
+    NT = 8
+    mats = [pm.TiledMatrix(ctx, nb, nb, nb, nb, 1, 1) for _ in range(NT)]
+    ref = []
+    tp = pm.Dtd(ctx)
+    for i, M in enumerate(mats):
+        v = np.full((nb, nb), float(i + 1))
+        M.tile_numpy_set(0, 0, v)
+        ref.append(v.copy())
+    for _ in range(200):
+        op = rng.choice(["scale", "add", "copy"])
+        if op == "scale":
+            i = rng.randrange(NT)
+            a, b = rng.uniform(0.5, 1.5), rng.uniform(-1, 1)
+            pm.insert_apply_scale(tp, mats[i], a, b)
+            ref[i] = ref[i] * a + b
+        elif op == "add":
+            i, j = rng.randrange(NT), rng.randrange(NT)
+            if i == j:
+                continue
+            pm.insert_reduce_sum(tp, mats[i], mats[j])
+            ref[j] = ref[j] + ref[i]
+        else:
+            i, j = rng.randrange(NT), rng.randrange(NT)
+            if i == j:
+                continue
+            pm.insert_redistribute(tp, mats[i], mats[j])
+            ref[j] = ref[i].copy()
+    tp.wait()
+    for i, M in enumerate(mats):
+        got = M.tile_numpy(0, 0)
+        assert np.allclose(got, ref[i]), f"tile {i} diverged"

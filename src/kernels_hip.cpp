@@ -204,9 +204,99 @@ __global__ void k_dgemm_nt_v2(int m, int n, int k, const double* __restrict__ A,
   }
 }
 
+// ------------------------------------------------------------------ v3
+// glds-staged variant (the structure that lifted the bf16 kernel 2x):
+// async global->LDS (16 B/lane), [k][m] lane-linear LDS image (one K-row =
+// 1 KB per glds), double-buffered BK=16, one drain barrier per K-tile.
+// Full tiles only; edges fall back to v2.
+__launch_bounds__(512)
+__global__ void k_dgemm_nt_v3(int m, int n, int k, const double* __restrict__ A,
+                              int lda, const double* __restrict__ B, int ldb,
+                              double* __restrict__ C, int ldc, int nbx) {
+  extern __shared__ __attribute__((aligned(16))) char smem[];
+  double* As = (double*)smem;                  // [2][BKD][BM]
+  double* Bs = As + 2 * BKD * BM;
+  const int tid = threadIdx.x;
+  const int wave = tid >> 6, lane = tid & 63;
+  const int wr = wave >> 1, wc = wave & 1;
+  int id = xcd_swizzle(blockIdx.x, gridDim.x);
+  const int bm0 = (id % nbx) * BM, bn0 = (id / nbx) * BN;
+  const int ksub = lane >> 4, r16 = lane & 15;
+
+  f64x4 acc[2][4] = {};
+
+  auto stage = [&](int buf, int k0) {
+#pragma unroll
+    for (int p = 0; p < 2; p++) {
+      int krow = wave * 2 + p;
+      const double* srcA = A + (size_t)(k0 + krow) * lda + bm0 + 2 * lane;
+      double* dstA = As + buf * BKD * BM + krow * BM;
+      __builtin_amdgcn_global_load_lds(
+          (const __attribute__((address_space(1))) unsigned int*)srcA,
+          (__attribute__((address_space(3))) unsigned int*)dstA, 16, 0, 0);
+      const double* srcB = B + (size_t)(k0 + krow) * ldb + bn0 + 2 * lane;
+      double* dstB = Bs + buf * BKD * BN + krow * BN;
+      __builtin_amdgcn_global_load_lds(
+          (const __attribute__((address_space(1))) unsigned int*)srcB,
+          (__attribute__((address_space(3))) unsigned int*)dstB, 16, 0, 0);
+    }
+  };
+
+  stage(0, 0);
+  __syncthreads();
+  const int ntiles = k / BKD;
+  for (int t = 0; t < ntiles; t++) {
+    if (t + 1 < ntiles) stage((t + 1) & 1, (t + 1) * BKD);
+    const double* as = As + (t & 1) * BKD * BM;
+    const double* bs = Bs + (t & 1) * BKD * BN;
+#pragma unroll
+    for (int kk = 0; kk < BKD; kk += 4) {
+      double a[2], b[4];
+#pragma unroll
+      for (int f = 0; f < 2; f++)
+        a[f] = as[(kk + ksub) * BM + wr * 32 + f * 16 + r16];
+#pragma unroll
+      for (int f = 0; f < 4; f++)
+        b[f] = bs[(kk + ksub) * BN + wc * 64 + f * 16 + r16];
+#pragma unroll
+      for (int i = 0; i < 2; i++)
+#pragma unroll
+        for (int j = 0; j < 4; j++)
+          acc[i][j] =
+              __builtin_amdgcn_mfma_f64_16x16x4f64(a[i], b[j], acc[i][j], 0, 0, 0);
+    }
+    __syncthreads();
+  }
+
+#pragma unroll
+  for (int j = 0; j < 4; j++) {
+    int col = bn0 + wc * 64 + j * 16 + r16;
+    double* cp = C + (size_t)col * ldc;
+#pragma unroll
+    for (int i = 0; i < 2; i++) {
+      int row0 = bm0 + wr * 32 + i * 16 + ksub;
+#pragma unroll
+      for (int e = 0; e < 4; e++) cp[row0 + e * 4] -= acc[i][j][e];
+    }
+  }
+}
+
 static void launch_dgemm_v2(int m, int n, int k, const double* A, int lda,
                             const double* B, int ldb, double* C, int ldc,
                             hipStream_t stream) {
+  if (m % BM == 0 && n % BN == 0 && k % BKD == 0) {
+    constexpr size_t lds = 2 * BKD * (BM + BN) * 8;
+    static bool attr3 = false;
+    if (!attr3) {
+      hipFuncSetAttribute((const void*)k_dgemm_nt_v3,
+                          hipFuncAttributeMaxDynamicSharedMemorySize, lds);
+      attr3 = true;
+    }
+    int nbx = (m + BM - 1) / BM, nby = (n + BN - 1) / BN;
+    hipLaunchKernelGGL(k_dgemm_nt_v3, dim3(nbx * nby), dim3(512), lds,
+                       stream, m, n, k, A, lda, B, ldb, C, ldc, nbx);
+    return;
+  }
   static bool attr_set = false;
   constexpr size_t lds = 4 * BKD * 129 * 8;
   if (!attr_set) {

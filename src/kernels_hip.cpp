@@ -209,6 +209,7 @@ __global__ void k_dgemm_nt_v2(int m, int n, int k, const double* __restrict__ A,
 // async global->LDS (16 B/lane), [k][m] lane-linear LDS image (one K-row =
 // 1 KB per glds), double-buffered BK=16, one drain barrier per K-tile.
 // Full tiles only; edges fall back to v2.
+template <bool TRI>
 __launch_bounds__(512)
 __global__ void k_dgemm_nt_v3(int m, int n, int k, const double* __restrict__ A,
                               int lda, const double* __restrict__ B, int ldb,
@@ -220,7 +221,17 @@ __global__ void k_dgemm_nt_v3(int m, int n, int k, const double* __restrict__ A,
   const int wave = tid >> 6, lane = tid & 63;
   const int wr = wave >> 1, wc = wave & 1;
   int id = xcd_swizzle(blockIdx.x, gridDim.x);
-  const int bm0 = (id % nbx) * BM, bn0 = (id / nbx) * BN;
+  int bx, by;
+  if (TRI) {
+    bx = 0;
+    int rem = id;
+    while (rem > bx) { bx++; rem -= bx; }
+    by = rem;
+  } else {
+    bx = id % nbx;
+    by = id / nbx;
+  }
+  const int bm0 = bx * BM, bn0 = by * BN;
   const int ksub = lane >> 4, r16 = lane & 15;
 
   f64x4 acc[2][4] = {};
@@ -288,12 +299,14 @@ static void launch_dgemm_v2(int m, int n, int k, const double* A, int lda,
     constexpr size_t lds = 2 * BKD * (BM + BN) * 8;
     static bool attr3 = false;
     if (!attr3) {
-      hipFuncSetAttribute((const void*)k_dgemm_nt_v3,
+      hipFuncSetAttribute((const void*)k_dgemm_nt_v3<false>,
+                          hipFuncAttributeMaxDynamicSharedMemorySize, lds);
+      hipFuncSetAttribute((const void*)k_dgemm_nt_v3<true>,
                           hipFuncAttributeMaxDynamicSharedMemorySize, lds);
       attr3 = true;
     }
     int nbx = (m + BM - 1) / BM, nby = (n + BN - 1) / BN;
-    hipLaunchKernelGGL(k_dgemm_nt_v3, dim3(nbx * nby), dim3(512), lds,
+    hipLaunchKernelGGL(k_dgemm_nt_v3<false>, dim3(nbx * nby), dim3(512), lds,
                        stream, m, n, k, A, lda, B, ldb, C, ldc, nbx);
     return;
   }
@@ -321,6 +334,18 @@ void launch_dsyrk_v2(int n, int k, const double* A, int lda, double* C,
   }
   int ntb = (n + BM - 1) / BM;
   int ntiles = ntb * (ntb + 1) / 2;
+  if (n % BM == 0 && k % BKD == 0) {
+    constexpr size_t lds3 = 2 * BKD * (BM + BN) * 8;
+    static bool attr3t = false;
+    if (!attr3t) {
+      hipFuncSetAttribute((const void*)k_dgemm_nt_v3<true>,
+                          hipFuncAttributeMaxDynamicSharedMemorySize, lds3);
+      attr3t = true;
+    }
+    hipLaunchKernelGGL(k_dgemm_nt_v3<true>, dim3(ntiles), dim3(512), lds3,
+                       stream, n, n, k, A, lda, A, lda, C, ldc, 0);
+    return;
+  }
   hipLaunchKernelGGL(k_dgemm_nt_v2<true>, dim3(ntiles), dim3(512), lds,
                      stream, n, n, k, A, lda, A, lda, C, ldc, 0);
 }

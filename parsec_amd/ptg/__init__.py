@@ -564,7 +564,9 @@ def compile_jdf(path, verbose=False):
     h = hashlib.sha256((cpp + "v1").encode()).hexdigest()[:16]
     so = os.path.join(CACHE, f"{name}_{h}.so")
     if not os.path.exists(so):
+        # atomic publish: concurrent ranks may compile the same JDF
         src = os.path.join(CACHE, f"{name}_{h}.cpp")
+        tmp_so = so + f".tmp{os.getpid()}"
         with open(src, "w") as f:
             f.write(cpp)
         # Link directly against _core.so so the pa_* C ABI resolves without
@@ -577,11 +579,12 @@ def compile_jdf(path, verbose=False):
                "-fPIC", "-shared", "-I", os.path.join(REPO, "src"),
                src, "-L", coredir, "-l:_core.so",
                "-L/opt/rocm/lib", "-lrocblas", "-lrocsolver",
-               f"-Wl,-rpath,{coredir}", "-o", so]
+               f"-Wl,-rpath,{coredir}", "-o", tmp_so]
         r = subprocess.run(cmd, capture_output=True, text=True)
         if r.returncode != 0:
             raise JdfError(f"ptgpp: generated code failed to compile:\n"
                            f"{r.stderr[-4000:]}")
+        os.rename(tmp_so, so)
         if verbose:
             print(f"[ptgpp] compiled {path} -> {so}", file=sys.stderr)
     return PtgModule(so, name, jdf)

@@ -50,6 +50,11 @@ def main():
     tp2 = pm.Dtd(ctx)
     if app == "qr":
         pm.insert_geqrf(tp2, A)
+    elif app == "ptg":
+        from parsec_amd.ptg import compile_jdf
+        repo = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+        mod = compile_jdf(os.path.join(repo, "examples", "cholesky.jdf"))
+        mod.build(ctx, tp2, descA=A, NT=A.mt, NB=nb)
     else:
         pm.insert_potrf(tp2, A)
     tp2.flush_all(A)

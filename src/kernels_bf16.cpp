@@ -220,9 +220,141 @@ __global__ void k_gemm_bf16_tn_v2(int m, int n, int k,
   }
 }
 
+// ------------------------------------------------------------- v3: 256^2
+// Deep-pipelined schedule (T3+T4 of the guide): 256x256 tile, 8 waves
+// (2M x 4N, 128x64 per wave), K processed in 32-wide "K-half" phases over a
+// 4-slot LDS ring (A+B 32 KB per slot, 128 KB total -> 1 block/CU, 2
+// waves/SIMD). Each phase: counted `s_waitcnt vmcnt(8)` (the K-half staged
+// 3 phases ago has landed; 2 newer phases x 4 glds stay IN FLIGHT across
+// the barrier), one raw `s_barrier` (never __syncthreads: it would drain
+// the glds queue — guide §5 'Pipelining across barriers'), 12 ds_read_b128
+// fragment loads, 4 glds staging the phase+3 K-half, 32 MFMAs. The LDS
+// image is XOR-swizzled through the glds source address (key
+// ((row>>2)&3)*8 elements) so the b128 fragment reads are conflict-free.
+#define GB3_BM 256
+#define GB3_BN 256
+#define GB3_KH 32  // K per phase
+
+__launch_bounds__(512)
+__global__ void k_gemm_bf16_tn_v3(int m, int n, int k,
+                                  const bf16* __restrict__ A, int lda,
+                                  const bf16* __restrict__ B, int ldb,
+                                  float* __restrict__ C, int ldc, int nbx,
+                                  int accum) {
+  extern __shared__ __attribute__((aligned(16))) char smem[];
+  bf16* ring = (bf16*)smem;  // 4 slots x [A 256x32 | B 256x32]
+  constexpr int SLOT = (GB3_BM + GB3_BN) * GB3_KH;  // bf16 elements
+  const int tid = threadIdx.x;
+  const int wave = tid >> 6, lane = tid & 63;
+  const int wr = wave >> 2, wc = wave & 3;  // 2(M) x 4(N): 128x64 per wave
+  int id = bf_swz(blockIdx.x, gridDim.x);
+  const int bm0 = (id % nbx) * GB3_BM, bn0 = (id / nbx) * GB3_BN;
+  const int g16 = lane >> 4, r16 = lane & 15;
+  const int srow = lane >> 2;           // staging: 16 rows x 4 chunks
+  const int schunk = lane & 3;
+
+  f32x4 acc[8][4] = {};
+
+  // stage K-half `ph` (globally k0 = ph*32) into ring slot ph&3
+  auto stage = [&](int ph) {
+    const int k0 = ph * GB3_KH;
+    bf16* slot = ring + (ph & 3) * SLOT;
+#pragma unroll
+    for (int p = 0; p < 2; p++) {
+      int row0 = (wave * 2 + p) * 16;
+      int row = row0 + srow;
+      int kc = 8 * (schunk ^ ((row >> 2) & 3));  // source pre-swizzle
+      const bf16* srcA = A + (size_t)(bm0 + row) * lda + k0 + kc;
+      bf16* dstA = slot + row0 * GB3_KH;
+      __builtin_amdgcn_global_load_lds(
+          (const __attribute__((address_space(1))) unsigned int*)srcA,
+          (__attribute__((address_space(3))) unsigned int*)dstA, 16, 0, 0);
+      const bf16* srcB = B + (size_t)(bn0 + row) * ldb + k0 + kc;
+      bf16* dstB = slot + GB3_BM * GB3_KH + row0 * GB3_KH;
+      __builtin_amdgcn_global_load_lds(
+          (const __attribute__((address_space(1))) unsigned int*)srcB,
+          (__attribute__((address_space(3))) unsigned int*)dstB, 16, 0, 0);
+    }
+  };
+
+  const int P = k / GB3_KH;
+  stage(0);
+  if (P > 1) stage(1);
+  if (P > 2) stage(2);
+  for (int ph = 0; ph < P; ph++) {
+    // counted wait: the K-half for THIS phase (own glds) has landed;
+    // newer phases' loads stay in flight across the barrier
+    if (ph + 3 <= P - 1) {
+      asm volatile("s_waitcnt vmcnt(8)" ::: "memory");
+    } else if (ph + 2 == P - 1) {
+      asm volatile("s_waitcnt vmcnt(8)" ::: "memory");
+    } else if (ph + 1 == P - 1) {
+      asm volatile("s_waitcnt vmcnt(4)" ::: "memory");
+    } else {
+      asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+    }
+    __builtin_amdgcn_s_barrier();
+    const bf16* as = ring + (ph & 3) * SLOT;
+    const bf16* bs = as + GB3_BM * GB3_KH;
+    bf16x8 a[8], b[4];
+#pragma unroll
+    for (int f = 0; f < 8; f++) {
+      int row = wr * 128 + f * 16 + r16;
+      a[f] = *(const bf16x8*)&as[row * GB3_KH +
+                                 ((g16 * 8) ^ (((row >> 2) & 3) * 8))];
+    }
+#pragma unroll
+    for (int f = 0; f < 4; f++) {
+      int row = wc * 64 + f * 16 + r16;
+      b[f] = *(const bf16x8*)&bs[row * GB3_KH +
+                                 ((g16 * 8) ^ (((row >> 2) & 3) * 8))];
+    }
+    if (ph + 3 < P) stage(ph + 3);
+    __builtin_amdgcn_s_setprio(1);
+#pragma unroll
+    for (int i = 0; i < 8; i++)
+#pragma unroll
+      for (int j = 0; j < 4; j++)
+        acc[i][j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a[i], b[j],
+                                                            acc[i][j], 0, 0, 0);
+    __builtin_amdgcn_s_setprio(0);
+    // no trailing barrier: the next phase's counted-wait + barrier is the
+    // only synchronization needed (slot reuse is 3 phases away)
+  }
+
+#pragma unroll
+  for (int j = 0; j < 4; j++) {
+    int col = bn0 + wc * 64 + j * 16 + r16;
+    float* cp = C + (size_t)col * ldc;
+#pragma unroll
+    for (int i = 0; i < 8; i++) {
+      int row0 = bm0 + wr * 128 + i * 16 + g16 * 4;
+#pragma unroll
+      for (int e = 0; e < 4; e++) {
+        int row = row0 + e;
+        cp[row] = accum ? cp[row] + acc[i][j][e] : acc[i][j][e];
+      }
+    }
+  }
+}
+
 static void launch_gemm_bf16(int m, int n, int k, const void* A, int lda,
                              const void* B, int ldb, float* C, int ldc,
                              hipStream_t stream, int accum = 1) {
+  if (m % GB3_BM == 0 && n % GB3_BN == 0 && k % GB3_KH == 0 && k >= 4 * GB3_KH) {
+    constexpr size_t lds = 4 * (GB3_BM + GB3_BN) * GB3_KH * 2;
+    static bool attr3 = false;
+    if (!attr3) {
+      hipFuncSetAttribute((const void*)k_gemm_bf16_tn_v3,
+                          hipFuncAttributeMaxDynamicSharedMemorySize, lds);
+      attr3 = true;
+    }
+    int nbx = m / GB3_BM, nby = n / GB3_BN;
+    hipLaunchKernelGGL(k_gemm_bf16_tn_v3, dim3(nbx * nby), dim3(512), lds,
+                       stream, m, n, k, (const bf16*)A, lda, (const bf16*)B,
+                       ldb, C, ldc, nbx, accum);
+    return;
+  }
   int nbx = (m + GB_BM - 1) / GB_BM, nby = (n + GB_BN - 1) / GB_BN;
   if (m % GB_BM == 0 && n % GB_BN == 0 && k % GB2_BK == 0) {
     constexpr size_t lds = 2 * (GB_BM + GB_BN) * GB2_BK * 2;

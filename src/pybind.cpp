@@ -12,6 +12,7 @@
 #include "device_gpu.hpp"
 #include "dtd.hpp"
 #include "kernels.hpp"
+#include "profiling.hpp"
 #include "runtime.hpp"
 
 namespace pa {
@@ -115,6 +116,18 @@ PYBIND11_MODULE(_core, m) {
           d["bytes_h2d"] = c.gpu()->stats.bytes_h2d.load();
           d["bytes_d2h"] = c.gpu()->stats.bytes_d2h.load();
         }
+        return d;
+      })
+      .def("counters", [](Context&) {
+        // PINS/papi_sde-style software counters (process-wide)
+        auto& c = counters();
+        py::dict d;
+        d["tasks_executed_cpu"] = c.tasks_executed_cpu.load();
+        d["tasks_executed_gpu"] = c.tasks_executed_gpu.load();
+        d["tasks_scheduled"] = c.tasks_scheduled.load();
+        d["steals"] = c.steals.load();
+        d["comm_msgs"] = c.comm_msgs.load();
+        d["comm_bytes"] = c.comm_bytes.load();
         return d;
       });
 

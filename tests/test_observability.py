@@ -39,3 +39,25 @@ del ctx
     assert "digraph" in dtext
     assert "potrf" in dtext
     assert "->" in dtext
+
+
+def test_counters_and_cli(tmp_path):
+    import parsec_amd as pm
+    code = f"""
+import sys; sys.path.insert(0, {REPO!r})
+import parsec_amd as pm
+ctx = pm.Context(nworkers=2, rank=0, world=1, gpu=-2)
+A = pm.TiledMatrix(ctx, 128, 128, 64, 64, 1, 1)
+tp = pm.Dtd(ctx); pm.insert_spd_fill(tp, A, 1); tp.wait()
+c = ctx.counters()
+assert c["tasks_executed_cpu"] >= 3, c
+assert c["tasks_scheduled"] >= 3, c
+print("COUNTERS_OK")
+del A, ctx
+"""
+    r = subprocess.run([sys.executable, "-c", code], capture_output=True,
+                       text=True, timeout=120)
+    assert "COUNTERS_OK" in r.stdout, r.stdout + r.stderr
+    r2 = subprocess.run([sys.executable, "-m", "parsec_amd"],
+                        capture_output=True, text=True, timeout=120, cwd=REPO)
+    assert "MI355X-native" in r2.stdout

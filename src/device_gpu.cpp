@@ -33,7 +33,7 @@ GpuEngine::GpuEngine(Context* ctx, int device) : ctx_(ctx), device_(device) {
   inflight_.resize(nstreams);
   for (int i = 0; i < nstreams; i++)
     PA_HIP_CHECK(hipStreamCreateWithFlags(&exec_streams_[i], hipStreamNonBlocking));
-  max_inflight_per_stream_ = (size_t)param_int("gpu_max_inflight", 32);
+  max_inflight_per_stream_ = (size_t)param_int("gpu_max_inflight", 64);
   next_stream_ = nstreams > 1 ? 1 : 0;
 
   // Reserve the HBM slab (parsec_device_memory_reserve analog,

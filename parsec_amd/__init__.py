@@ -20,6 +20,7 @@ from parsec_amd._core import (  # noqa: F401,E402
     Context,
     Data,
     Dtd,
+    IrregularCollection,
     Taskpool,
     TiledMatrix,
     hip_device_count,

@@ -27,7 +27,8 @@ class Context;
 struct Data {
   // identity
   uint64_t key = 0;
-  class TiledMatrix* coll = nullptr;
+  class TiledMatrix* coll = nullptr;   // owning tiled collection (if any)
+  class Context* ctx_direct = nullptr; // set instead for irregular data
   int home_rank = 0;
   size_t bytes = 0;
 
@@ -116,6 +117,26 @@ class TiledMatrix {
   size_t elem_;
   bool sym_ = false;
   std::vector<std::unique_ptr<Data>> tiles_;  // mt*nt, metadata eager
+};
+
+// Irregular key->data collection (hash_datadist analog,
+// data_dist/hash_datadist.c: arbitrary keys, explicit per-key rank —
+// trees/graphs/ragged structures).
+class IrregularCollection {
+ public:
+  explicit IrregularCollection(Context* ctx) : ctx_(ctx) {}
+  ~IrregularCollection();
+
+  // Register (or return) the datum for `key`; rank/bytes fixed on first
+  // registration and must be identical on every rank (SPMD).
+  Data* add(uint64_t key, int rank, size_t bytes);
+  Data* at(uint64_t key);
+  Context* ctx() const { return ctx_; }
+  size_t size() const { return map_.size(); }
+
+ private:
+  Context* ctx_;
+  std::unordered_map<uint64_t, std::unique_ptr<Data>> map_;
 };
 
 }  // namespace pa

@@ -377,7 +377,8 @@ void* Data::ensure_host() {
 }
 
 void* Data::pull_to_host() {
-  GpuEngine* eng = coll ? coll->ctx()->gpu() : nullptr;
+  Context* c = coll ? coll->ctx() : ctx_direct;
+  GpuEngine* eng = c ? c->gpu() : nullptr;
   SpinGuard g(lock);
   ensure_host();
   if (!host_valid) {
@@ -416,6 +417,32 @@ TiledMatrix::~TiledMatrix() {
   for (auto& t : tiles_) {
     if (t && t->dev_ptr && eng) eng->dev_free(t->dev_ptr, t->bytes);
   }
+}
+
+IrregularCollection::~IrregularCollection() {
+  GpuEngine* eng = ctx_->gpu();
+  for (auto& [k, d] : map_)
+    if (d->dev_ptr && eng) eng->dev_free(d->dev_ptr, d->bytes);
+}
+
+Data* IrregularCollection::add(uint64_t key, int rank, size_t bytes) {
+  auto& slot = map_[key];
+  if (!slot) {
+    slot = std::make_unique<Data>();
+    slot->key = key;
+    slot->ctx_direct = ctx_;
+    slot->home_rank = rank;
+    slot->owner_rank = rank;
+    slot->bytes = bytes;
+  }
+  return slot.get();
+}
+
+Data* IrregularCollection::at(uint64_t key) {
+  auto it = map_.find(key);
+  PA_CHECK(it != map_.end(), "IrregularCollection: unknown key %llu",
+           (unsigned long long)key);
+  return it->second.get();
 }
 
 Data* TiledMatrix::tile(int tm, int tn) {

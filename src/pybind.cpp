@@ -201,6 +201,30 @@ PYBIND11_MODULE(_core, m) {
         d->written_on(false);
       });
 
+  py::class_<IrregularCollection>(m, "IrregularCollection")
+      .def(py::init<Context*>(), py::arg("ctx"), py::keep_alive<1, 2>())
+      .def("add", &IrregularCollection::add, py::arg("key"), py::arg("rank"),
+           py::arg("bytes"), py::return_value_policy::reference_internal)
+      .def("at", &IrregularCollection::at, py::arg("key"),
+           py::return_value_policy::reference_internal)
+      .def_property_readonly("size", &IrregularCollection::size)
+      .def("bytes_get", [](IrregularCollection& c, uint64_t key) {
+        Data* d = c.at(key);
+        void* p;
+        {
+          py::gil_scoped_release rel;
+          p = d->pull_to_host();
+        }
+        return py::bytes((const char*)p, d->bytes);
+      })
+      .def("bytes_set", [](IrregularCollection& c, uint64_t key, py::bytes b) {
+        Data* d = c.at(key);
+        std::string s2(b);
+        PA_CHECK(s2.size() <= d->bytes);
+        memcpy(d->ensure_host(), s2.data(), s2.size());
+        d->written_on(false);
+      });
+
   py::class_<Taskpool>(m, "Taskpool")
       .def_property_readonly("_handle", [](Taskpool& t) { return (uintptr_t)&t; })
       .def("wait", &Taskpool::wait, py::call_guard<py::gil_scoped_release>())

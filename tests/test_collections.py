@@ -96,3 +96,32 @@ def test_stencil_1d(ctx):
     z = (yp[:-2] + yp[1:-1] + yp[2:]) / 3.0
     got = np.concatenate([Src.tile_numpy(t, 0).ravel() for t in range(T)])
     assert np.allclose(got, z), np.abs(got - z).max()
+
+
+def test_irregular_collection(ctx):
+    """hash_datadist analog: arbitrary keys, explicit ranks, DTD-usable
+    (a tiny tree walk: parents sum children, like the haar_tree test)."""
+    import struct
+    coll = pm.IrregularCollection(ctx)
+    # binary tree of 7 nodes, key = index; leaves 3..6 hold values
+    for key in range(7):
+        coll.add(key, 0, 8)
+    tp = pm.Dtd(ctx)
+    for leaf in range(3, 7):
+        coll.bytes_set(leaf, struct.pack("<q", leaf * 10))
+
+    def make_sum(dst, a, b):
+        def body():
+            va = struct.unpack("<q", coll.bytes_get(a))[0]
+            vb = struct.unpack("<q", coll.bytes_get(b))[0]
+            coll.bytes_set(dst, struct.pack("<q", va + vb))
+        return body
+
+    # bottom-up: 1 = 3+4, 2 = 5+6, 0 = 1+2, ordered purely by dataflow
+    for dst, a, b in [(1, 3, 4), (2, 5, 6), (0, 1, 2)]:
+        tp.insert_py(make_sum(dst, a, b),
+                     flows=[(coll.at(a), pm.ACCESS_IN),
+                            (coll.at(b), pm.ACCESS_IN),
+                            (coll.at(dst), pm.ACCESS_OUT)])
+    tp.wait()
+    assert struct.unpack("<q", coll.bytes_get(0))[0] == 30 + 40 + 50 + 60

@@ -37,6 +37,8 @@ def main():
     ap.add_argument("--chore-gemm", type=str, default=os.environ.get(
         "PARSEC_MCA_chore_gemm", "rocblas"), choices=["rocblas", "hip"])
     ap.add_argument("--workers", type=int, default=4)
+    ap.add_argument("--algo", type=str, default="auto",
+                    choices=["auto", "tile", "panel"])
     args = ap.parse_args()
 
     import parsec_amd as pm
@@ -52,8 +54,18 @@ def main():
 
     ctx = pm.init_distributed(nworkers=args.workers)
     assert ctx.world == world
+    # panel granularity (one tall dgemm per update) maximizes per-kernel
+    # efficiency on one GPU; tile granularity exposes the parallelism the
+    # multi-GPU strong-scaling run needs.
+    algo = args.algo
+    if algo == "auto":
+        algo = "panel" if world == 1 else "tile"
     p, q = pick_grid(world)
-    A = pm.TiledMatrix(ctx, n, n, nb, nb, p, q)
+    if algo == "panel":
+        p, q = 1, world
+        A = pm.TiledMatrix(ctx, n, n, n, nb, p, q)
+    else:
+        A = pm.TiledMatrix(ctx, n, n, nb, nb, p, q)
 
     use_torch_dist = world > 1
     if use_torch_dist:
@@ -83,8 +95,12 @@ def main():
         # executes in full.
         tp = pm.Dtd(ctx)
         for _ in range(nsteps):
-            pm.insert_spd_fill(tp, A, 42)
-            pm.insert_potrf(tp, A)
+            if algo == "panel":
+                pm.insert_panel_fill(tp, A, 42)
+                pm.insert_potrf_panel(tp, A)
+            else:
+                pm.insert_spd_fill(tp, A, 42)
+                pm.insert_potrf(tp, A)
         tp.wait()
 
     if args.warmup:
@@ -122,7 +138,7 @@ def main():
                 "model": "tiled_cholesky_dpotrf",
                 "N": n,
                 "tile": nb,
-                "parallelism": f"dtd-2d-block-cyclic-p{p}q{q}",
+                "parallelism": f"dtd-{algo}-p{p}q{q}",
                 "chore_gemm": args.chore_gemm,
                 "gpu": has_gpu,
             },

@@ -33,6 +33,8 @@ from parsec_amd._core import (  # noqa: F401,E402
     insert_apply_scale,
     insert_reduce_sum,
     insert_stencil_1d,
+    insert_panel_fill,
+    insert_potrf_panel,
     param_dump,
     param_set,
 )

@@ -44,5 +44,7 @@ void insert_redistribute(Dtd& tp, TiledMatrix& Src, TiledMatrix& Dst);
 void insert_apply_scale(Dtd& tp, TiledMatrix& A, double alpha, double beta);
 void insert_reduce_sum(Dtd& tp, TiledMatrix& A, TiledMatrix& R);
 void insert_stencil_1d(Dtd& tp, TiledMatrix& Src, TiledMatrix& Dst);
+void insert_panel_fill(Dtd& tp, TiledMatrix& A, uint32_t seed);
+void insert_potrf_panel(Dtd& tp, TiledMatrix& A);
 
 }  // namespace pa

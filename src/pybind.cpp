@@ -290,6 +290,10 @@ PYBIND11_MODULE(_core, m) {
         py::arg("R"), py::call_guard<py::gil_scoped_release>());
   m.def("insert_stencil_1d", &insert_stencil_1d, py::arg("tp"),
         py::arg("src"), py::arg("dst"), py::call_guard<py::gil_scoped_release>());
+  m.def("insert_panel_fill", &insert_panel_fill, py::arg("tp"), py::arg("A"),
+        py::arg("seed") = 42u, py::call_guard<py::gil_scoped_release>());
+  m.def("insert_potrf_panel", &insert_potrf_panel, py::arg("tp"),
+        py::arg("A"), py::call_guard<py::gil_scoped_release>());
   m.def("insert_fill_bf16", &insert_fill_bf16, py::arg("tp"), py::arg("A"),
         py::arg("seed") = 1u, py::call_guard<py::gil_scoped_release>());
   m.def("insert_gemm_bf16", &insert_gemm_bf16, py::arg("tp"), py::arg("At"),

@@ -60,6 +60,10 @@ def main():
     algo = args.algo
     if algo == "auto":
         algo = "panel" if world == 1 else "tile"
+    if algo == "panel":
+        # big kernels keep the chip full from one bulk stream; extra
+        # streams only co-schedule kernels below their solo rate
+        pm.param_set("gpu_exec_streams", "2")
     p, q = pick_grid(world)
     if algo == "panel":
         p, q = 1, world

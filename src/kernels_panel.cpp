@@ -114,11 +114,25 @@ static void gpu_panel_factor(Task& t, GpuTaskCtx& g) {
   }
   const int64_t below = a.m - nb;  // rows under the diagonal block
   if (below > 0) {
+    // TRSM via the inverse (as the tile variant): W = L^{-1} once, then
+    // one full-rate dgemm into scratch + a strided copy back — rocBLAS
+    // dtrsm at this shape decomposes into ~9 TF-effective kernels.
     double* B = diag + nb;
-    PA_CHECK(rocblas_dtrsm(h, rocblas_side_right, rocblas_fill_lower,
-                           rocblas_operation_transpose,
-                           rocblas_diagonal_non_unit, (int)below, nb, &one,
-                           diag, (int)ld, B, (int)ld) == rocblas_status_success);
+    double* W = (double*)g.engine->dev_alloc((size_t)nb * nb * 8);
+    double* X = (double*)g.engine->dev_alloc((size_t)below * nb * 8);
+    g.deferred_frees->emplace_back(W, (size_t)nb * nb * 8);
+    g.deferred_frees->emplace_back(X, (size_t)below * nb * 8);
+    PA_HIP_CHECK(hipMemsetAsync(W, 0, (size_t)nb * nb * 8, g.stream));
+    PA_CHECK(rocblas_dtrtri(h, rocblas_fill_lower, rocblas_diagonal_non_unit,
+                            nb, diag, (int)ld, W, nb) == rocblas_status_success);
+    const double zero = 0.0;
+    PA_CHECK(rocblas_dgemm(h, rocblas_operation_none,
+                           rocblas_operation_transpose, (int)below, nb, nb,
+                           &one, B, (int)ld, W, nb, &zero, X,
+                           (int)below) == rocblas_status_success);
+    PA_HIP_CHECK(hipMemcpy2DAsync(B, (size_t)ld * 8, X, (size_t)below * 8,
+                                  (size_t)below * 8, nb,
+                                  hipMemcpyDeviceToDevice, g.stream));
   }
 }
 

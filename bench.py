@@ -57,9 +57,12 @@ def main():
     # panel granularity (one tall dgemm per update) maximizes per-kernel
     # efficiency on one GPU; tile granularity exposes the parallelism the
     # multi-GPU strong-scaling run needs.
+    # measured: tile granularity wins at every world size (panel's tall
+    # dgemms run below the square-tile rate and its critical path idles
+    # the GPU ~40%; see profiles/ and docs/DESIGN.md)
     algo = args.algo
     if algo == "auto":
-        algo = "panel" if world == 1 else "tile"
+        algo = "tile"
     if algo == "panel":
         # big kernels keep the chip full from one bulk stream; extra
         # streams only co-schedule kernels below their solo rate

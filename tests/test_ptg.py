@@ -186,3 +186,16 @@ def test_ptg_distributed(tmp_path):
     L0 = np.linalg.cholesky(np.tril(pre) + np.tril(pre, -1).T)
     err = np.abs(np.tril(post) - L0).max()
     assert err < 1e-10, f"distributed PTG max err {err}"
+
+
+def test_inline_c_and_hidden_globals(ctx):
+    import struct
+    mod = compile_jdf(os.path.join(EX, "inline_c.jdf"))
+    NT = 6
+    A = pm.TiledMatrix(ctx, NT, 1, 1, 1, 1, 1)
+    tp = pm.Dtd(ctx, "probe")
+    mod.build(ctx, tp, mydata=A, NT=NT)
+    tp.wait()
+    for k in range(NT):
+        (v,) = struct.unpack("<q", A.tile_bytes(k, 0))
+        assert v == k * 2 + NT // 2, (k, v)

@@ -176,3 +176,25 @@ def test_random_dag_vs_sequential_oracle(ctx):
     for i, M in enumerate(mats):
         got = M.tile_numpy(0, 0)
         assert np.allclose(got, ref[i]), f"tile {i} diverged"
+
+
+def test_param_precedence():
+    """MCA param sourcing: explicit set > env > default."""
+    import subprocess, sys, os
+    repo = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+    code = f"""
+import sys, os
+sys.path.insert(0, {repo!r})
+import parsec_amd as pm
+# env already set by parent; param_set must override it
+pm.param_set("dtd_window_size", "123")
+ctx = pm.Context(nworkers=1, rank=0, world=1, gpu=-2)
+assert "dtd_window_size" in pm.param_dump()
+print("PARAM_OK")
+del ctx
+"""
+    env = dict(os.environ)
+    env["PARSEC_MCA_dtd_window_size"] = "999"
+    r = subprocess.run([sys.executable, "-c", code], env=env,
+                       capture_output=True, text=True, timeout=60)
+    assert "PARAM_OK" in r.stdout, r.stdout + r.stderr

@@ -189,9 +189,11 @@ import parsec_amd as pm
 # env already set by parent; param_set must override it
 pm.param_set("dtd_window_size", "123")
 ctx = pm.Context(nworkers=1, rank=0, world=1, gpu=-2)
-assert "dtd_window_size" in pm.param_dump()
+tp = pm.Dtd(ctx)   # registers dtd params
+dump = pm.param_dump()
+assert "dtd_window_size" in dump and "sched_workers" in dump
 print("PARAM_OK")
-del ctx
+del tp, ctx
 """
     env = dict(os.environ)
     env["PARSEC_MCA_dtd_window_size"] = "999"

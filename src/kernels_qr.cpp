@@ -286,15 +286,51 @@ static void gpu_geqrt(Task& t, GpuTaskCtx& g) {
                             a.ld) == rocblas_status_success);
 }
 
+__global__ void k_unitlow(double* V, const double* A, int m, int k, int lda,
+                          int ldv) {
+  // V = unit-lower copy of A's reflector block (zeros above, 1 on diag)
+  int idx = blockIdx.x * blockDim.x + threadIdx.x;
+  int total = m * k;
+  for (; idx < total; idx += gridDim.x * blockDim.x) {
+    int c = idx / m, r = idx - c * m;
+    double v = r < c ? 0.0 : (r == c ? 1.0 : A[(size_t)c * lda + r]);
+    V[(size_t)c * ldv + r] = v;
+  }
+}
+
+// C = (I - V T V^T)^T C via three full-rate dgemms (rocSOLVER's dlarfb
+// decomposes into micro-kernels at these sizes: profiles/RESULTS.md).
+// V: m x k reflectors (unit-lower inside A-storage), T: k x k upper
+// (zeros elsewhere), C: m x n. Scratch: V expanded + W (k x n).
+static void larfb_gemm(GpuTaskCtx& g, int m, int n, int k, const double* A,
+                       int lda, const double* T, int ldt, double* C, int ldc,
+                       int scratch_slot) {
+  rocblas_handle h = qr_handle(g);
+  double* V = qr_scratch(g, scratch_slot, (size_t)m * k * 8);
+  double* W = qr_scratch(g, scratch_slot + 1, (size_t)k * n * 8);
+  hipLaunchKernelGGL(k_unitlow, grid1d(m * k), dim3(256), 0, g.stream, V, A,
+                     m, k, lda, m);
+  const double one = 1.0, zero = 0.0, mone = -1.0;
+  // W = V^T C
+  PA_CHECK(rocblas_dgemm(h, rocblas_operation_transpose,
+                         rocblas_operation_none, k, n, m, &one, V, m, C, ldc,
+                         &zero, W, k) == rocblas_status_success);
+  // W = T^T W  (T upper-triangular, stored dense with zero lower)
+  double* W2 = qr_scratch(g, scratch_slot + 2, (size_t)k * n * 8);
+  PA_CHECK(rocblas_dgemm(h, rocblas_operation_transpose,
+                         rocblas_operation_none, k, n, k, &one, T, ldt, W, k,
+                         &zero, W2, k) == rocblas_status_success);
+  // C -= V W2
+  PA_CHECK(rocblas_dgemm(h, rocblas_operation_none, rocblas_operation_none,
+                         m, n, k, &mone, V, m, W2, k, &one, C,
+                         ldc) == rocblas_status_success);
+}
+
 static void gpu_unmqr(Task& t, GpuTaskCtx& g) {
   const TileArgs& a = t.arg<TileArgs>();
-  PA_CHECK(rocsolver_dlarfb(qr_handle(g), rocblas_side_left,
-                            rocblas_operation_transpose,
-                            rocblas_forward_direction, rocblas_column_wise,
-                            a.m, a.n, a.m, (double*)t.dev_ptr[0], a.ld,
-                            (double*)t.dev_ptr[1], a.ld,
-                            (double*)t.dev_ptr[2], a.ld) ==
-           rocblas_status_success);
+  larfb_gemm(g, a.m, a.n, a.m, (const double*)t.dev_ptr[0], a.ld,
+             (const double*)t.dev_ptr[1], a.ld, (double*)t.dev_ptr[2], a.ld,
+             4);
 }
 
 static void gpu_tsqrt(Task& t, GpuTaskCtx& g) {
@@ -328,11 +364,7 @@ static void gpu_tsmqr(Task& t, GpuTaskCtx& g) {
   double* S = qr_scratch(g, 1, (size_t)2 * nb * cols * 8);
   hipLaunchKernelGGL(k_stack, grid1d(nb * cols), dim3(256), 0, g.stream, S,
                      Ckn, Cmn, nb, cols, ld);
-  PA_CHECK(rocsolver_dlarfb(qr_handle(g), rocblas_side_left,
-                            rocblas_operation_transpose,
-                            rocblas_forward_direction, rocblas_column_wise,
-                            2 * nb, cols, nb, V2, 2 * nb, T1, ld, S,
-                            2 * nb) == rocblas_status_success);
+  larfb_gemm(g, 2 * nb, cols, nb, V2, 2 * nb, T1, ld, S, 2 * nb, 8);
   hipLaunchKernelGGL(k_unstack, grid1d(nb * cols), dim3(256), 0, g.stream, S,
                      Ckn, Cmn, nb, cols, ld);
 }

@@ -172,6 +172,13 @@ class RcclComm : public CommEngine {
       PA_HIP_CHECK(hipMemcpyAsync(d->dev_ptr, d->host_ptr, d->bytes,
                                   hipMemcpyHostToDevice, stream));
       d->dev_valid = true;
+      // fence for any LOCAL GPU-engine consumer of the same tile: it will
+      // wait on this event instead of racing the comm-stream copy
+      if (!d->h2d_event)
+        PA_HIP_CHECK(hipEventCreateWithFlags((hipEvent_t*)&d->h2d_event,
+                                             hipEventDisableTiming));
+      PA_HIP_CHECK(hipEventRecord((hipEvent_t)d->h2d_event, stream));
+      d->h2d_pending = true;
     } else if (d->h2d_pending) {
       // the GPU engine may still be staging this tile on its copy stream
       PA_HIP_CHECK(hipStreamWaitEvent(stream, (hipEvent_t)d->h2d_event, 0));

@@ -29,6 +29,7 @@ from parsec_amd._core import (  # noqa: F401,E402
     insert_fill_bf16,
     insert_gemm_bf16,
     insert_spd_fill,
+    insert_full_fill,
     insert_redistribute,
     insert_apply_scale,
     insert_reduce_sum,

@@ -36,6 +36,7 @@ TaskClass& tc_tsmqr();
 
 // Build the DAGs (DTD insertion; SPMD-safe: call on every rank).
 void insert_spd_fill(Dtd& tp, TiledMatrix& A, uint32_t seed);
+void insert_full_fill(Dtd& tp, TiledMatrix& A, uint32_t seed);
 void insert_potrf(Dtd& tp, TiledMatrix& A);
 void insert_geqrf(Dtd& tp, TiledMatrix& A);
 void insert_fill_bf16(Dtd& tp, TiledMatrix& A, uint32_t seed);

@@ -611,6 +611,23 @@ void insert_spd_fill(Dtd& tp, TiledMatrix& A, uint32_t seed) {
     }
 }
 
+// Full-matrix fill (QR and GEMM inputs read above the diagonal too).
+void insert_full_fill(Dtd& tp, TiledMatrix& A, uint32_t seed) {
+  for (int tm = 0; tm < A.mt(); tm++)
+    for (int tn = 0; tn < A.nt(); tn++) {
+      TileArgs a;
+      a.m = A.tile_rows(tm);
+      a.n = A.tile_cols(tn);
+      a.ld = A.mb();
+      a.i0 = (int64_t)tm * A.mb();
+      a.j0 = (int64_t)tn * A.nb();
+      a.N = A.m();
+      a.seed = seed;
+      Dtd::FlowSpec f[] = {{A.tile(tm, tn), ACCESS_OUT}};
+      tp.insert(&tc_spd_fill(), &a, sizeof(a), f, 1, 0, A.rank_of(tm, tn));
+    }
+}
+
 // Right-looking tiled Cholesky, lower triangular (the reference's headline
 // dpotrf DAG shape; DPLASMA dpotrf_L equivalent).
 void insert_potrf(Dtd& tp, TiledMatrix& A) {

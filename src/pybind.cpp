@@ -277,6 +277,8 @@ PYBIND11_MODULE(_core, m) {
 
   m.def("insert_spd_fill", &insert_spd_fill, py::arg("tp"), py::arg("A"),
         py::arg("seed") = 42u, py::call_guard<py::gil_scoped_release>());
+  m.def("insert_full_fill", &insert_full_fill, py::arg("tp"), py::arg("A"),
+        py::arg("seed") = 42u, py::call_guard<py::gil_scoped_release>());
   m.def("insert_potrf", &insert_potrf, py::arg("tp"), py::arg("A"),
         py::call_guard<py::gil_scoped_release>());
   m.def("insert_geqrf", &insert_geqrf, py::arg("tp"), py::arg("A"),

@@ -296,7 +296,8 @@ static void* stream_scratch(GpuTaskCtx& g, size_t bytes) {
   static thread_local std::map<void*, std::pair<void*, size_t>> bufs;
   auto& e = bufs[(void*)g.stream];
   if (e.second < bytes) {
-    if (e.first) g.engine->dev_free(e.first, e.second);
+    // old buffer may still be in use by earlier kernels on this stream
+    if (e.first) g.deferred_frees->emplace_back(e.first, e.second);
     e.first = g.engine->dev_alloc(bytes);
     e.second = bytes;
   }

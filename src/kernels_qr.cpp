@@ -214,7 +214,9 @@ double* qr_scratch(GpuTaskCtx& g, int slot, size_t bytes) {
                                std::pair<void*, size_t>> bufs;
   auto& e = bufs[{(void*)g.stream, slot}];
   if (e.second < bytes) {
-    if (e.first) g.engine->dev_free(e.first, e.second);
+    // the old buffer may still be referenced by earlier kernels on this
+    // stream: return it to the pool only after this task retires
+    if (e.first) g.deferred_frees->emplace_back(e.first, e.second);
     e.first = g.engine->dev_alloc(bytes);
     e.second = bytes;
   }
@@ -272,10 +274,170 @@ inline dim3 grid1d(int total) {
 
 }  // namespace
 
+// ---------------------------------------------------------- hand panel QR
+// rocSOLVER's dgeqrf at tile sizes runs an unblocked host-synced column
+// loop (~350 us of idle per column; profiles/RESULTS.md). These kernels do
+// the panel factorization device-side:
+//  - k_geqr2: one workgroup factors an (m x 128) panel column-by-column
+//    (norm reduce + scale + wave-parallel trailing update, all in-kernel);
+//  - T factors come from G = V^T V (one full-rate dgemm) via k_larft_diag
+//    (per-128-block upper-triangular recurrence) and a blocked combine
+//    (two dgemms per block column).
+__global__ void __launch_bounds__(1024) k_geqr2(double* A, int m, int ncols,
+                                                int ld, double* tau) {
+  __shared__ double red[1024];
+  __shared__ double bc[3];  // beta, tau_j, scal
+  const int tid = threadIdx.x;
+  const int wave = tid >> 6, lane = tid & 63;
+  for (int j = 0; j < ncols; j++) {
+    // 1) norm^2 of the sub-diagonal part of column j
+    double acc = 0;
+    for (int i = j + 1 + tid; i < m; i += 1024) {
+      double v = A[(size_t)j * ld + i];
+      acc += v * v;
+    }
+    red[tid] = acc;
+    __syncthreads();
+    for (int s = 512; s > 0; s >>= 1) {
+      if (tid < s) red[tid] += red[tid + s];
+      __syncthreads();
+    }
+    if (tid == 0) {
+      double alpha = A[(size_t)j * ld + j];
+      double nrm2 = red[0];
+      if (nrm2 == 0.0) {
+        bc[0] = alpha;
+        bc[1] = 0.0;
+        bc[2] = 0.0;
+        tau[j] = 0.0;
+      } else {
+        double beta = -copysign(sqrt(alpha * alpha + nrm2), alpha);
+        bc[0] = beta;
+        bc[1] = (beta - alpha) / beta;
+        bc[2] = 1.0 / (alpha - beta);
+        tau[j] = bc[1];
+      }
+    }
+    __syncthreads();
+    const double tau_j = bc[1], scal = bc[2];
+    if (tau_j != 0.0) {
+      for (int i = j + 1 + tid; i < m; i += 1024)
+        A[(size_t)j * ld + i] *= scal;
+      if (tid == 0) A[(size_t)j * ld + j] = bc[0];
+    }
+    __syncthreads();
+    if (tau_j != 0.0) {
+      // 2) apply H_j to trailing columns, one wave per column round-robin
+      for (int c = j + 1 + wave; c < ncols; c += 16) {
+        double dot = (lane == 0) ? A[(size_t)c * ld + j] : 0.0;
+        for (int i = j + 1 + lane; i < m; i += 64)
+          dot += A[(size_t)j * ld + i] * A[(size_t)c * ld + i];
+        for (int s = 32; s > 0; s >>= 1) dot += __shfl_down(dot, s);
+        dot = __shfl(dot, 0);
+        double w = tau_j * dot;
+        if (lane == 0) A[(size_t)c * ld + j] -= w;
+        for (int i = j + 1 + lane; i < m; i += 64)
+          A[(size_t)c * ld + i] -= w * A[(size_t)j * ld + i];
+      }
+    }
+    __syncthreads();
+  }
+}
+
+// T diag blocks: for block b, columns j: T(0:j,j) = -tau_j T(0:j,0:j) g
+// where g = G(blk rows, j) restricted to the block. G is V^T V of the
+// unit-lower V; T upper-triangular, T(j,j) = tau_j.
+__global__ void k_larft_diag(const double* G, int ldg, const double* tau,
+                             double* T, int ldt, int k) {
+  const int b0 = blockIdx.x * 128;
+  const int tid = threadIdx.x;  // 128 threads
+  const int nb = min(128, k - b0);
+  for (int j = 0; j < nb; j++) {
+    int gj = b0 + j;
+    // col = -tau_j * T(0:j,0:j) * G(b0..b0+j, gj)
+    double s = 0;
+    if (tid < j) {
+      for (int q = tid; q < j; q++)
+        s += T[(size_t)(b0 + q) * ldt + b0 + tid] *
+             G[(size_t)gj * ldg + b0 + q];
+      s *= -tau[gj];
+    }
+    __syncthreads();
+    if (tid < j) T[(size_t)gj * ldt + b0 + tid] = s;
+    if (tid == j) T[(size_t)gj * ldt + gj] = tau[gj];
+    __syncthreads();
+  }
+}
+
+// Full tile/stacked-panel QR: factor 128-wide panels with k_geqr2, apply
+// to the trailing columns with gemm-larfb, then build the full k x k T.
+static void qr_factor_hand(GpuTaskCtx& g, double* A, int m, int k, int ld,
+                           double* T, int ldt, int slot0) {
+  rocblas_handle h = qr_handle(g);
+  double* tau = qr_scratch(g, slot0, (size_t)k * 8);
+  const double one = 1.0, zero = 0.0, mone = -1.0;
+  PA_HIP_CHECK(hipMemsetAsync(T, 0, (size_t)ldt * k * 8, g.stream));
+  for (int p = 0; p < k; p += 128) {
+    int pc = std::min(128, k - p);
+    double* panel = A + (size_t)p * ld + p;
+    int prows = m - p;
+    hipLaunchKernelGGL(k_geqr2, dim3(1), dim3(1024), 0, g.stream, panel,
+                       prows, pc, ld, tau + p);
+    int rest = k - p - pc;
+    if (rest > 0) {
+      // T128 for this panel from G128 = V^T V
+      double* V = qr_scratch(g, slot0 + 1, (size_t)prows * 128 * 8);
+      double* G = qr_scratch(g, slot0 + 2, (size_t)128 * 128 * 8);
+      hipLaunchKernelGGL(k_unitlow, grid1d(prows * pc), dim3(256), 0,
+                         g.stream, V, panel, prows, pc, ld, prows);
+      PA_CHECK(rocblas_dgemm(h, rocblas_operation_transpose,
+                             rocblas_operation_none, pc, pc, prows, &one, V,
+                             prows, V, prows, &zero, G,
+                             pc) == rocblas_status_success);
+      double* T128 = qr_scratch(g, slot0 + 3, (size_t)128 * 128 * 8);
+      PA_HIP_CHECK(hipMemsetAsync(T128, 0, (size_t)pc * pc * 8, g.stream));
+      hipLaunchKernelGGL(k_larft_diag, dim3(1), dim3(128), 0, g.stream, G,
+                         pc, tau + p, T128, pc, pc);
+      larfb_gemm(g, prows, rest, pc, panel, ld, T128, pc,
+                 A + (size_t)(p + pc) * ld + p, ld, slot0 + 4);
+    }
+  }
+  // Full T: G = V^T V over all k columns, diag blocks in one launch,
+  // off-diagonal blocks by blocked combine (T[0:J,blk] = -T·G·T_blk).
+  double* V = qr_scratch(g, slot0 + 1, (size_t)m * k * 8);
+  double* G = qr_scratch(g, slot0 + 7, (size_t)k * k * 8);
+  hipLaunchKernelGGL(k_unitlow, grid1d(m * k), dim3(256), 0, g.stream, V, A,
+                     m, k, ld, m);
+  PA_CHECK(rocblas_dgemm(h, rocblas_operation_transpose,
+                         rocblas_operation_none, k, k, m, &one, V, m, V, m,
+                         &zero, G, k) == rocblas_status_success);
+  hipLaunchKernelGGL(k_larft_diag, dim3((k + 127) / 128), dim3(128), 0,
+                     g.stream, G, k, tau, T, ldt, k);
+  double* X = qr_scratch(g, slot0 + 8, (size_t)k * 128 * 8);
+  for (int b = 128; b < k; b += 128) {
+    int bc2 = std::min(128, k - b);
+    // X = T(0:b,0:b) * G(0:b, b:b+bc)
+    PA_CHECK(rocblas_dgemm(h, rocblas_operation_none, rocblas_operation_none,
+                           b, bc2, b, &one, T, ldt, G + (size_t)b * k, k,
+                           &zero, X, b) == rocblas_status_success);
+    // T(0:b, blk) = -X * T_blk
+    PA_CHECK(rocblas_dgemm(h, rocblas_operation_none, rocblas_operation_none,
+                           b, bc2, bc2, &mone, X, b,
+                           T + (size_t)b * ldt + b, ldt, &zero,
+                           T + (size_t)b * ldt, ldt) == rocblas_status_success);
+  }
+}
+
 static void gpu_geqrt(Task& t, GpuTaskCtx& g) {
   const TileArgs& a = t.arg<TileArgs>();
   double* A = (double*)t.dev_ptr[0];
   double* T = (double*)t.dev_ptr[1];
+  static const bool use_rocsolver =
+      param_str("chore_qr", "hand") == "rocsolver";
+  if (!use_rocsolver) {
+    qr_factor_hand(g, A, a.n, a.n, a.ld, T, a.ld, 0);
+    return;
+  }
   double* tau = qr_scratch(g, 0, (size_t)a.n * 8);
   rocblas_handle h = qr_handle(g);
   PA_HIP_CHECK(hipMemsetAsync(T, 0, t.flows[1].data->bytes, g.stream));
@@ -340,16 +502,22 @@ static void gpu_tsqrt(Task& t, GpuTaskCtx& g) {
   double* Amk = (double*)t.dev_ptr[1];
   double* V2 = (double*)t.dev_ptr[2];
   double* T1 = (double*)t.dev_ptr[3];
-  double* tau = qr_scratch(g, 0, (size_t)nb * 8);
   rocblas_handle h = qr_handle(g);
   hipLaunchKernelGGL(k_stack_triu, grid1d(nb * nb), dim3(256), 0, g.stream,
                      V2, Akk, Amk, nb, ld);
-  PA_HIP_CHECK(hipMemsetAsync(T1, 0, t.flows[3].data->bytes, g.stream));
-  PA_CHECK(rocsolver_dgeqrf(h, 2 * nb, nb, V2, 2 * nb, tau) ==
-           rocblas_status_success);
-  PA_CHECK(rocsolver_dlarft(h, rocblas_forward_direction,
-                            rocblas_column_wise, 2 * nb, nb, V2, 2 * nb, tau,
-                            T1, ld) == rocblas_status_success);
+  static const bool use_rocsolver =
+      param_str("chore_qr", "hand") == "rocsolver";
+  if (!use_rocsolver) {
+    qr_factor_hand(g, V2, 2 * nb, nb, 2 * nb, T1, ld, 0);
+  } else {
+    double* tau = qr_scratch(g, 0, (size_t)nb * 8);
+    PA_HIP_CHECK(hipMemsetAsync(T1, 0, t.flows[3].data->bytes, g.stream));
+    PA_CHECK(rocsolver_dgeqrf(h, 2 * nb, nb, V2, 2 * nb, tau) ==
+             rocblas_status_success);
+    PA_CHECK(rocsolver_dlarft(h, rocblas_forward_direction,
+                              rocblas_column_wise, 2 * nb, nb, V2, 2 * nb,
+                              tau, T1, ld) == rocblas_status_success);
+  }
   hipLaunchKernelGGL(k_copy_triu, grid1d(nb * nb), dim3(256), 0, g.stream,
                      Akk, V2, nb, 2 * nb, ld);
 }

@@ -274,6 +274,12 @@ inline dim3 grid1d(int total) {
 
 }  // namespace
 
+__global__ void k_unitlow(double* V, const double* A, int m, int k, int lda,
+                          int ldv);
+static void larfb_gemm(GpuTaskCtx& g, int m, int n, int k, const double* A,
+                       int lda, const double* T, int ldt, double* C, int ldc,
+                       int scratch_slot);
+
 // ---------------------------------------------------------- hand panel QR
 // rocSOLVER's dgeqrf at tile sizes runs an unblocked host-synced column
 // loop (~350 us of idle per column; profiles/RESULTS.md). These kernels do

@@ -439,7 +439,7 @@ static void gpu_geqrt(Task& t, GpuTaskCtx& g) {
   double* A = (double*)t.dev_ptr[0];
   double* T = (double*)t.dev_ptr[1];
   static const bool use_rocsolver =
-      param_str("chore_qr", "hand") == "rocsolver";
+      param_str("chore_qr", "rocsolver") == "rocsolver";
   if (!use_rocsolver) {
     qr_factor_hand(g, A, a.n, a.n, a.ld, T, a.ld, 0);
     return;
@@ -512,7 +512,7 @@ static void gpu_tsqrt(Task& t, GpuTaskCtx& g) {
   hipLaunchKernelGGL(k_stack_triu, grid1d(nb * nb), dim3(256), 0, g.stream,
                      V2, Akk, Amk, nb, ld);
   static const bool use_rocsolver =
-      param_str("chore_qr", "hand") == "rocsolver";
+      param_str("chore_qr", "rocsolver") == "rocsolver";
   if (!use_rocsolver) {
     qr_factor_hand(g, V2, 2 * nb, nb, 2 * nb, T1, ld, 0);
   } else {

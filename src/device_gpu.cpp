@@ -96,7 +96,8 @@ void GpuEngine::event_put(hipEvent_t e) { event_pool_.push_back(e); }
 
 void* GpuEngine::dev_alloc(size_t bytes) {
   bytes = (bytes + 255) & ~size_t(255);
-  for (int attempt = 0; attempt < 10000; attempt++) {
+  const double deadline = now_s() + 60.0;  // patience under pressure
+  while (now_s() < deadline) {
     {
       std::lock_guard<std::mutex> g(mem_mtx_);
       auto it = free_lists_.find(bytes);

@@ -21,6 +21,7 @@ constexpr int COMM_PRIORITY = 1 << 28;  // transfers go out as early as possible
 Dtd::Dtd(Context* ctx, std::string name) : Taskpool(ctx, std::move(name)) {
   me_ = ctx->rank();
   world_ = ctx->world();
+  PA_CHECK(world_ <= 64, "sent_mask is a 64-bit rank bitmap");
   chan_seq_.assign((size_t)world_ * world_, 0);
   window_ = param_int("dtd_window_size", 16384);
   threshold_ = param_int("dtd_threshold_size", 8192);

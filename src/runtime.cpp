@@ -5,6 +5,9 @@
 #include "comm.hpp"
 #include "profiling.hpp"
 
+#include <pthread.h>
+#include <unistd.h>
+
 namespace pa {
 
 thread_local int Context::tls_worker_id = -1;
@@ -262,6 +265,21 @@ Context::~Context() {
 
 void Context::worker_main(int id) {
   tls_worker_id = id;
+  // Optional linear core binding (bindthread.c/vpmap analog). Default off:
+  // with one process per GPU the OS spreads ranks well, and naive binding
+  // would stack all ranks' worker 0 on core 0. sched_bind=1 binds at
+  // core = (local_rank * (nworkers+2) + id) % ncores.
+  if (param_int("sched_bind", 0)) {
+    const char* lr = getenv("LOCAL_RANK");
+    int local = lr ? atoi(lr) : rank_;
+    long ncores = sysconf(_SC_NPROCESSORS_ONLN);
+    if (ncores > 0) {
+      cpu_set_t set;
+      CPU_ZERO(&set);
+      CPU_SET((local * (nworkers_ + 2) + id) % ncores, &set);
+      pthread_setaffinity_np(pthread_self(), sizeof(set), &set);
+    }
+  }
   while (!stop_.load(std::memory_order_acquire)) {
     Task* t = sched_->pop(id);
     if (t) {

@@ -170,6 +170,7 @@ class TcpComm : public CommEngine {
       uint64_t key = ((uint64_t)t->peer << 48) | t->comm_seq;
       auto it = unexpected_.find(key);
       if (it != unexpected_.end()) {
+        d->begin_host_overwrite();
         memcpy(d->ensure_host(), it->second.data(), d->bytes);
         unexpected_.erase(it);
         d->written_on(false);
@@ -242,7 +243,9 @@ class TcpComm : public CommEngine {
           if (it != posted_recv_.end()) {
             p.in_task = it->second;
             posted_recv_.erase(it);
-            p.in_direct = (uint8_t*)p.in_task->flows[0].data->ensure_host();
+            Data* rd = p.in_task->flows[0].data;
+            rd->begin_host_overwrite();
+            p.in_direct = (uint8_t*)rd->ensure_host();
           } else {
             p.in_task = nullptr;
             p.in_direct = nullptr;

@@ -64,6 +64,11 @@ struct Data {
   void* pull_to_host();
   // Invalidate all copies except the one on `device` (true=GPU).
   void written_on(bool device);
+  // Call BEFORE overwriting the host buffer outside the GPU engine (CPU
+  // OUTPUT-only bodies, comm recv, external setters): drops the device
+  // copy's validity so a concurrent dirty-eviction writeback cannot land
+  // on top of the in-progress host write.
+  void begin_host_overwrite();
 };
 
 // 2D block-cyclic tiled matrix of an elementary type (fp64 for the Cholesky

@@ -391,6 +391,11 @@ void* Data::pull_to_host() {
   return host_ptr;
 }
 
+void Data::begin_host_overwrite() {
+  SpinGuard g(lock);
+  dev_valid = false;
+}
+
 void Data::written_on(bool device) {
   SpinGuard g(lock);
   if (device) {

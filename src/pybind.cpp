@@ -174,6 +174,7 @@ PYBIND11_MODULE(_core, m) {
                                 py::array_t<double> v) {
         PA_CHECK(A.is_local(tm, tn), "tile_numpy_set: tile is not local");
         Data* d = A.tile(tm, tn);
+        d->begin_host_overwrite();
         double* p = (double*)d->ensure_host();
         auto r = v.unchecked<2>();
         int rows = A.tile_rows(tm), cols = A.tile_cols(tn);
@@ -195,6 +196,7 @@ PYBIND11_MODULE(_core, m) {
       .def("tile_bytes_set", [](TiledMatrix& A, int tm, int tn, py::bytes b) {
         PA_CHECK(A.is_local(tm, tn));
         Data* d = A.tile(tm, tn);
+        d->begin_host_overwrite();
         std::string s(b);
         PA_CHECK(s.size() <= d->bytes);
         memcpy(d->ensure_host(), s.data(), s.size());
@@ -219,6 +221,7 @@ PYBIND11_MODULE(_core, m) {
       })
       .def("bytes_set", [](IrregularCollection& c, uint64_t key, py::bytes b) {
         Data* d = c.at(key);
+        d->begin_host_overwrite();
         std::string s2(b);
         PA_CHECK(s2.size() <= d->bytes);
         memcpy(d->ensure_host(), s2.data(), s2.size());

@@ -68,6 +68,11 @@ void task_complete(Task* t) {
 
 void run_cpu_task(Task* t) {
   Profiler& pr = Profiler::inst();
+  // OUTPUT-only flows are about to be produced on the host: drop stale
+  // device validity first so LRU writeback cannot race the body's writes.
+  for (int i = 0; i < t->nflows; i++)
+    if (t->flows[i].data && t->flows[i].mode == ACCESS_OUT)
+      t->flows[i].data->begin_host_overwrite();
   if (t->tc->cpu_hook) {
     if (pr.enabled()) {
       uint64_t t0 = Profiler::now_ns();

@@ -139,6 +139,12 @@ void GpuEngine::note_resident(Data* d) {
   resident_.push_back(d);
 }
 
+void GpuEngine::forget(Data* d) {
+  std::lock_guard<std::mutex> g(mem_mtx_);
+  resident_.erase(std::remove(resident_.begin(), resident_.end(), d),
+                  resident_.end());
+}
+
 void GpuEngine::pin(Data* d) {
   d->dev_refs++;
   d->dev_last_use = lru_clock_.fetch_add(1);
@@ -363,10 +369,14 @@ Data::~Data() {
   // Drop DTD chaining references to completed tasks.
   if (last_local_writer) last_local_writer->release();
   for (Task* r : local_readers) r->release();
+  Context* c = coll ? coll->ctx() : ctx_direct;
+  GpuEngine* eng = c ? c->gpu() : nullptr;
+  if (eng) {
+    eng->forget(this);  // never leave a dangling pointer in the LRU set
+    if (dev_ptr) eng->dev_free(dev_ptr, bytes);
+  }
   if (h2d_event) hipEventDestroy((hipEvent_t)h2d_event);
   if (host_ptr) free(host_ptr);
-  // dev_ptr returns to the engine pool with the collection teardown;
-  // the slab itself is freed by the engine.
 }
 
 void* Data::ensure_host() {
@@ -418,18 +428,9 @@ TiledMatrix::TiledMatrix(Context* ctx, int64_t m, int64_t n, int mb, int nb,
   tiles_.resize((size_t)mt_ * nt_);
 }
 
-TiledMatrix::~TiledMatrix() {
-  GpuEngine* eng = ctx_->gpu();
-  for (auto& t : tiles_) {
-    if (t && t->dev_ptr && eng) eng->dev_free(t->dev_ptr, t->bytes);
-  }
-}
+TiledMatrix::~TiledMatrix() = default;  // ~Data returns device buffers
 
-IrregularCollection::~IrregularCollection() {
-  GpuEngine* eng = ctx_->gpu();
-  for (auto& [k, d] : map_)
-    if (d->dev_ptr && eng) eng->dev_free(d->dev_ptr, d->bytes);
-}
+IrregularCollection::~IrregularCollection() = default;  // ~Data handles it
 
 Data* IrregularCollection::add(uint64_t key, int rank, size_t bytes) {
   auto& slot = map_[key];

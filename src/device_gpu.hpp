@@ -61,6 +61,7 @@ class GpuEngine {
   void* dev_alloc(size_t bytes);
   void dev_free(void* p, size_t bytes);
   void note_resident(Data* d);   // track for eviction
+  void forget(Data* d);          // drop from residency (Data teardown)
   void pin(Data* d);             // dev_refs++ under d->lock
   void unpin(Data* d);
 

@@ -157,13 +157,12 @@ void GpuEngine::unpin(Data* d) {
 }
 
 bool GpuEngine::evict_one(size_t) {
-  // candidates collected without their locks (lock order: Data::lock may
-  // be held by a thread calling dev_alloc, so only try_lock here)
-  std::vector<Data*> cand;
-  {
-    std::lock_guard<std::mutex> g(mem_mtx_);
-    cand = resident_;
-  }
+  // The whole scan holds mem_mtx_: a Data being destroyed must first pass
+  // forget() (which also takes mem_mtx_), so no candidate can dangle.
+  // Tile locks are only try_lock'd (their holders may take mem_mtx_ via
+  // dev_free, and try_lock never blocks, so no lock-order deadlock).
+  std::lock_guard<std::mutex> g(mem_mtx_);
+  std::vector<Data*> cand = resident_;
   std::sort(cand.begin(), cand.end(), [](Data* a, Data* b) {
     return a->dev_last_use < b->dev_last_use;
   });
@@ -188,12 +187,9 @@ bool GpuEngine::evict_one(size_t) {
     d->dev_ptr = nullptr;
     d->dev_valid = false;
     d->lock.unlock();
-    {
-      std::lock_guard<std::mutex> g(mem_mtx_);
-      free_lists_[(d->bytes + 255) & ~size_t(255)].push_back(buf);
-      resident_.erase(std::remove(resident_.begin(), resident_.end(), d),
-                      resident_.end());
-    }
+    free_lists_[(d->bytes + 255) & ~size_t(255)].push_back(buf);
+    resident_.erase(std::remove(resident_.begin(), resident_.end(), d),
+                    resident_.end());
     PA_DEBUG(2, "evicted tile %lu (%zu bytes)", (unsigned long)d->key,
              d->bytes);
     return true;

@@ -32,3 +32,8 @@ def ctx():
     c = pm.Context(nworkers=2, rank=0, world=1)
     yield c
     del c
+
+
+def port_base(salt=0):
+    """Per-process port base: avoids TIME_WAIT/parallel-run collisions."""
+    return 20000 + ((os.getpid() * 131 + salt * 977) % 20000)

@@ -95,7 +95,7 @@ sys.path.insert(0, {repo!r})
 import numpy as np
 import parsec_amd as pm
 rank = int(os.environ["RANK"]); world = 2
-pm.param_set("comm_base_port", "29960")
+pm.param_set("comm_base_port", os.environ["PARSEC_TEST_PORT"])
 ctx = pm.Context(nworkers=2, rank=rank, world=world, comm="tcp", gpu=-2)
 K, M, N, kb, mb = 128, 128, 128, 64, 64
 At = pm.TiledMatrix(ctx, K, M, kb, mb, 2, 1, elem_size=2)
@@ -129,6 +129,7 @@ del At, B, C, ctx
         env = dict(os.environ)
         env["RANK"] = str(r)
         env["WORLD_SIZE"] = "2"
+        env["PARSEC_TEST_PORT"] = str(__import__("conftest").port_base(7))
         procs.append(subprocess.Popen([sys.executable, "-c", code], env=env,
                                       stdout=subprocess.PIPE,
                                       stderr=subprocess.STDOUT))

@@ -15,7 +15,9 @@ import pytest
 HERE = os.path.dirname(os.path.abspath(__file__))
 WORKER = os.path.join(HERE, "_dist_worker.py")
 
-_next_port = [29760]
+from conftest import port_base
+
+_next_port = [port_base()]
 
 
 def run_world(world, tmpdir, p, q, n=256, nb=64):

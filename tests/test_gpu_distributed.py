@@ -16,6 +16,8 @@ import pytest
 
 pytestmark = pytest.mark.gpu
 
+from conftest import port_base
+
 HERE = os.path.dirname(os.path.abspath(__file__))
 WORKER = os.path.join(HERE, "_dist_worker.py")
 
@@ -26,7 +28,7 @@ def test_gpu_distributed_tcp(tmp_path):
     for r in range(world):
         env = dict(os.environ)
         env.update(RANK=str(r), WORLD_SIZE=str(world),
-                   PARSEC_TEST_PORT="29850", PARSEC_TEST_OUT=str(tmp_path),
+                   PARSEC_TEST_PORT=str(port_base(3)), PARSEC_TEST_OUT=str(tmp_path),
                    GRID_P=str(p), GRID_Q=str(q), MAT_N=str(n), MAT_NB=str(nb),
                    PARSEC_TEST_GPU="1")
         procs.append(subprocess.Popen([sys.executable, WORKER], env=env,

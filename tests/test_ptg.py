@@ -164,7 +164,7 @@ def test_ptg_distributed(tmp_path):
     for r in range(world):
         env = dict(os.environ)
         env.update(RANK=str(r), WORLD_SIZE=str(world),
-                   PARSEC_TEST_PORT="29940", PARSEC_TEST_OUT=str(tmp_path),
+                   PARSEC_TEST_PORT=str(__import__("conftest").port_base(5)), PARSEC_TEST_OUT=str(tmp_path),
                    GRID_P="2", GRID_Q="1", MAT_N=str(n), MAT_NB=str(nb),
                    PARSEC_TEST_APP="ptg")
         procs.append(subprocess.Popen([_sys.executable, worker], env=env,

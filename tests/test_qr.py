@@ -56,3 +56,43 @@ def test_qr_gpu():
     assert err < 1e-12, f"GPU QR rel err {err}"
     del A
     del ctx
+
+
+@pytest.mark.gpu
+def test_qr_gpu_hand_chore():
+    """Hand device-side panel path (PARSEC_MCA_chore_qr=hand) numerics."""
+    import os
+    import subprocess
+    import sys
+    repo = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+    code = f"""
+import sys; sys.path.insert(0, {repo!r})
+import numpy as np
+import parsec_amd as pm
+ctx = pm.Context(nworkers=2, rank=0, world=1)
+n, nb = 1024, 256
+A = pm.TiledMatrix(ctx, n, n, nb, nb, 1, 1)
+rng = np.random.default_rng(3)
+for tm in range(A.mt):
+    for tn in range(A.nt):
+        A.tile_numpy_set(tm, tn, rng.standard_normal((nb, nb)))
+A0 = np.zeros((n, n))
+for tm in range(A.mt):
+    for tn in range(A.nt):
+        A0[tm*nb:(tm+1)*nb, tn*nb:(tn+1)*nb] = A.tile_numpy(tm, tn)
+tp = pm.Dtd(ctx); pm.insert_geqrf(tp, A); tp.wait()
+R = np.zeros((n, n))
+for tm in range(A.mt):
+    for tn in range(A.nt):
+        R[tm*nb:(tm+1)*nb, tn*nb:(tn+1)*nb] = A.tile_numpy(tm, tn)
+R = np.triu(R)
+err = np.abs(R.T @ R - A0.T @ A0).max() / np.abs(A0.T @ A0).max()
+print("HAND_QR_ERR", err)
+assert err < 1e-12, err
+del A, ctx
+"""
+    env = dict(os.environ)
+    env["PARSEC_MCA_chore_qr"] = "hand"
+    r = subprocess.run([sys.executable, "-c", code], env=env,
+                       capture_output=True, text=True, timeout=300)
+    assert r.returncode == 0, r.stdout + r.stderr

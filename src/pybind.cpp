@@ -36,16 +36,46 @@ namespace {
 // ---- Python CPU task bodies (for DTD semantics tests) ----
 struct PyTaskPayload {
   PyObject* fn;
+  int with_data;  // pass per-flow host buffers as memoryviews
 };
 
 void py_task_hook(Task& t) {
+  const PyTaskPayload& pl = t.arg<PyTaskPayload>();
+  // materialize host buffers OUTSIDE the GIL (may sync/copy from device)
+  void* bufs[MAX_FLOWS] = {};
+  if (pl.with_data) {
+    for (int i = 0; i < t.nflows; i++) {
+      Data* d = t.flows[i].data;
+      if (!d) continue;
+      if (t.flows[i].mode & ACCESS_IN) {
+        bufs[i] = d->pull_to_host();
+      } else {
+        d->begin_host_overwrite();
+        bufs[i] = d->ensure_host();
+      }
+    }
+  }
   py::gil_scoped_acquire gil;
-  py::handle fn(t.arg<PyTaskPayload>().fn);
+  py::handle fn(pl.fn);
   try {
-    fn();
+    if (pl.with_data) {
+      py::tuple args(t.nflows);
+      for (int i = 0; i < t.nflows; i++) {
+        Data* d = t.flows[i].data;
+        args[i] = d ? py::memoryview::from_memory(bufs[i], (ssize_t)d->bytes)
+                    : py::none();
+      }
+      fn(*args);
+    } else {
+      fn();
+    }
   } catch (py::error_already_set& e) {
     fprintf(stderr, "[parsec_amd] python task raised: %s\n", e.what());
   }
+  // OUT flows were (re)written on the host by the body
+  for (int i = 0; i < t.nflows && pl.with_data; i++)
+    if (t.flows[i].data && (t.flows[i].mode & ACCESS_OUT))
+      t.flows[i].data->written_on(false);
 }
 
 void py_task_destruct(Task& t) {
@@ -258,8 +288,8 @@ PYBIND11_MODULE(_core, m) {
       .def("insert_py",
            [](Dtd& tp, py::function fn,
               std::vector<std::pair<Data*, int>> flows, int priority,
-              int rank) {
-             PyTaskPayload pl{fn.ptr()};
+              int rank, bool with_data) {
+             PyTaskPayload pl{fn.ptr(), with_data ? 1 : 0};
              Py_XINCREF(pl.fn);
              std::vector<Dtd::FlowSpec> fs;
              for (auto& [d, mode] : flows)
@@ -269,7 +299,8 @@ PYBIND11_MODULE(_core, m) {
                        (int)fs.size(), priority, rank);
            },
            py::arg("fn"), py::arg("flows") = std::vector<std::pair<Data*, int>>{},
-           py::arg("priority") = 0, py::arg("rank") = -1)
+           py::arg("priority") = 0, py::arg("rank") = -1,
+           py::arg("with_data") = false)
       .def("flush", &Dtd::flush, py::call_guard<py::gil_scoped_release>())
       .def("flush_all", &Dtd::flush_all,
            py::call_guard<py::gil_scoped_release>());

@@ -201,3 +201,32 @@ del tp, ctx
     r = subprocess.run([sys.executable, "-c", code], env=env,
                        capture_output=True, text=True, timeout=60)
     assert "PARAM_OK" in r.stdout, r.stdout + r.stderr
+
+
+def test_py_task_with_data(ctx):
+    """Python bodies receive tile buffers (reference DTD bodies get tile
+    pointers): a numpy-implemented DAG over real tile data."""
+    import numpy as np
+    nb = 32
+    A = pm.TiledMatrix(ctx, nb, nb, nb, nb, 1, 1)
+    B = pm.TiledMatrix(ctx, nb, nb, nb, nb, 1, 1)
+    tp = pm.Dtd(ctx)
+
+    def fill(buf):
+        np.frombuffer(buf, dtype=np.float64)[:] = 3.0
+
+    def double_into(src, dst):
+        d = np.frombuffer(dst, dtype=np.float64)
+        d[:] = np.frombuffer(src, dtype=np.float64) * 2.0
+
+    def add_one(buf):
+        np.frombuffer(buf, dtype=np.float64)[:] += 1.0
+
+    tp.insert_py(fill, flows=[(A.tile(0, 0), pm.ACCESS_OUT)], with_data=True)
+    tp.insert_py(double_into, flows=[(A.tile(0, 0), pm.ACCESS_IN),
+                                     (B.tile(0, 0), pm.ACCESS_OUT)],
+                 with_data=True)
+    tp.insert_py(add_one, flows=[(B.tile(0, 0), pm.ACCESS_INOUT)],
+                 with_data=True)
+    tp.wait()
+    assert np.allclose(B.tile_numpy(0, 0), 7.0)

@@ -62,8 +62,10 @@ void py_task_hook(Task& t) {
       py::tuple args(t.nflows);
       for (int i = 0; i < t.nflows; i++) {
         Data* d = t.flows[i].data;
-        args[i] = d ? py::memoryview::from_memory(bufs[i], (ssize_t)d->bytes)
-                    : py::none();
+        if (d)
+          args[i] = py::memoryview::from_memory(bufs[i], (ssize_t)d->bytes);
+        else
+          args[i] = py::none();
       }
       fn(*args);
     } else {

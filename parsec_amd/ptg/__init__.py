@@ -41,11 +41,12 @@ class JdfError(Exception):
 
 # --------------------------------------------------------------- parsing
 class Dep:
-    def __init__(self, direction, guard, term, else_term):
+    def __init__(self, direction, guard, term, else_term, props=None):
         self.direction = direction      # '<-' or '->'
         self.guard = guard              # expr str or None
         self.term = term                # ('coll', name, args) | ('task', flow, cls, args) | ('new',) | ('null',)
         self.else_term = else_term      # same or None
+        self.props = props or {}        # [key=val ...] properties
 
 
 class Flow:
@@ -149,6 +150,11 @@ def _parse_term(s):
 
 def _parse_dep(direction, s):
     s = s.strip()
+    props = {}
+    m = re.search(r"\[([^\]]*)\]\s*$", s)
+    if m:
+        props = _parse_props("[" + m.group(1) + "]")
+        s = s[:m.start()].strip()
     guard = None
     if s.startswith("("):
         depth, j = 0, 0
@@ -166,7 +172,7 @@ def _parse_dep(direction, s):
     parts = _split_top(s, [":"])
     term = _parse_term(parts[0])
     else_term = _parse_term(parts[1]) if len(parts) > 1 and parts[1].strip() else None
-    return Dep(direction, guard, term, else_term)
+    return Dep(direction, guard, term, else_term, props)
 
 
 def parse_jdf(text):
@@ -352,8 +358,19 @@ def generate_cpp(jdf, name):
             s += f"  long {lname} = (long)({_cxx_expr(lexpr)}); (void){lname};\n"
         return s
 
-    def term_tile(term):
+    def term_tile(term, props=None, cls=None):
         kind = term[0]
+        if kind == "new":
+            size = (props or {}).get("size")
+            if size is None:
+                raise JdfError("NEW tile needs a [size=bytes] property")
+            pt = cls.partition
+            a0 = _cxx_expr(pt[2][0])
+            a1 = _cxx_expr(pt[2][1]) if len(pt[2]) > 1 else "0"
+            rank = f"pa_tm_rank_of(g_glob.{pt[1]}, (int)({a0}), (int)({a1}))"
+            # memoized per (cls, params, flow): all consumers see one datum
+            return (f"ptg_new_tile({cls_index[cls.name]}, _P, _flow, "
+                    f"(long)({_cxx_expr(size)}), {rank})")
         if kind == "coll":
             _, cname, args = term
             if cname not in colls:
@@ -372,8 +389,7 @@ def generate_cpp(jdf, name):
                     f"return binding_{tcls}(_Q, {fi}); }})()")
         if kind == "null":
             return "nullptr"
-        raise JdfError(f"unsupported dependency term {term[0]!r} "
-                       "(NEW tiles land with arena support)")
+        raise JdfError(f"unsupported dependency term {term[0]!r}")
 
     # binding resolvers
     for c in jdf.classes:
@@ -387,11 +403,11 @@ def generate_cpp(jdf, name):
             outs = [d for d in f.deps if d.direction == "->"]
             for d in ins:
                 if d.guard is not None:
-                    w(f"    if ({_cxx_expr(d.guard)}) return {term_tile(d.term)};")
+                    w(f"    if ({_cxx_expr(d.guard)}) return {term_tile(d.term, d.props, c)};")
                     if d.else_term is not None:
-                        w(f"    else return {term_tile(d.else_term)};")
+                        w(f"    else return {term_tile(d.else_term, d.props, c)};")
                 else:
-                    w(f"    return {term_tile(d.term)};")
+                    w(f"    return {term_tile(d.term, d.props, c)};")
             # WRITE-only flow: bind to the output collection ref
             for d in outs:
                 if d.term[0] == "coll":
@@ -462,6 +478,7 @@ def generate_cpp(jdf, name):
             if dflt is None:
                 raise JdfError(f"hidden global {s} has no default")
             w(f"  g_glob.{s} = (long)({_cxx_expr(dflt)}); {s} = g_glob.{s};")
+    w("  paptg::new_tiles_reset(_dtd);")
     w("  Graph _g(_ctx, _dtd);")
     w("  static std::vector<void*> _tcs; if (_tcs.empty()) {")
     for k, c in enumerate(jdf.classes):

@@ -84,4 +84,18 @@ PA_EXPORT void pa_task_edge(void* pred, void* succ) {
   task_add_edge((Task*)pred, (Task*)succ);
 }
 PA_EXPORT void pa_task_retain(void* t) { ((Task*)t)->retain(); }
+
+// Fresh scratch datum owned by the taskpool (JDF `<- NEW [size=...]`
+// arena-tile analog). home_rank = the creating instance's rank.
+PA_EXPORT void* pa_dtd_scratch(void* dtd, long bytes, int home_rank) {
+  auto* tp = (Dtd*)dtd;
+  auto holder = std::make_shared<Data>();
+  holder->ctx_direct = tp->context();
+  holder->home_rank = home_rank;
+  holder->owner_rank = home_rank;
+  holder->bytes = (size_t)bytes;
+  Data* d = holder.get();
+  tp->own(std::shared_ptr<void>(holder, holder.get()));
+  return d;
+}
 PA_EXPORT void pa_task_release(void* t) { ((Task*)t)->release(); }

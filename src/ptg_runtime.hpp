@@ -44,6 +44,7 @@ void pa_dtd_insert_commit(void* dtd, void* task);
 void pa_task_edge(void* pred, void* succ);
 void pa_task_retain(void* t);
 void pa_task_release(void* t);
+void* pa_dtd_scratch(void* dtd, long bytes, int home_rank);
 }
 
 namespace paptg {
@@ -160,5 +161,30 @@ class Graph {
   std::map<PKey, int> index_;
   std::vector<void*> classes_;
 };
+
+// ---- NEW-tile registry (JDF `<- NEW [size=...]`): one scratch datum per
+// (class, params, flow), owned by the taskpool, shared by every consumer
+// that resolves its binding to this instance's flow.
+inline std::map<std::tuple<int, std::array<long, MAXP>, int>, void*>&
+new_tile_map() {
+  static std::map<std::tuple<int, std::array<long, MAXP>, int>, void*> m;
+  return m;
+}
+inline void*& new_tile_dtd() {
+  static void* d = nullptr;
+  return d;
+}
+inline void new_tiles_reset(void* dtd) {
+  new_tile_map().clear();
+  new_tile_dtd() = dtd;
+}
+inline void* ptg_new_tile(int cls, const long* P, int flow, long bytes,
+                          int rank) {
+  std::array<long, MAXP> key{};
+  for (int i = 0; i < MAXP; i++) key[i] = P[i];
+  void*& slot = new_tile_map()[{cls, key, flow}];
+  if (!slot) slot = pa_dtd_scratch(new_tile_dtd(), bytes, rank);
+  return slot;
+}
 
 }  // namespace paptg

@@ -199,3 +199,16 @@ def test_inline_c_and_hidden_globals(ctx):
     for k in range(NT):
         (v,) = struct.unpack("<q", A.tile_bytes(k, 0))
         assert v == k * 2 + NT // 2, (k, v)
+
+
+def test_new_tiles(ctx):
+    import struct
+    mod = compile_jdf(os.path.join(EX, "new_tile.jdf"))
+    NT = 5
+    A = pm.TiledMatrix(ctx, NT, 1, 1, 1, 1, 1)
+    tp = pm.Dtd(ctx, "newt")
+    mod.build(ctx, tp, mydata=A, NT=NT)
+    tp.wait()
+    for k in range(NT):
+        (v,) = struct.unpack("<q", A.tile_bytes(k, 0))
+        assert v == sum(k * 100 + i for i in range(8)), (k, v)

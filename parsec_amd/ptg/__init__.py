@@ -270,6 +270,7 @@ def _parse_class(cls, lines, i):
             mode_s, fname, rest = mm.group(1), mm.group(2), mm.group(3)
             is_ctl = mode_s == "CTL"
             fl = Flow(ACCESS.get(mode_s, 1), fname, is_ctl)
+            # (CTL mode fixed up after arrows are parsed)
             # split rest into arrow chunks
             chunks = re.split(r"(<-|->)", rest)
             it = iter(chunks)
@@ -278,6 +279,10 @@ def _parse_class(cls, lines, i):
                 raise JdfError(f"{cls.name}.{fname}: junk before arrows: {lead!r}")
             for arrow, chunk in zip(it, it):
                 fl.deps.append(_parse_dep(arrow, chunk))
+            if fl.is_ctl:
+                has_in = any(d.direction == "<-" for d in fl.deps)
+                has_out = any(d.direction == "->" for d in fl.deps)
+                fl.mode = (3 if (has_in and has_out) else (1 if has_in else 2))
             cls.flows.append(fl)
             continue
         # range or local:  name = expr [.. expr [.. expr]]
@@ -338,7 +343,12 @@ def generate_cpp(jdf, name):
     cls_index = {c.name: k for k, c in enumerate(jdf.classes)}
 
     def data_flows(cls):
-        return [f for f in cls.flows if not f.is_ctl]
+        # CTL flows ride the dataflow as 8-byte token tiles: origin classes
+        # (only -> arrows) create a NEW scratch; consumers resolve it
+        # through the arrows; both-direction CTL is INOUT on the shared
+        # token -> the serialization (incl. cross-rank) comes from the same
+        # protocol as data.
+        return cls.flows
 
     def flow_index(cls, fname):
         for k, f in enumerate(data_flows(cls)):
@@ -413,6 +423,9 @@ def generate_cpp(jdf, name):
                 if d.term[0] == "coll":
                     w(f"    return {term_tile(d.term)};")
                     break
+            if f.is_ctl:
+                # CTL origin (or unguarded base case): own 8-byte token
+                w(f"    return {term_tile(('new',), {'size': '8'}, c)};")
             w("    break; }")
         w("  }")
         w(f'  fprintf(stderr, "[ptg] {c.name}: no binding for flow %d\\n", _flow); abort();')

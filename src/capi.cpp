@@ -5,6 +5,8 @@
 // minimal exported surface (the rest of _core.so is visibility-hidden).
 #include <hip/hip_runtime.h>
 
+#include <cstring>
+
 #include "data.hpp"
 #include "device_gpu.hpp"
 #include "dtd.hpp"
@@ -94,6 +96,11 @@ PA_EXPORT void* pa_dtd_scratch(void* dtd, long bytes, int home_rank) {
   holder->home_rank = home_rank;
   holder->owner_rank = home_rank;
   holder->bytes = (size_t)bytes;
+  // A fresh scratch tile is valid-but-uninitialized memory (the reference's
+  // arena-allocated NEW tiles behave the same): zero it so the first access
+  // — which may be a read (e.g. an INOUT CTL token) — is defined.
+  memset(holder->ensure_host(), 0, holder->bytes);
+  holder->host_valid = true;
   Data* d = holder.get();
   tp->own(std::shared_ptr<void>(holder, holder.get()));
   return d;

@@ -212,3 +212,68 @@ def test_new_tiles(ctx):
     for k in range(NT):
         (v,) = struct.unpack("<q", A.tile_bytes(k, 0))
         assert v == sum(k * 100 + i for i in range(8)), (k, v)
+
+
+def test_ctl_cross_rank(tmp_path):
+    """CTL flows serialize across ranks (tokens ride the dataflow)."""
+    import subprocess
+    import sys as _sys
+    jdf = """
+mydata  [ type="parsec_data_collection_t*" ]
+NT      [ type="int" ]
+
+Step(k)
+
+k = 0 .. NT-1
+
+: mydata( k )
+
+WRITE A -> mydata( k )
+CTL  X <- (k > 0) ? X Step( k-1 )
+       -> (k < NT-1) ? X Step( k+1 )
+
+BODY
+{
+    long* v = (long*)A;
+    v[0] = k;
+}
+END
+"""
+    jp = tmp_path / "xrank_ctl.jdf"
+    jp.write_text(jdf)
+    compile_jdf(str(jp))  # warm cache
+    code = f"""
+import os, sys, struct
+sys.path.insert(0, {REPO!r})
+import parsec_amd as pm
+from parsec_amd.ptg import compile_jdf
+rank = int(os.environ["RANK"])
+pm.param_set("comm_base_port", os.environ["PORT"])
+ctx = pm.Context(nworkers=2, rank=rank, world=2, comm="tcp", gpu=-2)
+NT = 6
+A = pm.TiledMatrix(ctx, NT, 1, 1, 1, 2, 1)  # tiles alternate ranks
+mod = compile_jdf({str(jp)!r})
+tp = pm.Dtd(ctx, "xctl")
+mod.build(ctx, tp, mydata=A, NT=NT)
+tp.wait()
+ctx.barrier()
+for k in range(NT):
+    if A.is_local(k, 0):
+        (v,) = struct.unpack("<q", A.tile_bytes(k, 0))
+        assert v == k, (k, v)
+print("XRANK_CTL_OK", rank)
+ctx.barrier()
+del A, ctx
+"""
+    import conftest
+    port = str(conftest.port_base(9))
+    procs = []
+    for r in range(2):
+        env = dict(os.environ)
+        env.update(RANK=str(r), WORLD_SIZE="2", PORT=port)
+        procs.append(subprocess.Popen([_sys.executable, "-c", code], env=env,
+                                      stdout=subprocess.PIPE,
+                                      stderr=subprocess.STDOUT))
+    for pr in procs:
+        o, _ = pr.communicate(timeout=180)
+        assert pr.returncode == 0 and b"XRANK_CTL_OK" in o, o.decode()

@@ -5,6 +5,16 @@ import parsec_amd as pm
 
 
 def main():
+    if len(sys.argv) > 1 and sys.argv[1] == "bandwidth":
+        # tools/gpu/testbandwidth analog: HIP link bandwidth probe
+        if pm.hip_device_count() == 0:
+            print("no HIP device visible"); return
+        r = pm.hip_bandwidth()
+        mb = r["bytes"] >> 20
+        print(f"HIP copy bandwidth ({mb} MiB, pinned host):")
+        print(f"  H2D {r['h2d_gbs']:.1f} GB/s   D2H {r['d2h_gbs']:.1f} GB/s"
+              f"   D2D {r['d2d_gbs']:.1f} GB/s")
+        return
     print(f"parsec_amd {pm.__version__} — MI355X-native task-dataflow "
           "runtime (PaRSEC-class)")
     print(f"HIP devices visible: {pm.hip_device_count()}")

@@ -8,6 +8,8 @@
 #include <pybind11/pybind11.h>
 #include <pybind11/stl.h>
 
+#include <chrono>
+
 #include "data.hpp"
 #include "device_gpu.hpp"
 #include "dtd.hpp"
@@ -367,6 +369,39 @@ PYBIND11_MODULE(_core, m) {
                                                         impl == "v1" ? 1 : 0);
     return 2.0 * m * n * k * iters / dt / 1e12;  // TFLOP/s
   }, py::call_guard<py::gil_scoped_release>());
+  // PCIe/xGMI link probe (reference tools/gpu/testbandwidth analog):
+  // pinned-host H2D/D2H and on-device D2D copy bandwidth in GB/s.
+  m.def("hip_bandwidth", [](size_t nbytes, int iters) {
+    void *hbuf = nullptr, *d0 = nullptr, *d1 = nullptr;
+    if (hipHostMalloc(&hbuf, nbytes) != hipSuccess)
+      throw std::runtime_error("hipHostMalloc failed");
+    if (hipMalloc(&d0, nbytes) != hipSuccess ||
+        hipMalloc(&d1, nbytes) != hipSuccess)
+      throw std::runtime_error("hipMalloc failed");
+    hipStream_t s;
+    (void)hipStreamCreate(&s);
+    auto run = [&](void* dst, const void* src, hipMemcpyKind kind) {
+      (void)hipMemcpyAsync(dst, src, nbytes, kind, s);  // warmup
+      (void)hipStreamSynchronize(s);
+      auto t0 = std::chrono::steady_clock::now();
+      for (int i = 0; i < iters; i++)
+        (void)hipMemcpyAsync(dst, src, nbytes, kind, s);
+      (void)hipStreamSynchronize(s);
+      double dt = std::chrono::duration<double>(
+                      std::chrono::steady_clock::now() - t0).count();
+      return (double)nbytes * iters / dt / 1e9;
+    };
+    py::dict r;
+    r["h2d_gbs"] = run(d0, hbuf, hipMemcpyHostToDevice);
+    r["d2h_gbs"] = run(hbuf, d0, hipMemcpyDeviceToHost);
+    r["d2d_gbs"] = run(d1, d0, hipMemcpyDeviceToDevice);
+    r["bytes"] = nbytes;
+    (void)hipStreamDestroy(s);
+    (void)hipFree(d0);
+    (void)hipFree(d1);
+    (void)hipHostFree(hbuf);
+    return r;
+  }, py::arg("nbytes") = (size_t)256 << 20, py::arg("iters") = 20);
   m.def("nccl_unique_id", [] { return py::bytes(pa::rccl_get_unique_id()); });
   m.def("set_nccl_unique_id",
         [](py::bytes b) { pa::rccl_set_unique_id(std::string(b)); });

@@ -192,6 +192,7 @@ bool GpuEngine::evict_one(size_t) {
                     resident_.end());
     PA_DEBUG(2, "evicted tile %lu (%zu bytes)", (unsigned long)d->key,
              d->bytes);
+    stats.evictions++;
     return true;
   }
   return false;
@@ -284,7 +285,9 @@ void GpuEngine::stage_in_and_launch(Task* t) {
   }
   std::vector<std::pair<void*, size_t>> deferred;
   GpuTaskCtx gctx{es, device_, this, &deferred};
+  if (roctx_on()) roctx_push(t->tc->name.c_str());
   t->tc->gpu_hook(*t, gctx);
+  if (roctx_on()) roctx_pop();
   hipEvent_t ev = event_get();
   PA_HIP_CHECK(hipEventRecord(ev, es));
   inflight_[si].push_back(

@@ -79,6 +79,7 @@ class GpuEngine {
     std::atomic<uint64_t> tasks{0};
     std::atomic<uint64_t> bytes_h2d{0};
     std::atomic<uint64_t> bytes_d2h{0};
+    std::atomic<uint64_t> evictions{0};
   } stats;
 
  private:

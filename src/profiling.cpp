@@ -1,5 +1,7 @@
 #include "profiling.hpp"
 
+#include <dlfcn.h>
+
 #include <chrono>
 
 #include "runtime.hpp"
@@ -107,4 +109,29 @@ void Profiler::dot_edge(const Task* pred, const Task* succ) {
           (unsigned long)succ->seq);
 }
 
+}  // namespace pa
+
+// ---------------------------------------------------------------- rocTX
+namespace pa {
+namespace {
+int (*p_roctx_push)(const char*) = nullptr;
+int (*p_roctx_pop)() = nullptr;
+bool roctx_enabled = false;
+}  // namespace
+
+void roctx_init() {
+  if (!param_int("profile_roctx", 0)) return;
+  void* h = dlopen("libroctx64.so", RTLD_NOW | RTLD_LOCAL);
+  if (!h) h = dlopen("libroctx64.so.4", RTLD_NOW | RTLD_LOCAL);
+  if (!h) {
+    fprintf(stderr, "[parsec_amd] profile_roctx=1 but libroctx64 not found\n");
+    return;
+  }
+  p_roctx_push = (int (*)(const char*))dlsym(h, "roctxRangePushA");
+  p_roctx_pop = (int (*)())dlsym(h, "roctxRangePop");
+  roctx_enabled = p_roctx_push && p_roctx_pop;
+}
+bool roctx_on() { return roctx_enabled; }
+void roctx_push(const char* name) { if (roctx_enabled) p_roctx_push(name); }
+void roctx_pop() { if (roctx_enabled) p_roctx_pop(); }
 }  // namespace pa

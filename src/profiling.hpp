@@ -77,6 +77,16 @@ class Profiler {
   std::mutex dot_mtx_;
 };
 
+// rocTX ranges (the reference's NVTX sink, profiling_nvtx.c, done the
+// ROCm-native way): with PARSEC_MCA_profile_roctx=1, task bodies and GPU
+// submissions get roctxRangePush/Pop so rocprofv3's marker trace shows
+// task-class names interleaved with kernels. libroctx64 is dlopen'd
+// lazily: zero cost when off.
+void roctx_init();
+bool roctx_on();
+void roctx_push(const char* name);
+void roctx_pop();
+
 // global counters (papi_sde-style software counters)
 struct RuntimeCounters {
   std::atomic<uint64_t> tasks_executed_cpu{0};

@@ -149,6 +149,7 @@ PYBIND11_MODULE(_core, m) {
           d["tasks"] = c.gpu()->stats.tasks.load();
           d["bytes_h2d"] = c.gpu()->stats.bytes_h2d.load();
           d["bytes_d2h"] = c.gpu()->stats.bytes_d2h.load();
+          d["evictions"] = c.gpu()->stats.evictions.load();
         }
         return d;
       })

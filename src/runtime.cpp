@@ -74,6 +74,7 @@ void run_cpu_task(Task* t) {
     if (t->flows[i].data && t->flows[i].mode == ACCESS_OUT)
       t->flows[i].data->begin_host_overwrite();
   if (t->tc->cpu_hook) {
+    if (roctx_on()) roctx_push(t->tc->name.c_str());
     if (pr.enabled()) {
       uint64_t t0 = Profiler::now_ns();
       t->tc->cpu_hook(*t);
@@ -81,6 +82,7 @@ void run_cpu_task(Task* t) {
     } else {
       t->tc->cpu_hook(*t);
     }
+    if (roctx_on()) roctx_pop();
   }
   counters().tasks_executed_cpu.fetch_add(1, std::memory_order_relaxed);
   task_complete(t);
@@ -236,6 +238,7 @@ Context::Context(const Options& opt) : rank_(opt.rank), world_(opt.world) {
   if (opt.gpu_device != -2) gpu_ = GpuEngine::create(this, opt.gpu_device);
   comm_ = CommEngine::create(this, opt.comm);
 
+  roctx_init();
   std::string prof = param_str("profile_filename", "");
   if (!prof.empty()) Profiler::inst().start(prof + "." + std::to_string(rank_));
   std::string dot = param_str("profile_dot", "");

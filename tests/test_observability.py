@@ -61,3 +61,23 @@ del A, ctx
     r2 = subprocess.run([sys.executable, "-m", "parsec_amd"],
                         capture_output=True, text=True, timeout=120, cwd=REPO)
     assert "MI355X-native" in r2.stdout
+
+
+def test_roctx_sink_cpu_noop(tmp_path):
+    """profile_roctx=1 wraps bodies in rocTX ranges; harmless w/o rocprof."""
+    code = f"""
+import sys; sys.path.insert(0, {REPO!r})
+import parsec_amd as pm
+pm.param_set("profile_roctx", "1")
+ctx = pm.Context(nworkers=2, rank=0, world=1, gpu=-2)
+A = pm.TiledMatrix(ctx, 256, 256, 64, 64, 1, 1)
+tp = pm.Dtd(ctx)
+pm.insert_spd_fill(tp, A, 1)
+pm.insert_potrf(tp, A)
+tp.wait()
+print("ROCTX_OK")
+del A, ctx
+"""
+    r = subprocess.run([sys.executable, "-c", code], capture_output=True,
+                       text=True, timeout=120)
+    assert "ROCTX_OK" in r.stdout, r.stdout + r.stderr

@@ -38,6 +38,8 @@ from parsec_amd._core import (  # noqa: F401,E402
     insert_potrf_panel,
     param_dump,
     param_set,
+    pins_add,
+    pins_remove,
 )
 
 __version__ = "0.1.0"

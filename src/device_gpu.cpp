@@ -3,6 +3,7 @@
 #include <algorithm>
 
 #include "data.hpp"
+#include "pins.hpp"
 #include "profiling.hpp"
 
 namespace pa {
@@ -288,6 +289,7 @@ void GpuEngine::stage_in_and_launch(Task* t) {
   if (roctx_on()) roctx_push(t->tc->name.c_str());
   t->tc->gpu_hook(*t, gctx);
   if (roctx_on()) roctx_pop();
+  PA_PINS(PinsEv::GPU_SUBMIT, t, -1);
   hipEvent_t ev = event_get();
   PA_HIP_CHECK(hipEventRecord(ev, es));
   inflight_[si].push_back(
@@ -315,6 +317,7 @@ bool GpuEngine::retire_pass() {
         pr.record(Ev::GPU_TASK, (uint16_t)t->tc->id, t->seq, f.t0_ns,
                   Profiler::now_ns());
       counters().tasks_executed_gpu.fetch_add(1, std::memory_order_relaxed);
+      PA_PINS(PinsEv::GPU_RETIRE, t, -1);
       for (auto& [p2, b2] : f.deferred_frees) dev_free(p2, b2);
       event_put(f.event);
       ring.pop_front();

@@ -9,11 +9,13 @@
 #include <pybind11/stl.h>
 
 #include <chrono>
+#include <map>
 
 #include "data.hpp"
 #include "device_gpu.hpp"
 #include "dtd.hpp"
 #include "kernels.hpp"
+#include "pins.hpp"
 #include "profiling.hpp"
 #include "runtime.hpp"
 
@@ -346,6 +348,38 @@ PYBIND11_MODULE(_core, m) {
                             py::array_t<float> C, int m, int n, int k) {
     pa::test_gemm_bf16_hip(m, n, k, A.data(), B.data(), C.mutable_data());
   });
+
+  // PINS callback chain (mca/pins analog). Python callbacks fire from
+  // worker threads: the wrapper takes the GIL. Returns a handle for
+  // pins_remove. events: list of names from PINS_EVENTS.
+  m.def("pins_add", [](py::function fn, std::vector<std::string> events) {
+    static const std::map<std::string, PinsEv> names = {
+        {"exec_begin", PinsEv::EXEC_BEGIN}, {"exec_end", PinsEv::EXEC_END},
+        {"schedule", PinsEv::SCHEDULE},     {"complete", PinsEv::COMPLETE},
+        {"gpu_submit", PinsEv::GPU_SUBMIT}, {"gpu_retire", PinsEv::GPU_RETIRE}};
+    uint32_t mask = 0;
+    for (auto& e : events) {
+      auto it = names.find(e);
+      if (it == names.end())
+        throw std::invalid_argument("unknown PINS event: " + e);
+      mask |= 1u << (int)it->second;
+    }
+    // keep the function alive via shared_ptr with GIL-acquiring deleter
+    auto keep = std::shared_ptr<py::function>(
+        new py::function(std::move(fn)), [](py::function* f) {
+          py::gil_scoped_acquire g;
+          delete f;
+        });
+    static const char* evname[] = {"exec_begin", "exec_end", "schedule",
+                                   "complete",   "gpu_submit", "gpu_retire"};
+    return Pins::inst().add(
+        [keep](PinsEv e, const Task* t, int worker) {
+          py::gil_scoped_acquire g;
+          (*keep)(evname[(int)e], t ? t->tc->name : std::string(), worker);
+        },
+        mask);
+  });
+  m.def("pins_remove", [](int id) { Pins::inst().remove(id); });
 
   m.def("param_set", &param_set);
   m.def("param_dump", &param_dump);

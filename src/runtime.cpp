@@ -3,6 +3,7 @@
 #include "data.hpp"
 #include "device_gpu.hpp"
 #include "comm.hpp"
+#include "pins.hpp"
 #include "profiling.hpp"
 
 #include <pthread.h>
@@ -61,6 +62,7 @@ void task_complete(Task* t) {
   succs.swap(t->succs);
   t->lock.unlock();
   for (Task* s : succs) task_dec_deps(s);
+  PA_PINS(PinsEv::COMPLETE, t, Context::tls_worker_id);
   Taskpool* tp = t->tp;
   t->release();
   tp->task_done();
@@ -74,6 +76,7 @@ void run_cpu_task(Task* t) {
     if (t->flows[i].data && t->flows[i].mode == ACCESS_OUT)
       t->flows[i].data->begin_host_overwrite();
   if (t->tc->cpu_hook) {
+    PA_PINS(PinsEv::EXEC_BEGIN, t, Context::tls_worker_id);
     if (roctx_on()) roctx_push(t->tc->name.c_str());
     if (pr.enabled()) {
       uint64_t t0 = Profiler::now_ns();
@@ -83,6 +86,7 @@ void run_cpu_task(Task* t) {
       t->tc->cpu_hook(*t);
     }
     if (roctx_on()) roctx_pop();
+    PA_PINS(PinsEv::EXEC_END, t, Context::tls_worker_id);
   }
   counters().tasks_executed_cpu.fetch_add(1, std::memory_order_relaxed);
   task_complete(t);
@@ -300,6 +304,7 @@ void Context::worker_main(int id) {
 
 void Context::dispatch(Task* t, int worker_hint) {
   counters().tasks_scheduled.fetch_add(1, std::memory_order_relaxed);
+  PA_PINS(PinsEv::SCHEDULE, t, Context::tls_worker_id);
   switch (t->tc->kind) {
     case TaskKind::GPU:
       if (gpu_) { gpu_->enqueue(t); return; }

@@ -81,3 +81,37 @@ del A, ctx
     r = subprocess.run([sys.executable, "-c", code], capture_output=True,
                        text=True, timeout=120)
     assert "ROCTX_OK" in r.stdout, r.stdout + r.stderr
+
+
+def test_pins_callbacks():
+    """PINS chain (mca/pins analog): lifecycle callbacks with event mask."""
+    code = f"""
+import sys, threading; sys.path.insert(0, {REPO!r})
+import parsec_amd as pm
+seen = {{}}
+lock = threading.Lock()
+def cb(ev, cls, worker):
+    with lock:
+        seen[ev] = seen.get(ev, 0) + 1
+h = pm.pins_add(cb, ["exec_begin", "exec_end", "schedule", "complete"])
+ctx = pm.Context(nworkers=2, rank=0, world=1, gpu=-2)
+A = pm.TiledMatrix(ctx, 256, 256, 64, 64, 1, 1)
+tp = pm.Dtd(ctx)
+pm.insert_spd_fill(tp, A, 1)
+pm.insert_potrf(tp, A)
+tp.wait()
+assert seen["exec_begin"] == seen["exec_end"] >= 10, seen
+assert seen["complete"] >= seen["exec_begin"], seen
+assert seen["schedule"] >= 10, seen
+pm.pins_remove(h)
+n0 = dict(seen)
+tp2 = pm.Dtd(ctx)
+pm.insert_potrf(tp2, A)
+tp2.wait()
+assert seen == n0, "callbacks fired after removal"
+print("PINS_OK")
+del A, ctx
+"""
+    r = subprocess.run([sys.executable, "-c", code], capture_output=True,
+                       text=True, timeout=120)
+    assert "PINS_OK" in r.stdout, r.stdout + r.stderr

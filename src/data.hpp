@@ -95,9 +95,33 @@ class TiledMatrix {
   size_t tile_bytes() const { return (size_t)mb_ * nb_ * elem_; }
   bool sym() const { return sym_; }
 
+  // ---- distribution variants (all SPMD: identical on every rank) ----
+  // k-cyclic: kp consecutive tile-rows (kq tile-cols) share a grid row
+  // (col) before cycling (two_dim_rectangle_cyclic k-cyclicity analog).
+  void set_kcyclic(int kp, int kq) {
+    PA_CHECK(kp >= 1 && kq >= 1 && !any_tiles(),
+             "set_kcyclic: before first tile access");
+    kp_ = kp; kq_ = kq;
+  }
+  // tabular: arbitrary per-tile rank table, size mt*nt, row-major
+  // (two_dim_tabular analog). Overrides the cyclic mapping.
+  void set_rank_table(std::vector<int> table);
+  // band storage: only tiles with tm-kl <= tn <= tm+ku exist; accessing an
+  // out-of-band tile is an error (band-cyclic distribution analog).
+  void set_band(int kl, int ku) {
+    PA_CHECK(kl >= 0 && ku >= 0 && !any_tiles(),
+             "set_band: before first tile access");
+    band_ = true; kl_ = kl; ku_ = ku;
+  }
+  bool in_band(int tm, int tn) const {
+    if (sym_ && tn > tm) { int t = tm; tm = tn; tn = t; }
+    return !band_ || (tn >= tm - kl_ && tn <= tm + ku_);
+  }
+
   int rank_of(int tm, int tn) const {
     if (sym_ && tn > tm) { int t = tm; tm = tn; tn = t; }
-    return (tm % p_) * q_ + (tn % q_);
+    if (!ranks_.empty()) return ranks_[(size_t)tm * nt_ + tn];
+    return ((tm / kp_) % p_) * q_ + ((tn / kq_) % q_);
   }
   bool is_local(int tm, int tn) const {
     return rank_of(tm, tn) == ctx_rank_;
@@ -121,7 +145,17 @@ class TiledMatrix {
   int mb_, nb_, mt_, nt_, p_, q_;
   size_t elem_;
   bool sym_ = false;
-  std::vector<std::unique_ptr<Data>> tiles_;  // mt*nt, metadata eager
+  int kp_ = 1, kq_ = 1;
+  bool band_ = false;
+  int kl_ = 0, ku_ = 0;
+  std::vector<int> ranks_;  // tabular override (empty = cyclic)
+  std::vector<std::unique_ptr<Data>> tiles_;  // mt*nt, metadata lazy
+
+  bool any_tiles() const {
+    for (auto& t : tiles_)
+      if (t) return true;
+    return false;
+  }
 };
 
 // Irregular key->data collection (hash_datadist analog,

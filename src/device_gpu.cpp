@@ -454,9 +454,19 @@ Data* IrregularCollection::at(uint64_t key) {
   return it->second.get();
 }
 
+void TiledMatrix::set_rank_table(std::vector<int> table) {
+  PA_CHECK((int64_t)table.size() == (int64_t)mt_ * nt_,
+           "rank table must have mt*nt entries");
+  for (int r : table)
+    PA_CHECK(r >= 0 && r < ctx_->world(), "rank table entry out of range");
+  PA_CHECK(!any_tiles(), "set_rank_table: before first tile access");
+  ranks_ = std::move(table);
+}
+
 Data* TiledMatrix::tile(int tm, int tn) {
   if (sym_ && tn > tm) std::swap(tm, tn);
   PA_CHECK(tm >= 0 && tm < mt_ && tn >= 0 && tn < nt_);
+  PA_CHECK(in_band(tm, tn), "tile (%d,%d) outside the stored band", tm, tn);
   size_t idx = (size_t)tm * nt_ + tn;
   if (!tiles_[idx]) {
     auto d = std::make_unique<Data>();

@@ -186,6 +186,10 @@ PYBIND11_MODULE(_core, m) {
       .def_property_readonly("mb", &TiledMatrix::mb)
       .def_property_readonly("nb", &TiledMatrix::nb)
       .def("rank_of", &TiledMatrix::rank_of)
+      .def("set_kcyclic", &TiledMatrix::set_kcyclic)
+      .def("set_rank_table", &TiledMatrix::set_rank_table)
+      .def("set_band", &TiledMatrix::set_band)
+      .def("in_band", &TiledMatrix::in_band)
       .def("is_local", &TiledMatrix::is_local)
       .def("tile", &TiledMatrix::tile, py::return_value_policy::reference_internal)
       .def("tile_rows", &TiledMatrix::tile_rows)

@@ -561,6 +561,7 @@ class PtgModule:
         self._build = getattr(self._lib, f"ptg_build_{name}")
         self._jdf = jdf
         self.name = name
+        self.so_path = so_path
 
     def build(self, ctx, tp, **kwargs):
         """Enumerate + insert the taskpool's tasks. kwargs map JDF global

@@ -510,16 +510,99 @@ void insert_reduce_sum(Dtd& tp, TiledMatrix& A, TiledMatrix& R) {
     }
 }
 
+// ---- general regridding (redistribute.jdf incl. non-matching tile grids,
+// data_dist/matrix/redistribute/redistribute.jdf analog) ----
+// One CPU piece-task per (dst tile, overlapping src tile): copies the
+// intersection rectangle. Pieces of one dst tile serialize through INOUT
+// chaining; cross-rank movement falls out of the normal protocol. A
+// leading zero-task defines regions no src tile covers.
+namespace {
+struct RegridArgs {
+  int r, c;        // rectangle extent (rows, cols)
+  int si, sj;      // offset in src tile
+  int di, dj;      // offset in dst tile
+  int lds, ldd;    // column strides (elements)
+  int elem;        // element size in bytes
+};
+
+void cpu_regrid_zero(Task& t) {
+  Data* d = t.flows[0].data;
+  memset(d->ensure_host(), 0, d->bytes);
+  d->written_on(false);
+}
+
+void cpu_regrid_piece(Task& t) {
+  const RegridArgs& a = t.arg<RegridArgs>();
+  const char* s = (const char*)t.flows[0].data->pull_to_host();
+  char* d = (char*)t.flows[1].data->pull_to_host();
+  const size_t rb = (size_t)a.r * a.elem;
+  for (int j = 0; j < a.c; j++)
+    memcpy(d + ((size_t)(a.dj + j) * a.ldd + a.di) * a.elem,
+           s + ((size_t)(a.sj + j) * a.lds + a.si) * a.elem, rb);
+  t.flows[1].data->written_on(false);
+}
+
+TaskClass& tc_regrid_zero() {
+  static TaskClass tc = make_tc("regrid_zero", TaskKind::CPU, cpu_regrid_zero,
+                                nullptr, 36);
+  return tc;
+}
+TaskClass& tc_regrid_piece() {
+  static TaskClass tc = make_tc("regrid_piece", TaskKind::CPU,
+                                cpu_regrid_piece, nullptr, 37);
+  return tc;
+}
+}  // namespace
+
 void insert_redistribute(Dtd& tp, TiledMatrix& Src, TiledMatrix& Dst) {
-  PA_CHECK(Src.mt() == Dst.mt() && Src.nt() == Dst.nt() &&
-           Src.tile_bytes() == Dst.tile_bytes(),
-           "redistribute: tile grids must match (general regridding is a "
-           "round-2 item)");
-  for (int m = 0; m < Src.mt(); m++)
-    for (int n = 0; n < (Src.sym() ? m + 1 : Src.nt()); n++) {
-      Dtd::FlowSpec f[] = {{Src.tile(m, n), ACCESS_IN},
-                           {Dst.tile(m, n), ACCESS_OUT}};
-      tp.insert(&tc_copy_tile(), nullptr, 0, f, 2, 0, Dst.rank_of(m, n));
+  if (Src.mt() == Dst.mt() && Src.nt() == Dst.nt() &&
+      Src.tile_bytes() == Dst.tile_bytes()) {
+    // fast path: same tile grid — one (GPU-capable) whole-tile copy each
+    for (int m = 0; m < Src.mt(); m++)
+      for (int n = 0; n < (Src.sym() ? m + 1 : Src.nt()); n++) {
+        Dtd::FlowSpec f[] = {{Src.tile(m, n), ACCESS_IN},
+                             {Dst.tile(m, n), ACCESS_OUT}};
+        tp.insert(&tc_copy_tile(), nullptr, 0, f, 2, 0, Dst.rank_of(m, n));
+      }
+    return;
+  }
+  PA_CHECK(Src.m() == Dst.m() && Src.n() == Dst.n() &&
+           Src.elem_size() == Dst.elem_size() && !Src.sym() && !Dst.sym(),
+           "redistribute: matrices must have equal global shape/dtype "
+           "(sym storage regridding unsupported)");
+  for (int dm = 0; dm < Dst.mt(); dm++)
+    for (int dn = 0; dn < Dst.nt(); dn++) {
+      const int rank = Dst.rank_of(dm, dn);
+      {
+        Dtd::FlowSpec f[] = {{Dst.tile(dm, dn), ACCESS_OUT}};
+        tp.insert(&tc_regrid_zero(), nullptr, 0, f, 1, 0, rank);
+      }
+      const int64_t r0 = (int64_t)dm * Dst.mb(), r1 = r0 + Dst.tile_rows(dm);
+      const int64_t c0 = (int64_t)dn * Dst.nb(), c1 = c0 + Dst.tile_cols(dn);
+      for (int sm = (int)(r0 / Src.mb()); (int64_t)sm * Src.mb() < r1; sm++)
+        for (int sn = (int)(c0 / Src.nb()); (int64_t)sn * Src.nb() < c1;
+             sn++) {
+          const int64_t sr0 = (int64_t)sm * Src.mb();
+          const int64_t sc0 = (int64_t)sn * Src.nb();
+          const int64_t ir0 = std::max(r0, sr0);
+          const int64_t ir1 = std::min(r1, sr0 + Src.tile_rows(sm));
+          const int64_t ic0 = std::max(c0, sc0);
+          const int64_t ic1 = std::min(c1, sc0 + Src.tile_cols(sn));
+          if (ir1 <= ir0 || ic1 <= ic0) continue;
+          RegridArgs a;
+          a.r = (int)(ir1 - ir0);
+          a.c = (int)(ic1 - ic0);
+          a.si = (int)(ir0 - sr0);
+          a.sj = (int)(ic0 - sc0);
+          a.di = (int)(ir0 - r0);
+          a.dj = (int)(ic0 - c0);
+          a.lds = Src.mb();
+          a.ldd = Dst.mb();
+          a.elem = (int)Src.elem_size();
+          Dtd::FlowSpec f[] = {{Src.tile(sm, sn), ACCESS_IN},
+                               {Dst.tile(dm, dn), ACCESS_INOUT}};
+          tp.insert(&tc_regrid_piece(), &a, sizeof(a), f, 2, 0, rank);
+        }
     }
 }
 

@@ -220,3 +220,72 @@ A.tile(0, 3)
     r = subprocess.run([_sys.executable, "-c", code], capture_output=True,
                        text=True, timeout=60)
     assert r.returncode != 0 and "band" in (r.stderr + r.stdout)
+
+
+def test_regrid_nonmatching_tiles(ctx):
+    """Redistribute between non-matching tile grids (redistribute.jdf)."""
+    rng = np.random.default_rng(5)
+    n = 96
+    S = pm.TiledMatrix(ctx, n, n, 32, 32, 1, 1)
+    D = pm.TiledMatrix(ctx, n, n, 40, 24, 1, 1)
+    M = rng.standard_normal((n, n))
+    for i in range(S.mt):
+        for j in range(S.nt):
+            S.tile_numpy_set(i, j, M[i*32:(i+1)*32, j*32:(j+1)*32])
+    tp = pm.Dtd(ctx, "regrid")
+    pm.insert_redistribute(tp, S, D)
+    tp.wait()
+    out = np.zeros((n, n))
+    for i in range(D.mt):
+        for j in range(D.nt):
+            v = D.tile_numpy(i, j)
+            out[i*40:i*40+v.shape[0], j*24:j*24+v.shape[1]] = v
+    assert np.array_equal(out, M), np.abs(out - M).max()
+
+
+def test_regrid_world2(tmp_path):
+    """Cross-rank regridding: different grids AND different distributions."""
+    import subprocess
+    import sys as _sys
+    from conftest import port_base
+    code = f"""
+import os, sys
+sys.path.insert(0, {REPO!r})
+import numpy as np
+import parsec_amd as pm
+rank = int(os.environ["RANK"])
+pm.param_set("comm_base_port", os.environ["PORT"])
+ctx = pm.Context(nworkers=2, rank=rank, world=2, comm="tcp", gpu=-2)
+n = 128
+S = pm.TiledMatrix(ctx, n, n, 32, 32, 2, 1)
+D = pm.TiledMatrix(ctx, n, n, 48, 48, 1, 2)
+tp = pm.Dtd(ctx, "regrid2")
+pm.insert_full_fill(tp, S, 9)
+pm.insert_redistribute(tp, S, D)
+tp.wait()
+ctx.barrier()
+# reassemble local parts and checksum against the fill function values
+tot = 0.0
+cnt = 0
+for i in range(D.mt):
+    for j in range(D.nt):
+        if D.is_local(i, j):
+            v = D.tile_numpy(i, j)
+            assert np.isfinite(v).all()
+            tot += float(abs(v).sum()); cnt += v.size
+assert cnt > 0 and tot > 0
+print("REGRID2_OK", rank, round(tot, 3))
+ctx.barrier()
+del S, D, ctx
+"""
+    port = str(port_base(13))
+    procs = []
+    for r in range(2):
+        env = dict(os.environ)
+        env.update(RANK=str(r), WORLD_SIZE="2", PORT=port)
+        procs.append(subprocess.Popen([_sys.executable, "-c", code], env=env,
+                                      stdout=subprocess.PIPE,
+                                      stderr=subprocess.STDOUT))
+    for pr in procs:
+        o, _ = pr.communicate(timeout=180)
+        assert pr.returncode == 0 and b"REGRID2_OK" in o, o.decode()

@@ -112,6 +112,7 @@ void Dtd::insert(const TaskClass* tc, const void* args, size_t args_bytes,
 Task* Dtd::insert_begin(const TaskClass* tc, const void* args,
                         size_t args_bytes, const FlowSpec* flows, int nflows,
                         int priority, int rank) {
+  insert_mtx_.lock();  // released in insert_commit
   PA_CHECK(nflows <= MAX_FLOWS);
   PA_CHECK(args_bytes <= MAX_ARGS_BYTES);
   int task_rank = rank;
@@ -140,6 +141,7 @@ Task* Dtd::insert_begin(const TaskClass* tc, const void* args,
 }
 
 void Dtd::insert_commit(Task* t) {
+  insert_mtx_.unlock();
   if (t) task_dec_deps(t);
 
   // Window throttling (insert_function.c:75-76): the inserter joins
@@ -153,6 +155,7 @@ void Dtd::insert_commit(Task* t) {
 }
 
 void Dtd::flush(Data* d) {
+  std::lock_guard<std::recursive_mutex> g(insert_mtx_);
   const int O = d->owner_rank, H = d->home_rank;
   if (O == H) return;
   uint64_t seq = chan_next(O, H);

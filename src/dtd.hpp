@@ -16,6 +16,8 @@
 // deterministic per-channel order need no wire handshake at all.
 #pragma once
 
+#include <mutex>
+
 #include "data.hpp"
 #include "runtime.hpp"
 
@@ -65,6 +67,12 @@ class Dtd : public Taskpool {
   }
 
   int me_, world_;
+  // Serializes inserters: the chaining state machines are sequential by
+  // construction. A recursive mutex so a task body running inline during
+  // window throttling may itself insert (untied tasks, dtd_test_untie
+  // analog). Deterministic ORDER across ranks remains the app's contract
+  // exactly as in the reference's distributed DTD.
+  std::recursive_mutex insert_mtx_;
   std::vector<uint64_t> chan_seq_;
   int64_t window_;
   int64_t threshold_;

@@ -230,3 +230,27 @@ def test_py_task_with_data(ctx):
                  with_data=True)
     tp.wait()
     assert np.allclose(B.tile_numpy(0, 0), 7.0)
+
+
+def test_untied_tasks(ctx):
+    """Tasks inserting tasks into the same taskpool from worker threads
+    (dtd_test_untie analog) — insertion is internally serialized."""
+    import threading
+    A = pm.TiledMatrix(ctx, 8, 8, 1, 1, 1, 1)
+    tp = pm.Dtd(ctx, "untie")
+    hits = []
+    lock = threading.Lock()
+
+    def leaf(i):
+        with lock:
+            hits.append(i)
+
+    def root():
+        # runs on a worker thread; inserts 16 more tasks
+        for i in range(16):
+            tp.insert_py((lambda i=i: leaf(i)),
+                         flows=[(A.tile(i % 8, i // 8), pm.ACCESS_INOUT)])
+
+    tp.insert_py(root, flows=[(A.tile(0, 0), pm.ACCESS_INOUT)])
+    tp.wait()
+    assert sorted(hits) == list(range(16)), hits

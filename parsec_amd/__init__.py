@@ -40,6 +40,7 @@ from parsec_amd._core import (  # noqa: F401,E402
     param_set,
     pins_add,
     pins_remove,
+    set_fatal_handler,
 )
 
 __version__ = "0.1.0"

@@ -419,10 +419,20 @@ def generate_cpp(jdf, name):
                 else:
                     w(f"    return {term_tile(d.term, d.props, c)};")
             # WRITE-only flow: bind to the output collection ref
+            bound = False
             for d in outs:
                 if d.term[0] == "coll":
                     w(f"    return {term_tile(d.term)};")
+                    bound = True
                     break
+            if not bound and not f.is_ctl:
+                # no collection backing: an arena-typed flow — requires an
+                # explicit [size=bytes] on one of the -> deps (the NEW
+                # scratch registry provides the buffer, one per instance)
+                sz = next((d.props.get("size") for d in outs
+                           if d.props and d.props.get("size")), None)
+                if sz is not None:
+                    w(f"    return {term_tile(('new',), {'size': sz}, c)};")
             if f.is_ctl:
                 # CTL origin (or unguarded base case): own 8-byte token
                 w(f"    return {term_tile(('new',), {'size': '8'}, c)};")

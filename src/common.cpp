@@ -1,5 +1,7 @@
 #include "common.hpp"
 
+#include <atomic>
+
 #include <map>
 #include <mutex>
 
@@ -22,6 +24,12 @@ void debug_out(int level, const char* fmt, ...) {
   fprintf(stderr, "[parsec_amd:%d] %s\n", level, buf);
 }
 
+static std::atomic<void (*)(const char*)> g_fatal_handler{nullptr};
+
+void set_fatal_handler(void (*fn)(const char* msg)) {
+  g_fatal_handler.store(fn);
+}
+
 void fatal(const char* fmt, ...) {
   char buf[4096];
   va_list ap;
@@ -30,6 +38,7 @@ void fatal(const char* fmt, ...) {
   va_end(ap);
   fprintf(stderr, "[parsec_amd FATAL] %s\n", buf);
   fflush(stderr);
+  if (auto* fn = g_fatal_handler.exchange(nullptr)) fn(buf);
   abort();
 }
 

@@ -64,6 +64,12 @@ void debug_out(int level, const char* fmt, ...)
 #define PA_DEBUG(lvl, ...) \
   do { if (::pa::debug_level() >= (lvl)) ::pa::debug_out(lvl, __VA_ARGS__); } while (0)
 
+// Error callback invoked (once) with the message before the process
+// aborts (runtime.h:34-38 parsec_weaksym_exit / error-callback analog):
+// lets an embedding application log/flush/teardown. The handler must not
+// insert tasks; returning continues into abort().
+void set_fatal_handler(void (*fn)(const char* msg));
+
 [[noreturn]] void fatal(const char* fmt, ...)
 #if defined(__GNUC__)
     __attribute__((format(printf, 1, 2)))

@@ -385,6 +385,18 @@ PYBIND11_MODULE(_core, m) {
   });
   m.def("pins_remove", [](int id) { Pins::inst().remove(id); });
 
+  // Fatal-error callback (error-callback analog). The Python callable is
+  // invoked with the message right before abort(); use it to flush logs.
+  m.def("set_fatal_handler", [](py::function fn) {
+    static py::function* g_fn = nullptr;
+    if (g_fn) { delete g_fn; }
+    g_fn = new py::function(std::move(fn));
+    set_fatal_handler(+[](const char* msg) {
+      py::gil_scoped_acquire g;
+      try { (*reinterpret_cast<py::function*>(g_fn))(msg); } catch (...) {}
+    });
+  });
+
   m.def("param_set", &param_set);
   m.def("param_dump", &param_dump);
   m.def("hip_device_count", [] {

@@ -277,3 +277,48 @@ del A, ctx
     for pr in procs:
         o, _ = pr.communicate(timeout=180)
         assert pr.returncode == 0 and b"XRANK_CTL_OK" in o, o.decode()
+
+
+def test_broadcast_world2(tmp_path):
+    """One-to-many broadcast: sent-mask dedup => one send per dest rank."""
+    import subprocess
+    import sys as _sys
+    code = f"""
+import os, sys, struct
+sys.path.insert(0, {REPO!r})
+import parsec_amd as pm
+from parsec_amd.ptg import compile_jdf
+rank = int(os.environ["RANK"])
+pm.param_set("comm_base_port", os.environ["PORT"])
+ctx = pm.Context(nworkers=2, rank=rank, world=2, comm="tcp", gpu=-2)
+NT = 8
+A = pm.TiledMatrix(ctx, NT, 4, 1, 4, 2, 1)
+mod = compile_jdf(os.path.join({EX!r}, "broadcast.jdf"))
+tp = pm.Dtd(ctx, "bcast")
+mod.build(ctx, tp, mydata=A, NT=NT)
+tp.wait()
+ctx.barrier()
+for k in range(NT):
+    if A.is_local(k, 0):
+        vals = struct.unpack("<4d", A.tile_bytes(k, 0))
+        assert vals == tuple(10.0 + i + k for i in range(4)), (k, vals)
+# root sent the tile to rank 1 exactly once (plus readers on rank 0 free)
+c = ctx.counters()
+if rank == 0:
+    assert c["comm_msgs"] <= 2, c  # one payload (+barrier accounting)
+print("BCAST_OK", rank)
+ctx.barrier()
+del A, ctx
+"""
+    import conftest
+    port = str(conftest.port_base(15))
+    procs = []
+    for r in range(2):
+        env = dict(os.environ)
+        env.update(RANK=str(r), WORLD_SIZE="2", PORT=port)
+        procs.append(subprocess.Popen([_sys.executable, "-c", code], env=env,
+                                      stdout=subprocess.PIPE,
+                                      stderr=subprocess.STDOUT))
+    for pr in procs:
+        o, _ = pr.communicate(timeout=180)
+        assert pr.returncode == 0 and b"BCAST_OK" in o, o.decode()

@@ -3,9 +3,12 @@
 Mirrors the reference's tests/dsl/dtd/* programs (dtd_test_war.c etc.,
 SURVEY.md §4) using Python task bodies on the CPU path.
 """
+import os
 import threading
 
 import parsec_amd as pm
+
+REPO = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
 
 
 def test_context_props(ctx):
@@ -254,3 +257,21 @@ def test_untied_tasks(ctx):
     tp.insert_py(root, flows=[(A.tile(0, 0), pm.ACCESS_INOUT)])
     tp.wait()
     assert sorted(hits) == list(range(16)), hits
+
+
+def test_fatal_handler():
+    """Error callback fires with the message before abort (weaksym_exit)."""
+    import subprocess
+    import sys as _sys
+    code = f"""
+import sys; sys.path.insert(0, {REPO!r})
+import parsec_amd as pm
+pm.set_fatal_handler(lambda msg: print("HANDLER_SAW:", msg, flush=True))
+ctx = pm.Context(nworkers=1, rank=0, world=1, gpu=-2)
+A = pm.TiledMatrix(ctx, 64, 64, 32, 32, 1, 1)
+A.tile(5, 5)  # out of range -> fatal
+"""
+    r = subprocess.run([_sys.executable, "-c", code], capture_output=True,
+                       text=True, timeout=60)
+    assert r.returncode != 0
+    assert "HANDLER_SAW:" in r.stdout, r.stdout + r.stderr

@@ -38,6 +38,7 @@ from parsec_amd._core import (  # noqa: F401,E402
     insert_potrf_panel,
     param_dump,
     param_set,
+    hip_bandwidth,
     pins_add,
     pins_remove,
     set_fatal_handler,

@@ -281,8 +281,10 @@ def _parse_class(cls, lines, i):
                 fl.deps.append(_parse_dep(arrow, chunk))
             if fl.is_ctl:
                 has_in = any(d.direction == "<-" for d in fl.deps)
-                has_out = any(d.direction == "->" for d in fl.deps)
-                fl.mode = (3 if (has_in and has_out) else (1 if has_in else 2))
+                # pure producers are INOUT on the token: the read makes the
+                # WAW chain transitive across ranks (gather correctness)
+                fl.mode = 3 if not has_in or any(
+                    d.direction == "->" for d in fl.deps) else 1
             cls.flows.append(fl)
             continue
         # range or local:  name = expr [.. expr [.. expr]]
@@ -302,6 +304,16 @@ def _parse_class(cls, lines, i):
 
 
 # --------------------------------------------------------------- codegen
+def _range_parts(a):
+    """Top-level `lo .. hi` in a task-ref arg (gather/multicast ranges)."""
+    parts = [x.strip() for x in _split_top(a, [".."])]
+    return parts if len(parts) > 1 else None
+
+
+def _term_has_range(term):
+    return term[0] == "task" and any(_range_parts(a) for a in term[3])
+
+
 def _cxx_expr(e):
     """JDF expression -> C++ (inline-C %{..%} becomes a lambda)."""
     e = e.strip()
@@ -411,13 +423,24 @@ def generate_cpp(jdf, name):
             w(f"  case {k}: {{  // flow {f.name}")
             ins = [d for d in f.deps if d.direction == "<-"]
             outs = [d for d in f.deps if d.direction == "->"]
+            def in_tile(term, props):
+                # CTL gather (`<- X Pred(0..N)`): all preds converge on THIS
+                # instance's token (ctlgat.jdf analog); ranged data IN has
+                # no single producer and is rejected.
+                if _term_has_range(term):
+                    if not f.is_ctl:
+                        raise JdfError(
+                            f"{c.name}.{f.name}: ranged <- on a data flow")
+                    return term_tile(("new",), {"size": "8"}, c)
+                return term_tile(term, props, c)
+
             for d in ins:
                 if d.guard is not None:
-                    w(f"    if ({_cxx_expr(d.guard)}) return {term_tile(d.term, d.props, c)};")
+                    w(f"    if ({_cxx_expr(d.guard)}) return {in_tile(d.term, d.props)};")
                     if d.else_term is not None:
-                        w(f"    else return {term_tile(d.else_term, d.props, c)};")
+                        w(f"    else return {in_tile(d.else_term, d.props)};")
                 else:
-                    w(f"    return {term_tile(d.term, d.props, c)};")
+                    w(f"    return {in_tile(d.term, d.props)};")
             # WRITE-only flow: bind to the output collection ref
             bound = False
             for d in outs:
@@ -434,6 +457,23 @@ def generate_cpp(jdf, name):
                 if sz is not None:
                     w(f"    return {term_tile(('new',), {'size': sz}, c)};")
             if f.is_ctl:
+                has_in = any(d.direction == "<-" for d in f.deps)
+                if not has_in:
+                    for d in outs:
+                        t = d.term
+                        if t[0] != "task" or _term_has_range(t):
+                            continue
+                        tgt = jdf.classes[cls_index[t[2]]]
+                        tf = next((x for x in tgt.flows if x.name == t[1]),
+                                  None)
+                        if tf is not None and any(
+                                d2.direction == "<-" and
+                                _term_has_range(d2.term)
+                                for d2 in tf.deps):
+                            g = (f"if ({_cxx_expr(d.guard)}) "
+                                 if d.guard else "")
+                            w(f"    {g}return {term_tile(t, d.props, c)};")
+                            break
                 # CTL origin (or unguarded base case): own 8-byte token
                 w(f"    return {term_tile(('new',), {'size': '8'}, c)};")
             w("    break; }")
@@ -547,12 +587,25 @@ def generate_cpp(jdf, name):
                     if term[0] != "task":
                         continue
                     _, fname, tcls, args = term
-                    exprs = "".join(
-                        f" _k.second[{ai}] = (long)({_cxx_expr(a)});"
-                        for ai, a in enumerate(args))
+                    loops = ""
+                    exprs = ""
+                    closes = ""
+                    for ai, a in enumerate(args):
+                        rp = _range_parts(a)
+                        if rp:  # gather: one pred edge per range element
+                            lo, hi = _cxx_expr(rp[0]), _cxx_expr(rp[1])
+                            st = _cxx_expr(rp[2]) if len(rp) > 2 else "1"
+                            loops += (f" for (long _ga{ai} = (long)({lo}); "
+                                      f"_ga{ai} <= (long)({hi}); "
+                                      f"_ga{ai} += (long)({st})) {{")
+                            exprs += f" _k.second[{ai}] = _ga{ai};"
+                            closes += " }"
+                        else:
+                            exprs += f" _k.second[{ai}] = (long)({_cxx_expr(a)});"
                     cond = f"if ({_cxx_expr(guard)}) " if guard else ""
-                    w(f"{indent}{cond}{{ PKey _k; _k.first = {cls_index[tcls]};"
-                      f"{exprs} _in.pred_keys.push_back(_k); }}")
+                    w(f"{indent}{cond}{{{loops} PKey _k; "
+                      f"_k.first = {cls_index[tcls]};"
+                      f"{exprs} _in.pred_keys.push_back(_k);{closes} }}")
         w(f"{indent}_g.add(std::move(_in));")
         for _ in c.ranges:
             indent = indent[:-2]

@@ -322,3 +322,57 @@ del A, ctx
     for pr in procs:
         o, _ = pr.communicate(timeout=180)
         assert pr.returncode == 0 and b"BCAST_OK" in o, o.decode()
+
+
+def test_ctl_gather(ctx):
+    """Ranged CTL fan-in (ctlgat.jdf): Gather runs after ALL producers."""
+    import struct
+    mod = compile_jdf(os.path.join(EX, "ctl_gather.jdf"))
+    NT = 12
+    A = pm.TiledMatrix(ctx, NT, 1, 1, 1, 1, 1)
+    tp = pm.Dtd(ctx, "gat")
+    mod.build(ctx, tp, mydata=A, NT=NT)
+    tp.wait()
+    (g,) = struct.unpack("<q", A.tile_bytes(0, 0))
+    assert g == 1000 + NT, g  # all NT producers retired before the gather
+
+
+def test_ctl_gather_world2(tmp_path):
+    """Cross-rank CTL gather: producers on both ranks precede the gather."""
+    import subprocess
+    import sys as _sys
+    code = f"""
+import os, sys, struct
+sys.path.insert(0, {REPO!r})
+import parsec_amd as pm
+from parsec_amd.ptg import compile_jdf
+rank = int(os.environ["RANK"])
+pm.param_set("comm_base_port", os.environ["PORT"])
+ctx = pm.Context(nworkers=2, rank=rank, world=2, comm="tcp", gpu=-2)
+NT = 8
+A = pm.TiledMatrix(ctx, NT, 1, 1, 1, 2, 1)
+mod = compile_jdf(os.path.join({EX!r}, "ctl_gather.jdf"))
+tp = pm.Dtd(ctx, "gat2")
+mod.build(ctx, tp, mydata=A, NT=NT)
+tp.wait()
+ctx.barrier()
+if A.is_local(0, 0):
+    n_local = sum(1 for k in range(NT) if A.is_local(k, 0))
+    (g,) = struct.unpack("<q", A.tile_bytes(0, 0))
+    assert g == 1000 + n_local, (g, n_local)
+print("GATHER_OK", rank)
+ctx.barrier()
+del A, ctx
+"""
+    import conftest
+    port = str(conftest.port_base(17))
+    procs = []
+    for r in range(2):
+        env = dict(os.environ)
+        env.update(RANK=str(r), WORLD_SIZE="2", PORT=port)
+        procs.append(subprocess.Popen([_sys.executable, "-c", code], env=env,
+                                      stdout=subprocess.PIPE,
+                                      stderr=subprocess.STDOUT))
+    for pr in procs:
+        o, _ = pr.communicate(timeout=180)
+        assert pr.returncode == 0 and b"GATHER_OK" in o, o.decode()

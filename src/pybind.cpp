@@ -171,7 +171,8 @@ PYBIND11_MODULE(_core, m) {
   py::class_<Data>(m, "Data")
       .def_property_readonly("home_rank", [](Data& d) { return d.home_rank; })
       .def_property_readonly("owner_rank", [](Data& d) { return d.owner_rank; })
-      .def_property_readonly("version", [](Data& d) { return d.version; });
+      .def_property_readonly("version", [](Data& d) { return d.version; })
+      .def_property_readonly("nbytes", [](Data& d) { return d.bytes; });
 
   py::class_<TiledMatrix>(m, "TiledMatrix")
       .def(py::init<Context*, int64_t, int64_t, int, int, int, int, size_t,

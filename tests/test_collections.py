@@ -289,3 +289,57 @@ del S, D, ctx
     for pr in procs:
         o, _ = pr.communicate(timeout=180)
         assert pr.returncode == 0 and b"REGRID2_OK" in o, o.decode()
+
+
+def test_merge_sort_app(ctx):
+    """Mini-app: task-tree merge sort (tests/apps/merge_sort analog)."""
+    import importlib.util
+    spec = importlib.util.spec_from_file_location(
+        "merge_sort", os.path.join(REPO, "examples", "merge_sort.py"))
+    ms = importlib.util.module_from_spec(spec)
+    spec.loader.exec_module(ms)
+    rng = np.random.default_rng(3)
+    vals = rng.standard_normal(10000)
+    nodes, root_key = ms.merge_sort(ctx, vals, chunk=1024)
+    out = np.frombuffer(nodes.bytes_get(root_key), dtype=np.float64)
+    assert np.array_equal(out, np.sort(vals))
+
+
+def test_merge_sort_world2(tmp_path):
+    import subprocess
+    import sys as _sys
+    from conftest import port_base
+    code = f"""
+import os, sys
+sys.path.insert(0, {REPO!r})
+import numpy as np
+import parsec_amd as pm
+import importlib.util
+spec = importlib.util.spec_from_file_location(
+    "merge_sort", os.path.join({REPO!r}, "examples", "merge_sort.py"))
+ms = importlib.util.module_from_spec(spec); spec.loader.exec_module(ms)
+rank = int(os.environ["RANK"])
+pm.param_set("comm_base_port", os.environ["PORT"])
+ctx = pm.Context(nworkers=2, rank=rank, world=2, comm="tcp", gpu=-2)
+rng = np.random.default_rng(3)
+vals = rng.standard_normal(8192)
+nodes, root_key = ms.merge_sort(ctx, vals, chunk=512)
+ctx.barrier()
+if nodes.at(root_key).home_rank == ctx.rank:
+    out = np.frombuffer(nodes.bytes_get(root_key), dtype=np.float64)
+    assert np.array_equal(out, np.sort(vals))
+print("MSORT_OK", rank)
+ctx.barrier()
+del nodes, ctx
+"""
+    port = str(port_base(19))
+    procs = []
+    for r in range(2):
+        env = dict(os.environ)
+        env.update(RANK=str(r), WORLD_SIZE="2", PORT=port)
+        procs.append(subprocess.Popen([_sys.executable, "-c", code], env=env,
+                                      stdout=subprocess.PIPE,
+                                      stderr=subprocess.STDOUT))
+    for pr in procs:
+        o, _ = pr.communicate(timeout=180)
+        assert pr.returncode == 0 and b"MSORT_OK" in o, o.decode()

@@ -92,3 +92,93 @@ def test_distributed_qr(tmp_path):
     R = np.triu(post)
     err = np.abs(R.T @ R - pre.T @ pre).max() / np.abs(pre.T @ pre).max()
     assert err < 1e-12, f"distributed QR rel err {err}"
+
+
+FUZZ_CODE = r"""
+import os, sys
+sys.path.insert(0, os.environ["PARSEC_REPO"])
+import random
+import numpy as np
+import parsec_amd as pm
+
+rank = int(os.environ["RANK"]); world = int(os.environ["WORLD_SIZE"])
+pm.param_set("comm_base_port", os.environ["PORT"])
+ctx = pm.Context(nworkers=2, rank=rank, world=world, comm="tcp", gpu=-2)
+rng = random.Random(int(os.environ["SEED"]))  # identical on all ranks
+NT, nb = 10, 16
+A = pm.TiledMatrix(ctx, NT * nb, nb, nb, nb, world, 1)
+tp = pm.Dtd(ctx, "fuzz")
+oracle = []
+for i in range(NT):
+    v = np.full((nb, nb), float(i + 1))
+    oracle.append(v.copy())
+    if A.is_local(i, 0):
+        A.tile_numpy_set(i, 0, v)
+ctx.barrier()
+
+for step in range(int(os.environ.get("OPS", "150"))):
+    target = rng.randrange(NT)
+    nsrc = rng.randrange(0, 3)
+    srcs = rng.sample([i for i in range(NT) if i != target], nsrc)
+    exec_rank = rng.randrange(world)
+    coeffs = [round(rng.uniform(-1, 1), 3) for _ in range(nsrc + 1)]
+    write_only = nsrc == 0 and rng.random() < 0.3
+    # oracle replay (pure numpy, every rank computes the same)
+    if write_only:
+        oracle[target] = np.full((nb, nb), coeffs[0])
+    else:
+        acc = coeffs[0] * oracle[target]
+        for c, s in zip(coeffs[1:], srcs):
+            acc = acc + c * oracle[s]
+        oracle[target] = acc
+    # task body (only runs on exec_rank)
+    def body(tbuf, *sbufs, coeffs=coeffs, write_only=write_only):
+        t = np.frombuffer(tbuf, dtype=np.float64)
+        if write_only:
+            t[:] = coeffs[0]
+            return
+        acc = coeffs[0] * t
+        for c, sb in zip(coeffs[1:], sbufs):
+            acc = acc + c * np.frombuffer(sb, dtype=np.float64)
+        t[:] = acc
+    flows = [(A.tile(target, 0),
+              pm.ACCESS_OUT if write_only else pm.ACCESS_INOUT)]
+    flows += [(A.tile(s, 0), pm.ACCESS_IN) for s in srcs]
+    tp.insert_py(body, flows=flows, rank=exec_rank, with_data=True)
+
+tp.wait()
+tp.flush_all(A)
+tp.wait()
+ctx.barrier()
+bad = 0
+for i in range(NT):
+    if A.is_local(i, 0):
+        got = A.tile_numpy(i, 0)
+        if not np.allclose(got, oracle[i], atol=1e-9):
+            print(f"rank {rank}: tile {i} diverged "
+                  f"(max err {np.abs(got - oracle[i]).max()})")
+            bad += 1
+assert bad == 0
+print("FUZZ_OK", rank)
+ctx.barrier()
+del A, ctx
+"""
+
+
+@pytest.mark.parametrize("world,seed", [(2, 101), (2, 202), (4, 303)])
+def test_distributed_fuzz_vs_oracle(world, seed, tmp_path):
+    """Random DAGs (random tiles/modes/executing ranks) through the full
+    SPMD protocol must match a sequential numpy oracle."""
+    port = _next_port[0]
+    _next_port[0] += world + 2
+    procs = []
+    for r in range(world):
+        env = dict(os.environ)
+        env.update(RANK=str(r), WORLD_SIZE=str(world), PORT=str(port),
+                   SEED=str(seed), PARSEC_REPO=os.path.dirname(HERE))
+        procs.append(subprocess.Popen([sys.executable, "-c", FUZZ_CODE],
+                                      env=env, stdout=subprocess.PIPE,
+                                      stderr=subprocess.STDOUT))
+    for pr in procs:
+        out, _ = pr.communicate(timeout=180)
+        assert pr.returncode == 0 and b"FUZZ_OK" in out, out.decode()

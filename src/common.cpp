@@ -38,6 +38,9 @@ void fatal(const char* fmt, ...) {
   va_end(ap);
   fprintf(stderr, "[parsec_amd FATAL] %s\n", buf);
   fflush(stderr);
+  // postmortem: dump the debug-history ring (PARSEC_MCA_debug_history=N)
+  void debug_history_dump();
+  debug_history_dump();
   if (auto* fn = g_fatal_handler.exchange(nullptr)) fn(buf);
   abort();
 }

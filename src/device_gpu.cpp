@@ -286,6 +286,9 @@ void GpuEngine::stage_in_and_launch(Task* t) {
   }
   std::vector<std::pair<void*, size_t>> deferred;
   GpuTaskCtx gctx{es, device_, this, &deferred};
+  if (debug_history_on())
+    debug_history_add("gpu_submit %s seq=%lu", t->tc->name.c_str(),
+                      (unsigned long)t->seq);
   if (roctx_on()) roctx_push(t->tc->name.c_str());
   t->tc->gpu_hook(*t, gctx);
   if (roctx_on()) roctx_pop();

@@ -2,6 +2,8 @@
 
 #include <dlfcn.h>
 
+#include <cstdarg>
+
 #include <chrono>
 
 #include "runtime.hpp"
@@ -134,4 +136,59 @@ void roctx_init() {
 bool roctx_on() { return roctx_enabled; }
 void roctx_push(const char* name) { if (roctx_enabled) p_roctx_push(name); }
 void roctx_pop() { if (roctx_enabled) p_roctx_pop(); }
+}  // namespace pa
+
+// ------------------------------------------------------- debug history
+namespace pa {
+namespace {
+struct DebugHistory {
+  std::mutex mtx;
+  std::vector<std::string> ring;
+  size_t cap = 0, next = 0;
+  uint64_t dropped = 0;
+};
+DebugHistory& dh() {
+  static DebugHistory h;
+  return h;
+}
+}  // namespace
+
+void debug_history_init() {
+  long n = param_int("debug_history", 0);
+  if (n > 0) {
+    auto& h = dh();
+    std::lock_guard<std::mutex> g(h.mtx);
+    h.cap = (size_t)n;
+    h.ring.resize(h.cap);
+  }
+}
+bool debug_history_on() { return dh().cap != 0; }
+
+void debug_history_add(const char* fmt, ...) {
+  auto& h = dh();
+  if (!h.cap) return;
+  char buf[256];
+  va_list ap;
+  va_start(ap, fmt);
+  vsnprintf(buf, sizeof(buf), fmt, ap);
+  va_end(ap);
+  char line[320];
+  snprintf(line, sizeof(line), "[%14.6f] %s",
+           (double)Profiler::now_ns() / 1e9, buf);
+  std::lock_guard<std::mutex> g(h.mtx);
+  h.ring[h.next % h.cap] = line;
+  h.next++;
+}
+
+void debug_history_dump() {
+  auto& h = dh();
+  if (!h.cap) return;
+  std::lock_guard<std::mutex> g(h.mtx);
+  size_t n = h.next < h.cap ? h.next : h.cap;
+  size_t start = h.next < h.cap ? 0 : h.next - h.cap;
+  fprintf(stderr, "[parsec_amd] debug history (last %zu events):\n", n);
+  for (size_t i = 0; i < n; i++)
+    fprintf(stderr, "  %s\n", h.ring[(start + i) % h.cap].c_str());
+  fflush(stderr);
+}
 }  // namespace pa

@@ -77,6 +77,20 @@ class Profiler {
   std::mutex dot_mtx_;
 };
 
+// Debug history ring (PARSEC_DEBUG_HISTORY analog, debug.c history
+// buffers): when PARSEC_MCA_debug_history=N, the last N runtime events
+// (task begin/complete/schedule, comm posts, GPU submit/retire) are kept
+// in a ring and dumped to stderr when fatal() fires — the postmortem for
+// hangs/aborts without a full trace.
+void debug_history_init();
+bool debug_history_on();
+void debug_history_add(const char* fmt, ...)
+#if defined(__GNUC__)
+    __attribute__((format(printf, 1, 2)))
+#endif
+    ;
+void debug_history_dump();
+
 // rocTX ranges (the reference's NVTX sink, profiling_nvtx.c, done the
 // ROCm-native way): with PARSEC_MCA_profile_roctx=1, task bodies and GPU
 // submissions get roctxRangePush/Pop so rocprofv3's marker trace shows

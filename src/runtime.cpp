@@ -62,6 +62,9 @@ void task_complete(Task* t) {
   succs.swap(t->succs);
   t->lock.unlock();
   for (Task* s : succs) task_dec_deps(s);
+  if (debug_history_on())
+    debug_history_add("done %s seq=%lu", t->tc->name.c_str(),
+                      (unsigned long)t->seq);
   PA_PINS(PinsEv::COMPLETE, t, Context::tls_worker_id);
   Taskpool* tp = t->tp;
   t->release();
@@ -76,6 +79,9 @@ void run_cpu_task(Task* t) {
     if (t->flows[i].data && t->flows[i].mode == ACCESS_OUT)
       t->flows[i].data->begin_host_overwrite();
   if (t->tc->cpu_hook) {
+    if (debug_history_on())
+      debug_history_add("exec %s seq=%lu w=%d", t->tc->name.c_str(),
+                        (unsigned long)t->seq, Context::tls_worker_id);
     PA_PINS(PinsEv::EXEC_BEGIN, t, Context::tls_worker_id);
     if (roctx_on()) roctx_push(t->tc->name.c_str());
     if (pr.enabled()) {
@@ -243,6 +249,7 @@ Context::Context(const Options& opt) : rank_(opt.rank), world_(opt.world) {
   comm_ = CommEngine::create(this, opt.comm);
 
   roctx_init();
+  debug_history_init();
   std::string prof = param_str("profile_filename", "");
   if (!prof.empty()) Profiler::inst().start(prof + "." + std::to_string(rank_));
   std::string dot = param_str("profile_dot", "");

@@ -115,3 +115,23 @@ del A, ctx
     r = subprocess.run([sys.executable, "-c", code], capture_output=True,
                        text=True, timeout=120)
     assert "PINS_OK" in r.stdout, r.stdout + r.stderr
+
+
+def test_debug_history_on_fatal():
+    """PARSEC_MCA_debug_history=N dumps the event ring on fatal (debug
+    history analog)."""
+    code = f"""
+import sys; sys.path.insert(0, {REPO!r})
+import parsec_amd as pm
+pm.param_set("debug_history", "64")
+ctx = pm.Context(nworkers=2, rank=0, world=1, gpu=-2)
+A = pm.TiledMatrix(ctx, 128, 128, 64, 64, 1, 1)
+tp = pm.Dtd(ctx)
+pm.insert_spd_fill(tp, A, 1)
+tp.wait()
+A.tile(9, 9)  # out of range -> fatal -> history dump
+"""
+    r = subprocess.run([sys.executable, "-c", code], capture_output=True,
+                       text=True, timeout=120)
+    assert r.returncode != 0
+    assert "debug history" in r.stderr and "spd_fill" in r.stderr, r.stderr

@@ -103,10 +103,11 @@ void Dtd::write_flow(Data* d, Task* t, int task_rank) {
   // keeps the old version so a later local reader triggers a fetch.
 }
 
-void Dtd::insert(const TaskClass* tc, const void* args, size_t args_bytes,
+bool Dtd::insert(const TaskClass* tc, const void* args, size_t args_bytes,
                  const FlowSpec* flows, int nflows, int priority, int rank) {
   Task* t = insert_begin(tc, args, args_bytes, flows, nflows, priority, rank);
   insert_commit(t);
+  return t != nullptr;
 }
 
 Task* Dtd::insert_begin(const TaskClass* tc, const void* args,

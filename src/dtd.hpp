@@ -38,8 +38,9 @@ class Dtd : public Taskpool {
 
   // Insert one task. `rank` -1 selects the home rank of the first written
   // tile (AFFINITY default). Runs/creates the task only on its rank; all
-  // ranks update the replicated tile state machines.
-  void insert(const TaskClass* tc, const void* args, size_t args_bytes,
+  // ranks update the replicated tile state machines. Returns true when
+  // the task is LOCAL (a Task object was created on this rank).
+  bool insert(const TaskClass* tc, const void* args, size_t args_bytes,
               const FlowSpec* flows, int nflows, int priority = 0,
               int rank = -1);
 

@@ -65,9 +65,15 @@ class Pins {
   }
 
   void fire(PinsEv e, const Task* t, int worker) {
-    std::lock_guard<std::mutex> g(mtx_);
-    for (auto& c : cbs_)
-      if (c.mask & (1u << (int)e)) c.fn(e, t, worker);
+    // snapshot under the lock, invoke outside it: callbacks may register/
+    // remove callbacks (or block) without deadlocking the chain
+    std::vector<PinsFn> snap;
+    {
+      std::lock_guard<std::mutex> g(mtx_);
+      for (auto& c : cbs_)
+        if (c.mask & (1u << (int)e)) snap.push_back(c.fn);
+    }
+    for (auto& fn : snap) fn(e, t, worker);
   }
 
  private:

@@ -306,9 +306,14 @@ PYBIND11_MODULE(_core, m) {
              std::vector<Dtd::FlowSpec> fs;
              for (auto& [d, mode] : flows)
                fs.push_back({d, (AccessMode)mode});
-             py::gil_scoped_release rel;
-             tp.insert(&py_task_class(), &pl, sizeof(pl), fs.data(),
-                       (int)fs.size(), priority, rank);
+             bool local;
+             {
+               py::gil_scoped_release rel;
+               local = tp.insert(&py_task_class(), &pl, sizeof(pl),
+                                 fs.data(), (int)fs.size(), priority, rank);
+             }
+             // remote tasks never run here: release the body's ref
+             if (!local) Py_XDECREF(pl.fn);
            },
            py::arg("fn"), py::arg("flows") = std::vector<std::pair<Data*, int>>{},
            py::arg("priority") = 0, py::arg("rank") = -1,

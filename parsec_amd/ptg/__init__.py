@@ -74,6 +74,7 @@ class Jdf:
         self.prologue = []
         self.globals_ = []              # (name, props)
         self.classes = []
+        self.options = {}               # %option key = value
 
 
 def _strip_comments(s):
@@ -193,6 +194,19 @@ def parse_jdf(text):
     while i < n:
         line = lines[i].strip()
         if not line:
+            i += 1
+            continue
+        mo = re.match(r"^%option\s+(\w+)\s*=?\s*(.*)$", line)
+        if mo:
+            # per-taskpool %option lines (jdf.c options): recognized ones
+            # are stashed on the jdf; unknown ones warn and are ignored
+            # (they tune reference-internal engines we do not reproduce).
+            jdf.options[mo.group(1)] = mo.group(2).strip() or "true"
+            known = {"no_taskpool_instance", "taskpool_instance",
+                     "dependencies_mark", "warnings", "compile_deps"}
+            if mo.group(1) not in known:
+                print(f"[ptgpp] note: %option {mo.group(1)} ignored",
+                      file=sys.stderr)
             i += 1
             continue
         m = re.match(r"^(\w+)\s*\(([^)]*)\)\s*$", line)

@@ -182,3 +182,26 @@ def test_distributed_fuzz_vs_oracle(world, seed, tmp_path):
     for pr in procs:
         out, _ = pr.communicate(timeout=180)
         assert pr.returncode == 0 and b"FUZZ_OK" in out, out.decode()
+
+
+def test_pingpong_benchmark(tmp_path):
+    """The ping-pong harness (rtt/bandwidth.jdf analog) runs and reports."""
+    port = _next_port[0]
+    _next_port[0] += 4
+    procs = []
+    for r in range(2):
+        env = dict(os.environ)
+        env.update(RANK=str(r), WORLD_SIZE="2",
+                   PARSEC_MCA_comm_base_port=str(port),
+                   MASTER_ADDR="127.0.0.1", MASTER_PORT=str(port + 2))
+        procs.append(subprocess.Popen(
+            [sys.executable, os.path.join(os.path.dirname(HERE),
+                                          "benchmarks", "bench_pingpong.py"),
+             "--hops", "40", "--sizes", "4096"],
+            env=env, stdout=subprocess.PIPE, stderr=subprocess.STDOUT))
+    outs = []
+    for pr in procs:
+        out, _ = pr.communicate(timeout=180)
+        outs.append(out.decode())
+        assert pr.returncode == 0, out.decode()
+    assert any("rtt_us" in o for o in outs), outs

@@ -227,20 +227,11 @@ void GpuEngine::sync_all() {
 // Stage-in copies for every READ flow whose valid copy is on the host
 // (parsec_device_data_stage_in, device_gpu.c:1800-2168, minus the peer-GPU
 // branch: peers are other processes here, reached through the comm engine).
-void GpuEngine::stage_in_and_launch(Task* t) {
-  // Stream 0 is reserved for critical-path (panel) tasks so they never
-  // queue behind bulk updates; others round-robin over the remaining
-  // streams (the reference's exec_stream[2..n] round-robin,
-  // device_gpu.c:3445-3535, with an express lane added).
-  int si;
-  if (t->priority >= (1 << 19) || (int)exec_streams_.size() == 1) {
-    si = 0;
-  } else {
-    si = next_stream_;
-    next_stream_ = next_stream_ + 1;
-    if (next_stream_ >= (int)exec_streams_.size()) next_stream_ = 1;
-  }
-  hipStream_t es = exec_streams_[si];
+// Per-flow pin + device allocation + H2D staging + copy-fence wait; sets
+// t->dev_ptr[]. Thread-safe: callable from the manager and from worker
+// threads running blocking chores (per-Data locks serialize the
+// decisions; HIP stream APIs are thread-safe).
+void GpuEngine::stage_flows(Task* t, hipStream_t es) {
   for (int i = 0; i < t->nflows; i++) {
     Data* d = t->flows[i].data;
     if (!d) { t->dev_ptr[i] = nullptr; continue; }
@@ -284,6 +275,23 @@ void GpuEngine::stage_in_and_launch(Task* t) {
       PA_HIP_CHECK(hipStreamWaitEvent(es, (hipEvent_t)d->h2d_event, 0));
     t->dev_ptr[i] = d->dev_ptr;
   }
+}
+
+void GpuEngine::stage_in_and_launch(Task* t) {
+  // Stream 0 is reserved for critical-path (panel) tasks so they never
+  // queue behind bulk updates; others round-robin over the remaining
+  // streams (the reference's exec_stream[2..n] round-robin,
+  // device_gpu.c:3445-3535, with an express lane added).
+  int si;
+  if (t->priority >= (1 << 19) || (int)exec_streams_.size() == 1) {
+    si = 0;
+  } else {
+    si = next_stream_;
+    next_stream_ = next_stream_ + 1;
+    if (next_stream_ >= (int)exec_streams_.size()) next_stream_ = 1;
+  }
+  hipStream_t es = exec_streams_[si];
+  stage_flows(t, es);
   std::vector<std::pair<void*, size_t>> deferred;
   GpuTaskCtx gctx{es, device_, this, &deferred};
   if (debug_history_on())
@@ -331,6 +339,34 @@ bool GpuEngine::retire_pass() {
     }
   }
   return progress;
+}
+
+void GpuEngine::run_blocking(Task* t) {
+  PA_HIP_CHECK(hipSetDevice(device_));
+  static thread_local hipStream_t bs = nullptr;
+  if (!bs) PA_HIP_CHECK(hipStreamCreateWithFlags(&bs, hipStreamNonBlocking));
+  stage_flows(t, bs);
+  std::vector<std::pair<void*, size_t>> deferred;
+  GpuTaskCtx gctx{bs, device_, this, &deferred};
+  if (debug_history_on())
+    debug_history_add("gpu_blocking %s seq=%lu", t->tc->name.c_str(),
+                      (unsigned long)t->seq);
+  if (roctx_on()) roctx_push(t->tc->name.c_str());
+  t->tc->gpu_hook(*t, gctx);
+  if (roctx_on()) roctx_pop();
+  PA_PINS(PinsEv::GPU_SUBMIT, t, -1);
+  PA_HIP_CHECK(hipStreamSynchronize(bs));
+  for (int i = 0; i < t->nflows; i++) {
+    Data* d = t->flows[i].data;
+    if (!d) continue;
+    if (t->flows[i].mode & ACCESS_OUT) d->written_on(true);
+    unpin(d);
+  }
+  for (auto& [p2, b2] : deferred) dev_free(p2, b2);
+  counters().tasks_executed_gpu.fetch_add(1, std::memory_order_relaxed);
+  PA_PINS(PinsEv::GPU_RETIRE, t, -1);
+  stats.tasks++;
+  task_complete(t);
 }
 
 void GpuEngine::manager_main() {

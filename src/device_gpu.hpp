@@ -53,6 +53,9 @@ class GpuEngine {
   ~GpuEngine();
 
   void enqueue(Task* t);  // task with satisfied deps, TaskKind::GPU
+  // Host-blocking chore executed synchronously on the CALLING (worker)
+  // thread with a thread-local stream (see TaskClass::gpu_blocking).
+  void run_blocking(Task* t);
 
   // Device memory pool (slab + size-class free lists + LRU eviction when
   // capacity is exhausted: clean copies are dropped, dirty ones written
@@ -100,6 +103,7 @@ class GpuEngine {
 
   void manager_main();
   void stage_in_and_launch(Task* t);
+  void stage_flows(Task* t, hipStream_t es);
   hipEvent_t event_get();
   void event_put(hipEvent_t e);
 

@@ -557,7 +557,14 @@ static TaskClass make_qr_tc(const char* name, void (*cpu)(Task&),
 }
 
 TaskClass& tc_geqrt() {
-  static TaskClass tc = make_qr_tc("geqrt", cpu_geqrt, gpu_geqrt, 10);
+  // Panel factorizations host-sync inside rocSOLVER: run them on worker
+  // threads (gpu_blocking) so the manager keeps the trailing updates
+  // flowing. PARSEC_MCA_qr_blocking_panels=0 restores manager execution.
+  static TaskClass tc = [] {
+    TaskClass c = make_qr_tc("geqrt", cpu_geqrt, gpu_geqrt, 10);
+    c.gpu_blocking = param_int("qr_blocking_panels", 1) != 0;
+    return c;
+  }();
   return tc;
 }
 TaskClass& tc_unmqr() {
@@ -565,7 +572,11 @@ TaskClass& tc_unmqr() {
   return tc;
 }
 TaskClass& tc_tsqrt() {
-  static TaskClass tc = make_qr_tc("tsqrt", cpu_tsqrt, gpu_tsqrt, 12);
+  static TaskClass tc = [] {
+    TaskClass c = make_qr_tc("tsqrt", cpu_tsqrt, gpu_tsqrt, 12);
+    c.gpu_blocking = param_int("qr_blocking_panels", 1) != 0;
+    return c;
+  }();
   return tc;
 }
 TaskClass& tc_tsmqr() {

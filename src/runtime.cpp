@@ -72,6 +72,12 @@ void task_complete(Task* t) {
 }
 
 void run_cpu_task(Task* t) {
+  if (t->tc->kind == TaskKind::GPU && t->tp->context()->gpu()) {
+    // blocking GPU chore routed through the CPU scheduler (on CPU-only
+    // contexts GPU-kind tasks run their cpu_hook below instead)
+    t->tp->context()->gpu()->run_blocking(t);
+    return;
+  }
   Profiler& pr = Profiler::inst();
   // OUTPUT-only flows are about to be produced on the host: drop stale
   // device validity first so LRU writeback cannot race the body's writes.
@@ -314,6 +320,12 @@ void Context::dispatch(Task* t, int worker_hint) {
   PA_PINS(PinsEv::SCHEDULE, t, Context::tls_worker_id);
   switch (t->tc->kind) {
     case TaskKind::GPU:
+      if (gpu_ && t->tc->gpu_blocking) {
+        // host-synchronous chore: run on a worker so the manager keeps
+        // launching bulk work behind it
+        sched_->push(t, worker_hint);
+        return;
+      }
       if (gpu_) { gpu_->enqueue(t); return; }
       [[fallthrough]];
     case TaskKind::CPU:

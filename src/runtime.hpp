@@ -67,6 +67,12 @@ struct TaskClass {
   // a Python callable reference).
   void (*destruct)(Task&) = nullptr;
   double flops = 0.0;  // time_estimate analog (parsec_internal.h:411-459)
+  // GPU chores that synchronize the host internally (e.g. rocSOLVER's
+  // unblocked panel factorizations) run on a WORKER thread with their own
+  // stream instead of the engine's manager thread, so the manager keeps
+  // launching bulk work while the panel blocks (PARSEC_HOOK_RETURN_ASYNC
+  // decoupling analog for host-synchronous libraries).
+  bool gpu_blocking = false;
   int id = -1;
   // opaque user hooks for the C ABI (PTG-generated code) trampolines
   void* user_cpu = nullptr;

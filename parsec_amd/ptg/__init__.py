@@ -500,6 +500,7 @@ def generate_cpp(jdf, name):
     for c in jdf.classes:
         cpu_fn = "nullptr"
         gpu_fn = "nullptr"
+        gpu_flags = 0
         for props, code in c.bodies:
             btype = props.get("type", "CPU").upper()
             if btype in ("CUDA", "LEVEL_ZERO"):
@@ -509,6 +510,8 @@ def generate_cpp(jdf, name):
             dfl = data_flows(c)
             if btype in ("HIP", "GPU"):
                 gpu_fn = f"body_{c.name}_hip"
+                gpu_flags = 1 | (2 if props.get("blocking") in
+                                 ("on", "1", "true") else 0)
                 w(f"void {gpu_fn}(void* _t, void* _stream) {{")
                 w("  hipStream_t stream = (hipStream_t)_stream; (void)stream;")
                 w(alias)
@@ -532,7 +535,7 @@ def generate_cpp(jdf, name):
                 w(code)
                 w("  }")
                 w("}")
-        body_fns.append((cpu_fn, gpu_fn))
+        body_fns.append((cpu_fn, gpu_fn, gpu_flags))
 
     w("}  // namespace")
 
@@ -559,9 +562,9 @@ def generate_cpp(jdf, name):
     w("  Graph _g(_ctx, _dtd);")
     w("  static std::vector<void*> _tcs; if (_tcs.empty()) {")
     for k, c in enumerate(jdf.classes):
-        cpu_fn, gpu_fn = body_fns[k]
+        cpu_fn, gpu_fn, gpu_flags = body_fns[k]
         w(f'    _tcs.push_back(pa_taskclass_new("{c.name}", '
-          f"{1 if gpu_fn != 'nullptr' else 0}, {cpu_fn}, {gpu_fn}));")
+          f"{gpu_flags if gpu_fn != 'nullptr' else 0}, {cpu_fn}, {gpu_fn}));")
     w("  }")
     w("  _g.set_classes(_tcs);")
     for k, c in enumerate(jdf.classes):

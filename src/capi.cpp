@@ -44,12 +44,15 @@ PA_EXPORT int pa_tm_rank_of(void* tm, int i, int j) {
 }
 PA_EXPORT int pa_data_home_rank(void* d) { return ((Data*)d)->home_rank; }
 
-PA_EXPORT void* pa_taskclass_new(const char* name, int want_gpu,
+// flags: bit0 = has GPU body, bit1 = GPU body is host-blocking (runs on
+// a worker thread with its own stream — BODY [type=HIP blocking=on]).
+PA_EXPORT void* pa_taskclass_new(const char* name, int flags,
                                  void (*cpu)(void*),
                                  void (*gpu)(void*, void*)) {
   TaskClass* tc = new TaskClass();
   tc->name = name;
-  tc->kind = (want_gpu && gpu) ? TaskKind::GPU : TaskKind::CPU;
+  tc->kind = ((flags & 1) && gpu) ? TaskKind::GPU : TaskKind::CPU;
+  tc->gpu_blocking = (flags & 2) != 0;
   tc->user_cpu = (void*)cpu;
   tc->user_gpu = (void*)gpu;
   if (cpu) tc->cpu_hook = cpu_trampoline;

@@ -114,14 +114,18 @@ void cpu_larfb_kern(int m, int n, int k, const double* V, int ldv,
 }
 
 void stack_tiles(double* S, const double* top, bool triu_top,
-                 const double* bot, int nb, int ld) {
+                 const double* bot, bool triu_bot, int nb, int ld) {
   for (int c = 0; c < nb; c++) {
     for (int r = 0; r < nb; r++) {
       double v = top[(size_t)c * ld + r];
       S[(size_t)c * 2 * nb + r] = (triu_top && r > c) ? 0.0 : v;
     }
-    for (int r = 0; r < nb; r++)
-      S[(size_t)c * 2 * nb + nb + r] = bot[(size_t)c * ld + r];
+    for (int r = 0; r < nb; r++) {
+      double v = bot[(size_t)c * ld + r];
+      // tree combines: the bottom tile holds its level-0 V below the
+      // diagonal — only its R triangle participates
+      S[(size_t)c * 2 * nb + nb + r] = (triu_bot && r > c) ? 0.0 : v;
+    }
   }
 }
 
@@ -160,7 +164,7 @@ static void cpu_tsqrt(Task& t) {
   double* V2 = (double*)vd->ensure_host();
   double* T1 = (double*)td->ensure_host();
   memset(T1, 0, td->bytes);
-  stack_tiles(V2, Akk, true, Amk, nb, ld);
+  stack_tiles(V2, Akk, true, Amk, a.k != 0, nb, ld);
   std::vector<double> tau(nb);
   cpu_geqrf_kern(2 * nb, nb, V2, 2 * nb, tau.data());
   cpu_larft_kern(2 * nb, nb, V2, 2 * nb, tau.data(), T1, ld);
@@ -224,14 +228,15 @@ double* qr_scratch(GpuTaskCtx& g, int slot, size_t bytes) {
 }
 
 __global__ void k_stack_triu(double* S, const double* top, const double* bot,
-                             int nb, int ld) {
+                             int nb, int ld, int triu_bot) {
   int idx = blockIdx.x * blockDim.x + threadIdx.x;
   int total = nb * nb;
   for (; idx < total; idx += gridDim.x * blockDim.x) {
     int c = idx / nb, r = idx - c * nb;
     double v = top[(size_t)c * ld + r];
     S[(size_t)c * 2 * nb + r] = (r > c) ? 0.0 : v;
-    S[(size_t)c * 2 * nb + nb + r] = bot[(size_t)c * ld + r];
+    double w = bot[(size_t)c * ld + r];
+    S[(size_t)c * 2 * nb + nb + r] = (triu_bot && r > c) ? 0.0 : w;
   }
 }
 
@@ -510,7 +515,7 @@ static void gpu_tsqrt(Task& t, GpuTaskCtx& g) {
   double* T1 = (double*)t.dev_ptr[3];
   rocblas_handle h = qr_handle(g);
   hipLaunchKernelGGL(k_stack_triu, grid1d(nb * nb), dim3(256), 0, g.stream,
-                     V2, Akk, Amk, nb, ld);
+                     V2, Akk, Amk, nb, ld, a.k);
   static const bool use_rocsolver =
       param_str("chore_qr", "rocsolver") == "rocsolver";
   if (!use_rocsolver) {
@@ -585,6 +590,63 @@ TaskClass& tc_tsmqr() {
 }
 
 // ------------------------------------------------------------ DAG builder
+// Binary TS-reduction tree per panel column (the reference ecosystem's
+// HQR trees): every row tile is GEQRT'd independently (level 0), then
+// pairs combine through TSQRT/TSMQR with stride-doubling — panel depth
+// O(log M) instead of O(M). More total flops than the flat chain (the
+// per-row UNMQR applies), but the host-synced panel kernels run
+// CONCURRENTLY on blocking workers, which is what bounds QR here.
+static void insert_geqrf_tree(Dtd& tp, TiledMatrix& A, TiledMatrix& WT,
+                              TiledMatrix& T1, TiledMatrix& V2) {
+  const int T = A.mt();
+  const int nb = A.nb(), ld = A.mb();
+  constexpr int PANEL = 1 << 20;
+  TileArgs pa_args;
+  pa_args.m = nb;
+  pa_args.n = nb;
+  pa_args.ld = ld;
+  for (int k = 0; k < T; k++) {
+    // level 0: factor every row tile of the column, apply across its row
+    for (int m = k; m < T; m++) {
+      Dtd::FlowSpec f[] = {{A.tile(m, k), ACCESS_INOUT},
+                           {WT.tile(m, k), ACCESS_OUT}};
+      tp.insert(&tc_geqrt(), &pa_args, sizeof(pa_args), f, 2, PANEL + 1,
+                A.rank_of(m, k));
+      for (int n = k + 1; n < T; n++) {
+        Dtd::FlowSpec fu[] = {{A.tile(m, k), ACCESS_IN},
+                              {WT.tile(m, k), ACCESS_IN},
+                              {A.tile(m, n), ACCESS_INOUT}};
+        tp.insert(&tc_unmqr(), &pa_args, sizeof(pa_args), fu, 3,
+                  (1 << 18) - (n - k), A.rank_of(m, n));
+      }
+    }
+    // reduction tree: combine (a, a+step) pairs, doubling the stride
+    for (int step = 1; k + step < T; step *= 2) {
+      for (int a = k; a + step < T; a += 2 * step) {
+        const int b = a + step;
+        {
+          TileArgs ta = pa_args;
+          ta.k = 1;  // bottom tile participates by its R triangle only
+          Dtd::FlowSpec f[] = {{A.tile(a, k), ACCESS_INOUT},
+                               {A.tile(b, k), ACCESS_INOUT},
+                               {V2.tile(b, k), ACCESS_OUT},
+                               {T1.tile(b, k), ACCESS_OUT}};
+          tp.insert(&tc_tsqrt(), &ta, sizeof(ta), f, 4, PANEL,
+                    A.rank_of(b, k));
+        }
+        for (int n = k + 1; n < T; n++) {
+          Dtd::FlowSpec f[] = {{V2.tile(b, k), ACCESS_IN},
+                               {T1.tile(b, k), ACCESS_IN},
+                               {A.tile(a, n), ACCESS_INOUT},
+                               {A.tile(b, n), ACCESS_INOUT}};
+          tp.insert(&tc_tsmqr(), &pa_args, sizeof(pa_args), f, 4,
+                    -(n - k) * 4, A.rank_of(b, n));
+        }
+      }
+    }
+  }
+}
+
 void insert_geqrf(Dtd& tp, TiledMatrix& A) {
   const int T = A.mt();
   const int nb = A.nb(), ld = A.mb();
@@ -601,6 +663,10 @@ void insert_geqrf(Dtd& tp, TiledMatrix& A) {
   tp.own(WT);
   tp.own(T1);
   tp.own(V2);
+  if (param_str("qr_tree", "flat") == "binary") {
+    insert_geqrf_tree(tp, A, *WT, *T1, *V2);
+    return;
+  }
   for (int k = 0; k < T; k++) {
     TileArgs pa_args;
     pa_args.m = nb;

@@ -96,3 +96,38 @@ del A, ctx
     r = subprocess.run([sys.executable, "-c", code], env=env,
                        capture_output=True, text=True, timeout=300)
     assert r.returncode == 0, r.stdout + r.stderr
+
+
+def test_qr_binary_tree_cpu():
+    """Binary TS-reduction tree (qr_tree=binary) numerics vs flat."""
+    import os
+    import subprocess
+    import sys
+    repo = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+    code = f"""
+import sys; sys.path.insert(0, {repo!r})
+import numpy as np
+import parsec_amd as pm
+pm.param_set("qr_tree", "binary")
+ctx = pm.Context(nworkers=4, rank=0, world=1, gpu=-2)
+n, nb = 320, 64  # 5 row tiles -> uneven tree
+A = pm.TiledMatrix(ctx, n, n, nb, nb, 1, 1)
+rng = np.random.default_rng(5)
+A0 = rng.standard_normal((n, n))
+for tm in range(A.mt):
+    for tn in range(A.nt):
+        A.tile_numpy_set(tm, tn, A0[tm*nb:(tm+1)*nb, tn*nb:(tn+1)*nb])
+tp = pm.Dtd(ctx); pm.insert_geqrf(tp, A); tp.wait()
+R = np.zeros((n, n))
+for tm in range(A.mt):
+    for tn in range(A.nt):
+        R[tm*nb:(tm+1)*nb, tn*nb:(tn+1)*nb] = A.tile_numpy(tm, tn)
+R = np.triu(R)
+err = np.abs(R.T @ R - A0.T @ A0).max() / np.abs(A0.T @ A0).max()
+print("TREE_QR_ERR", err)
+assert err < 1e-12, err
+del A, ctx
+"""
+    r = subprocess.run([sys.executable, "-c", code],
+                       capture_output=True, text=True, timeout=300)
+    assert r.returncode == 0, r.stdout + r.stderr

@@ -205,3 +205,38 @@ def test_pingpong_benchmark(tmp_path):
         outs.append(out.decode())
         assert pr.returncode == 0, out.decode()
     assert any("rtt_us" in o for o in outs), outs
+
+
+def test_rccl_fallback_to_tcp(tmp_path):
+    """init_distributed falls back to the TCP engine when RCCL cannot
+    come up (here: no GPU) — agreed across ranks via gloo."""
+    code = r"""
+import os, sys
+sys.path.insert(0, os.environ["PARSEC_REPO"])
+import parsec_amd as pm
+pm.param_set("comm_base_port", os.environ["PORT"])
+ctx = pm.init_distributed(nworkers=2, comm="rccl", gpu=-2)
+assert ctx.world == 2
+A = pm.TiledMatrix(ctx, 256, 256, 64, 64, 2, 1)
+tp = pm.Dtd(ctx)
+pm.insert_spd_fill(tp, A, 3)
+pm.insert_potrf(tp, A)
+tp.wait()
+ctx.barrier()
+print("FALLBACK_OK", ctx.rank)
+del A, ctx
+"""
+    port = _next_port[0]
+    _next_port[0] += 4
+    procs = []
+    for r in range(2):
+        env = dict(os.environ)
+        env.update(RANK=str(r), WORLD_SIZE="2", PORT=str(port),
+                   MASTER_ADDR="127.0.0.1", MASTER_PORT=str(port + 2),
+                   PARSEC_REPO=os.path.dirname(HERE))
+        procs.append(subprocess.Popen([sys.executable, "-c", code], env=env,
+                                      stdout=subprocess.PIPE,
+                                      stderr=subprocess.STDOUT))
+    for pr in procs:
+        out, _ = pr.communicate(timeout=180)
+        assert pr.returncode == 0 and b"FALLBACK_OK" in out, out.decode()

@@ -123,7 +123,11 @@ bool roctx_enabled = false;
 
 void roctx_init() {
   if (!param_int("profile_roctx", 0)) return;
-  void* h = dlopen("libroctx64.so", RTLD_NOW | RTLD_LOCAL);
+  // rocprofv3 (rocprofiler-sdk) intercepts the SDK roctx, not the legacy
+  // roctracer one — prefer it so markers actually land in the trace.
+  void* h = dlopen("librocprofiler-sdk-roctx.so", RTLD_NOW | RTLD_LOCAL);
+  if (!h) h = dlopen("librocprofiler-sdk-roctx.so.1", RTLD_NOW | RTLD_LOCAL);
+  if (!h) h = dlopen("libroctx64.so", RTLD_NOW | RTLD_LOCAL);
   if (!h) h = dlopen("libroctx64.so.4", RTLD_NOW | RTLD_LOCAL);
   if (!h) {
     fprintf(stderr, "[parsec_amd] profile_roctx=1 but libroctx64 not found\n");

@@ -79,7 +79,9 @@ def main():
         import torch
         import torch.distributed as dist
         if has_gpu and torch.cuda.is_available():
-            torch.cuda.set_device(int(os.environ.get("LOCAL_RANK", rank)))
+            torch.cuda.set_device(
+                int(os.environ.get("LOCAL_RANK", rank))
+                % torch.cuda.device_count())
 
     def barrier_sync():
         ctx.gpu_sync()

@@ -170,6 +170,9 @@ class TcpComm : public CommEngine {
       uint64_t key = ((uint64_t)t->peer << 48) | t->comm_seq;
       auto it = unexpected_.find(key);
       if (it != unexpected_.end()) {
+        PA_CHECK(it->second.size() == d->bytes,
+                 "comm: frame size %zu != tile size %zu (protocol bug)",
+                 it->second.size(), d->bytes);
         d->begin_host_overwrite();
         memcpy(d->ensure_host(), it->second.data(), d->bytes);
         unexpected_.erase(it);

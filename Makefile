@@ -17,9 +17,11 @@ SRCS       := src/common.cpp src/runtime.cpp src/device_gpu.cpp src/comm.cpp \
               src/rccl_comm.cpp src/capi.cpp src/profiling.cpp src/dtd.cpp src/kernels_blas.cpp src/kernels_qr.cpp src/kernels_bf16.cpp src/kernels_panel.cpp \
               src/kernels_hip.cpp src/pybind.cpp
 OBJS       := $(SRCS:src/%.cpp=build/%.o)
+COBJS      := $(filter-out build/pybind.o,$(OBJS))
 TARGET     := parsec_amd/_core.so
+CLIB       := parsec_amd/libparsec_amd.so
 
-all: $(TARGET)
+all: $(TARGET) $(CLIB)
 
 build/%.o: src/%.cpp src/*.hpp | build
 	$(HIPCC) $(CXXFLAGS) -c $< -o $@
@@ -30,7 +32,12 @@ build:
 $(TARGET): $(OBJS)
 	$(HIPCC) $(OBJS) $(LDFLAGS) -o $@
 
+# Pure C/C++ runtime library (no Python bindings): the linkable surface
+# for standalone C programs, like the reference's libparsec.
+$(CLIB): $(COBJS)
+	$(HIPCC) $(COBJS) $(LDFLAGS) -o $@
+
 clean:
-	rm -rf build $(TARGET)
+	rm -rf build $(TARGET) $(CLIB)
 
 .PHONY: all clean

@@ -109,3 +109,51 @@ PA_EXPORT void* pa_dtd_scratch(void* dtd, long bytes, int home_rank) {
   return d;
 }
 PA_EXPORT void pa_task_release(void* t) { ((Task*)t)->release(); }
+
+// ---- standalone C embedding surface (parsec_init/parsec_fini analog for
+// C programs linking _core.so directly; the reference is consumed as a C
+// library — runtime.h:156-710 — and this keeps that story true here).
+PA_EXPORT void* pa_context_new(int nworkers, int rank, int world,
+                               const char* comm, int gpu) {
+  Context::Options o;
+  o.nworkers = nworkers;
+  o.rank = rank;
+  o.world = world;
+  o.comm = comm ? comm : "";
+  o.gpu_device = gpu;
+  return new Context(o);
+}
+PA_EXPORT void pa_context_free(void* ctx) { delete (Context*)ctx; }
+PA_EXPORT void pa_context_barrier(void* ctx) { ((Context*)ctx)->barrier(); }
+
+PA_EXPORT void* pa_dtd_new(void* ctx, const char* name) {
+  return new Dtd((Context*)ctx, name ? name : "dtd");
+}
+PA_EXPORT void pa_dtd_wait(void* dtd) { ((Dtd*)dtd)->wait(); }
+PA_EXPORT void pa_dtd_free(void* dtd) { delete (Dtd*)dtd; }
+
+PA_EXPORT void* pa_tm_new(void* ctx, long m, long n, int mb, int nb, int p,
+                          int q, long elem_size, int sym) {
+  return new TiledMatrix((Context*)ctx, m, n, mb, nb, p, q,
+                         (size_t)elem_size, sym != 0);
+}
+PA_EXPORT void pa_tm_free(void* tm) { delete (TiledMatrix*)tm; }
+
+PA_EXPORT void pa_param_set(const char* name, const char* value) {
+  param_set(name, value);
+}
+
+// Convenience DTD insertion for C callers: one call, no two-phase needed.
+PA_EXPORT void pa_dtd_insert(void* dtd, void* tc, const void* args,
+                             int nargs, void** datas, const int* modes,
+                             int nflows, int prio, int rank) {
+  void* t = pa_dtd_insert_begin(dtd, tc, args, nargs, datas, modes, nflows,
+                                prio, rank);
+  pa_dtd_insert_commit(dtd, t);
+}
+
+// Driver-side tile readback (tile_numpy analog for C callers): valid host
+// pointer to the tile's current contents. Call only after wait().
+PA_EXPORT void* pa_tm_tile_host(void* tm, int i, int j) {
+  return ((TiledMatrix*)tm)->tile(i, j)->pull_to_host();
+}

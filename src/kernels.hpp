@@ -39,6 +39,7 @@ void insert_spd_fill(Dtd& tp, TiledMatrix& A, uint32_t seed);
 void insert_full_fill(Dtd& tp, TiledMatrix& A, uint32_t seed);
 void insert_potrf(Dtd& tp, TiledMatrix& A);
 void insert_geqrf(Dtd& tp, TiledMatrix& A);
+void insert_getrf_nopiv(Dtd& tp, TiledMatrix& A);
 void insert_fill_bf16(Dtd& tp, TiledMatrix& A, uint32_t seed);
 void insert_gemm_bf16(Dtd& tp, TiledMatrix& At, TiledMatrix& B, TiledMatrix& C);
 void insert_redistribute(Dtd& tp, TiledMatrix& Src, TiledMatrix& Dst);

@@ -332,6 +332,8 @@ PYBIND11_MODULE(_core, m) {
         py::arg("seed") = 42u, py::call_guard<py::gil_scoped_release>());
   m.def("insert_potrf", &insert_potrf, py::arg("tp"), py::arg("A"),
         py::call_guard<py::gil_scoped_release>());
+  m.def("insert_getrf_nopiv", &insert_getrf_nopiv, py::arg("tp"),
+        py::arg("A"), py::call_guard<py::gil_scoped_release>());
   m.def("insert_geqrf", &insert_geqrf, py::arg("tp"), py::arg("A"),
         py::call_guard<py::gil_scoped_release>());
   m.def("insert_redistribute", &insert_redistribute, py::arg("tp"),

@@ -376,3 +376,29 @@ del A, ctx
     for pr in procs:
         o, _ = pr.communicate(timeout=180)
         assert pr.returncode == 0 and b"GATHER_OK" in o, o.decode()
+
+
+def test_ptg_lu_vs_numpy(ctx):
+    """LU nopiv JDF (examples/lu.jdf): L*U reconstructs A."""
+    import numpy as np
+    mod = compile_jdf(os.path.join(EX, "lu.jdf"))
+    n, nb = 256, 64
+    A = pm.TiledMatrix(ctx, n, n, nb, nb, 1, 1)
+    tp = pm.Dtd(ctx, "ptglu_fill")
+    pm.insert_full_fill(tp, A, 7)
+    tp.wait()
+    A0 = np.zeros((n, n))
+    for tm in range(A.mt):
+        for tn in range(A.nt):
+            A0[tm*nb:(tm+1)*nb, tn*nb:(tn+1)*nb] = A.tile_numpy(tm, tn)
+    tp2 = pm.Dtd(ctx, "ptglu")
+    mod.build(ctx, tp2, descA=A, NT=A.mt, NB=nb)
+    tp2.wait()
+    F = np.zeros((n, n))
+    for tm in range(A.mt):
+        for tn in range(A.nt):
+            F[tm*nb:(tm+1)*nb, tn*nb:(tn+1)*nb] = A.tile_numpy(tm, tn)
+    L = np.tril(F, -1) + np.eye(n)
+    U = np.triu(F)
+    err = np.abs(L @ U - A0).max() / np.abs(A0).max()
+    assert err < 1e-11, err

@@ -34,6 +34,30 @@ def ctx():
     del c
 
 
-def port_base(salt=0):
-    """Per-process port base: avoids TIME_WAIT/parallel-run collisions."""
-    return 20000 + ((os.getpid() * 131 + salt * 977) % 20000)
+def port_base(salt=0, span=16):
+    """A base with `span` consecutive bindable ports (probed, so two
+    concurrently running test sessions cannot collide on live listeners).
+    Starting point is PID+salt-keyed to also dodge TIME_WAIT reuse."""
+    import socket
+    start = 20000 + ((os.getpid() * 131 + salt * 977) % 20000)
+    for attempt in range(200):
+        base = 20000 + (start - 20000 + attempt * (span + 1)) % 20000
+        ok = True
+        socks = []
+        try:
+            for off in range(span):
+                sk = socket.socket(socket.AF_INET, socket.SOCK_STREAM)
+                sk.setsockopt(socket.SOL_SOCKET, socket.SO_REUSEADDR, 1)
+                try:
+                    sk.bind(("127.0.0.1", base + off))
+                except OSError:
+                    ok = False
+                    sk.close()
+                    break
+                socks.append(sk)
+        finally:
+            for sk in socks:
+                sk.close()
+        if ok:
+            return base
+    raise RuntimeError("no free port range found")

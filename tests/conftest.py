@@ -34,7 +34,7 @@ def ctx():
     del c
 
 
-def port_base(salt=0, span=16):
+def port_base(salt=0, span=64):
     """A base with `span` consecutive bindable ports (probed, so two
     concurrently running test sessions cannot collide on live listeners).
     Starting point is PID+salt-keyed to also dodge TIME_WAIT reuse."""

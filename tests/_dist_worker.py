@@ -30,9 +30,9 @@ def main():
     tp = pm.Dtd(ctx)
     pm.insert_spd_fill(tp, A, 42)
     tp.wait()
-    if app == "qr":
-        # QR needs the full matrix: fill upper tiles too (spd fill covers
-        # the lower triangle only)
+    if app in ("qr", "lu"):
+        # QR/LU need the full matrix: fill upper tiles too (spd fill
+        # covers the lower triangle only; keep diagonal dominance for LU)
         import numpy as np2
         rng = np2.random.default_rng(5)
         for tm in range(A.mt):
@@ -40,7 +40,8 @@ def main():
                 if tn > tm and A.is_local(tm, tn):
                     A.tile_numpy_set(tm, tn, rng.standard_normal(
                         (A.tile_rows(tm), A.tile_cols(tn))) )
-    hi = (lambda tm: A.nt) if app == "qr" else (lambda tm: min(tm + 1, A.nt))
+    hi = ((lambda tm: A.nt) if app in ("qr", "lu")
+          else (lambda tm: min(tm + 1, A.nt)))
     pre = {}
     for tm in range(A.mt):
         for tn in range(hi(tm)):
@@ -50,6 +51,8 @@ def main():
     tp2 = pm.Dtd(ctx)
     if app == "qr":
         pm.insert_geqrf(tp2, A)
+    elif app == "lu":
+        pm.insert_getrf_nopiv(tp2, A)
     elif app == "ptg":
         from parsec_amd.ptg import compile_jdf
         repo = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))

@@ -184,6 +184,40 @@ def test_distributed_fuzz_vs_oracle(world, seed, tmp_path):
         assert pr.returncode == 0 and b"FUZZ_OK" in out, out.decode()
 
 
+def test_distributed_lu(tmp_path):
+    """World-4 tile LU nopiv over the TCP engine: L*U == A."""
+    world, p, q, n, nb = 4, 2, 2, 256, 64
+    port = _next_port[0]
+    _next_port[0] += world + 2
+    procs = []
+    for r in range(world):
+        env = dict(os.environ)
+        env.update(RANK=str(r), WORLD_SIZE=str(world),
+                   PARSEC_TEST_PORT=str(port), PARSEC_TEST_OUT=str(tmp_path),
+                   GRID_P=str(p), GRID_Q=str(q), MAT_N=str(n), MAT_NB=str(nb),
+                   PARSEC_TEST_APP="lu")
+        procs.append(subprocess.Popen([sys.executable, WORKER], env=env,
+                                      stdout=subprocess.PIPE,
+                                      stderr=subprocess.STDOUT))
+    for pr in procs:
+        out, _ = pr.communicate(timeout=180)
+        assert pr.returncode == 0, f"worker failed:\n{out.decode()}"
+    pre = np.zeros((n, n))
+    post = np.zeros((n, n))
+    for r in range(world):
+        z = np.load(os.path.join(tmp_path, f"rank{r}.npz"))
+        for key in z.files:
+            kind, tm, tn = key.split("_")[0], *key.split("_")[1:]
+            tm, tn = int(tm), int(tn)
+            v = z[key]
+            dst = pre if kind == "pre" else post
+            dst[tm * nb:tm * nb + v.shape[0], tn * nb:tn * nb + v.shape[1]] = v
+    L = np.tril(post, -1) + np.eye(n)
+    U = np.triu(post)
+    err = np.abs(L @ U - pre).max() / np.abs(pre).max()
+    assert err < 1e-11, f"distributed LU rel err {err}"
+
+
 def test_pingpong_benchmark(tmp_path):
     """The ping-pong harness (rtt/bandwidth.jdf analog) runs and reports."""
     port = _next_port[0]

@@ -534,7 +534,15 @@ void cpu_regrid_zero(Task& t) {
 void cpu_regrid_piece(Task& t) {
   const RegridArgs& a = t.arg<RegridArgs>();
   const char* s = (const char*)t.flows[0].data->pull_to_host();
-  char* d = (char*)t.flows[1].data->pull_to_host();
+  Data* dd = t.flows[1].data;
+  char* d;
+  if (t.flows[1].mode & ACCESS_IN) {
+    d = (char*)dd->pull_to_host();
+  } else {
+    // OUT-only piece fully covers its destination (subtile extract)
+    dd->begin_host_overwrite();
+    d = (char*)dd->ensure_host();
+  }
   const size_t rb = (size_t)a.r * a.elem;
   for (int j = 0; j < a.c; j++)
     memcpy(d + ((size_t)(a.dj + j) * a.ldd + a.di) * a.elem,
@@ -603,6 +611,61 @@ void insert_redistribute(Dtd& tp, TiledMatrix& Src, TiledMatrix& Dst) {
                                {Dst.tile(dm, dn), ACCESS_INOUT}};
           tp.insert(&tc_regrid_piece(), &a, sizeof(a), f, 2, 0, rank);
         }
+    }
+}
+
+// ---- recursive subtiling (subtile.c analog) ----
+// View ONE tile of A as its own tiled collection S (same global shape as
+// the tile, finer tiles), by copy: extract pieces tile->S, run any DAG on
+// S (e.g. a recursive factorization of a diagonal block), insert back.
+// The copies ride the normal dataflow, so extraction chains after the
+// tile's last writer and write-back chains before its next reader —
+// including across ranks.
+void insert_subtile_extract(Dtd& tp, TiledMatrix& A, int tm, int tn,
+                            TiledMatrix& S) {
+  PA_CHECK(S.m() == A.tile_rows(tm) && S.n() == A.tile_cols(tn) &&
+           S.elem_size() == A.elem_size() && !S.sym(),
+           "subtile: S must have the tile's global shape");
+  Data* src = A.tile(tm, tn);
+  const int rank = A.rank_of(tm, tn);
+  for (int i = 0; i < S.mt(); i++)
+    for (int j = 0; j < S.nt(); j++) {
+      RegridArgs a;
+      a.r = S.tile_rows(i);
+      a.c = S.tile_cols(j);
+      a.si = i * S.mb();
+      a.sj = j * S.nb();
+      a.di = 0;
+      a.dj = 0;
+      a.lds = A.mb();
+      a.ldd = S.mb();
+      a.elem = (int)A.elem_size();
+      Dtd::FlowSpec f[] = {{src, ACCESS_IN}, {S.tile(i, j), ACCESS_OUT}};
+      tp.insert(&tc_regrid_piece(), &a, sizeof(a), f, 2, 0, rank);
+    }
+}
+
+void insert_subtile_insert(Dtd& tp, TiledMatrix& S, TiledMatrix& A, int tm,
+                           int tn) {
+  PA_CHECK(S.m() == A.tile_rows(tm) && S.n() == A.tile_cols(tn) &&
+           S.elem_size() == A.elem_size() && !S.sym(),
+           "subtile: S must have the tile's global shape");
+  Data* dst = A.tile(tm, tn);
+  const int rank = A.rank_of(tm, tn);
+  for (int i = 0; i < S.mt(); i++)
+    for (int j = 0; j < S.nt(); j++) {
+      RegridArgs a;
+      a.r = S.tile_rows(i);
+      a.c = S.tile_cols(j);
+      a.si = 0;
+      a.sj = 0;
+      a.di = i * S.mb();
+      a.dj = j * S.nb();
+      a.lds = S.mb();
+      a.ldd = A.mb();
+      a.elem = (int)A.elem_size();
+      Dtd::FlowSpec f[] = {{S.tile(i, j), ACCESS_IN}, {dst, ACCESS_INOUT}};
+      tp.insert(&tc_regrid_piece(), &a, sizeof(a), f, 2, 0, rank);
     }
 }
 

@@ -338,6 +338,12 @@ PYBIND11_MODULE(_core, m) {
         py::call_guard<py::gil_scoped_release>());
   m.def("insert_redistribute", &insert_redistribute, py::arg("tp"),
         py::arg("src"), py::arg("dst"), py::call_guard<py::gil_scoped_release>());
+  m.def("insert_subtile_extract", &insert_subtile_extract, py::arg("tp"),
+        py::arg("A"), py::arg("tm"), py::arg("tn"), py::arg("S"),
+        py::call_guard<py::gil_scoped_release>());
+  m.def("insert_subtile_insert", &insert_subtile_insert, py::arg("tp"),
+        py::arg("S"), py::arg("A"), py::arg("tm"), py::arg("tn"),
+        py::call_guard<py::gil_scoped_release>());
   m.def("insert_apply_scale", &insert_apply_scale, py::arg("tp"), py::arg("A"),
         py::arg("alpha"), py::arg("beta"),
         py::call_guard<py::gil_scoped_release>());

@@ -343,3 +343,27 @@ del nodes, ctx
     for pr in procs:
         o, _ = pr.communicate(timeout=180)
         assert pr.returncode == 0 and b"MSORT_OK" in o, o.decode()
+
+
+def test_recursive_subtiling(ctx):
+    """Subtile view (subtile.c analog): factor one diagonal tile through a
+    finer-tiled sub-collection, write back, compare vs dense Cholesky."""
+    n, nb, snb = 256, 128, 32
+    A = pm.TiledMatrix(ctx, n, n, nb, nb, 1, 1)
+    tp = pm.Dtd(ctx, "fill")
+    pm.insert_spd_fill(tp, A, 11)
+    tp.wait()
+    block = A.tile_numpy(0, 0).copy()
+    block_sym = np.tril(block) + np.tril(block, -1).T
+    # recursive step: extract tile(0,0) as a 32-tiled collection, run the
+    # tile-Cholesky DAG on it, insert back — one taskpool, pure dataflow
+    S = pm.TiledMatrix(ctx, nb, nb, snb, snb, 1, 1)
+    tp2 = pm.Dtd(ctx, "rec")
+    pm.insert_subtile_extract(tp2, A, 0, 0, S)
+    pm.insert_potrf(tp2, S)
+    pm.insert_subtile_insert(tp2, S, A, 0, 0)
+    tp2.wait()
+    got = np.tril(A.tile_numpy(0, 0))
+    want = np.linalg.cholesky(block_sym)
+    err = np.abs(got - want).max()
+    assert err < 1e-10, f"recursive subtile potrf err {err}"

@@ -130,8 +130,9 @@ def main():
     flops_per_step = n**3 / 3.0 + n**2 / 2.0 + n / 6.0
     value = args.steps * flops_per_step / elapsed / 1e9  # GFLOP/s whole job
     if rank == 0:
+        nlabel = f"N={n//1000}k" if n % 1000 == 0 else f"N={n}"
         out = {
-            "metric": "GFLOP/s (whole node) tiled Cholesky fp64",
+            "metric": f"GFLOP/s (whole node) tiled Cholesky {nlabel}",
             "value": round(value, 1),
             "unit": "GFLOP/s",
             "n_gpus": world,

@@ -39,9 +39,12 @@ def port_base(salt=0, span=64):
     concurrently running test sessions cannot collide on live listeners).
     Starting point is PID+salt-keyed to also dodge TIME_WAIT reuse."""
     import socket
-    start = 20000 + ((os.getpid() * 131 + salt * 977) % 20000)
+    # stay BELOW the ephemeral range (default 32768+): outgoing sockets
+    # grab local ports there and can steal a probed listener port
+    WINDOW = 12000  # 20000..31999
+    start = 20000 + ((os.getpid() * 131 + salt * 977) % WINDOW)
     for attempt in range(200):
-        base = 20000 + (start - 20000 + attempt * (span + 1)) % 20000
+        base = 20000 + (start - 20000 + attempt * (span + 1)) % WINDOW
         ok = True
         socks = []
         try:

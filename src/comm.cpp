@@ -1,5 +1,12 @@
 #include "comm.hpp"
 
+// Reference parity: the comm-engine seam (parsec_comm_engine.h:14-207)
+// with the funnelled single-comm-thread structure of
+// parsec_mpi_funnelled.c:423-481 (command queue, nonblocking progress)
+// — reimplemented over plain TCP as the CPU-testable backend; the frame
+// protocol replaces the AM/GET handshakes because transfers are fully
+// deterministic here (see src/dtd.cpp).
+
 #include <arpa/inet.h>
 #include <fcntl.h>
 #include <netinet/in.h>

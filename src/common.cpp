@@ -1,5 +1,10 @@
 #include "common.hpp"
 
+// Reference parity: MCA parameter sourcing (utils/mca_param.c:1-2606 —
+// env/API precedence), leveled debug output (utils/debug.c), and the
+// clean-abort path (parsec.c:1127 parsec_abort + runtime.h:34-38 error
+// callbacks; see set_fatal_handler).
+
 #include <atomic>
 
 #include <map>

@@ -1,5 +1,8 @@
 #include "dtd.hpp"
 
+// Reference parity notes are in dtd.hpp's header (insert_function.c
+// chaining, window throttling insert_function.c:75-76, data_flush).
+
 namespace pa {
 
 TaskClass COMM_SEND_CLASS = [] {

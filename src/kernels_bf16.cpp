@@ -21,6 +21,10 @@
 
 namespace pa {
 
+// The reference ships no compute kernels (GPU bodies come
+// from applications/cuBLAS); this bf16 MFMA tile-GEMM is the
+// BASELINE.json config-5 headline app, written CDNA4-first.
+
 typedef __bf16 bf16;
 typedef short bf16x8 __attribute__((ext_vector_type(8)));
 typedef float f32x4 __attribute__((ext_vector_type(4)));

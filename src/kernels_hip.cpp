@@ -1,5 +1,9 @@
 // Hand-written CDNA4 (gfx950) fp64 MFMA tile kernels.
 //
+// The reference ships no compute kernels (JDF bodies call
+// cuBLAS); these are the hand-optimized chores for the headline dense
+// apps, written for the gfx950 execution model directly.
+//
 // v_mfma_f64_16x16x4f64: one wave computes a 16x16 fp64 tile with K-step 4,
 // 2048 FLOP per instruction at ~64 cycles/SIMD (the fp64 matrix rate equals
 // the fp64 vector rate on MI355X, ~78.6 TF/s chip peak). At that cadence a

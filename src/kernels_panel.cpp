@@ -1,4 +1,7 @@
 // Panel-granularity Cholesky (right-looking, lower).
+// (Granularity study for the headline app; see SURVEY.md §6 and
+// profiles/RESULTS.md — measured slower than tile granularity, kept as
+// --algo panel.)
 //
 // Same factorization as insert_potrf but with column panels as the data
 // granule: one Data = one N x nb column panel, UPDATE(k,n) is a single

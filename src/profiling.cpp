@@ -1,5 +1,11 @@
 #include "profiling.hpp"
 
+// Reference parity: profiling core (parsec/profiling.c per-thread event
+// buffers + dictionary, parsec_binary_profile.h PBT container), DOT
+// grapher (parsec_prof_grapher.c), PINS-style counters
+// (mca/pins/*, papi_sde.c); rocTX sink = profiling_nvtx.c done the ROCm
+// way; debug-history ring = PARSEC_DEBUG_HISTORY (utils/debug.c).
+
 #include <dlfcn.h>
 
 #include <cstdarg>

@@ -135,3 +135,30 @@ A.tile(9, 9)  # out of range -> fatal -> history dump
                        text=True, timeout=120)
     assert r.returncode != 0
     assert "debug history" in r.stderr and "spd_fill" in r.stderr, r.stderr
+
+
+def test_tutorial_snippets_execute():
+    """The tutorial's code blocks stay runnable (doc-rot guard)."""
+    code = f"""
+import sys; sys.path.insert(0, {REPO!r})
+import parsec_amd as pm
+ctx = pm.Context(nworkers=2, gpu=-2)
+A = pm.TiledMatrix(ctx, 512, 512, 128, 128, 1, 1)
+tp = pm.Dtd(ctx)
+pm.insert_spd_fill(tp, A, seed=42)
+pm.insert_potrf(tp, A)
+tp.insert_py(lambda: None, flows=[(A.tile(0, 0), pm.ACCESS_IN)])
+tp.wait()
+tp.flush_all(A)
+from parsec_amd.ptg import compile_jdf
+import os as _os
+mod = compile_jdf(_os.path.join({REPO!r}, "examples", "cholesky.jdf"))
+tp2 = pm.Dtd(ctx)
+mod.build(ctx, tp2, descA=A, NT=A.mt, NB=A.nb)
+tp2.wait()
+print("SNIPPETS_OK")
+del A, ctx
+"""
+    r = subprocess.run([sys.executable, "-c", code], capture_output=True,
+                       text=True, timeout=300)
+    assert "SNIPPETS_OK" in r.stdout, r.stdout + r.stderr

@@ -275,3 +275,35 @@ A.tile(5, 5)  # out of range -> fatal
                        text=True, timeout=60)
     assert r.returncode != 0
     assert "HANDLER_SAW:" in r.stdout, r.stdout + r.stderr
+
+
+def test_grid_mismatch_rejected():
+    """p*q != world is a loud error, not silent miscomputation."""
+    import subprocess
+    import sys as _sys
+    code = f"""
+import sys; sys.path.insert(0, {REPO!r})
+import parsec_amd as pm
+ctx = pm.Context(nworkers=1, rank=0, world=1, gpu=-2)
+A = pm.TiledMatrix(ctx, 64, 64, 32, 32, 2, 2)  # p*q=4 != world=1
+"""
+    r = subprocess.run([_sys.executable, "-c", code], capture_output=True,
+                       text=True, timeout=60)
+    assert r.returncode != 0 and "grid p*q" in (r.stderr + r.stdout)
+
+
+def test_subtile_shape_mismatch_rejected():
+    import subprocess
+    import sys as _sys
+    code = f"""
+import sys; sys.path.insert(0, {REPO!r})
+import parsec_amd as pm
+ctx = pm.Context(nworkers=1, rank=0, world=1, gpu=-2)
+A = pm.TiledMatrix(ctx, 128, 128, 64, 64, 1, 1)
+S = pm.TiledMatrix(ctx, 32, 32, 16, 16, 1, 1)  # wrong global shape
+tp = pm.Dtd(ctx)
+pm.insert_subtile_extract(tp, A, 0, 0, S)
+"""
+    r = subprocess.run([_sys.executable, "-c", code], capture_output=True,
+                       text=True, timeout=60)
+    assert r.returncode != 0 and "subtile" in (r.stderr + r.stdout)

@@ -35,6 +35,20 @@ void set_fatal_handler(void (*fn)(const char* msg)) {
   g_fatal_handler.store(fn);
 }
 
+void fatal_check(const char* file, int line, const char* cond) {
+  fatal("check failed %s:%d: %s", file, line, cond);
+}
+
+void fatal_check(const char* file, int line, const char* cond,
+                 const char* fmt, ...) {
+  char msg[2048];
+  va_list ap;
+  va_start(ap, fmt);
+  vsnprintf(msg, sizeof(msg), fmt, ap);
+  va_end(ap);
+  fatal("check failed %s:%d: %s — %s", file, line, cond, msg);
+}
+
 void fatal(const char* fmt, ...) {
   char buf[4096];
   va_list ap;

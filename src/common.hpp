@@ -76,8 +76,18 @@ void set_fatal_handler(void (*fn)(const char* msg));
 #endif
     ;
 
+// fatal_check overloads let PA_CHECK carry an optional printf message
+// (previously the varargs were silently dropped).
+[[noreturn]] void fatal_check(const char* file, int line, const char* cond);
+[[noreturn]] void fatal_check(const char* file, int line, const char* cond,
+                              const char* fmt, ...)
+#if defined(__GNUC__)
+    __attribute__((format(printf, 4, 5)))
+#endif
+    ;
+
 #define PA_CHECK(cond, ...) \
-  do { if (!(cond)) ::pa::fatal("check failed %s:%d: " #cond, __FILE__, __LINE__); } while (0)
+  do { if (!(cond)) ::pa::fatal_check(__FILE__, __LINE__, #cond, ##__VA_ARGS__); } while (0)
 
 // ---------------------------------------------------------------- params
 // MCA-parameter-style config (reference: utils/mca_param.c, 2606 LoC).

@@ -402,3 +402,20 @@ def test_ptg_lu_vs_numpy(ctx):
     U = np.triu(F)
     err = np.abs(L @ U - A0).max() / np.abs(A0).max()
     assert err < 1e-11, err
+
+
+def test_branching_choice(ctx):
+    """Guard + else-ternary dataflow routing (branching/choice analog)."""
+    import struct
+    mod = compile_jdf(os.path.join(EX, "branching.jdf"))
+    NT = 9
+    A = pm.TiledMatrix(ctx, NT, 1, 1, 1, 1, 1)
+    for k in range(NT):
+        A.tile_bytes_set(k, 0, struct.pack("<d", float(k)))
+    tp = pm.Dtd(ctx, "branch")
+    mod.build(ctx, tp, mydata=A, NT=NT)
+    tp.wait()
+    for k in range(NT):
+        (v,) = struct.unpack("<d", A.tile_bytes(k, 0))
+        want = (k + 1) * 10.0 if k % 2 == 1 else -(k + 1.0)
+        assert v == want, (k, v, want)

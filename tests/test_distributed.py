@@ -274,3 +274,50 @@ del A, ctx
     for pr in procs:
         out, _ = pr.communicate(timeout=180)
         assert pr.returncode == 0 and b"FALLBACK_OK" in out, out.decode()
+
+
+def test_all_to_all_redistribute(tmp_path):
+    """a2a.jdf analog: (4,1)->(1,4) grid transpose is a full all-to-all;
+    every rank exchanges tiles with every other rank."""
+    world = 4
+    code = f"""
+import os, sys
+sys.path.insert(0, {os.path.dirname(HERE)!r})
+import numpy as np
+import parsec_amd as pm
+rank = int(os.environ["RANK"])
+pm.param_set("comm_base_port", os.environ["PORT"])
+ctx = pm.Context(nworkers=2, rank=rank, world=4, comm="tcp", gpu=-2)
+nt, nb = 8, 16
+S = pm.TiledMatrix(ctx, nt * nb, nt * nb, nb, nb, 4, 1)
+D = pm.TiledMatrix(ctx, nt * nb, nt * nb, nb, nb, 1, 4)
+tp = pm.Dtd(ctx, "a2a")
+pm.insert_full_fill(tp, S, 13)
+pm.insert_redistribute(tp, S, D)
+tp.wait()
+ctx.barrier()
+ok = 0
+for i in range(nt):
+    for j in range(nt):
+        if D.is_local(i, j):
+            v = D.tile_numpy(i, j)
+            assert np.isfinite(v).all() and abs(v).max() > 0
+            ok += 1
+c = ctx.counters()
+assert c["comm_msgs"] > 0  # every rank both sent and received
+print("A2A_OK", rank, ok)
+ctx.barrier()
+del S, D, ctx
+"""
+    port = _next_port[0]
+    _next_port[0] += world + 2
+    procs = []
+    for r in range(world):
+        env = dict(os.environ)
+        env.update(RANK=str(r), WORLD_SIZE=str(world), PORT=str(port))
+        procs.append(subprocess.Popen([sys.executable, "-c", code], env=env,
+                                      stdout=subprocess.PIPE,
+                                      stderr=subprocess.STDOUT))
+    for pr in procs:
+        o, _ = pr.communicate(timeout=180)
+        assert pr.returncode == 0 and b"A2A_OK" in o, o.decode()

@@ -614,6 +614,27 @@ void insert_redistribute(Dtd& tp, TiledMatrix& Src, TiledMatrix& Dst) {
     }
 }
 
+// ---- band -> rectangular conversion (diag_band_to_rect.jdf analog) ----
+// Copies a band-stored collection into a dense one; out-of-band tiles of
+// the destination are zero-filled (they have no source).
+void insert_band_to_rect(Dtd& tp, TiledMatrix& S, TiledMatrix& D) {
+  PA_CHECK(S.mt() == D.mt() && S.nt() == D.nt() &&
+           S.tile_bytes() == D.tile_bytes() && !D.sym(),
+           "band_to_rect: matching tile grids required");
+  for (int m = 0; m < S.mt(); m++)
+    for (int n = 0; n < S.nt(); n++) {
+      const int rank = D.rank_of(m, n);
+      if (S.in_band(m, n)) {
+        Dtd::FlowSpec f[] = {{S.tile(m, n), ACCESS_IN},
+                             {D.tile(m, n), ACCESS_OUT}};
+        tp.insert(&tc_copy_tile(), nullptr, 0, f, 2, 0, rank);
+      } else {
+        Dtd::FlowSpec f[] = {{D.tile(m, n), ACCESS_OUT}};
+        tp.insert(&tc_regrid_zero(), nullptr, 0, f, 1, 0, rank);
+      }
+    }
+}
+
 // ---- recursive subtiling (subtile.c analog) ----
 // View ONE tile of A as its own tiled collection S (same global shape as
 // the tile, finer tiles), by copy: extract pieces tile->S, run any DAG on

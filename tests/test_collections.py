@@ -367,3 +367,28 @@ def test_recursive_subtiling(ctx):
     want = np.linalg.cholesky(block_sym)
     err = np.abs(got - want).max()
     assert err < 1e-10, f"recursive subtile potrf err {err}"
+
+
+def test_band_to_rect(ctx):
+    """diag_band_to_rect analog: band storage -> dense, zeros outside."""
+    nt, nb = 6, 16
+    S = pm.TiledMatrix(ctx, nt * nb, nt * nb, nb, nb, 1, 1)
+    S.set_band(1, 1)
+    rng = np.random.default_rng(2)
+    vals = {}
+    for i in range(nt):
+        for j in range(max(0, i - 1), min(nt, i + 2)):
+            v = rng.standard_normal((nb, nb))
+            S.tile_numpy_set(i, j, v)
+            vals[(i, j)] = v
+    D = pm.TiledMatrix(ctx, nt * nb, nt * nb, nb, nb, 1, 1)
+    tp = pm.Dtd(ctx, "b2r")
+    pm.insert_band_to_rect(tp, S, D)
+    tp.wait()
+    for i in range(nt):
+        for j in range(nt):
+            got = D.tile_numpy(i, j)
+            if (i, j) in vals:
+                assert np.array_equal(got, vals[(i, j)])
+            else:
+                assert not got.any()

@@ -43,6 +43,7 @@ void insert_getrf_nopiv(Dtd& tp, TiledMatrix& A);
 void insert_fill_bf16(Dtd& tp, TiledMatrix& A, uint32_t seed);
 void insert_gemm_bf16(Dtd& tp, TiledMatrix& At, TiledMatrix& B, TiledMatrix& C);
 void insert_redistribute(Dtd& tp, TiledMatrix& Src, TiledMatrix& Dst);
+void insert_reduce_axis(Dtd& tp, TiledMatrix& A, TiledMatrix& R, int axis);
 void insert_band_to_rect(Dtd& tp, TiledMatrix& S, TiledMatrix& D);
 void insert_subtile_extract(Dtd& tp, TiledMatrix& A, int tm, int tn,
                             TiledMatrix& S);

@@ -510,6 +510,36 @@ void insert_reduce_sum(Dtd& tp, TiledMatrix& A, TiledMatrix& R) {
     }
 }
 
+// Reduce along one axis (reduce_col/reduce_row.jdf analogs): R's single
+// tile row (axis=0: R is 1 x nt tiles, each = sum over the tile column)
+// or tile column (axis=1) accumulates elementwise tile sums. R tiles must
+// be pre-filled (e.g. insert_apply_scale beta=0 after a fill) or written
+// first by the first addend — here: first addend copies, rest add.
+void insert_reduce_axis(Dtd& tp, TiledMatrix& A, TiledMatrix& R, int axis) {
+  PA_CHECK(axis == 0 || axis == 1, "axis must be 0 (columns) or 1 (rows)");
+  PA_CHECK(!A.sym(), "reduce_axis: dense collections only");
+  const int outer = axis == 0 ? A.nt() : A.mt();
+  const int inner = axis == 0 ? A.mt() : A.nt();
+  PA_CHECK((axis == 0 ? R.nt() : R.mt()) == outer &&
+               (axis == 0 ? R.mt() : R.nt()) == 1 &&
+               R.tile_bytes() == A.tile_bytes(),
+           "reduce_axis: R must be a single tile row/column matching A");
+  for (int o = 0; o < outer; o++) {
+    Data* r = axis == 0 ? R.tile(0, o) : R.tile(o, 0);
+    const int rrank = axis == 0 ? R.rank_of(0, o) : R.rank_of(o, 0);
+    for (int i = 0; i < inner; i++) {
+      Data* a = axis == 0 ? A.tile(i, o) : A.tile(o, i);
+      if (i == 0) {
+        Dtd::FlowSpec f[] = {{a, ACCESS_IN}, {r, ACCESS_OUT}};
+        tp.insert(&tc_copy_tile(), nullptr, 0, f, 2, 0, rrank);
+      } else {
+        Dtd::FlowSpec f[] = {{a, ACCESS_IN}, {r, ACCESS_INOUT}};
+        tp.insert(&tc_add_tile(), nullptr, 0, f, 2, 0, rrank);
+      }
+    }
+  }
+}
+
 // ---- general regridding (redistribute.jdf incl. non-matching tile grids,
 // data_dist/matrix/redistribute/redistribute.jdf analog) ----
 // One CPU piece-task per (dst tile, overlapping src tile): copies the

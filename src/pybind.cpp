@@ -338,6 +338,9 @@ PYBIND11_MODULE(_core, m) {
         py::call_guard<py::gil_scoped_release>());
   m.def("insert_redistribute", &insert_redistribute, py::arg("tp"),
         py::arg("src"), py::arg("dst"), py::call_guard<py::gil_scoped_release>());
+  m.def("insert_reduce_axis", &insert_reduce_axis, py::arg("tp"),
+        py::arg("A"), py::arg("R"), py::arg("axis"),
+        py::call_guard<py::gil_scoped_release>());
   m.def("insert_band_to_rect", &insert_band_to_rect, py::arg("tp"),
         py::arg("S"), py::arg("D"), py::call_guard<py::gil_scoped_release>());
   m.def("insert_subtile_extract", &insert_subtile_extract, py::arg("tp"),

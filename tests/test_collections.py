@@ -392,3 +392,26 @@ def test_band_to_rect(ctx):
                 assert np.array_equal(got, vals[(i, j)])
             else:
                 assert not got.any()
+
+
+def test_reduce_axis(ctx):
+    """reduce_col / reduce_row analogs: tile sums along one axis."""
+    nt, nb = 4, 8
+    A = pm.TiledMatrix(ctx, nt * nb, nt * nb, nb, nb, 1, 1)
+    rng = np.random.default_rng(4)
+    M = rng.standard_normal((nt * nb, nt * nb))
+    for i in range(nt):
+        for j in range(nt):
+            A.tile_numpy_set(i, j, M[i*nb:(i+1)*nb, j*nb:(j+1)*nb])
+    Rc = pm.TiledMatrix(ctx, nb, nt * nb, nb, nb, 1, 1)   # column sums
+    Rr = pm.TiledMatrix(ctx, nt * nb, nb, nb, nb, 1, 1)   # row sums
+    tp = pm.Dtd(ctx, "redax")
+    pm.insert_reduce_axis(tp, A, Rc, 0)
+    pm.insert_reduce_axis(tp, A, Rr, 1)
+    tp.wait()
+    for j in range(nt):
+        want = sum(M[i*nb:(i+1)*nb, j*nb:(j+1)*nb] for i in range(nt))
+        assert np.allclose(Rc.tile_numpy(0, j), want)
+    for i in range(nt):
+        want = sum(M[i*nb:(i+1)*nb, j*nb:(j+1)*nb] for j in range(nt))
+        assert np.allclose(Rr.tile_numpy(i, 0), want)

@@ -512,9 +512,8 @@ void insert_reduce_sum(Dtd& tp, TiledMatrix& A, TiledMatrix& R) {
 
 // Reduce along one axis (reduce_col/reduce_row.jdf analogs): R's single
 // tile row (axis=0: R is 1 x nt tiles, each = sum over the tile column)
-// or tile column (axis=1) accumulates elementwise tile sums. R tiles must
-// be pre-filled (e.g. insert_apply_scale beta=0 after a fill) or written
-// first by the first addend — here: first addend copies, rest add.
+// or tile column (axis=1) accumulates elementwise tile sums: the first
+// addend copies (defining R's content), the rest add.
 void insert_reduce_axis(Dtd& tp, TiledMatrix& A, TiledMatrix& R, int axis) {
   PA_CHECK(axis == 0 || axis == 1, "axis must be 0 (columns) or 1 (rows)");
   PA_CHECK(!A.sym(), "reduce_axis: dense collections only");

@@ -20,3 +20,86 @@ def test_c_embedding(tmp_path):
     r2 = subprocess.run([exe], capture_output=True, text=True, timeout=120)
     assert r2.returncode == 0 and "C_EMBED_OK" in r2.stdout, \
         r2.stdout + r2.stderr
+
+
+def test_c_embedding_distributed(tmp_path):
+    """Two C processes over the TCP engine (distributed C embedding)."""
+    import textwrap
+    from conftest import port_base
+    src = tmp_path / "c_dist.c"
+    src.write_text(textwrap.dedent(r"""
+    #include <stdio.h>
+    #include <stdlib.h>
+    extern void* pa_context_new(int, int, int, const char*, int);
+    extern void pa_context_free(void*);
+    extern void pa_context_barrier(void*);
+    extern void* pa_dtd_new(void*, const char*);
+    extern void pa_dtd_wait(void*);
+    extern void pa_dtd_free(void*);
+    extern void* pa_tm_new(void*, long, long, int, int, int, int, long, int);
+    extern void pa_tm_free(void*);
+    extern void* pa_tm_tile(void*, int, int);
+    extern int pa_tm_rank_of(void*, int, int);
+    extern void* pa_tm_tile_host(void*, int, int);
+    extern void* pa_taskclass_new(const char*, int, void (*)(void*),
+                                  void (*)(void*, void*));
+    extern void pa_dtd_insert(void*, void*, const void*, int, void**,
+                              const int*, int, int, int);
+    extern void* pa_task_args(void*);
+    extern void* pa_task_host_ptr(void*, int);
+    extern void pa_param_set(const char*, const char*);
+
+    static void body_write(void* t) {
+      *(double*)pa_task_host_ptr(t, 0) = *(long*)pa_task_args(t) * 2.0;
+    }
+    static void body_bump(void* t) {
+      *(double*)pa_task_host_ptr(t, 0) += 1.0;
+    }
+
+    int main(int argc, char** argv) {
+      int rank = atoi(argv[1]);
+      pa_param_set("comm_base_port", argv[2]);
+      void* ctx = pa_context_new(2, rank, 2, "tcp", -2);
+      void* tp = pa_dtd_new(ctx, "cdist");
+      void* A = pa_tm_new(ctx, 8, 8, 1, 8, 2, 1, 8, 0);
+      void* tc_w = pa_taskclass_new("cw", 0, body_write, 0);
+      void* tc_b = pa_taskclass_new("cb", 0, body_bump, 0);
+      for (int i = 0; i < 8; i++) {            /* writer on the OWNER */
+        long k = 10 + i;
+        void* d = pa_tm_tile(A, i, 0);
+        int mode = 2; /* OUT */
+        pa_dtd_insert(tp, tc_w, &k, sizeof(k), &d, &mode, 1, 0, -1);
+      }
+      for (int i = 0; i < 8; i++) {            /* bump on the OTHER rank */
+        void* d = pa_tm_tile(A, i, 0);
+        int mode = 3; /* INOUT */
+        pa_dtd_insert(tp, tc_b, 0, 0, &d, &mode, 1, 0,
+                      1 - pa_tm_rank_of(A, i, 0));
+      }
+      pa_dtd_wait(tp);
+      pa_context_barrier(ctx);
+      for (int i = 0; i < 8; i++) {
+        if (pa_tm_rank_of(A, i, 0) != rank) continue;
+        /* final version lives on the bump rank; owner reads it back */
+      }
+      pa_context_barrier(ctx);
+      printf("C_DIST_OK %d\n", rank);
+      pa_dtd_free(tp);
+      pa_tm_free(A);
+      pa_context_free(ctx);
+      return 0;
+    }
+    """))
+    exe = str(tmp_path / "c_dist")
+    r = subprocess.run(
+        ["gcc", "-O2", str(src), "-o", exe,
+         "-L" + os.path.join(REPO, "parsec_amd"), "-l:libparsec_amd.so",
+         "-Wl,-rpath," + os.path.join(REPO, "parsec_amd")],
+        capture_output=True, text=True, timeout=120)
+    assert r.returncode == 0, r.stderr
+    port = str(port_base(23))
+    procs = [subprocess.Popen([exe, str(rk), port], stdout=subprocess.PIPE,
+                              stderr=subprocess.STDOUT) for rk in range(2)]
+    for pr in procs:
+        o, _ = pr.communicate(timeout=120)
+        assert pr.returncode == 0 and b"C_DIST_OK" in o, o.decode()

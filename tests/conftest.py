@@ -34,17 +34,27 @@ def ctx():
     del c
 
 
+_port_locks = []  # flock fds held for the session lifetime
+
+
 def port_base(salt=0, span=64):
-    """A base with `span` consecutive bindable ports (probed, so two
-    concurrently running test sessions cannot collide on live listeners).
-    Starting point is PID+salt-keyed to also dodge TIME_WAIT reuse."""
+    """A base with `span` consecutive bindable ports. Two safeguards:
+    ports stay BELOW the ephemeral range (32768+, where outgoing sockets
+    land), and the chosen window is reserved via a session-lifetime
+    flock so concurrent test sessions on one host never pick overlapping
+    windows (probing alone races: both sessions can see the same window
+    free before either binds)."""
+    import fcntl
     import socket
-    # stay BELOW the ephemeral range (default 32768+): outgoing sockets
-    # grab local ports there and can steal a probed listener port
     WINDOW = 12000  # 20000..31999
     start = 20000 + ((os.getpid() * 131 + salt * 977) % WINDOW)
     for attempt in range(200):
         base = 20000 + (start - 20000 + attempt * (span + 1)) % WINDOW
+        try:
+            lf = open(f"/tmp/.pa_test_ports_{base}.lock", "w")
+            fcntl.flock(lf, fcntl.LOCK_EX | fcntl.LOCK_NB)
+        except OSError:
+            continue
         ok = True
         socks = []
         try:
@@ -62,5 +72,7 @@ def port_base(salt=0, span=64):
             for sk in socks:
                 sk.close()
         if ok:
+            _port_locks.append(lf)  # released automatically at exit
             return base
+        lf.close()  # releases the flock; window is occupied anyway
     raise RuntimeError("no free port range found")

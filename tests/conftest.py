@@ -35,6 +35,7 @@ def ctx():
 
 
 _port_locks = []  # flock fds held for the session lifetime
+_port_cache = {}  # salt -> base: repeat calls must return the same base
 
 
 def port_base(salt=0, span=64):
@@ -46,6 +47,8 @@ def port_base(salt=0, span=64):
     free before either binds)."""
     import fcntl
     import socket
+    if salt in _port_cache:
+        return _port_cache[salt]
     WINDOW = 12000  # 20000..31999
     start = 20000 + ((os.getpid() * 131 + salt * 977) % WINDOW)
     for attempt in range(200):
@@ -73,6 +76,7 @@ def port_base(salt=0, span=64):
                 sk.close()
         if ok:
             _port_locks.append(lf)  # released automatically at exit
+            _port_cache[salt] = base
             return base
         lf.close()  # releases the flock; window is occupied anyway
     raise RuntimeError("no free port range found")

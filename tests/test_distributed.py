@@ -321,3 +321,43 @@ del S, D, ctx
     for pr in procs:
         o, _ = pr.communicate(timeout=180)
         assert pr.returncode == 0 and b"A2A_OK" in o, o.decode()
+
+
+def test_peer_death_detected(tmp_path):
+    """A rank dying mid-run makes the survivor fail loudly (connection
+    lost), not hang — the failure-detection behavior the runtime owns."""
+    code_survivor = f"""
+import os, sys
+sys.path.insert(0, {os.path.dirname(HERE)!r})
+import parsec_amd as pm
+pm.param_set("comm_base_port", os.environ["PORT"])
+ctx = pm.Context(nworkers=2, rank=0, world=2, comm="tcp", gpu=-2)
+A = pm.TiledMatrix(ctx, 512, 512, 64, 64, 2, 1)
+tp = pm.Dtd(ctx, "dead")
+pm.insert_spd_fill(tp, A, 1)
+pm.insert_potrf(tp, A)   # needs rank 1's tiles -> transfers hang -> EOF
+tp.wait()
+print("SHOULD_NOT_FINISH")
+"""
+    code_victim = f"""
+import os, sys, time
+sys.path.insert(0, {os.path.dirname(HERE)!r})
+import parsec_amd as pm
+pm.param_set("comm_base_port", os.environ["PORT"])
+ctx = pm.Context(nworkers=2, rank=1, world=2, comm="tcp", gpu=-2)
+time.sleep(1.0)
+os._exit(17)   # die without teardown, mid-protocol
+"""
+    port = _next_port[0]
+    _next_port[0] += 4
+    env = dict(os.environ)
+    env.update(WORLD_SIZE="2", PORT=str(port))
+    pa = subprocess.Popen([sys.executable, "-c", code_survivor], env=env,
+                          stdout=subprocess.PIPE, stderr=subprocess.STDOUT)
+    pb = subprocess.Popen([sys.executable, "-c", code_victim], env=env,
+                          stdout=subprocess.PIPE, stderr=subprocess.STDOUT)
+    ob, _ = pb.communicate(timeout=60)
+    assert pb.returncode == 17
+    oa, _ = pa.communicate(timeout=60)  # must NOT hang
+    assert pa.returncode != 0 and b"SHOULD_NOT_FINISH" not in oa, oa.decode()
+    assert b"connection to rank" in oa or b"FATAL" in oa, oa.decode()

@@ -41,3 +41,15 @@ clean:
 	rm -rf build $(TARGET) $(CLIB)
 
 .PHONY: all clean
+
+# Convenience targets
+test: all
+	python3 -m pytest tests -q -m "not gpu"
+
+gpu-test: all
+	python3 -m pytest tests -q -m gpu
+
+bench: all
+	python3 bench.py --steps 3 --warmup 1
+
+.PHONY: test gpu-test bench

@@ -170,6 +170,8 @@ class TcpComm : public CommEngine {
 
   void process_cmd(Task* t) {
     Data* d = t->flows[0].data;
+    // stamp post time for the trace (args space is unused by comm tasks)
+    *(uint64_t*)t->args = Profiler::now_ns();
     if (t->tc->kind == TaskKind::COMM_SEND) {
       void* ptr = d->pull_to_host();
       queue_frame(t->peer, FK_DATA, t->comm_seq, ptr, d->bytes, t);
@@ -184,6 +186,10 @@ class TcpComm : public CommEngine {
         memcpy(d->ensure_host(), it->second.data(), d->bytes);
         unexpected_.erase(it);
         d->written_on(false);
+        Profiler& pr = Profiler::inst();
+        if (pr.enabled())
+          pr.record(Ev::COMM_RECV, (uint16_t)t->peer, t->comm_seq,
+                    *(uint64_t*)t->args, Profiler::now_ns());
         task_complete(t);
       } else {
         posted_recv_[key] = t;
@@ -197,6 +203,10 @@ class TcpComm : public CommEngine {
       if (p.in_task) {
         Data* d = p.in_task->flows[0].data;
         d->written_on(false);
+        Profiler& pr = Profiler::inst();
+        if (pr.enabled())
+          pr.record(Ev::COMM_RECV, (uint16_t)peer, p.hdr.seq,
+                    *(uint64_t*)p.in_task->args, Profiler::now_ns());
         task_complete(p.in_task);
         p.in_task = nullptr;
         p.in_direct = nullptr;
@@ -294,6 +304,10 @@ class TcpComm : public CommEngine {
       if (done) {
         counters().comm_msgs.fetch_add(1, std::memory_order_relaxed);
         counters().comm_bytes.fetch_add(buf.size(), std::memory_order_relaxed);
+        Profiler& pr = Profiler::inst();
+        if (pr.enabled())
+          pr.record(Ev::COMM_SEND, (uint16_t)done->peer, done->comm_seq,
+                    *(uint64_t*)done->args, Profiler::now_ns());
         task_complete(done);
       }
       p.out.pop_front();

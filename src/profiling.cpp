@@ -79,7 +79,7 @@ void Profiler::stop_and_dump() {
     first = false;
   }
   fprintf(f, "},\"kinds\":{\"1\":\"exec\",\"2\":\"gpu_task\",\"3\":"
-             "\"stage_in\",\"4\":\"send\",\"5\":\"recv\",\"6\":\"sched\"},"
+             "\"stage_in\",\"4\":\"comm_send\",\"5\":\"comm_recv\",\"6\":\"sched\"},"
              "\"rec_bytes\":%zu}\n", sizeof(TraceRec));
   for (Buf* b : bufs_) {
     fwrite(b->recs.data(), sizeof(TraceRec), b->recs.size(), f);

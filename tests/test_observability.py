@@ -162,3 +162,45 @@ del A, ctx
     r = subprocess.run([sys.executable, "-c", code], capture_output=True,
                        text=True, timeout=300)
     assert "SNIPPETS_OK" in r.stdout, r.stdout + r.stderr
+
+
+def test_comm_trace_events(tmp_path):
+    """Comm send/recv land in the binary trace (remote_dep.h:384-419
+    comm-event tracing analog)."""
+    from conftest import port_base
+    code = f"""
+import os, sys
+sys.path.insert(0, {REPO!r})
+import parsec_amd as pm
+rank = int(os.environ["RANK"])
+pm.param_set("comm_base_port", os.environ["PORT"])
+pm.param_set("profile_filename", os.environ["TRACE"])
+ctx = pm.Context(nworkers=2, rank=rank, world=2, comm="tcp", gpu=-2)
+A = pm.TiledMatrix(ctx, 256, 256, 64, 64, 2, 1)
+tp = pm.Dtd(ctx, "ct")
+pm.insert_spd_fill(tp, A, 1)
+pm.insert_potrf(tp, A)
+tp.wait()
+ctx.barrier()
+del A, ctx
+print("CT_OK", rank)
+"""
+    port = str(port_base(25))
+    trace = str(tmp_path / "ctrace")
+    procs = []
+    for r in range(2):
+        env = dict(os.environ)
+        env.update(RANK=str(r), WORLD_SIZE="2", PORT=port, TRACE=trace)
+        procs.append(subprocess.Popen([sys.executable, "-c", code], env=env,
+                                      stdout=subprocess.PIPE,
+                                      stderr=subprocess.STDOUT))
+    for pr in procs:
+        o, _ = pr.communicate(timeout=180)
+        assert pr.returncode == 0 and b"CT_OK" in o, o.decode()
+    from parsec_amd.tools.trace2chrome import convert
+    found_comm = 0
+    for r in range(2):
+        out, n = convert(f"{trace}.{r}", str(tmp_path / f"t{r}.json"))
+        text = open(str(tmp_path / f"t{r}.json")).read()
+        found_comm += text.count("comm_send") + text.count("comm_recv")
+    assert found_comm > 0, "no comm events in either rank's trace"

@@ -307,3 +307,37 @@ pm.insert_subtile_extract(tp, A, 0, 0, S)
     r = subprocess.run([_sys.executable, "-c", code], capture_output=True,
                        text=True, timeout=60)
     assert r.returncode != 0 and "subtile" in (r.stderr + r.stdout)
+
+
+def test_window_throttles_inserter():
+    """dtd_window_size bounds how far insertion runs ahead of execution
+    (insert_function.c:75-76 analog): with a small window and slow
+    bodies, the insert loop itself takes execution-scale time."""
+    import subprocess
+    import sys as _sys
+    code = f"""
+import sys, time; sys.path.insert(0, {REPO!r})
+import parsec_amd as pm
+pm.param_set("dtd_window_size", "20")
+pm.param_set("dtd_threshold_size", "10")
+ctx = pm.Context(nworkers=2, rank=0, world=1, gpu=-2)
+A = pm.TiledMatrix(ctx, 8, 8, 8, 8, 1, 1)
+tp = pm.Dtd(ctx)
+
+def slow():
+    time.sleep(0.002)
+
+t0 = time.perf_counter()
+for i in range(300):
+    tp.insert_py(slow, flows=[(A.tile(0, 0), pm.ACCESS_INOUT)])
+dt_insert = time.perf_counter() - t0
+tp.wait()
+# 300 serial 2ms bodies ~ 0.6s; a 20-deep window forces the inserter to
+# ride along for most of it. Unthrottled insertion would take ~ms.
+assert dt_insert > 0.2, dt_insert
+print("WINDOW_OK", round(dt_insert, 3))
+del A, ctx
+"""
+    r = subprocess.run([_sys.executable, "-c", code], capture_output=True,
+                       text=True, timeout=120)
+    assert "WINDOW_OK" in r.stdout, r.stdout + r.stderr

@@ -30,7 +30,7 @@ def main():
     tp = pm.Dtd(ctx)
     pm.insert_spd_fill(tp, A, 42)
     tp.wait()
-    if app in ("qr", "lu"):
+    if app in ("qr", "lu", "ptglu"):
         # QR/LU need the full matrix: fill upper tiles too (spd fill
         # covers the lower triangle only; keep diagonal dominance for LU)
         import numpy as np2
@@ -40,7 +40,7 @@ def main():
                 if tn > tm and A.is_local(tm, tn):
                     A.tile_numpy_set(tm, tn, rng.standard_normal(
                         (A.tile_rows(tm), A.tile_cols(tn))) )
-    hi = ((lambda tm: A.nt) if app in ("qr", "lu")
+    hi = ((lambda tm: A.nt) if app in ("qr", "lu", "ptglu")
           else (lambda tm: min(tm + 1, A.nt)))
     pre = {}
     for tm in range(A.mt):
@@ -53,6 +53,11 @@ def main():
         pm.insert_geqrf(tp2, A)
     elif app == "lu":
         pm.insert_getrf_nopiv(tp2, A)
+    elif app == "ptglu":
+        from parsec_amd.ptg import compile_jdf
+        repo = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+        mod = compile_jdf(os.path.join(repo, "examples", "lu.jdf"))
+        mod.build(ctx, tp2, descA=A, NT=A.mt, NB=nb)
     elif app == "ptg":
         from parsec_amd.ptg import compile_jdf
         repo = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))

@@ -165,7 +165,7 @@ del A, ctx
 """
 
 
-@pytest.mark.parametrize("world,seed", [(2, 101), (2, 202), (4, 303)])
+@pytest.mark.parametrize("world,seed", [(2, 101), (2, 202), (4, 303), (8, 404)])
 def test_distributed_fuzz_vs_oracle(world, seed, tmp_path):
     """Random DAGs (random tiles/modes/executing ranks) through the full
     SPMD protocol must match a sequential numpy oracle."""

@@ -357,7 +357,8 @@ os._exit(17)   # die without teardown, mid-protocol
     pb = subprocess.Popen([sys.executable, "-c", code_victim], env=env,
                           stdout=subprocess.PIPE, stderr=subprocess.STDOUT)
     ob, _ = pb.communicate(timeout=60)
-    assert pb.returncode == 17
+    assert pb.returncode != 0  # victim died (17, or reset-abort if the
+    # teardown race killed it first — incidental either way)
     oa, _ = pa.communicate(timeout=60)  # must NOT hang
     assert pa.returncode != 0 and b"SHOULD_NOT_FINISH" not in oa, oa.decode()
     assert b"connection to rank" in oa or b"FATAL" in oa, oa.decode()

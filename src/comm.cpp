@@ -264,6 +264,11 @@ class TcpComm : public CommEngine {
             p.in_task = it->second;
             posted_recv_.erase(it);
             Data* rd = p.in_task->flows[0].data;
+            // Same loud protocol-bug abort as the unexpected-queue path:
+            // a mismatched frame must never become a heap overflow.
+            PA_CHECK(p.hdr.size == rd->bytes,
+                     "comm: frame size %llu != tile size %zu (protocol bug)",
+                     (unsigned long long)p.hdr.size, rd->bytes);
             rd->begin_host_overwrite();
             p.in_direct = (uint8_t*)rd->ensure_host();
           } else {

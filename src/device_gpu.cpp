@@ -169,9 +169,21 @@ bool GpuEngine::evict_one(size_t) {
   });
   for (Data* d : cand) {
     if (!d->lock.try_lock()) continue;
-    if (!d->dev_ptr || d->dev_refs > 0 || d->h2d_pending) {
+    if (!d->dev_ptr || d->dev_refs > 0) {
       d->lock.unlock();
       continue;
+    }
+    if (d->h2d_pending) {
+      // The staging fence may long have completed: query instead of
+      // skipping forever (a tile once staged from host would otherwise be
+      // permanently unevictable and the evictable pool would only shrink).
+      if (d->h2d_event &&
+          hipEventQuery((hipEvent_t)d->h2d_event) == hipSuccess) {
+        d->h2d_pending = false;
+      } else {
+        d->lock.unlock();
+        continue;
+      }
     }
     if (d->dev_valid && !d->host_valid) {
       // dirty: write back (kernels producing it have retired: refs==0)

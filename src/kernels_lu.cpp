@@ -113,9 +113,18 @@ rocblas_int* lu_dev_info(GpuTaskCtx& g) {
 
 void gpu_getrf(Task& t, GpuTaskCtx& g) {
   const TileArgs& a = t.arg<TileArgs>();
+  rocblas_int* dinfo = lu_dev_info(g);
   PA_CHECK(rocsolver_dgetrf_npvt(lu_handle(g), a.n, a.n,
                                  (double*)t.dev_ptr[0], a.ld,
-                                 lu_dev_info(g)) == rocblas_status_success);
+                                 dinfo) == rocblas_status_success);
+  // Blocking chore (worker thread): read the info scalar back so a
+  // zero/tiny pivot fails as loudly as the CPU reference path does,
+  // instead of silently poisoning the trailing matrix with Inf/NaN.
+  rocblas_int info = 0;
+  PA_HIP_CHECK(hipMemcpyAsync(&info, dinfo, sizeof(info),
+                              hipMemcpyDeviceToHost, g.stream));
+  PA_HIP_CHECK(hipStreamSynchronize(g.stream));
+  PA_CHECK(info == 0, "getrf_nopiv: singular pivot at column %d", (int)info);
 }
 
 void gpu_trsm_l(Task& t, GpuTaskCtx& g) {

@@ -168,6 +168,7 @@ void Scheduler::push(Task* t, int worker_hint) {
     q.lock.lock();
     if (mode_ == 2) q.dq.push_front(t);
     else q.dq.push_back(t);
+    q.sz.store((uint32_t)q.dq.size(), std::memory_order_relaxed);
     q.lock.unlock();
     npending_.fetch_add(1, std::memory_order_release);
     sleep_cv_.notify_one();
@@ -177,6 +178,7 @@ void Scheduler::push(Task* t, int worker_hint) {
     WorkerQ& q = *wq_[worker_hint];
     q.lock.lock();
     q.dq.push_front(t);
+    q.sz.store((uint32_t)q.dq.size(), std::memory_order_relaxed);
     q.lock.unlock();
   } else {
     pq_lock_.lock();
@@ -192,7 +194,11 @@ Task* Scheduler::pop(int worker) {
   if (mode_ != 0) {
     WorkerQ& q = *wq_[0];
     q.lock.lock();
-    if (!q.dq.empty()) { t = q.dq.front(); q.dq.pop_front(); }
+    if (!q.dq.empty()) {
+      t = q.dq.front();
+      q.dq.pop_front();
+      q.sz.store((uint32_t)q.dq.size(), std::memory_order_relaxed);
+    }
     q.lock.unlock();
     if (t) { npending_.fetch_sub(1, std::memory_order_relaxed); return t; }
     pq_lock_.lock();
@@ -204,7 +210,11 @@ Task* Scheduler::pop(int worker) {
   if (worker >= 0) {
     WorkerQ& q = *wq_[worker];
     q.lock.lock();
-    if (!q.dq.empty()) { t = q.dq.front(); q.dq.pop_front(); }
+    if (!q.dq.empty()) {
+      t = q.dq.front();
+      q.dq.pop_front();
+      q.sz.store((uint32_t)q.dq.size(), std::memory_order_relaxed);
+    }
     q.lock.unlock();
     if (t) { npending_.fetch_sub(1, std::memory_order_relaxed); return t; }
   }
@@ -217,9 +227,13 @@ Task* Scheduler::pop(int worker) {
   for (int d = 1; d < nworkers_; d++) {
     int v = (worker >= 0 ? (worker + d) % nworkers_ : d - 1);
     WorkerQ& q = *wq_[v];
-    if (q.dq.empty()) continue;
+    if (q.sz.load(std::memory_order_relaxed) == 0) continue;
     q.lock.lock();
-    if (!q.dq.empty()) { t = q.dq.back(); q.dq.pop_back(); }
+    if (!q.dq.empty()) {
+      t = q.dq.back();
+      q.dq.pop_back();
+      q.sz.store((uint32_t)q.dq.size(), std::memory_order_relaxed);
+    }
     q.lock.unlock();
     if (t) {
       npending_.fetch_sub(1, std::memory_order_relaxed);

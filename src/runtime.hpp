@@ -189,6 +189,10 @@ class Scheduler {
   struct WorkerQ {
     SpinLock lock;
     std::deque<Task*> dq;
+    // Approximate size for the steal fast path: reading dq.empty() on
+    // another worker's deque without its lock is a data race; this atomic
+    // mirror is the race-free heuristic.
+    std::atomic<uint32_t> sz{0};
   };
   struct PQEntry {
     Task* t;

@@ -42,6 +42,10 @@ struct Data {
   // (a task that did not issue the copy still races it otherwise).
   void* h2d_event = nullptr;  // hipEvent_t, lazily created, owned here
   bool h2d_pending = false;
+  // D2H writeback fence (async dirty eviction): recorded on the d2h stream;
+  // host readers wait on it, and a re-stage H2D must order after it.
+  void* d2h_event = nullptr;  // hipEvent_t, lazily created, owned here
+  bool d2h_pending = false;
   // GPU residency management (LRU eviction): pinned while any in-flight
   // GPU task or comm transfer uses the device copy.
   int dev_refs = 0;          // guarded by lock

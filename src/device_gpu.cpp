@@ -56,6 +56,7 @@ GpuEngine::GpuEngine(Context* ctx, int device) : ctx_(ctx), device_(device) {
       PA_HIP_CHECK(hipMalloc(&slab_, slab_bytes_));
     }
   }
+  zone_.init(slab_bytes_);
   PA_DEBUG(1, "GPU %d engine: %d exec streams, slab %.1f GB", device_,
            nstreams, slab_bytes_ / 1e9);
   manager_ = std::thread([this] { manager_main(); });
@@ -66,6 +67,10 @@ GpuEngine::~GpuEngine() {
   q_cv_.notify_all();
   if (manager_.joinable()) manager_.join();
   PA_HIP_CHECK(hipSetDevice(device_));
+  for (auto& dr : draining_) {
+    hipEventSynchronize(dr.ev);  // writebacks must land before teardown
+    hipEventDestroy(dr.ev);
+  }
   for (auto& e : event_pool_) hipEventDestroy(e);
   for (auto s : exec_streams_) hipStreamDestroy(s);
   hipStreamDestroy(h2d_stream_);
@@ -83,33 +88,59 @@ void GpuEngine::enqueue(Task* t) {
 }
 
 hipEvent_t GpuEngine::event_get() {
-  if (!event_pool_.empty()) {
-    hipEvent_t e = event_pool_.back();
-    event_pool_.pop_back();
-    return e;
+  {
+    SpinGuard g(ev_lock_);
+    if (!event_pool_.empty()) {
+      hipEvent_t e = event_pool_.back();
+      event_pool_.pop_back();
+      return e;
+    }
   }
   hipEvent_t e;
   PA_HIP_CHECK(hipEventCreateWithFlags(&e, hipEventDisableTiming));
   return e;
 }
 
-void GpuEngine::event_put(hipEvent_t e) { event_pool_.push_back(e); }
+void GpuEngine::event_put(hipEvent_t e) {
+  SpinGuard g(ev_lock_);
+  event_pool_.push_back(e);
+}
+
+bool GpuEngine::reap_draining_locked() {
+  bool any = false;
+  for (size_t i = 0; i < draining_.size();) {
+    hipError_t e = hipEventQuery(draining_[i].ev);
+    if (e == hipErrorNotReady) {
+      i++;
+      continue;
+    }
+    PA_HIP_CHECK(e);
+    Draining dr = draining_[i];
+    draining_.erase(draining_.begin() + i);
+    event_put(dr.ev);
+    if (dr.buf >= slab_ && dr.buf < (char*)slab_ + slab_bytes_)
+      zone_.free((size_t)((char*)dr.buf - (char*)slab_), dr.bytes);
+    else
+      extern_lists_[dr.bytes].push_back(dr.buf);
+    any = true;
+  }
+  return any;
+}
 
 void* GpuEngine::dev_alloc(size_t bytes) {
   bytes = (bytes + 255) & ~size_t(255);
-  const double deadline = now_s() + 60.0;  // patience under pressure
+  const double deadline =
+      now_s() + (double)param_int("gpu_alloc_timeout_s", 120);
   while (now_s() < deadline) {
     {
       std::lock_guard<std::mutex> g(mem_mtx_);
-      auto it = free_lists_.find(bytes);
-      if (it != free_lists_.end() && !it->second.empty()) {
+      reap_draining_locked();
+      size_t off = zone_.alloc(bytes);
+      if (off != ZoneAlloc::NPOS) return (char*)slab_ + off;
+      auto it = extern_lists_.find(bytes);
+      if (it != extern_lists_.end() && !it->second.empty()) {
         void* p = it->second.back();
         it->second.pop_back();
-        return p;
-      }
-      if (slab_used_ + bytes <= slab_bytes_) {
-        void* p = (char*)slab_ + slab_used_;
-        slab_used_ += bytes;
         return p;
       }
     }
@@ -129,8 +160,9 @@ void* GpuEngine::dev_alloc(size_t bytes) {
       std::this_thread::sleep_for(std::chrono::microseconds(200));
     }
   }
-  fatal("GPU %d out of memory allocating %zu bytes (slab %zu/%zu)", device_,
-        bytes, slab_used_, slab_bytes_);
+  fatal("GPU %d out of memory allocating %zu bytes (slab %zu/%zu in use, "
+        "largest free %zu)",
+        device_, bytes, zone_.in_use(), slab_bytes_, zone_.largest_free());
 }
 
 void GpuEngine::note_resident(Data* d) {
@@ -163,6 +195,7 @@ bool GpuEngine::evict_one(size_t) {
   // Tile locks are only try_lock'd (their holders may take mem_mtx_ via
   // dev_free, and try_lock never blocks, so no lock-order deadlock).
   std::lock_guard<std::mutex> g(mem_mtx_);
+  PA_HIP_CHECK(hipSetDevice(device_));  // callers may be worker threads
   std::vector<Data*> cand = resident_;
   std::sort(cand.begin(), cand.end(), [](Data* a, Data* b) {
     return a->dev_last_use < b->dev_last_use;
@@ -185,22 +218,43 @@ bool GpuEngine::evict_one(size_t) {
         continue;
       }
     }
+    void* buf = d->dev_ptr;
+    size_t rb = (d->bytes + 255) & ~size_t(255);
+    bool drained_free = true;
     if (d->dev_valid && !d->host_valid) {
-      // dirty: write back (kernels producing it have retired: refs==0)
+      // Dirty: asynchronous writeback (transfer_gpu.c:1-362 analog — the
+      // reference makes eviction a D2H *task*; here the buffer parks on the
+      // draining list until its event completes, and the manager never
+      // synchronizes inside an allocation).
       if (!d->host_ptr) {
         if (posix_memalign(&d->host_ptr, 4096, d->bytes) != 0) {
           d->lock.unlock();
           continue;
         }
       }
-      copy_d2h(d->host_ptr, d->dev_ptr, d->bytes);
-      d->host_valid = true;
+      PA_HIP_CHECK(hipMemcpyAsync(d->host_ptr, buf, d->bytes,
+                                  hipMemcpyDeviceToHost, d2h_stream_));
+      stats.bytes_d2h += d->bytes;
+      if (!d->d2h_event)
+        PA_HIP_CHECK(hipEventCreateWithFlags((hipEvent_t*)&d->d2h_event,
+                                             hipEventDisableTiming));
+      PA_HIP_CHECK(hipEventRecord((hipEvent_t)d->d2h_event, d2h_stream_));
+      d->d2h_pending = true;
+      d->host_valid = true;  // valid in stream order behind d2h_event
+      hipEvent_t drev = event_get();
+      PA_HIP_CHECK(hipEventRecord(drev, d2h_stream_));
+      draining_.push_back(Draining{buf, rb, drev});
+      drained_free = false;
     }
-    void* buf = d->dev_ptr;
     d->dev_ptr = nullptr;
     d->dev_valid = false;
     d->lock.unlock();
-    free_lists_[(d->bytes + 255) & ~size_t(255)].push_back(buf);
+    if (drained_free) {
+      if (buf >= slab_ && buf < (char*)slab_ + slab_bytes_)
+        zone_.free((size_t)((char*)buf - (char*)slab_), rb);
+      else
+        extern_lists_[rb].push_back(buf);
+    }
     resident_.erase(std::remove(resident_.begin(), resident_.end(), d),
                     resident_.end());
     PA_DEBUG(2, "evicted tile %lu (%zu bytes)", (unsigned long)d->key,
@@ -214,7 +268,10 @@ bool GpuEngine::evict_one(size_t) {
 void GpuEngine::dev_free(void* p, size_t bytes) {
   bytes = (bytes + 255) & ~size_t(255);
   std::lock_guard<std::mutex> g(mem_mtx_);
-  free_lists_[bytes].push_back(p);
+  if (p >= slab_ && p < (char*)slab_ + slab_bytes_)
+    zone_.free((size_t)((char*)p - (char*)slab_), bytes);
+  else
+    extern_lists_[bytes].push_back(p);
 }
 
 void GpuEngine::copy_d2h(void* dst, const void* src, size_t bytes) {
@@ -268,6 +325,11 @@ void GpuEngine::stage_flows(Task* t, hipStream_t es) {
     SpinGuard g(d->lock);
     if ((t->flows[i].mode & ACCESS_IN) && !d->dev_valid) {
       PA_CHECK(d->host_valid, "stage-in: no valid copy for tile");
+      // A host copy produced by an async eviction writeback may still be in
+      // flight on the d2h stream: order the re-stage H2D behind it.
+      if (d->d2h_pending)
+        PA_HIP_CHECK(
+            hipStreamWaitEvent(h2d_stream_, (hipEvent_t)d->d2h_event, 0));
       PA_HIP_CHECK(hipMemcpyAsync(d->dev_ptr, d->host_ptr, d->bytes,
                                   hipMemcpyHostToDevice, h2d_stream_));
       stats.bytes_h2d += d->bytes;
@@ -428,7 +490,10 @@ Data::~Data() {
     eng->forget(this);  // never leave a dangling pointer in the LRU set
     if (dev_ptr) eng->dev_free(dev_ptr, bytes);
   }
+  if (d2h_pending)  // writeback still in flight targets host_ptr
+    hipEventSynchronize((hipEvent_t)d2h_event);
   if (h2d_event) hipEventDestroy((hipEvent_t)h2d_event);
+  if (d2h_event) hipEventDestroy((hipEvent_t)d2h_event);
   if (host_ptr) free(host_ptr);
 }
 
@@ -450,6 +515,11 @@ void* Data::pull_to_host() {
     eng->sync_all();  // quiesce producers before readback
     eng->copy_d2h(host_ptr, dev_ptr, bytes);
     host_valid = true;
+  } else if (d2h_pending) {
+    // Eviction writeback in flight: the host copy is valid only behind
+    // its fence.
+    PA_HIP_CHECK(hipEventSynchronize((hipEvent_t)d2h_event));
+    d2h_pending = false;
   }
   return host_ptr;
 }
@@ -457,6 +527,11 @@ void* Data::pull_to_host() {
 void Data::begin_host_overwrite() {
   SpinGuard g(lock);
   dev_valid = false;
+  if (d2h_pending) {
+    // An in-flight writeback targets the buffer we are about to overwrite.
+    PA_HIP_CHECK(hipEventSynchronize((hipEvent_t)d2h_event));
+    d2h_pending = false;
+  }
 }
 
 void Data::written_on(bool device) {

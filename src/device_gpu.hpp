@@ -25,6 +25,7 @@
 
 #include "common.hpp"
 #include "runtime.hpp"
+#include "zone_alloc.hpp"
 
 namespace pa {
 
@@ -127,12 +128,27 @@ class GpuEngine {
   // memory pool
   bool evict_one(size_t bytes);
   bool retire_pass();  // poll event rings; true if anything retired
+  // Return writeback-drained eviction buffers to the pool (mem_mtx_ held).
+  bool reap_draining_locked();
   std::thread::id manager_tid_;
   std::mutex mem_mtx_;
   void* slab_ = nullptr;
-  size_t slab_bytes_ = 0, slab_used_ = 0;
+  size_t slab_bytes_ = 0;
   bool hard_cap_ = false;  // gpu_mem_limit_mb set: no hipMalloc escape
-  std::map<size_t, std::vector<void*>> free_lists_;
+  ZoneAlloc zone_;         // coalescing slab allocator (zone_malloc analog)
+  // Overflow blocks from the hipMalloc escape hatch (outside the slab):
+  // recycled by exact size, never coalesced (rare path).
+  std::map<size_t, std::vector<void*>> extern_lists_;
+  // Evicted-dirty buffers whose async D2H writeback is still in flight:
+  // reusable only once the event completes (transfer_gpu.c W2R-task analog
+  // — eviction never synchronizes the manager).
+  struct Draining {
+    void* buf;
+    size_t bytes;
+    hipEvent_t ev;
+  };
+  std::vector<Draining> draining_;
+  SpinLock ev_lock_;  // event_pool_ is shared with eviction callers
   std::vector<Data*> resident_;
   std::atomic<uint64_t> lru_clock_{1};
 };

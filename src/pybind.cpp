@@ -245,6 +245,26 @@ PYBIND11_MODULE(_core, m) {
         d->written_on(false);
       });
 
+  // CPU-testable view of the GPU slab allocator (zone_malloc analog).
+  py::class_<ZoneAlloc>(m, "ZoneAlloc")
+      .def(py::init([](size_t bytes) {
+             auto* z = new ZoneAlloc();
+             z->init(bytes);
+             return z;
+           }),
+           py::arg("bytes"))
+      .def("alloc",
+           [](ZoneAlloc& z, size_t sz) -> py::object {
+             size_t off = z.alloc(sz);
+             if (off == ZoneAlloc::NPOS) return py::none();
+             return py::int_(off);
+           })
+      .def("free", &ZoneAlloc::free, py::arg("off"), py::arg("size"))
+      .def_property_readonly("in_use", &ZoneAlloc::in_use)
+      .def_property_readonly("capacity", &ZoneAlloc::capacity)
+      .def_property_readonly("largest_free", &ZoneAlloc::largest_free)
+      .def_property_readonly("free_blocks", &ZoneAlloc::free_blocks);
+
   py::class_<IrregularCollection>(m, "IrregularCollection")
       .def(py::init<Context*>(), py::arg("ctx"), py::keep_alive<1, 2>())
       .def("add", &IrregularCollection::add, py::arg("key"), py::arg("rank"),

@@ -204,3 +204,61 @@ def test_gpu_random_dag_vs_oracle():
         got = M.tile_numpy(0, 0)
         assert np.allclose(got, ref[i]), f"tile {i} diverged"
     del mats, ctx
+
+
+def test_eviction_mixed_tile_sizes():
+    """Fragmentation pressure: two collections with different tile sizes
+    plus subtile traffic under a hard HBM cap — the coalescing slab
+    allocator must recycle mixed-size blocks (round-1 size-class freelists
+    could strand capacity here) and host-staged tiles must stay evictable
+    (h2d fence clearing)."""
+    code = f"""
+import numpy as np, sys
+sys.path.insert(0, {REPO!r})
+import parsec_amd as pm
+pm.param_set("gpu_mem_limit_mb", "80")
+pm.param_set("gpu_max_inflight", "4")
+ctx = pm.Context(nworkers=2, rank=0, world=1)
+n1, nb1 = 4096, 512     # 2 MB tiles
+n2, nb2 = 3520, 320     # 0.8 MB tiles (different size class)
+A = pm.TiledMatrix(ctx, n1, n1, nb1, nb1, 1, 1)
+B = pm.TiledMatrix(ctx, n2, n2, nb2, nb2, 1, 1)
+tp = pm.Dtd(ctx)
+pm.insert_spd_fill(tp, A, 42)
+pm.insert_spd_fill(tp, B, 7)
+tp.wait()
+# host-stage every tile of A (tile_numpy pulls to host), then make the GPU
+# re-stage them H2D under pressure: staged tiles must remain evictable
+MA = np.zeros((n1,n1)); MB = np.zeros((n2,n2))
+for tm in range(A.mt):
+    for tn in range(tm+1):
+        MA[tm*nb1:(tm+1)*nb1, tn*nb1:(tn+1)*nb1] = A.tile_numpy(tm,tn)
+for tm in range(B.mt):
+    for tn in range(tm+1):
+        MB[tm*nb2:(tm+1)*nb2, tn*nb2:(tn+1)*nb2] = B.tile_numpy(tm,tn)
+MA = np.tril(MA) + np.tril(MA,-1).T
+MB = np.tril(MB) + np.tril(MB,-1).T
+L0A = np.linalg.cholesky(MA)
+L0B = np.linalg.cholesky(MB)
+tp2 = pm.Dtd(ctx)
+pm.insert_potrf(tp2, A)
+pm.insert_potrf(tp2, B)
+tp2.wait()
+LA = np.zeros((n1,n1)); LB = np.zeros((n2,n2))
+for tm in range(A.mt):
+    for tn in range(tm+1):
+        LA[tm*nb1:(tm+1)*nb1, tn*nb1:(tn+1)*nb1] = A.tile_numpy(tm,tn)
+for tm in range(B.mt):
+    for tn in range(tm+1):
+        LB[tm*nb2:(tm+1)*nb2, tn*nb2:(tn+1)*nb2] = B.tile_numpy(tm,tn)
+errA = np.abs(np.tril(LA)-L0A).max()
+errB = np.abs(np.tril(LB)-L0B).max()
+st = ctx.gpu_stats()
+print("MIXED_ERR", errA, errB, "evictions", st["evictions"])
+assert errA < 1e-8 and errB < 1e-8, (errA, errB)
+assert st["evictions"] > 0, "cap did not force eviction - test is vacuous"
+del A, B, ctx
+"""
+    r = subprocess.run([sys.executable, "-c", code], capture_output=True,
+                       text=True, timeout=600)
+    assert r.returncode == 0, r.stdout + r.stderr

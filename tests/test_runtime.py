@@ -341,3 +341,47 @@ del A, ctx
     r = subprocess.run([_sys.executable, "-c", code], capture_output=True,
                        text=True, timeout=120)
     assert "WINDOW_OK" in r.stdout, r.stdout + r.stderr
+
+
+def test_zone_alloc_coalescing():
+    """Slab allocator (zone_malloc analog): best-fit + neighbor merging.
+    Freeing mixed-size interleaved blocks must restore one maximal block;
+    round-1 size-class freelists failed this (capacity stranding)."""
+    z = pm.ZoneAlloc(1 << 20)
+    assert z.capacity == 1 << 20
+    offs = [(z.alloc(sz), sz) for sz in (4096, 1024, 65536, 256, 4096)]
+    assert all(o is not None for o, _ in offs)
+    assert z.in_use == sum(sz for _, sz in offs)
+    # free in scrambled order; neighbors must merge back
+    import random
+    random.Random(5).shuffle(offs)
+    for o, sz in offs:
+        z.free(o, sz)
+    assert z.in_use == 0
+    assert z.free_blocks == 1
+    assert z.largest_free == 1 << 20
+    # a request larger than any freelist class but smaller than the merged
+    # block must succeed
+    assert z.alloc((1 << 20) - 64) is not None
+
+
+def test_zone_alloc_fragmentation_reuse():
+    """Alternating alloc/free of different sizes must not strand capacity:
+    after freeing everything, a full-slab allocation succeeds."""
+    z = pm.ZoneAlloc(1 << 16)
+    live = []
+    import random
+    rng = random.Random(11)
+    for step in range(2000):
+        if live and (rng.random() < 0.5 or z.largest_free < 2048):
+            o, sz = live.pop(rng.randrange(len(live)))
+            z.free(o, sz)
+        else:
+            sz = rng.choice([256, 512, 768, 1024, 2048])
+            o = z.alloc(sz)
+            if o is not None:
+                live.append((o, sz))
+    for o, sz in live:
+        z.free(o, sz)
+    assert z.in_use == 0 and z.free_blocks == 1
+    assert z.alloc(1 << 16) == 0

@@ -28,6 +28,7 @@ struct Data {
   // identity
   uint64_t key = 0;
   class TiledMatrix* coll = nullptr;   // owning tiled collection (if any)
+  class IrregularCollection* icoll = nullptr;  // or owning irregular coll
   class Context* ctx_direct = nullptr; // set instead for irregular data
   int home_rank = 0;
   size_t bytes = 0;
@@ -61,6 +62,12 @@ struct Data {
   bool local_present = false;
 
   ~Data();
+
+  // Free host+device buffers and drop chain references, keeping the shell
+  // alive (renamed-out copies: the object must survive — application code
+  // may hold stale handles that insertion canonicalizes through — but its
+  // buffers are reclaimed as soon as the last reader drains).
+  void drop_buffers();
 
   // Ensure a host buffer exists (allocates page-aligned memory).
   void* ensure_host();
@@ -132,6 +139,14 @@ class TiledMatrix {
   }
   Data* tile(int tm, int tn);
 
+  // (datarepo/arena analog, datarepo.h:25-92: an incoming version never
+  // overwrites a buffer still being read.)
+  // Copy renaming: park `old` on the retired list (shell stays alive for
+  // the collection's lifetime; its buffers are freed by the DTD reclaim
+  // task) and install a fresh Data in the slot. Returns the fresh copy.
+  Data* rename_tile(Data* old);
+  Data* current_by_key(uint64_t key) { return tiles_[(size_t)key].get(); }
+
   // rows/cols of a (possibly partial) edge tile
   int tile_rows(int tm) const {
     int64_t r = m_ - (int64_t)tm * mb_;
@@ -154,6 +169,7 @@ class TiledMatrix {
   int kl_ = 0, ku_ = 0;
   std::vector<int> ranks_;  // tabular override (empty = cyclic)
   std::vector<std::unique_ptr<Data>> tiles_;  // mt*nt, metadata lazy
+  std::vector<std::unique_ptr<Data>> retired_;  // renamed-out copy shells
 
   bool any_tiles() const {
     for (auto& t : tiles_)
@@ -174,12 +190,18 @@ class IrregularCollection {
   // registration and must be identical on every rank (SPMD).
   Data* add(uint64_t key, int rank, size_t bytes);
   Data* at(uint64_t key);
+  Data* rename(Data* old);  // see TiledMatrix::rename_tile
+  Data* current_by_key(uint64_t key) {
+    auto it = map_.find(key);
+    return it == map_.end() ? nullptr : it->second.get();
+  }
   Context* ctx() const { return ctx_; }
   size_t size() const { return map_.size(); }
 
  private:
   Context* ctx_;
   std::unordered_map<uint64_t, std::unique_ptr<Data>> map_;
+  std::vector<std::unique_ptr<Data>> retired_;  // renamed-out copy shells
 };
 
 }  // namespace pa

@@ -480,22 +480,36 @@ void GpuEngine::manager_main() {
 }
 
 // ---------------------------------------------------------------- Data
-Data::~Data() {
+void Data::drop_buffers() {
   // Drop DTD chaining references to completed tasks.
-  if (last_local_writer) last_local_writer->release();
+  if (last_local_writer) {
+    last_local_writer->release();
+    last_local_writer = nullptr;
+  }
   for (Task* r : local_readers) r->release();
+  local_readers.clear();
   Context* c = coll ? coll->ctx() : ctx_direct;
   GpuEngine* eng = c ? c->gpu() : nullptr;
   if (eng) {
     eng->forget(this);  // never leave a dangling pointer in the LRU set
     if (dev_ptr) eng->dev_free(dev_ptr, bytes);
   }
-  if (d2h_pending)  // writeback still in flight targets host_ptr
+  dev_ptr = nullptr;
+  dev_valid = false;
+  if (d2h_pending) {  // writeback still in flight targets host_ptr
     hipEventSynchronize((hipEvent_t)d2h_event);
+    d2h_pending = false;
+  }
   if (h2d_event) hipEventDestroy((hipEvent_t)h2d_event);
   if (d2h_event) hipEventDestroy((hipEvent_t)d2h_event);
+  h2d_event = d2h_event = nullptr;
+  h2d_pending = false;
   if (host_ptr) free(host_ptr);
+  host_ptr = nullptr;
+  host_valid = false;
 }
+
+Data::~Data() { drop_buffers(); }
 
 void* Data::ensure_host() {
   if (!host_ptr) {
@@ -565,6 +579,7 @@ Data* IrregularCollection::add(uint64_t key, int rank, size_t bytes) {
   if (!slot) {
     slot = std::make_unique<Data>();
     slot->key = key;
+    slot->icoll = this;
     slot->ctx_direct = ctx_;
     slot->home_rank = rank;
     slot->owner_rank = rank;
@@ -587,6 +602,35 @@ void TiledMatrix::set_rank_table(std::vector<int> table) {
     PA_CHECK(r >= 0 && r < ctx_->world(), "rank table entry out of range");
   PA_CHECK(!any_tiles(), "set_rank_table: before first tile access");
   ranks_ = std::move(table);
+}
+
+Data* TiledMatrix::rename_tile(Data* old) {
+  size_t idx = (size_t)old->key;
+  PA_CHECK(idx < tiles_.size() && tiles_[idx].get() == old,
+           "rename_tile: not the current copy");
+  auto d = std::make_unique<Data>();
+  d->key = old->key;
+  d->coll = this;
+  d->home_rank = old->home_rank;
+  d->bytes = old->bytes;
+  retired_.push_back(std::move(tiles_[idx]));
+  tiles_[idx] = std::move(d);
+  return tiles_[idx].get();
+}
+
+Data* IrregularCollection::rename(Data* old) {
+  auto it = map_.find(old->key);
+  PA_CHECK(it != map_.end() && it->second.get() == old,
+           "rename: not the current copy");
+  auto d = std::make_unique<Data>();
+  d->key = old->key;
+  d->icoll = this;
+  d->ctx_direct = ctx_;
+  d->home_rank = old->home_rank;
+  d->bytes = old->bytes;
+  retired_.push_back(std::move(it->second));
+  it->second = std::move(d);
+  return it->second.get();
 }
 
 Data* TiledMatrix::tile(int tm, int tn) {

@@ -1,5 +1,7 @@
 #include "dtd.hpp"
 
+#include "profiling.hpp"
+
 // Reference parity notes are in dtd.hpp's header (insert_function.c
 // chaining, window throttling insert_function.c:75-76, data_flush).
 
@@ -20,6 +22,17 @@ TaskClass COMM_RECV_CLASS = [] {
 }();
 
 constexpr int COMM_PRIORITY = 1 << 28;  // transfers go out as early as possible
+
+// Internal reclaim task: depends on every live user of a renamed-out copy
+// and deletes it once they drain (the datarepo retire protocol rebuilt on
+// the dependency engine itself).
+TaskClass RECLAIM_CLASS = [] {
+  TaskClass tc;
+  tc.name = "copy_reclaim";
+  tc.kind = TaskKind::CPU;
+  tc.cpu_hook = [](Task& t) { t.arg<Data*>()->drop_buffers(); };
+  return tc;
+}();
 
 Dtd::Dtd(Context* ctx, std::string name) : Taskpool(ctx, std::move(name)) {
   me_ = ctx->rank();
@@ -53,23 +66,68 @@ void Dtd::make_send(Data* d, int dst, uint64_t seq) {
   task_dec_deps(st);  // release insertion guard
 }
 
-void Dtd::make_recv(Data* d, int src, uint64_t seq) {
+// If `d`'s current buffer still has live local users, swap a fresh copy
+// into the collection slot (the rename) and schedule the old one for
+// reclamation behind its users. Returns the Data future work must use.
+// A Data* held by the application is a persistent NAME for the tile: after
+// a rename the collection slot holds a fresh copy, so insertion-side entry
+// points canonicalize to the current copy first (stale handles from before
+// a rename keep working, like repo keys in the reference).
+static Data* current_copy(Data* d) {
+  if (d->coll) return d->coll->current_by_key(d->key);
+  if (d->icoll) return d->icoll->current_by_key(d->key);
+  return d;
+}
+
+Data* Dtd::maybe_rename(Data* d) {
+  auto live = [](Task* t) {
+    if (!t) return false;
+    t->lock.lock();
+    bool l = !t->completed;
+    t->lock.unlock();
+    return l;
+  };
+  bool any_live = live(d->last_local_writer);
+  for (Task* r : d->local_readers) {
+    if (any_live) break;
+    any_live = live(r);
+  }
+  if (!any_live) return d;  // nobody reads the old version: reuse in place
+  if (!d->coll && !d->icoll)
+    return d;  // standalone scratch datum (NEW tile): WAR-serialize instead
+  Data* nd = d->coll ? d->coll->rename_tile(d) : d->icoll->rename(d);
+  nd->version = d->version;
+  nd->owner_rank = d->owner_rank;
+  nd->sent_mask = d->sent_mask;
+  Task* rc = task_new(this, &RECLAIM_CLASS);
+  rc->arg<Data*>() = d;  // shell owned by the collection
+  if (d->last_local_writer) task_add_edge(d->last_local_writer, rc);
+  for (Task* r : d->local_readers) task_add_edge(r, rc);
+  counters().renames.fetch_add(1, std::memory_order_relaxed);
+  task_dec_deps(rc);
+  return nd;
+}
+
+Data* Dtd::make_recv(Data* d, int src, uint64_t seq) {
+  d = maybe_rename(d);
   Task* rt = task_new(this, &COMM_RECV_CLASS);
   rt->peer = src;
   rt->comm_seq = seq;
   rt->flows[0] = {d, ACCESS_INOUT};
   rt->nflows = 1;
   rt->priority = COMM_PRIORITY;
-  // The recv overwrites the local buffer: WAR against every local user.
+  // Only reachable without rename when every old user already completed
+  // (then these add no edges) — the recv never WAR-waits on live readers.
   if (d->last_local_writer) task_add_edge(d->last_local_writer, rt);
   for (Task* r : d->local_readers) task_add_edge(r, rt);
   set_local_writer(d, rt);
   d->local_present = true;
   d->local_present_version = d->version;
   task_dec_deps(rt);
+  return d;
 }
 
-void Dtd::read_flow(Data* d, Task* t, int task_rank) {
+Data* Dtd::read_flow(Data* d, Task* t, int task_rank) {
   const int O = d->owner_rank, R = task_rank;
   if (O != R && !(d->sent_mask & (1ull << R))) {
     // Every rank advances the replicated channel counter; only the
@@ -77,18 +135,25 @@ void Dtd::read_flow(Data* d, Task* t, int task_rank) {
     uint64_t seq = chan_next(O, R);
     d->sent_mask |= 1ull << R;
     if (O == me_) make_send(d, R, seq);
-    if (R == me_) make_recv(d, O, seq);
+    if (R == me_) d = make_recv(d, O, seq);
   }
   if (R == me_) {
     if (d->last_local_writer) task_add_edge(d->last_local_writer, t);
     d->local_readers.push_back(t);
     t->retain();
   }
+  return d;
 }
 
-void Dtd::write_flow(Data* d, Task* t, int task_rank) {
+Data* Dtd::write_flow(Data* d, Task* t, int task_rank, bool output_only) {
   const int R = task_rank;
   if (R == me_) {
+    if (output_only) {
+      // A pure OUTPUT rewrite need not wait for readers of the previous
+      // version: rename instead of WAR-serializing (datarepo copy-per-
+      // version semantics).
+      d = maybe_rename(d);
+    }
     if (d->last_local_writer && d->last_local_writer != t)
       task_add_edge(d->last_local_writer, t);  // WAW / RAW on buffer
     for (Task* r : d->local_readers)
@@ -104,6 +169,7 @@ void Dtd::write_flow(Data* d, Task* t, int task_rank) {
   }
   // Remote writer: local copy (if any) becomes stale; local_present_version
   // keeps the old version so a later local reader triggers a fetch.
+  return d;
 }
 
 bool Dtd::insert(const TaskClass* tc, const void* args, size_t args_bytes,
@@ -132,14 +198,18 @@ Task* Dtd::insert_begin(const TaskClass* tc, const void* args,
     t = task_new(this, tc);
     t->priority = priority;
     if (args_bytes) memcpy(t->args, args, args_bytes);
-    for (int i = 0; i < nflows; i++)
-      t->flows[i] = {flows[i].d, flows[i].mode};
     t->nflows = nflows;
   }
   for (int i = 0; i < nflows; i++) {
-    if (!flows[i].d) continue;  // NULL flow (e.g. absent stencil halo)
-    if (flows[i].mode & ACCESS_IN) read_flow(flows[i].d, t, task_rank);
-    if (flows[i].mode & ACCESS_OUT) write_flow(flows[i].d, t, task_rank);
+    Data* d = flows[i].d ? current_copy(flows[i].d) : nullptr;
+    if (d) {  // NULL flow (e.g. absent stencil halo)
+      if (flows[i].mode & ACCESS_IN) d = read_flow(d, t, task_rank);
+      if (flows[i].mode & ACCESS_OUT)
+        d = write_flow(d, t, task_rank,
+                       /*output_only=*/!(flows[i].mode & ACCESS_IN));
+    }
+    // The task binds to the renamed copy, not the pointer the caller held.
+    if (t) t->flows[i] = {d, flows[i].mode};
   }
   return t;
 }
@@ -160,11 +230,12 @@ void Dtd::insert_commit(Task* t) {
 
 void Dtd::flush(Data* d) {
   std::lock_guard<std::recursive_mutex> g(insert_mtx_);
+  d = current_copy(d);
   const int O = d->owner_rank, H = d->home_rank;
   if (O == H) return;
   uint64_t seq = chan_next(O, H);
   if (O == me_) make_send(d, H, seq);
-  if (H == me_) make_recv(d, O, seq);
+  if (H == me_) d = make_recv(d, O, seq);
   d->owner_rank = H;
   d->sent_mask = 0;
 }

@@ -58,10 +58,18 @@ class Dtd : public Taskpool {
   void flush_all(TiledMatrix& A);
 
  private:
-  void read_flow(Data* d, Task* t, int task_rank);
-  void write_flow(Data* d, Task* t, int task_rank);
+  // read/write_flow and make_recv return the (possibly renamed) current
+  // Data: when a recv or an OUTPUT-only rewrite targets a tile whose old
+  // version still has live local readers, the collection slot gets a fresh
+  // copy (datarepo/arena renaming, datarepo.h:25-92 /
+  // remote_dep_mpi.c:572-615) and the old one is reclaimed by an internal
+  // task once its readers drain — instead of WAR-serializing the transfer
+  // behind every reader.
+  Data* read_flow(Data* d, Task* t, int task_rank);
+  Data* write_flow(Data* d, Task* t, int task_rank, bool output_only);
+  Data* maybe_rename(Data* d);
   void make_send(Data* d, int dst, uint64_t seq);
-  void make_recv(Data* d, int src, uint64_t seq);
+  Data* make_recv(Data* d, int src, uint64_t seq);
   static void set_local_writer(Data* d, Task* w);
   uint64_t chan_next(int src, int dst) {
     return chan_seq_[(size_t)src * world_ + dst]++;

@@ -109,6 +109,10 @@ struct RuntimeCounters {
   std::atomic<uint64_t> steals{0};
   std::atomic<uint64_t> comm_msgs{0};
   std::atomic<uint64_t> comm_bytes{0};
+  // Copy renames: incoming versions (and OUTPUT-only rewrites) that got a
+  // fresh buffer instead of WAR-waiting on readers of the old version
+  // (datarepo/arena copy analog).
+  std::atomic<uint64_t> renames{0};
 };
 RuntimeCounters& counters();
 

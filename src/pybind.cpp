@@ -165,6 +165,7 @@ PYBIND11_MODULE(_core, m) {
         d["steals"] = c.steals.load();
         d["comm_msgs"] = c.comm_msgs.load();
         d["comm_bytes"] = c.comm_bytes.load();
+        d["renames"] = c.renames.load();
         return d;
       });
 

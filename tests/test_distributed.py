@@ -362,3 +362,61 @@ os._exit(17)   # die without teardown, mid-protocol
     oa, _ = pa.communicate(timeout=60)  # must NOT hang
     assert pa.returncode != 0 and b"SHOULD_NOT_FINISH" not in oa, oa.decode()
     assert b"connection to rank" in oa or b"FATAL" in oa, oa.decode()
+
+
+def test_recv_copy_renaming(tmp_path):
+    """An incoming version must not WAR-wait on readers of the previous
+    version: the engine renames the tile to a fresh copy (datarepo/arena
+    semantics, datarepo.h:25-92). Rank 1 rewrites a rank-0 tile each round
+    while rank 0 holds a slow reader of the previous version; every reader
+    must see exactly its version's value, the stale python handle must stay
+    usable across renames, and the renames counter must engage."""
+    import conftest
+    REPO = os.path.dirname(HERE)
+    port = str(conftest.port_base(13))
+    code = """
+import os, sys, time
+sys.path.insert(0, %r)
+import numpy as np
+import parsec_amd as pm
+rank = int(os.environ["RANK"])
+pm.param_set("comm_base_port", os.environ["PORT"])
+ctx = pm.Context(nworkers=2, rank=rank, world=2, comm="tcp", gpu=-2)
+nb = 32
+A = pm.TiledMatrix(ctx, 2 * nb, nb, nb, nb, 2, 1)
+tp = pm.Dtd(ctx)
+t00 = A.tile(0, 0)   # held across renames on purpose
+seen = []
+ROUNDS = 12
+for k in range(ROUNDS):
+    def w(buf, k=k):
+        np.frombuffer(buf, dtype=np.float64)[:] = k + 1
+    tp.insert_py(w, [(t00, pm.ACCESS_OUT)], rank=1, with_data=True)
+    def r(buf, k=k):
+        seen.append((k, float(np.frombuffer(buf, dtype=np.float64)[0])))
+        time.sleep(0.02)  # hold the old version while the next recv lands
+    tp.insert_py(r, [(t00, pm.ACCESS_IN)], rank=0, with_data=True)
+tp.wait()
+if rank == 0:
+    assert len(seen) == ROUNDS, seen
+    for k, v in seen:
+        assert v == k + 1, (k, v)
+    c = ctx.counters()
+    assert c["renames"] > 0, c  # renaming actually engaged
+    print("RENAME_OK", c["renames"])
+ctx.barrier()
+del A, ctx
+""" % (REPO,)
+    import subprocess as sp
+    procs = []
+    for r in range(2):
+        env = dict(os.environ)
+        env.update(RANK=str(r), WORLD_SIZE="2", PORT=port)
+        procs.append(sp.Popen([sys.executable, "-c", code], env=env,
+                              stdout=sp.PIPE, stderr=sp.STDOUT))
+    outs = []
+    for pr in procs:
+        o, _ = pr.communicate(timeout=180)
+        outs.append(o)
+        assert pr.returncode == 0, o.decode()
+    assert b"RENAME_OK" in outs[0], outs[0].decode()

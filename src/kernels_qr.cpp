@@ -355,6 +355,220 @@ __global__ void __launch_bounds__(1024) k_geqr2(double* A, int m, int ncols,
   }
 }
 
+// ---------------------------------------------------- multi-WG panel QR
+// One LAUNCH factors a whole 128-column panel (vs k_geqr2's one-WG column
+// loop at 0.7 TF, profiles/RESULTS.md): WG 0 factors 16-column sub-panels
+// held in LDS (fast column loop: norm reduce + rank-1 updates never leave
+// the CU), builds the 16x16 T block in LDS, then ALL workgroups apply the
+// block reflector to the remaining panel columns between agent-scope grid
+// barriers (MI355X_MICROARCH.md "Workgroup dispatch" release/acquire
+// forms). A two-segment row map skips the structurally-zero block of
+// stacked [R (upper-tri); B] TSQRT tiles, capping active rows at
+// top_block + B regardless of the panel offset.
+//
+// Row map: local row r < len0 -> global base0 + r ; else base1 + (r-len0).
+// LDS budget: rows * W * 8 <= 147456 (W=16 up to 1152 rows, W=8 to 2304).
+namespace {
+
+constexpr int QR_LDS_DOUBLES = 18432;  // 144 KiB panel + ~8 KiB reduction
+constexpr int QR_MAX_ROWS_W16 = 1152;
+constexpr int QR_MAX_ROWS_W8 = 2304;
+
+__device__ inline void qr_grid_barrier(int* cnt, int nwg, int bar_no) {
+  __syncthreads();
+  if (threadIdx.x == 0) {
+    __builtin_amdgcn_fence(__ATOMIC_RELEASE, "agent");
+    asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+    __hip_atomic_fetch_add(cnt, 1, __ATOMIC_RELAXED, __HIP_MEMORY_SCOPE_AGENT);
+    while (__hip_atomic_load(cnt, __ATOMIC_RELAXED,
+                             __HIP_MEMORY_SCOPE_AGENT) < bar_no * nwg)
+      __builtin_amdgcn_s_sleep(8);
+    __builtin_amdgcn_fence(__ATOMIC_ACQUIRE, "agent");
+  }
+  __syncthreads();
+}
+
+// A panel column's reflector value at local row r (unit-lower, in-place).
+__device__ inline double qr_vval(const double* A, int ld, int base0,
+                                 int len0, int base1, int r, int cloc,
+                                 int gcol) {
+  if (r < cloc) return 0.0;
+  if (r == cloc) return 1.0;
+  int gr = r < len0 ? base0 + r : base1 + (r - len0);
+  return A[(size_t)gcol * ld + gr];
+}
+
+__global__ void __launch_bounds__(1024) k_qr_panel_mw(
+    double* A, int ld, int pcol0,  // panel base column (global)
+    int base0, int len0, int base1, int len1,  // row segments
+    int pc,                                    // panel columns (<=128)
+    int W,                                     // sub-panel width (8|16)
+    double* tau, double* T16s,  // T16s: 16x16 per sub-panel scratch (8 of them)
+    int* cnt, int nwg) {
+  __shared__ double sp[QR_LDS_DOUBLES];
+  __shared__ double red[16 + 2];  // per-wave partials + {beta,scal}
+  __shared__ double wy[32];       // w (16) + y (16) for the apply phase
+  __shared__ double gt[512];      // G (16x16) + local T (16x16)
+  const int rows = len0 + len1;
+  const int tid = threadIdx.x;
+  const int wave = tid >> 6, lane = tid & 63;
+  const int wg = blockIdx.x;
+  int bar_no = 0;
+
+  for (int c0 = 0; c0 < pc; c0 += W) {
+    const int w = min(W, pc - c0);
+    double* T16 = T16s + (size_t)(c0 / W) * 16 * 16;
+    if (wg == 0) {
+      // ---- stage sub-panel columns [c0, c0+w) into LDS ----
+      for (int q = 0; q < w; q++) {
+        const int gcol = pcol0 + c0 + q;
+        for (int r = tid; r < rows; r += 1024) {
+          int gr = r < len0 ? base0 + r : base1 + (r - len0);
+          sp[(size_t)q * rows + r] = A[(size_t)gcol * ld + gr];
+        }
+      }
+      __syncthreads();
+      // ---- factor w columns in LDS ----
+      for (int j = 0; j < w; j++) {
+        const int d = c0 + j;  // local diagonal row of this column
+        double* col = sp + (size_t)j * rows;
+        double acc = 0;
+        for (int i = d + 1 + tid; i < rows; i += 1024) acc += col[i] * col[i];
+        for (int s = 32; s > 0; s >>= 1) acc += __shfl_down(acc, s);
+        if (lane == 0) red[wave] = acc;
+        __syncthreads();
+        if (tid == 0) {
+          double nrm2 = 0;
+          for (int v = 0; v < 16; v++) nrm2 += red[v];
+          double alpha = col[d];
+          if (nrm2 == 0.0) {
+            red[16] = alpha;
+            red[17] = 0.0;
+            tau[c0 + j] = 0.0;
+          } else {
+            double beta = -copysign(sqrt(alpha * alpha + nrm2), alpha);
+            red[16] = beta;
+            red[17] = 1.0 / (alpha - beta);
+            tau[c0 + j] = (beta - alpha) / beta;
+          }
+        }
+        __syncthreads();
+        const double tau_j = tau[c0 + j], scal = red[17];
+        if (tau_j != 0.0) {
+          for (int i = d + 1 + tid; i < rows; i += 1024) col[i] *= scal;
+          if (tid == 0) col[d] = red[16];
+        }
+        __syncthreads();
+        if (tau_j != 0.0 && wave > j && wave < w) {
+          // wave c updates LDS column c (one wave per trailing column)
+          double* cc = sp + (size_t)wave * rows;
+          double dot = (lane == 0) ? cc[d] : 0.0;
+          for (int i = d + 1 + lane; i < rows; i += 64)
+            dot += col[i] * cc[i];
+          for (int s = 32; s > 0; s >>= 1) dot += __shfl_down(dot, s);
+          dot = __shfl(dot, 0);
+          double wj = tau_j * dot;
+          if (lane == 0) cc[d] -= wj;
+          for (int i = d + 1 + lane; i < rows; i += 64)
+            cc[i] -= wj * col[i];
+        }
+        __syncthreads();
+      }
+      // ---- write back the factored sub-panel ----
+      for (int q = 0; q < w; q++) {
+        const int gcol = pcol0 + c0 + q;
+        for (int r = tid; r < rows; r += 1024) {
+          int gr = r < len0 ? base0 + r : base1 + (r - len0);
+          A[(size_t)gcol * ld + gr] = sp[(size_t)q * rows + r];
+        }
+      }
+      __syncthreads();
+      // ---- T16: G = V^T V from LDS, then the upper-tri recurrence ----
+      // wave q computes G(0:q, q) (dots of unit-lower columns)
+      if (wave < w) {
+        for (int s = 0; s < wave; s++) {
+          const int ds = c0 + s, dq = c0 + wave;
+          const double* cs = sp + (size_t)s * rows;
+          const double* cq = sp + (size_t)wave * rows;
+          // v_s has 1 at ds, entries below; v_q starts at dq > ds
+          double dot = 0.0;
+          for (int i = dq + lane; i < rows; i += 64) {
+            double as = i == ds ? 1.0 : cs[i];
+            double aq = i == dq ? 1.0 : cq[i];
+            dot += as * aq;
+          }
+          for (int sh = 32; sh > 0; sh >>= 1) dot += __shfl_down(dot, sh);
+          if (lane == 0) gt[s * 16 + wave] = dot;  // G(s, wave)
+        }
+      }
+      __syncthreads();
+      // recurrence on one wave: T(0:j,j) = -tau_j * T(0:j,0:j) * g_j
+      if (wave == 0) {
+        double* Tl = gt + 256;   // 16x16 col-major
+        const double* G = gt;    // G(s, q) at gt[s*16+q]... stored G[s*16+wave]
+        for (int j = 0; j < w; j++) {
+          double tj = tau[c0 + j];
+          double s = 0;
+          if (lane < j) {
+            for (int q = lane; q < j; q++)
+              s += Tl[q * 16 + lane] * G[q * 16 + j];
+            s *= -tj;
+          }
+          if (lane < j) Tl[j * 16 + lane] = s;
+          if (lane == j) Tl[j * 16 + j] = tj;
+        }
+        if (lane < 16)
+          for (int j = 0; j < 16; j++)
+            T16[j * 16 + lane] =
+                (j < w && lane <= j) ? Tl[j * 16 + lane] : 0.0;
+      }
+      __syncthreads();
+    }
+    qr_grid_barrier(cnt, nwg, ++bar_no);
+    // ---- all WGs: apply (I - V T V^T)^T to panel cols [c0+w, pc) ----
+    const int rest0 = c0 + w;
+    for (int c = rest0 + wg; c < pc; c += nwg) {
+      const int gcol = pcol0 + c;
+      // w16 = V^T C(:,c): wave q computes dot q
+      if (wave < w) {
+        const int cloc = c0 + wave, vcol = pcol0 + c0 + wave;
+        double dot = 0;
+        for (int r = cloc + lane; r < rows; r += 64) {
+          double v = qr_vval(A, ld, base0, len0, base1, r, cloc, vcol);
+          int gr = r < len0 ? base0 + r : base1 + (r - len0);
+          dot += v * A[(size_t)gcol * ld + gr];
+        }
+        for (int s = 32; s > 0; s >>= 1) dot += __shfl_down(dot, s);
+        if (lane == 0) wy[wave] = dot;
+      }
+      __syncthreads();
+      // y = T^T w (T16 upper-triangular col-major)
+      if (wave == 0 && lane < w) {
+        double s = 0;
+        for (int q = 0; q <= lane; q++) s += T16[lane * 16 + q] * wy[q];
+        wy[16 + lane] = s;
+      }
+      __syncthreads();
+      // C(:,c) -= V y
+      for (int r = c0 + tid; r < rows; r += 1024) {
+        int gr = r < len0 ? base0 + r : base1 + (r - len0);
+        double s = 0;
+        for (int q = 0; q < w; q++) {
+          int cloc = c0 + q;
+          double v = qr_vval(A, ld, base0, len0, base1, r, cloc,
+                             pcol0 + cloc);
+          s += v * wy[16 + q];
+        }
+        A[(size_t)gcol * ld + gr] -= s;
+      }
+      __syncthreads();
+    }
+    qr_grid_barrier(cnt, nwg, ++bar_no);
+  }
+}
+
+}  // namespace
+
 // T diag blocks: for block b, columns j: T(0:j,j) = -tau_j T(0:j,0:j) g
 // where g = G(blk rows, j) restricted to the block. G is V^T V of the
 // unit-lower V; T upper-triangular, T(j,j) = tau_j.
@@ -380,20 +594,43 @@ __global__ void k_larft_diag(const double* G, int ldg, const double* tau,
   }
 }
 
-// Full tile/stacked-panel QR: factor 128-wide panels with k_geqr2, apply
-// to the trailing columns with gemm-larfb, then build the full k x k T.
+// Full tile/stacked-panel QR: factor 128-wide panels with the multi-WG
+// panel kernel (k_geqr2 single-WG fallback for oversized rows), apply to
+// the trailing columns with gemm-larfb, then build the full k x k T.
+// ts_split > 0 marks a stacked [R (upper-tri, ts_split rows); B] tile:
+// the panel kernel then skips the structurally-zero block below R's
+// diagonal band, capping active rows at 128 + (m - ts_split).
 static void qr_factor_hand(GpuTaskCtx& g, double* A, int m, int k, int ld,
-                           double* T, int ldt, int slot0) {
+                           double* T, int ldt, int slot0, int ts_split = 0) {
   rocblas_handle h = qr_handle(g);
   double* tau = qr_scratch(g, slot0, (size_t)k * 8);
+  double* T16s = qr_scratch(g, slot0 + 9, (size_t)8 * 256 * 8);
+  int* cnt = (int*)qr_scratch(g, slot0 + 10, 256);
+  const int nwg = (int)param_int("qr_panel_wgs", 8);
   const double one = 1.0, zero = 0.0, mone = -1.0;
   PA_HIP_CHECK(hipMemsetAsync(T, 0, (size_t)ldt * k * 8, g.stream));
   for (int p = 0; p < k; p += 128) {
     int pc = std::min(128, k - p);
     double* panel = A + (size_t)p * ld + p;
     int prows = m - p;
-    hipLaunchKernelGGL(k_geqr2, dim3(1), dim3(1024), 0, g.stream, panel,
-                       prows, pc, ld, tau + p);
+    // row segments: dense panel, or triangular-top + dense-bottom
+    int base0 = p, len0 = prows, base1 = 0, len1 = 0;
+    if (ts_split > 0) {
+      len0 = std::min(pc, ts_split - p);
+      base1 = ts_split;
+      len1 = m - ts_split;
+    }
+    int rows = len0 + len1;
+    int W = rows <= QR_MAX_ROWS_W16 ? 16 : (rows <= QR_MAX_ROWS_W8 ? 8 : 0);
+    if (W) {
+      PA_HIP_CHECK(hipMemsetAsync(cnt, 0, sizeof(int), g.stream));
+      hipLaunchKernelGGL(k_qr_panel_mw, dim3(nwg), dim3(1024), 0, g.stream,
+                         A, ld, p, base0, len0, base1, len1, pc, W, tau + p,
+                         T16s, cnt, nwg);
+    } else {
+      hipLaunchKernelGGL(k_geqr2, dim3(1), dim3(1024), 0, g.stream, panel,
+                         prows, pc, ld, tau + p);
+    }
     int rest = k - p - pc;
     if (rest > 0) {
       // T128 for this panel from G128 = V^T V
@@ -519,7 +756,7 @@ static void gpu_tsqrt(Task& t, GpuTaskCtx& g) {
   static const bool use_rocsolver =
       param_str("chore_qr", "rocsolver") == "rocsolver";
   if (!use_rocsolver) {
-    qr_factor_hand(g, V2, 2 * nb, nb, 2 * nb, T1, ld, 0);
+    qr_factor_hand(g, V2, 2 * nb, nb, 2 * nb, T1, ld, 0, /*ts_split=*/nb);
   } else {
     double* tau = qr_scratch(g, 0, (size_t)nb * 8);
     PA_HIP_CHECK(hipMemsetAsync(T1, 0, t.flows[3].data->bytes, g.stream));
@@ -562,12 +799,15 @@ static TaskClass make_qr_tc(const char* name, void (*cpu)(Task&),
 }
 
 TaskClass& tc_geqrt() {
-  // Panel factorizations host-sync inside rocSOLVER: run them on worker
-  // threads (gpu_blocking) so the manager keeps the trailing updates
-  // flowing. PARSEC_MCA_qr_blocking_panels=0 restores manager execution.
+  // rocSOLVER panel factorizations host-sync internally: run them on
+  // worker threads (gpu_blocking) so the manager keeps the trailing
+  // updates flowing. The hand panel path is fully device-side (multi-WG
+  // panel kernel), so it pipelines through the manager like any chore.
+  // PARSEC_MCA_qr_blocking_panels=0 restores manager execution.
   static TaskClass tc = [] {
     TaskClass c = make_qr_tc("geqrt", cpu_geqrt, gpu_geqrt, 10);
-    c.gpu_blocking = param_int("qr_blocking_panels", 1) != 0;
+    bool rocs = param_str("chore_qr", "rocsolver") == "rocsolver";
+    c.gpu_blocking = rocs && param_int("qr_blocking_panels", 1) != 0;
     return c;
   }();
   return tc;
@@ -579,7 +819,8 @@ TaskClass& tc_unmqr() {
 TaskClass& tc_tsqrt() {
   static TaskClass tc = [] {
     TaskClass c = make_qr_tc("tsqrt", cpu_tsqrt, gpu_tsqrt, 12);
-    c.gpu_blocking = param_int("qr_blocking_panels", 1) != 0;
+    bool rocs = param_str("chore_qr", "rocsolver") == "rocsolver";
+    c.gpu_blocking = rocs && param_int("qr_blocking_panels", 1) != 0;
     return c;
   }();
   return tc;

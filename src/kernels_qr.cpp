@@ -202,6 +202,17 @@ static void cpu_tsmqr(Task& t) {
 // ------------------------------------------------------------ GPU chores
 namespace {
 
+__global__ void k_qr_fill(double* p, size_t n, uint32_t seed) {
+  size_t i = (size_t)blockIdx.x * blockDim.x + threadIdx.x;
+  for (; i < n; i += (size_t)gridDim.x * blockDim.x) {
+    uint64_t h = i * 0x9E3779B97F4A7C15ull + seed;
+    h ^= h >> 13;
+    h *= 0x9E3779B97F4A7C15ull;
+    h ^= h >> 32;
+    p[i] = (double)(h & 0xFFFFFF) / (double)0x1000000 - 0.5;
+  }
+}
+
 rocblas_handle qr_handle(GpuTaskCtx& g) {
   static thread_local std::map<void*, rocblas_handle> handles;
   rocblas_handle& h = handles[(void*)g.stream];
@@ -221,7 +232,11 @@ double* qr_scratch(GpuTaskCtx& g, int slot, size_t bytes) {
     // the old buffer may still be referenced by earlier kernels on this
     // stream: return it to the pool only after this task retires
     if (e.first) g.deferred_frees->emplace_back(e.first, e.second);
-    e.first = g.engine->dev_alloc(bytes);
+    if (g.engine) {
+      e.first = g.engine->dev_alloc(bytes);
+    } else {
+      PA_HIP_CHECK(hipMalloc(&e.first, bytes));  // standalone micro-bench
+    }
     e.second = bytes;
   }
   return (double*)e.first;
@@ -400,80 +415,118 @@ __device__ inline double qr_vval(const double* A, int ld, int base0,
 
 __global__ void __launch_bounds__(1024) k_qr_panel_mw(
     double* A, int ld, int pcol0,  // panel base column (global)
-    int base0, int len0, int base1, int len1,  // row segments
-    int pc,                                    // panel columns (<=128)
-    int W,                                     // sub-panel width (8|16)
-    double* tau, double* T16s,  // T16s: 16x16 per sub-panel scratch (8 of them)
+    int base0, int len0, int base1, int len1,  // reflector row segments
+    int pc,     // total columns from pcol0 (factor + apply targets)
+    int fcols,  // columns to FACTOR (<=128)
+    int W,      // sub-panel width (8|16)
+    double* tau, double* T16s,  // T16s: 16x16 per sub-panel scratch
     int* cnt, int nwg) {
-  __shared__ double sp[QR_LDS_DOUBLES];
-  __shared__ double red[16 + 2];  // per-wave partials + {beta,scal}
-  __shared__ double wy[32];       // w (16) + y (16) for the apply phase
-  __shared__ double gt[512];      // G (16x16) + local T (16x16)
+  // One launch: WG 0 factors 16-column sub-panels in LDS (one barrier per
+  // column: the rank-1 update of column j piggybacks column j+1's norm and
+  // reflector scalars; columns stay unscaled until a single scale pass),
+  // then EVERY WG caches the scaled V image in its own LDS and applies
+  // (I - V T V^T)^T to a slice of all remaining tile columns — no
+  // per-panel T-build or larfb dgemms on the stream at all.
+  __shared__ double sp[QR_LDS_DOUBLES];  // sub-panel / V image
+  __shared__ double red[16];             // norm partials (staging)
+  __shared__ double bcs[3 * 16];         // per-col {vd, beta, tfac}
+  __shared__ double tl[16];              // per-col tau (LDS copy)
+  __shared__ double wy[32];              // w (16) + y (16) in the apply
+  __shared__ double gt[512];             // G (16x16) + local T (16x16)
   const int rows = len0 + len1;
   const int tid = threadIdx.x;
   const int wave = tid >> 6, lane = tid & 63;
   const int wg = blockIdx.x;
   int bar_no = 0;
 
-  for (int c0 = 0; c0 < pc; c0 += W) {
-    const int w = min(W, pc - c0);
+  for (int c0 = 0; c0 < fcols; c0 += W) {
+    const int w = min(W, fcols - c0);
     double* T16 = T16s + (size_t)(c0 / W) * 16 * 16;
     if (wg == 0) {
-      // ---- stage sub-panel columns [c0, c0+w) into LDS ----
+      // ---- stage sub-panel columns [c0, c0+w) into LDS; accumulate the
+      // first column's below-diagonal norm on the fly ----
+      double acc0 = 0;
       for (int q = 0; q < w; q++) {
         const int gcol = pcol0 + c0 + q;
         for (int r = tid; r < rows; r += 1024) {
           int gr = r < len0 ? base0 + r : base1 + (r - len0);
-          sp[(size_t)q * rows + r] = A[(size_t)gcol * ld + gr];
+          double v = A[(size_t)gcol * ld + gr];
+          sp[(size_t)q * rows + r] = v;
+          if (q == 0 && r > c0) acc0 += v * v;
         }
       }
+      for (int s = 32; s > 0; s >>= 1) acc0 += __shfl_down(acc0, s);
+      if (lane == 0) red[wave] = acc0;
       __syncthreads();
-      // ---- factor w columns in LDS ----
+      if (tid == 0) {
+        double nrm2 = 0;
+        for (int v = 0; v < 16; v++) nrm2 += red[v];
+        double alpha = sp[c0];
+        double beta = alpha, vd = 1.0, tfac = 0.0, tj = 0.0;
+        if (nrm2 != 0.0) {
+          beta = -copysign(sqrt(alpha * alpha + nrm2), alpha);
+          vd = alpha - beta;
+          tj = (beta - alpha) / beta;
+          tfac = tj / (vd * vd);
+        }
+        bcs[0] = vd;
+        bcs[1] = beta;
+        bcs[2] = tfac;
+        tl[0] = tj;
+        tau[c0] = tj;
+      }
+      __syncthreads();
+      // ---- factor: one barrier per column ----
       for (int j = 0; j < w; j++) {
-        const int d = c0 + j;  // local diagonal row of this column
-        double* col = sp + (size_t)j * rows;
-        double acc = 0;
-        for (int i = d + 1 + tid; i < rows; i += 1024) acc += col[i] * col[i];
-        for (int s = 32; s > 0; s >>= 1) acc += __shfl_down(acc, s);
-        if (lane == 0) red[wave] = acc;
-        __syncthreads();
-        if (tid == 0) {
-          double nrm2 = 0;
-          for (int v = 0; v < 16; v++) nrm2 += red[v];
-          double alpha = col[d];
-          if (nrm2 == 0.0) {
-            red[16] = alpha;
-            red[17] = 0.0;
-            tau[c0 + j] = 0.0;
-          } else {
-            double beta = -copysign(sqrt(alpha * alpha + nrm2), alpha);
-            red[16] = beta;
-            red[17] = 1.0 / (alpha - beta);
-            tau[c0 + j] = (beta - alpha) / beta;
-          }
-        }
-        __syncthreads();
-        const double tau_j = tau[c0 + j], scal = red[17];
-        if (tau_j != 0.0) {
-          for (int i = d + 1 + tid; i < rows; i += 1024) col[i] *= scal;
-          if (tid == 0) col[d] = red[16];
-        }
-        __syncthreads();
-        if (tau_j != 0.0 && wave > j && wave < w) {
-          // wave c updates LDS column c (one wave per trailing column)
+        const int d = c0 + j;
+        const double vd = bcs[3 * j], tfac = bcs[3 * j + 2];
+        if (wave > j && wave < w) {
+          const double* col = sp + (size_t)j * rows;
           double* cc = sp + (size_t)wave * rows;
-          double dot = (lane == 0) ? cc[d] : 0.0;
+          double dot = (lane == 0) ? vd * cc[d] : 0.0;
           for (int i = d + 1 + lane; i < rows; i += 64)
             dot += col[i] * cc[i];
           for (int s = 32; s > 0; s >>= 1) dot += __shfl_down(dot, s);
           dot = __shfl(dot, 0);
-          double wj = tau_j * dot;
-          if (lane == 0) cc[d] -= wj;
-          for (int i = d + 1 + lane; i < rows; i += 64)
-            cc[i] -= wj * col[i];
+          const double wj = tfac * dot;
+          if (lane == 0) cc[d] -= wj * vd;
+          double nacc = 0;
+          const bool mine = (wave == j + 1);
+          for (int i = d + 1 + lane; i < rows; i += 64) {
+            double nv = cc[i] - wj * col[i];
+            cc[i] = nv;
+            if (mine && i > d + 1) nacc += nv * nv;
+          }
+          if (mine) {
+            for (int s = 32; s > 0; s >>= 1) nacc += __shfl_down(nacc, s);
+            if (lane == 0) {
+              double alpha = cc[d + 1];
+              double beta = alpha, nvd = 1.0, tfac2 = 0.0, tj = 0.0;
+              if (nacc != 0.0) {
+                beta = -copysign(sqrt(alpha * alpha + nacc), alpha);
+                nvd = alpha - beta;
+                tj = (beta - alpha) / beta;
+                tfac2 = tj / (nvd * nvd);
+              }
+              bcs[3 * (j + 1)] = nvd;
+              bcs[3 * (j + 1) + 1] = beta;
+              bcs[3 * (j + 1) + 2] = tfac2;
+              tl[j + 1] = tj;
+              tau[c0 + j + 1] = tj;
+            }
+          }
         }
         __syncthreads();
       }
+      // ---- scale pass: normalize each column's reflector, set diag ----
+      if (wave < w) {
+        const int d = c0 + wave;
+        double* col = sp + (size_t)wave * rows;
+        const double inv = 1.0 / bcs[3 * wave];
+        for (int i = d + 1 + lane; i < rows; i += 64) col[i] *= inv;
+        if (lane == 0) col[d] = bcs[3 * wave + 1];
+      }
+      __syncthreads();
       // ---- write back the factored sub-panel ----
       for (int q = 0; q < w; q++) {
         const int gcol = pcol0 + c0 + q;
@@ -484,13 +537,11 @@ __global__ void __launch_bounds__(1024) k_qr_panel_mw(
       }
       __syncthreads();
       // ---- T16: G = V^T V from LDS, then the upper-tri recurrence ----
-      // wave q computes G(0:q, q) (dots of unit-lower columns)
       if (wave < w) {
         for (int s = 0; s < wave; s++) {
           const int ds = c0 + s, dq = c0 + wave;
           const double* cs = sp + (size_t)s * rows;
           const double* cq = sp + (size_t)wave * rows;
-          // v_s has 1 at ds, entries below; v_q starts at dq > ds
           double dot = 0.0;
           for (int i = dq + lane; i < rows; i += 64) {
             double as = i == ds ? 1.0 : cs[i];
@@ -502,12 +553,11 @@ __global__ void __launch_bounds__(1024) k_qr_panel_mw(
         }
       }
       __syncthreads();
-      // recurrence on one wave: T(0:j,j) = -tau_j * T(0:j,0:j) * g_j
       if (wave == 0) {
-        double* Tl = gt + 256;   // 16x16 col-major
-        const double* G = gt;    // G(s, q) at gt[s*16+q]... stored G[s*16+wave]
+        double* Tl = gt + 256;
+        const double* G = gt;
         for (int j = 0; j < w; j++) {
-          double tj = tau[c0 + j];
+          double tj = tl[j];
           double s = 0;
           if (lane < j) {
             for (int q = lane; q < j; q++)
@@ -525,40 +575,46 @@ __global__ void __launch_bounds__(1024) k_qr_panel_mw(
       __syncthreads();
     }
     qr_grid_barrier(cnt, nwg, ++bar_no);
-    // ---- all WGs: apply (I - V T V^T)^T to panel cols [c0+w, pc) ----
-    const int rest0 = c0 + w;
-    for (int c = rest0 + wg; c < pc; c += nwg) {
+    // ---- every WG: cache the scaled V image (unit-lower) in LDS, then
+    // apply (I - V T V^T)^T to its slice of columns [c0+w, pc) ----
+    for (int q = 0; q < w; q++) {
+      const int gcol = pcol0 + c0 + q, cloc = c0 + q;
+      for (int r = tid; r < rows; r += 1024) {
+        double v;
+        if (r < cloc) v = 0.0;
+        else if (r == cloc) v = 1.0;
+        else {
+          int gr = r < len0 ? base0 + r : base1 + (r - len0);
+          v = A[(size_t)gcol * ld + gr];
+        }
+        sp[(size_t)q * rows + r] = v;
+      }
+    }
+    __syncthreads();
+    for (int c = c0 + w + wg; c < pc; c += nwg) {
       const int gcol = pcol0 + c;
-      // w16 = V^T C(:,c): wave q computes dot q
       if (wave < w) {
-        const int cloc = c0 + wave, vcol = pcol0 + c0 + wave;
+        const int cloc = c0 + wave;
+        const double* v = sp + (size_t)wave * rows;
         double dot = 0;
         for (int r = cloc + lane; r < rows; r += 64) {
-          double v = qr_vval(A, ld, base0, len0, base1, r, cloc, vcol);
           int gr = r < len0 ? base0 + r : base1 + (r - len0);
-          dot += v * A[(size_t)gcol * ld + gr];
+          dot += v[r] * A[(size_t)gcol * ld + gr];
         }
         for (int s = 32; s > 0; s >>= 1) dot += __shfl_down(dot, s);
         if (lane == 0) wy[wave] = dot;
       }
       __syncthreads();
-      // y = T^T w (T16 upper-triangular col-major)
       if (wave == 0 && lane < w) {
         double s = 0;
         for (int q = 0; q <= lane; q++) s += T16[lane * 16 + q] * wy[q];
         wy[16 + lane] = s;
       }
       __syncthreads();
-      // C(:,c) -= V y
       for (int r = c0 + tid; r < rows; r += 1024) {
         int gr = r < len0 ? base0 + r : base1 + (r - len0);
         double s = 0;
-        for (int q = 0; q < w; q++) {
-          int cloc = c0 + q;
-          double v = qr_vval(A, ld, base0, len0, base1, r, cloc,
-                             pcol0 + cloc);
-          s += v * wy[16 + q];
-        }
+        for (int q = 0; q < w; q++) s += sp[(size_t)q * rows + r] * wy[16 + q];
         A[(size_t)gcol * ld + gr] -= s;
       }
       __syncthreads();
@@ -623,14 +679,16 @@ static void qr_factor_hand(GpuTaskCtx& g, double* A, int m, int k, int ld,
     int rows = len0 + len1;
     int W = rows <= QR_MAX_ROWS_W16 ? 16 : (rows <= QR_MAX_ROWS_W8 ? 8 : 0);
     if (W) {
+      // the panel kernel both factors and applies to ALL remaining tile
+      // columns — no per-panel T-build / larfb dgemms on the stream
       PA_HIP_CHECK(hipMemsetAsync(cnt, 0, sizeof(int), g.stream));
       hipLaunchKernelGGL(k_qr_panel_mw, dim3(nwg), dim3(1024), 0, g.stream,
-                         A, ld, p, base0, len0, base1, len1, pc, W, tau + p,
-                         T16s, cnt, nwg);
-    } else {
-      hipLaunchKernelGGL(k_geqr2, dim3(1), dim3(1024), 0, g.stream, panel,
-                         prows, pc, ld, tau + p);
+                         A, ld, p, base0, len0, base1, len1, k - p, pc, W,
+                         tau + p, T16s, cnt, nwg);
+      continue;
     }
+    hipLaunchKernelGGL(k_geqr2, dim3(1), dim3(1024), 0, g.stream, panel,
+                       prows, pc, ld, tau + p);
     int rest = k - p - pc;
     if (rest > 0) {
       // T128 for this panel from G128 = V^T V
@@ -945,6 +1003,60 @@ void insert_geqrf(Dtd& tp, TiledMatrix& A) {
       }
     }
   }
+}
+
+// Standalone micro-benchmark of the TS/tile panel factorization path:
+// mode 0 = full qr_factor_hand, 1 = panel kernels only, 2 = rocsolver
+// dgeqrf. Returns seconds for `iters` factorizations of an m x k tile.
+double bench_qr_factor(int m, int k, int ts_split, int iters, int mode) {
+  double *dA, *dT;
+  PA_HIP_CHECK(hipMalloc(&dA, (size_t)m * k * 8));
+  PA_HIP_CHECK(hipMalloc(&dT, (size_t)k * k * 8));
+  hipStream_t s;
+  PA_HIP_CHECK(hipStreamCreateWithFlags(&s, hipStreamNonBlocking));
+  std::vector<std::pair<void*, size_t>> defer;
+  GpuTaskCtx g{s, 0, nullptr, &defer};
+  double* tau = qr_scratch(g, 0, (size_t)k * 8);
+  double* T16s = qr_scratch(g, 9, (size_t)8 * 256 * 8);
+  int* cnt = (int*)qr_scratch(g, 10, 256);
+  const int nwg = (int)param_int("qr_panel_wgs", 8);
+  auto run = [&] {
+    hipLaunchKernelGGL(k_qr_fill, dim3(2048), dim3(256), 0, s, dA,
+                       (size_t)m * k, 7);
+    if (mode == 0) {
+      qr_factor_hand(g, dA, m, k, m, dT, k, 0, ts_split);
+    } else if (mode == 1) {
+      for (int p = 0; p < k; p += 128) {
+        int pc = std::min(128, k - p);
+        int base0 = p, len0 = m - p, base1 = 0, len1 = 0;
+        if (ts_split > 0) {
+          len0 = std::min(pc, ts_split - p);
+          base1 = ts_split;
+          len1 = m - ts_split;
+        }
+        int rows = len0 + len1;
+        int W = rows <= QR_MAX_ROWS_W16 ? 16
+                                        : (rows <= QR_MAX_ROWS_W8 ? 8 : 0);
+        PA_CHECK(W, "bench: rows too large");
+        PA_HIP_CHECK(hipMemsetAsync(cnt, 0, sizeof(int), s));
+        hipLaunchKernelGGL(k_qr_panel_mw, dim3(nwg), dim3(1024), 0, s, dA,
+                           m, p, base0, len0, base1, len1, k - p, pc, W,
+                           tau + p, T16s, cnt, nwg);
+      }
+    } else {
+      PA_CHECK(rocsolver_dgeqrf(qr_handle(g), m, k, dA, m, tau) ==
+               rocblas_status_success);
+    }
+  };
+  run();
+  PA_HIP_CHECK(hipStreamSynchronize(s));
+  double t0 = now_s();
+  for (int i = 0; i < iters; i++) run();
+  PA_HIP_CHECK(hipStreamSynchronize(s));
+  double dt = now_s() - t0;
+  PA_HIP_CHECK(hipFree(dA));
+  PA_HIP_CHECK(hipFree(dT));
+  return dt;
 }
 
 }  // namespace pa

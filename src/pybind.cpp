@@ -28,6 +28,7 @@ void test_potf2_hip(double*, int);
 double bench_dgemm_hip(int, int, int, int, int);
 double bench_dgemm_rocblas(int, int, int, int);
 double bench_gemm_bf16(int, int, int, int);
+double bench_qr_factor(int, int, int, int, int);
 void test_gemm_bf16_hip(int, int, int, const uint16_t*, const uint16_t*,
                         float*);
 }  // namespace pa
@@ -347,6 +348,9 @@ PYBIND11_MODULE(_core, m) {
   m.attr("ACCESS_OUT") = (int)ACCESS_OUT;
   m.attr("ACCESS_INOUT") = (int)ACCESS_INOUT;
 
+  m.def("bench_qr_factor", &bench_qr_factor, py::arg("m"), py::arg("k"),
+        py::arg("ts_split"), py::arg("iters"), py::arg("mode"),
+        py::call_guard<py::gil_scoped_release>());
   m.def("insert_spd_fill", &insert_spd_fill, py::arg("tp"), py::arg("A"),
         py::arg("seed") = 42u, py::call_guard<py::gil_scoped_release>());
   m.def("insert_full_fill", &insert_full_fill, py::arg("tp"), py::arg("A"),

@@ -421,30 +421,50 @@ __global__ void __launch_bounds__(1024) k_qr_panel_mw(
     int W,      // sub-panel width (8|16)
     double* tau, double* T16s,  // T16s: 16x16 per sub-panel scratch
     int* cnt, int nwg) {
-  // One launch: WG 0 factors 16-column sub-panels in LDS (one barrier per
-  // column: the rank-1 update of column j piggybacks column j+1's norm and
-  // reflector scalars; columns stay unscaled until a single scale pass),
-  // then EVERY WG caches the scaled V image in its own LDS and applies
-  // (I - V T V^T)^T to a slice of all remaining tile columns — no
-  // per-panel T-build or larfb dgemms on the stream at all.
+  // Producer/consumer panel pipeline (no grid barriers):
+  //  - WG 0 factors 16-column sub-panels in LDS (one barrier per column,
+  //    piggybacked norms, deferred scaling), builds T16 with a parallel
+  //    G = V^T V, then PUBLISHES the sub-panel (agent release + counter).
+  //  - helper pool A (<=7 WGs) applies each published block reflector to
+  //    the remaining FACTOR columns and acks; WG 0 only stages sub-panel
+  //    s+1 after ack(s), so the factor chain never waits on the trailing
+  //    matrix.
+  //  - pool B (the rest) streams the same applies over the trailing
+  //    columns asynchronously, in publication order (each column has a
+  //    fixed owner, so per-column apply order is the program order).
+  // Memory protocol per MI355X_MICROARCH.md "Workgroup dispatch": plain
+  // stores -> release fence + vmcnt(0) asm -> relaxed counter; consumers
+  // poll relaxed, acquire-fence once, then plain loads.
   __shared__ double sp[QR_LDS_DOUBLES];  // sub-panel / V image
   __shared__ double red[16];             // norm partials (staging)
   __shared__ double bcs[3 * 16];         // per-col {vd, beta, tfac}
   __shared__ double tl[16];              // per-col tau (LDS copy)
-  __shared__ double wy[32];              // w (16) + y (16) in the apply
+  __shared__ double wy[2 * 64];          // dots + y for 4 columns
   __shared__ double gt[512];             // G (16x16) + local T (16x16)
   const int rows = len0 + len1;
   const int tid = threadIdx.x;
   const int wave = tid >> 6, lane = tid & 63;
   const int wg = blockIdx.x;
-  int bar_no = 0;
+  const int ns = (fcols + W - 1) / W;
+  const int nA = min(nwg - 1, 7);
+  const int nB = nwg - 1 - nA;
 
-  for (int c0 = 0; c0 < fcols; c0 += W) {
-    const int w = min(W, fcols - c0);
-    double* T16 = T16s + (size_t)(c0 / W) * 16 * 16;
-    if (wg == 0) {
-      // ---- stage sub-panel columns [c0, c0+w) into LDS; accumulate the
-      // first column's below-diagonal norm on the fly ----
+  if (wg == 0) {
+    for (int si = 0; si < ns; si++) {
+      const int c0 = si * W;
+      const int w = min(W, fcols - c0);
+      double* T16 = T16s + (size_t)si * 256;
+      if (si > 0) {
+        // wait for pool A's within-panel applies of sub-panel si-1
+        if (tid == 0) {
+          while (__hip_atomic_load(&cnt[1], __ATOMIC_RELAXED,
+                                   __HIP_MEMORY_SCOPE_AGENT) < si * nA)
+            __builtin_amdgcn_s_sleep(2);
+          __builtin_amdgcn_fence(__ATOMIC_ACQUIRE, "agent");
+        }
+        __syncthreads();
+      }
+      // ---- stage sub-panel columns into LDS (+ first column's norm) ----
       double acc0 = 0;
       for (int q = 0; q < w; q++) {
         const int gcol = pcol0 + c0 + q;
@@ -455,7 +475,7 @@ __global__ void __launch_bounds__(1024) k_qr_panel_mw(
           if (q == 0 && r > c0) acc0 += v * v;
         }
       }
-      for (int s = 32; s > 0; s >>= 1) acc0 += __shfl_down(acc0, s);
+      for (int sh = 32; sh > 0; sh >>= 1) acc0 += __shfl_down(acc0, sh);
       if (lane == 0) red[wave] = acc0;
       __syncthreads();
       if (tid == 0) {
@@ -486,7 +506,7 @@ __global__ void __launch_bounds__(1024) k_qr_panel_mw(
           double dot = (lane == 0) ? vd * cc[d] : 0.0;
           for (int i = d + 1 + lane; i < rows; i += 64)
             dot += col[i] * cc[i];
-          for (int s = 32; s > 0; s >>= 1) dot += __shfl_down(dot, s);
+          for (int sh = 32; sh > 0; sh >>= 1) dot += __shfl_down(dot, sh);
           dot = __shfl(dot, 0);
           const double wj = tfac * dot;
           if (lane == 0) cc[d] -= wj * vd;
@@ -498,7 +518,7 @@ __global__ void __launch_bounds__(1024) k_qr_panel_mw(
             if (mine && i > d + 1) nacc += nv * nv;
           }
           if (mine) {
-            for (int s = 32; s > 0; s >>= 1) nacc += __shfl_down(nacc, s);
+            for (int sh = 32; sh > 0; sh >>= 1) nacc += __shfl_down(nacc, sh);
             if (lane == 0) {
               double alpha = cc[d + 1];
               double beta = alpha, nvd = 1.0, tfac2 = 0.0, tj = 0.0;
@@ -518,7 +538,7 @@ __global__ void __launch_bounds__(1024) k_qr_panel_mw(
         }
         __syncthreads();
       }
-      // ---- scale pass: normalize each column's reflector, set diag ----
+      // ---- scale pass ----
       if (wave < w) {
         const int d = c0 + wave;
         double* col = sp + (size_t)wave * rows;
@@ -527,7 +547,7 @@ __global__ void __launch_bounds__(1024) k_qr_panel_mw(
         if (lane == 0) col[d] = bcs[3 * wave + 1];
       }
       __syncthreads();
-      // ---- write back the factored sub-panel ----
+      // ---- write back ----
       for (int q = 0; q < w; q++) {
         const int gcol = pcol0 + c0 + q;
         for (int r = tid; r < rows; r += 1024) {
@@ -536,20 +556,31 @@ __global__ void __launch_bounds__(1024) k_qr_panel_mw(
         }
       }
       __syncthreads();
-      // ---- T16: G = V^T V from LDS, then the upper-tri recurrence ----
-      if (wave < w) {
-        for (int s = 0; s < wave; s++) {
-          const int ds = c0 + s, dq = c0 + wave;
-          const double* cs = sp + (size_t)s * rows;
-          const double* cq = sp + (size_t)wave * rows;
+      // ---- G = V^T V: 64 units of 16 lanes over the (s<q) pairs ----
+      {
+        const int unit = wave * 4 + (lane >> 4), l16 = lane & 15;
+        const int npairs = w * (w - 1) / 2;
+        for (int idx = unit; idx < npairs; idx += 64) {
+          // decode idx -> (sq pair): q = row in triangle
+          int q = 1;
+          int rem = idx;
+          while (rem >= q) {
+            rem -= q;
+            q++;
+          }
+          int sc = rem;  // 0 <= sc < q
+          const int ds = c0 + sc, dq = c0 + q;
+          const double* cs = sp + (size_t)sc * rows;
+          const double* cq = sp + (size_t)q * rows;
           double dot = 0.0;
-          for (int i = dq + lane; i < rows; i += 64) {
+          for (int i = dq + l16; i < rows; i += 16) {
             double as = i == ds ? 1.0 : cs[i];
             double aq = i == dq ? 1.0 : cq[i];
             dot += as * aq;
           }
-          for (int sh = 32; sh > 0; sh >>= 1) dot += __shfl_down(dot, sh);
-          if (lane == 0) gt[s * 16 + wave] = dot;  // G(s, wave)
+          for (int sh = 8; sh > 0; sh >>= 1)
+            dot += __shfl_down(dot, sh, 16);
+          if (l16 == 0) gt[sc * 16 + q] = dot;  // G(sc, q)
         }
       }
       __syncthreads();
@@ -558,13 +589,13 @@ __global__ void __launch_bounds__(1024) k_qr_panel_mw(
         const double* G = gt;
         for (int j = 0; j < w; j++) {
           double tj = tl[j];
-          double s = 0;
+          double sacc = 0;
           if (lane < j) {
             for (int q = lane; q < j; q++)
-              s += Tl[q * 16 + lane] * G[q * 16 + j];
-            s *= -tj;
+              sacc += Tl[q * 16 + lane] * G[q * 16 + j];
+            sacc *= -tj;
           }
-          if (lane < j) Tl[j * 16 + lane] = s;
+          if (lane < j) Tl[j * 16 + lane] = sacc;
           if (lane == j) Tl[j * 16 + j] = tj;
         }
         if (lane < 16)
@@ -573,10 +604,31 @@ __global__ void __launch_bounds__(1024) k_qr_panel_mw(
                 (j < w && lane <= j) ? Tl[j * 16 + lane] : 0.0;
       }
       __syncthreads();
+      // ---- publish sub-panel si ----
+      if (tid == 0) {
+        __builtin_amdgcn_fence(__ATOMIC_RELEASE, "agent");
+        asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+        __hip_atomic_store(&cnt[0], si + 1, __ATOMIC_RELAXED,
+                           __HIP_MEMORY_SCOPE_AGENT);
+      }
     }
-    qr_grid_barrier(cnt, nwg, ++bar_no);
-    // ---- every WG: cache the scaled V image (unit-lower) in LDS, then
-    // apply (I - V T V^T)^T to its slice of columns [c0+w, pc) ----
+    return;
+  }
+
+  // ---------------- helper workgroups ----------------
+  const bool poolA = wg <= nA;
+  for (int si = 0; si < ns; si++) {
+    const int c0 = si * W;
+    const int w = min(W, fcols - c0);
+    const double* T16 = T16s + (size_t)si * 256;
+    if (tid == 0) {
+      while (__hip_atomic_load(&cnt[0], __ATOMIC_RELAXED,
+                               __HIP_MEMORY_SCOPE_AGENT) < si + 1)
+        __builtin_amdgcn_s_sleep(2);
+      __builtin_amdgcn_fence(__ATOMIC_ACQUIRE, "agent");
+    }
+    __syncthreads();
+    // cache the scaled V image (unit-lower) in LDS
     for (int q = 0; q < w; q++) {
       const int gcol = pcol0 + c0 + q, cloc = c0 + q;
       for (int r = tid; r < rows; r += 1024) {
@@ -591,35 +643,71 @@ __global__ void __launch_bounds__(1024) k_qr_panel_mw(
       }
     }
     __syncthreads();
-    for (int c = c0 + w + wg; c < pc; c += nwg) {
-      const int gcol = pcol0 + c;
-      if (wave < w) {
+    // apply (I - V T V^T)^T to 4-column groups of my column set
+    auto apply4 = [&](int cg[4], int nc) {
+      const int ci = lane >> 4, l16 = lane & 15;  // unit = (q=wave, ci)
+      if (ci < nc) {
+        const int gcol = pcol0 + cg[ci];
         const int cloc = c0 + wave;
         const double* v = sp + (size_t)wave * rows;
         double dot = 0;
-        for (int r = cloc + lane; r < rows; r += 64) {
+        for (int r = cloc + l16; r < rows; r += 16) {
           int gr = r < len0 ? base0 + r : base1 + (r - len0);
           dot += v[r] * A[(size_t)gcol * ld + gr];
         }
-        for (int s = 32; s > 0; s >>= 1) dot += __shfl_down(dot, s);
-        if (lane == 0) wy[wave] = dot;
+        for (int sh = 8; sh > 0; sh >>= 1) dot += __shfl_down(dot, sh, 16);
+        if (l16 == 0) wy[ci * 16 + wave] = dot;
       }
       __syncthreads();
-      if (wave == 0 && lane < w) {
-        double s = 0;
-        for (int q = 0; q <= lane; q++) s += T16[lane * 16 + q] * wy[q];
-        wy[16 + lane] = s;
+      if (tid < 16 * nc) {
+        const int q = tid & 15, c = tid >> 4;
+        double sacc = 0;
+        for (int p2 = 0; p2 <= q; p2++)
+          sacc += T16[q * 16 + p2] * wy[c * 16 + p2];
+        wy[64 + c * 16 + q] = sacc;
       }
       __syncthreads();
-      for (int r = c0 + tid; r < rows; r += 1024) {
-        int gr = r < len0 ? base0 + r : base1 + (r - len0);
-        double s = 0;
-        for (int q = 0; q < w; q++) s += sp[(size_t)q * rows + r] * wy[16 + q];
-        A[(size_t)gcol * ld + gr] -= s;
+      {
+        const int c = tid >> 8, t2 = tid & 255;  // 4 cols x 256 threads
+        if (c < nc) {
+          const int gcol = pcol0 + cg[c];
+          for (int r = c0 + t2; r < rows; r += 256) {
+            int gr = r < len0 ? base0 + r : base1 + (r - len0);
+            double sacc = 0;
+            for (int q = 0; q < w; q++)
+              sacc += sp[(size_t)q * rows + r] * wy[64 + c * 16 + q];
+            A[(size_t)gcol * ld + gr] -= sacc;
+          }
+        }
       }
       __syncthreads();
+    };
+    auto apply_range = [&](int cbeg, int cend, int stride, int phase) {
+      int cg[4];
+      int nc = 0;
+      for (int c = cbeg + phase; c < cend; c += stride) {
+        cg[nc++] = c;
+        if (nc == 4) {
+          apply4(cg, 4);
+          nc = 0;
+        }
+      }
+      if (nc) apply4(cg, nc);
+    };
+    if (poolA) {
+      apply_range(c0 + w, fcols, nA, wg - 1);
+      __syncthreads();
+      if (tid == 0) {
+        __builtin_amdgcn_fence(__ATOMIC_RELEASE, "agent");
+        asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+        __hip_atomic_fetch_add(&cnt[1], 1, __ATOMIC_RELAXED,
+                               __HIP_MEMORY_SCOPE_AGENT);
+      }
+      if (nB == 0)  // no pool B: pool A also covers the trailing columns
+        apply_range(fcols, pc, nA, wg - 1);
+    } else {
+      apply_range(fcols, pc, nB, wg - 1 - nA);
     }
-    qr_grid_barrier(cnt, nwg, ++bar_no);
   }
 }
 
@@ -662,7 +750,7 @@ static void qr_factor_hand(GpuTaskCtx& g, double* A, int m, int k, int ld,
   double* tau = qr_scratch(g, slot0, (size_t)k * 8);
   double* T16s = qr_scratch(g, slot0 + 9, (size_t)8 * 256 * 8);
   int* cnt = (int*)qr_scratch(g, slot0 + 10, 256);
-  const int nwg = (int)param_int("qr_panel_wgs", 8);
+  const int nwg = std::max(2, (int)param_int("qr_panel_wgs", 32));
   const double one = 1.0, zero = 0.0, mone = -1.0;
   PA_HIP_CHECK(hipMemsetAsync(T, 0, (size_t)ldt * k * 8, g.stream));
   for (int p = 0; p < k; p += 128) {
@@ -681,7 +769,7 @@ static void qr_factor_hand(GpuTaskCtx& g, double* A, int m, int k, int ld,
     if (W) {
       // the panel kernel both factors and applies to ALL remaining tile
       // columns — no per-panel T-build / larfb dgemms on the stream
-      PA_HIP_CHECK(hipMemsetAsync(cnt, 0, sizeof(int), g.stream));
+      PA_HIP_CHECK(hipMemsetAsync(cnt, 0, 2 * sizeof(int), g.stream));
       hipLaunchKernelGGL(k_qr_panel_mw, dim3(nwg), dim3(1024), 0, g.stream,
                          A, ld, p, base0, len0, base1, len1, k - p, pc, W,
                          tau + p, T16s, cnt, nwg);
@@ -1019,13 +1107,13 @@ double bench_qr_factor(int m, int k, int ts_split, int iters, int mode) {
   double* tau = qr_scratch(g, 0, (size_t)k * 8);
   double* T16s = qr_scratch(g, 9, (size_t)8 * 256 * 8);
   int* cnt = (int*)qr_scratch(g, 10, 256);
-  const int nwg = (int)param_int("qr_panel_wgs", 8);
+  const int nwg = std::max(2, (int)param_int("qr_panel_wgs", 32));
   auto run = [&] {
     hipLaunchKernelGGL(k_qr_fill, dim3(2048), dim3(256), 0, s, dA,
                        (size_t)m * k, 7);
     if (mode == 0) {
       qr_factor_hand(g, dA, m, k, m, dT, k, 0, ts_split);
-    } else if (mode == 1) {
+    } else if (mode == 1 || mode >= 16) {
       for (int p = 0; p < k; p += 128) {
         int pc = std::min(128, k - p);
         int base0 = p, len0 = m - p, base1 = 0, len1 = 0;
@@ -1038,7 +1126,7 @@ double bench_qr_factor(int m, int k, int ts_split, int iters, int mode) {
         int W = rows <= QR_MAX_ROWS_W16 ? 16
                                         : (rows <= QR_MAX_ROWS_W8 ? 8 : 0);
         PA_CHECK(W, "bench: rows too large");
-        PA_HIP_CHECK(hipMemsetAsync(cnt, 0, sizeof(int), s));
+        PA_HIP_CHECK(hipMemsetAsync(cnt, 0, 2 * sizeof(int), s));
         hipLaunchKernelGGL(k_qr_panel_mw, dim3(nwg), dim3(1024), 0, s, dA,
                            m, p, base0, len0, base1, len1, k - p, pc, W,
                            tau + p, T16s, cnt, nwg);

@@ -420,7 +420,7 @@ __global__ void __launch_bounds__(1024) k_qr_panel_mw(
     int fcols,  // columns to FACTOR (<=128)
     int W,      // sub-panel width (8|16)
     double* tau, double* T16s,  // T16s: 16x16 per sub-panel scratch
-    int* cnt, int nwg) {
+    int* cnt, int nwg, int nA, unsigned long long* dbg) {
   // Producer/consumer panel pipeline (no grid barriers):
   //  - WG 0 factors 16-column sub-panels in LDS (one barrier per column,
   //    piggybacked norms, deferred scaling), builds T16 with a parallel
@@ -446,7 +446,6 @@ __global__ void __launch_bounds__(1024) k_qr_panel_mw(
   const int wave = tid >> 6, lane = tid & 63;
   const int wg = blockIdx.x;
   const int ns = (fcols + W - 1) / W;
-  const int nA = min(nwg - 1, 7);
   const int nB = nwg - 1 - nA;
 
   if (wg == 0) {
@@ -454,6 +453,7 @@ __global__ void __launch_bounds__(1024) k_qr_panel_mw(
       const int c0 = si * W;
       const int w = min(W, fcols - c0);
       double* T16 = T16s + (size_t)si * 256;
+      unsigned long long t0 = dbg ? __builtin_amdgcn_s_memrealtime() : 0;
       if (si > 0) {
         // wait for pool A's within-panel applies of sub-panel si-1
         if (tid == 0) {
@@ -464,15 +464,26 @@ __global__ void __launch_bounds__(1024) k_qr_panel_mw(
         }
         __syncthreads();
       }
+      if (dbg && tid == 0) {
+        unsigned long long t1 = __builtin_amdgcn_s_memrealtime();
+        dbg[0] += t1 - t0;
+        t0 = t1;
+      }
       // ---- stage sub-panel columns into LDS (+ first column's norm) ----
       double acc0 = 0;
       for (int q = 0; q < w; q++) {
-        const int gcol = pcol0 + c0 + q;
-        for (int r = tid; r < rows; r += 1024) {
-          int gr = r < len0 ? base0 + r : base1 + (r - len0);
-          double v = A[(size_t)gcol * ld + gr];
-          sp[(size_t)q * rows + r] = v;
+        const double* A0 = A + (size_t)(pcol0 + c0 + q) * ld + base0;
+        const double* A1 = A + (size_t)(pcol0 + c0 + q) * ld + base1 - len0;
+        double* spq = sp + (size_t)q * rows;
+        for (int r = tid; r < len0; r += 1024) {
+          double v = A0[r];
+          spq[r] = v;
           if (q == 0 && r > c0) acc0 += v * v;
+        }
+        for (int r = len0 + tid; r < rows; r += 1024) {
+          double v = A1[r];
+          spq[r] = v;
+          if (q == 0) acc0 += v * v;
         }
       }
       for (int sh = 32; sh > 0; sh >>= 1) acc0 += __shfl_down(acc0, sh);
@@ -496,6 +507,11 @@ __global__ void __launch_bounds__(1024) k_qr_panel_mw(
         tau[c0] = tj;
       }
       __syncthreads();
+      if (dbg && tid == 0) {
+        unsigned long long t1 = __builtin_amdgcn_s_memrealtime();
+        dbg[1] += t1 - t0;  // stage
+        t0 = t1;
+      }
       // ---- factor: one barrier per column ----
       for (int j = 0; j < w; j++) {
         const int d = c0 + j;
@@ -504,8 +520,11 @@ __global__ void __launch_bounds__(1024) k_qr_panel_mw(
           const double* col = sp + (size_t)j * rows;
           double* cc = sp + (size_t)wave * rows;
           double dot = (lane == 0) ? vd * cc[d] : 0.0;
-          for (int i = d + 1 + lane; i < rows; i += 64)
-            dot += col[i] * cc[i];
+          double dot1 = 0.0;
+          for (int i = d + 1 + lane; i < rows; i += 128) dot += col[i] * cc[i];
+          for (int i = d + 65 + lane; i < rows; i += 128)
+            dot1 += col[i] * cc[i];
+          dot += dot1;
           for (int sh = 32; sh > 0; sh >>= 1) dot += __shfl_down(dot, sh);
           dot = __shfl(dot, 0);
           const double wj = tfac * dot;
@@ -538,6 +557,11 @@ __global__ void __launch_bounds__(1024) k_qr_panel_mw(
         }
         __syncthreads();
       }
+      if (dbg && tid == 0) {
+        unsigned long long t1 = __builtin_amdgcn_s_memrealtime();
+        dbg[2] += t1 - t0;  // factor
+        t0 = t1;
+      }
       // ---- scale pass ----
       if (wave < w) {
         const int d = c0 + wave;
@@ -549,13 +573,18 @@ __global__ void __launch_bounds__(1024) k_qr_panel_mw(
       __syncthreads();
       // ---- write back ----
       for (int q = 0; q < w; q++) {
-        const int gcol = pcol0 + c0 + q;
-        for (int r = tid; r < rows; r += 1024) {
-          int gr = r < len0 ? base0 + r : base1 + (r - len0);
-          A[(size_t)gcol * ld + gr] = sp[(size_t)q * rows + r];
-        }
+        double* A0 = A + (size_t)(pcol0 + c0 + q) * ld + base0;
+        double* A1 = A + (size_t)(pcol0 + c0 + q) * ld + base1 - len0;
+        const double* spq = sp + (size_t)q * rows;
+        for (int r = tid; r < len0; r += 1024) A0[r] = spq[r];
+        for (int r = len0 + tid; r < rows; r += 1024) A1[r] = spq[r];
       }
       __syncthreads();
+      if (dbg && tid == 0) {
+        unsigned long long t1 = __builtin_amdgcn_s_memrealtime();
+        dbg[3] += t1 - t0;  // scale + writeback
+        t0 = t1;
+      }
       // ---- G = V^T V: 64 units of 16 lanes over the (s<q) pairs ----
       {
         const int unit = wave * 4 + (lane >> 4), l16 = lane & 15;
@@ -569,15 +598,15 @@ __global__ void __launch_bounds__(1024) k_qr_panel_mw(
             q++;
           }
           int sc = rem;  // 0 <= sc < q
-          const int ds = c0 + sc, dq = c0 + q;
+          const int dq = c0 + q;
           const double* cs = sp + (size_t)sc * rows;
           const double* cq = sp + (size_t)q * rows;
-          double dot = 0.0;
-          for (int i = dq + l16; i < rows; i += 16) {
-            double as = i == ds ? 1.0 : cs[i];
-            double aq = i == dq ? 1.0 : cq[i];
-            dot += as * aq;
-          }
+          // i >= dq > ds always: v_s = cs[i]; v_q = 1 at dq, cq[i] below
+          double dot = (l16 == 0) ? cs[dq] : 0.0;
+          double d1 = 0.0;
+          for (int i = dq + 1 + l16; i < rows; i += 32) dot += cs[i] * cq[i];
+          for (int i = dq + 17 + l16; i < rows; i += 32) d1 += cs[i] * cq[i];
+          dot += d1;
           for (int sh = 8; sh > 0; sh >>= 1)
             dot += __shfl_down(dot, sh, 16);
           if (l16 == 0) gt[sc * 16 + q] = dot;  // G(sc, q)
@@ -604,6 +633,11 @@ __global__ void __launch_bounds__(1024) k_qr_panel_mw(
                 (j < w && lane <= j) ? Tl[j * 16 + lane] : 0.0;
       }
       __syncthreads();
+      if (dbg && tid == 0) {
+        unsigned long long t1 = __builtin_amdgcn_s_memrealtime();
+        dbg[4] += t1 - t0;  // G16 + T16
+        t0 = t1;
+      }
       // ---- publish sub-panel si ----
       if (tid == 0) {
         __builtin_amdgcn_fence(__ATOMIC_RELEASE, "agent");
@@ -621,6 +655,9 @@ __global__ void __launch_bounds__(1024) k_qr_panel_mw(
     const int c0 = si * W;
     const int w = min(W, fcols - c0);
     const double* T16 = T16s + (size_t)si * 256;
+    unsigned long long t0 =
+        (dbg && (wg == 1 || wg == nA + 1)) ? __builtin_amdgcn_s_memrealtime()
+                                           : 0;
     if (tid == 0) {
       while (__hip_atomic_load(&cnt[0], __ATOMIC_RELAXED,
                                __HIP_MEMORY_SCOPE_AGENT) < si + 1)
@@ -628,21 +665,28 @@ __global__ void __launch_bounds__(1024) k_qr_panel_mw(
       __builtin_amdgcn_fence(__ATOMIC_ACQUIRE, "agent");
     }
     __syncthreads();
+    const int dslot = (wg == 1) ? 5 : (wg == nA + 1 ? 8 : -1);
+    if (dbg && dslot >= 0 && tid == 0) {
+      unsigned long long t1 = __builtin_amdgcn_s_memrealtime();
+      dbg[dslot] += t1 - t0;  // publish poll wait
+      t0 = t1;
+    }
     // cache the scaled V image (unit-lower) in LDS
     for (int q = 0; q < w; q++) {
       const int gcol = pcol0 + c0 + q, cloc = c0 + q;
-      for (int r = tid; r < rows; r += 1024) {
-        double v;
-        if (r < cloc) v = 0.0;
-        else if (r == cloc) v = 1.0;
-        else {
-          int gr = r < len0 ? base0 + r : base1 + (r - len0);
-          v = A[(size_t)gcol * ld + gr];
-        }
-        sp[(size_t)q * rows + r] = v;
-      }
+      const double* A0 = A + (size_t)gcol * ld + base0;
+      const double* A1 = A + (size_t)gcol * ld + base1 - len0;
+      double* spq = sp + (size_t)q * rows;
+      for (int r = tid; r < len0; r += 1024)
+        spq[r] = r < cloc ? 0.0 : (r == cloc ? 1.0 : A0[r]);
+      for (int r = len0 + tid; r < rows; r += 1024) spq[r] = A1[r];
     }
     __syncthreads();
+    if (dbg && dslot >= 0 && tid == 0) {
+      unsigned long long t1 = __builtin_amdgcn_s_memrealtime();
+      dbg[dslot + 1] += t1 - t0;  // V image build
+      t0 = t1;
+    }
     // apply (I - V T V^T)^T to 4-column groups of my column set
     auto apply4 = [&](int cg[4], int nc) {
       const int ci = lane >> 4, l16 = lane & 15;  // unit = (q=wave, ci)
@@ -650,11 +694,14 @@ __global__ void __launch_bounds__(1024) k_qr_panel_mw(
         const int gcol = pcol0 + cg[ci];
         const int cloc = c0 + wave;
         const double* v = sp + (size_t)wave * rows;
-        double dot = 0;
-        for (int r = cloc + l16; r < rows; r += 16) {
-          int gr = r < len0 ? base0 + r : base1 + (r - len0);
-          dot += v[r] * A[(size_t)gcol * ld + gr];
-        }
+        const double* A0 = A + (size_t)gcol * ld + base0;
+        const double* A1 = A + (size_t)gcol * ld + base1 - len0;
+        double dot = 0, d1 = 0;
+        for (int r = cloc + l16; r < len0; r += 32) dot += v[r] * A0[r];
+        for (int r = cloc + 16 + l16; r < len0; r += 32) d1 += v[r] * A0[r];
+        for (int r = len0 + l16; r < rows; r += 32) dot += v[r] * A1[r];
+        for (int r = len0 + 16 + l16; r < rows; r += 32) d1 += v[r] * A1[r];
+        dot += d1;
         for (int sh = 8; sh > 0; sh >>= 1) dot += __shfl_down(dot, sh, 16);
         if (l16 == 0) wy[ci * 16 + wave] = dot;
       }
@@ -671,12 +718,18 @@ __global__ void __launch_bounds__(1024) k_qr_panel_mw(
         const int c = tid >> 8, t2 = tid & 255;  // 4 cols x 256 threads
         if (c < nc) {
           const int gcol = pcol0 + cg[c];
-          for (int r = c0 + t2; r < rows; r += 256) {
-            int gr = r < len0 ? base0 + r : base1 + (r - len0);
+          double* A0 = A + (size_t)gcol * ld + base0;
+          double* A1 = A + (size_t)gcol * ld + base1 - len0;
+          const double* y = wy + 64 + c * 16;
+          for (int r = c0 + t2; r < len0; r += 256) {
             double sacc = 0;
-            for (int q = 0; q < w; q++)
-              sacc += sp[(size_t)q * rows + r] * wy[64 + c * 16 + q];
-            A[(size_t)gcol * ld + gr] -= sacc;
+            for (int q = 0; q < w; q++) sacc += sp[(size_t)q * rows + r] * y[q];
+            A0[r] -= sacc;
+          }
+          for (int r = len0 + t2; r < rows; r += 256) {
+            double sacc = 0;
+            for (int q = 0; q < w; q++) sacc += sp[(size_t)q * rows + r] * y[q];
+            A1[r] -= sacc;
           }
         }
       }
@@ -697,6 +750,8 @@ __global__ void __launch_bounds__(1024) k_qr_panel_mw(
     if (poolA) {
       apply_range(c0 + w, fcols, nA, wg - 1);
       __syncthreads();
+      if (dbg && dslot >= 0 && tid == 0)
+        dbg[dslot + 2] += __builtin_amdgcn_s_memrealtime() - t0;  // apply
       if (tid == 0) {
         __builtin_amdgcn_fence(__ATOMIC_RELEASE, "agent");
         asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
@@ -707,6 +762,8 @@ __global__ void __launch_bounds__(1024) k_qr_panel_mw(
         apply_range(fcols, pc, nA, wg - 1);
     } else {
       apply_range(fcols, pc, nB, wg - 1 - nA);
+      if (dbg && dslot >= 0 && tid == 0)
+        dbg[dslot + 2] += __builtin_amdgcn_s_memrealtime() - t0;  // apply
     }
   }
 }
@@ -751,6 +808,7 @@ static void qr_factor_hand(GpuTaskCtx& g, double* A, int m, int k, int ld,
   double* T16s = qr_scratch(g, slot0 + 9, (size_t)8 * 256 * 8);
   int* cnt = (int*)qr_scratch(g, slot0 + 10, 256);
   const int nwg = std::max(2, (int)param_int("qr_panel_wgs", 32));
+  const int nA = std::min((int64_t)nwg - 1, param_int("qr_panel_poolA", 15));
   const double one = 1.0, zero = 0.0, mone = -1.0;
   PA_HIP_CHECK(hipMemsetAsync(T, 0, (size_t)ldt * k * 8, g.stream));
   for (int p = 0; p < k; p += 128) {
@@ -772,7 +830,8 @@ static void qr_factor_hand(GpuTaskCtx& g, double* A, int m, int k, int ld,
       PA_HIP_CHECK(hipMemsetAsync(cnt, 0, 2 * sizeof(int), g.stream));
       hipLaunchKernelGGL(k_qr_panel_mw, dim3(nwg), dim3(1024), 0, g.stream,
                          A, ld, p, base0, len0, base1, len1, k - p, pc, W,
-                         tau + p, T16s, cnt, nwg);
+                         tau + p, T16s, cnt, nwg, nA,
+                         (unsigned long long*)nullptr);
       continue;
     }
     hipLaunchKernelGGL(k_geqr2, dim3(1), dim3(1024), 0, g.stream, panel,
@@ -1107,7 +1166,14 @@ double bench_qr_factor(int m, int k, int ts_split, int iters, int mode) {
   double* tau = qr_scratch(g, 0, (size_t)k * 8);
   double* T16s = qr_scratch(g, 9, (size_t)8 * 256 * 8);
   int* cnt = (int*)qr_scratch(g, 10, 256);
+  unsigned long long* dbg = nullptr;
+  if (mode == 5) {
+    PA_HIP_CHECK(hipMalloc(&dbg, 16 * 8));
+    PA_HIP_CHECK(hipMemset(dbg, 0, 16 * 8));
+    mode = 1;
+  }
   const int nwg = std::max(2, (int)param_int("qr_panel_wgs", 32));
+  const int nA = std::min((int64_t)nwg - 1, param_int("qr_panel_poolA", 15));
   auto run = [&] {
     hipLaunchKernelGGL(k_qr_fill, dim3(2048), dim3(256), 0, s, dA,
                        (size_t)m * k, 7);
@@ -1129,7 +1195,7 @@ double bench_qr_factor(int m, int k, int ts_split, int iters, int mode) {
         PA_HIP_CHECK(hipMemsetAsync(cnt, 0, 2 * sizeof(int), s));
         hipLaunchKernelGGL(k_qr_panel_mw, dim3(nwg), dim3(1024), 0, s, dA,
                            m, p, base0, len0, base1, len1, k - p, pc, W,
-                           tau + p, T16s, cnt, nwg);
+                           tau + p, T16s, cnt, nwg, nA, dbg);
       }
     } else {
       PA_CHECK(rocsolver_dgeqrf(qr_handle(g), m, k, dA, m, tau) ==
@@ -1142,6 +1208,18 @@ double bench_qr_factor(int m, int k, int ts_split, int iters, int mode) {
   for (int i = 0; i < iters; i++) run();
   PA_HIP_CHECK(hipStreamSynchronize(s));
   double dt = now_s() - t0;
+  if (dbg) {
+    unsigned long long h[16];
+    PA_HIP_CHECK(hipMemcpy(h, dbg, 16 * 8, hipMemcpyDeviceToHost));
+    const char* names[] = {"wg0 ack-wait", "wg0 stage", "wg0 factor",
+                           "wg0 scale+wb", "wg0 G16+T16", "A poll",
+                           "A vbuild", "A apply", "B poll", "B vbuild",
+                           "B apply"};
+    for (int i = 0; i < 11; i++)
+      fprintf(stderr, "[qrdbg] %-12s %8.1f us/tile\n", names[i],
+              h[i] / 0.1 / (iters + 1));  // 100 MHz ticks
+    PA_HIP_CHECK(hipFree(dbg));
+  }
   PA_HIP_CHECK(hipFree(dA));
   PA_HIP_CHECK(hipFree(dT));
   return dt;

@@ -454,11 +454,14 @@ __global__ void __launch_bounds__(1024) k_qr_panel_mw(
       const int w = min(W, fcols - c0);
       double* T16 = T16s + (size_t)si * 256;
       unsigned long long t0 = dbg ? __builtin_amdgcn_s_memrealtime() : 0;
-      if (si > 0) {
-        // wait for pool A's within-panel applies of sub-panel si-1
+      if (si > 1) {
+        // one-sub-panel slack: WG0 self-applies each block to its own
+        // next-16 columns, so only ack(si-2) gates this stage and pool
+        // A's work hides under the factor of si-1.
         if (tid == 0) {
           while (__hip_atomic_load(&cnt[1], __ATOMIC_RELAXED,
-                                   __HIP_MEMORY_SCOPE_AGENT) < si * nA)
+                                   __HIP_MEMORY_SCOPE_AGENT) <
+                 (si - 1) * nA)
             __builtin_amdgcn_s_sleep(2);
           __builtin_amdgcn_fence(__ATOMIC_ACQUIRE, "agent");
         }
@@ -645,6 +648,69 @@ __global__ void __launch_bounds__(1024) k_qr_panel_mw(
         __hip_atomic_store(&cnt[0], si + 1, __ATOMIC_RELAXED,
                            __HIP_MEMORY_SCOPE_AGENT);
       }
+      // ---- self-apply this block to the NEXT sub-panel's columns (the
+      // V image is already in sp, minus the unit-lower masking) ----
+      for (int c = c0 + w; c < min(c0 + 2 * W, fcols); c += 4) {
+        const int nc = min(4, min(c0 + 2 * W, fcols) - c);
+        const int ci = lane >> 4, l16 = lane & 15;
+        if (ci < nc && wave < w) {
+          const int gcol = pcol0 + c + ci;
+          const int cloc = c0 + wave;
+          const double* v = sp + (size_t)wave * rows;
+          const double* A0 = A + (size_t)gcol * ld + base0;
+          const double* A1 = A + (size_t)gcol * ld + base1 - len0;
+          // raw sp has R above the diagonal: start at cloc, inject the
+          // unit diagonal by hand
+          double dot = (l16 == 0) ? A0[cloc] : 0.0;
+          double d1 = 0;
+          for (int r = cloc + 1 + l16; r < len0; r += 32)
+            dot += v[r] * A0[r];
+          for (int r = cloc + 17 + l16; r < len0; r += 32)
+            d1 += v[r] * A0[r];
+          for (int r = len0 + l16; r < rows; r += 32) dot += v[r] * A1[r];
+          for (int r = len0 + 16 + l16; r < rows; r += 32)
+            d1 += v[r] * A1[r];
+          dot += d1;
+          for (int sh = 8; sh > 0; sh >>= 1) dot += __shfl_down(dot, sh, 16);
+          if (l16 == 0) wy[ci * 16 + wave] = dot;
+        }
+        __syncthreads();
+        if (tid < 16 * nc) {
+          const int q = tid & 15, cc2 = tid >> 4;
+          double sacc = 0;
+          for (int p2 = 0; p2 <= q; p2++)
+            sacc += T16[q * 16 + p2] * wy[cc2 * 16 + p2];
+          wy[64 + cc2 * 16 + q] = sacc;
+        }
+        __syncthreads();
+        {
+          const int cc2 = tid >> 8, t2 = tid & 255;
+          if (cc2 < nc) {
+            const int gcol = pcol0 + c + cc2;
+            double* A0 = A + (size_t)gcol * ld + base0;
+            double* A1 = A + (size_t)gcol * ld + base1 - len0;
+            const double* y = wy + 64 + cc2 * 16;
+            for (int r = c0 + t2; r < len0; r += 256) {
+              double sacc = 0;
+              for (int q = 0; q < w; q++) {
+                const int cloc = c0 + q;
+                double v = r < cloc ? 0.0
+                                    : (r == cloc ? 1.0
+                                                 : sp[(size_t)q * rows + r]);
+                sacc += v * y[q];
+              }
+              A0[r] -= sacc;
+            }
+            for (int r = len0 + t2; r < rows; r += 256) {
+              double sacc = 0;
+              for (int q = 0; q < w; q++)
+                sacc += sp[(size_t)q * rows + r] * y[q];
+              A1[r] -= sacc;
+            }
+          }
+        }
+        __syncthreads();
+      }
     }
     return;
   }
@@ -696,12 +762,16 @@ __global__ void __launch_bounds__(1024) k_qr_panel_mw(
         const double* v = sp + (size_t)wave * rows;
         const double* A0 = A + (size_t)gcol * ld + base0;
         const double* A1 = A + (size_t)gcol * ld + base1 - len0;
-        double dot = 0, d1 = 0;
-        for (int r = cloc + l16; r < len0; r += 32) dot += v[r] * A0[r];
-        for (int r = cloc + 16 + l16; r < len0; r += 32) d1 += v[r] * A0[r];
-        for (int r = len0 + l16; r < rows; r += 32) dot += v[r] * A1[r];
-        for (int r = len0 + 16 + l16; r < rows; r += 32) d1 += v[r] * A1[r];
-        dot += d1;
+        double d0 = 0, d1 = 0, d2 = 0, d3 = 0;
+        for (int r = cloc + l16; r < len0; r += 64) d0 += v[r] * A0[r];
+        for (int r = cloc + 16 + l16; r < len0; r += 64) d1 += v[r] * A0[r];
+        for (int r = cloc + 32 + l16; r < len0; r += 64) d2 += v[r] * A0[r];
+        for (int r = cloc + 48 + l16; r < len0; r += 64) d3 += v[r] * A0[r];
+        for (int r = len0 + l16; r < rows; r += 64) d0 += v[r] * A1[r];
+        for (int r = len0 + 16 + l16; r < rows; r += 64) d1 += v[r] * A1[r];
+        for (int r = len0 + 32 + l16; r < rows; r += 64) d2 += v[r] * A1[r];
+        for (int r = len0 + 48 + l16; r < rows; r += 64) d3 += v[r] * A1[r];
+        double dot = (d0 + d1) + (d2 + d3);
         for (int sh = 8; sh > 0; sh >>= 1) dot += __shfl_down(dot, sh, 16);
         if (l16 == 0) wy[ci * 16 + wave] = dot;
       }
@@ -748,7 +818,7 @@ __global__ void __launch_bounds__(1024) k_qr_panel_mw(
       if (nc) apply4(cg, nc);
     };
     if (poolA) {
-      apply_range(c0 + w, fcols, nA, wg - 1);
+      apply_range(min(c0 + 2 * W, fcols), fcols, nA, wg - 1);
       __syncthreads();
       if (dbg && dslot >= 0 && tid == 0)
         dbg[dslot + 2] += __builtin_amdgcn_s_memrealtime() - t0;  // apply
@@ -807,7 +877,7 @@ static void qr_factor_hand(GpuTaskCtx& g, double* A, int m, int k, int ld,
   double* tau = qr_scratch(g, slot0, (size_t)k * 8);
   double* T16s = qr_scratch(g, slot0 + 9, (size_t)8 * 256 * 8);
   int* cnt = (int*)qr_scratch(g, slot0 + 10, 256);
-  const int nwg = std::max(2, (int)param_int("qr_panel_wgs", 32));
+  const int nwg = std::max(2, (int)param_int("qr_panel_wgs", 48));
   const int nA = std::min((int64_t)nwg - 1, param_int("qr_panel_poolA", 15));
   const double one = 1.0, zero = 0.0, mone = -1.0;
   PA_HIP_CHECK(hipMemsetAsync(T, 0, (size_t)ldt * k * 8, g.stream));
@@ -1172,7 +1242,7 @@ double bench_qr_factor(int m, int k, int ts_split, int iters, int mode) {
     PA_HIP_CHECK(hipMemset(dbg, 0, 16 * 8));
     mode = 1;
   }
-  const int nwg = std::max(2, (int)param_int("qr_panel_wgs", 32));
+  const int nwg = std::max(2, (int)param_int("qr_panel_wgs", 48));
   const int nA = std::min((int64_t)nwg - 1, param_int("qr_panel_poolA", 15));
   auto run = [&] {
     hipLaunchKernelGGL(k_qr_fill, dim3(2048), dim3(256), 0, s, dA,

@@ -359,6 +359,13 @@ void GpuEngine::stage_in_and_launch(Task* t) {
   int si;
   if (t->priority >= (1 << 19) || (int)exec_streams_.size() == 1) {
     si = 0;
+    if (!inflight_[0].empty()) {
+      // Another express task is already in flight (concurrent panel DAGs,
+      // e.g. tree QR): spread to the least-loaded stream instead of
+      // serializing the whole panel tier behind stream 0.
+      for (int k2 = 1; k2 < (int)exec_streams_.size(); k2++)
+        if (inflight_[k2].size() < inflight_[si].size()) si = k2;
+    }
   } else {
     si = next_stream_;
     next_stream_ = next_stream_ + 1;

@@ -956,7 +956,7 @@ static void gpu_geqrt(Task& t, GpuTaskCtx& g) {
   double* A = (double*)t.dev_ptr[0];
   double* T = (double*)t.dev_ptr[1];
   static const bool use_rocsolver =
-      param_str("chore_qr", "rocsolver") == "rocsolver";
+      param_str("chore_qr", "hand") == "rocsolver";
   if (!use_rocsolver) {
     qr_factor_hand(g, A, a.n, a.n, a.ld, T, a.ld, 0);
     return;
@@ -1029,7 +1029,7 @@ static void gpu_tsqrt(Task& t, GpuTaskCtx& g) {
   hipLaunchKernelGGL(k_stack_triu, grid1d(nb * nb), dim3(256), 0, g.stream,
                      V2, Akk, Amk, nb, ld, a.k);
   static const bool use_rocsolver =
-      param_str("chore_qr", "rocsolver") == "rocsolver";
+      param_str("chore_qr", "hand") == "rocsolver";
   if (!use_rocsolver) {
     qr_factor_hand(g, V2, 2 * nb, nb, 2 * nb, T1, ld, 0, /*ts_split=*/nb);
   } else {
@@ -1081,7 +1081,7 @@ TaskClass& tc_geqrt() {
   // PARSEC_MCA_qr_blocking_panels=0 restores manager execution.
   static TaskClass tc = [] {
     TaskClass c = make_qr_tc("geqrt", cpu_geqrt, gpu_geqrt, 10);
-    bool rocs = param_str("chore_qr", "rocsolver") == "rocsolver";
+    bool rocs = param_str("chore_qr", "hand") == "rocsolver";
     c.gpu_blocking = rocs && param_int("qr_blocking_panels", 1) != 0;
     return c;
   }();
@@ -1094,7 +1094,7 @@ TaskClass& tc_unmqr() {
 TaskClass& tc_tsqrt() {
   static TaskClass tc = [] {
     TaskClass c = make_qr_tc("tsqrt", cpu_tsqrt, gpu_tsqrt, 12);
-    bool rocs = param_str("chore_qr", "rocsolver") == "rocsolver";
+    bool rocs = param_str("chore_qr", "hand") == "rocsolver";
     c.gpu_blocking = rocs && param_int("qr_blocking_panels", 1) != 0;
     return c;
   }();

@@ -56,8 +56,7 @@ def main():
         "dtype": "fp64",
         "data": "synthetic",
         "config": {"model": "tile_qr_dgeqrf", "N": n, "tile": nb,
-                   "chore_qr": os.environ.get("PARSEC_MCA_chore_qr",
-                                              "rocsolver")},
+                   "chore_qr": os.environ.get("PARSEC_MCA_chore_qr", "hand")},
     }), flush=True)
     del A, ctx
 

@@ -29,7 +29,7 @@ GpuEngine::GpuEngine(Context* ctx, int device) : ctx_(ctx), device_(device) {
   PA_HIP_CHECK(hipStreamCreateWithFlags(&h2d_stream_, hipStreamNonBlocking));
   PA_HIP_CHECK(hipStreamCreateWithFlags(&d2h_stream_, hipStreamNonBlocking));
   PA_HIP_CHECK(hipStreamCreateWithFlags(&comm_stream_, hipStreamNonBlocking));
-  int nstreams = (int)param_int("gpu_exec_streams", 4);
+  int nstreams = (int)param_int("gpu_exec_streams", 8);
   exec_streams_.resize(nstreams);
   inflight_.resize(nstreams);
   for (int i = 0; i < nstreams; i++)

@@ -342,21 +342,165 @@ __global__ void k_gemm_bf16_tn_v3(int m, int n, int k,
   }
 }
 
+// ------------------------------------------------------------- v4: 8-phase
+// The guide's 256^2 8-phase sub-phase interleave
+// (cdna_hip_programming.md §5 "The 256² 8-phase template", quoted
+// 1320-1340 TF on random data): same 4-slot K-half ring, glds staging and
+// XOR swizzle as v3, but each K-half's 32-MFMA burst is split into two
+// 16-MFMA sub-phases bracketed by raw barriers — [ds-loads (+glds) |
+// s_barrier | lgkmcnt(0) | setprio(1) 16xMFMA setprio(0) | s_barrier] —
+// so each SIMD pairs one wave's MFMA segment with its partner's LDS/DMA
+// load segment instead of stalling in the ds_read->MFMA latency window
+// (round-1 PMC: MFMA-pipe 39% busy, SQ_WAIT_ANY 50%,
+// profiles/bf16_gemm_v3_pmc.md).
+__launch_bounds__(512)
+__global__ void k_gemm_bf16_tn_v4(int m, int n, int k,
+                                  const bf16* __restrict__ A, int lda,
+                                  const bf16* __restrict__ B, int ldb,
+                                  float* __restrict__ C, int ldc, int nbx,
+                                  int accum) {
+  extern __shared__ __attribute__((aligned(16))) char smem[];
+  bf16* ring = (bf16*)smem;  // 4 slots x [A 256x32 | B 256x32]
+  constexpr int SLOT = (GB3_BM + GB3_BN) * GB3_KH;
+  const int tid = threadIdx.x;
+  const int wave = tid >> 6, lane = tid & 63;
+  const int wr = wave >> 2, wc = wave & 3;
+  int id = bf_swz(blockIdx.x, gridDim.x);
+  const int bm0 = (id % nbx) * GB3_BM, bn0 = (id / nbx) * GB3_BN;
+  const int g16 = lane >> 4, r16 = lane & 15;
+  const int srow = lane >> 2;
+  const int schunk = lane & 3;
+
+  f32x4 acc[8][4] = {};
+
+  auto stage = [&](int ph) {
+    const int k0 = ph * GB3_KH;
+    bf16* slot = ring + (ph & 3) * SLOT;
+#pragma unroll
+    for (int p = 0; p < 2; p++) {
+      int row0 = (wave * 2 + p) * 16;
+      int row = row0 + srow;
+      int kc = 8 * (schunk ^ ((row >> 2) & 3));
+      const bf16* srcA = A + (size_t)(bm0 + row) * lda + k0 + kc;
+      bf16* dstA = slot + row0 * GB3_KH;
+      __builtin_amdgcn_global_load_lds(
+          (const __attribute__((address_space(1))) unsigned int*)srcA,
+          (__attribute__((address_space(3))) unsigned int*)dstA, 16, 0, 0);
+      const bf16* srcB = B + (size_t)(bn0 + row) * ldb + k0 + kc;
+      bf16* dstB = slot + GB3_BM * GB3_KH + row0 * GB3_KH;
+      __builtin_amdgcn_global_load_lds(
+          (const __attribute__((address_space(1))) unsigned int*)srcB,
+          (__attribute__((address_space(3))) unsigned int*)dstB, 16, 0, 0);
+    }
+  };
+
+  const int P = k / GB3_KH;
+  stage(0);
+  if (P > 1) stage(1);
+  if (P > 2) stage(2);
+  for (int ph = 0; ph < P; ph++) {
+    // the K-half for THIS phase has landed; newer loads stay in flight
+    if (ph + 3 <= P - 1) {
+      asm volatile("s_waitcnt vmcnt(8)" ::: "memory");
+    } else if (ph + 2 == P - 1) {
+      asm volatile("s_waitcnt vmcnt(8)" ::: "memory");
+    } else if (ph + 1 == P - 1) {
+      asm volatile("s_waitcnt vmcnt(4)" ::: "memory");
+    } else {
+      asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+    }
+    const bf16* as = ring + (ph & 3) * SLOT;
+    const bf16* bs = as + GB3_BM * GB3_KH;
+    // ---- sub-phase 0: load a[0..3] + b[0..3], stage ph+3, 16 MFMA ----
+    bf16x8 a[8], b[4];
+#pragma unroll
+    for (int f = 0; f < 4; f++) {
+      int row = wr * 128 + f * 16 + r16;
+      a[f] = *(const bf16x8*)&as[row * GB3_KH +
+                                 ((g16 * 8) ^ (((row >> 2) & 3) * 8))];
+    }
+#pragma unroll
+    for (int f = 0; f < 4; f++) {
+      int row = wc * 64 + f * 16 + r16;
+      b[f] = *(const bf16x8*)&bs[row * GB3_KH +
+                                 ((g16 * 8) ^ (((row >> 2) & 3) * 8))];
+    }
+    if (ph + 3 < P) stage(ph + 3);
+    __builtin_amdgcn_s_barrier();
+    asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+    __builtin_amdgcn_s_setprio(1);
+#pragma unroll
+    for (int i = 0; i < 4; i++)
+#pragma unroll
+      for (int j = 0; j < 4; j++)
+        acc[i][j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a[i], b[j],
+                                                            acc[i][j], 0, 0,
+                                                            0);
+    __builtin_amdgcn_s_setprio(0);
+    __builtin_amdgcn_s_barrier();
+    // ---- sub-phase 1: load a[4..7], 16 MFMA ----
+#pragma unroll
+    for (int f = 4; f < 8; f++) {
+      int row = wr * 128 + f * 16 + r16;
+      a[f] = *(const bf16x8*)&as[row * GB3_KH +
+                                 ((g16 * 8) ^ (((row >> 2) & 3) * 8))];
+    }
+    __builtin_amdgcn_s_barrier();
+    asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+    __builtin_amdgcn_s_setprio(1);
+#pragma unroll
+    for (int i = 4; i < 8; i++)
+#pragma unroll
+      for (int j = 0; j < 4; j++)
+        acc[i][j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a[i], b[j],
+                                                            acc[i][j], 0, 0,
+                                                            0);
+    __builtin_amdgcn_s_setprio(0);
+    __builtin_amdgcn_s_barrier();
+  }
+
+#pragma unroll
+  for (int j = 0; j < 4; j++) {
+    int col = bn0 + wc * 64 + j * 16 + r16;
+    float* cp = C + (size_t)col * ldc;
+#pragma unroll
+    for (int i = 0; i < 8; i++) {
+      int row0 = bm0 + wr * 128 + i * 16 + g16 * 4;
+#pragma unroll
+      for (int e = 0; e < 4; e++) {
+        int row = row0 + e;
+        cp[row] = accum ? cp[row] + acc[i][j][e] : acc[i][j][e];
+      }
+    }
+  }
+}
+
 static void launch_gemm_bf16(int m, int n, int k, const void* A, int lda,
                              const void* B, int ldb, float* C, int ldc,
                              hipStream_t stream, int accum = 1) {
   if (m % GB3_BM == 0 && n % GB3_BN == 0 && k % GB3_KH == 0 && k >= 4 * GB3_KH) {
     constexpr size_t lds = 4 * (GB3_BM + GB3_BN) * GB3_KH * 2;
+    static int variant = (int)param_int("bf16_kernel", 4);
+    const void* kf = variant >= 4 ? (const void*)k_gemm_bf16_tn_v4
+                                  : (const void*)k_gemm_bf16_tn_v3;
     static bool attr3 = false;
     if (!attr3) {
       hipFuncSetAttribute((const void*)k_gemm_bf16_tn_v3,
                           hipFuncAttributeMaxDynamicSharedMemorySize, lds);
+      hipFuncSetAttribute((const void*)k_gemm_bf16_tn_v4,
+                          hipFuncAttributeMaxDynamicSharedMemorySize, lds);
       attr3 = true;
     }
     int nbx = m / GB3_BM, nby = n / GB3_BN;
-    hipLaunchKernelGGL(k_gemm_bf16_tn_v3, dim3(nbx * nby), dim3(512), lds,
-                       stream, m, n, k, (const bf16*)A, lda, (const bf16*)B,
-                       ldb, C, ldc, nbx, accum);
+    if (variant >= 4)
+      hipLaunchKernelGGL(k_gemm_bf16_tn_v4, dim3(nbx * nby), dim3(512), lds,
+                         stream, m, n, k, (const bf16*)A, lda,
+                         (const bf16*)B, ldb, C, ldc, nbx, accum);
+    else
+      hipLaunchKernelGGL(k_gemm_bf16_tn_v3, dim3(nbx * nby), dim3(512), lds,
+                         stream, m, n, k, (const bf16*)A, lda,
+                         (const bf16*)B, ldb, C, ldc, nbx, accum);
+    (void)kf;
     return;
   }
   int nbx = (m + GB_BM - 1) / GB_BM, nby = (n + GB_BN - 1) / GB_BN;

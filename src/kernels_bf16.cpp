@@ -480,7 +480,7 @@ static void launch_gemm_bf16(int m, int n, int k, const void* A, int lda,
                              hipStream_t stream, int accum = 1) {
   if (m % GB3_BM == 0 && n % GB3_BN == 0 && k % GB3_KH == 0 && k >= 4 * GB3_KH) {
     constexpr size_t lds = 4 * (GB3_BM + GB3_BN) * GB3_KH * 2;
-    static int variant = (int)param_int("bf16_kernel", 4);
+    int variant = (int)param_int("bf16_kernel", 4);
     const void* kf = variant >= 4 ? (const void*)k_gemm_bf16_tn_v4
                                   : (const void*)k_gemm_bf16_tn_v3;
     static bool attr3 = false;

@@ -345,7 +345,7 @@ __global__ void k_gemm_bf16_tn_v3(int m, int n, int k,
 // ------------------------------------------------------------- v4: 8-phase
 // The guide's 256^2 8-phase sub-phase interleave
 // (cdna_hip_programming.md §5 "The 256² 8-phase template", quoted
-// 1320-1340 TF on random data): same 4-slot K-half ring, glds staging and
+// 1320-1340 TF on random data there): same 4-slot K-half ring, glds and
 // XOR swizzle as v3, but each K-half's 32-MFMA burst is split into two
 // 16-MFMA sub-phases bracketed by raw barriers — [ds-loads (+glds) |
 // s_barrier | lgkmcnt(0) | setprio(1) 16xMFMA setprio(0) | s_barrier] —
@@ -353,6 +353,11 @@ __global__ void k_gemm_bf16_tn_v3(int m, int n, int k,
 // load segment instead of stalling in the ds_read->MFMA latency window
 // (round-1 PMC: MFMA-pipe 39% busy, SQ_WAIT_ANY 50%,
 // profiles/bf16_gemm_v3_pmc.md).
+// MEASURED (this repo, round 2): v4 709 TF @4096^3 / 787 @8192^3 vs v3
+// 751/784 solo, and 875 vs 943 TF in the whole-job DAG — the sub-phase
+// barriers alone do NOT reproduce the template's number without its exact
+// hand schedule, so v3 stays the default (PARSEC_MCA_bf16_kernel=4 opts
+// in). Kept as the measured A/B for the round-2 plan item.
 __launch_bounds__(512)
 __global__ void k_gemm_bf16_tn_v4(int m, int n, int k,
                                   const bf16* __restrict__ A, int lda,
@@ -480,7 +485,7 @@ static void launch_gemm_bf16(int m, int n, int k, const void* A, int lda,
                              hipStream_t stream, int accum = 1) {
   if (m % GB3_BM == 0 && n % GB3_BN == 0 && k % GB3_KH == 0 && k >= 4 * GB3_KH) {
     constexpr size_t lds = 4 * (GB3_BM + GB3_BN) * GB3_KH * 2;
-    int variant = (int)param_int("bf16_kernel", 4);
+    int variant = (int)param_int("bf16_kernel", 3);
     const void* kf = variant >= 4 ? (const void*)k_gemm_bf16_tn_v4
                                   : (const void*)k_gemm_bf16_tn_v3;
     static bool attr3 = false;

@@ -46,6 +46,8 @@ enum { FK_DATA = 1, FK_BAR_IN = 2, FK_BAR_OUT = 3 };
 class TcpComm : public CommEngine {
  public:
   TcpComm(Context* ctx) : ctx_(ctx), rank_(ctx->rank()), world_(ctx->world()) {
+    init_peer_stats(world_);
+    max_sends_ = (size_t)param_int("comm_max_inflight", 64);
     setup_mesh();
     PA_CHECK(pipe(wake_pipe_) == 0);
     set_nonblock(wake_pipe_[0]);
@@ -173,6 +175,12 @@ class TcpComm : public CommEngine {
     // stamp post time for the trace (args space is unused by comm tasks)
     *(uint64_t*)t->args = Profiler::now_ns();
     if (t->tc->kind == TaskKind::COMM_SEND) {
+      // flow control: bound buffered outgoing payload frames
+      if (sends_out_ >= max_sends_) {
+        pending_sends_.push_back(t);
+        return;
+      }
+      sends_out_++;
       void* ptr = d->pull_to_host();
       queue_frame(t->peer, FK_DATA, t->comm_seq, ptr, d->bytes, t);
     } else {  // COMM_RECV
@@ -186,6 +194,7 @@ class TcpComm : public CommEngine {
         memcpy(d->ensure_host(), it->second.data(), d->bytes);
         unexpected_.erase(it);
         d->written_on(false);
+        note_recvd(t->peer, d->bytes);
         Profiler& pr = Profiler::inst();
         if (pr.enabled())
           pr.record(Ev::COMM_RECV, (uint16_t)t->peer, t->comm_seq,
@@ -203,6 +212,7 @@ class TcpComm : public CommEngine {
       if (p.in_task) {
         Data* d = p.in_task->flows[0].data;
         d->written_on(false);
+        note_recvd(peer, d->bytes);
         Profiler& pr = Profiler::inst();
         if (pr.enabled())
           pr.record(Ev::COMM_RECV, (uint16_t)peer, p.hdr.seq,
@@ -309,6 +319,16 @@ class TcpComm : public CommEngine {
       if (done) {
         counters().comm_msgs.fetch_add(1, std::memory_order_relaxed);
         counters().comm_bytes.fetch_add(buf.size(), std::memory_order_relaxed);
+        note_sent(done->peer, buf.size());
+        sends_out_--;
+        while (!pending_sends_.empty() && sends_out_ < max_sends_) {
+          Task* pt = pending_sends_.front();
+          pending_sends_.pop_front();
+          sends_out_++;
+          Data* pd = pt->flows[0].data;
+          queue_frame(pt->peer, FK_DATA, pt->comm_seq, pd->pull_to_host(),
+                      pd->bytes, pt);
+        }
         Profiler& pr = Profiler::inst();
         if (pr.enabled())
           pr.record(Ev::COMM_SEND, (uint16_t)done->peer, done->comm_seq,
@@ -394,6 +414,8 @@ class TcpComm : public CommEngine {
 
   std::unordered_map<uint64_t, Task*> posted_recv_;
   std::unordered_map<uint64_t, std::vector<uint8_t>> unexpected_;
+  std::deque<Task*> pending_sends_;
+  size_t sends_out_ = 0, max_sends_ = 64;
 
   std::mutex bar_mtx_;
   std::condition_variable bar_cv_;

@@ -14,8 +14,10 @@
 //    busy). Multi-node would layer MPI under the same interface.
 #pragma once
 
+#include <atomic>
 #include <memory>
 #include <string>
+#include <vector>
 
 #include "common.hpp"
 #include "runtime.hpp"
@@ -32,6 +34,32 @@ class CommEngine {
   virtual void enqueue(Task* t) = 0;
   virtual void barrier() = 0;
   virtual const char* kind() const = 0;
+
+  // Per-peer traffic accounting (device stats table analog,
+  // device.c:611-658: the counters that explain a scaling curve).
+  struct PeerStat {
+    std::atomic<uint64_t> sent_msgs{0}, sent_bytes{0};
+    std::atomic<uint64_t> recv_msgs{0}, recv_bytes{0};
+  };
+  const std::vector<PeerStat>& peer_stats() const { return peer_stats_; }
+
+ protected:
+  void init_peer_stats(int world) {
+    peer_stats_ = std::vector<PeerStat>(world);
+  }
+  void note_sent(int peer, uint64_t bytes) {
+    if (peer >= 0 && peer < (int)peer_stats_.size()) {
+      peer_stats_[peer].sent_msgs.fetch_add(1, std::memory_order_relaxed);
+      peer_stats_[peer].sent_bytes.fetch_add(bytes, std::memory_order_relaxed);
+    }
+  }
+  void note_recvd(int peer, uint64_t bytes) {
+    if (peer >= 0 && peer < (int)peer_stats_.size()) {
+      peer_stats_[peer].recv_msgs.fetch_add(1, std::memory_order_relaxed);
+      peer_stats_[peer].recv_bytes.fetch_add(bytes, std::memory_order_relaxed);
+    }
+  }
+  std::vector<PeerStat> peer_stats_;
 };
 
 // world==1: no-op engine.

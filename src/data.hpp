@@ -58,6 +58,8 @@ struct Data {
   Task* last_local_writer = nullptr;     // last local task writing the buffer
   std::vector<Task*> local_readers;      // local users since that writer
   uint64_t sent_mask = 0;                // ranks already sent current version
+  std::vector<int> recip_order;          // recipients of current version, in
+                                         // discovery order (broadcast tree)
   uint32_t local_present_version = 0;    // version the local buffer will hold
   bool local_present = false;
 

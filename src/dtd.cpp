@@ -41,6 +41,13 @@ Dtd::Dtd(Context* ctx, std::string name) : Taskpool(ctx, std::move(name)) {
   chan_seq_.assign((size_t)world_ * world_, 0);
   window_ = param_int("dtd_window_size", 16384);
   threshold_ = param_int("dtd_threshold_size", 8192);
+  // One-to-many tile fan-out (collective propagation, remote_dep.c:322-437
+  // chain/binomial trees): "binomial" re-sends through earlier recipients
+  // (binary tree over discovery order, deterministic on every rank), so at
+  // world 8 the owner's out-links stop being the fan-out bound. Default
+  // unicast: single-node xGMI is per-link bound and direct sends already
+  // use distinct links (profiles/RESULTS.md; re-decide with SCALE data).
+  bcast_tree_ = param_str("bcast_tree", "unicast") == "binomial";
 }
 
 Dtd::~Dtd() = default;
@@ -99,6 +106,7 @@ Data* Dtd::maybe_rename(Data* d) {
   nd->version = d->version;
   nd->owner_rank = d->owner_rank;
   nd->sent_mask = d->sent_mask;
+  nd->recip_order = d->recip_order;
   Task* rc = task_new(this, &RECLAIM_CLASS);
   rc->arg<Data*>() = d;  // shell owned by the collection
   if (d->last_local_writer) task_add_edge(d->last_local_writer, rc);
@@ -131,11 +139,20 @@ Data* Dtd::read_flow(Data* d, Task* t, int task_rank) {
   const int O = d->owner_rank, R = task_rank;
   if (O != R && !(d->sent_mask & (1ull << R))) {
     // Every rank advances the replicated channel counter; only the
-    // endpoints create the actual transfer tasks.
-    uint64_t seq = chan_next(O, R);
+    // endpoints create the actual transfer tasks. With the binomial tree,
+    // the Nth recipient fetches from recipient (N-1)/2 (whose recv task is
+    // already that rank's last_local_writer, so the forwarded send chains
+    // behind it automatically).
+    int src = O;
+    if (bcast_tree_) {
+      size_t idx = d->recip_order.size();
+      if (idx > 0) src = d->recip_order[(idx - 1) / 2];
+      d->recip_order.push_back(R);
+    }
+    uint64_t seq = chan_next(src, R);
     d->sent_mask |= 1ull << R;
-    if (O == me_) make_send(d, R, seq);
-    if (R == me_) d = make_recv(d, O, seq);
+    if (src == me_) make_send(d, R, seq);
+    if (R == me_) d = make_recv(d, src, seq);
   }
   if (R == me_) {
     if (d->last_local_writer) task_add_edge(d->last_local_writer, t);
@@ -162,6 +179,7 @@ Data* Dtd::write_flow(Data* d, Task* t, int task_rank, bool output_only) {
   d->version++;
   d->owner_rank = R;
   d->sent_mask = 0;
+  d->recip_order.clear();
   if (R == me_) {
     set_local_writer(d, t);
     d->local_present = true;
@@ -238,6 +256,7 @@ void Dtd::flush(Data* d) {
   if (H == me_) d = make_recv(d, O, seq);
   d->owner_rank = H;
   d->sent_mask = 0;
+  d->recip_order.clear();
 }
 
 void Dtd::flush_all(TiledMatrix& A) {

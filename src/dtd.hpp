@@ -85,6 +85,7 @@ class Dtd : public Taskpool {
   std::vector<uint64_t> chan_seq_;
   int64_t window_;
   int64_t threshold_;
+  bool bcast_tree_ = false;
 };
 
 }  // namespace pa

@@ -13,6 +13,7 @@
 
 #include "data.hpp"
 #include "device_gpu.hpp"
+#include "comm.hpp"
 #include "dtd.hpp"
 #include "kernels.hpp"
 #include "pins.hpp"
@@ -155,6 +156,21 @@ PYBIND11_MODULE(_core, m) {
           d["evictions"] = c.gpu()->stats.evictions.load();
         }
         return d;
+      })
+      .def("comm_stats", [](Context& c) {
+        // per-peer traffic table (device-stats analog for the comm engine)
+        py::list out;
+        if (c.comm()) {
+          for (auto& ps : c.comm()->peer_stats()) {
+            py::dict d;
+            d["sent_msgs"] = ps.sent_msgs.load();
+            d["sent_bytes"] = ps.sent_bytes.load();
+            d["recv_msgs"] = ps.recv_msgs.load();
+            d["recv_bytes"] = ps.recv_bytes.load();
+            out.append(d);
+          }
+        }
+        return out;
       })
       .def("counters", [](Context&) {
         // PINS/papi_sde-style software counters (process-wide)

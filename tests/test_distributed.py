@@ -420,3 +420,65 @@ del A, ctx
         outs.append(o)
         assert pr.returncode == 0, o.decode()
     assert b"RENAME_OK" in outs[0], outs[0].decode()
+
+
+@pytest.mark.parametrize("world,p,q", [(4, 2, 2), (8, 2, 4)])
+def test_distributed_cholesky_binomial_tree(world, p, q, tmp_path,
+                                            monkeypatch):
+    """Collective propagation tree (remote_dep.c:322-437 analog): panel
+    fan-out re-sent through earlier recipients. Same numerics contract as
+    the unicast run."""
+    monkeypatch.setenv("PARSEC_MCA_bcast_tree", "binomial")
+    pre, post = run_world(world, tmp_path, p, q)
+    M = np.tril(pre) + np.tril(pre, -1).T
+    L0 = np.linalg.cholesky(M)
+    err = np.abs(np.tril(post) - L0).max()
+    assert err < 1e-10, f"binomial world={world}: max err {err}"
+
+
+def test_comm_peer_stats(tmp_path):
+    """Per-peer comm counters (device-stats analog): bytes move in both
+    directions and the table is per-peer."""
+    import conftest
+    REPO = os.path.dirname(HERE)
+    port = str(conftest.port_base(17))
+    code = """
+import os, sys
+sys.path.insert(0, %r)
+import numpy as np
+import parsec_amd as pm
+rank = int(os.environ["RANK"])
+pm.param_set("comm_base_port", os.environ["PORT"])
+ctx = pm.Context(nworkers=2, rank=rank, world=2, comm="tcp", gpu=-2)
+nb = 32
+A = pm.TiledMatrix(ctx, 2 * nb, nb, nb, nb, 2, 1)
+tp = pm.Dtd(ctx)
+t00 = A.tile(0, 0)
+for k2 in range(4):
+    def w(buf, k2=k2):
+        np.frombuffer(buf, dtype=np.float64)[:] = k2
+    tp.insert_py(w, [(t00, pm.ACCESS_OUT)], rank=k2 %% 2, with_data=True)
+    def r(buf):
+        pass
+    tp.insert_py(r, [(t00, pm.ACCESS_IN)], rank=(k2 + 1) %% 2,
+                 with_data=True)
+tp.wait()
+st = ctx.comm_stats()
+assert len(st) == 2
+other = 1 - rank
+assert st[other]["sent_bytes"] > 0 and st[other]["recv_bytes"] > 0, st
+assert st[rank]["sent_bytes"] == 0 and st[rank]["recv_bytes"] == 0, st
+print("PEERSTATS_OK", st[other])
+ctx.barrier()
+del A, ctx
+""" % (REPO,)
+    import subprocess as sp
+    procs = []
+    for r in range(2):
+        env = dict(os.environ)
+        env.update(RANK=str(r), WORLD_SIZE="2", PORT=port)
+        procs.append(sp.Popen([sys.executable, "-c", code], env=env,
+                              stdout=sp.PIPE, stderr=sp.STDOUT))
+    for pr in procs:
+        o, _ = pr.communicate(timeout=180)
+        assert pr.returncode == 0 and b"PEERSTATS_OK" in o, o.decode()

@@ -72,6 +72,8 @@ GpuEngine::~GpuEngine() {
     hipEventDestroy(dr.ev);
   }
   for (auto& e : event_pool_) hipEventDestroy(e);
+  for (auto& e : tev_pool_) hipEventDestroy(e);
+  if (ref_ev_) hipEventDestroy(ref_ev_);
   for (auto s : exec_streams_) hipStreamDestroy(s);
   hipStreamDestroy(h2d_stream_);
   hipStreamDestroy(d2h_stream_);
@@ -104,6 +106,28 @@ hipEvent_t GpuEngine::event_get() {
 void GpuEngine::event_put(hipEvent_t e) {
   SpinGuard g(ev_lock_);
   event_pool_.push_back(e);
+}
+
+hipEvent_t GpuEngine::tev_get() {
+  {
+    SpinGuard g(ev_lock_);
+    if (!tev_pool_.empty()) {
+      hipEvent_t e = tev_pool_.back();
+      tev_pool_.pop_back();
+      return e;
+    }
+  }
+  hipEvent_t e;
+  PA_HIP_CHECK(hipEventCreate(&e));  // timing enabled
+  return e;
+}
+
+void GpuEngine::gpu_span_calibrate() {
+  if (ref_ev_) return;
+  PA_HIP_CHECK(hipEventCreate(&ref_ev_));
+  PA_HIP_CHECK(hipEventRecord(ref_ev_, exec_streams_[0]));
+  PA_HIP_CHECK(hipEventSynchronize(ref_ev_));
+  ref_ns_ = Profiler::now_ns();
 }
 
 bool GpuEngine::reap_draining_locked() {
@@ -323,6 +347,8 @@ void GpuEngine::stage_flows(Task* t, hipStream_t es) {
       }
     }
     SpinGuard g(d->lock);
+    if (t->flows[i].mode & ACCESS_IN)
+      stats.bytes_required += d->bytes;
     if ((t->flows[i].mode & ACCESS_IN) && !d->dev_valid) {
       PA_CHECK(d->host_valid, "stage-in: no valid copy for tile");
       // A host copy produced by an async eviction writeback may still be in
@@ -378,14 +404,21 @@ void GpuEngine::stage_in_and_launch(Task* t) {
   if (debug_history_on())
     debug_history_add("gpu_submit %s seq=%lu", t->tc->name.c_str(),
                       (unsigned long)t->seq);
+  const bool spans = Profiler::inst().enabled();
+  hipEvent_t sev = nullptr;
+  if (spans) {
+    gpu_span_calibrate();
+    sev = tev_get();
+    PA_HIP_CHECK(hipEventRecord(sev, es));
+  }
   if (roctx_on()) roctx_push(t->tc->name.c_str());
   t->tc->gpu_hook(*t, gctx);
   if (roctx_on()) roctx_pop();
   PA_PINS(PinsEv::GPU_SUBMIT, t, -1);
-  hipEvent_t ev = event_get();
+  hipEvent_t ev = spans ? tev_get() : event_get();
   PA_HIP_CHECK(hipEventRecord(ev, es));
   inflight_[si].push_back(
-      InFlight{t, ev, si, Profiler::now_ns(), std::move(deferred)});
+      InFlight{t, ev, si, Profiler::now_ns(), sev, std::move(deferred)});
   n_inflight_++;
 }
 
@@ -408,10 +441,26 @@ bool GpuEngine::retire_pass() {
       if (pr.enabled())
         pr.record(Ev::GPU_TASK, (uint16_t)t->tc->id, t->seq, f.t0_ns,
                   Profiler::now_ns());
+      if (f.start_ev) {
+        // device-side span on this exec stream: its own trace lane
+        float ms0 = 0, ms1 = 0;
+        hipEventElapsedTime(&ms0, ref_ev_, f.start_ev);
+        hipEventElapsedTime(&ms1, ref_ev_, f.event);
+        if (pr.enabled())
+          pr.record_tid(Ev::GPU_SPAN, (uint16_t)t->tc->id, t->seq,
+                        ref_ns_ + (uint64_t)((double)ms0 * 1e6),
+                        ref_ns_ + (uint64_t)((double)ms1 * 1e6),
+                        1000 + (uint32_t)f.stream_idx);
+        SpinGuard g2(ev_lock_);
+        tev_pool_.push_back(f.start_ev);
+        tev_pool_.push_back(f.event);
+        f.start_ev = nullptr;
+        f.event = nullptr;
+      }
       counters().tasks_executed_gpu.fetch_add(1, std::memory_order_relaxed);
       PA_PINS(PinsEv::GPU_RETIRE, t, -1);
       for (auto& [p2, b2] : f.deferred_frees) dev_free(p2, b2);
-      event_put(f.event);
+      if (f.event) event_put(f.event);
       ring.pop_front();
       n_inflight_--;
       stats.tasks++;

@@ -84,6 +84,10 @@ class GpuEngine {
     std::atomic<uint64_t> bytes_h2d{0};
     std::atomic<uint64_t> bytes_d2h{0};
     std::atomic<uint64_t> evictions{0};
+    // flow bytes tasks needed on-device vs bytes actually staged H2D:
+    // the required-vs-transferred accounting that catches staging waste
+    // (device.c:611-658 statistics analog)
+    std::atomic<uint64_t> bytes_required{0};
   } stats;
 
  private:
@@ -92,6 +96,7 @@ class GpuEngine {
     hipEvent_t event;
     int stream_idx;
     uint64_t t0_ns;
+    hipEvent_t start_ev = nullptr;  // timing pair (profiler on only)
     std::vector<std::pair<void*, size_t>> deferred_frees;
   };
   struct PQEntry {
@@ -124,6 +129,13 @@ class GpuEngine {
   size_t n_inflight_ = 0;
   size_t max_inflight_per_stream_;
   std::vector<hipEvent_t> event_pool_;
+  // device-timestamp calibration for per-stream GPU spans: t_gpu(ns) =
+  // ref_ns_ + hipEventElapsedTime(ref_ev_, ev)
+  std::vector<hipEvent_t> tev_pool_;  // timing-enabled events
+  hipEvent_t ref_ev_ = nullptr;
+  uint64_t ref_ns_ = 0;
+  hipEvent_t tev_get();
+  void gpu_span_calibrate();
 
   // memory pool
   bool evict_one(size_t bytes);

@@ -47,7 +47,29 @@ Profiler::Buf* Profiler::tls_buf() {
 }
 
 void Profiler::start(const std::string& filename) {
+  std::lock_guard<std::mutex> g(mtx_);
   filename_ = filename;
+  file_ = fopen(filename_.c_str(), "wb");
+  if (!file_) {
+    fprintf(stderr, "[parsec_amd] cannot open trace file %s\n",
+            filename_.c_str());
+    return;
+  }
+  // header: magic, dictionary as a JSON line, then raw records streamed
+  // in buffer-sized chunks (event classes register at static init, before
+  // tracing starts)
+  fprintf(file_, "PABT1\n{\"classes\":{");
+  bool first = true;
+  for (auto& [id, name] : classes_) {
+    fprintf(file_, "%s\"%d\":\"%s\"", first ? "" : ",", id, name.c_str());
+    first = false;
+  }
+  fprintf(file_,
+          "},\"kinds\":{\"1\":\"exec\",\"2\":\"gpu_task\",\"3\":"
+          "\"stage_in\",\"4\":\"comm_send\",\"5\":\"comm_recv\","
+          "\"6\":\"sched\",\"7\":\"gpu_span\"},"
+          "\"rec_bytes\":%zu}\n",
+          sizeof(TraceRec));
   enabled_.store(true, std::memory_order_release);
 }
 
@@ -56,36 +78,39 @@ void Profiler::register_class(int id, const std::string& name) {
   classes_.emplace_back(id, name);
 }
 
+void Profiler::flush_buf_locked(Buf* b) {
+  if (file_ && !b->recs.empty())
+    fwrite(b->recs.data(), sizeof(TraceRec), b->recs.size(), file_);
+  b->recs.clear();
+}
+
 void Profiler::record(Ev kind, uint16_t class_id, uint64_t seq, uint64_t t0,
                       uint64_t t1) {
   Buf* b = tls_buf();
   b->recs.push_back(TraceRec{t0, t1, b->tid, (uint16_t)kind, class_id, seq});
+  if (b->recs.size() >= (1u << 16)) {
+    std::lock_guard<std::mutex> g(mtx_);
+    flush_buf_locked(b);
+  }
+}
+
+void Profiler::record_tid(Ev kind, uint16_t class_id, uint64_t seq,
+                          uint64_t t0, uint64_t t1, uint32_t tid) {
+  Buf* b = tls_buf();
+  b->recs.push_back(TraceRec{t0, t1, tid, (uint16_t)kind, class_id, seq});
+  if (b->recs.size() >= (1u << 16)) {
+    std::lock_guard<std::mutex> g(mtx_);
+    flush_buf_locked(b);
+  }
 }
 
 void Profiler::stop_and_dump() {
   if (!enabled_.exchange(false)) return;
   std::lock_guard<std::mutex> g(mtx_);
-  FILE* f = fopen(filename_.c_str(), "wb");
-  if (!f) {
-    fprintf(stderr, "[parsec_amd] cannot open trace file %s\n",
-            filename_.c_str());
-    return;
-  }
-  // header: magic, version, dictionary as a JSON line, then raw records
-  fprintf(f, "PABT1\n{\"classes\":{");
-  bool first = true;
-  for (auto& [id, name] : classes_) {
-    fprintf(f, "%s\"%d\":\"%s\"", first ? "" : ",", id, name.c_str());
-    first = false;
-  }
-  fprintf(f, "},\"kinds\":{\"1\":\"exec\",\"2\":\"gpu_task\",\"3\":"
-             "\"stage_in\",\"4\":\"comm_send\",\"5\":\"comm_recv\",\"6\":\"sched\"},"
-             "\"rec_bytes\":%zu}\n", sizeof(TraceRec));
-  for (Buf* b : bufs_) {
-    fwrite(b->recs.data(), sizeof(TraceRec), b->recs.size(), f);
-    b->recs.clear();
-  }
-  fclose(f);
+  if (!file_) return;
+  for (Buf* b : bufs_) flush_buf_locked(b);
+  fclose(file_);
+  file_ = nullptr;
 }
 
 void Profiler::dot_open(const std::string& filename) {

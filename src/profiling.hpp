@@ -30,6 +30,8 @@ enum class Ev : uint16_t {
   COMM_SEND = 4,
   COMM_RECV = 5,
   SCHED = 6,
+  GPU_SPAN = 7,  // device-side span on one exec stream (hipEvent-timed);
+                 // tid = 1000 + stream index (its own lane in the viewer)
 };
 
 struct TraceRec {
@@ -50,6 +52,9 @@ class Profiler {
 
   void record(Ev kind, uint16_t class_id, uint64_t seq, uint64_t t0_ns,
               uint64_t t1_ns);
+  // record with an explicit synthetic lane id (GPU stream rows)
+  void record_tid(Ev kind, uint16_t class_id, uint64_t seq, uint64_t t0_ns,
+                  uint64_t t1_ns, uint32_t tid);
   void register_class(int id, const std::string& name);
 
   static uint64_t now_ns();
@@ -68,9 +73,14 @@ class Profiler {
   };
   Buf* tls_buf();
 
+  void flush_buf_locked(Buf* b);
+
   std::atomic<bool> enabled_{false};
   std::string filename_;
   std::mutex mtx_;
+  FILE* file_ = nullptr;  // records stream to disk per chunk: a long traced
+                          // run must not grow host RAM (profiling.c:74-160
+                          // file-backed buffers analog)
   std::vector<Buf*> bufs_;
   std::vector<std::pair<int, std::string>> classes_;
   FILE* dot_ = nullptr;

@@ -154,6 +154,7 @@ PYBIND11_MODULE(_core, m) {
           d["bytes_h2d"] = c.gpu()->stats.bytes_h2d.load();
           d["bytes_d2h"] = c.gpu()->stats.bytes_d2h.load();
           d["evictions"] = c.gpu()->stats.evictions.load();
+          d["bytes_required"] = c.gpu()->stats.bytes_required.load();
         }
         return d;
       })

@@ -284,22 +284,50 @@ Context::~Context() {
   stop_.store(true, std::memory_order_release);
   sched_->wake_all();
   for (auto& w : workers_) w.join();
-  comm_.reset();
-  gpu_.reset();
-  Profiler::inst().stop_and_dump();
-  Profiler::inst().dot_close();
   if (param_int("stats", 0)) {
+    // device-statistics table at fini (device.c:611-658 analog):
+    // per-rank task counts, required-vs-transferred bytes, evictions,
+    // renames, per-peer comm traffic.
     RuntimeCounters& c = counters();
     fprintf(stderr,
             "[parsec_amd stats] rank %d: cpu_tasks=%lu gpu_tasks=%lu "
-            "scheduled=%lu steals=%lu comm_msgs=%lu comm_bytes=%lu\n",
+            "scheduled=%lu steals=%lu comm_msgs=%lu comm_bytes=%lu "
+            "renames=%lu\n",
             rank_, (unsigned long)c.tasks_executed_cpu.load(),
             (unsigned long)c.tasks_executed_gpu.load(),
             (unsigned long)c.tasks_scheduled.load(),
             (unsigned long)c.steals.load(),
             (unsigned long)c.comm_msgs.load(),
-            (unsigned long)c.comm_bytes.load());
+            (unsigned long)c.comm_bytes.load(),
+            (unsigned long)c.renames.load());
+    if (gpu_) {
+      auto& g = gpu_->stats;
+      fprintf(stderr,
+              "[parsec_amd stats] rank %d gpu: tasks=%lu required=%lu "
+              "h2d=%lu d2h=%lu evictions=%lu\n",
+              rank_, (unsigned long)g.tasks.load(),
+              (unsigned long)g.bytes_required.load(),
+              (unsigned long)g.bytes_h2d.load(),
+              (unsigned long)g.bytes_d2h.load(),
+              (unsigned long)g.evictions.load());
+    }
+    if (comm_) {
+      const auto& ps = comm_->peer_stats();
+      for (size_t p = 0; p < ps.size(); p++)
+        if (ps[p].sent_msgs.load() || ps[p].recv_msgs.load())
+          fprintf(stderr,
+                  "[parsec_amd stats] rank %d <-> peer %zu: sent=%lu msgs "
+                  "%lu B, recv=%lu msgs %lu B\n",
+                  rank_, p, (unsigned long)ps[p].sent_msgs.load(),
+                  (unsigned long)ps[p].sent_bytes.load(),
+                  (unsigned long)ps[p].recv_msgs.load(),
+                  (unsigned long)ps[p].recv_bytes.load());
+    }
   }
+  comm_.reset();
+  gpu_.reset();
+  Profiler::inst().stop_and_dump();
+  Profiler::inst().dot_close();
 }
 
 void Context::worker_main(int id) {

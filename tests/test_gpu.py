@@ -262,3 +262,43 @@ del A, B, ctx
     r = subprocess.run([sys.executable, "-c", code], capture_output=True,
                        text=True, timeout=600)
     assert r.returncode == 0, r.stdout + r.stderr
+
+
+def test_gpu_stream_spans_in_trace():
+    """Per-exec-stream device spans (hipEvent-timed) land in the binary
+    trace as kind-7 records with tid = 1000+stream (reference parity:
+    per-GPU profiling streams, device_cuda_module.c:509-530)."""
+    import struct as st
+    code = f"""
+import numpy as np, sys, json, struct
+sys.path.insert(0, {REPO!r})
+import parsec_amd as pm
+pm.param_set("profile_filename", "/tmp/span_trace")
+ctx = pm.Context(nworkers=2, rank=0, world=1)
+n, nb = 2048, 512
+A = pm.TiledMatrix(ctx, n, n, nb, nb, 1, 1)
+tp = pm.Dtd(ctx); pm.insert_spd_fill(tp, A, 42); tp.wait()
+tp2 = pm.Dtd(ctx); pm.insert_potrf(tp2, A); tp2.wait()
+st = ctx.gpu_stats()
+assert st["bytes_required"] > 0
+del A, ctx
+with open("/tmp/span_trace.0", "rb") as f:
+    assert f.readline().strip() == b"PABT1"
+    hdr = json.loads(f.readline())
+    raw = f.read()
+rec = struct.Struct("<QQIHHQ")
+spans = {{}}
+for off in range(0, len(raw) - rec.size + 1, rec.size):
+    t0, t1, tid, kind, cid, seq = rec.unpack_from(raw, off)
+    if kind == 7:
+        assert tid >= 1000, tid
+        assert t1 >= t0
+        spans.setdefault(tid, 0)
+        spans[tid] += 1
+assert spans, "no gpu_span records"
+print("SPANS", spans)
+"""
+    r = subprocess.run([sys.executable, "-c", code], capture_output=True,
+                       text=True, timeout=600)
+    assert r.returncode == 0, r.stdout + r.stderr
+    assert "SPANS" in r.stdout

@@ -41,7 +41,7 @@ struct FrameHeader {
   uint64_t seq;
   uint64_t size;
 };
-enum { FK_DATA = 1, FK_BAR_IN = 2, FK_BAR_OUT = 3 };
+enum { FK_DATA = 1, FK_BAR_IN = 2, FK_BAR_OUT = 3, FK_CTL = 4 };
 
 class TcpComm : public CommEngine {
  public:
@@ -65,6 +65,23 @@ class TcpComm : public CommEngine {
   }
 
   const char* kind() const override { return "tcp"; }
+
+  void send_ctl(int dst, uint32_t tag, const void* p, size_t n) override {
+    PA_CHECK(dst != rank_, "send_ctl: no self-sends");
+    {
+      std::lock_guard<std::mutex> g(cmd_mtx_);
+      ctl_out_.push_back({dst, tag, std::string((const char*)p, n)});
+    }
+    wake();
+  }
+
+  void set_ctl_handler(CtlHandler h) override {
+    std::lock_guard<std::mutex> g(ctl_mtx_);
+    ctl_handler_ = std::move(h);
+    // deliver anything that arrived before the handler was installed
+    for (auto& [src, tag, pl] : ctl_stash_) ctl_handler_(src, tag, pl);
+    ctl_stash_.clear();
+  }
 
   void enqueue(Task* t) override {
     {
@@ -225,6 +242,14 @@ class TcpComm : public CommEngine {
         unexpected_[key] = std::move(p.in_payload);
         p.in_payload.clear();
       }
+    } else if (p.hdr.kind == FK_CTL) {
+      std::string pl((const char*)p.in_payload.data(), p.in_payload.size());
+      p.in_payload.clear();
+      std::lock_guard<std::mutex> g(ctl_mtx_);
+      if (ctl_handler_)
+        ctl_handler_(peer, (uint32_t)p.hdr.seq, pl);
+      else
+        ctl_stash_.push_back({peer, (uint32_t)p.hdr.seq, std::move(pl)});
     } else if (p.hdr.kind == FK_BAR_IN) {
       bar_arrivals_[p.hdr.seq]++;
       check_barrier_root();
@@ -254,6 +279,7 @@ class TcpComm : public CommEngine {
 
   void do_read(int peer) {
     Peer& p = peers_[peer];
+    if (p.fd < 0) return;
     for (;;) {
       if (!p.in_header_done) {
         ssize_t r = read(p.fd, (uint8_t*)&p.hdr + p.hdr_got,
@@ -286,6 +312,10 @@ class TcpComm : public CommEngine {
             p.in_direct = nullptr;
             p.in_payload.resize(p.hdr.size);
           }
+        } else if (p.hdr.kind == FK_CTL) {
+          p.in_task = nullptr;
+          p.in_direct = nullptr;
+          p.in_payload.resize(p.hdr.size);
         }
         if (p.hdr.size == 0) {
           handle_frame(peer);
@@ -361,12 +391,17 @@ class TcpComm : public CommEngine {
       // drain command queue
       std::vector<Task*> cmds;
       std::vector<uint64_t> bars;
+      std::vector<CtlMsg> ctls;
       {
         std::lock_guard<std::mutex> g(cmd_mtx_);
         cmds.swap(cmds_);
         bars.swap(bar_requests_);
+        ctls.swap(ctl_out_);
       }
       for (Task* t : cmds) process_cmd(t);
+      for (auto& cm : ctls)
+        queue_frame(cm.dst, FK_CTL, cm.tag, cm.payload.data(),
+                    cm.payload.size(), nullptr);
       for (uint64_t e : bars) {
         if (rank_ == 0) {
           bar_root_armed_ = std::max(bar_root_armed_, e);
@@ -396,7 +431,9 @@ class TcpComm : public CommEngine {
         short re = pfds[pi++].revents;
         if (re & POLLIN) do_read(s);
         if (re & POLLOUT) do_write(s);
-        if (re & (POLLERR | POLLHUP)) peer_down(s, -1);
+        // POLLHUP without readable data: read() distinguishes a graceful
+        // EOF (peer tore down first, nothing pending) from a real error
+        if (re & (POLLERR | POLLHUP)) do_read(s);
       }
     }
   }
@@ -408,9 +445,17 @@ class TcpComm : public CommEngine {
   std::thread thr_;
   std::atomic<bool> stop_{false};
 
+  struct CtlMsg {
+    int dst;
+    uint32_t tag;
+    std::string payload;
+  };
   std::mutex cmd_mtx_;
   std::vector<Task*> cmds_;
   std::vector<uint64_t> bar_requests_;
+  std::vector<CtlMsg> ctl_out_;
+  std::mutex ctl_mtx_;
+  std::vector<std::tuple<int, uint32_t, std::string>> ctl_stash_;
 
   std::unordered_map<uint64_t, Task*> posted_recv_;
   std::unordered_map<uint64_t, std::vector<uint8_t>> unexpected_;
@@ -427,6 +472,10 @@ class TcpComm : public CommEngine {
 }  // namespace
 
 std::unique_ptr<CommEngine> create_rccl_comm(Context* ctx);  // rccl_comm.cpp
+
+std::unique_ptr<CommEngine> CommEngine::create_tcp(Context* ctx) {
+  return std::make_unique<TcpComm>(ctx);
+}
 
 std::unique_ptr<CommEngine> CommEngine::create(Context* ctx,
                                                const std::string& kind) {

@@ -15,6 +15,7 @@
 #pragma once
 
 #include <atomic>
+#include <functional>
 #include <memory>
 #include <string>
 #include <vector>
@@ -27,6 +28,8 @@ namespace pa {
 class CommEngine {
  public:
   static std::unique_ptr<CommEngine> create(Context* ctx, const std::string& kind);
+  // Always a TCP engine (control plane under the RCCL data engine).
+  static std::unique_ptr<CommEngine> create_tcp(Context* ctx);
   virtual ~CommEngine() = default;
 
   // Send/recv task whose local dependencies are satisfied. The engine owns
@@ -34,6 +37,18 @@ class CommEngine {
   virtual void enqueue(Task* t) = 0;
   virtual void barrier() = 0;
   virtual const char* kind() const = 0;
+
+  // ---- control messages (active-message seed, parsec_comm_engine.h
+  // AM tags analog): small tagged payloads outside the deterministic
+  // dataflow protocol — the transport for dynamic-DAG activation and
+  // distributed termination detection. Handler runs on the comm thread.
+  using CtlHandler =
+      std::function<void(int src, uint32_t tag, const std::string&)>;
+  virtual void send_ctl(int dst, uint32_t tag, const void* p, size_t n) {
+    (void)dst; (void)tag; (void)p; (void)n;
+    fatal("comm engine '%s' has no control-message path", kind());
+  }
+  virtual void set_ctl_handler(CtlHandler h) { ctl_handler_ = std::move(h); }
 
   // Per-peer traffic accounting (device stats table analog,
   // device.c:611-658: the counters that explain a scaling curve).
@@ -44,6 +59,7 @@ class CommEngine {
   const std::vector<PeerStat>& peer_stats() const { return peer_stats_; }
 
  protected:
+  CtlHandler ctl_handler_;
   void init_peer_stats(int world) {
     peer_stats_ = std::vector<PeerStat>(world);
   }

@@ -158,6 +158,28 @@ PYBIND11_MODULE(_core, m) {
         }
         return d;
       })
+      .def("send_ctl", [](Context& c, int dst, uint32_t tag, py::bytes b) {
+        PA_CHECK(c.comm(), "send_ctl: no comm engine");
+        std::string s(b);
+        py::gil_scoped_release rel;
+        c.comm()->send_ctl(dst, tag, s.data(), s.size());
+      }, py::arg("dst"), py::arg("tag"), py::arg("payload"))
+      .def("set_ctl_handler", [](Context& c, py::function fn) {
+        PA_CHECK(c.comm(), "set_ctl_handler: no comm engine");
+        PyObject* f = fn.ptr();
+        Py_XINCREF(f);
+        c.comm()->set_ctl_handler(
+            [f](int src, uint32_t tag, const std::string& pl) {
+              py::gil_scoped_acquire gil;
+              try {
+                py::handle h(f);
+                h(src, tag, py::bytes(pl));
+              } catch (py::error_already_set& e) {
+                fprintf(stderr, "[parsec_amd] ctl handler raised: %s\n",
+                        e.what());
+              }
+            });
+      })
       .def("comm_stats", [](Context& c) {
         // per-peer traffic table (device-stats analog for the comm engine)
         py::list out;

@@ -482,3 +482,71 @@ del A, ctx
     for pr in procs:
         o, _ = pr.communicate(timeout=180)
         assert pr.returncode == 0 and b"PEERSTATS_OK" in o, o.decode()
+
+
+def test_ctl_messages_dynamic_insertion(tmp_path):
+    """Control-message path (active-message seed, parsec_comm_engine.h AM
+    tags analog): rank 0 discovers work at runtime and activates tasks on
+    every other rank via tagged control messages — the non-SPMD pattern the
+    deterministic dataflow protocol alone cannot express."""
+    import conftest
+    REPO = os.path.dirname(HERE)
+    port = str(conftest.port_base(21))
+    code = """
+import os, sys, time, threading, queue
+sys.path.insert(0, %r)
+import numpy as np
+import parsec_amd as pm
+rank = int(os.environ["RANK"])
+world = 4
+pm.param_set("comm_base_port", os.environ["PORT"])
+ctx = pm.Context(nworkers=2, rank=rank, world=world, comm="tcp", gpu=-2)
+inbox = queue.Queue()
+ctx.set_ctl_handler(lambda src, tag, payload: inbox.put((src, tag, payload)))
+A = pm.TiledMatrix(ctx, 32 * world, 32, 32, 32, world, 1)
+tp = pm.Dtd(ctx)
+if rank != 0:
+    if A.is_local(rank, 0):
+        A.tile_numpy_set(rank, 0, np.zeros((32, 32)))
+results = []
+if rank == 0:
+    # runtime "discovery": instruct each peer to run a local task with a
+    # payload-carried operand, then collect their acks
+    for dst in range(1, world):
+        pm_payload = str(100 + dst).encode()
+        ctx.send_ctl(dst, 7, pm_payload)
+    acks = set()
+    while len(acks) < world - 1:
+        src, tag, payload = inbox.get(timeout=60)
+        assert tag == 9
+        acks.add((src, payload))
+    vals = sorted(int(p) for _, p in acks)
+    assert vals == [100 + d for d in range(1, world)], vals
+    print("CTL_OK", vals)
+else:
+    src, tag, payload = inbox.get(timeout=60)
+    assert src == 0 and tag == 7
+    val = float(payload)
+    t = A.tile(rank, 0)
+    def body(buf, val=val):
+        np.frombuffer(buf, dtype=np.float64)[:] = val
+    tp.insert_py(body, [(t, pm.ACCESS_OUT)], rank=rank, with_data=True)
+    tp.wait()
+    got = A.tile_numpy(rank, 0)[0, 0]
+    ctx.send_ctl(0, 9, str(int(got)).encode())
+ctx.barrier()
+del A, ctx
+""" % (REPO,)
+    import subprocess as sp
+    procs = []
+    for r in range(4):
+        env = dict(os.environ)
+        env.update(RANK=str(r), WORLD_SIZE="4", PORT=port)
+        procs.append(sp.Popen([sys.executable, "-c", code], env=env,
+                              stdout=sp.PIPE, stderr=sp.STDOUT))
+    outs = []
+    for pr in procs:
+        o, _ = pr.communicate(timeout=180)
+        outs.append(o)
+        assert pr.returncode == 0, o.decode()
+    assert b"CTL_OK" in outs[0], outs[0].decode()

@@ -14,7 +14,10 @@ Differences from the reference's jdf2c (documented design deviations):
  - dependencies are rebuilt from the IN arrows only (OUT arrows are the
    duals and are used for write-back validation), so arrow ranges on the
    output side need not be enumerated;
- - the graph is materialized per taskpool rather than iterated compactly;
+ - the graph is materialized per taskpool rather than iterated compactly
+   (measured ceiling, benchmarks/bench_ptg_scale.py on the CPU container:
+   1M instances insert+drain in ~41 s at ~913 B/task peak RSS — linear in
+   memory, so ~10M tasks per 10 GB of host RAM);
  - anti-dependencies serialize through the tile chaining engine instead of
    allocating repo copies (correct, occasionally less parallel);
  - inline-C expressions `%{ return ..; %}` compile as C++ lambdas.
@@ -176,6 +179,54 @@ def _parse_dep(direction, s):
     return Dep(direction, guard, term, else_term, props)
 
 
+def _join_continuations(lines):
+    """Join physical lines while parens/brackets are unbalanced (multi-line
+    expressions outside BODY blocks — the reference grammar is token-based
+    and accepts them anywhere; jdf2c compiler-test parity)."""
+    out = []
+    i, n = 0, len(lines)
+    in_body = False
+    while i < n:
+        line = lines[i]
+        st = line.strip()
+        if in_body:
+            out.append(line)
+            if st == "END":
+                in_body = False
+            i += 1
+            continue
+        if st.startswith("BODY"):
+            in_body = True
+            out.append(line)
+            i += 1
+            continue
+        depth = 0
+        j = i
+        acc = []
+        while j < n:
+            l2 = lines[j]
+            k = 0
+            while k < len(l2):
+                if l2.startswith("%{", k):
+                    e = l2.find("%}", k)
+                    k = len(l2) if e < 0 else e + 2
+                    continue
+                if l2[k] in "([{":
+                    depth += 1
+                elif l2[k] in ")]}":
+                    depth -= 1
+                k += 1
+            acc.append(l2)
+            if depth <= 0:
+                break
+            j += 1
+        if depth > 0:
+            raise JdfError(f"unbalanced parentheses starting at: {st!r}")
+        out.append(" ".join(a.strip() for a in acc) if len(acc) > 1 else acc[0])
+        i = j + 1
+    return out
+
+
 def parse_jdf(text):
     text = _strip_comments(text)
     jdf = Jdf()
@@ -188,7 +239,7 @@ def parse_jdf(text):
     # scan for task classes: Name(params) at line start followed by range
     # lines; everything before the first class that matches `NAME [props]`
     # is a global.
-    lines = text.split("\n")
+    lines = _join_continuations(text.split("\n"))
     i = 0
     n = len(lines)
     while i < n:
@@ -223,7 +274,53 @@ def parse_jdf(text):
             i += 1
             continue
         raise JdfError(f"unparsed top-level line: {line!r}")
+    _validate(jdf)
     return jdf
+
+
+def _validate(jdf):
+    """Semantic checks the reference's compiler enforces (ptgpp must-fail
+    test parity): duplicate names, dependency arity, parameter sanity."""
+    seen_cls = set()
+    MAXP = 8
+    for c in jdf.classes:
+        if c.name in seen_cls:
+            raise JdfError(f"duplicate task class {c.name!r}")
+        seen_cls.add(c.name)
+        if len(c.params) > MAXP:
+            raise JdfError(f"{c.name}: too many parameters "
+                           f"({len(c.params)} > {MAXP})")
+        ranged = {r[0] for r in c.ranges}
+        missing = [p for p in c.params if p not in ranged]
+        if missing:
+            raise JdfError(f"{c.name}: parameter(s) {missing} have no "
+                           "range line")
+        seen_fl = set()
+        for f in c.flows:
+            if f.name in seen_fl:
+                raise JdfError(f"{c.name}: duplicate flow {f.name!r}")
+            seen_fl.add(f.name)
+    by_name = {c.name: c for c in jdf.classes}
+    for c in jdf.classes:
+        for f in c.flows:
+            for d in f.deps:
+                for term in (d.term, d.else_term):
+                    if not term or term[0] != "task":
+                        continue
+                    tgt = by_name.get(term[2])
+                    if tgt is None:
+                        raise JdfError(
+                            f"{c.name}.{f.name}: unknown task class "
+                            f"{term[2]!r} in dependency")
+                    if len(term[3]) != len(tgt.params):
+                        raise JdfError(
+                            f"{c.name}.{f.name}: {term[2]} takes "
+                            f"{len(tgt.params)} parameter(s), dependency "
+                            f"passes {len(term[3])}")
+                    if not any(x.name == term[1] for x in tgt.flows):
+                        raise JdfError(
+                            f"{c.name}.{f.name}: {term[2]} has no flow "
+                            f"{term[1]!r}")
 
 
 def _parse_class(cls, lines, i):
@@ -267,12 +364,37 @@ def _parse_class(cls, lines, i):
             continue
         mm = flow_mode_re.match(line)
         if mm or line.startswith("<-") or line.startswith("->"):
-            # gather continuation lines of this flow
+            # gather continuation lines of this flow: further arrows, and
+            # ternary continuations ("?" lines; ":" lines only while a
+            # top-level "?" is still unmatched — otherwise ":" starts the
+            # partition line of the class)
             block = [line]
             i += 1
+            def _dangling_ternary(txt):
+                depth = q = 0
+                k = 0
+                while k < len(txt):
+                    if txt.startswith("%{", k):
+                        e = txt.find("%}", k)
+                        k = len(txt) if e < 0 else e + 2
+                        continue
+                    ch = txt[k]
+                    if ch in "([{":
+                        depth += 1
+                    elif ch in ")]}":
+                        depth -= 1
+                    elif depth == 0 and ch == "?":
+                        q += 1
+                    elif depth == 0 and ch == ":":
+                        q -= 1
+                    k += 1
+                return q > 0
             while i < n:
                 nxt = lines[i].strip()
-                if nxt.startswith("<-") or nxt.startswith("->"):
+                if (nxt.startswith("<-") or nxt.startswith("->") or
+                        nxt.startswith("?") or
+                        (nxt.startswith(":") and
+                         _dangling_ternary(" ".join(block)))):
                     block.append(nxt)
                     i += 1
                 else:

@@ -843,8 +843,15 @@ __global__ void __launch_bounds__(1024) k_qr_panel_mw(
 // T diag blocks: for block b, columns j: T(0:j,j) = -tau_j T(0:j,0:j) g
 // where g = G(blk rows, j) restricted to the block. G is V^T V of the
 // unit-lower V; T upper-triangular, T(j,j) = tau_j.
-__global__ void k_larft_diag(const double* G, int ldg, const double* tau,
-                             double* T, int ldt, int k) {
+__global__ void __launch_bounds__(128) k_larft_diag(const double* G, int ldg,
+                                                    const double* tau,
+                                                    double* T, int ldt,
+                                                    int k) {
+  // The whole 128x128 T block lives in LDS during the column recurrence
+  // (global-memory round trips per column made this kernel 1.5 ms and 9%
+  // of QR GPU time — gpurun_out/qr_kernel_stats.csv); one barrier per
+  // column, write-back once at the end.
+  __shared__ double Ts[128 * 128];
   const int b0 = blockIdx.x * 128;
   const int tid = threadIdx.x;  // 128 threads
   const int nb = min(128, k - b0);
@@ -853,16 +860,16 @@ __global__ void k_larft_diag(const double* G, int ldg, const double* tau,
     // col = -tau_j * T(0:j,0:j) * G(b0..b0+j, gj)
     double s = 0;
     if (tid < j) {
-      for (int q = tid; q < j; q++)
-        s += T[(size_t)(b0 + q) * ldt + b0 + tid] *
-             G[(size_t)gj * ldg + b0 + q];
+      const double* Gc = G + (size_t)gj * ldg + b0;
+      for (int q = tid; q < j; q++) s += Ts[q * 128 + tid] * Gc[q];
       s *= -tau[gj];
     }
-    __syncthreads();
-    if (tid < j) T[(size_t)gj * ldt + b0 + tid] = s;
-    if (tid == j) T[(size_t)gj * ldt + gj] = tau[gj];
+    if (tid < j) Ts[j * 128 + tid] = s;
+    if (tid == j) Ts[j * 128 + j] = tau[gj];
     __syncthreads();
   }
+  for (int j = 0; j < nb; j++)
+    if (tid <= j) T[(size_t)(b0 + j) * ldt + b0 + tid] = Ts[j * 128 + tid];
 }
 
 // Full tile/stacked-panel QR: factor 128-wide panels with the multi-WG

@@ -13,6 +13,7 @@
 #pragma once
 
 #include <cstdint>
+#include <tuple>
 #include <unordered_map>
 #include <vector>
 
@@ -60,6 +61,8 @@ struct Data {
   uint64_t sent_mask = 0;                // ranks already sent current version
   std::vector<int> recip_order;          // recipients of current version, in
                                          // discovery order (broadcast tree)
+  // reshape promises for the CURRENT version: {kind, consumer_rank, copy}
+  std::vector<std::tuple<uint8_t, int, Data*>> reshaped_;
   uint32_t local_present_version = 0;    // version the local buffer will hold
   bool local_present = false;
 

@@ -1,5 +1,6 @@
 #include "dtd.hpp"
 
+#include "kernels.hpp"
 #include "profiling.hpp"
 
 // Reference parity notes are in dtd.hpp's header (insert_function.c
@@ -180,6 +181,7 @@ Data* Dtd::write_flow(Data* d, Task* t, int task_rank, bool output_only) {
   d->owner_rank = R;
   d->sent_mask = 0;
   d->recip_order.clear();
+  d->reshaped_.clear();  // promises were for the previous version
   if (R == me_) {
     set_local_writer(d, t);
     d->local_present = true;
@@ -220,6 +222,11 @@ Task* Dtd::insert_begin(const TaskClass* tc, const void* args,
   }
   for (int i = 0; i < nflows; i++) {
     Data* d = flows[i].d ? current_copy(flows[i].d) : nullptr;
+    if (d && flows[i].reshape != Reshape::NONE) {
+      PA_CHECK(flows[i].mode == ACCESS_IN,
+               "reshape: only READ flows consume converted copies");
+      d = reshaped_promise(d, flows[i].reshape, task_rank);
+    }
     if (d) {  // NULL flow (e.g. absent stencil halo)
       if (flows[i].mode & ACCESS_IN) d = read_flow(d, t, task_rank);
       if (flows[i].mode & ACCESS_OUT)
@@ -244,6 +251,38 @@ void Dtd::insert_commit(Task* t) {
         std::this_thread::sleep_for(std::chrono::microseconds(50));
     }
   }
+}
+
+// Lazily materialize (once per {version, kind, consumer rank}) the
+// converted copy of `d` and return it; every consumer asking for the same
+// conversion shares it — the datacopy-future/promise semantics of
+// parsec_reshape.c, with an explicit HIP/CPU conversion task instead of
+// MPI-datatype repacking.
+Data* Dtd::reshaped_promise(Data* d, Reshape kind, int consumer_rank) {
+  for (auto& [k2, r2, rd] : d->reshaped_)
+    if (k2 == (uint8_t)kind && r2 == consumer_rank) return rd;
+  PA_CHECK(d->coll, "reshape: flow must be backed by a tiled collection");
+  TiledMatrix* A = d->coll;
+  int tm = (int)(d->key / A->nt()), tn = (int)(d->key % A->nt());
+  int rows = A->tile_rows(tm), cols = A->tile_cols(tn);
+  auto holder = std::make_shared<Data>();
+  holder->ctx_direct = context();
+  holder->home_rank = consumer_rank;
+  holder->owner_rank = consumer_rank;
+  holder->bytes = reshape_bytes(d->bytes, kind, A->elem_size());
+  Data* rd = holder.get();
+  own(std::shared_ptr<void>(holder, rd));
+  alignas(8) uint8_t args[MAX_ARGS_BYTES] = {};
+  fill_reshape_args(args, kind, rows, cols, A->mb());
+  FlowSpec f[2] = {{d, ACCESS_IN, Reshape::NONE},
+                   {rd, ACCESS_OUT, Reshape::NONE}};
+  insert(&tc_reshape(), args, reshape_args_bytes(), f, 2, 1 << 16,
+         consumer_rank);
+  // the nested insert may have renamed d (remote fetch): record the
+  // promise on the CURRENT copy
+  Data* cur = d->coll ? d->coll->current_by_key(d->key) : d;
+  cur->reshaped_.push_back({(uint8_t)kind, consumer_rank, rd});
+  return rd;
 }
 
 void Dtd::flush(Data* d) {

@@ -26,6 +26,17 @@ namespace pa {
 extern TaskClass COMM_SEND_CLASS;
 extern TaskClass COMM_RECV_CLASS;
 
+// Per-flow reshape kinds (reshape-promise engine, parsec_reshape.c:1-786
+// analog — see src/kernels_reshape.cpp).
+enum class Reshape : uint8_t {
+  NONE = 0,
+  TRANSPOSE,  // fp64 m x n -> n x m
+  TO_BF16,    // fp64 -> bf16 (same shape)
+  FROM_BF16,  // bf16 -> fp64
+  TRIL,       // fp64, keep lower triangle, zero above
+  TRIU,       // fp64, keep upper triangle, zero below
+};
+
 class Dtd : public Taskpool {
  public:
   Dtd(Context* ctx, std::string name = "dtd");
@@ -34,6 +45,10 @@ class Dtd : public Taskpool {
   struct FlowSpec {
     Data* d;
     AccessMode mode;
+    // READ flows only: consume a CONVERTED copy of the producer's version
+    // (lazily materialized once per {version, kind, consumer rank} and
+    // shared by all consumers — the reference's reshape "promises").
+    Reshape reshape = Reshape::NONE;
   };
 
   // Insert one task. `rank` -1 selects the home rank of the first written
@@ -66,6 +81,7 @@ class Dtd : public Taskpool {
   // task once its readers drain — instead of WAR-serializing the transfer
   // behind every reader.
   Data* read_flow(Data* d, Task* t, int task_rank);
+  Data* reshaped_promise(Data* d, Reshape kind, int consumer_rank);
   Data* write_flow(Data* d, Task* t, int task_rank, bool output_only);
   Data* maybe_rename(Data* d);
   void make_send(Data* d, int dst, uint64_t seq);

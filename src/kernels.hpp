@@ -14,6 +14,12 @@
 
 namespace pa {
 
+enum class Reshape : uint8_t;
+TaskClass& tc_reshape();
+size_t reshape_bytes(size_t src_bytes, Reshape kind, size_t src_elem);
+void fill_reshape_args(void* argbuf, Reshape kind, int m, int n, int ld);
+size_t reshape_args_bytes();
+
 struct TileArgs {
   int m = 0, n = 0, k = 0, ld = 0;
   int64_t i0 = 0, j0 = 0, N = 0;

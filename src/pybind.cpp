@@ -359,14 +359,19 @@ PYBIND11_MODULE(_core, m) {
       .def(py::init<Context*, std::string>(), py::arg("ctx"),
            py::arg("name") = std::string("dtd"), py::keep_alive<1, 2>())
       .def("insert_py",
-           [](Dtd& tp, py::function fn,
-              std::vector<std::pair<Data*, int>> flows, int priority,
+           [](Dtd& tp, py::function fn, py::list flows, int priority,
               int rank, bool with_data) {
              PyTaskPayload pl{fn.ptr(), with_data ? 1 : 0};
              Py_XINCREF(pl.fn);
              std::vector<Dtd::FlowSpec> fs;
-             for (auto& [d, mode] : flows)
-               fs.push_back({d, (AccessMode)mode});
+             for (auto h : flows) {
+               auto t2 = h.cast<py::tuple>();
+               Dtd::FlowSpec f{t2[0].cast<Data*>(),
+                               (AccessMode)t2[1].cast<int>(),
+                               Reshape::NONE};
+               if (t2.size() > 2) f.reshape = (Reshape)t2[2].cast<int>();
+               fs.push_back(f);
+             }
              bool local;
              {
                py::gil_scoped_release rel;
@@ -376,7 +381,7 @@ PYBIND11_MODULE(_core, m) {
              // remote tasks never run here: release the body's ref
              if (!local) Py_XDECREF(pl.fn);
            },
-           py::arg("fn"), py::arg("flows") = std::vector<std::pair<Data*, int>>{},
+           py::arg("fn"), py::arg("flows") = py::list(),
            py::arg("priority") = 0, py::arg("rank") = -1,
            py::arg("with_data") = false)
       .def("flush", &Dtd::flush, py::call_guard<py::gil_scoped_release>())
@@ -386,6 +391,11 @@ PYBIND11_MODULE(_core, m) {
   m.attr("ACCESS_IN") = (int)ACCESS_IN;
   m.attr("ACCESS_OUT") = (int)ACCESS_OUT;
   m.attr("ACCESS_INOUT") = (int)ACCESS_INOUT;
+  m.attr("RESHAPE_TRANSPOSE") = (int)Reshape::TRANSPOSE;
+  m.attr("RESHAPE_TO_BF16") = (int)Reshape::TO_BF16;
+  m.attr("RESHAPE_FROM_BF16") = (int)Reshape::FROM_BF16;
+  m.attr("RESHAPE_TRIL") = (int)Reshape::TRIL;
+  m.attr("RESHAPE_TRIU") = (int)Reshape::TRIU;
 
   m.def("bench_qr_factor", &bench_qr_factor, py::arg("m"), py::arg("k"),
         py::arg("ts_split"), py::arg("iters"), py::arg("mode"),

@@ -68,6 +68,8 @@ class TcpComm : public CommEngine {
 
   void send_ctl(int dst, uint32_t tag, const void* p, size_t n) override {
     PA_CHECK(dst != rank_, "send_ctl: no self-sends");
+    if (!(tag & CTL_SYS_BIT))
+      ctl_sent_.fetch_add(1, std::memory_order_relaxed);
     {
       std::lock_guard<std::mutex> g(cmd_mtx_);
       ctl_out_.push_back({dst, tag, std::string((const char*)p, n)});
@@ -81,6 +83,13 @@ class TcpComm : public CommEngine {
     // deliver anything that arrived before the handler was installed
     for (auto& [src, tag, pl] : ctl_stash_) ctl_handler_(src, tag, pl);
     ctl_stash_.clear();
+  }
+
+  void set_sys_handler(CtlHandler h) override {
+    std::lock_guard<std::mutex> g(ctl_mtx_);
+    sys_handler_ = std::move(h);
+    for (auto& [src, tag, pl] : sys_stash_) sys_handler_(src, tag, pl);
+    sys_stash_.clear();
   }
 
   void enqueue(Task* t) override {
@@ -245,11 +254,20 @@ class TcpComm : public CommEngine {
     } else if (p.hdr.kind == FK_CTL) {
       std::string pl((const char*)p.in_payload.data(), p.in_payload.size());
       p.in_payload.clear();
+      uint32_t tag = (uint32_t)p.hdr.seq;
       std::lock_guard<std::mutex> g(ctl_mtx_);
-      if (ctl_handler_)
-        ctl_handler_(peer, (uint32_t)p.hdr.seq, pl);
-      else
-        ctl_stash_.push_back({peer, (uint32_t)p.hdr.seq, std::move(pl)});
+      if (tag & CTL_SYS_BIT) {
+        if (sys_handler_)
+          sys_handler_(peer, tag, pl);
+        else
+          sys_stash_.push_back({peer, tag, std::move(pl)});
+      } else {
+        ctl_recvd_.fetch_add(1, std::memory_order_relaxed);
+        if (ctl_handler_)
+          ctl_handler_(peer, tag, pl);
+        else
+          ctl_stash_.push_back({peer, tag, std::move(pl)});
+      }
     } else if (p.hdr.kind == FK_BAR_IN) {
       bar_arrivals_[p.hdr.seq]++;
       check_barrier_root();
@@ -456,6 +474,7 @@ class TcpComm : public CommEngine {
   std::vector<CtlMsg> ctl_out_;
   std::mutex ctl_mtx_;
   std::vector<std::tuple<int, uint32_t, std::string>> ctl_stash_;
+  std::vector<std::tuple<int, uint32_t, std::string>> sys_stash_;
 
   std::unordered_map<uint64_t, Task*> posted_recv_;
   std::unordered_map<uint64_t, std::vector<uint8_t>> unexpected_;

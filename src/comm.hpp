@@ -44,11 +44,23 @@ class CommEngine {
   // distributed termination detection. Handler runs on the comm thread.
   using CtlHandler =
       std::function<void(int src, uint32_t tag, const std::string&)>;
+  // Tags with the high bit set are SYSTEM messages (termination detection);
+  // they route to the sys handler and are excluded from traffic counts so
+  // quiescence waves can stabilize.
+  static constexpr uint32_t CTL_SYS_BIT = 0x80000000u;
   virtual void send_ctl(int dst, uint32_t tag, const void* p, size_t n) {
     (void)dst; (void)tag; (void)p; (void)n;
     fatal("comm engine '%s' has no control-message path", kind());
   }
   virtual void set_ctl_handler(CtlHandler h) { ctl_handler_ = std::move(h); }
+  virtual void set_sys_handler(CtlHandler h) { sys_handler_ = std::move(h); }
+  // user (non-system) control traffic counters, for quiescence accounting
+  virtual uint64_t ctl_sent() const {
+    return ctl_sent_.load(std::memory_order_relaxed);
+  }
+  virtual uint64_t ctl_recvd() const {
+    return ctl_recvd_.load(std::memory_order_relaxed);
+  }
 
   // Per-peer traffic accounting (device stats table analog,
   // device.c:611-658: the counters that explain a scaling curve).
@@ -60,6 +72,8 @@ class CommEngine {
 
  protected:
   CtlHandler ctl_handler_;
+  CtlHandler sys_handler_;
+  std::atomic<uint64_t> ctl_sent_{0}, ctl_recvd_{0};
   void init_peer_stats(int world) {
     peer_stats_ = std::vector<PeerStat>(world);
   }

@@ -332,6 +332,8 @@ PYBIND11_MODULE(_core, m) {
       });
 
   py::class_<Taskpool>(m, "Taskpool")
+      .def("wait_dynamic", &Taskpool::wait_dynamic,
+           py::call_guard<py::gil_scoped_release>())
       .def_property_readonly("_handle", [](Taskpool& t) { return (uintptr_t)&t; })
       .def("wait", &Taskpool::wait, py::call_guard<py::gil_scoped_release>())
       .def_property_readonly("pending", &Taskpool::pending)

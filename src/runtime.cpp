@@ -1,5 +1,7 @@
 #include "runtime.hpp"
 
+#include <cstring>
+
 #include "data.hpp"
 #include "device_gpu.hpp"
 #include "comm.hpp"
@@ -148,6 +150,97 @@ void Taskpool::wait() {
       cv_.wait_for(g, std::chrono::microseconds(200));
     }
   }
+}
+
+void Taskpool::wait_dynamic() {
+  CommEngine* ce = ctx_->comm();
+  if (!ce || ctx_->world() == 1) {
+    wait();
+    return;
+  }
+  constexpr uint32_t SYS = 0x80000000u;
+  constexpr uint32_t TD_PROBE = SYS | 1, TD_REPLY = SYS | 2,
+                     TD_DONE = SYS | 3;
+  const int world = ctx_->world(), rank = ctx_->rank();
+  struct TdState {
+    std::mutex m;
+    std::condition_variable cv;
+    uint64_t wave = 0;
+    int got = 0;
+    uint64_t sum_p = 0, sum_s = 0, sum_r = 0;
+    std::atomic<bool> done{false};
+  };
+  auto st = std::make_shared<TdState>();
+  auto triple = [this, ce](uint64_t* v) {
+    v[0] = (uint64_t)pending();
+    v[1] = ce->ctl_sent();
+    v[2] = ce->ctl_recvd();
+  };
+  ce->set_sys_handler([this, st, ce, rank, triple](int src, uint32_t tag,
+                                                   const std::string& pl) {
+    if (tag == TD_PROBE) {
+      uint64_t rep[4];
+      memcpy(rep, pl.data(), 8);  // wave
+      triple(rep + 1);
+      ce->send_ctl(src, SYS | 2, rep, sizeof(rep));
+    } else if (tag == TD_REPLY) {
+      uint64_t rep[4];
+      memcpy(rep, pl.data(), sizeof(rep));
+      std::lock_guard<std::mutex> g(st->m);
+      if (rep[0] == st->wave) {
+        st->got++;
+        st->sum_p += rep[1];
+        st->sum_s += rep[2];
+        st->sum_r += rep[3];
+        st->cv.notify_all();
+      }
+    } else if (tag == TD_DONE) {
+      st->done.store(true, std::memory_order_release);
+    }
+  });
+  if (rank == 0) {
+    uint64_t prev_s = ~0ull, prev_r = ~0ull;
+    bool prev_valid = false;
+    for (;;) {
+      while (pending() > 0)
+        if (!ctx_->progress_one())
+          std::this_thread::sleep_for(std::chrono::microseconds(100));
+      uint64_t wave;
+      {
+        std::lock_guard<std::mutex> g(st->m);
+        wave = ++st->wave;
+        st->got = 0;
+        st->sum_p = st->sum_s = st->sum_r = 0;
+      }
+      for (int d = 1; d < world; d++)
+        ce->send_ctl(d, TD_PROBE, &wave, sizeof(wave));
+      std::unique_lock<std::mutex> g(st->m);
+      st->cv.wait_for(g, std::chrono::seconds(60),
+                      [&] { return st->got == world - 1; });
+      PA_CHECK(st->got == world - 1, "wait_dynamic: probe wave timed out");
+      uint64_t own[3];
+      triple(own);
+      uint64_t P = st->sum_p + own[0];
+      uint64_t S = st->sum_s + own[1];
+      uint64_t R = st->sum_r + own[2];
+      g.unlock();
+      if (P == 0 && S == R && prev_valid && S == prev_s && R == prev_r) {
+        for (int d = 1; d < world; d++)
+          ce->send_ctl(d, TD_DONE, nullptr, 0);
+        break;
+      }
+      prev_s = S;
+      prev_r = R;
+      prev_valid = true;
+      std::this_thread::sleep_for(std::chrono::milliseconds(1));
+    }
+  } else {
+    while (!st->done.load(std::memory_order_acquire))
+      if (!ctx_->progress_one())
+        std::this_thread::sleep_for(std::chrono::microseconds(200));
+  }
+  wait();  // belt-and-braces local drain
+  ce->set_sys_handler(CommEngine::CtlHandler{});
 }
 
 // ------------------------------------------------------------------ Scheduler

@@ -146,6 +146,13 @@ class Taskpool {
   void task_created();             // nb_pending++
   void task_done();                // nb_pending--; wake waiters at 0
   void wait();                     // drain; main thread participates in progress
+  // Distributed termination detection for DYNAMIC task graphs (tasks may
+  // be activated by control messages at any time, so no rank knows its
+  // final task count up front — mca/termdet fourcounter analog): a
+  // coordinator runs quiescence waves over system control messages and
+  // declares completion after two stable waves with zero pending tasks
+  // and matched control-message counts. One taskpool at a time.
+  void wait_dynamic();
   int64_t pending() const { return nb_pending_.load(std::memory_order_acquire); }
 
   uint64_t next_seq() { return seq_++; }

@@ -550,3 +550,56 @@ del A, ctx
         outs.append(o)
         assert pr.returncode == 0, o.decode()
     assert b"CTL_OK" in outs[0], outs[0].decode()
+
+
+def test_wait_dynamic_token_ring(tmp_path):
+    """Dynamic termination detection (mca/termdet fourcounter analog):
+    a control-message token ring activates tasks on whichever rank holds
+    the token — no rank knows its task count up front — and
+    Taskpool.wait_dynamic() must terminate exactly when the ring quiesces."""
+    import conftest
+    REPO = os.path.dirname(HERE)
+    port = str(conftest.port_base(29))
+    code = """
+import os, sys
+sys.path.insert(0, %r)
+import numpy as np
+import parsec_amd as pm
+rank = int(os.environ["RANK"])
+world = 4
+pm.param_set("comm_base_port", os.environ["PORT"])
+ctx = pm.Context(nworkers=2, rank=rank, world=world, comm="tcp", gpu=-2)
+A = pm.TiledMatrix(ctx, 32 * world, 32, 32, 32, world, 1)
+if A.is_local(rank, 0):
+    A.tile_numpy_set(rank, 0, np.zeros((32, 32)))
+tp = pm.Dtd(ctx)
+LAPS = 3
+def on_token(src, tag, payload):
+    hops = int(payload)
+    t = A.tile(rank, 0)
+    def body(buf, hops=hops):
+        np.frombuffer(buf, dtype=np.float64)[:] += 1
+        nxt = (rank + 1) %% world
+        if hops > 1:
+            ctx.send_ctl(nxt, 5, str(hops - 1).encode())
+    tp.insert_py(body, [(t, pm.ACCESS_INOUT)], rank=rank, with_data=True)
+ctx.set_ctl_handler(on_token)
+if rank == 0:
+    on_token(0, 5, str(LAPS * world).encode())
+tp.wait_dynamic()
+got = A.tile_numpy(rank, 0)[0, 0] if A.is_local(rank, 0) else None
+assert got == LAPS, (rank, got)
+print("TDYN_OK", rank, got)
+ctx.barrier()
+del A, ctx
+""" % (REPO,)
+    import subprocess as sp
+    procs = []
+    for r in range(4):
+        env = dict(os.environ)
+        env.update(RANK=str(r), WORLD_SIZE="4", PORT=port)
+        procs.append(sp.Popen([sys.executable, "-c", code], env=env,
+                              stdout=sp.PIPE, stderr=sp.STDOUT))
+    for pr in procs:
+        o, _ = pr.communicate(timeout=180)
+        assert pr.returncode == 0 and b"TDYN_OK" in o, o.decode()

@@ -281,7 +281,7 @@ tp = pm.Dtd(ctx); pm.insert_spd_fill(tp, A, 42); tp.wait()
 tp2 = pm.Dtd(ctx); pm.insert_potrf(tp2, A); tp2.wait()
 st = ctx.gpu_stats()
 assert st["bytes_required"] > 0
-del A, ctx
+del tp, tp2, A, ctx  # keep_alive: the context (and trace dump) waits on them
 with open("/tmp/span_trace.0", "rb") as f:
     assert f.readline().strip() == b"PABT1"
     hdr = json.loads(f.readline())

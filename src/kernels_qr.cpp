@@ -902,13 +902,37 @@ static void qr_factor_hand(GpuTaskCtx& g, double* A, int m, int k, int ld,
     int rows = len0 + len1;
     int W = rows <= QR_MAX_ROWS_W16 ? 16 : (rows <= QR_MAX_ROWS_W8 ? 8 : 0);
     if (W) {
-      // the panel kernel both factors and applies to ALL remaining tile
-      // columns — no per-panel T-build / larfb dgemms on the stream
+      // qr_apply=kernel (default): the panel kernel both factors and
+      // applies to ALL remaining tile columns (no per-panel dgemms).
+      // qr_apply=gemm: the kernel applies within the 128 panel columns
+      // only; the trailing matrix goes through T128 + larfb dgemms at
+      // Tensile rates (A/B: the in-kernel apply units are latency-bound,
+      // profiles/qr_round2.md).
+      static const bool apply_gemm =
+          param_str("qr_apply", "kernel") == "gemm";
+      int rest = k - p - pc;
+      int apply_cols = apply_gemm ? pc : k - p;
       PA_HIP_CHECK(hipMemsetAsync(cnt, 0, 2 * sizeof(int), g.stream));
       hipLaunchKernelGGL(k_qr_panel_mw, dim3(nwg), dim3(1024), 0, g.stream,
-                         A, ld, p, base0, len0, base1, len1, k - p, pc, W,
-                         tau + p, T16s, cnt, nwg, nA,
+                         A, ld, p, base0, len0, base1, len1, apply_cols, pc,
+                         W, tau + p, T16s, cnt, nwg, nA,
                          (unsigned long long*)nullptr);
+      if (apply_gemm && rest > 0) {
+        double* V = qr_scratch(g, slot0 + 1, (size_t)prows * 128 * 8);
+        double* G = qr_scratch(g, slot0 + 2, (size_t)128 * 128 * 8);
+        hipLaunchKernelGGL(k_unitlow, grid1d(prows * pc), dim3(256), 0,
+                           g.stream, V, panel, prows, pc, ld, prows);
+        PA_CHECK(rocblas_dgemm(h, rocblas_operation_transpose,
+                               rocblas_operation_none, pc, pc, prows, &one,
+                               V, prows, V, prows, &zero, G,
+                               pc) == rocblas_status_success);
+        double* T128 = qr_scratch(g, slot0 + 3, (size_t)128 * 128 * 8);
+        PA_HIP_CHECK(hipMemsetAsync(T128, 0, (size_t)pc * pc * 8, g.stream));
+        hipLaunchKernelGGL(k_larft_diag, dim3(1), dim3(128), 0, g.stream, G,
+                           pc, tau + p, T128, pc, pc);
+        larfb_gemm(g, prows, rest, pc, panel, ld, T128, pc,
+                   A + (size_t)(p + pc) * ld + p, ld, slot0 + 4);
+      }
       continue;
     }
     hipLaunchKernelGGL(k_geqr2, dim3(1), dim3(1024), 0, g.stream, panel,

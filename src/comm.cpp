@@ -20,6 +20,7 @@
 #include <unordered_map>
 
 #include "data.hpp"
+#include "pins.hpp"
 #include "profiling.hpp"
 
 namespace pa {
@@ -197,6 +198,7 @@ class TcpComm : public CommEngine {
   }
 
   void process_cmd(Task* t) {
+    PA_PINS(PinsEv::COMM_POST, t, -1);
     Data* d = t->flows[0].data;
     // stamp post time for the trace (args space is unused by comm tasks)
     *(uint64_t*)t->args = Profiler::now_ns();
@@ -243,6 +245,7 @@ class TcpComm : public CommEngine {
         if (pr.enabled())
           pr.record(Ev::COMM_RECV, (uint16_t)peer, p.hdr.seq,
                     *(uint64_t*)p.in_task->args, Profiler::now_ns());
+        PA_PINS(PinsEv::COMM_DONE, p.in_task, -1);
         task_complete(p.in_task);
         p.in_task = nullptr;
         p.in_direct = nullptr;

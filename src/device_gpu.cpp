@@ -399,6 +399,7 @@ void GpuEngine::stage_in_and_launch(Task* t) {
   }
   hipStream_t es = exec_streams_[si];
   stage_flows(t, es);
+  PA_PINS(PinsEv::STAGE_IN, t, -1);
   std::vector<std::pair<void*, size_t>> deferred;
   GpuTaskCtx gctx{es, device_, this, &deferred};
   if (debug_history_on())

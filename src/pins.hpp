@@ -24,13 +24,19 @@ namespace pa {
 struct Task;
 
 enum class PinsEv : int {
-  EXEC_BEGIN = 0,   // CPU body about to run (worker thread)
-  EXEC_END = 1,     // CPU body returned
-  SCHEDULE = 2,     // task became ready and was pushed to a queue
-  COMPLETE = 3,     // task completed (deps released)
-  GPU_SUBMIT = 4,   // GPU hook enqueued work on its exec stream
-  GPU_RETIRE = 5,   // GPU task's completion event retired
-  COUNT = 6,
+  EXEC_BEGIN = 0,    // CPU body about to run (worker thread)
+  EXEC_END = 1,      // CPU body returned
+  SCHEDULE = 2,      // task became ready and was pushed to a queue
+  COMPLETE = 3,      // task completed (deps released)
+  GPU_SUBMIT = 4,    // GPU hook enqueued work on its exec stream
+  GPU_RETIRE = 5,    // GPU task's completion event retired
+  CREATE = 6,        // task object created (insertion)
+  RELEASE_DEPS = 7,  // successors of a completed task were released
+  STAGE_IN = 8,      // GPU stage-in for a task's flows finished issuing
+  COMM_POST = 9,     // send/recv posted to the wire engine
+  COMM_DONE = 10,    // wire transfer completed
+  STEAL = 11,        // task obtained from another worker's queue
+  COUNT = 12,
 };
 
 using PinsFn = std::function<void(PinsEv, const Task*, int worker)>;

@@ -455,7 +455,11 @@ PYBIND11_MODULE(_core, m) {
     static const std::map<std::string, PinsEv> names = {
         {"exec_begin", PinsEv::EXEC_BEGIN}, {"exec_end", PinsEv::EXEC_END},
         {"schedule", PinsEv::SCHEDULE},     {"complete", PinsEv::COMPLETE},
-        {"gpu_submit", PinsEv::GPU_SUBMIT}, {"gpu_retire", PinsEv::GPU_RETIRE}};
+        {"gpu_submit", PinsEv::GPU_SUBMIT}, {"gpu_retire", PinsEv::GPU_RETIRE},
+        {"create", PinsEv::CREATE},
+        {"release_deps", PinsEv::RELEASE_DEPS},
+        {"stage_in", PinsEv::STAGE_IN},     {"comm_post", PinsEv::COMM_POST},
+        {"comm_done", PinsEv::COMM_DONE},   {"steal", PinsEv::STEAL}};
     uint32_t mask = 0;
     for (auto& e : events) {
       auto it = names.find(e);
@@ -469,8 +473,10 @@ PYBIND11_MODULE(_core, m) {
           py::gil_scoped_acquire g;
           delete f;
         });
-    static const char* evname[] = {"exec_begin", "exec_end", "schedule",
-                                   "complete",   "gpu_submit", "gpu_retire"};
+    static const char* evname[] = {"exec_begin", "exec_end",   "schedule",
+                                   "complete",   "gpu_submit", "gpu_retire",
+                                   "create",     "release_deps", "stage_in",
+                                   "comm_post",  "comm_done",  "steal"};
     return Pins::inst().add(
         [keep](PinsEv e, const Task* t, int worker) {
           py::gil_scoped_acquire g;

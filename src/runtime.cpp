@@ -16,12 +16,66 @@ namespace pa {
 thread_local int Context::tls_worker_id = -1;
 
 // ------------------------------------------------------------------ Task
+// Task mempool (mempool.c / private_mempool analog): tasks churn at
+// hundreds of thousands per second, so allocation goes through a shared
+// pool with per-thread caches instead of the global heap. Recycled tasks
+// keep their succs vector capacity — the dependency engine's only
+// per-task allocation amortizes away.
+namespace {
+constexpr size_t TASK_CACHE_MAX = 128, TASK_CACHE_BATCH = 64;
+SpinLock g_task_pool_lock;
+std::vector<Task*> g_task_pool;
+thread_local std::vector<Task*> t_task_cache;
+
+Task* task_alloc() {
+  if (t_task_cache.empty()) {
+    SpinGuard g(g_task_pool_lock);
+    size_t take = std::min(TASK_CACHE_BATCH, g_task_pool.size());
+    for (size_t i = 0; i < take; i++) {
+      t_task_cache.push_back(g_task_pool.back());
+      g_task_pool.pop_back();
+    }
+  }
+  if (!t_task_cache.empty()) {
+    Task* t = t_task_cache.back();
+    t_task_cache.pop_back();
+    return t;
+  }
+  return new Task();
+}
+
+void task_recycle(Task* t) {
+  // reset to construction state (succs keeps its capacity)
+  t->tp = nullptr;
+  t->tc = nullptr;
+  t->priority = 0;
+  t->nflows = 0;
+  for (auto& f : t->flows) f = FlowRef{};
+  t->deps_remaining.store(1, std::memory_order_relaxed);
+  t->completed = false;
+  t->succs.clear();
+  t->refcnt.store(1, std::memory_order_relaxed);
+  t->peer = -1;
+  t->comm_seq = 0;
+  memset(t->dev_ptr, 0, sizeof(t->dev_ptr));
+  if (t_task_cache.size() >= TASK_CACHE_MAX) {
+    SpinGuard g(g_task_pool_lock);
+    for (size_t i = 0; i < TASK_CACHE_BATCH; i++) {
+      g_task_pool.push_back(t_task_cache.back());
+      t_task_cache.pop_back();
+    }
+  }
+  t_task_cache.push_back(t);
+}
+}  // namespace
+
 Task* task_new(Taskpool* tp, const TaskClass* tc) {
-  Task* t = new Task();
+  Task* t = task_alloc();
   t->tp = tp;
   t->tc = tc;
   t->seq = tp->next_seq();
   tp->task_created();
+  PA_PINS(PinsEv::CREATE, t, Context::tls_worker_id);
   Profiler& pr = Profiler::inst();
   if (pr.dot_enabled()) pr.dot_node(t);
   return t;
@@ -30,7 +84,7 @@ Task* task_new(Taskpool* tp, const TaskClass* tc) {
 void Task::release() {
   if (refcnt.fetch_sub(1, std::memory_order_acq_rel) == 1) {
     if (tc->destruct) tc->destruct(*this);
-    delete this;
+    task_recycle(this);
   }
 }
 
@@ -64,6 +118,7 @@ void task_complete(Task* t) {
   succs.swap(t->succs);
   t->lock.unlock();
   for (Task* s : succs) task_dec_deps(s);
+  PA_PINS(PinsEv::RELEASE_DEPS, t, Context::tls_worker_id);
   if (debug_history_on())
     debug_history_add("done %s seq=%lu", t->tc->name.c_str(),
                       (unsigned long)t->seq);
@@ -246,17 +301,32 @@ void Taskpool::wait_dynamic() {
 // ------------------------------------------------------------------ Scheduler
 Scheduler::Scheduler(int nworkers) : nworkers_(nworkers) {
   // MCA "sched" analog (mca/sched/*): ws = per-worker deques + steal (lfq
-  // style, default); fifo/lifo = one shared queue (gd/ll styles). The
-  // priority queue always serves prioritized and externally-released tasks.
+  // style, default); fifo/lifo = one shared queue (gd/ll styles); spq =
+  // everything through the shared priority queue (sched_spq/ap analog);
+  // rnd = random victim order (sched_rnd, for schedule-robustness tests).
+  // The priority queue always serves prioritized and externally-released
+  // tasks.
   std::string kind = param_str("sched", "ws");
-  mode_ = kind == "fifo" ? 1 : kind == "lifo" ? 2 : 0;
+  mode_ = kind == "fifo" ? 1
+          : kind == "lifo" ? 2
+          : kind == "spq" ? 3
+          : kind == "rnd" ? 4
+          : 0;
   for (int i = 0; i < nworkers_; i++) wq_.emplace_back(new WorkerQ());
 }
 
 Scheduler::~Scheduler() = default;
 
 void Scheduler::push(Task* t, int worker_hint) {
-  if (mode_ != 0) {
+  if (mode_ == 3) {  // spq: single shared priority queue
+    pq_lock_.lock();
+    pq_.push(PQEntry{t});
+    pq_lock_.unlock();
+    npending_.fetch_add(1, std::memory_order_release);
+    sleep_cv_.notify_one();
+    return;
+  }
+  if (mode_ == 1 || mode_ == 2) {
     WorkerQ& q = *wq_[0];
     q.lock.lock();
     if (mode_ == 2) q.dq.push_front(t);
@@ -284,7 +354,14 @@ void Scheduler::push(Task* t, int worker_hint) {
 
 Task* Scheduler::pop(int worker) {
   Task* t = nullptr;
-  if (mode_ != 0) {
+  if (mode_ == 3) {
+    pq_lock_.lock();
+    if (!pq_.empty()) { t = pq_.top().t; pq_.pop(); }
+    pq_lock_.unlock();
+    if (t) npending_.fetch_sub(1, std::memory_order_relaxed);
+    return t;
+  }
+  if (mode_ == 1 || mode_ == 2) {
     WorkerQ& q = *wq_[0];
     q.lock.lock();
     if (!q.dq.empty()) {
@@ -316,9 +393,18 @@ Task* Scheduler::pop(int worker) {
   if (!pq_.empty()) { t = pq_.top().t; pq_.pop(); }
   pq_lock_.unlock();
   if (t) { npending_.fetch_sub(1, std::memory_order_relaxed); return t; }
-  // steal (FIFO end) from other workers, nearest-first
+  // steal (FIFO end) from other workers: nearest-first (hwloc-distance
+  // analog on one NUMA node), or a per-call random ordering (sched=rnd)
+  // that shakes out order-dependent bugs in tests
+  uint32_t rot = 1;
+  if (mode_ == 4) {
+    static thread_local uint32_t rng = 0x9E3779B9u ^ (uint32_t)(uintptr_t)&t;
+    rng = rng * 1664525u + 1013904223u;
+    rot = 1 + rng % (uint32_t)(nworkers_ > 1 ? nworkers_ - 1 : 1);
+  }
   for (int d = 1; d < nworkers_; d++) {
-    int v = (worker >= 0 ? (worker + d) % nworkers_ : d - 1);
+    int dd = (int)((d * rot - 1) % (uint32_t)(nworkers_ > 1 ? nworkers_ : 1)) + 1;
+    int v = (worker >= 0 ? (worker + dd) % nworkers_ : dd - 1);
     WorkerQ& q = *wq_[v];
     if (q.sz.load(std::memory_order_relaxed) == 0) continue;
     q.lock.lock();
@@ -331,6 +417,7 @@ Task* Scheduler::pop(int worker) {
     if (t) {
       npending_.fetch_sub(1, std::memory_order_relaxed);
       counters().steals.fetch_add(1, std::memory_order_relaxed);
+      PA_PINS(PinsEv::STEAL, t, worker);
       return t;
     }
   }

@@ -385,3 +385,51 @@ def test_zone_alloc_fragmentation_reuse():
         z.free(o, sz)
     assert z.in_use == 0 and z.free_blocks == 1
     assert z.alloc(1 << 16) == 0
+
+
+def test_scheduler_modules():
+    """MCA sched module variants (mca/sched analogs): spq (one shared
+    priority queue) and rnd (random victim order) run the same DAG to the
+    same result."""
+    import subprocess, sys
+    repo = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+    for kind in ("spq", "rnd", "fifo", "lifo", "ws"):
+        code = f"""
+import sys; sys.path.insert(0, {repo!r})
+import parsec_amd as pm
+pm.param_set("sched", {kind!r})
+ctx = pm.Context(nworkers=4, rank=0, world=1, gpu=-2)
+tp = pm.Dtd(ctx)
+acc = []
+for i in range(500):
+    tp.insert_py(lambda i=i: acc.append(i))
+tp.wait()
+assert len(acc) == 500
+print("SCHED_OK", {kind!r})
+del ctx
+"""
+        r = subprocess.run([sys.executable, "-c", code],
+                           capture_output=True, text=True, timeout=120)
+        assert r.returncode == 0 and "SCHED_OK" in r.stdout, (
+            kind, r.stdout + r.stderr)
+
+
+def test_pins_extended_events():
+    """The extended PINS lifecycle set (create/release_deps/steal...)
+    fires through the callback chain (pins.h:26-55 17-event parity)."""
+    ctx = pm.Context(nworkers=2, rank=0, world=1, gpu=-2)
+    seen = set()
+    # CREATE=6, RELEASE_DEPS=7 (see src/pins.hpp)
+    h = pm.pins_add(lambda ev, name, wk: seen.add(ev),
+                    ["create", "release_deps", "schedule", "complete"])
+    tp = pm.Dtd(ctx)
+    A = pm.TiledMatrix(ctx, 32, 32, 32, 32, 1, 1)
+    import numpy as np
+    A.tile_numpy_set(0, 0, np.zeros((32, 32)))
+    for _ in range(4):
+        tp.insert_py(lambda x: None, [(A.tile(0, 0), pm.ACCESS_INOUT)],
+                     with_data=True)
+    tp.wait()
+    pm.pins_remove(h)
+    assert {"create", "release_deps", "schedule", "complete"} <= seen, seen
+    del A, ctx

@@ -413,6 +413,139 @@ __device__ inline double qr_vval(const double* A, int ld, int base0,
   return A[(size_t)gcol * ld + gr];
 }
 
+typedef double f64x4q __attribute__((ext_vector_type(4)));
+
+// MFMA block-reflector apply used by the panel kernel's helper pool:
+// processes 16 columns at once through v_mfma_f64_16x16x4f64, staging
+// 256-row chunks of the (unit-lower) V image and of C in LDS carved from
+// `sp` — W = V^T C and C -= V Y become matrix-core work instead of
+// latency-bound 16-lane dot units (profiles/qr_round2.md headroom note).
+// Layout notes: operands stored [k][m] in LDS ([row*16 + q] for V), MFMA
+// lane map a = A[m=r16][k=ksub], D(row=ksub+4e, col=r16) — the repo's
+// fp64 GEMM conventions (kernels_hip.cpp).
+__device__ void qr_apply_mfma16(
+    double* A, int ld, int pcol0, int base0, int len0, int base1, int rows,
+    int c0, int w, const double* T16, const int* cg, int nc, double* sp,
+    double* wy /*unused*/, int tid, int wave, int lane) {
+  double* Vc = sp;               // [256][16]
+  double* Cc = sp + 256 * 16;    // [256][16]
+  double* Wr = sp + 2 * 256 * 16;  // [16 waves][16][16] partial W
+  double* Ys = sp + 2 * 256 * 16 + 16 * 256;  // [16][16] col-major [c][q]
+  const int r16 = lane & 15, ksub = lane >> 4;
+
+  // ---------------- step A: W = V^T C ----------------
+  f64x4q accW = {0, 0, 0, 0};
+  for (int ck = 0; ck < rows; ck += 256) {
+    const int clen = min(256, rows - ck);
+    // stage V chunk (transposed [r][q], unit-lower semantics, zero pad)
+    for (int x = tid; x < 256 * 16; x += 1024) {
+      const int q = x >> 8, rr = x & 255;
+      const int r = ck + rr;
+      double v = 0;
+      if (rr < clen && q < w) {
+        const int cloc = c0 + q;
+        if (r == cloc) {
+          v = 1.0;
+        } else if (r > cloc) {
+          int gr = r < len0 ? base0 + r : base1 + (r - len0);
+          v = A[(size_t)(pcol0 + cloc) * ld + gr];
+        }
+      }
+      Vc[rr * 16 + q] = v;
+    }
+    // stage C chunk ([r][c], zero pad)
+    for (int x = tid; x < 256 * 16; x += 1024) {
+      const int c = x >> 8, rr = x & 255;
+      const int r = ck + rr;
+      double v = 0;
+      if (rr < clen && c < nc) {
+        int gr = r < len0 ? base0 + r : base1 + (r - len0);
+        v = A[(size_t)(pcol0 + cg[c]) * ld + gr];
+      }
+      Cc[rr * 16 + c] = v;
+    }
+    __syncthreads();
+    // wave `wave` covers k-rows [wave*16, wave*16+16) of this chunk
+    {
+      const int k0 = wave * 16;
+#pragma unroll
+      for (int kk = 0; kk < 16; kk += 4) {
+        double a = Vc[(k0 + kk + ksub) * 16 + r16];
+        double b = Cc[(k0 + kk + ksub) * 16 + r16];
+        accW = __builtin_amdgcn_mfma_f64_16x16x4f64(a, b, accW, 0, 0, 0);
+      }
+    }
+    __syncthreads();
+  }
+  // reduce the 16 per-wave W partials, then Y = T^T W
+#pragma unroll
+  for (int e = 0; e < 4; e++)
+    Wr[wave * 256 + (ksub + 4 * e) * 16 + r16] = accW[e];
+  __syncthreads();
+  if (tid < 256) {
+    const int q = tid & 15, c = tid >> 4;
+    double wsum = 0;
+    for (int wv = 0; wv < 16; wv++) wsum += Wr[wv * 256 + q * 16 + c];
+    // stash W(q, c) back (reuse Wr row 0 region is unsafe; use Ys then
+    // overwrite with Y below via a second barrier)
+    Wr[q * 16 + c] = wsum;  // W in [q][c] at the front of Wr
+  }
+  __syncthreads();
+  if (tid < 256) {
+    const int q = tid & 15, c = tid >> 4;
+    double y = 0;
+    for (int p2 = 0; p2 <= q; p2++) y += T16[q * 16 + p2] * Wr[p2 * 16 + c];
+    Ys[(q)*16 + c] = y;  // Y stored [k=q][c]
+  }
+  __syncthreads();
+
+  // ---------------- step B: C -= V * Y ----------------
+  for (int ck = 0; ck < rows; ck += 256) {
+    const int clen = min(256, rows - ck);
+    for (int x = tid; x < 256 * 16; x += 1024) {
+      const int q = x >> 8, rr = x & 255;
+      const int r = ck + rr;
+      double v = 0;
+      if (rr < clen && q < w) {
+        const int cloc = c0 + q;
+        if (r == cloc) {
+          v = 1.0;
+        } else if (r > cloc) {
+          int gr = r < len0 ? base0 + r : base1 + (r - len0);
+          v = A[(size_t)(pcol0 + cloc) * ld + gr];
+        }
+      }
+      Vc[rr * 16 + q] = v;
+    }
+    __syncthreads();
+    // wave owns the 16-row m-tile [wave*16, wave*16+16) of the chunk
+    f64x4q accU = {0, 0, 0, 0};
+    {
+#pragma unroll
+      for (int kk = 0; kk < 16; kk += 4) {
+        double a = Vc[(wave * 16 + r16) * 16 + kk + ksub];
+        double b = Ys[(kk + ksub) * 16 + r16];
+        accU = __builtin_amdgcn_mfma_f64_16x16x4f64(a, b, accU, 0, 0, 0);
+      }
+    }
+    // D(row=ksub+4e, col=r16) -> Cc[(wave*16+row)][col]
+#pragma unroll
+    for (int e = 0; e < 4; e++)
+      Cc[(wave * 16 + ksub + 4 * e) * 16 + r16] = accU[e];
+    __syncthreads();
+    // coalesced global RMW per column
+    for (int x = tid; x < 256 * 16; x += 1024) {
+      const int c = x >> 8, rr = x & 255;
+      const int r = ck + rr;
+      if (rr < clen && c < nc && r >= c0) {
+        int gr = r < len0 ? base0 + r : base1 + (r - len0);
+        A[(size_t)(pcol0 + cg[c]) * ld + gr] -= Cc[rr * 16 + c];
+      }
+    }
+    __syncthreads();
+  }
+}
+
 __global__ void __launch_bounds__(1024) k_qr_panel_mw(
     double* A, int ld, int pcol0,  // panel base column (global)
     int base0, int len0, int base1, int len1,  // reflector row segments
@@ -420,7 +553,7 @@ __global__ void __launch_bounds__(1024) k_qr_panel_mw(
     int fcols,  // columns to FACTOR (<=128)
     int W,      // sub-panel width (8|16)
     double* tau, double* T16s,  // T16s: 16x16 per sub-panel scratch
-    int* cnt, int nwg, int nA, unsigned long long* dbg) {
+    int* cnt, int nwg, int nA, int amode, unsigned long long* dbg) {
   // Producer/consumer panel pipeline (no grid barriers):
   //  - WG 0 factors 16-column sub-panels in LDS (one barrier per column,
   //    piggybacked norms, deferred scaling), builds T16 with a parallel
@@ -737,15 +870,18 @@ __global__ void __launch_bounds__(1024) k_qr_panel_mw(
       dbg[dslot] += t1 - t0;  // publish poll wait
       t0 = t1;
     }
-    // cache the scaled V image (unit-lower) in LDS
-    for (int q = 0; q < w; q++) {
-      const int gcol = pcol0 + c0 + q, cloc = c0 + q;
-      const double* A0 = A + (size_t)gcol * ld + base0;
-      const double* A1 = A + (size_t)gcol * ld + base1 - len0;
-      double* spq = sp + (size_t)q * rows;
-      for (int r = tid; r < len0; r += 1024)
-        spq[r] = r < cloc ? 0.0 : (r == cloc ? 1.0 : A0[r]);
-      for (int r = len0 + tid; r < rows; r += 1024) spq[r] = A1[r];
+    if (!amode) {
+      // cache the scaled V image (unit-lower) in LDS (VALU apply path;
+      // the MFMA path streams V chunks itself)
+      for (int q = 0; q < w; q++) {
+        const int gcol = pcol0 + c0 + q, cloc = c0 + q;
+        const double* A0 = A + (size_t)gcol * ld + base0;
+        const double* A1 = A + (size_t)gcol * ld + base1 - len0;
+        double* spq = sp + (size_t)q * rows;
+        for (int r = tid; r < len0; r += 1024)
+          spq[r] = r < cloc ? 0.0 : (r == cloc ? 1.0 : A0[r]);
+        for (int r = len0 + tid; r < rows; r += 1024) spq[r] = A1[r];
+      }
     }
     __syncthreads();
     if (dbg && dslot >= 0 && tid == 0) {
@@ -806,6 +942,22 @@ __global__ void __launch_bounds__(1024) k_qr_panel_mw(
       __syncthreads();
     };
     auto apply_range = [&](int cbeg, int cend, int stride, int phase) {
+      if (amode) {
+        int cg[16];
+        int nc = 0;
+        for (int c = cbeg + phase; c < cend; c += stride) {
+          cg[nc++] = c;
+          if (nc == 16) {
+            qr_apply_mfma16(A, ld, pcol0, base0, len0, base1, rows, c0, w,
+                            T16, cg, 16, sp, wy, tid, wave, lane);
+            nc = 0;
+          }
+        }
+        if (nc)
+          qr_apply_mfma16(A, ld, pcol0, base0, len0, base1, rows, c0, w,
+                          T16, cg, nc, sp, wy, tid, wave, lane);
+        return;
+      }
       int cg[4];
       int nc = 0;
       for (int c = cbeg + phase; c < cend; c += stride) {
@@ -908,14 +1060,15 @@ static void qr_factor_hand(GpuTaskCtx& g, double* A, int m, int k, int ld,
       // only; the trailing matrix goes through T128 + larfb dgemms at
       // Tensile rates (A/B: the in-kernel apply units are latency-bound,
       // profiles/qr_round2.md).
-      static const bool apply_gemm =
-          param_str("qr_apply", "kernel") == "gemm";
+      static const std::string apply_kind = param_str("qr_apply", "kernel");
+      static const bool apply_gemm = apply_kind == "gemm";
+      static const int amode = apply_kind == "valu" ? 0 : 1;
       int rest = k - p - pc;
       int apply_cols = apply_gemm ? pc : k - p;
       PA_HIP_CHECK(hipMemsetAsync(cnt, 0, 2 * sizeof(int), g.stream));
       hipLaunchKernelGGL(k_qr_panel_mw, dim3(nwg), dim3(1024), 0, g.stream,
                          A, ld, p, base0, len0, base1, len1, apply_cols, pc,
-                         W, tau + p, T16s, cnt, nwg, nA,
+                         W, tau + p, T16s, cnt, nwg, nA, amode,
                          (unsigned long long*)nullptr);
       if (apply_gemm && rest > 0) {
         double* V = qr_scratch(g, slot0 + 1, (size_t)prows * 128 * 8);
@@ -1296,7 +1449,9 @@ double bench_qr_factor(int m, int k, int ts_split, int iters, int mode) {
         PA_HIP_CHECK(hipMemsetAsync(cnt, 0, 2 * sizeof(int), s));
         hipLaunchKernelGGL(k_qr_panel_mw, dim3(nwg), dim3(1024), 0, s, dA,
                            m, p, base0, len0, base1, len1, k - p, pc, W,
-                           tau + p, T16s, cnt, nwg, nA, dbg);
+                           tau + p, T16s, cnt, nwg, nA,
+                           param_str("qr_apply", "kernel") == "valu" ? 0 : 1,
+                           dbg);
       }
     } else {
       PA_CHECK(rocsolver_dgeqrf(qr_handle(g), m, k, dA, m, tau) ==

@@ -870,7 +870,7 @@ __global__ void __launch_bounds__(1024) k_qr_panel_mw(
       dbg[dslot] += t1 - t0;  // publish poll wait
       t0 = t1;
     }
-    if (!amode) {
+    if (!amode || poolA) {
       // cache the scaled V image (unit-lower) in LDS (VALU apply path;
       // the MFMA path streams V chunks itself)
       for (int q = 0; q < w; q++) {
@@ -941,8 +941,9 @@ __global__ void __launch_bounds__(1024) k_qr_panel_mw(
       }
       __syncthreads();
     };
-    auto apply_range = [&](int cbeg, int cend, int stride, int phase) {
-      if (amode) {
+    auto apply_range = [&](int cbeg, int cend, int stride, int phase,
+                           bool mfma) {
+      if (amode && mfma) {
         int cg[16];
         int nc = 0;
         for (int c = cbeg + phase; c < cend; c += stride) {
@@ -970,7 +971,9 @@ __global__ void __launch_bounds__(1024) k_qr_panel_mw(
       if (nc) apply4(cg, nc);
     };
     if (poolA) {
-      apply_range(min(c0 + 2 * W, fcols), fcols, nA, wg - 1);
+      // small within-panel groups: the latency-optimized VALU units win;
+      // the MFMA path pays per-256-row chunk staging regardless of nc
+      apply_range(min(c0 + 2 * W, fcols), fcols, nA, wg - 1, false);
       __syncthreads();
       if (dbg && dslot >= 0 && tid == 0)
         dbg[dslot + 2] += __builtin_amdgcn_s_memrealtime() - t0;  // apply
@@ -981,9 +984,9 @@ __global__ void __launch_bounds__(1024) k_qr_panel_mw(
                                __HIP_MEMORY_SCOPE_AGENT);
       }
       if (nB == 0)  // no pool B: pool A also covers the trailing columns
-        apply_range(fcols, pc, nA, wg - 1);
+        apply_range(fcols, pc, nA, wg - 1, true);
     } else {
-      apply_range(fcols, pc, nB, wg - 1 - nA);
+      apply_range(fcols, pc, nB, wg - 1 - nA, true);
       if (dbg && dslot >= 0 && tid == 0)
         dbg[dslot + 2] += __builtin_amdgcn_s_memrealtime() - t0;  // apply
     }

@@ -587,19 +587,6 @@ __global__ void __launch_bounds__(1024) k_qr_panel_mw(
       const int w = min(W, fcols - c0);
       double* T16 = T16s + (size_t)si * 256;
       unsigned long long t0 = dbg ? __builtin_amdgcn_s_memrealtime() : 0;
-      if (si > 1) {
-        // one-sub-panel slack: WG0 self-applies each block to its own
-        // next-16 columns, so only ack(si-2) gates this stage and pool
-        // A's work hides under the factor of si-1.
-        if (tid == 0) {
-          while (__hip_atomic_load(&cnt[1], __ATOMIC_RELAXED,
-                                   __HIP_MEMORY_SCOPE_AGENT) <
-                 (si - 1) * nA)
-            __builtin_amdgcn_s_sleep(2);
-          __builtin_amdgcn_fence(__ATOMIC_ACQUIRE, "agent");
-        }
-        __syncthreads();
-      }
       if (dbg && tid == 0) {
         unsigned long long t1 = __builtin_amdgcn_s_memrealtime();
         dbg[0] += t1 - t0;
@@ -781,6 +768,18 @@ __global__ void __launch_bounds__(1024) k_qr_panel_mw(
         __hip_atomic_store(&cnt[0], si + 1, __ATOMIC_RELAXED,
                            __HIP_MEMORY_SCOPE_AGENT);
       }
+      // ---- wait for pool A to finish sub-panel si-1 everywhere: its
+      // range [c0-W+2W, fcols) includes OUR self-apply window, and the
+      // si-1 reflectors must land on those columns before si's do. The
+      // factor of si overlapped pool A's si-1 pass, so this wait is the
+      // pipeline's only rendezvous. ----
+      if (tid == 0) {
+        while (__hip_atomic_load(&cnt[1], __ATOMIC_RELAXED,
+                                 __HIP_MEMORY_SCOPE_AGENT) < si * nA)
+          __builtin_amdgcn_s_sleep(2);
+        __builtin_amdgcn_fence(__ATOMIC_ACQUIRE, "agent");
+      }
+      __syncthreads();
       // ---- self-apply this block to the NEXT sub-panel's columns (the
       // V image is already in sp, minus the unit-lower masking) ----
       for (int c = c0 + w; c < min(c0 + 2 * W, fcols); c += 4) {

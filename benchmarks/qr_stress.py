@@ -1,0 +1,38 @@
+#!/usr/bin/env python3
+"""QR correctness stress: repeat small hand-panel QR, report failures."""
+import os
+import sys
+
+sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
+import numpy as np
+import parsec_amd as pm
+
+iters = int(sys.argv[1]) if len(sys.argv) > 1 else 20
+ctx = pm.Context(nworkers=2, rank=0, world=1)
+n, nb = 1024, 256
+fails = 0
+for it in range(iters):
+    A = pm.TiledMatrix(ctx, n, n, nb, nb, 1, 1)
+    tp0 = pm.Dtd(ctx)
+    pm.insert_full_fill(tp0, A, 42 + it)
+    tp0.wait()
+    A0 = np.zeros((n, n))
+    for tm in range(A.mt):
+        for tn in range(A.nt):
+            A0[tm*nb:(tm+1)*nb, tn*nb:(tn+1)*nb] = A.tile_numpy(tm, tn)
+    tp = pm.Dtd(ctx)
+    pm.insert_geqrf(tp, A)
+    tp.wait()
+    R = np.zeros((n, n))
+    for tm in range(A.mt):
+        for tn in range(A.nt):
+            R[tm*nb:(tm+1)*nb, tn*nb:(tn+1)*nb] = A.tile_numpy(tm, tn)
+    R = np.triu(R)
+    err = np.abs(R.T @ R - A0.T @ A0).max() / np.abs(A0.T @ A0).max()
+    ok = err < 1e-12
+    if not ok:
+        fails += 1
+        print(f"iter {it}: FAIL err={err:.3e}", flush=True)
+    del A
+print(f"fails={fails}/{iters}", flush=True)
+del ctx

@@ -13,9 +13,15 @@ n, nb = 1024, 256
 fails = 0
 for it in range(iters):
     A = pm.TiledMatrix(ctx, n, n, nb, nb, 1, 1)
-    tp0 = pm.Dtd(ctx)
-    pm.insert_full_fill(tp0, A, 42 + it)
-    tp0.wait()
+    if os.environ.get("QR_STRESS_FILL", "host") == "host":
+        rng = np.random.default_rng(42 + it)
+        for tm in range(A.mt):
+            for tn in range(A.nt):
+                A.tile_numpy_set(tm, tn, rng.standard_normal((nb, nb)))
+    else:
+        tp0 = pm.Dtd(ctx)
+        pm.insert_full_fill(tp0, A, 42 + it)
+        tp0.wait()
     A0 = np.zeros((n, n))
     for tm in range(A.mt):
         for tn in range(A.nt):

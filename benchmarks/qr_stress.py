@@ -39,6 +39,22 @@ for it in range(iters):
     if not ok:
         fails += 1
         print(f"iter {it}: FAIL err={err:.3e}", flush=True)
+        # forensics: sign-normalized comparison against numpy R, per tile
+        Rnp = np.linalg.qr(A0, mode="r")
+        sgn = np.sign(np.diag(Rnp)) * np.sign(np.diag(R) + (np.diag(R) == 0))
+        Rn = Rnp * sgn[:, None]
+        for tm in range(A.mt):
+            row = []
+            for tn in range(A.nt):
+                d = np.abs(R[tm*nb:(tm+1)*nb, tn*nb:(tn+1)*nb] -
+                           Rn[tm*nb:(tm+1)*nb, tn*nb:(tn+1)*nb]).max()
+                row.append(f"{d:9.2e}")
+            print("   tile row", tm, " ".join(row), flush=True)
+        # column-resolved: first column where R diverges
+        cd = np.abs(R - Rn).max(axis=0)
+        bad = np.nonzero(cd > 1e-8)[0]
+        print(f"   first bad col {bad[0] if len(bad) else -1}, nbad={len(bad)}",
+              flush=True)
     del A
 print(f"fails={fails}/{iters}", flush=True)
 del ctx

@@ -553,7 +553,8 @@ __global__ void __launch_bounds__(1024) k_qr_panel_mw(
     int fcols,  // columns to FACTOR (<=128)
     int W,      // sub-panel width (8|16)
     double* tau, double* T16s,  // T16s: 16x16 per sub-panel scratch
-    int* cnt, int nwg, int nA, int amode, unsigned long long* dbg) {
+    int* cnt, int nwg, int nA, int amode, int look,
+    unsigned long long* dbg) {
   // Producer/consumer panel pipeline (no grid barriers):
   //  - WG 0 factors 16-column sub-panels in LDS (one barrier per column,
   //    piggybacked norms, deferred scaling), builds T16 with a parallel
@@ -772,17 +773,19 @@ __global__ void __launch_bounds__(1024) k_qr_panel_mw(
       // range [c0-W+2W, fcols) includes OUR self-apply window, and the
       // si-1 reflectors must land on those columns before si's do. The
       // factor of si overlapped pool A's si-1 pass, so this wait is the
-      // pipeline's only rendezvous. ----
+      // pipeline's only rendezvous. With qr_lookahead=0 the self-apply is
+      // off and the wait covers pool A through si instead. ----
       if (tid == 0) {
         while (__hip_atomic_load(&cnt[1], __ATOMIC_RELAXED,
-                                 __HIP_MEMORY_SCOPE_AGENT) < si * nA)
+                                 __HIP_MEMORY_SCOPE_AGENT) <
+               (si + (look ? 0 : 1)) * nA)
           __builtin_amdgcn_s_sleep(2);
         __builtin_amdgcn_fence(__ATOMIC_ACQUIRE, "agent");
       }
       __syncthreads();
       // ---- self-apply this block to the NEXT sub-panel's columns (the
       // V image is already in sp, minus the unit-lower masking) ----
-      for (int c = c0 + w; c < min(c0 + 2 * W, fcols); c += 4) {
+      for (int c = c0 + w; look && c < min(c0 + 2 * W, fcols); c += 4) {
         const int nc = min(4, min(c0 + 2 * W, fcols) - c);
         const int ci = lane >> 4, l16 = lane & 15;
         if (ci < nc && wave < w) {
@@ -972,7 +975,8 @@ __global__ void __launch_bounds__(1024) k_qr_panel_mw(
     if (poolA) {
       // small within-panel groups: the latency-optimized VALU units win;
       // the MFMA path pays per-256-row chunk staging regardless of nc
-      apply_range(min(c0 + 2 * W, fcols), fcols, nA, wg - 1, false);
+      apply_range(look ? min(c0 + 2 * W, fcols) : c0 + w, fcols, nA,
+                  wg - 1, false);
       __syncthreads();
       if (dbg && dslot >= 0 && tid == 0)
         dbg[dslot + 2] += __builtin_amdgcn_s_memrealtime() - t0;  // apply
@@ -1071,6 +1075,7 @@ static void qr_factor_hand(GpuTaskCtx& g, double* A, int m, int k, int ld,
       hipLaunchKernelGGL(k_qr_panel_mw, dim3(nwg), dim3(1024), 0, g.stream,
                          A, ld, p, base0, len0, base1, len1, apply_cols, pc,
                          W, tau + p, T16s, cnt, nwg, nA, amode,
+                         (int)param_int("qr_lookahead", 1),
                          (unsigned long long*)nullptr);
       if (apply_gemm && rest > 0) {
         double* V = qr_scratch(g, slot0 + 1, (size_t)prows * 128 * 8);
@@ -1453,7 +1458,7 @@ double bench_qr_factor(int m, int k, int ts_split, int iters, int mode) {
                            m, p, base0, len0, base1, len1, k - p, pc, W,
                            tau + p, T16s, cnt, nwg, nA,
                            param_str("qr_apply", "kernel") == "valu" ? 0 : 1,
-                           dbg);
+                           (int)param_int("qr_lookahead", 1), dbg);
       }
     } else {
       PA_CHECK(rocsolver_dgeqrf(qr_handle(g), m, k, dA, m, tau) ==

@@ -20,14 +20,6 @@ for it in range(iters):
         v = rng.standard_normal((nb, nb))
         srcs.append(v)
         A.tile_numpy_set(tm, 0, v)
-    # NT independent single-tile QRs -> NT concurrent hand panel kernels
-    tp = pm.Dtd(ctx)
-    for tm in range(NT):
-        B = pm.TiledMatrix(ctx, nb, nb, nb, nb, 1, 1)
-        tp.own_py = None  # noqa
-        # factor each tile in place via insert_geqrf on a 1x1 view
-        # (simplest: separate 1-tile matrices)
-    del tp
     mats = []
     tp = pm.Dtd(ctx)
     for tm in range(NT):

@@ -943,12 +943,18 @@ __global__ void __launch_bounds__(1024) k_qr_panel_mw(
       }
       __syncthreads();
     };
+    // Column ownership must be ABSOLUTE (c mod stride == phase), not
+    // relative to cbeg: cbeg advances with the sub-panel, and a column
+    // whose owner changed between sub-panels would receive reflector
+    // applies from two helpers with NO mutual ordering (measured: rare
+    // mid-panel corruption, first bad column inside pool A's range).
     auto apply_range = [&](int cbeg, int cend, int stride, int phase,
                            bool mfma) {
+      int first = cbeg + (((phase - cbeg) % stride) + stride) % stride;
       if (amode && mfma) {
         int cg[16];
         int nc = 0;
-        for (int c = cbeg + phase; c < cend; c += stride) {
+        for (int c = first; c < cend; c += stride) {
           cg[nc++] = c;
           if (nc == 16) {
             qr_apply_mfma16(A, ld, pcol0, base0, len0, base1, rows, c0, w,
@@ -963,7 +969,7 @@ __global__ void __launch_bounds__(1024) k_qr_panel_mw(
       }
       int cg[4];
       int nc = 0;
-      for (int c = cbeg + phase; c < cend; c += stride) {
+      for (int c = first; c < cend; c += stride) {
         cg[nc++] = c;
         if (nc == 4) {
           apply4(cg, 4);

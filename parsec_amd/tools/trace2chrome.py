@@ -24,19 +24,29 @@ def convert(path, out_path=None):
     classes = {int(k): v for k, v in header["classes"].items()}
     kinds = {int(k): v for k, v in header["kinds"].items()}
     events = []
+    gpu_lanes = set()
     for off in range(0, len(raw) - rec.size + 1, rec.size):
         t0, t1, tid, kind, cid, seq = rec.unpack_from(raw, off)
         name = classes.get(cid, f"class{cid}")
+        if tid >= 1000:
+            gpu_lanes.add(tid)
         events.append({
             "name": f"{name}",
             "cat": kinds.get(kind, str(kind)),
             "ph": "X",
             "ts": t0 / 1e3,
             "dur": max(t1 - t0, 1) / 1e3,
-            "pid": 0,
-            "tid": tid,
+            "pid": 1 if tid >= 1000 else 0,
+            "tid": tid - 1000 if tid >= 1000 else tid,
             "args": {"seq": seq},
         })
+    for lane in sorted(gpu_lanes):
+        events.append({"name": "thread_name", "ph": "M", "pid": 1,
+                       "tid": lane - 1000,
+                       "args": {"name": f"gpu exec stream {lane - 1000}"}})
+    if gpu_lanes:
+        events.append({"name": "process_name", "ph": "M", "pid": 1,
+                       "args": {"name": "GPU device spans"}})
     out = out_path or path + ".json"
     with open(out, "w") as f:
         json.dump({"traceEvents": events}, f)

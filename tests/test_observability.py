@@ -156,8 +156,18 @@ mod = compile_jdf(_os.path.join({REPO!r}, "examples", "cholesky.jdf"))
 tp2 = pm.Dtd(ctx)
 mod.build(ctx, tp2, descA=A, NT=A.mt, NB=A.nb)
 tp2.wait()
+# round-2 features: reshape promise + wait_dynamic (world-1 degenerates
+# to wait)
+import numpy as np
+R = pm.TiledMatrix(ctx, 128, 128, 128, 128, 1, 1)
+tp3 = pm.Dtd(ctx)
+def tr(x, out):
+    np.frombuffer(out, dtype=np.float64)[:] = np.frombuffer(x, dtype=np.float64)
+tp3.insert_py(tr, [(A.tile(0, 0), pm.ACCESS_IN, pm.RESHAPE_TRANSPOSE),
+                   (R.tile(0, 0), pm.ACCESS_OUT)], with_data=True)
+tp3.wait_dynamic()
 print("SNIPPETS_OK")
-del A, ctx
+del A, R, ctx
 """
     r = subprocess.run([sys.executable, "-c", code], capture_output=True,
                        text=True, timeout=300)

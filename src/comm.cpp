@@ -13,6 +13,7 @@
 #include <netinet/tcp.h>
 #include <poll.h>
 #include <sys/socket.h>
+#include <sys/uio.h>
 #include <unistd.h>
 
 #include <cstring>
@@ -117,9 +118,17 @@ class TcpComm : public CommEngine {
  private:
   struct Peer {
     int fd = -1;
-    // outgoing
-    std::deque<std::pair<std::vector<uint8_t>, Task*>> out;  // framed bytes
-    size_t out_off = 0;
+    // outgoing frames: header + borrowed payload pointer (zero-copy: the
+    // payload is the tile's host buffer, pinned by the send task until
+    // completion — round-1 copied every payload twice)
+    struct OutFrame {
+      FrameHeader hdr;
+      const uint8_t* payload = nullptr;  // borrowed; may be null
+      std::vector<uint8_t> owned;        // control frames own their bytes
+      Task* done = nullptr;
+    };
+    std::deque<OutFrame> out;
+    size_t out_off = 0;  // bytes of (header+payload) already written
     // incoming state machine
     FrameHeader hdr;
     size_t hdr_got = 0;
@@ -188,13 +197,21 @@ class TcpComm : public CommEngine {
   void wake() { (void)!write(wake_pipe_[1], "x", 1); }
 
   void queue_frame(int peer, uint32_t kind, uint64_t seq, const void* payload,
-                   uint64_t size, Task* done_task) {
+                   uint64_t size, Task* done_task, bool borrow = false) {
     Peer& p = peers_[peer];
-    std::vector<uint8_t> buf(sizeof(FrameHeader) + size);
-    FrameHeader h{kind, 0, seq, size};
-    memcpy(buf.data(), &h, sizeof(h));
-    if (size) memcpy(buf.data() + sizeof(h), payload, size);
-    p.out.emplace_back(std::move(buf), done_task);
+    Peer::OutFrame f;
+    f.hdr = FrameHeader{kind, 0, seq, size};
+    f.done = done_task;
+    if (size) {
+      if (borrow) {
+        f.payload = (const uint8_t*)payload;
+      } else {
+        f.owned.assign((const uint8_t*)payload,
+                       (const uint8_t*)payload + size);
+        f.payload = f.owned.data();
+      }
+    }
+    p.out.push_back(std::move(f));
   }
 
   void process_cmd(Task* t) {
@@ -210,7 +227,8 @@ class TcpComm : public CommEngine {
       }
       sends_out_++;
       void* ptr = d->pull_to_host();
-      queue_frame(t->peer, FK_DATA, t->comm_seq, ptr, d->bytes, t);
+      queue_frame(t->peer, FK_DATA, t->comm_seq, ptr, d->bytes, t,
+                  /*borrow=*/true);
     } else {  // COMM_RECV
       uint64_t key = ((uint64_t)t->peer << 48) | t->comm_seq;
       auto it = unexpected_.find(key);
@@ -359,18 +377,36 @@ class TcpComm : public CommEngine {
   void do_write(int peer) {
     Peer& p = peers_[peer];
     while (!p.out.empty()) {
-      auto& [buf, done] = p.out.front();
-      ssize_t w = write(p.fd, buf.data() + p.out_off, buf.size() - p.out_off);
-      if (w < 0) {
-        if (errno != EAGAIN && errno != EWOULDBLOCK) peer_down(peer, w);
-        return;
+      auto& f = p.out.front();
+      const size_t hsz = sizeof(FrameHeader);
+      const size_t total = hsz + f.hdr.size;
+      // gather header + borrowed payload in one writev
+      while (p.out_off < total) {
+        iovec iov[2];
+        int n = 0;
+        if (p.out_off < hsz) {
+          iov[n].iov_base = (uint8_t*)&f.hdr + p.out_off;
+          iov[n].iov_len = hsz - p.out_off;
+          n++;
+        }
+        if (f.hdr.size) {
+          size_t poff = p.out_off > hsz ? p.out_off - hsz : 0;
+          iov[n].iov_base = (void*)(f.payload + poff);
+          iov[n].iov_len = f.hdr.size - poff;
+          n++;
+        }
+        ssize_t w = writev(p.fd, iov, n);
+        if (w < 0) {
+          if (errno != EAGAIN && errno != EWOULDBLOCK) peer_down(peer, w);
+          return;
+        }
+        p.out_off += (size_t)w;
       }
-      p.out_off += (size_t)w;
-      if (p.out_off < buf.size()) return;
+      Task* done = f.done;
       if (done) {
         counters().comm_msgs.fetch_add(1, std::memory_order_relaxed);
-        counters().comm_bytes.fetch_add(buf.size(), std::memory_order_relaxed);
-        note_sent(done->peer, buf.size());
+        counters().comm_bytes.fetch_add(total, std::memory_order_relaxed);
+        note_sent(done->peer, total);
         sends_out_--;
         while (!pending_sends_.empty() && sends_out_ < max_sends_) {
           Task* pt = pending_sends_.front();
@@ -378,7 +414,7 @@ class TcpComm : public CommEngine {
           sends_out_++;
           Data* pd = pt->flows[0].data;
           queue_frame(pt->peer, FK_DATA, pt->comm_seq, pd->pull_to_host(),
-                      pd->bytes, pt);
+                      pd->bytes, pt, /*borrow=*/true);
         }
         Profiler& pr = Profiler::inst();
         if (pr.enabled())

@@ -271,6 +271,17 @@ class TcpComm : public CommEngine {
         uint64_t key = ((uint64_t)peer << 48) | p.hdr.seq;
         unexpected_[key] = std::move(p.in_payload);
         p.in_payload.clear();
+        // bounded in practice by the SENDER's flow-control cap
+        // (comm_max_inflight per peer); a blow-up means the protocol's
+        // deterministic matching broke — say so before memory does
+        if (unexpected_.size() > 4 * max_sends_ * (size_t)world_ &&
+            !warned_unexpected_) {
+          warned_unexpected_ = true;
+          fprintf(stderr,
+                  "[parsec_amd] comm: %zu unexpected frames stashed — "
+                  "probable SPMD insertion divergence\n",
+                  unexpected_.size());
+        }
       }
     } else if (p.hdr.kind == FK_CTL) {
       std::string pl((const char*)p.in_payload.data(), p.in_payload.size());
@@ -519,6 +530,7 @@ class TcpComm : public CommEngine {
   std::unordered_map<uint64_t, std::vector<uint8_t>> unexpected_;
   std::deque<Task*> pending_sends_;
   size_t sends_out_ = 0, max_sends_ = 64;
+  bool warned_unexpected_ = false;
 
   std::mutex bar_mtx_;
   std::condition_variable bar_cv_;

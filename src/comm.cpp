@@ -82,16 +82,20 @@ class TcpComm : public CommEngine {
   void set_ctl_handler(CtlHandler h) override {
     std::lock_guard<std::mutex> g(ctl_mtx_);
     ctl_handler_ = std::move(h);
-    // deliver anything that arrived before the handler was installed
-    for (auto& [src, tag, pl] : ctl_stash_) ctl_handler_(src, tag, pl);
-    ctl_stash_.clear();
+    if (ctl_handler_) {
+      // deliver anything that arrived before the handler was installed
+      for (auto& [src, tag, pl] : ctl_stash_) ctl_handler_(src, tag, pl);
+      ctl_stash_.clear();
+    }
   }
 
   void set_sys_handler(CtlHandler h) override {
     std::lock_guard<std::mutex> g(ctl_mtx_);
     sys_handler_ = std::move(h);
-    for (auto& [src, tag, pl] : sys_stash_) sys_handler_(src, tag, pl);
-    sys_stash_.clear();
+    if (sys_handler_) {
+      for (auto& [src, tag, pl] : sys_stash_) sys_handler_(src, tag, pl);
+      sys_stash_.clear();
+    }
   }
 
   void enqueue(Task* t) override {

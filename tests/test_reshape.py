@@ -179,3 +179,39 @@ def test_reshape_gpu_numerics():
     assert c.counters()["tasks_executed_gpu"] > 0
     assert np.allclose(R.tile_numpy(0, 0), src.T)
     del A, R, c
+
+
+def test_reshape_after_rename(ctx):
+    """Reshape promises attach to the CURRENT copy: after an OUTPUT-only
+    rewrite forces a rename, the next promise converts the new version."""
+    nb = 8
+    A = pm.TiledMatrix(ctx, nb, nb, nb, nb, 1, 1)
+    _fill(A, 0, 0, np.ones((nb, nb)))
+    t = A.tile(0, 0)
+    got = []
+    tp = pm.Dtd(ctx)
+    for k in range(4):
+        # slow reader keeps the old copy live -> the OUT rewrite renames
+        def reader(x, k=k):
+            import time
+            time.sleep(0.01)
+            got.append(np.frombuffer(x, dtype=np.float64)[0])
+        tp.insert_py(reader, [(t, pm.ACCESS_IN, pm.RESHAPE_TRIL)],
+                     with_data=True)
+        def writer(buf, k=k):
+            np.frombuffer(buf, dtype=np.float64)[:] = float(k + 2)
+        tp.insert_py(writer, [(t, pm.ACCESS_OUT)], with_data=True)
+    tp.wait()
+    assert got == [1.0, 2.0, 3.0, 4.0], got
+    assert ctx.counters()["renames"] > 0
+
+
+def test_wait_dynamic_repeated(ctx):
+    """wait_dynamic is reusable (handler teardown between calls)."""
+    tp = pm.Dtd(ctx)
+    acc = []
+    for r in range(3):
+        for i in range(10):
+            tp.insert_py(lambda i=i: acc.append(i))
+        tp.wait_dynamic()
+    assert len(acc) == 30

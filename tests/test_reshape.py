@@ -202,7 +202,9 @@ def test_reshape_after_rename(ctx):
             np.frombuffer(buf, dtype=np.float64)[:] = float(k + 2)
         tp.insert_py(writer, [(t, pm.ACCESS_OUT)], with_data=True)
     tp.wait()
-    assert got == [1.0, 2.0, 3.0, 4.0], got
+    # renaming makes readers of successive versions CONCURRENT, so the
+    # completion order is free — each must still see its own version
+    assert sorted(got) == [1.0, 2.0, 3.0, 4.0], got
     assert ctx.counters()["renames"] > 0
 
 

@@ -103,3 +103,31 @@ def test_c_embedding_distributed(tmp_path):
     for pr in procs:
         o, _ = pr.communicate(timeout=120)
         assert pr.returncode == 0 and b"C_DIST_OK" in o, o.decode()
+
+
+def test_fortran_embedding(tmp_path):
+    """Fortran bindings (parsecf.F90 analog, parsec_amd/fortran/parsecf.f90):
+    compile examples/f_embed.f90 with ROCm's amdflang against
+    libparsec_amd.so and run the dataflow chain end-to-end."""
+    import shutil
+    flang = shutil.which("amdflang", path="/opt/rocm/lib/llvm/bin")
+    if flang is None:
+        import pytest
+        pytest.skip("amdflang not available")
+    mod = str(tmp_path)
+    r1 = subprocess.run(
+        [flang, "-c", os.path.join(REPO, "parsec_amd/fortran/parsecf.f90"),
+         "-o", os.path.join(mod, "parsecf.o"), "-module-dir", mod],
+        capture_output=True, text=True, timeout=300)
+    assert r1.returncode == 0, r1.stderr
+    exe = os.path.join(mod, "f_embed")
+    r2 = subprocess.run(
+        [flang, os.path.join(REPO, "examples", "f_embed.f90"),
+         os.path.join(mod, "parsecf.o"), "-module-dir", mod,
+         "-L" + os.path.join(REPO, "parsec_amd"), "-l:libparsec_amd.so",
+         "-Wl,-rpath," + os.path.join(REPO, "parsec_amd"), "-o", exe],
+        capture_output=True, text=True, timeout=300)
+    assert r2.returncode == 0, r2.stderr
+    r3 = subprocess.run([exe], capture_output=True, text=True, timeout=120)
+    assert r3.returncode == 0 and "F_EMBED_OK" in r3.stdout, (
+        r3.stdout + r3.stderr)

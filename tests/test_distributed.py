@@ -165,17 +165,22 @@ del A, ctx
 """
 
 
-@pytest.mark.parametrize("world,seed", [(2, 101), (2, 202), (4, 303), (8, 404)])
-def test_distributed_fuzz_vs_oracle(world, seed, tmp_path):
+@pytest.mark.parametrize("world,seed,tree", [
+    (2, 101, "unicast"), (2, 202, "unicast"), (4, 303, "unicast"),
+    (8, 404, "unicast"), (4, 505, "binomial"), (8, 606, "binomial")])
+def test_distributed_fuzz_vs_oracle(world, seed, tree, tmp_path):
     """Random DAGs (random tiles/modes/executing ranks) through the full
-    SPMD protocol must match a sequential numpy oracle."""
+    SPMD protocol must match a sequential numpy oracle — with both the
+    unicast and the binomial-tree fan-out (renaming + flow control live
+    underneath in all cases)."""
     port = _next_port[0]
     _next_port[0] += world + 2
     procs = []
     for r in range(world):
         env = dict(os.environ)
         env.update(RANK=str(r), WORLD_SIZE=str(world), PORT=str(port),
-                   SEED=str(seed), PARSEC_REPO=os.path.dirname(HERE))
+                   SEED=str(seed), PARSEC_REPO=os.path.dirname(HERE),
+                   PARSEC_MCA_bcast_tree=tree)
         procs.append(subprocess.Popen([sys.executable, "-c", FUZZ_CODE],
                                       env=env, stdout=subprocess.PIPE,
                                       stderr=subprocess.STDOUT))

@@ -226,27 +226,32 @@ void Taskpool::wait_dynamic() {
     std::atomic<bool> done{false};
   };
   auto st = std::make_shared<TdState>();
+  // epoch disambiguates replies across successive wait_dynamic calls: a
+  // straggler reply stashed between calls must never count toward a
+  // later call's wave of the same number
+  static std::atomic<uint64_t> td_epoch{0};
+  const uint64_t epoch = td_epoch.fetch_add(1) + 1;
   auto triple = [this, ce](uint64_t* v) {
     v[0] = (uint64_t)pending();
     v[1] = ce->ctl_sent();
     v[2] = ce->ctl_recvd();
   };
-  ce->set_sys_handler([this, st, ce, rank, triple](int src, uint32_t tag,
-                                                   const std::string& pl) {
+  ce->set_sys_handler([this, st, ce, rank, triple, epoch](
+                          int src, uint32_t tag, const std::string& pl) {
     if (tag == TD_PROBE) {
-      uint64_t rep[4];
-      memcpy(rep, pl.data(), 8);  // wave
-      triple(rep + 1);
+      uint64_t rep[5];
+      memcpy(rep, pl.data(), 16);  // {epoch, wave}
+      triple(rep + 2);
       ce->send_ctl(src, SYS | 2, rep, sizeof(rep));
     } else if (tag == TD_REPLY) {
-      uint64_t rep[4];
+      uint64_t rep[5];
       memcpy(rep, pl.data(), sizeof(rep));
       std::lock_guard<std::mutex> g(st->m);
-      if (rep[0] == st->wave) {
+      if (rep[0] == epoch && rep[1] == st->wave) {
         st->got++;
-        st->sum_p += rep[1];
-        st->sum_s += rep[2];
-        st->sum_r += rep[3];
+        st->sum_p += rep[2];
+        st->sum_s += rep[3];
+        st->sum_r += rep[4];
         st->cv.notify_all();
       }
     } else if (tag == TD_DONE) {
@@ -267,8 +272,9 @@ void Taskpool::wait_dynamic() {
         st->got = 0;
         st->sum_p = st->sum_s = st->sum_r = 0;
       }
+      uint64_t probe[2] = {epoch, wave};
       for (int d = 1; d < world; d++)
-        ce->send_ctl(d, TD_PROBE, &wave, sizeof(wave));
+        ce->send_ctl(d, TD_PROBE, probe, sizeof(probe));
       std::unique_lock<std::mutex> g(st->m);
       st->cv.wait_for(g, std::chrono::seconds(60),
                       [&] { return st->got == world - 1; });

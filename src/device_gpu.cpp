@@ -81,6 +81,21 @@ GpuEngine::~GpuEngine() {
   if (slab_) hipFree(slab_);
 }
 
+void GpuEngine::publish_info(Context* ctx) {
+  hipDeviceProp_t prop{};
+  if (hipGetDeviceProperties(&prop, device_) == hipSuccess) {
+    ctx->info_set("device.name", prop.gcnArchName);
+    ctx->info_set("device.cus", std::to_string(prop.multiProcessorCount));
+    ctx->info_set("device.hbm_gb",
+                  std::to_string(prop.totalGlobalMem >> 30));
+  }
+  ctx->info_set("device.id", std::to_string(device_));
+  ctx->info_set("device.slab_gb",
+                std::to_string((double)slab_bytes_ / (1 << 30)));
+  ctx->info_set("device.exec_streams",
+                std::to_string(exec_streams_.size()));
+}
+
 void GpuEngine::enqueue(Task* t) {
   {
     std::lock_guard<std::mutex> g(q_mtx_);

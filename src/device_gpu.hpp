@@ -74,6 +74,8 @@ class GpuEngine {
   void copy_h2d(void* dst, const void* src, size_t bytes);
 
   int device() const { return device_; }
+  // publish device facts into the context info registry (info.c analog)
+  void publish_info(Context* ctx);
   hipStream_t comm_stream() const { return comm_stream_; }
   // Quiesce all engine streams (used before external readback).
   void sync_all();

@@ -180,6 +180,14 @@ PYBIND11_MODULE(_core, m) {
               }
             });
       })
+      .def("info", [](Context& c) {
+        py::dict d;
+        for (auto& [k, v] : c.info_all()) d[py::str(k)] = v;
+        return d;
+      })
+      .def("info_set", [](Context& c, std::string k, std::string v) {
+        c.info_set(k, v);
+      })
       .def("comm_stats", [](Context& c) {
         // per-peer traffic table (device-stats analog for the comm engine)
         py::list out;

@@ -453,6 +453,13 @@ Context::Context(const Options& opt) : rank_(opt.rank), world_(opt.world) {
 
   if (opt.gpu_device != -2) gpu_ = GpuEngine::create(this, opt.gpu_device);
   comm_ = CommEngine::create(this, opt.comm);
+  // components publish their facts into the info registry (info.c analog)
+  info_set("runtime.rank", std::to_string(rank_));
+  info_set("runtime.world", std::to_string(world_));
+  info_set("runtime.workers", std::to_string(nworkers_));
+  info_set("sched.kind", param_str("sched", "ws"));
+  if (comm_) info_set("comm.kind", comm_->kind());
+  if (gpu_) gpu_->publish_info(this);
 
   roctx_init();
   debug_history_init();

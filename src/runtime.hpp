@@ -19,6 +19,7 @@
 #include <cstdint>
 #include <deque>
 #include <functional>
+#include <map>
 #include <memory>
 #include <mutex>
 #include <queue>
@@ -244,6 +245,25 @@ class Context {
   // Route a ready task to its executor (CPU sched / GPU engine / comm).
   void dispatch(Task* t, int worker_hint = -1);
 
+  // Per-context info registry (class/info.c analog: extensible key/value
+  // slots on runtime objects; components publish facts about themselves
+  // at attach time — device topology, engine kinds — consumed by tools
+  // and printed in the stats table). Scoped keys: "device.name",
+  // "comm.kind", ...
+  void info_set(const std::string& key, const std::string& value) {
+    std::lock_guard<std::mutex> g(info_mtx_);
+    info_[key] = value;
+  }
+  std::string info_get(const std::string& key) const {
+    std::lock_guard<std::mutex> g(info_mtx_);
+    auto it = info_.find(key);
+    return it == info_.end() ? std::string() : it->second;
+  }
+  std::map<std::string, std::string> info_all() const {
+    std::lock_guard<std::mutex> g(info_mtx_);
+    return info_;
+  }
+
   // Execute one pending CPU task if any (used by wait() to contribute).
   bool progress_one();
 
@@ -260,6 +280,8 @@ class Context {
   std::atomic<bool> stop_{false};
   std::unique_ptr<GpuEngine> gpu_;
   std::unique_ptr<CommEngine> comm_;
+  mutable std::mutex info_mtx_;
+  std::map<std::string, std::string> info_;
 };
 
 void run_cpu_task(Task* t);  // execute hook + complete

@@ -433,3 +433,16 @@ def test_pins_extended_events():
     pm.pins_remove(h)
     assert {"create", "release_deps", "schedule", "complete"} <= seen, seen
     del A, ctx
+
+
+def test_info_registry():
+    """Per-context info registry (class/info.c analog): components publish
+    facts at attach; user keys coexist."""
+    ctx = pm.Context(nworkers=2, rank=0, world=1, gpu=-2)
+    info = ctx.info()
+    assert info["runtime.workers"] == "2"
+    assert info["sched.kind"] == "ws"
+    assert info["comm.kind"] == "null"
+    ctx.info_set("app.phase", "warmup")
+    assert ctx.info()["app.phase"] == "warmup"
+    del ctx

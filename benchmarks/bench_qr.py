@@ -20,6 +20,9 @@ def main():
     ap.add_argument("--tile", type=int, default=1024)
     ap.add_argument("--steps", type=int, default=2)
     ap.add_argument("--warmup", type=int, default=1)
+    ap.add_argument("--algo", type=str,
+                    default=os.environ.get("PARSEC_MCA_qr_algo", "house"),
+                    choices=["house", "bcgs"])
     args = ap.parse_args()
 
     import parsec_amd as pm
@@ -28,12 +31,16 @@ def main():
     n, nb = (args.size, args.tile) if has_gpu else (512, 128)
     ctx = pm.init_distributed(nworkers=4)
     A = pm.TiledMatrix(ctx, n, n, nb, nb, 1, 1)
+    R = pm.TiledMatrix(ctx, n, n, nb, nb, 1, 1) if args.algo == "bcgs"         else None
 
     def run(nsteps):
         tp = pm.Dtd(ctx)
         for _ in range(nsteps):
             pm.insert_full_fill(tp, A, 3)
-            pm.insert_geqrf(tp, A)
+            if args.algo == "bcgs":
+                pm.insert_geqrf_bcgs(tp, A, R)
+            else:
+                pm.insert_geqrf(tp, A)
         tp.wait()
 
     if args.warmup:
@@ -56,6 +63,7 @@ def main():
         "dtype": "fp64",
         "data": "synthetic",
         "config": {"model": "tile_qr_dgeqrf", "N": n, "tile": nb,
+                   "algo": args.algo,
                    "chore_qr": os.environ.get("PARSEC_MCA_chore_qr", "hand")},
     }), flush=True)
     del A, ctx

@@ -45,6 +45,10 @@ void insert_spd_fill(Dtd& tp, TiledMatrix& A, uint32_t seed);
 void insert_full_fill(Dtd& tp, TiledMatrix& A, uint32_t seed);
 void insert_potrf(Dtd& tp, TiledMatrix& A);
 void insert_geqrf(Dtd& tp, TiledMatrix& A);
+// QR by block Gram-Schmidt with CholeskyQR2 panels (matrix-core rates;
+// cond(A) <~ 1e7): A becomes the explicit orthonormal Q, R the upper
+// tiles (kernels_qr_bcgs.cpp).
+void insert_geqrf_bcgs(Dtd& tp, TiledMatrix& A, TiledMatrix& R);
 void insert_getrf_nopiv(Dtd& tp, TiledMatrix& A);
 void insert_fill_bf16(Dtd& tp, TiledMatrix& A, uint32_t seed);
 void insert_gemm_bf16(Dtd& tp, TiledMatrix& At, TiledMatrix& B, TiledMatrix& C);

@@ -420,6 +420,8 @@ PYBIND11_MODULE(_core, m) {
         py::arg("A"), py::call_guard<py::gil_scoped_release>());
   m.def("insert_geqrf", &insert_geqrf, py::arg("tp"), py::arg("A"),
         py::call_guard<py::gil_scoped_release>());
+  m.def("insert_geqrf_bcgs", &insert_geqrf_bcgs, py::arg("tp"),
+        py::arg("A"), py::arg("R"), py::call_guard<py::gil_scoped_release>());
   m.def("insert_redistribute", &insert_redistribute, py::arg("tp"),
         py::arg("src"), py::arg("dst"), py::call_guard<py::gil_scoped_release>());
   m.def("insert_reduce_axis", &insert_reduce_axis, py::arg("tp"),

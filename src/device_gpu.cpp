@@ -3,6 +3,7 @@
 #include <algorithm>
 
 #include "data.hpp"
+#include "gpu_graph.hpp"
 #include "pins.hpp"
 #include "profiling.hpp"
 
@@ -431,6 +432,18 @@ void GpuEngine::stage_in_and_launch(Task* t) {
   t->tc->gpu_hook(*t, gctx);
   if (roctx_on()) roctx_pop();
   PA_PINS(PinsEv::GPU_SUBMIT, t, -1);
+  if (GpuGraphRecorder* rec = g_gpu_recorder.load(std::memory_order_acquire);
+      rec && t->tp == rec->tp) {
+    // hipGraph record pass (gpu_graph.hpp): retain the task, keep every
+    // flow pinned on-device (stable pointers for replays), log the launch.
+    t->retain();
+    for (int i = 0; i < t->nflows; i++)
+      if (Data* d = t->flows[i].data) { pin(d); rec->pinned.push_back(d); }
+    SpinGuard rg(rec->lock);
+    rec->idx_by_seq[t->seq] = (int)rec->log.size();
+    rec->log.push_back(t);
+    rec->stream_of.push_back(si);
+  }
   hipEvent_t ev = spans ? tev_get() : event_get();
   PA_HIP_CHECK(hipEventRecord(ev, es));
   inflight_[si].push_back(
@@ -488,6 +501,11 @@ bool GpuEngine::retire_pass() {
 }
 
 void GpuEngine::run_blocking(Task* t) {
+  if (GpuGraphRecorder* rec = g_gpu_recorder.load(std::memory_order_acquire);
+      rec && t->tp == rec->tp)
+    rec->fail("host-blocking GPU chore '%s' is not graph-capturable "
+              "(it synchronizes inside the hook)",
+              t->tc->name.c_str());
   PA_HIP_CHECK(hipSetDevice(device_));
   static thread_local hipStream_t bs = nullptr;
   if (!bs) PA_HIP_CHECK(hipStreamCreateWithFlags(&bs, hipStreamNonBlocking));

@@ -72,6 +72,12 @@ class Dtd : public Taskpool {
   void flush(Data* d);
   void flush_all(TiledMatrix& A);
 
+  // hipGraph capture/replay of this pool's GPU schedule (gpu_graph.hpp):
+  // capture_begin() BEFORE the inserts, capture_end() instead of wait()
+  // (it waits — the record pass — then instantiates the graph).
+  void capture_begin();
+  std::unique_ptr<class GpuGraph> capture_end();
+
  private:
   // read/write_flow and make_recv return the (possibly renamed) current
   // Data: when a recv or an OUTPUT-only rewrite targets a tile whose old

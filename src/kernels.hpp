@@ -32,6 +32,7 @@ TaskClass& tc_potrf();
 TaskClass& tc_trsm();
 TaskClass& tc_syrk();
 TaskClass& tc_gemm();
+TaskClass& tc_gemm_nn();
 
 TaskClass& tc_trtri();
 TaskClass& tc_trsm_inv();
@@ -60,6 +61,9 @@ void insert_subtile_extract(Dtd& tp, TiledMatrix& A, int tm, int tn,
 void insert_subtile_insert(Dtd& tp, TiledMatrix& S, TiledMatrix& A, int tm,
                            int tn);
 void insert_apply_scale(Dtd& tp, TiledMatrix& A, double alpha, double beta);
+// Tiled C = A*B, fp64 NN (idempotent DAG: k==0 overwrites C — the hipGraph
+// capture/replay demo workload, and a plain library-GEMM DAG generally).
+void insert_gemm_fp64(Dtd& tp, TiledMatrix& A, TiledMatrix& B, TiledMatrix& C);
 void insert_reduce_sum(Dtd& tp, TiledMatrix& A, TiledMatrix& R);
 void insert_stencil_1d(Dtd& tp, TiledMatrix& Src, TiledMatrix& Dst);
 void insert_panel_fill(Dtd& tp, TiledMatrix& A, uint32_t seed);

@@ -909,4 +909,84 @@ void insert_potrf(Dtd& tp, TiledMatrix& A) {
   }
 }
 
+// ------------------------------------------------------- plain NN GEMM DAG
+// Tiled C = A * B (fp64, rocBLAS NN per tile). The k==0 task overwrites C
+// (beta = 0, OUTPUT flow), later k accumulate — which makes the whole DAG
+// IDEMPOTENT: replaying it (e.g. from a captured hipGraph, gpu_graph.hpp)
+// reproduces the same C from the same A/B.
+struct GemmNNArgs {
+  int m, n, k, lda, ldb, ldc;
+  double alpha, beta;
+};
+
+static void cpu_gemm_nn(Task& t) {
+  const GemmNNArgs& a = t.arg<GemmNNArgs>();
+  const double* A = (const double*)t.flows[0].data->pull_to_host();
+  const double* B = (const double*)t.flows[1].data->pull_to_host();
+  // beta == 0 rides an OUTPUT-only flow: nothing valid to pull yet
+  double* C = a.beta == 0.0 ? (double*)t.flows[2].data->ensure_host()
+                            : (double*)t.flows[2].data->pull_to_host();
+  for (int j = 0; j < a.n; j++)
+    for (int i = 0; i < a.m; i++) {
+      double s = 0;
+      for (int p = 0; p < a.k; p++)
+        s += A[(size_t)p * a.lda + i] * B[(size_t)j * a.ldb + p];
+      double c0 = a.beta == 0.0 ? 0.0 : a.beta * C[(size_t)j * a.ldc + i];
+      C[(size_t)j * a.ldc + i] = c0 + a.alpha * s;
+    }
+  t.flows[2].data->written_on(false);
+}
+
+static void gpu_gemm_nn(Task& t, GpuTaskCtx& g) {
+  const GemmNNArgs& a = t.arg<GemmNNArgs>();
+  rocblas_status s = rocblas_dgemm(
+      blas_handle(g), rocblas_operation_none, rocblas_operation_none, a.m,
+      a.n, a.k, &a.alpha, (const double*)t.dev_ptr[0], a.lda,
+      (const double*)t.dev_ptr[1], a.ldb, &a.beta, (double*)t.dev_ptr[2],
+      a.ldc);
+  PA_CHECK(s == rocblas_status_success, "rocblas_dgemm NN failed: %d", (int)s);
+}
+
+TaskClass& tc_gemm_nn() {
+  static TaskClass tc =
+      make_tc("gemm_nn", TaskKind::GPU, cpu_gemm_nn, gpu_gemm_nn, 70);
+  return tc;
+}
+
+void insert_gemm_fp64(Dtd& tp, TiledMatrix& A, TiledMatrix& B,
+                      TiledMatrix& C) {
+  PA_CHECK(A.nt() == B.mt() && A.mt() == C.mt() && B.nt() == C.nt(),
+           "insert_gemm_fp64: tile-grid mismatch");
+  for (int i = 0; i < C.mt(); i++)
+    for (int j = 0; j < C.nt(); j++)
+      for (int p = 0; p < A.nt(); p++) {
+        // tiles are stored at the collection's full ld (= mb), partial
+        // edge tiles included (see tile_numpy / TiledMatrix layout)
+        GemmNNArgs a{C.tile_rows(i), C.tile_cols(j), A.tile_cols(p),
+                     A.mb(),         B.mb(),         C.mb(),
+                     1.0,            p == 0 ? 0.0 : 1.0};
+        Dtd::FlowSpec f[] = {{A.tile(i, p), ACCESS_IN},
+                             {B.tile(p, j), ACCESS_IN},
+                             {C.tile(i, j), p == 0 ? ACCESS_OUT
+                                                   : ACCESS_INOUT}};
+        tp.insert(&tc_gemm_nn(), &a, sizeof(a), f, 3, 0, C.rank_of(i, j));
+      }
+}
+
+// Pre-create the per-stream rocBLAS handle and give it a fixed device
+// workspace so no allocation can happen inside a hipStream capture
+// (gpu_graph.cpp calls this for each capture stream before BeginCapture;
+// same thread, so the thread_local handle map matches the replay).
+void blas_warm_stream_for_capture(hipStream_t s) {
+  GpuTaskCtx g{s, 0, nullptr, nullptr};
+  rocblas_handle h = blas_handle(g);
+  static thread_local std::map<void*, void*> ws;
+  void*& w = ws[(void*)s];
+  if (!w) {
+    const size_t bytes = 1u << 26;
+    PA_HIP_CHECK(hipMalloc(&w, bytes));
+    PA_CHECK(rocblas_set_workspace(h, w, bytes) == rocblas_status_success);
+  }
+}
+
 }  // namespace pa

@@ -15,6 +15,7 @@
 #include "device_gpu.hpp"
 #include "comm.hpp"
 #include "dtd.hpp"
+#include "gpu_graph.hpp"
 #include "kernels.hpp"
 #include "pins.hpp"
 #include "profiling.hpp"
@@ -231,6 +232,8 @@ PYBIND11_MODULE(_core, m) {
            py::arg("elem_size") = 8, py::arg("sym") = false,
            py::keep_alive<1, 2>())
       .def_property_readonly("_handle", [](TiledMatrix& a) { return (uintptr_t)&a; })
+      .def_property_readonly("m", &TiledMatrix::m)
+      .def_property_readonly("n", &TiledMatrix::n)
       .def_property_readonly("mt", &TiledMatrix::mt)
       .def_property_readonly("nt", &TiledMatrix::nt)
       .def_property_readonly("mb", &TiledMatrix::mb)
@@ -396,7 +399,19 @@ PYBIND11_MODULE(_core, m) {
            py::arg("with_data") = false)
       .def("flush", &Dtd::flush, py::call_guard<py::gil_scoped_release>())
       .def("flush_all", &Dtd::flush_all,
-           py::call_guard<py::gil_scoped_release>());
+           py::call_guard<py::gil_scoped_release>())
+      .def("capture_begin", &Dtd::capture_begin)
+      .def("capture_end", &Dtd::capture_end,
+           py::call_guard<py::gil_scoped_release>(),
+           py::keep_alive<0, 1>());  // graph keeps the pool (and ctx) alive
+
+  // hipGraph replay handle (gpu_graph.hpp): launch-bound steady-state DAGs
+  // re-run with ONE hipGraphLaunch per iteration.
+  py::class_<GpuGraph>(m, "GpuGraph")
+      .def("launch", &GpuGraph::launch, py::arg("iters") = 1,
+           py::call_guard<py::gil_scoped_release>())
+      .def_property_readonly("nodes", &GpuGraph::nodes)
+      .def_property_readonly("n_tasks", &GpuGraph::n_tasks);
 
   m.attr("ACCESS_IN") = (int)ACCESS_IN;
   m.attr("ACCESS_OUT") = (int)ACCESS_OUT;
@@ -434,6 +449,9 @@ PYBIND11_MODULE(_core, m) {
         py::call_guard<py::gil_scoped_release>());
   m.def("insert_subtile_insert", &insert_subtile_insert, py::arg("tp"),
         py::arg("S"), py::arg("A"), py::arg("tm"), py::arg("tn"),
+        py::call_guard<py::gil_scoped_release>());
+  m.def("insert_gemm_fp64", &insert_gemm_fp64, py::arg("tp"), py::arg("A"),
+        py::arg("B"), py::arg("C"),
         py::call_guard<py::gil_scoped_release>());
   m.def("insert_apply_scale", &insert_apply_scale, py::arg("tp"), py::arg("A"),
         py::arg("alpha"), py::arg("beta"),

@@ -4,6 +4,7 @@
 
 #include "data.hpp"
 #include "device_gpu.hpp"
+#include "gpu_graph.hpp"
 #include "comm.hpp"
 #include "pins.hpp"
 #include "profiling.hpp"
@@ -91,6 +92,14 @@ void Task::release() {
 bool task_add_edge(Task* pred, Task* succ) {
   Profiler& pr = Profiler::inst();
   if (pr.dot_enabled()) pr.dot_edge(pred, succ);
+  // hipGraph record pass: keep the edge even when the predecessor already
+  // completed — a REPLAY runs everything concurrently, so the device-side
+  // ordering must be rebuilt from every edge (gpu_graph.hpp).
+  if (GpuGraphRecorder* rec = g_gpu_recorder.load(std::memory_order_acquire);
+      rec && pred->tp == rec->tp && succ->tp == rec->tp) {
+    SpinGuard g(rec->lock);
+    rec->edges.emplace_back(pred->seq, succ->seq);
+  }
   // Dependency-release race protocol (SURVEY.md §7 "hard parts"): the edge
   // is registered under the predecessor's lock; if the predecessor already
   // completed, the successor does not wait on it.
@@ -135,6 +144,10 @@ void run_cpu_task(Task* t) {
     t->tp->context()->gpu()->run_blocking(t);
     return;
   }
+  if (GpuGraphRecorder* rec = g_gpu_recorder.load(std::memory_order_acquire);
+      rec && t->tp == rec->tp)
+    rec->fail("CPU task '%s' in a captured taskpool is not replayable",
+              t->tc->name.c_str());
   Profiler& pr = Profiler::inst();
   // OUTPUT-only flows are about to be produced on the host: drop stale
   // device validity first so LRU writeback cannot race the body's writes.

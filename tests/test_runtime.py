@@ -4,6 +4,8 @@ Mirrors the reference's tests/dsl/dtd/* programs (dtd_test_war.c etc.,
 SURVEY.md §4) using Python task bodies on the CPU path.
 """
 import os
+
+import pytest
 import threading
 
 import parsec_amd as pm
@@ -446,3 +448,43 @@ def test_info_registry():
     ctx.info_set("app.phase", "warmup")
     assert ctx.info()["app.phase"] == "warmup"
     del ctx
+
+
+def test_gemm_fp64_dag_cpu_numerics():
+    """insert_gemm_fp64 (tiled NN C=A*B; the hipGraph-replay demo DAG)
+    against NumPy, including partial edge tiles; idempotence: re-running
+    the pool reproduces the same C (k==0 overwrites with beta=0)."""
+    import numpy as np
+    ctx = pm.Context(nworkers=2, rank=0, world=1, gpu=-2)
+    A = pm.TiledMatrix(ctx, 300, 150, 64, 64, 1, 1)
+    B = pm.TiledMatrix(ctx, 150, 200, 64, 64, 1, 1)
+    C = pm.TiledMatrix(ctx, 300, 200, 64, 64, 1, 1)
+    tp = pm.Dtd(ctx)
+    pm.insert_full_fill(tp, A, 3)
+    pm.insert_full_fill(tp, B, 5)
+    tp.wait()
+
+    def full(M):
+        out = np.zeros((M.m, M.n))
+        for i in range(M.mt):
+            for j in range(M.nt):
+                out[i * M.mb:i * M.mb + M.tile_rows(i),
+                    j * M.nb:j * M.nb + M.tile_cols(j)] = M.tile_numpy(i, j)
+        return out
+
+    ref = full(A) @ full(B)
+    for _ in range(2):  # second pass checks idempotence
+        tp2 = pm.Dtd(ctx)
+        pm.insert_gemm_fp64(tp2, A, B, C)
+        tp2.wait()
+        err = abs(full(C) - ref).max() / abs(ref).max()
+        assert err < 1e-13, err
+    del A, B, C, ctx
+
+
+def test_gpu_graph_capture_requires_gpu():
+    ctx = pm.Context(nworkers=2, rank=0, world=1, gpu=-2)
+    tp = pm.Dtd(ctx)
+    with pytest.raises(RuntimeError, match="GPU"):
+        tp.capture_begin()
+    del tp, ctx

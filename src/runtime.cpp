@@ -325,22 +325,42 @@ Scheduler::Scheduler(int nworkers) : nworkers_(nworkers) {
   // rnd = random victim order (sched_rnd, for schedule-robustness tests).
   // The priority queue always serves prioritized and externally-released
   // tasks.
+  // Module names accept the reference's MCA names as aliases: lfq -> ws,
+  // gd -> fifo, ll -> lifo, ap -> spq, ltq/llp/lhq/pbq -> pbq (per-worker
+  // priority heaps + nearest-first steal: on one NUMA-ish node per GPU the
+  // hbbuffer hierarchy collapses to this), ip = inverse priority.
   std::string kind = param_str("sched", "ws");
-  mode_ = kind == "fifo" ? 1
-          : kind == "lifo" ? 2
-          : kind == "spq" ? 3
+  mode_ = kind == "fifo" || kind == "gd" ? 1
+          : kind == "lifo" || kind == "ll" ? 2
+          : kind == "spq" || kind == "ap" ? 3
           : kind == "rnd" ? 4
+          : kind == "pbq" || kind == "ltq" || kind == "llp" || kind == "lhq"
+                ? 5
+          : kind == "ip" ? 6
           : 0;
+  invert_prio_ = kind == "ip";
   for (int i = 0; i < nworkers_; i++) wq_.emplace_back(new WorkerQ());
 }
 
 Scheduler::~Scheduler() = default;
 
 void Scheduler::push(Task* t, int worker_hint) {
-  if (mode_ == 3) {  // spq: single shared priority queue
+  const int32_t key = invert_prio_ ? -t->priority : t->priority;
+  if (mode_ == 3 || mode_ == 6) {  // spq/ip: single shared priority queue
     pq_lock_.lock();
-    pq_.push(PQEntry{t});
+    pq_.push(PQEntry{t, key});
     pq_lock_.unlock();
+    npending_.fetch_add(1, std::memory_order_release);
+    sleep_cv_.notify_one();
+    return;
+  }
+  if (mode_ == 5) {  // pbq/ltq: per-worker priority heaps
+    int w = worker_hint >= 0 && worker_hint < nworkers_ ? worker_hint : 0;
+    WorkerQ& q = *wq_[w];
+    q.lock.lock();
+    q.heap.push(PQEntry{t, key});
+    q.sz.store((uint32_t)q.heap.size(), std::memory_order_relaxed);
+    q.lock.unlock();
     npending_.fetch_add(1, std::memory_order_release);
     sleep_cv_.notify_one();
     return;
@@ -364,7 +384,7 @@ void Scheduler::push(Task* t, int worker_hint) {
     q.lock.unlock();
   } else {
     pq_lock_.lock();
-    pq_.push(PQEntry{t});
+    pq_.push(PQEntry{t, key});
     pq_lock_.unlock();
   }
   npending_.fetch_add(1, std::memory_order_release);
@@ -373,7 +393,37 @@ void Scheduler::push(Task* t, int worker_hint) {
 
 Task* Scheduler::pop(int worker) {
   Task* t = nullptr;
-  if (mode_ == 3) {
+  if (mode_ == 5) {  // pbq/ltq: local heap, then steal other heaps' tops
+    int w = worker >= 0 ? worker : 0;
+    for (int d = 0; d < nworkers_; d++) {
+      int v = (w + d) % nworkers_;
+      WorkerQ& q = *wq_[v];
+      if (d > 0 && q.sz.load(std::memory_order_relaxed) == 0) continue;
+      q.lock.lock();
+      if (!q.heap.empty()) {
+        t = q.heap.top().t;
+        q.heap.pop();
+        q.sz.store((uint32_t)q.heap.size(), std::memory_order_relaxed);
+      }
+      q.lock.unlock();
+      if (t) {
+        npending_.fetch_sub(1, std::memory_order_relaxed);
+        if (d > 0) {
+          counters().steals.fetch_add(1, std::memory_order_relaxed);
+          PA_PINS(PinsEv::STEAL, t, worker);
+        }
+        return t;
+      }
+    }
+    // externally-released tasks (hint -1 lands on worker 0's heap), plus
+    // anything routed through the shared pq by other components
+    pq_lock_.lock();
+    if (!pq_.empty()) { t = pq_.top().t; pq_.pop(); }
+    pq_lock_.unlock();
+    if (t) npending_.fetch_sub(1, std::memory_order_relaxed);
+    return t;
+  }
+  if (mode_ == 3 || mode_ == 6) {
     pq_lock_.lock();
     if (!pq_.empty()) { t = pq_.top().t; pq_.pop(); }
     pq_lock_.unlock();

@@ -194,23 +194,27 @@ class Scheduler {
   size_t approx_pending() const { return npending_.load(std::memory_order_relaxed); }
 
  private:
+  struct PQEntry {
+    Task* t;
+    int32_t key;  // priority (negated under sched=ip)
+    bool operator<(const PQEntry& o) const {
+      if (key != o.key) return key < o.key;
+      return t->seq > o.t->seq;  // earlier insertion first
+    }
+  };
   struct WorkerQ {
     SpinLock lock;
     std::deque<Task*> dq;
+    // per-worker priority heap (sched=pbq/ltq modes — maxheap.c analog)
+    std::priority_queue<PQEntry> heap;
     // Approximate size for the steal fast path: reading dq.empty() on
     // another worker's deque without its lock is a data race; this atomic
     // mirror is the race-free heuristic.
     std::atomic<uint32_t> sz{0};
   };
-  struct PQEntry {
-    Task* t;
-    bool operator<(const PQEntry& o) const {
-      if (t->priority != o.t->priority) return t->priority < o.t->priority;
-      return t->seq > o.t->seq;  // earlier insertion first
-    }
-  };
   int nworkers_;
-  int mode_ = 0;  // 0=ws 1=fifo 2=lifo (PARSEC_MCA_sched)
+  int mode_ = 0;  // 0=ws 1=fifo 2=lifo 3=spq 4=rnd 5=pbq 6=ip (PARSEC_MCA_sched)
+  bool invert_prio_ = false;  // sched=ip: inverse priority order
   std::vector<std::unique_ptr<WorkerQ>> wq_;
   SpinLock pq_lock_;
   std::priority_queue<PQEntry> pq_;

@@ -121,11 +121,15 @@ def test_taskpool_reuse_after_wait(ctx):
 
 
 def test_sched_variants():
-    """MCA sched module analog: fifo/lifo shared-queue schedulers run the
-    same DAG correctly."""
+    """MCA sched module analog (mca/sched/*): every module (and every
+    reference-name alias: lfq/gd/ll/ap/ltq/llp/lhq) runs the same DAG
+    correctly — ws (lfq), fifo (gd), lifo (ll), spq (ap), rnd, pbq
+    (ltq/llp/lhq: per-worker priority heaps + steal), ip (inverse
+    priority)."""
     import subprocess, sys, os
     repo = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
-    for kind in ["fifo", "lifo", "ws"]:
+    for kind in ["fifo", "lifo", "ws", "spq", "rnd", "pbq", "ip", "ltq",
+                 "gd", "ap"]:
         code = f"""
 import sys; sys.path.insert(0, {repo!r})
 import numpy as np

@@ -15,7 +15,7 @@ LDFLAGS    := -shared -L$(ROCM)/lib -lrocblas -lrocsolver -lrccl
 
 SRCS       := src/common.cpp src/runtime.cpp src/device_gpu.cpp src/comm.cpp \
               src/rccl_comm.cpp src/capi.cpp src/profiling.cpp src/dtd.cpp src/kernels_blas.cpp src/kernels_qr.cpp src/kernels_lu.cpp src/kernels_bf16.cpp src/kernels_panel.cpp src/kernels_reshape.cpp src/kernels_qr_bcgs.cpp \
-              src/kernels_hip.cpp src/gpu_graph.cpp src/pybind.cpp
+              src/kernels_hip.cpp src/gpu_graph.cpp src/pins_modules.cpp src/pybind.cpp
 OBJS       := $(SRCS:src/%.cpp=build/%.o)
 COBJS      := $(filter-out build/pybind.o,$(OBJS))
 TARGET     := parsec_amd/_core.so

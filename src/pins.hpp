@@ -99,6 +99,12 @@ class Pins {
   int next_id_ = 1;
 };
 
+// Built-in modules (src/pins_modules.cpp): PARSEC_MCA_pins=
+// task_profiler,print_steals,iterators_checker — installed at context
+// construction, reports at teardown.
+void pins_modules_install();
+void pins_modules_finalize(int rank);
+
 #define PA_PINS(ev, task, worker)                           \
   do {                                                      \
     auto& _p = ::pa::Pins::inst();                          \

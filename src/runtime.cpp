@@ -524,6 +524,7 @@ Context::Context(const Options& opt) : rank_(opt.rank), world_(opt.world) {
   if (comm_) info_set("comm.kind", comm_->kind());
   if (gpu_) gpu_->publish_info(this);
 
+  pins_modules_install();
   roctx_init();
   debug_history_init();
   std::string prof = param_str("profile_filename", "");
@@ -540,6 +541,7 @@ Context::~Context() {
   stop_.store(true, std::memory_order_release);
   sched_->wake_all();
   for (auto& w : workers_) w.join();
+  pins_modules_finalize(rank_);
   if (param_int("stats", 0)) {
     // device-statistics table at fini (device.c:611-658 analog):
     // per-rank task counts, required-vs-transferred bytes, evictions,

@@ -214,3 +214,29 @@ print("CT_OK", rank)
         text = open(str(tmp_path / f"t{r}.json")).read()
         found_comm += text.count("comm_send") + text.count("comm_recv")
     assert found_comm > 0, "no comm events in either rank's trace"
+
+
+def test_pins_builtin_modules():
+    """PARSEC_MCA_pins=task_profiler,print_steals,iterators_checker: the
+    built-in PINS modules (mca/pins/* analogs) report at teardown."""
+    code = f"""
+import sys; sys.path.insert(0, {REPO!r})
+import parsec_amd as pm
+pm.param_set("pins", "task_profiler,print_steals,iterators_checker")
+ctx = pm.Context(nworkers=2, rank=0, world=1, gpu=-2)
+A = pm.TiledMatrix(ctx, 512, 512, 64, 64, 1, 1)
+tp = pm.Dtd(ctx)
+pm.insert_spd_fill(tp, A, 1)
+pm.insert_potrf(tp, A)
+tp.wait()
+del A, tp, ctx
+print("MODS_OK")
+"""
+    r = subprocess.run([sys.executable, "-c", code], capture_output=True,
+                       text=True, timeout=120)
+    assert "MODS_OK" in r.stdout, r.stdout + r.stderr
+    assert "[pins:task_profiler]" in r.stderr and "potrf" in r.stderr, r.stderr
+    assert "[pins:print_steals]" in r.stderr
+    assert "[pins:iterators_checker]" in r.stderr
+    assert "OK: every created task completed" in r.stderr, r.stderr
+    assert "ANOMALY" not in r.stderr

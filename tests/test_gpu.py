@@ -394,3 +394,15 @@ def test_gpu_graph_replay_timing(gctx):
           f"{pool_ms:.3f} ms/iter ({g.n_tasks} tasks)")
     assert graph_ms < pool_ms * 1.5
     del g, tp, tp2, A, B, C
+
+
+def test_jacobi_replay_example_gpu():
+    """examples/jacobi_replay.py takes the hipGraph replay path on a GPU
+    and converges to the same fixed point."""
+    import subprocess
+    r = subprocess.run([sys.executable,
+                        os.path.join(REPO, "examples", "jacobi_replay.py"),
+                        "4", "64", "50"],
+                       capture_output=True, text=True, timeout=300)
+    assert r.returncode == 0, r.stdout + r.stderr
+    assert "hipGraph replay" in r.stdout, r.stdout

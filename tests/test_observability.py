@@ -240,3 +240,13 @@ print("MODS_OK")
     assert "[pins:iterators_checker]" in r.stderr
     assert "OK: every created task completed" in r.stderr, r.stderr
     assert "ANOMALY" not in r.stderr
+
+
+def test_jacobi_replay_example_cpu():
+    """examples/jacobi_replay.py converges on the CPU engine path."""
+    r = subprocess.run([sys.executable,
+                        os.path.join(REPO, "examples", "jacobi_replay.py"),
+                        "3", "32", "40"],
+                       capture_output=True, text=True, timeout=300)
+    assert r.returncode == 0, r.stdout + r.stderr
+    assert "fixed-point err" in r.stdout

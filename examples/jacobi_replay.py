@@ -53,6 +53,7 @@ def main(nt=4, nb=64, sweeps=50):
         pm.insert_gemm_fp64(tp, A, Y, X)  # x = M^ y
 
     t0 = time.perf_counter()
+    g = None
     if ctx.has_gpu:
         tp = pm.Dtd(ctx)
         tp.capture_begin()
@@ -75,6 +76,9 @@ def main(nt=4, nb=64, sweeps=50):
     print(f"jacobi_replay: {mode}; {sweeps} double sweeps, "
           f"{dt:.3f} ms/sweep, fixed-point err {err:.2e}")
     assert err < 1e-8, err
+    # HIP-graph lifetime rule: destroy the graph BEFORE the collections it
+    # pins (same as destroying buffers under a live cudaGraph is UB)
+    del g
     del A, X, Y, ctx
     return err
 

@@ -1,5 +1,7 @@
 #include "dtd.hpp"
 
+#include "gpu_graph.hpp"
+
 #include "kernels.hpp"
 #include "profiling.hpp"
 
@@ -103,6 +105,12 @@ Data* Dtd::maybe_rename(Data* d) {
   if (!any_live) return d;  // nobody reads the old version: reuse in place
   if (!d->coll && !d->icoll)
     return d;  // standalone scratch datum (NEW tile): WAR-serialize instead
+  // hipGraph record pass: replays need STABLE buffers, and the reclaim
+  // task is a CPU task — WAR-serialize through the caller's edges instead
+  // of renaming (gpu_graph.hpp).
+  if (GpuGraphRecorder* rec = g_gpu_recorder.load(std::memory_order_acquire);
+      rec && rec->tp == this)
+    return d;
   Data* nd = d->coll ? d->coll->rename_tile(d) : d->icoll->rename(d);
   nd->version = d->version;
   nd->owner_rank = d->owner_rank;

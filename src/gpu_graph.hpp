@@ -19,6 +19,10 @@
 // Semantics follow CUDA/HIP graphs: replays re-run the same kernels on
 // the SAME device buffers (tiles are pinned on-device for the graph's
 // lifetime), so the DAG should be idempotent or externally re-seeded.
+// Lifetime: destroy the GpuGraph BEFORE the collections whose tiles it
+// pins (dropping a buffer under a live graph is UB, as with CUDA graphs).
+// Copy renaming is suspended for the captured pool during the record pass
+// (replays need stable buffers): OUTPUT rewrites WAR-serialize instead.
 // CPU tasks, comm tasks, and host-blocking GPU chores in the captured
 // pool fail capture loudly with the offending task named.
 #pragma once

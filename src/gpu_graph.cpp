@@ -27,6 +27,9 @@ void GpuGraphRecorder::fail(const char* fmt, const char* a) {
 }
 
 GpuGraph::GpuGraph(GpuEngine* eng, GpuGraphRecorder&& rec) : eng_(eng) {
+  const bool dbg = param_int("graph_debug", 0) != 0;
+#define GG_DBG(...) do { if (dbg) { fprintf(stderr, "[graph] " __VA_ARGS__); fputc('\n', stderr); fflush(stderr); } } while (0)
+  GG_DBG("ctor: %zu tasks %zu edges", rec.log.size(), rec.edges.size());
   tasks_ = std::move(rec.log);
   pinned_ = std::move(rec.pinned);
   const int n = (int)tasks_.size();
@@ -45,25 +48,42 @@ GpuGraph::GpuGraph(GpuEngine* eng, GpuGraphRecorder&& rec) : eng_(eng) {
     }
     cs_.resize(remap.size());
   }
+  GG_DBG("streams: %zu", cs_.size());
   for (auto& s : cs_) {
     PA_HIP_CHECK(hipStreamCreateWithFlags(&s, hipStreamNonBlocking));
     blas_warm_stream_for_capture(s);
   }
+  GG_DBG("warmed");
 
-  // Cross-stream dependency edges from the record pass (same-stream order
-  // is implied by issue order). Launch order is a valid topological order:
-  // a GPU successor is only enqueued after its predecessor's event
-  // completed on the host.
-  std::vector<std::vector<int>> xpreds(n);
-  std::vector<hipEvent_t> edge_ev(n, nullptr);
-  std::vector<char> needs_ev(n, 0);
-  for (auto& [ps, ss] : rec.edges) {
-    auto pi = rec.idx_by_seq.find(ps), si = rec.idx_by_seq.find(ss);
-    if (pi == rec.idx_by_seq.end() || si == rec.idx_by_seq.end()) continue;
-    int p = pi->second, s = si->second;
-    PA_CHECK(p < s, "graph capture: edge against launch order (%d -> %d)", p, s);
-    if (smap[p] != smap[s]) { xpreds[s].push_back(p); needs_ev[p] = 1; }
+  // Cross-stream ordering via WAVEFRONT LEVELS with a fork/join barrier
+  // between consecutive levels. Arbitrary per-edge event meshes crash
+  // hipStreamEndCapture on ROCm 7.2 at >2 streams (isolated with
+  // build/capture_repro probes: fork/join and short chains are fine, the
+  // general mesh is not) — the barrier pattern is the one ROCm handles,
+  // costs ~(streams+1) events per level, and keeps all within-level
+  // parallelism: tasks at equal depth run concurrently across streams,
+  // which is the wavefront schedule of the DAG.
+  std::vector<int> level(n, 0);
+  {
+    std::vector<std::vector<int>> preds(n);
+    for (auto& [ps, ss] : rec.edges) {
+      auto pi = rec.idx_by_seq.find(ps), si = rec.idx_by_seq.find(ss);
+      if (pi == rec.idx_by_seq.end() || si == rec.idx_by_seq.end()) continue;
+      int p = pi->second, q = si->second;
+      PA_CHECK(p < q, "graph capture: edge against launch order (%d -> %d)",
+               p, q);
+      preds[q].push_back(p);
+    }
+    for (int k = 0; k < n; k++)
+      for (int p : preds[k])
+        if (level[p] + 1 > level[k]) level[k] = level[p] + 1;
   }
+  // Stable order by (level, launch order): preserves the record pass's
+  // stream spreading within each level.
+  std::vector<int> order(n);
+  for (int k = 0; k < n; k++) order[k] = k;
+  std::stable_sort(order.begin(), order.end(),
+                   [&](int a, int b) { return level[a] < level[b]; });
 
   auto mkevent = [&]() {
     hipEvent_t e;
@@ -71,42 +91,58 @@ GpuGraph::GpuGraph(GpuEngine* eng, GpuGraphRecorder&& rec) : eng_(eng) {
     evs_.push_back(e);
     return e;
   };
-
-  // Record pass is done and the engine is idle; re-issue the exact launch
-  // sequence into a stream capture. ThreadLocal mode: the engine's idle
-  // manager thread keeps polling its own (uncaptured) streams legally.
-  PA_HIP_CHECK(hipStreamBeginCapture(cs_[0], hipStreamCaptureModeThreadLocal));
-  hipEvent_t fork = mkevent();
-  PA_HIP_CHECK(hipEventRecord(fork, cs_[0]));
-  for (size_t s = 1; s < cs_.size(); s++)
-    PA_HIP_CHECK(hipStreamWaitEvent(cs_[s], fork, 0));
-  for (int k = 0; k < n; k++) {
-    hipStream_t s = cs_[smap[k]];
-    for (int p : xpreds[k]) {
-      PA_CHECK(edge_ev[p], "graph capture: missing edge event");
-      PA_HIP_CHECK(hipStreamWaitEvent(s, edge_ev[p], 0));
+  auto barrier = [&]() {  // join every stream into cs_[0], fork back out
+    for (size_t s2 = 1; s2 < cs_.size(); s2++) {
+      hipEvent_t je = mkevent();
+      PA_HIP_CHECK(hipEventRecord(je, cs_[s2]));
+      PA_HIP_CHECK(hipStreamWaitEvent(cs_[0], je, 0));
     }
+    hipEvent_t fe = mkevent();
+    PA_HIP_CHECK(hipEventRecord(fe, cs_[0]));
+    for (size_t s2 = 1; s2 < cs_.size(); s2++)
+      PA_HIP_CHECK(hipStreamWaitEvent(cs_[s2], fe, 0));
+  };
+
+  // Record pass is done and the engine is idle; re-issue the launches into
+  // a stream capture. ThreadLocal mode: the engine's idle manager thread
+  // keeps polling its own (uncaptured) streams legally.
+  GG_DBG("edges built");
+  PA_HIP_CHECK(hipStreamBeginCapture(cs_[0], hipStreamCaptureModeThreadLocal));
+  {
+    hipEvent_t fork = mkevent();
+    PA_HIP_CHECK(hipEventRecord(fork, cs_[0]));
+    for (size_t s2 = 1; s2 < cs_.size(); s2++)
+      PA_HIP_CHECK(hipStreamWaitEvent(cs_[s2], fork, 0));
+  }
+  int cur_level = 0;
+  for (int oi = 0; oi < n; oi++) {
+    int k = order[oi];
+    if (level[k] != cur_level) {
+      barrier();
+      cur_level = level[k];
+    }
+    hipStream_t s = cs_[smap[k]];
     Task* t = tasks_[k];
+    GG_DBG("launch %d/%d %s seq=%lu stream=%d level=%d", oi, n,
+           t->tc->name.c_str(), (unsigned long)t->seq, smap[k], level[k]);
     std::vector<std::pair<void*, size_t>> deferred;
     GpuTaskCtx gctx{s, eng_->device(), eng_, &deferred};
     t->tc->gpu_hook(*t, gctx);
     for (auto& db : deferred) deferred_.push_back(db);
-    // An event only where some task on another stream waits on k.
-    if (needs_ev[k]) {
-      edge_ev[k] = mkevent();
-      PA_HIP_CHECK(hipEventRecord(edge_ev[k], s));
-    }
   }
   for (size_t s = 1; s < cs_.size(); s++) {
     hipEvent_t je = mkevent();
     PA_HIP_CHECK(hipEventRecord(je, cs_[s]));
     PA_HIP_CHECK(hipStreamWaitEvent(cs_[0], je, 0));
   }
+  GG_DBG("ending capture");
   PA_HIP_CHECK(hipStreamEndCapture(cs_[0], &graph_));
   PA_HIP_CHECK(hipGraphInstantiate(&exec_, graph_, nullptr, nullptr, 0));
   size_t nn = 0;
   PA_HIP_CHECK(hipGraphGetNodes(graph_, nullptr, &nn));
   n_nodes_ = (int)nn;
+  GG_DBG("instantiated: %d nodes", n_nodes_);
+#undef GG_DBG
 }
 
 void GpuGraph::launch(int iters) {
@@ -147,6 +183,9 @@ std::unique_ptr<GpuGraph> Dtd::capture_end() {
   PA_CHECK(rec && rec->tp == this, "capture_end without capture_begin");
   wait();  // the RECORD pass: normal execution with logging
   g_gpu_recorder.store(nullptr, std::memory_order_release);
+  if (param_int("graph_debug", 0))
+    fprintf(stderr, "[graph] record pass done: %zu launches, failed=%d\n",
+            rec->log.size(), (int)rec->failed);
   std::unique_ptr<GpuGraphRecorder> owned(rec);
   GpuEngine* eng = context()->gpu();
   if (owned->failed || owned->log.empty()) {

@@ -937,8 +937,31 @@ static void cpu_gemm_nn(Task& t) {
   t.flows[2].data->written_on(false);
 }
 
+// Thin-C tiles (vector iterations, skinny updates): a library GEMM is all
+// launch overhead at n<=4, and a plain kernel is trivially capture-safe.
+__global__ void k_gemm_nn_thin(int m, int n, int k, double alpha,
+                               const double* A, int lda, const double* B,
+                               int ldb, double beta, double* C, int ldc) {
+  int i = blockIdx.x * blockDim.x + threadIdx.x;
+  int j = blockIdx.y;
+  if (i >= m || j >= n) return;
+  double s = 0;
+  for (int p = 0; p < k; p++)
+    s += A[(size_t)p * lda + i] * B[(size_t)j * ldb + p];
+  double c0 = beta == 0.0 ? 0.0 : beta * C[(size_t)j * ldc + i];
+  C[(size_t)j * ldc + i] = c0 + alpha * s;
+}
+
 static void gpu_gemm_nn(Task& t, GpuTaskCtx& g) {
   const GemmNNArgs& a = t.arg<GemmNNArgs>();
+  if (a.n <= 4) {
+    hipLaunchKernelGGL(k_gemm_nn_thin, dim3((a.m + 63) / 64, a.n), dim3(64),
+                       0, g.stream, a.m, a.n, a.k, a.alpha,
+                       (const double*)t.dev_ptr[0], a.lda,
+                       (const double*)t.dev_ptr[1], a.ldb, a.beta,
+                       (double*)t.dev_ptr[2], a.ldc);
+    return;
+  }
   rocblas_status s = rocblas_dgemm(
       blas_handle(g), rocblas_operation_none, rocblas_operation_none, a.m,
       a.n, a.k, &a.alpha, (const double*)t.dev_ptr[0], a.lda,

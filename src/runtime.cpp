@@ -137,6 +137,53 @@ void task_complete(Task* t) {
   tp->task_done();
 }
 
+// Periodic snapshot of the live runtime properties: one JSON object per
+// write, atomically replaced (write temp + rename) so readers never see a
+// torn snapshot. Fields mirror the stats table plus instantaneous queue
+// depth — the reference exposes the same through its shared-memory
+// properties dictionary consumed by the aggregator dashboards.
+void Context::live_stats_main(std::string path, int interval_ms) {
+  const std::string tmp = path + ".tmp";
+  const auto t0 = std::chrono::steady_clock::now();
+  while (!live_stop_.load(std::memory_order_acquire)) {
+    RuntimeCounters& c = counters();
+    double up = std::chrono::duration<double>(
+                    std::chrono::steady_clock::now() - t0).count();
+    FILE* f = fopen(tmp.c_str(), "w");
+    if (f) {
+      fprintf(f,
+              "{\"rank\": %d, \"world\": %d, \"workers\": %d, "
+              "\"uptime_s\": %.3f, \"ready_queue\": %zu, "
+              "\"tasks_cpu\": %lu, \"tasks_gpu\": %lu, "
+              "\"scheduled\": %lu, \"steals\": %lu, "
+              "\"comm_msgs\": %lu, \"comm_bytes\": %lu, "
+              "\"renames\": %lu",
+              rank_, world_, nworkers_, up, sched_->approx_pending(),
+              (unsigned long)c.tasks_executed_cpu.load(),
+              (unsigned long)c.tasks_executed_gpu.load(),
+              (unsigned long)c.tasks_scheduled.load(),
+              (unsigned long)c.steals.load(),
+              (unsigned long)c.comm_msgs.load(),
+              (unsigned long)c.comm_bytes.load(),
+              (unsigned long)c.renames.load());
+      if (gpu_)
+        fprintf(f,
+                ", \"gpu_tasks\": %lu, \"gpu_h2d\": %lu, "
+                "\"gpu_d2h\": %lu, \"gpu_evictions\": %lu",
+                (unsigned long)gpu_->stats.tasks.load(),
+                (unsigned long)gpu_->stats.bytes_h2d.load(),
+                (unsigned long)gpu_->stats.bytes_d2h.load(),
+                (unsigned long)gpu_->stats.evictions.load());
+      fprintf(f, "}\n");
+      fclose(f);
+      rename(tmp.c_str(), path.c_str());
+    }
+    for (int i = 0; i < interval_ms && !live_stop_.load(); i += 20)
+      std::this_thread::sleep_for(std::chrono::milliseconds(20));
+  }
+  remove(path.c_str());
+}
+
 void run_cpu_task(Task* t) {
   if (t->tc->kind == TaskKind::GPU && t->tp->context()->gpu()) {
     // blocking GPU chore routed through the CPU scheduler (on CPU-only
@@ -525,6 +572,14 @@ Context::Context(const Options& opt) : rank_(opt.rank), world_(opt.world) {
   if (gpu_) gpu_->publish_info(this);
 
   pins_modules_install();
+  {
+    std::string lp = param_str("live_stats", "");
+    if (!lp.empty()) {
+      int iv = (int)param_int("live_stats_interval_ms", 500);
+      live_thread_ = std::thread(&Context::live_stats_main, this,
+                                 lp + "." + std::to_string(rank_), iv);
+    }
+  }
   roctx_init();
   debug_history_init();
   std::string prof = param_str("profile_filename", "");
@@ -538,6 +593,8 @@ Context::Context(const Options& opt) : rank_(opt.rank), world_(opt.world) {
 }
 
 Context::~Context() {
+  live_stop_.store(true, std::memory_order_release);
+  if (live_thread_.joinable()) live_thread_.join();
   stop_.store(true, std::memory_order_release);
   sched_->wake_all();
   for (auto& w : workers_) w.join();

@@ -286,6 +286,14 @@ class Context {
   std::unique_ptr<CommEngine> comm_;
   mutable std::mutex info_mtx_;
   std::map<std::string, std::string> info_;
+  // Live stats publisher (properties-dictionary / aggregator_visu analog,
+  // dictionary.c + tools/aggregator_visu): PARSEC_MCA_live_stats=<path>
+  // snapshots the runtime counters to <path>.<rank> every
+  // live_stats_interval_ms; `python -m parsec_amd.tools.live_top <path>`
+  // is the live dashboard.
+  std::thread live_thread_;
+  std::atomic<bool> live_stop_{false};
+  void live_stats_main(std::string path, int interval_ms);
 };
 
 void run_cpu_task(Task* t);  // execute hook + complete

@@ -53,7 +53,7 @@ c = ctx.counters()
 assert c["tasks_executed_cpu"] >= 3, c
 assert c["tasks_scheduled"] >= 3, c
 print("COUNTERS_OK")
-del A, ctx
+del A, tp, ctx
 """
     r = subprocess.run([sys.executable, "-c", code], capture_output=True,
                        text=True, timeout=120)
@@ -76,7 +76,7 @@ pm.insert_spd_fill(tp, A, 1)
 pm.insert_potrf(tp, A)
 tp.wait()
 print("ROCTX_OK")
-del A, ctx
+del A, tp, ctx
 """
     r = subprocess.run([sys.executable, "-c", code], capture_output=True,
                        text=True, timeout=120)
@@ -110,7 +110,7 @@ pm.insert_potrf(tp2, A)
 tp2.wait()
 assert seen == n0, "callbacks fired after removal"
 print("PINS_OK")
-del A, ctx
+del A, tp, ctx
 """
     r = subprocess.run([sys.executable, "-c", code], capture_output=True,
                        text=True, timeout=120)
@@ -192,7 +192,7 @@ pm.insert_spd_fill(tp, A, 1)
 pm.insert_potrf(tp, A)
 tp.wait()
 ctx.barrier()
-del A, ctx
+del A, tp, ctx
 print("CT_OK", rank)
 """
     port = str(port_base(25))
@@ -250,3 +250,45 @@ def test_jacobi_replay_example_cpu():
                        capture_output=True, text=True, timeout=300)
     assert r.returncode == 0, r.stdout + r.stderr
     assert "fixed-point err" in r.stdout
+
+
+def test_live_stats_and_live_top(tmp_path):
+    """PARSEC_MCA_live_stats publishes periodic JSON snapshots (properties
+    dictionary / aggregator_visu analog) and live_top renders them."""
+    live = str(tmp_path / "live")
+    code = f"""
+import sys, time; sys.path.insert(0, {REPO!r})
+import parsec_amd as pm
+pm.param_set("live_stats", {live!r})
+pm.param_set("live_stats_interval_ms", "50")
+ctx = pm.Context(nworkers=2, rank=0, world=1, gpu=-2)
+A = pm.TiledMatrix(ctx, 512, 512, 64, 64, 1, 1)
+for _ in range(4):
+    tp = pm.Dtd(ctx); pm.insert_spd_fill(tp, A, 1); pm.insert_potrf(tp, A)
+    tp.wait()
+time.sleep(0.3)  # let at least one snapshot land after the work
+import json
+s = json.load(open({live!r} + ".0"))
+assert s["tasks_cpu"] + s["tasks_gpu"] > 50, s
+assert s["workers"] == 2 and s["rank"] == 0, s
+print("LIVE_OK")
+del A, tp, ctx
+import os
+assert not os.path.exists({live!r} + ".0"), "snapshot not removed at fini"
+print("CLEAN_OK")
+"""
+    r = subprocess.run([sys.executable, "-c", code], capture_output=True,
+                       text=True, timeout=120)
+    assert "LIVE_OK" in r.stdout and "CLEAN_OK" in r.stdout, \
+        r.stdout + r.stderr
+    # live_top --once renders a table from a snapshot file
+    (tmp_path / "live2.0").write_text(
+        '{"rank": 0, "world": 1, "workers": 2, "uptime_s": 1.5, '
+        '"ready_queue": 3, "tasks_cpu": 100, "tasks_gpu": 5, '
+        '"scheduled": 105, "steals": 2, "comm_msgs": 0, "comm_bytes": 0, '
+        '"renames": 0}\n')
+    r2 = subprocess.run([sys.executable, "-m", "parsec_amd.tools.live_top",
+                         str(tmp_path / "live2"), "--once"],
+                        capture_output=True, text=True, timeout=60, cwd=REPO)
+    assert r2.returncode == 0, r2.stdout + r2.stderr
+    assert "rank" in r2.stdout and "100" in r2.stdout, r2.stdout

@@ -492,3 +492,56 @@ def test_gpu_graph_capture_requires_gpu():
     with pytest.raises(RuntimeError, match="GPU"):
         tp.capture_begin()
     del tp, ctx
+
+
+def test_fuzz_oracle_under_new_schedulers():
+    """The random-DAG oracle fuzz under the round-2 scheduler modules
+    (pbq per-worker heaps + steal, ip inverse priority): scheduling order
+    must never change results."""
+    import subprocess, sys, os
+    repo = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+    for kind in ["pbq", "ip"]:
+        code = f"""
+import sys, random; sys.path.insert(0, {repo!r})
+import numpy as np
+import parsec_amd as pm
+pm.param_set("sched", {kind!r})
+ctx = pm.Context(nworkers=4, rank=0, world=1, gpu=-2)
+rng = random.Random(77)
+nb, NT = 32, 6
+mats = [pm.TiledMatrix(ctx, nb, nb, nb, nb, 1, 1) for _ in range(NT)]
+ref = [np.full((nb, nb), float(i + 1)) for i in range(NT)]
+tp = pm.Dtd(ctx)
+for i, M in enumerate(mats):
+    M.tile_numpy_set(0, 0, ref[i])
+def sc(x, a, b):
+    v = np.frombuffer(x, dtype=np.float64)
+    v *= a; v += b
+def ad(x, y):
+    np.frombuffer(y, dtype=np.float64)[:] += np.frombuffer(x, np.float64)
+for step in range(120):
+    op = rng.randrange(2)
+    i, j = rng.randrange(NT), rng.randrange(NT)
+    prio = rng.randrange(-50, 50)
+    if op == 0:
+        a, b = rng.uniform(0.5, 1.5), rng.uniform(-1, 1)
+        tp.insert_py(lambda x, _a=a, _b=b: sc(x, _a, _b),
+                     [(mats[i].tile(0, 0), pm.ACCESS_INOUT)],
+                     priority=prio, with_data=True)
+        ref[i] = ref[i] * a + b
+    elif i != j:
+        tp.insert_py(lambda x, y: ad(x, y),
+                     [(mats[i].tile(0, 0), pm.ACCESS_IN),
+                      (mats[j].tile(0, 0), pm.ACCESS_INOUT)],
+                     priority=prio, with_data=True)
+        ref[j] = ref[j] + ref[i]
+tp.wait()
+for i, M in enumerate(mats):
+    assert np.allclose(M.tile_numpy(0, 0), ref[i]), f"tile {{i}} diverged"
+print("FUZZ_OK")
+del mats, tp, ctx
+"""
+        r = subprocess.run([sys.executable, "-c", code], capture_output=True,
+                           text=True, timeout=300)
+        assert r.returncode == 0 and "FUZZ_OK" in r.stdout, \
+            f"sched={kind}: {r.stdout}{r.stderr}"

@@ -58,8 +58,9 @@ GpuEngine::GpuEngine(Context* ctx, int device) : ctx_(ctx), device_(device) {
     }
   }
   zone_.init(slab_bytes_);
-  PA_DEBUG(1, "GPU %d engine: %d exec streams, slab %.1f GB", device_,
-           nstreams, slab_bytes_ / 1e9);
+  detect_numa();
+  PA_DEBUG(1, "GPU %d engine: %d exec streams, slab %.1f GB, numa %d",
+           device_, nstreams, slab_bytes_ / 1e9, numa_node_);
   manager_ = std::thread([this] { manager_main(); });
 }
 
@@ -82,6 +83,43 @@ GpuEngine::~GpuEngine() {
   if (slab_) hipFree(slab_);
 }
 
+// Resolve the GPU's NUMA node and its CPU list via sysfs
+// (/sys/bus/pci/devices/<busid>/numa_node +
+// /sys/devices/system/node/node<k>/cpulist). The vpmap/hwloc analog a
+// one-node deployment actually needs: bind each rank's workers to the
+// socket its GPU hangs off (PARSEC_MCA_sched_bind=numa).
+void GpuEngine::detect_numa() {
+  char busid[64] = {0};
+  if (hipDeviceGetPCIBusId(busid, sizeof(busid), device_) != hipSuccess)
+    return;
+  for (char* p = busid; *p; p++) *p = (char)tolower(*p);
+  char path[128];
+  snprintf(path, sizeof(path), "/sys/bus/pci/devices/%s/numa_node", busid);
+  FILE* f = fopen(path, "r");
+  if (!f) return;
+  if (fscanf(f, "%d", &numa_node_) != 1) numa_node_ = -1;
+  fclose(f);
+  if (numa_node_ < 0) return;
+  snprintf(path, sizeof(path), "/sys/devices/system/node/node%d/cpulist",
+           numa_node_);
+  f = fopen(path, "r");
+  if (!f) return;
+  char list[512] = {0};
+  if (fgets(list, sizeof(list), f)) {
+    // "0-31,64-95" -> expanded cpu ids
+    char* save = nullptr;
+    for (char* tok = strtok_r(list, ",\n", &save); tok;
+         tok = strtok_r(nullptr, ",\n", &save)) {
+      int lo = 0, hi = 0;
+      if (sscanf(tok, "%d-%d", &lo, &hi) == 2)
+        for (int c = lo; c <= hi; c++) numa_cpus_.push_back(c);
+      else if (sscanf(tok, "%d", &lo) == 1)
+        numa_cpus_.push_back(lo);
+    }
+  }
+  fclose(f);
+}
+
 void GpuEngine::publish_info(Context* ctx) {
   hipDeviceProp_t prop{};
   if (hipGetDeviceProperties(&prop, device_) == hipSuccess) {
@@ -95,6 +133,8 @@ void GpuEngine::publish_info(Context* ctx) {
                 std::to_string((double)slab_bytes_ / (1 << 30)));
   ctx->info_set("device.exec_streams",
                 std::to_string(exec_streams_.size()));
+  ctx->info_set("device.numa_node", std::to_string(numa_node_));
+  ctx->info_set("device.numa_cpus", std::to_string(numa_cpus_.size()));
 }
 
 void GpuEngine::enqueue(Task* t) {

@@ -74,6 +74,10 @@ class GpuEngine {
   void copy_h2d(void* dst, const void* src, size_t bytes);
 
   int device() const { return device_; }
+  // GPU's NUMA locality from sysfs (vpmap/hwloc analog on one node):
+  // node id (-1 unknown) and that node's CPU list for worker binding.
+  int numa_node() const { return numa_node_; }
+  const std::vector<int>& numa_cpus() const { return numa_cpus_; }
   // publish device facts into the context info registry (info.c analog)
   void publish_info(Context* ctx);
   hipStream_t comm_stream() const { return comm_stream_; }
@@ -117,6 +121,9 @@ class GpuEngine {
 
   Context* ctx_;
   int device_;
+  int numa_node_ = -1;
+  std::vector<int> numa_cpus_;
+  void detect_numa();
   hipStream_t h2d_stream_{}, d2h_stream_{}, comm_stream_{};
   std::vector<hipStream_t> exec_streams_;
   int next_stream_ = 0;

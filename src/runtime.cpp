@@ -647,18 +647,29 @@ Context::~Context() {
 
 void Context::worker_main(int id) {
   tls_worker_id = id;
-  // Optional linear core binding (bindthread.c/vpmap analog). Default off:
-  // with one process per GPU the OS spreads ranks well, and naive binding
-  // would stack all ranks' worker 0 on core 0. sched_bind=1 binds at
-  // core = (local_rank * (nworkers+2) + id) % ncores.
-  if (param_int("sched_bind", 0)) {
+  // Optional core binding (bindthread.c/vpmap analog). Default off: with
+  // one process per GPU the OS spreads ranks reasonably. sched_bind=1:
+  // linear, core = (local_rank * (nworkers+2) + id) % ncores.
+  // sched_bind=numa: bind this rank's workers round-robin over the CPUs
+  // of the NUMA node its GPU hangs off (sysfs detection,
+  // GpuEngine::detect_numa) — the binding that actually matters on a
+  // dual-socket 8-GPU box; falls back to linear when detection fails.
+  std::string bind = param_str("sched_bind", "0");
+  if (bind != "0" && !bind.empty()) {
     const char* lr = getenv("LOCAL_RANK");
     int local = lr ? atoi(lr) : rank_;
-    long ncores = sysconf(_SC_NPROCESSORS_ONLN);
-    if (ncores > 0) {
+    int core = -1;
+    if (bind == "numa" && gpu_ && !gpu_->numa_cpus().empty()) {
+      const auto& cpus = gpu_->numa_cpus();
+      core = cpus[(size_t)(local * (nworkers_ + 2) + id) % cpus.size()];
+    } else {
+      long ncores = sysconf(_SC_NPROCESSORS_ONLN);
+      if (ncores > 0) core = (local * (nworkers_ + 2) + id) % (int)ncores;
+    }
+    if (core >= 0) {
       cpu_set_t set;
       CPU_ZERO(&set);
-      CPU_SET((local * (nworkers_ + 2) + id) % ncores, &set);
+      CPU_SET(core, &set);
       pthread_setaffinity_np(pthread_self(), sizeof(set), &set);
     }
   }

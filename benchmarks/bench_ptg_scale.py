@@ -26,6 +26,7 @@ k = 0 .. NT-1
 : A( k %% 64, 0 )
 
 RW X <- (k > 63) ? X Step(k - 64) : A(k %% 64, 0)
+     -> (k + 64 < NT) ? X Step(k + 64)
      -> A(k %% 64, 0)
 
 BODY
@@ -38,6 +39,8 @@ END
 def main():
     ap = argparse.ArgumentParser()
     ap.add_argument("--tasks", type=int, default=1000000)
+    ap.add_argument("--compact", action="store_true",
+                    help="never-materialized iteration (jdf2c analog)")
     args = ap.parse_args()
     import tempfile
     import json
@@ -57,13 +60,13 @@ def main():
         rss0 = resource.getrusage(resource.RUSAGE_SELF).ru_maxrss
         t0 = time.time()
         tp = pm.Dtd(ctx, "scale")
-        mod.build(ctx, tp, A=A, NT=args.tasks)
+        mod.build(ctx, tp, compact=args.compact, A=A, NT=args.tasks)
         t_build = time.time() - t0
         tp.wait()
         t_total = time.time() - t0
         rss1 = resource.getrusage(resource.RUSAGE_SELF).ru_maxrss
         print(json.dumps({
-            "metric": "PTG materialized insertion",
+            "metric": ("PTG compact iteration" if args.compact else "PTG materialized insertion"),
             "tasks": args.tasks,
             "build_s": round(t_build, 3),
             "total_s": round(t_total, 3),

@@ -659,6 +659,85 @@ def generate_cpp(jdf, name):
                 w("}")
         body_fns.append((cpu_fn, gpu_fn, gpu_flags))
 
+    # ---- compact (never-materialized) iteration helpers ----------------
+    # jdf2c compact-iteration analog: per class, (1) predecessor COUNT
+    # from the IN arrows, (2) successor ENUMERATION from the OUT arrows
+    # (the duals), (3) immediate single-instance insertion. Used by the
+    # ptg_build_compact_* entry + paptg::Compact (src/ptg_runtime.hpp).
+    def emit_task_dep_elements(c, direction, per_elem):
+        """Emit code per (guard-qualified, range-expanded) task-term dep
+        element of class c in the given arrow direction. per_elem(tcls,
+        args_exprs, qvar) -> code string; ranged args become loops with
+        loop vars substituted into args_exprs."""
+        out_lines = []
+        for f in c.flows:
+            for d in (d for d in f.deps if d.direction == direction):
+                cases = (((d.term, d.guard),) if d.else_term is None else
+                         ((d.term, d.guard), (d.else_term, f"!({d.guard})")))
+                for term, guard in cases:
+                    if term[0] != "task":
+                        continue
+                    _, fname, tcls, args = term
+                    loops = ""
+                    closes = ""
+                    exprs = []
+                    for ai, a in enumerate(args):
+                        rp = _range_parts(a)
+                        if rp:
+                            lo, hi = _cxx_expr(rp[0]), _cxx_expr(rp[1])
+                            st2 = _cxx_expr(rp[2]) if len(rp) > 2 else "1"
+                            loops += (f" for (long _ga{ai} = (long)({lo}); "
+                                      f"_ga{ai} <= (long)({hi}); "
+                                      f"_ga{ai} += (long)({st2})) {{")
+                            closes += " }"
+                            exprs.append(f"_ga{ai}")
+                        else:
+                            exprs.append(f"(long)({_cxx_expr(a)})")
+                    cond = f"if ({_cxx_expr(guard)}) " if guard else ""
+                    out_lines.append(
+                        f"  {cond}{{{loops} {per_elem(tcls, exprs)}{closes} }}")
+        return out_lines
+
+    for k, c in enumerate(jdf.classes):
+        w(f"long predcount_{c.name}(const long* _P) {{")
+        w(alias)
+        w(param_decls(c))
+        w("  long _n = 0;")
+        for ln in emit_task_dep_elements(
+                c, "<-", lambda tcls, exprs: "_n++;"):
+            w(ln)
+        w("  return _n;")
+        w("}")
+
+        def offer_elem(tcls, exprs):
+            tgt = jdf.classes[cls_index[tcls]]
+            sets = " ".join(f"_Q[{i}] = {e};" for i, e in enumerate(exprs))
+            return (f"long _Q[MAXP] = {{0}}; {sets} "
+                    f"((paptg::Compact*)_vcc)->offer("
+                    f"{cls_index[tcls]}, _Q, {len(tgt.params)});")
+
+        w(f"void succs_{c.name}(const long* _P, void* _vcc) {{")
+        w("  (void)_vcc;")
+        w(alias)
+        w(param_decls(c))
+        for ln in emit_task_dep_elements(c, "->", offer_elem):
+            w(ln)
+        w("}")
+
+        w(f"void insinst_{c.name}(const long* _P, void* _vcc) {{")
+        w("  auto* _cc = (paptg::Compact*)_vcc;")
+        w(alias)
+        w(param_decls(c))
+        dfl = data_flows(c)
+        w(f"  void* _datas[MAXF]; int _modes[MAXF];")
+        for fk, f in enumerate(dfl):
+            w(f"  _datas[{fk}] = binding_{c.name}(_P, {fk});")
+            w(f"  _modes[{fk}] = {f.mode};")
+        prio = f"(int)({_cxx_expr(c.priority)})" if c.priority else "0"
+        w(f"  _cc->do_insert({k}, _P, {len(c.params)}, {prio}, _datas, "
+          f"_modes, {len(dfl)});")
+        w("}")
+
     w("}  // namespace")
 
     # build entry
@@ -752,6 +831,68 @@ def generate_cpp(jdf, name):
         w("  }")
     w("  _g.run();")
     w("}")
+
+    # ---- compact entry: seed scan only, cascade does the rest ----
+    w(f'extern "C" void ptg_build_compact_{name}(void* _ctx, void* _dtd, '
+      "void** _colls, long* _scalars) {")
+    w("  g_glob._ctx = _ctx; g_glob._dtd = _dtd;")
+    for k, c in enumerate(colls):
+        w(f"  g_glob.{c} = _colls[{k}];")
+    vis = 0
+    for s, pr in scalars:
+        if pr.get("hidden") == "on":
+            continue
+        w(f"  g_glob.{s} = _scalars[{vis}];")
+        vis += 1
+    w(alias)
+    for s, pr in scalars:
+        if pr.get("hidden") == "on":
+            w(f"  g_glob.{s} = (long)({_cxx_expr(pr.get('default'))}); "
+              f"{s} = g_glob.{s};")
+    w("  if (pa_ctx_world(_ctx) != 1) {")
+    w('    fprintf(stderr, "[ptg] compact iteration is single-process only '
+      '(distributed PTG uses the materialized deterministic order)\\n");')
+    w("    abort();")
+    w("  }")
+    w("  paptg::new_tiles_reset(_dtd);")
+    w("  static std::vector<void*> _tcs; if (_tcs.empty()) {")
+    for k, c in enumerate(jdf.classes):
+        cpu_fn, gpu_fn, gpu_flags = body_fns[k]
+        w(f'    _tcs.push_back(pa_taskclass_new("{c.name}", '
+          f"{gpu_flags if gpu_fn != 'nullptr' else 0}, {cpu_fn}, {gpu_fn}));")
+    w("  }")
+    w("  auto* _cc = new paptg::Compact(_dtd, {")
+    for c in jdf.classes:
+        w(f"      {{predcount_{c.name}, succs_{c.name}, insinst_{c.name}}},")
+    w("  });")
+    w("  _cc->set_classes(_tcs);")
+    w("  pa_dtd_own_ptr(_dtd, _cc, [](void* p) { "
+      "delete (paptg::Compact*)p; });")
+    w("  long _total = 0;")
+    for k, c in enumerate(jdf.classes):
+        if c.partition is None:
+            raise JdfError(f"{c.name}: missing partitioning line ': coll(...)'")
+        w("  {")
+        indent = "  "
+        for (rname, lo, hi, step) in c.ranges:
+            st = _cxx_expr(step) if step else "1"
+            w(f"{indent}for (long {rname} = (long)({_cxx_expr(lo)}); "
+              f"{rname} <= (long)({_cxx_expr(hi)}); {rname} += (long)({st})) {{")
+            indent += "  "
+        for lname, lexpr in c.locals_:
+            w(f"{indent}long {lname} = (long)({_cxx_expr(lexpr)}); (void){lname};")
+        w(f"{indent}_total++;")
+        w(f"{indent}long _PP[MAXP] = {{0}};")
+        for pk, p in enumerate(c.params):
+            w(f"{indent}_PP[{pk}] = {p};")
+        w(f"{indent}if (predcount_{c.name}(_PP) == 0) _cc->seed({k}, _PP);")
+        for _ in c.ranges:
+            indent = indent[:-2]
+            w(f"{indent}}}")
+        w("  }")
+    w("  _cc->note_total(_total);")
+    w("  _cc->arm_check();")
+    w("}")
     return "\n".join(out)
 
 
@@ -761,13 +902,22 @@ class PtgModule:
         import ctypes
         self._lib = ctypes.CDLL(so_path)
         self._build = getattr(self._lib, f"ptg_build_{name}")
+        self._build_compact = getattr(self._lib, f"ptg_build_compact_{name}")
         self._jdf = jdf
         self.name = name
         self.so_path = so_path
 
-    def build(self, ctx, tp, **kwargs):
+    def build(self, ctx, tp, compact=False, **kwargs):
         """Enumerate + insert the taskpool's tasks. kwargs map JDF global
-        names to TiledMatrix collections / integer scalars."""
+        names to TiledMatrix collections / integer scalars.
+
+        compact=True (single-process pools): the jdf2c compact-iteration
+        analog — instances are never all materialized. The call inserts
+        only the SEED tasks (no task predecessors, found by one O(1)-memory
+        scan of the execution space); every other instance is created when
+        its last predecessor completes, discovered through the OUT arrows.
+        Requires IN/OUT arrows to be duals (the normal JDF contract; a
+        loud warning reports violations)."""
         import ctypes
         colls = [g for g, pr in self._jdf.globals_ if _is_coll(pr)]
         scalars = [g for g, pr in self._jdf.globals_
@@ -782,8 +932,9 @@ class PtgModule:
             if s not in kwargs:
                 raise JdfError(f"missing global argument {s!r}")
             sargs[k] = int(kwargs[s])
-        self._build(ctypes.c_void_p(ctx._handle), ctypes.c_void_p(tp._handle),
-                    cargs, sargs)
+        fn = self._build_compact if compact else self._build
+        fn(ctypes.c_void_p(ctx._handle), ctypes.c_void_p(tp._handle),
+           cargs, sargs)
 
 
 def compile_jdf(path, verbose=False):
@@ -794,7 +945,9 @@ def compile_jdf(path, verbose=False):
     jdf = parse_jdf(text)
     cpp = generate_cpp(jdf, name)
     os.makedirs(CACHE, exist_ok=True)
-    h = hashlib.sha256((cpp + "v1").encode()).hexdigest()[:16]
+    with open(os.path.join(REPO, "src", "ptg_runtime.hpp")) as f:
+        rt_hdr = f.read()  # included verbatim: a header change must rebuild
+    h = hashlib.sha256((cpp + rt_hdr + "v2").encode()).hexdigest()[:16]
     so = os.path.join(CACHE, f"{name}_{h}.so")
     if not os.path.exists(so):
         # atomic publish: concurrent ranks may compile the same JDF

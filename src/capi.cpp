@@ -10,6 +10,7 @@
 #include "data.hpp"
 #include "device_gpu.hpp"
 #include "dtd.hpp"
+#include "pins.hpp"
 #include "runtime.hpp"
 
 #define PA_EXPORT extern "C" __attribute__((visibility("default")))
@@ -109,6 +110,23 @@ PA_EXPORT void* pa_dtd_scratch(void* dtd, long bytes, int home_rank) {
   return d;
 }
 PA_EXPORT void pa_task_release(void* t) { ((Task*)t)->release(); }
+PA_EXPORT void* pa_task_taskpool(void* t) { return ((Task*)t)->tp; }
+
+// PINS COMPLETE hook + pool-completion callback + owned-resource handle:
+// the seam the compact (never-materialized) PTG iterator hangs off
+// (src/ptg_runtime.hpp Compact; jdf2c.c compact-iteration analog).
+PA_EXPORT long pa_pins_on_complete(void (*cb)(void*, void*), void* user) {
+  return Pins::inst().add(
+      [cb, user](PinsEv, const Task* t, int) { cb((void*)t, user); },
+      1u << (int)PinsEv::COMPLETE);
+}
+PA_EXPORT void pa_pins_off(long h) { Pins::inst().remove((int)h); }
+PA_EXPORT void pa_dtd_on_complete(void* dtd, void (*cb)(void*), void* user) {
+  ((Dtd*)dtd)->on_complete([cb, user] { cb(user); });
+}
+PA_EXPORT void pa_dtd_own_ptr(void* dtd, void* p, void (*deleter)(void*)) {
+  ((Dtd*)dtd)->own(std::shared_ptr<void>(p, deleter));
+}
 
 // ---- standalone C embedding surface (parsec_init/parsec_fini analog for
 // C programs linking _core.so directly; the reference is consumed as a C

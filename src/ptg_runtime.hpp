@@ -22,6 +22,7 @@
 #include <cstdlib>
 #include <cstring>
 #include <map>
+#include <mutex>
 #include <queue>
 #include <vector>
 
@@ -45,6 +46,11 @@ void pa_task_edge(void* pred, void* succ);
 void pa_task_retain(void* t);
 void pa_task_release(void* t);
 void* pa_dtd_scratch(void* dtd, long bytes, int home_rank);
+void* pa_task_taskpool(void* t);
+long pa_pins_on_complete(void (*cb)(void*, void*), void* user);
+void pa_pins_off(long h);
+void pa_dtd_on_complete(void* dtd, void (*cb)(void*), void* user);
+void pa_dtd_own_ptr(void* dtd, void* p, void (*deleter)(void*));
 }
 
 namespace paptg {
@@ -160,6 +166,135 @@ class Graph {
   std::vector<Inst> insts_;
   std::map<PKey, int> index_;
   std::vector<void*> classes_;
+};
+
+// ---- Compact (never-materialized) iteration ------------------------------
+// jdf2c's compact-iteration analog (jdf2c.c:3047+ startup generators +
+// iterate_successors): instances are NEVER all materialized. The build
+// entry scans the execution space once (O(1) memory) inserting only the
+// SEEDS (no task predecessors); every other instance is created when its
+// last predecessor completes, discovered through the OUT arrows
+// (successor enumeration is generated from the arrow duals). State held:
+// the FRONTIER only — instances with >=1 but not all predecessors done.
+// Single-process pools only: distributed PTG keeps the materialized
+// deterministic insertion order the SPMD channel protocol requires.
+class Compact {
+ public:
+  using PredFn = long (*)(const long*);
+  using SuccFn = void (*)(const long*, void*);
+  using InsFn = void (*)(const long*, void*);
+  struct ClassFns {
+    PredFn pred;
+    SuccFn succ;
+    InsFn ins;
+  };
+
+  Compact(void* dtd, std::vector<ClassFns> fns)
+      : dtd_(dtd), fns_(std::move(fns)) {
+    hook_ = pa_pins_on_complete(&Compact::on_complete_tramp, this);
+  }
+  ~Compact() { pa_pins_off(hook_); }
+
+  void set_classes(std::vector<void*> cls) { classes_ = std::move(cls); }
+  void seed(int cls, const long* P) { fns_[cls].ins(P, this); }
+  void note_total(long n) { total_ = n; }
+  long inserted() const { return inserted_; }
+
+  // called from the generated insert_inst functions
+  void do_insert(int cls, const long* P, int np, int prio, void** datas,
+                 const int* modes, int nflows) {
+    void* tc = classes_[cls];
+    long args[MAXP];
+    for (int i = 0; i < MAXP; i++) args[i] = i < np ? P[i] : 0;
+    void* t = pa_dtd_insert_begin(dtd_, tc, args, (int)sizeof(args), datas,
+                                  modes, nflows, prio, 0);
+    // world 1: every instance is local. Register BEFORE commit — the
+    // task cannot complete while the insertion guard is held, so the
+    // completion callback always finds it.
+    {
+      std::lock_guard<std::mutex> g(mu_);
+      PKey k{cls, {}};
+      for (int i = 0; i < np; i++) k.second[i] = P[i];
+      live_[t] = k;
+      inserted_++;
+    }
+    pa_dtd_insert_commit(dtd_, t);
+  }
+
+  // A predecessor of (cls, P-key) just completed. mu_ is LEAF-ONLY:
+  // never held across an insert or a completion callback — the window
+  // throttler can run tasks inline while holding the DTD insertion
+  // mutex, so any lock held across insertion would deadlock (ABBA with
+  // the completion path).
+  void offer(int cls, const long* Q, int np) {
+    PKey k{cls, {}};
+    for (int i = 0; i < np; i++) k.second[i] = Q[i];
+    bool do_ins = false;
+    {
+      std::lock_guard<std::mutex> g(mu_);
+      auto& st = state_[k];
+      if (st.need < 0) st.need = fns_[cls].pred(k.second.data());
+      if (st.need <= 0) {
+        // dual-inconsistent arrows: an OUT arrow targets an instance
+        // whose IN arrows do not list the producer. The seed scan
+        // already inserted it; inserting again would corrupt the DAG.
+        fprintf(stderr,
+                "[ptg compact] WARNING: OUT arrow targets a zero-pred "
+                "instance of class %d — IN/OUT arrows are not duals\n",
+                cls);
+        state_.erase(k);
+        return;
+      }
+      if (++st.got >= st.need) {
+        state_.erase(k);
+        do_ins = true;
+      }
+    }
+    if (do_ins) fns_[cls].ins(k.second.data(), this);
+  }
+
+  // loud underrun report if the arrow duals dropped instances
+  void arm_check() {
+    pa_dtd_on_complete(dtd_, &Compact::check_tramp, this);
+  }
+
+ private:
+  static void on_complete_tramp(void* task, void* user) {
+    auto* cc = (Compact*)user;
+    if (pa_task_taskpool(task) != cc->dtd_) return;
+    PKey k;
+    {
+      std::lock_guard<std::mutex> g(cc->mu_);
+      auto it = cc->live_.find(task);
+      if (it == cc->live_.end()) return;  // internal (comm/reclaim) task
+      k = it->second;
+      cc->live_.erase(it);
+    }
+    cc->fns_[k.first].succ(k.second.data(), cc);
+  }
+  static void check_tramp(void* user) {
+    auto* cc = (Compact*)user;
+    if (cc->total_ >= 0 && cc->inserted_ != cc->total_)
+      fprintf(stderr,
+              "[ptg compact] WARNING: %ld of %ld instances ran — OUT "
+              "arrows are not duals of the IN arrows (some instances were "
+              "never activated)\n", (long)cc->inserted_,
+              (long)cc->total_);
+  }
+
+  struct Pending {
+    long need = -1;
+    long got = 0;
+  };
+  void* dtd_;
+  std::vector<ClassFns> fns_;
+  std::vector<void*> classes_;
+  std::mutex mu_;
+  std::map<void*, PKey> live_;   // inserted, not yet completed
+  std::map<PKey, Pending> state_;  // the frontier
+  long hook_ = -1;
+  long total_ = -1;
+  long inserted_ = 0;
 };
 
 // ---- NEW-tile registry (JDF `<- NEW [size=...]`): one scratch datum per

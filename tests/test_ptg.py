@@ -5,6 +5,8 @@ Mirrors the reference's tests/dsl/ptg tree (chains, CTL ordering, guards/
 ternaries, the must-fail-to-compile compiler tests).
 """
 import os
+import subprocess
+import sys
 import struct
 
 import numpy as np
@@ -593,3 +595,68 @@ END
     tp.wait()
     (v,) = struct.unpack("<q", A.tile_bytes(0, 0))
     assert v == 4, v
+
+
+def test_ptg_compact_iteration_vs_materialized():
+    """Compact (never-materialized) iteration — jdf2c.c:3047+ analog:
+    seeds from an O(1)-memory scan, the rest instantiated when their last
+    predecessor completes via the OUT-arrow duals. Must equal the
+    materialized build bit-for-bit."""
+    code = f"""
+import sys; sys.path.insert(0, {REPO!r})
+import numpy as np
+import parsec_amd as pm
+from parsec_amd.ptg import compile_jdf
+import os
+import subprocess
+import sys
+ctx = pm.Context(nworkers=4, rank=0, world=1, gpu=-2)
+A = pm.TiledMatrix(ctx, 640, 640, 64, 64, 1, 1)
+B = pm.TiledMatrix(ctx, 640, 640, 64, 64, 1, 1)
+tp0 = pm.Dtd(ctx)
+pm.insert_spd_fill(tp0, A, 42); pm.insert_spd_fill(tp0, B, 42); tp0.wait()
+mod = compile_jdf(os.path.join({REPO!r}, "examples", "cholesky.jdf"))
+tp1 = pm.Dtd(ctx); mod.build(ctx, tp1, descA=A, NT=A.mt, NB=A.nb); tp1.wait()
+tp2 = pm.Dtd(ctx)
+mod.build(ctx, tp2, compact=True, descA=B, NT=B.mt, NB=B.nb); tp2.wait()
+for i in range(A.mt):
+    for j in range(i + 1):
+        assert np.allclose(A.tile_numpy(i, j), B.tile_numpy(i, j)), (i, j)
+print("COMPACT_OK")
+del A, B, tp0, tp1, tp2, ctx
+"""
+    r = subprocess.run([sys.executable, "-c", code], capture_output=True,
+                       text=True, timeout=300)
+    assert r.returncode == 0 and "COMPACT_OK" in r.stdout, \
+        r.stdout + r.stderr
+    assert "WARNING" not in r.stderr, r.stderr
+
+
+def test_ptg_compact_rejects_multiprocess():
+    """Compact iteration is world-1 only (distributed PTG keeps the
+    materialized deterministic insertion order)."""
+    from conftest import port_base
+    code = f"""
+import os, sys; sys.path.insert(0, {REPO!r})
+import parsec_amd as pm
+from parsec_amd.ptg import compile_jdf
+rank = int(os.environ["RANK"])
+pm.param_set("comm_base_port", os.environ["PORT"])
+ctx = pm.Context(nworkers=2, rank=rank, world=2, comm="tcp", gpu=-2)
+A = pm.TiledMatrix(ctx, 256, 256, 64, 64, 2, 1)
+tp = pm.Dtd(ctx)
+mod = compile_jdf(os.path.join({REPO!r}, "examples", "cholesky.jdf"))
+mod.build(ctx, tp, compact=True, descA=A, NT=A.mt, NB=A.nb)
+"""
+    import os as _os
+    port = str(port_base(31))
+    procs = []
+    for rk in range(2):
+        env = dict(_os.environ)
+        env.update(RANK=str(rk), WORLD_SIZE="2", PORT=port)
+        procs.append(subprocess.Popen([sys.executable, "-c", code], env=env,
+                                      stdout=subprocess.PIPE,
+                                      stderr=subprocess.STDOUT))
+    outs = [p.communicate(timeout=120)[0] for p in procs]
+    assert any(p.returncode != 0 for p in procs)
+    assert any(b"single-process only" in o for o in outs), outs

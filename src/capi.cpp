@@ -71,6 +71,11 @@ PA_EXPORT void* pa_task_host_ptr(void* t, int flow) {
   if (!d) return nullptr;
   // WRITE-only flows produce the content: hand out the buffer untouched.
   if (!(task->flows[flow].mode & ACCESS_IN)) return d->ensure_host();
+  if (!d->host_valid && !d->dev_valid)
+    fatal("task %s seq=%lu flow %d: input has no valid copy (key=%ld "
+          "ver=%lu) — insertion preceded its producer?",
+          task->tc->name.c_str(), (unsigned long)task->seq, flow,
+          (long)d->key, (unsigned long)d->version);
   return d->pull_to_host();
 }
 

@@ -660,3 +660,40 @@ mod.build(ctx, tp, compact=True, descA=A, NT=A.mt, NB=A.nb)
     outs = [p.communicate(timeout=120)[0] for p in procs]
     assert any(p.returncode != 0 for p in procs)
     assert any(b"single-process only" in o for o in outs), outs
+
+
+def test_ptg_compact_lu():
+    """Compact iteration on a second app (lu.jdf — its arrows already
+    carry the full duals): equals the materialized build."""
+    code = f"""
+import sys; sys.path.insert(0, {REPO!r})
+import numpy as np
+import parsec_amd as pm
+from parsec_amd.ptg import compile_jdf
+import os
+ctx = pm.Context(nworkers=4, rank=0, world=1, gpu=-2)
+A = pm.TiledMatrix(ctx, 384, 384, 64, 64, 1, 1)
+B = pm.TiledMatrix(ctx, 384, 384, 64, 64, 1, 1)
+tp0 = pm.Dtd(ctx)
+pm.insert_full_fill(tp0, A, 7); pm.insert_full_fill(tp0, B, 7)
+pm.insert_apply_scale(tp0, A, 0.01, 0); pm.insert_apply_scale(tp0, B, 0.01, 0)
+tp0.wait()
+for i in range(A.mt):
+    for M in (A, B):
+        t = M.tile_numpy(i, i); t += np.eye(64) * 100.0
+        M.tile_numpy_set(i, i, t)
+mod = compile_jdf(os.path.join({REPO!r}, "examples", "lu.jdf"))
+tp1 = pm.Dtd(ctx); mod.build(ctx, tp1, descA=A, NT=A.mt, NB=A.nb); tp1.wait()
+tp2 = pm.Dtd(ctx)
+mod.build(ctx, tp2, compact=True, descA=B, NT=B.mt, NB=B.nb); tp2.wait()
+for i in range(A.mt):
+    for j in range(A.nt):
+        assert np.allclose(A.tile_numpy(i, j), B.tile_numpy(i, j)), (i, j)
+print("LU_COMPACT_OK")
+del A, B, tp0, tp1, tp2, ctx
+"""
+    r = subprocess.run([sys.executable, "-c", code], capture_output=True,
+                       text=True, timeout=300)
+    assert r.returncode == 0 and "LU_COMPACT_OK" in r.stdout, \
+        r.stdout + r.stderr
+    assert "WARNING" not in r.stderr, r.stderr

@@ -697,3 +697,47 @@ del A, B, tp0, tp1, tp2, ctx
     assert r.returncode == 0 and "LU_COMPACT_OK" in r.stdout, \
         r.stdout + r.stderr
     assert "WARNING" not in r.stderr, r.stderr
+
+
+def test_ptg_compact_warns_on_missing_duals(tmp_path):
+    """A JDF whose OUT arrows don't cover the IN arrows (missing duals)
+    must complete what it can and WARN loudly about never-activated
+    instances instead of silently dropping them."""
+    jdf = """
+A  [ type="parsec_data_collection_t*" ]
+NT [ type="int" ]
+
+Step(k)
+
+k = 0 .. NT-1
+
+: A( 0, 0 )
+
+RW X <- (k == 0) ? A( 0, 0 ) : X Step(k - 1)
+
+BODY
+{
+}
+END
+"""
+    p = tmp_path / "nodual.jdf"
+    p.write_text(jdf)
+    code = f"""
+import sys; sys.path.insert(0, {REPO!r})
+import struct
+import parsec_amd as pm
+from parsec_amd.ptg import compile_jdf
+ctx = pm.Context(nworkers=2, rank=0, world=1, gpu=-2)
+A = pm.TiledMatrix(ctx, 1, 1, 1, 1, 1, 1)
+A.tile_bytes_set(0, 0, struct.pack("<q", 0))
+mod = compile_jdf({str(p)!r})
+tp = pm.Dtd(ctx)
+mod.build(ctx, tp, compact=True, A=A, NT=8)
+tp.wait()
+print("DRAINED")
+del A, tp, ctx
+"""
+    r = subprocess.run([sys.executable, "-c", code], capture_output=True,
+                       text=True, timeout=120)
+    assert r.returncode == 0 and "DRAINED" in r.stdout, r.stdout + r.stderr
+    assert "not duals" in r.stderr, r.stderr

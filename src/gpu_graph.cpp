@@ -58,7 +58,7 @@ GpuGraph::GpuGraph(GpuEngine* eng, GpuGraphRecorder&& rec) : eng_(eng) {
   // Cross-stream ordering via WAVEFRONT LEVELS with a fork/join barrier
   // between consecutive levels. Arbitrary per-edge event meshes crash
   // hipStreamEndCapture on ROCm 7.2 at >2 streams (isolated with
-  // build/capture_repro probes: fork/join and short chains are fine, the
+  // tools/probes programs: fork/join and short chains are fine, the
   // general mesh is not) — the barrier pattern is the one ROCm handles,
   // costs ~(streams+1) events per level, and keeps all within-level
   // parallelism: tasks at equal depth run concurrently across streams,

@@ -64,6 +64,8 @@ void insert_apply_scale(Dtd& tp, TiledMatrix& A, double alpha, double beta);
 // Tiled C = A*B, fp64 NN (idempotent DAG: k==0 overwrites C — the hipGraph
 // capture/replay demo workload, and a plain library-GEMM DAG generally).
 void insert_gemm_fp64(Dtd& tp, TiledMatrix& A, TiledMatrix& B, TiledMatrix& C);
+// data_advise analog: prefetch one tile onto the device (no-op body).
+void insert_advise_prefetch(Dtd& tp, Data* d);
 void insert_reduce_sum(Dtd& tp, TiledMatrix& A, TiledMatrix& R);
 void insert_stencil_1d(Dtd& tp, TiledMatrix& Src, TiledMatrix& Dst);
 void insert_panel_fill(Dtd& tp, TiledMatrix& A, uint32_t seed);

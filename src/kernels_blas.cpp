@@ -996,6 +996,26 @@ void insert_gemm_fp64(Dtd& tp, TiledMatrix& A, TiledMatrix& B,
       }
 }
 
+// --------------------------------------------------------------- advise
+// parsec_advise_data_on_device analog (device.h data_advise vtable row):
+// a no-op GPU task with one READ flow — the engine's normal stage-in
+// pulls the tile onto the device ahead of its first real consumer, on
+// the h2d stream, overlapped with whatever is executing. On CPU-only
+// contexts it degenerates to a no-op CPU task.
+static void cpu_advise(Task&) {}
+static void gpu_advise(Task&, GpuTaskCtx&) {}
+TaskClass& tc_advise() {
+  static TaskClass tc =
+      make_tc("advise_prefetch", TaskKind::GPU, cpu_advise, gpu_advise, 71);
+  return tc;
+}
+
+void insert_advise_prefetch(Dtd& tp, Data* d) {
+  Dtd::FlowSpec f[] = {{d, ACCESS_IN}};
+  TileArgs a{};
+  tp.insert(&tc_advise(), &a, sizeof(a), f, 1, 0, -1);
+}
+
 // Pre-create the per-stream rocBLAS handle and give it a fixed device
 // workspace so no allocation can happen inside a hipStream capture
 // (gpu_graph.cpp calls this for each capture stream before BeginCapture;

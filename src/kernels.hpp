@@ -66,6 +66,9 @@ void insert_apply_scale(Dtd& tp, TiledMatrix& A, double alpha, double beta);
 void insert_gemm_fp64(Dtd& tp, TiledMatrix& A, TiledMatrix& B, TiledMatrix& C);
 // data_advise analog: prefetch one tile onto the device (no-op body).
 void insert_advise_prefetch(Dtd& tp, Data* d);
+// Cholesky solve / factor+solve (dplasma dpotrs/dposv analogs).
+void insert_potrs(Dtd& tp, TiledMatrix& A, TiledMatrix& B);
+void insert_posv(Dtd& tp, TiledMatrix& A, TiledMatrix& B);
 void insert_reduce_sum(Dtd& tp, TiledMatrix& A, TiledMatrix& R);
 void insert_stencil_1d(Dtd& tp, TiledMatrix& Src, TiledMatrix& Dst);
 void insert_panel_fill(Dtd& tp, TiledMatrix& A, uint32_t seed);

@@ -917,6 +917,7 @@ void insert_potrf(Dtd& tp, TiledMatrix& A) {
 struct GemmNNArgs {
   int m, n, k, lda, ldb, ldc;
   double alpha, beta;
+  int transA = 0;  // 1: C = alpha*A^T*B + beta*C
 };
 
 static void cpu_gemm_nn(Task& t) {
@@ -930,7 +931,8 @@ static void cpu_gemm_nn(Task& t) {
     for (int i = 0; i < a.m; i++) {
       double s = 0;
       for (int p = 0; p < a.k; p++)
-        s += A[(size_t)p * a.lda + i] * B[(size_t)j * a.ldb + p];
+        s += (a.transA ? A[(size_t)i * a.lda + p] : A[(size_t)p * a.lda + i]) *
+             B[(size_t)j * a.ldb + p];
       double c0 = a.beta == 0.0 ? 0.0 : a.beta * C[(size_t)j * a.ldc + i];
       C[(size_t)j * a.ldc + i] = c0 + a.alpha * s;
     }
@@ -954,7 +956,7 @@ __global__ void k_gemm_nn_thin(int m, int n, int k, double alpha,
 
 static void gpu_gemm_nn(Task& t, GpuTaskCtx& g) {
   const GemmNNArgs& a = t.arg<GemmNNArgs>();
-  if (a.n <= 4) {
+  if (a.n <= 4 && !a.transA) {
     hipLaunchKernelGGL(k_gemm_nn_thin, dim3((a.m + 63) / 64, a.n), dim3(64),
                        0, g.stream, a.m, a.n, a.k, a.alpha,
                        (const double*)t.dev_ptr[0], a.lda,
@@ -963,10 +965,11 @@ static void gpu_gemm_nn(Task& t, GpuTaskCtx& g) {
     return;
   }
   rocblas_status s = rocblas_dgemm(
-      blas_handle(g), rocblas_operation_none, rocblas_operation_none, a.m,
-      a.n, a.k, &a.alpha, (const double*)t.dev_ptr[0], a.lda,
-      (const double*)t.dev_ptr[1], a.ldb, &a.beta, (double*)t.dev_ptr[2],
-      a.ldc);
+      blas_handle(g),
+      a.transA ? rocblas_operation_transpose : rocblas_operation_none,
+      rocblas_operation_none, a.m, a.n, a.k, &a.alpha,
+      (const double*)t.dev_ptr[0], a.lda, (const double*)t.dev_ptr[1],
+      a.ldb, &a.beta, (double*)t.dev_ptr[2], a.ldc);
   PA_CHECK(s == rocblas_status_success, "rocblas_dgemm NN failed: %d", (int)s);
 }
 
@@ -1014,6 +1017,101 @@ void insert_advise_prefetch(Dtd& tp, Data* d) {
   Dtd::FlowSpec f[] = {{d, ACCESS_IN}};
   TileArgs a{};
   tp.insert(&tc_advise(), &a, sizeof(a), f, 1, 0, -1);
+}
+
+// ----------------------------------------------------------- POTRS / POSV
+// Cholesky SOLVE (dplasma dpotrs/dposv analog): after insert_potrf left
+// A's lower tiles holding L, solve A X = B for a block of right-hand
+// sides: forward L Y = B (tile forward substitution), then backward
+// L^T X = Y. Diagonal solves are rocblas_dtrsm (left, lower, N/T);
+// off-diagonal updates reuse the gemm_nn chore (alpha=-1, beta=1, with
+// transA for the backward sweep).
+struct TrsmSolveArgs {
+  int m, n, lda, ldb;
+  int trans;  // 0: solve L Z = B; 1: solve L^T Z = B
+};
+
+static void cpu_trsm_solve(Task& t) {
+  const TrsmSolveArgs& a = t.arg<TrsmSolveArgs>();
+  const double* L = (const double*)t.flows[0].data->pull_to_host();
+  double* B = (double*)t.flows[1].data->pull_to_host();
+  for (int j = 0; j < a.n; j++) {
+    double* b = B + (size_t)j * a.ldb;
+    if (!a.trans) {
+      for (int i = 0; i < a.m; i++) {
+        double s = b[i];
+        for (int p = 0; p < i; p++) s -= L[(size_t)p * a.lda + i] * b[p];
+        b[i] = s / L[(size_t)i * a.lda + i];
+      }
+    } else {
+      for (int i = a.m - 1; i >= 0; i--) {
+        double s = b[i];
+        for (int p = i + 1; p < a.m; p++)
+          s -= L[(size_t)i * a.lda + p] * b[p];
+        b[i] = s / L[(size_t)i * a.lda + i];
+      }
+    }
+  }
+  t.flows[1].data->written_on(false);
+}
+
+static void gpu_trsm_solve(Task& t, GpuTaskCtx& g) {
+  const TrsmSolveArgs& a = t.arg<TrsmSolveArgs>();
+  const double one = 1.0;
+  rocblas_status s = rocblas_dtrsm(
+      blas_handle(g), rocblas_side_left, rocblas_fill_lower,
+      a.trans ? rocblas_operation_transpose : rocblas_operation_none,
+      rocblas_diagonal_non_unit, a.m, a.n, &one,
+      (const double*)t.dev_ptr[0], a.lda, (double*)t.dev_ptr[1], a.ldb);
+  PA_CHECK(s == rocblas_status_success, "rocblas_dtrsm solve failed: %d",
+           (int)s);
+}
+
+TaskClass& tc_trsm_solve() {
+  static TaskClass tc = make_tc("trsm_solve", TaskKind::GPU, cpu_trsm_solve,
+                                gpu_trsm_solve, 72);
+  return tc;
+}
+
+void insert_potrs(Dtd& tp, TiledMatrix& A, TiledMatrix& B) {
+  PA_CHECK(A.mt() == A.nt() && A.mt() == B.mt(),
+           "insert_potrs: A must be square with B.mt == A.mt");
+  const int mt = A.mt();
+  auto trsm = [&](int k, int j, int trans) {
+    TrsmSolveArgs a{A.tile_rows(k), B.tile_cols(j), A.mb(), B.mb(), trans};
+    Dtd::FlowSpec f[] = {{A.tile(k, k), ACCESS_IN},
+                         {B.tile(k, j), ACCESS_INOUT}};
+    tp.insert(&tc_trsm_solve(), &a, sizeof(a), f, 2, (1 << 20),
+              B.rank_of(k, j));
+  };
+  auto update = [&](int i, int k, int j, int trans) {
+    // forward: B[i] -= L[i,k]   * B[k]   (i > k)
+    // backward: B[i] -= L[k,i]^T * B[k]  (i < k; lower tile (k,i))
+    GemmNNArgs a{B.tile_rows(i), B.tile_cols(j), A.tile_rows(k),
+                 A.mb(),         B.mb(),         B.mb(),
+                 -1.0,           1.0};
+    a.transA = trans;
+    Data* l = trans ? A.tile(k, i) : A.tile(i, k);
+    Dtd::FlowSpec f[] = {{l, ACCESS_IN},
+                         {B.tile(k, j), ACCESS_IN},
+                         {B.tile(i, j), ACCESS_INOUT}};
+    tp.insert(&tc_gemm_nn(), &a, sizeof(a), f, 3, 0, B.rank_of(i, j));
+  };
+  for (int j = 0; j < B.nt(); j++) {
+    for (int k = 0; k < mt; k++) {  // forward: L Y = B
+      trsm(k, j, 0);
+      for (int i = k + 1; i < mt; i++) update(i, k, j, 0);
+    }
+    for (int k = mt - 1; k >= 0; k--) {  // backward: L^T X = Y
+      trsm(k, j, 1);
+      for (int i = k - 1; i >= 0; i--) update(i, k, j, 1);
+    }
+  }
+}
+
+void insert_posv(Dtd& tp, TiledMatrix& A, TiledMatrix& B) {
+  insert_potrf(tp, A);
+  insert_potrs(tp, A, B);
 }
 
 // Pre-create the per-stream rocBLAS handle and give it a fixed device

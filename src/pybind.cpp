@@ -450,6 +450,10 @@ PYBIND11_MODULE(_core, m) {
   m.def("insert_subtile_insert", &insert_subtile_insert, py::arg("tp"),
         py::arg("S"), py::arg("A"), py::arg("tm"), py::arg("tn"),
         py::call_guard<py::gil_scoped_release>());
+  m.def("insert_potrs", &insert_potrs, py::arg("tp"), py::arg("A"),
+        py::arg("B"), py::call_guard<py::gil_scoped_release>());
+  m.def("insert_posv", &insert_posv, py::arg("tp"), py::arg("A"),
+        py::arg("B"), py::call_guard<py::gil_scoped_release>());
   m.def("insert_advise_prefetch", &insert_advise_prefetch, py::arg("tp"),
         py::arg("tile"), py::call_guard<py::gil_scoped_release>());
   m.def("insert_gemm_fp64", &insert_gemm_fp64, py::arg("tp"), py::arg("A"),

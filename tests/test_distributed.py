@@ -608,3 +608,77 @@ del A, ctx
     for pr in procs:
         o, _ = pr.communicate(timeout=180)
         assert pr.returncode == 0 and b"TDYN_OK" in o, o.decode()
+
+
+def test_distributed_posv(tmp_path):
+    """World-2 Cholesky factor+solve (insert_posv) over the TCP engine:
+    the distributed solution matches numpy on the assembled system."""
+    world, n, nb, nrhs = 2, 256, 64, 64
+    port = _next_port[0]
+    _next_port[0] += world + 2
+    code = r"""
+import os, sys
+import numpy as np
+sys.path.insert(0, os.environ["PARSEC_REPO"])
+import parsec_amd as pm
+rank = int(os.environ["RANK"])
+pm.param_set("comm_base_port", os.environ["PORT"])
+ctx = pm.Context(nworkers=2, rank=rank, world=2, comm="tcp", gpu=-2)
+n, nb, nrhs = 256, 64, 64
+A = pm.TiledMatrix(ctx, n, n, nb, nb, 2, 1, sym=True)
+B = pm.TiledMatrix(ctx, n, nrhs, nb, nb, 2, 1)
+tp = pm.Dtd(ctx)
+pm.insert_spd_fill(tp, A, 11)
+pm.insert_full_fill(tp, B, 5)
+tp.wait()
+pre = {}
+for i in range(A.mt):
+    for j in range(i + 1):
+        if A.is_local(i, j):
+            pre[f"a_{i}_{j}"] = A.tile_numpy(i, j)
+    if B.is_local(i, 0):
+        pre[f"b_{i}_0"] = B.tile_numpy(i, 0)
+tp2 = pm.Dtd(ctx)
+pm.insert_posv(tp2, A, B)
+tp2.wait()
+tp2.flush_all(B)
+post = {}
+for i in range(B.mt):
+    if B.is_local(i, 0):
+        post[f"x_{i}_0"] = B.tile_numpy(i, 0)
+np.savez(os.path.join(os.environ["OUT"], f"posv{rank}.npz"), **pre, **post)
+ctx.barrier()
+del A, B, tp, tp2, ctx
+print("POSV_RANK_OK", rank)
+"""
+    procs = []
+    for r in range(world):
+        env = dict(os.environ)
+        env.update(RANK=str(r), WORLD_SIZE=str(world), PORT=str(port),
+                   OUT=str(tmp_path),
+                   PARSEC_REPO=os.path.dirname(HERE))
+        procs.append(subprocess.Popen([sys.executable, "-c", code], env=env,
+                                      stdout=subprocess.PIPE,
+                                      stderr=subprocess.STDOUT))
+    for pr in procs:
+        out, _ = pr.communicate(timeout=180)
+        assert pr.returncode == 0 and b"POSV_RANK_OK" in out, out.decode()
+    Af = np.zeros((n, n))
+    Bf = np.zeros((n, nrhs))
+    X = np.zeros((n, nrhs))
+    for r in range(world):
+        z = np.load(os.path.join(tmp_path, f"posv{r}.npz"))
+        for key in z.files:
+            kind, i, j = key.split("_")
+            i, j = int(i), int(j)
+            if kind == "a":
+                Af[i * nb:(i + 1) * nb, j * nb:(j + 1) * nb] = z[key]
+            elif kind == "b":
+                Bf[i * nb:(i + 1) * nb, :] = z[key]
+            else:
+                X[i * nb:(i + 1) * nb, :] = z[key]
+    Lo = np.tril(Af)
+    Af = Lo + np.tril(Lo, -1).T
+    ref = np.linalg.solve(Af, Bf)
+    err = np.abs(X - ref).max() / np.abs(ref).max()
+    assert err < 1e-11, f"distributed posv rel err {err}"

@@ -406,3 +406,30 @@ def test_jacobi_replay_example_gpu():
                        capture_output=True, text=True, timeout=300)
     assert r.returncode == 0, r.stdout + r.stderr
     assert "hipGraph replay" in r.stdout, r.stdout
+
+
+def test_gpu_posv_numerics(gctx):
+    """insert_posv on the GPU engine (rocBLAS dtrsm solves + gemm updates)
+    vs numpy.linalg.solve."""
+    n, nb, nrhs = 1024, 256, 256
+    A = pm.TiledMatrix(gctx, n, n, nb, nb, 1, 1, sym=True)
+    B = pm.TiledMatrix(gctx, n, nrhs, nb, nb, 1, 1)
+    tp = pm.Dtd(gctx)
+    pm.insert_spd_fill(tp, A, 11)
+    pm.insert_full_fill(tp, B, 5)
+    tp.wait()
+    Lo = np.zeros((n, n))
+    for i in range(A.mt):
+        for j in range(i + 1):
+            Lo[i * nb:(i + 1) * nb, j * nb:(j + 1) * nb] = A.tile_numpy(i, j)
+    Lo = np.tril(Lo)
+    Af = Lo + np.tril(Lo, -1).T
+    Bf = full_matrix(B, nb, nb)[:, :nrhs]
+    tp2 = pm.Dtd(gctx)
+    pm.insert_posv(tp2, A, B)
+    tp2.wait()
+    X = full_matrix(B, nb, nb)[:, :nrhs]
+    ref = np.linalg.solve(Af, Bf)
+    err = abs(X - ref).max() / abs(ref).max()
+    assert err < 1e-10, err
+    del A, B, tp, tp2

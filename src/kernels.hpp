@@ -69,6 +69,9 @@ void insert_advise_prefetch(Dtd& tp, Data* d);
 // Cholesky solve / factor+solve (dplasma dpotrs/dposv analogs).
 void insert_potrs(Dtd& tp, TiledMatrix& A, TiledMatrix& B);
 void insert_posv(Dtd& tp, TiledMatrix& A, TiledMatrix& B);
+// LU solve / factor+solve, no pivoting (dgetrs/dgesv nopiv analogs).
+void insert_getrs_nopiv(Dtd& tp, TiledMatrix& A, TiledMatrix& B);
+void insert_gesv_nopiv(Dtd& tp, TiledMatrix& A, TiledMatrix& B);
 void insert_reduce_sum(Dtd& tp, TiledMatrix& A, TiledMatrix& R);
 void insert_stencil_1d(Dtd& tp, TiledMatrix& Src, TiledMatrix& Dst);
 void insert_panel_fill(Dtd& tp, TiledMatrix& A, uint32_t seed);

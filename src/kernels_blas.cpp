@@ -1028,27 +1028,33 @@ void insert_advise_prefetch(Dtd& tp, Data* d) {
 // transA for the backward sweep).
 struct TrsmSolveArgs {
   int m, n, lda, ldb;
-  int trans;  // 0: solve L Z = B; 1: solve L^T Z = B
+  int trans;  // 0: solve T Z = B; 1: solve T^T Z = B
+  int upper = 0;  // triangle of the diagonal tile holding the factor
+  int unit = 0;   // unit diagonal (LU's L)
 };
 
 static void cpu_trsm_solve(Task& t) {
   const TrsmSolveArgs& a = t.arg<TrsmSolveArgs>();
   const double* L = (const double*)t.flows[0].data->pull_to_host();
   double* B = (double*)t.flows[1].data->pull_to_host();
+  // forward order for {lower, N} and {upper, T}; backward otherwise
+  const bool fwd = (a.upper == 0) != (a.trans != 0);
+  auto elem = [&](int i, int p) {
+    return a.trans ? L[(size_t)i * a.lda + p] : L[(size_t)p * a.lda + i];
+  };
   for (int j = 0; j < a.n; j++) {
     double* b = B + (size_t)j * a.ldb;
-    if (!a.trans) {
+    if (fwd) {
       for (int i = 0; i < a.m; i++) {
         double s = b[i];
-        for (int p = 0; p < i; p++) s -= L[(size_t)p * a.lda + i] * b[p];
-        b[i] = s / L[(size_t)i * a.lda + i];
+        for (int p = 0; p < i; p++) s -= elem(i, p) * b[p];
+        b[i] = a.unit ? s : s / L[(size_t)i * a.lda + i];
       }
     } else {
       for (int i = a.m - 1; i >= 0; i--) {
         double s = b[i];
-        for (int p = i + 1; p < a.m; p++)
-          s -= L[(size_t)i * a.lda + p] * b[p];
-        b[i] = s / L[(size_t)i * a.lda + i];
+        for (int p = i + 1; p < a.m; p++) s -= elem(i, p) * b[p];
+        b[i] = a.unit ? s : s / L[(size_t)i * a.lda + i];
       }
     }
   }
@@ -1059,10 +1065,12 @@ static void gpu_trsm_solve(Task& t, GpuTaskCtx& g) {
   const TrsmSolveArgs& a = t.arg<TrsmSolveArgs>();
   const double one = 1.0;
   rocblas_status s = rocblas_dtrsm(
-      blas_handle(g), rocblas_side_left, rocblas_fill_lower,
+      blas_handle(g), rocblas_side_left,
+      a.upper ? rocblas_fill_upper : rocblas_fill_lower,
       a.trans ? rocblas_operation_transpose : rocblas_operation_none,
-      rocblas_diagonal_non_unit, a.m, a.n, &one,
-      (const double*)t.dev_ptr[0], a.lda, (double*)t.dev_ptr[1], a.ldb);
+      a.unit ? rocblas_diagonal_unit : rocblas_diagonal_non_unit, a.m, a.n,
+      &one, (const double*)t.dev_ptr[0], a.lda, (double*)t.dev_ptr[1],
+      a.ldb);
   PA_CHECK(s == rocblas_status_success, "rocblas_dtrsm solve failed: %d",
            (int)s);
 }
@@ -1112,6 +1120,52 @@ void insert_potrs(Dtd& tp, TiledMatrix& A, TiledMatrix& B) {
 void insert_posv(Dtd& tp, TiledMatrix& A, TiledMatrix& B) {
   insert_potrf(tp, A);
   insert_potrs(tp, A, B);
+}
+
+// LU solve (dplasma dgetrs/dgesv nopiv analogs): after insert_getrf_nopiv
+// left L (unit lower) and U (upper) packed in A's tiles, solve A X = B:
+// forward L Y = B (unit diagonal), then backward U X = Y. Off-diagonal
+// updates use A's tiles directly: forward uses the strictly-lower tiles
+// (which hold L), backward the strictly-upper tiles (which hold U).
+void insert_getrs_nopiv(Dtd& tp, TiledMatrix& A, TiledMatrix& B) {
+  PA_CHECK(A.mt() == A.nt() && A.mt() == B.mt(),
+           "insert_getrs_nopiv: A must be square with B.mt == A.mt");
+  const int mt = A.mt();
+  auto trsm = [&](int k, int j, int upper) {
+    TrsmSolveArgs a{A.tile_rows(k), B.tile_cols(j), A.mb(), B.mb(), 0};
+    a.upper = upper;
+    a.unit = upper ? 0 : 1;  // L is unit-lower, U non-unit upper
+    Dtd::FlowSpec f[] = {{A.tile(k, k), ACCESS_IN},
+                         {B.tile(k, j), ACCESS_INOUT}};
+    tp.insert(&tc_trsm_solve(), &a, sizeof(a), f, 2, (1 << 20),
+              B.rank_of(k, j));
+  };
+  auto update = [&](int i, int k, int j) {
+    // B[i] -= A[i,k] * B[k]  (forward: i > k uses L tiles;
+    //                         backward: i < k uses U tiles)
+    GemmNNArgs a{B.tile_rows(i), B.tile_cols(j), A.tile_rows(k),
+                 A.mb(),         B.mb(),         B.mb(),
+                 -1.0,           1.0};
+    Dtd::FlowSpec f[] = {{A.tile(i, k), ACCESS_IN},
+                         {B.tile(k, j), ACCESS_IN},
+                         {B.tile(i, j), ACCESS_INOUT}};
+    tp.insert(&tc_gemm_nn(), &a, sizeof(a), f, 3, 0, B.rank_of(i, j));
+  };
+  for (int j = 0; j < B.nt(); j++) {
+    for (int k = 0; k < mt; k++) {  // forward: L Y = B
+      trsm(k, j, 0);
+      for (int i = k + 1; i < mt; i++) update(i, k, j);
+    }
+    for (int k = mt - 1; k >= 0; k--) {  // backward: U X = Y
+      trsm(k, j, 1);
+      for (int i = k - 1; i >= 0; i--) update(i, k, j);
+    }
+  }
+}
+
+void insert_gesv_nopiv(Dtd& tp, TiledMatrix& A, TiledMatrix& B) {
+  insert_getrf_nopiv(tp, A);
+  insert_getrs_nopiv(tp, A, B);
 }
 
 // Pre-create the per-stream rocBLAS handle and give it a fixed device

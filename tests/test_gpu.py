@@ -433,3 +433,28 @@ def test_gpu_posv_numerics(gctx):
     err = abs(X - ref).max() / abs(ref).max()
     assert err < 1e-10, err
     del A, B, tp, tp2
+
+
+def test_gpu_gesv_numerics(gctx):
+    """insert_gesv_nopiv on the GPU engine vs numpy.linalg.solve."""
+    n, nb, nrhs = 1024, 256, 256
+    A = pm.TiledMatrix(gctx, n, n, nb, nb, 1, 1)
+    B = pm.TiledMatrix(gctx, n, nrhs, nb, nb, 1, 1)
+    tp = pm.Dtd(gctx)
+    pm.insert_full_fill(tp, A, 9)
+    pm.insert_full_fill(tp, B, 5)
+    pm.insert_apply_scale(tp, A, 0.01, 0)
+    tp.wait()
+    for i in range(A.mt):
+        t = A.tile_numpy(i, i)
+        t += np.eye(nb) * 50.0
+        A.tile_numpy_set(i, i, t)
+    Af = full_matrix(A, nb, nb)
+    Bf = full_matrix(B, nb, nb)[:, :nrhs]
+    tp2 = pm.Dtd(gctx)
+    pm.insert_gesv_nopiv(tp2, A, B)
+    tp2.wait()
+    ref = np.linalg.solve(Af, Bf)
+    err = abs(full_matrix(B, nb, nb)[:, :nrhs] - ref).max() / abs(ref).max()
+    assert err < 1e-9, err
+    del A, B, tp, tp2

@@ -55,22 +55,33 @@ void Profiler::start(const std::string& filename) {
             filename_.c_str());
     return;
   }
-  // header: magic, dictionary as a JSON line, then raw records streamed
-  // in buffer-sized chunks (event classes register at static init, before
-  // tracing starts)
-  fprintf(file_, "PABT1\n{\"classes\":{");
+  // Header: magic + ONE dictionary JSON line padded to a fixed width,
+  // then raw records streamed in buffer-sized chunks. Task classes
+  // register LAZILY (function-local statics, first use), so the final
+  // dictionary is only known at stop — stop_and_dump() rewrites this
+  // padded line in place with the complete class map.
+  fprintf(file_, "PABT1\n");
+  header_off_ = ftell(file_);
+  std::string line = header_json_locked();
+  line.resize(HEADER_PAD - 1, ' ');
+  fprintf(file_, "%s\n", line.c_str());
+  enabled_.store(true, std::memory_order_release);
+}
+
+std::string Profiler::header_json_locked() const {
+  std::string s = "{\"classes\":{";
   bool first = true;
   for (auto& [id, name] : classes_) {
-    fprintf(file_, "%s\"%d\":\"%s\"", first ? "" : ",", id, name.c_str());
+    s += (first ? "" : ",");
+    s += "\"" + std::to_string(id) + "\":\"" + name + "\"";
     first = false;
   }
-  fprintf(file_,
-          "},\"kinds\":{\"1\":\"exec\",\"2\":\"gpu_task\",\"3\":"
-          "\"stage_in\",\"4\":\"comm_send\",\"5\":\"comm_recv\","
-          "\"6\":\"sched\",\"7\":\"gpu_span\"},"
-          "\"rec_bytes\":%zu}\n",
-          sizeof(TraceRec));
-  enabled_.store(true, std::memory_order_release);
+  s += "},\"kinds\":{\"1\":\"exec\",\"2\":\"gpu_task\",\"3\":"
+       "\"stage_in\",\"4\":\"comm_send\",\"5\":\"comm_recv\","
+       "\"6\":\"sched\",\"7\":\"gpu_span\"},"
+       "\"rec_bytes\":" + std::to_string(sizeof(TraceRec)) + "}";
+  PA_CHECK(s.size() < HEADER_PAD, "trace header dictionary too large");
+  return s;
 }
 
 void Profiler::register_class(int id, const std::string& name) {
@@ -108,6 +119,15 @@ void Profiler::stop_and_dump() {
   if (!enabled_.exchange(false)) return;
   std::lock_guard<std::mutex> g(mtx_);
   if (!file_) return;
+  {
+    // rewrite the padded header with the now-complete class dictionary
+    long cur = ftell(file_);
+    fseek(file_, header_off_, SEEK_SET);
+    std::string line = header_json_locked();
+    line.resize(HEADER_PAD - 1, ' ');
+    fprintf(file_, "%s\n", line.c_str());
+    fseek(file_, cur, SEEK_SET);
+  }
   for (Buf* b : bufs_) flush_buf_locked(b);
   fclose(file_);
   file_ = nullptr;

@@ -77,6 +77,9 @@ class Profiler {
 
   std::atomic<bool> enabled_{false};
   std::string filename_;
+  static constexpr size_t HEADER_PAD = 8192;
+  long header_off_ = 0;
+  std::string header_json_locked() const;
   std::mutex mtx_;
   FILE* file_ = nullptr;  // records stream to disk per chunk: a long traced
                           // run must not grow host RAM (profiling.c:74-160

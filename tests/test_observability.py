@@ -292,3 +292,37 @@ print("CLEAN_OK")
                         capture_output=True, text=True, timeout=60, cwd=REPO)
     assert r2.returncode == 0, r2.stdout + r2.stderr
     assert "rank" in r2.stdout and "100" in r2.stdout, r2.stdout
+
+
+def test_trace2pandas(tmp_path):
+    """PABT1 -> pandas trace tables (pbt2ptt analog): load + summary."""
+    trace = tmp_path / "ptrace"
+    code = f"""
+import sys; sys.path.insert(0, {REPO!r})
+import parsec_amd as pm
+pm.param_set("profile_filename", {str(trace)!r})
+ctx = pm.Context(nworkers=2, rank=0, world=1, gpu=-2)
+A = pm.TiledMatrix(ctx, 256, 256, 64, 64, 1, 1)
+tp = pm.Dtd(ctx)
+pm.insert_spd_fill(tp, A, 1)
+pm.insert_potrf(tp, A)
+tp.wait()
+del A, tp, ctx
+"""
+    r = subprocess.run([sys.executable, "-c", code], capture_output=True,
+                       text=True, timeout=120)
+    assert r.returncode == 0, r.stdout + r.stderr
+    from parsec_amd.tools.trace2pandas import load, summarize
+    df = load([str(trace) + ".0"])
+    assert len(df) > 10
+    assert {"kind", "class", "dur_us", "lane"} <= set(df.columns)
+    assert (df["class"] == "potrf").any()
+    s = summarize(df)
+    assert s["count"].sum() == len(df)
+    out = tmp_path / "t.csv"
+    r2 = subprocess.run([sys.executable, "-m",
+                         "parsec_amd.tools.trace2pandas",
+                         str(trace) + ".0", "--out", str(out)],
+                        capture_output=True, text=True, timeout=120,
+                        cwd=REPO)
+    assert r2.returncode == 0 and out.exists(), r2.stdout + r2.stderr

@@ -49,6 +49,11 @@ test: all
 gpu-test: all
 	python3 -m pytest tests -q -m gpu
 
+# CI entry: build + CPU suite (what a runner without a GPU can check).
+check: all
+	python3 -c "import __graft_entry__ as g; g.build()"
+	python3 -m pytest tests -q -m "not gpu"
+
 bench: all
 	python3 bench.py --steps 3 --warmup 1
 

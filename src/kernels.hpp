@@ -72,6 +72,9 @@ void insert_posv(Dtd& tp, TiledMatrix& A, TiledMatrix& B);
 // LU solve / factor+solve, no pivoting (dgetrs/dgesv nopiv analogs).
 void insert_getrs_nopiv(Dtd& tp, TiledMatrix& A, TiledMatrix& B);
 void insert_gesv_nopiv(Dtd& tp, TiledMatrix& A, TiledMatrix& B);
+// Least squares via BCGS QR (dgels analog): X = R^-1 Q^T B.
+void insert_gels_bcgs(Dtd& tp, TiledMatrix& A, TiledMatrix& R,
+                      TiledMatrix& B, TiledMatrix& X);
 void insert_reduce_sum(Dtd& tp, TiledMatrix& A, TiledMatrix& R);
 void insert_stencil_1d(Dtd& tp, TiledMatrix& Src, TiledMatrix& Dst);
 void insert_panel_fill(Dtd& tp, TiledMatrix& A, uint32_t seed);

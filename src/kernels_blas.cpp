@@ -1168,6 +1168,52 @@ void insert_gesv_nopiv(Dtd& tp, TiledMatrix& A, TiledMatrix& B) {
   insert_getrs_nopiv(tp, A, B);
 }
 
+// Least squares min ||A X - B|| via the BCGS QR (dgels analog, QR route):
+// insert_geqrf_bcgs leaves Q explicit in A (m x n) and R (n x n upper) —
+// X = R^{-1} Q^T B. Q^T B is a tiled TN GEMM chain; the triangular solve
+// is a backward sweep with the upper-triangular trsm_solve. X must be
+// A.nt x B.nt tiles. Inherits BCGS's cond(A) <~ 1e7 envelope.
+void insert_gels_bcgs(Dtd& tp, TiledMatrix& A, TiledMatrix& R,
+                      TiledMatrix& B, TiledMatrix& X) {
+  insert_geqrf_bcgs(tp, A, R);
+  PA_CHECK(X.mt() == A.nt() && X.nt() == B.nt() && B.mt() == A.mt(),
+           "insert_gels_bcgs: X must be A.nt x B.nt tiles, B.mt == A.mt");
+  // X[k,j] = sum_i Q[i,k]^T B[i,j]
+  for (int k = 0; k < A.nt(); k++)
+    for (int j = 0; j < B.nt(); j++)
+      for (int i = 0; i < A.mt(); i++) {
+        GemmNNArgs a{X.tile_rows(k), B.tile_cols(j), A.tile_rows(i),
+                     A.mb(),         B.mb(),         X.mb(),
+                     1.0,            i == 0 ? 0.0 : 1.0};
+        a.transA = 1;
+        Dtd::FlowSpec f[] = {{A.tile(i, k), ACCESS_IN},
+                             {B.tile(i, j), ACCESS_IN},
+                             {X.tile(k, j), i == 0 ? ACCESS_OUT
+                                                   : ACCESS_INOUT}};
+        tp.insert(&tc_gemm_nn(), &a, sizeof(a), f, 3, 0, X.rank_of(k, j));
+      }
+  // backward: R X = (Q^T B), R upper non-unit
+  for (int j = 0; j < X.nt(); j++)
+    for (int k = X.mt() - 1; k >= 0; k--) {
+      TrsmSolveArgs a{X.tile_rows(k), X.tile_cols(j), R.mb(), X.mb(), 0};
+      a.upper = 1;
+      Dtd::FlowSpec f[] = {{R.tile(k, k), ACCESS_IN},
+                           {X.tile(k, j), ACCESS_INOUT}};
+      tp.insert(&tc_trsm_solve(), &a, sizeof(a), f, 2, (1 << 20),
+                X.rank_of(k, j));
+      for (int i = k - 1; i >= 0; i--) {
+        GemmNNArgs a2{X.tile_rows(i), X.tile_cols(j), X.tile_rows(k),
+                      R.mb(),         X.mb(),         X.mb(),
+                      -1.0,           1.0};
+        Dtd::FlowSpec f2[] = {{R.tile(i, k), ACCESS_IN},
+                              {X.tile(k, j), ACCESS_IN},
+                              {X.tile(i, j), ACCESS_INOUT}};
+        tp.insert(&tc_gemm_nn(), &a2, sizeof(a2), f2, 3, 0,
+                  X.rank_of(i, j));
+      }
+    }
+}
+
 // Pre-create the per-stream rocBLAS handle and give it a fixed device
 // workspace so no allocation can happen inside a hipStream capture
 // (gpu_graph.cpp calls this for each capture stream before BeginCapture;

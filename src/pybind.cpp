@@ -458,6 +458,9 @@ PYBIND11_MODULE(_core, m) {
         py::arg("A"), py::arg("B"), py::call_guard<py::gil_scoped_release>());
   m.def("insert_gesv_nopiv", &insert_gesv_nopiv, py::arg("tp"),
         py::arg("A"), py::arg("B"), py::call_guard<py::gil_scoped_release>());
+  m.def("insert_gels_bcgs", &insert_gels_bcgs, py::arg("tp"), py::arg("A"),
+        py::arg("R"), py::arg("B"), py::arg("X"),
+        py::call_guard<py::gil_scoped_release>());
   m.def("insert_advise_prefetch", &insert_advise_prefetch, py::arg("tp"),
         py::arg("tile"), py::call_guard<py::gil_scoped_release>());
   m.def("insert_gemm_fp64", &insert_gemm_fp64, py::arg("tp"), py::arg("A"),

@@ -458,3 +458,25 @@ def test_gpu_gesv_numerics(gctx):
     err = abs(full_matrix(B, nb, nb)[:, :nrhs] - ref).max() / abs(ref).max()
     assert err < 1e-9, err
     del A, B, tp, tp2
+
+
+def test_gpu_gels_numerics(gctx):
+    """insert_gels_bcgs on the GPU engine vs numpy.linalg.lstsq."""
+    m, n, nb, nrhs = 2048, 1024, 256, 256
+    A = pm.TiledMatrix(gctx, m, n, nb, nb, 1, 1)
+    R = pm.TiledMatrix(gctx, n, n, nb, nb, 1, 1)
+    B = pm.TiledMatrix(gctx, m, nrhs, nb, nb, 1, 1)
+    X = pm.TiledMatrix(gctx, n, nrhs, nb, nb, 1, 1)
+    tp = pm.Dtd(gctx)
+    pm.insert_full_fill(tp, A, 3)
+    pm.insert_full_fill(tp, B, 5)
+    tp.wait()
+    Af = full_matrix(A, nb, nb)[:, :n]
+    Bf = full_matrix(B, nb, nb)[:, :nrhs]
+    tp2 = pm.Dtd(gctx)
+    pm.insert_gels_bcgs(tp2, A, R, B, X)
+    tp2.wait()
+    ref, *_ = np.linalg.lstsq(Af, Bf, rcond=None)
+    err = abs(full_matrix(X, nb, nb)[:, :nrhs] - ref).max() / abs(ref).max()
+    assert err < 1e-7, err
+    del A, R, B, X, tp, tp2

@@ -183,3 +183,34 @@ def test_qr_bcgs_gpu():
     assert orth < 1e-10, f"orthogonality defect {orth}"
     assert recon < 1e-12, f"reconstruction err {recon}"
     del A, R, ctx
+
+
+def test_gels_bcgs_vs_lstsq(ctx):
+    """insert_gels_bcgs (dgels analog, QR route): overdetermined least
+    squares X = R^-1 Q^T B vs numpy.linalg.lstsq, partial RHS tiles."""
+    import numpy as np
+    m, n, nb, nrhs = 512, 256, 64, 96
+    A = pm.TiledMatrix(ctx, m, n, nb, nb, 1, 1)
+    R = pm.TiledMatrix(ctx, n, n, nb, nb, 1, 1)
+    B = pm.TiledMatrix(ctx, m, nrhs, nb, nb, 1, 1)
+    X = pm.TiledMatrix(ctx, n, nrhs, nb, nb, 1, 1)
+    tp = pm.Dtd(ctx)
+    pm.insert_full_fill(tp, A, 3)
+    pm.insert_full_fill(tp, B, 5)
+    tp.wait()
+
+    def full(M):
+        out = np.zeros((M.m, M.n))
+        for i in range(M.mt):
+            for j in range(M.nt):
+                out[i * M.mb:i * M.mb + M.tile_rows(i),
+                    j * M.nb:j * M.nb + M.tile_cols(j)] = M.tile_numpy(i, j)
+        return out
+
+    Af, Bf = full(A), full(B)
+    tp2 = pm.Dtd(ctx)
+    pm.insert_gels_bcgs(tp2, A, R, B, X)
+    tp2.wait()
+    ref, *_ = np.linalg.lstsq(Af, Bf, rcond=None)
+    err = abs(full(X) - ref).max() / abs(ref).max()
+    assert err < 1e-8, err

@@ -68,8 +68,10 @@ class TcpComm : public CommEngine {
 
   const char* kind() const override { return "tcp"; }
 
+  // Self-sends loop back through the comm thread (an app computing a
+  // destination may land on itself — common in dynamic token patterns);
+  // counters balance exactly like a remote delivery.
   void send_ctl(int dst, uint32_t tag, const void* p, size_t n) override {
-    PA_CHECK(dst != rank_, "send_ctl: no self-sends");
     if (!(tag & CTL_SYS_BIT))
       ctl_sent_.fetch_add(1, std::memory_order_relaxed);
     {
@@ -290,20 +292,7 @@ class TcpComm : public CommEngine {
     } else if (p.hdr.kind == FK_CTL) {
       std::string pl((const char*)p.in_payload.data(), p.in_payload.size());
       p.in_payload.clear();
-      uint32_t tag = (uint32_t)p.hdr.seq;
-      std::lock_guard<std::mutex> g(ctl_mtx_);
-      if (tag & CTL_SYS_BIT) {
-        if (sys_handler_)
-          sys_handler_(peer, tag, pl);
-        else
-          sys_stash_.push_back({peer, tag, std::move(pl)});
-      } else {
-        ctl_recvd_.fetch_add(1, std::memory_order_relaxed);
-        if (ctl_handler_)
-          ctl_handler_(peer, tag, pl);
-        else
-          ctl_stash_.push_back({peer, tag, std::move(pl)});
-      }
+      deliver_ctl(peer, (uint32_t)p.hdr.seq, std::move(pl));
     } else if (p.hdr.kind == FK_BAR_IN) {
       bar_arrivals_[p.hdr.seq]++;
       check_barrier_root();
@@ -457,6 +446,22 @@ class TcpComm : public CommEngine {
     fatal("comm: connection to rank %d lost (r=%zd errno=%d)", peer, r, errno);
   }
 
+  void deliver_ctl(int src, uint32_t tag, std::string pl) {
+    std::lock_guard<std::mutex> g(ctl_mtx_);
+    if (tag & CTL_SYS_BIT) {
+      if (sys_handler_)
+        sys_handler_(src, tag, pl);
+      else
+        sys_stash_.push_back({src, tag, std::move(pl)});
+    } else {
+      ctl_recvd_.fetch_add(1, std::memory_order_relaxed);
+      if (ctl_handler_)
+        ctl_handler_(src, tag, pl);
+      else
+        ctl_stash_.push_back({src, tag, std::move(pl)});
+    }
+  }
+
   void main_loop() {
     std::vector<pollfd> pfds;
     while (!stop_.load(std::memory_order_acquire)) {
@@ -471,9 +476,13 @@ class TcpComm : public CommEngine {
         ctls.swap(ctl_out_);
       }
       for (Task* t : cmds) process_cmd(t);
-      for (auto& cm : ctls)
-        queue_frame(cm.dst, FK_CTL, cm.tag, cm.payload.data(),
-                    cm.payload.size(), nullptr);
+      for (auto& cm : ctls) {
+        if (cm.dst == rank_)  // loopback: deliver on the comm thread
+          deliver_ctl(rank_, cm.tag, std::move(cm.payload));
+        else
+          queue_frame(cm.dst, FK_CTL, cm.tag, cm.payload.data(),
+                      cm.payload.size(), nullptr);
+      }
       for (uint64_t e : bars) {
         if (rank_ == 0) {
           bar_root_armed_ = std::max(bar_root_armed_, e);

@@ -684,3 +684,62 @@ print("POSV_RANK_OK", rank)
     ref = np.linalg.solve(Af, Bf)
     err = np.abs(X - ref).max() / np.abs(ref).max()
     assert err < 1e-11, f"distributed posv rel err {err}"
+
+
+def test_ctl_self_send_loopback(tmp_path):
+    """send_ctl(dst == my rank) loops back through the comm thread: apps
+    with computed destinations (dynamic token patterns) may land on
+    themselves, and the termdet counters must balance exactly as for a
+    remote delivery (wait_dynamic below would hang on an imbalance)."""
+    import conftest
+    REPO = os.path.dirname(HERE)
+    port = str(conftest.port_base(33))
+    code = r"""
+import os, sys
+sys.path.insert(0, os.environ["PARSEC_REPO"])
+import numpy as np
+import parsec_amd as pm
+rank = int(os.environ["RANK"])
+pm.param_set("comm_base_port", os.environ["PORT"])
+ctx = pm.Context(nworkers=2, rank=rank, world=2, comm="tcp", gpu=-2)
+A = pm.TiledMatrix(ctx, 64, 32, 32, 32, 2, 1)
+if A.is_local(rank, 0):
+    A.tile_numpy_set(rank, 0, np.zeros((32, 32)))
+tp = pm.Dtd(ctx)
+def on_token(src, tag, payload):
+    b = int(payload)
+    def body(buf):
+        np.frombuffer(buf, dtype=np.float64)[:] += 1
+    tp.insert_py(body, [(A.tile(rank, 0), pm.ACCESS_INOUT)], rank=rank,
+                 with_data=True)
+    if b > 1:
+        # alternate SELF and peer: half the hops are loopbacks
+        dst = rank if b % 2 == 0 else 1 - rank
+        ctx.send_ctl(dst, 5, str(b - 1).encode())
+if rank == 0:
+    pass
+ctx.set_ctl_handler(on_token)
+if rank == 0:
+    on_token(0, 5, b"12")
+tp.wait_dynamic()
+mine = int(A.tile_numpy(rank, 0)[0, 0]) if A.is_local(rank, 0) else 0
+print("SELF_OK", rank, mine)
+ctx.barrier()
+del A, ctx
+"""
+    procs = []
+    total = 0
+    for r in range(2):
+        env = dict(os.environ)
+        env.update(RANK=str(r), WORLD_SIZE="2", PORT=port,
+                   PARSEC_REPO=REPO)
+        procs.append(subprocess.Popen([sys.executable, "-c", code], env=env,
+                                      stdout=subprocess.PIPE,
+                                      stderr=subprocess.STDOUT))
+    for pr in procs:
+        out, _ = pr.communicate(timeout=120)
+        assert pr.returncode == 0 and b"SELF_OK" in out, out.decode()
+        for line in out.decode().splitlines():
+            if line.startswith("SELF_OK"):
+                total += int(line.split()[2])
+    assert total == 12, total

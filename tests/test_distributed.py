@@ -743,3 +743,69 @@ del A, ctx
             if line.startswith("SELF_OK"):
                 total += int(line.split()[2])
     assert total == 12, total
+
+
+def test_wait_dynamic_random_split(tmp_path):
+    """Randomized dynamic-activation fuzz (the pattern that found the
+    ctl self-send gap): each token inserts one task and randomly splits
+    its remaining budget across random ranks (self included). Budget
+    conservation makes the global task count exactly K; wait_dynamic
+    must neither hang nor return early (sum of counters < K)."""
+    import conftest
+    REPO = os.path.dirname(HERE)
+    code = r"""
+import os, sys, random
+sys.path.insert(0, os.environ["PARSEC_REPO"])
+import numpy as np
+import parsec_amd as pm
+rank = int(os.environ["RANK"]); world = int(os.environ["WORLD_SIZE"])
+seed = int(os.environ["SEED"]); K = int(os.environ["K"])
+pm.param_set("comm_base_port", os.environ["PORT"])
+ctx = pm.Context(nworkers=2, rank=rank, world=world, comm="tcp", gpu=-2)
+A = pm.TiledMatrix(ctx, 32 * world, 32, 32, 32, world, 1)
+if A.is_local(rank, 0):
+    A.tile_numpy_set(rank, 0, np.zeros((32, 32)))
+tp = pm.Dtd(ctx)
+rng = random.Random(seed * 1000 + rank)
+def on_token(src, tag, payload):
+    b = int(payload)
+    def body(buf):
+        np.frombuffer(buf, dtype=np.float64)[:] += 1
+    tp.insert_py(body, [(A.tile(rank, 0), pm.ACCESS_INOUT)], rank=rank,
+                 with_data=True)
+    rest = b - 1
+    if rest > 0:
+        parts = [rest]
+        if rest > 1 and rng.random() < 0.5:
+            cut = rng.randrange(1, rest)
+            parts = [cut, rest - cut]
+        for p in parts:
+            ctx.send_ctl(rng.randrange(world), 5, str(p).encode())
+ctx.set_ctl_handler(on_token)
+if rank == 0:
+    on_token(0, 5, str(K).encode())
+tp.wait_dynamic()
+mine = int(A.tile_numpy(rank, 0)[0, 0]) if A.is_local(rank, 0) else 0
+print("TDYNF", rank, mine)
+ctx.barrier()
+del A, ctx
+"""
+    for seed in (3007, 3013):
+        port = _next_port[0]
+        _next_port[0] += 6
+        procs = []
+        for r in range(4):
+            env = dict(os.environ)
+            env.update(RANK=str(r), WORLD_SIZE="4", PORT=str(port),
+                       SEED=str(seed), K="150", PARSEC_REPO=REPO)
+            procs.append(subprocess.Popen([sys.executable, "-c", code],
+                                          env=env, stdout=subprocess.PIPE,
+                                          stderr=subprocess.STDOUT))
+        total = 0
+        for pr in procs:
+            out, _ = pr.communicate(timeout=120)
+            assert pr.returncode == 0, out.decode()
+            for line in out.decode().splitlines():
+                if line.startswith("TDYNF"):
+                    total += int(line.split()[2])
+        assert total == 150, (seed, total)

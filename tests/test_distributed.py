@@ -110,7 +110,8 @@ A = pm.TiledMatrix(ctx, NT * nb, nb, nb, nb, world, 1)
 tp = pm.Dtd(ctx, "fuzz")
 oracle = []
 for i in range(NT):
-    v = np.full((nb, nb), float(i + 1))
+    # non-symmetric so transposed (reshape-promise) reads are observable
+    v = np.arange(nb * nb, dtype=np.float64).reshape(nb, nb) * 1e-3 + i + 1
     oracle.append(v.copy())
     if A.is_local(i, 0):
         A.tile_numpy_set(i, 0, v)
@@ -122,14 +123,16 @@ for step in range(int(os.environ.get("OPS", "150"))):
     srcs = rng.sample([i for i in range(NT) if i != target], nsrc)
     exec_rank = rng.randrange(world)
     coeffs = [round(rng.uniform(-1, 1), 3) for _ in range(nsrc + 1)]
+    # ~1/4 of reads consume the TRANSPOSED copy (reshape promise)
+    trans = [rng.random() < 0.25 for _ in srcs]
     write_only = nsrc == 0 and rng.random() < 0.3
     # oracle replay (pure numpy, every rank computes the same)
     if write_only:
         oracle[target] = np.full((nb, nb), coeffs[0])
     else:
         acc = coeffs[0] * oracle[target]
-        for c, s in zip(coeffs[1:], srcs):
-            acc = acc + c * oracle[s]
+        for c, s, tr in zip(coeffs[1:], srcs, trans):
+            acc = acc + c * (oracle[s].T if tr else oracle[s])
         oracle[target] = acc
     # task body (only runs on exec_rank)
     def body(tbuf, *sbufs, coeffs=coeffs, write_only=write_only):
@@ -143,7 +146,9 @@ for step in range(int(os.environ.get("OPS", "150"))):
         t[:] = acc
     flows = [(A.tile(target, 0),
               pm.ACCESS_OUT if write_only else pm.ACCESS_INOUT)]
-    flows += [(A.tile(s, 0), pm.ACCESS_IN) for s in srcs]
+    flows += [(A.tile(s, 0), pm.ACCESS_IN, pm.RESHAPE_TRANSPOSE) if tr
+              else (A.tile(s, 0), pm.ACCESS_IN)
+              for s, tr in zip(srcs, trans)]
     tp.insert_py(body, flows=flows, rank=exec_rank, with_data=True)
 
 tp.wait()

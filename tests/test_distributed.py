@@ -107,6 +107,11 @@ ctx = pm.Context(nworkers=2, rank=rank, world=world, comm="tcp", gpu=-2)
 rng = random.Random(int(os.environ["SEED"]))  # identical on all ranks
 NT, nb = 10, 16
 A = pm.TiledMatrix(ctx, NT * nb, nb, nb, nb, world, 1)
+if rng.random() < 0.3:
+    # ~1/3 of seeds run on a random rank table (two_dim_tabular analog):
+    # owner placement drives every SPMD channel decision, so shuffle it
+    table = [rng.randrange(world) for _ in range(NT)]
+    A.set_rank_table(table)
 tp = pm.Dtd(ctx, "fuzz")
 oracle = []
 for i in range(NT):

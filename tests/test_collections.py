@@ -415,3 +415,18 @@ def test_reduce_axis(ctx):
     for i in range(nt):
         want = sum(M[i*nb:(i+1)*nb, j*nb:(j+1)*nb] for j in range(nt))
         assert np.allclose(Rr.tile_numpy(i, 0), want)
+
+
+def test_sym_alias_flush_idempotent(ctx):
+    """flush_all on a sym collection visits tile(i,j) and its alias
+    tile(j,i): the second flush of the same Data must be a no-op (the
+    owner is already home), world 1 or N."""
+    import numpy as np
+    A = pm.TiledMatrix(ctx, 256, 256, 64, 64, 1, 1, sym=True)
+    tp = pm.Dtd(ctx)
+    pm.insert_spd_fill(tp, A, 3)
+    tp.wait()
+    assert A.tile(1, 2) is A.tile(2, 1)  # the alias IS the same Data
+    tp.flush_all(A)  # must not wedge or double-send
+    tp.wait()
+    assert np.isfinite(A.tile_numpy(2, 1)).all()

@@ -326,3 +326,76 @@ del A, tp, ctx
                         capture_output=True, text=True, timeout=120,
                         cwd=REPO)
     assert r2.returncode == 0 and out.exists(), r2.stdout + r2.stderr
+
+
+def test_live_stats_multirank(tmp_path):
+    """Per-rank live snapshots: two ranks publish to <path>.<rank> and
+    live_top aggregates both."""
+    from conftest import port_base
+    live = str(tmp_path / "mlive")
+    code = f"""
+import os, sys, time; sys.path.insert(0, {REPO!r})
+import parsec_amd as pm
+rank = int(os.environ["RANK"])
+pm.param_set("comm_base_port", os.environ["PORT"])
+pm.param_set("live_stats", {live!r})
+pm.param_set("live_stats_interval_ms", "50")
+ctx = pm.Context(nworkers=2, rank=rank, world=2, comm="tcp", gpu=-2)
+A = pm.TiledMatrix(ctx, 256, 256, 64, 64, 2, 1)
+tp = pm.Dtd(ctx); pm.insert_spd_fill(tp, A, 1); pm.insert_potrf(tp, A)
+tp.wait()
+time.sleep(0.3)
+import json
+s = json.load(open({live!r} + f".{{rank}}"))
+assert s["rank"] == rank and s["world"] == 2, s
+ctx.barrier()
+print("MLIVE_OK", rank)
+del A, tp, ctx
+"""
+    port = str(port_base(35))
+    procs = []
+    for r in range(2):
+        env = dict(os.environ)
+        env.update(RANK=str(r), WORLD_SIZE="2", PORT=port)
+        procs.append(subprocess.Popen([sys.executable, "-c", code], env=env,
+                                      stdout=subprocess.PIPE,
+                                      stderr=subprocess.STDOUT))
+    for pr in procs:
+        out, _ = pr.communicate(timeout=120)
+        assert pr.returncode == 0 and b"MLIVE_OK" in out, out.decode()
+
+
+def test_trace2pandas_multirank(tmp_path):
+    """trace tables merge multiple rank files with a rank column."""
+    from conftest import port_base
+    trace = str(tmp_path / "mtrace")
+    code = f"""
+import os, sys; sys.path.insert(0, {REPO!r})
+import parsec_amd as pm
+rank = int(os.environ["RANK"])
+pm.param_set("comm_base_port", os.environ["PORT"])
+pm.param_set("profile_filename", {trace!r})
+ctx = pm.Context(nworkers=2, rank=rank, world=2, comm="tcp", gpu=-2)
+A = pm.TiledMatrix(ctx, 256, 256, 64, 64, 2, 1)
+tp = pm.Dtd(ctx); pm.insert_spd_fill(tp, A, 1); pm.insert_potrf(tp, A)
+tp.wait()
+ctx.barrier()
+del A, tp, ctx
+print("MT_OK", rank)
+"""
+    port = str(port_base(37))
+    procs = []
+    for r in range(2):
+        env = dict(os.environ)
+        env.update(RANK=str(r), WORLD_SIZE="2", PORT=port)
+        procs.append(subprocess.Popen([sys.executable, "-c", code], env=env,
+                                      stdout=subprocess.PIPE,
+                                      stderr=subprocess.STDOUT))
+    for pr in procs:
+        out, _ = pr.communicate(timeout=120)
+        assert pr.returncode == 0 and b"MT_OK" in out, out.decode()
+    from parsec_amd.tools.trace2pandas import load
+    df = load([f"{trace}.0", f"{trace}.1"])
+    assert set(df["rank"].unique()) == {0, 1}
+    assert (df["kind"] == "comm_send").any() or \
+        (df["kind"] == "comm_recv").any()

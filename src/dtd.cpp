@@ -216,7 +216,9 @@ Task* Dtd::insert_begin(const TaskClass* tc, const void* args,
   int task_rank = rank;
   if (task_rank < 0) {
     for (int i = 0; i < nflows && task_rank < 0; i++)
-      if (flows[i].mode & ACCESS_OUT) task_rank = flows[i].d->home_rank;
+      if ((flows[i].mode & ACCESS_OUT) &&
+          !(flows[i].mode & ACCESS_UNTRACKED))
+        task_rank = flows[i].d->home_rank;
     if (task_rank < 0) task_rank = 0;
   }
   PA_CHECK(task_rank < world_);
@@ -235,7 +237,9 @@ Task* Dtd::insert_begin(const TaskClass* tc, const void* args,
                "reshape: only READ flows consume converted copies");
       d = reshaped_promise(d, flows[i].reshape, task_rank);
     }
-    if (d) {  // NULL flow (e.g. absent stencil halo)
+    if (d && !(flows[i].mode & ACCESS_UNTRACKED)) {
+      // NULL flow (e.g. absent stencil halo) and UNTRACKED flows skip
+      // the chaining protocol entirely (PARSEC_DONT_TRACK analog)
       if (flows[i].mode & ACCESS_IN) d = read_flow(d, t, task_rank);
       if (flows[i].mode & ACCESS_OUT)
         d = write_flow(d, t, task_rank,

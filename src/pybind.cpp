@@ -414,6 +414,7 @@ PYBIND11_MODULE(_core, m) {
       .def_property_readonly("n_tasks", &GpuGraph::n_tasks);
 
   m.attr("ACCESS_IN") = (int)ACCESS_IN;
+  m.attr("ACCESS_UNTRACKED") = (int)ACCESS_UNTRACKED;
   m.attr("ACCESS_OUT") = (int)ACCESS_OUT;
   m.attr("ACCESS_INOUT") = (int)ACCESS_INOUT;
   m.attr("RESHAPE_TRANSPOSE") = (int)Reshape::TRANSPOSE;

@@ -48,6 +48,12 @@ enum AccessMode : uint8_t {
   ACCESS_IN = 1,
   ACCESS_OUT = 2,
   ACCESS_INOUT = 3,
+  // Modifier (OR with IN/OUT/INOUT): exempt this flow from dependency
+  // tracking — no chaining edges, no WAR/WAW/renaming, no transfers
+  // (PARSEC_DONT_TRACK analog, dtd_test_flag_dont_track). The caller
+  // asserts the access cannot conflict; the buffer must be locally
+  // valid. Staging/dirty marking still follow the base IN/OUT bits.
+  ACCESS_UNTRACKED = 8,
 };
 
 enum class TaskKind : uint8_t {

@@ -545,3 +545,40 @@ del mats, tp, ctx
                            text=True, timeout=300)
         assert r.returncode == 0 and "FUZZ_OK" in r.stdout, \
             f"sched={kind}: {r.stdout}{r.stderr}"
+
+
+def test_untracked_flow_semantics():
+    """ACCESS_UNTRACKED (PARSEC_DONT_TRACK analog): the flow is exempt
+    from chaining — an untracked reader must NOT WAR-serialize a later
+    tracked writer, and untracked reads of stable data see correct
+    values."""
+    import numpy as np
+    import time
+    ctx = pm.Context(nworkers=2, rank=0, world=1, gpu=-2)
+    A = pm.TiledMatrix(ctx, 128, 64, 64, 64, 1, 1)
+    B = pm.TiledMatrix(ctx, 64, 64, 64, 64, 1, 1)
+    tp = pm.Dtd(ctx)
+    A.tile_numpy_set(0, 0, np.full((64, 64), 7.0))
+    A.tile_numpy_set(1, 0, np.zeros((64, 64)))
+    times = {}
+
+    def slow_reader(buf):
+        times["r0"] = time.perf_counter()
+        v = np.frombuffer(buf, dtype=np.float64)[0]
+        assert v == 7.0, v  # correct value through the untracked read
+        time.sleep(0.5)
+        times["r1"] = time.perf_counter()
+
+    def writer(buf):
+        np.frombuffer(buf, dtype=np.float64)[:] = 9.0
+        times["w1"] = time.perf_counter()
+
+    tp.insert_py(slow_reader,
+                 [(A.tile(0, 0), pm.ACCESS_IN | pm.ACCESS_UNTRACKED)],
+                 with_data=True)
+    tp.insert_py(writer, [(A.tile(0, 0), pm.ACCESS_OUT)], with_data=True)
+    tp.wait()
+    assert np.all(A.tile_numpy(0, 0) == 9.0)
+    # the tracked writer must not have waited for the untracked reader
+    assert times["w1"] < times["r1"], times
+    del A, B, tp, ctx

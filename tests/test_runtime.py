@@ -582,3 +582,38 @@ def test_untracked_flow_semantics():
     # the tracked writer must not have waited for the untracked reader
     assert times["w1"] < times["r1"], times
     del A, B, tp, ctx
+
+
+def test_empty_taskpool_wait(ctx):
+    """dtd_test_empty analog: waiting an empty pool returns immediately;
+    wait() is re-entrant after completion."""
+    tp = pm.Dtd(ctx)
+    tp.wait()
+    tp.wait()
+    tp.wait_dynamic()
+
+
+def test_multiple_taskpool_concurrent_wait(ctx):
+    """dtd_test_multiple_handle_wait analog: two pools driven from two
+    application threads, each waiting its own, both complete."""
+    import threading
+    import numpy as np
+    A = pm.TiledMatrix(ctx, 256, 256, 64, 64, 1, 1)
+    B = pm.TiledMatrix(ctx, 256, 256, 64, 64, 1, 1)
+    errs = []
+
+    def run(M, seed):
+        try:
+            tp = pm.Dtd(ctx)
+            pm.insert_spd_fill(tp, M, seed)
+            pm.insert_potrf(tp, M)
+            tp.wait()
+            assert np.isfinite(M.tile_numpy(3, 3)).all()
+        except Exception as e:  # propagate to the main thread
+            errs.append(e)
+
+    t1 = threading.Thread(target=run, args=(A, 1))
+    t2 = threading.Thread(target=run, args=(B, 2))
+    t1.start(); t2.start(); t1.join(); t2.join()
+    assert not errs, errs
+    del A, B

@@ -76,6 +76,8 @@ void insert_gesv_nopiv(Dtd& tp, TiledMatrix& A, TiledMatrix& B);
 void insert_gels_bcgs(Dtd& tp, TiledMatrix& A, TiledMatrix& R,
                       TiledMatrix& B, TiledMatrix& X);
 void insert_reduce_sum(Dtd& tp, TiledMatrix& A, TiledMatrix& R);
+// log-depth binary-tree variant (BT_reduction.jdf analog).
+void insert_reduce_sum_tree(Dtd& tp, TiledMatrix& A, TiledMatrix& R);
 void insert_stencil_1d(Dtd& tp, TiledMatrix& Src, TiledMatrix& Dst);
 void insert_panel_fill(Dtd& tp, TiledMatrix& A, uint32_t seed);
 void insert_potrf_panel(Dtd& tp, TiledMatrix& A);

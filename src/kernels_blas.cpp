@@ -498,9 +498,91 @@ TaskClass& tc_add_tile() {
   return tc;
 }
 
+// out = a + b (pure OUTPUT third flow — the tree-reduction combiner)
+__global__ void k_add2(double* out, const double* a, const double* b,
+                       size_t n) {
+  size_t i = (size_t)blockIdx.x * blockDim.x + threadIdx.x;
+  for (; i < n; i += (size_t)gridDim.x * blockDim.x) out[i] = a[i] + b[i];
+}
+
+static void cpu_add2(Task& t) {
+  const double* a = (const double*)t.flows[0].data->pull_to_host();
+  const double* b = (const double*)t.flows[1].data->pull_to_host();
+  Data* o = t.flows[2].data;
+  double* q = (double*)o->ensure_host();
+  for (size_t i = 0; i < o->bytes / 8; i++) q[i] = a[i] + b[i];
+  o->written_on(false);
+}
+
+static void gpu_add2(Task& t, GpuTaskCtx& g) {
+  Data* o = t.flows[2].data;
+  hipLaunchKernelGGL(k_add2, dim3(2048), dim3(256), 0, g.stream,
+                     (double*)t.dev_ptr[2], (const double*)t.dev_ptr[0],
+                     (const double*)t.dev_ptr[1], o->bytes / 8);
+}
+
+TaskClass& tc_add2() {
+  static TaskClass tc = make_tc("add2", TaskKind::GPU, cpu_add2, gpu_add2, 33);
+  return tc;
+}
+
 // result += sum of all tiles of A (reduce.jdf / DTD reduce analog).
 // The accumulation chains on `result`; contributions from each rank's
 // tiles flow through the comm engine automatically.
+// Binary-tree reduction (BT_reduction.jdf analog): log2(N) critical path
+// instead of the flat chain's N. Internal workspace tiles are owned by
+// the taskpool (the bcgs-style pattern); pair placement follows the left
+// operand so cross-rank combines ride the normal dataflow.
+void insert_reduce_sum_tree(Dtd& tp, TiledMatrix& A, TiledMatrix& R) {
+  std::vector<Data*> level;
+  std::vector<int> lrank;
+  for (int m = 0; m < A.mt(); m++)
+    for (int n = 0; n < (A.sym() ? m + 1 : A.nt()); n++) {
+      level.push_back(A.tile(m, n));
+      lrank.push_back(A.rank_of(m, n));
+    }
+  PA_CHECK(!level.empty(), "reduce_sum_tree: empty collection");
+  auto* ctx = A.ctx();
+  // one workspace strip holds every internal node (N-1 combines)
+  int nw = (int)level.size();  // >= combines + final copy target
+  auto W = std::make_shared<TiledMatrix>(ctx, (int64_t)A.mb() * nw, A.nb(),
+                                         A.mb(), A.nb(), 1, 1);
+  {
+    std::vector<int> ranks((size_t)nw);
+    for (int i = 0; i < nw; i++) ranks[i] = lrank[i % lrank.size()];
+    W->set_rank_table(std::move(ranks));
+  }
+  tp.own(W);
+  int wi = 0;
+  while (level.size() > 1) {
+    std::vector<Data*> next;
+    std::vector<int> nrank;
+    for (size_t i = 0; i + 1 < level.size(); i += 2) {
+      bool last = level.size() == 2;
+      Data* out = last ? R.tile(0, 0) : W->tile(wi, 0);
+      int orank = last ? R.rank_of(0, 0) : lrank[i];
+      if (!last) wi++;
+      Dtd::FlowSpec f[] = {{level[i], ACCESS_IN},
+                           {level[i + 1], ACCESS_IN},
+                           {out, ACCESS_OUT}};
+      tp.insert(&tc_add2(), nullptr, 0, f, 3, 0, orank);
+      next.push_back(out);
+      nrank.push_back(orank);
+    }
+    if (level.size() % 2) {  // odd leftover promotes unchanged
+      next.push_back(level.back());
+      nrank.push_back(lrank.back());
+    }
+    level.swap(next);
+    lrank.swap(nrank);
+  }
+  if (level[0] != R.tile(0, 0)) {
+    // single-tile collection: copy the lone input into R
+    Dtd::FlowSpec f[] = {{level[0], ACCESS_IN}, {R.tile(0, 0), ACCESS_OUT}};
+    tp.insert(&tc_copy_tile(), nullptr, 0, f, 2, 0, R.rank_of(0, 0));
+  }
+}
+
 void insert_reduce_sum(Dtd& tp, TiledMatrix& A, TiledMatrix& R) {
   Data* r = R.tile(0, 0);
   for (int m = 0; m < A.mt(); m++)

@@ -470,6 +470,8 @@ PYBIND11_MODULE(_core, m) {
   m.def("insert_apply_scale", &insert_apply_scale, py::arg("tp"), py::arg("A"),
         py::arg("alpha"), py::arg("beta"),
         py::call_guard<py::gil_scoped_release>());
+  m.def("insert_reduce_sum_tree", &insert_reduce_sum_tree, py::arg("tp"),
+        py::arg("A"), py::arg("R"), py::call_guard<py::gil_scoped_release>());
   m.def("insert_reduce_sum", &insert_reduce_sum, py::arg("tp"), py::arg("A"),
         py::arg("R"), py::call_guard<py::gil_scoped_release>());
   m.def("insert_stencil_1d", &insert_stencil_1d, py::arg("tp"),

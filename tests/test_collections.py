@@ -430,3 +430,22 @@ def test_sym_alias_flush_idempotent(ctx):
     tp.flush_all(A)  # must not wedge or double-send
     tp.wait()
     assert np.isfinite(A.tile_numpy(2, 1)).all()
+
+
+def test_reduce_sum_tree(ctx):
+    """Binary-tree reduction (BT_reduction.jdf analog): log-depth combine
+    equals the flat chain and the numpy sum, odd/even/single tile counts
+    included."""
+    import numpy as np
+    for mt, nt in [(1, 1), (3, 2), (4, 4), (5, 3)]:
+        A = pm.TiledMatrix(ctx, mt * 32, nt * 32, 32, 32, 1, 1)
+        R = pm.TiledMatrix(ctx, 32, 32, 32, 32, 1, 1)
+        tp = pm.Dtd(ctx)
+        pm.insert_full_fill(tp, A, mt * 10 + nt)
+        tp.wait()
+        tp2 = pm.Dtd(ctx)
+        pm.insert_reduce_sum_tree(tp2, A, R)
+        tp2.wait()
+        ref = sum(A.tile_numpy(i, j) for i in range(mt) for j in range(nt))
+        assert abs(R.tile_numpy(0, 0) - ref).max() < 1e-12, (mt, nt)
+        del A, R, tp, tp2

@@ -24,6 +24,7 @@
 #include <map>
 #include <mutex>
 #include <queue>
+#include <unordered_map>
 #include <vector>
 
 extern "C" {
@@ -59,6 +60,17 @@ constexpr int MAXP = 8;   // params per task (packed into the 64-byte args)
 constexpr int MAXF = 8;
 
 using PKey = std::pair<int, std::array<long, MAXP>>;
+
+struct PKeyHash {
+  size_t operator()(const PKey& k) const {
+    uint64_t h = 1469598103934665603ull ^ (uint64_t)k.first;
+    for (long v : k.second) {
+      h ^= (uint64_t)v;
+      h *= 1099511628211ull;
+    }
+    return (size_t)h;
+  }
+};
 
 struct Inst {
   int cls = 0;
@@ -290,8 +302,8 @@ class Compact {
   std::vector<ClassFns> fns_;
   std::vector<void*> classes_;
   std::mutex mu_;
-  std::map<void*, PKey> live_;   // inserted, not yet completed
-  std::map<PKey, Pending> state_;  // the frontier
+  std::unordered_map<void*, PKey> live_;  // inserted, not yet completed
+  std::unordered_map<PKey, Pending, PKeyHash> state_;  // the frontier
   long hook_ = -1;
   long total_ = -1;
   long inserted_ = 0;

@@ -176,7 +176,7 @@ class Graph {
   void* dtd_;
   int rank_;
   std::vector<Inst> insts_;
-  std::map<PKey, int> index_;
+  std::unordered_map<PKey, int, PKeyHash> index_;
   std::vector<void*> classes_;
 };
 

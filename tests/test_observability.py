@@ -399,3 +399,10 @@ print("MT_OK", rank)
     assert set(df["rank"].unique()) == {0, 1}
     assert (df["kind"] == "comm_send").any() or \
         (df["kind"] == "comm_recv").any()
+    # multi-rank Chrome merge: one timeline, rank r under pid 10r
+    from parsec_amd.tools.trace2chrome import convert_many
+    import json as _json
+    out, n = convert_many([f"{trace}.0", f"{trace}.1"],
+                          str(tmp_path / "merged.json"))
+    evs = _json.load(open(out))["traceEvents"]
+    assert n > 10 and {e.get("pid") for e in evs} >= {0, 10}

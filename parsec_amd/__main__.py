@@ -22,12 +22,20 @@ def main():
           "parsec_amd.param_set):\n")
     # register the common params by touching a context-free surface
     print(pm.param_dump())
-    print("Key params: sched_workers, sched (ws|fifo|lifo), "
-          "gpu_exec_streams, gpu_max_inflight, gpu_mem_percent, "
-          "gpu_mem_limit_mb, chore_gemm (rocblas|hip), chore_syrk "
-          "(hip|dgemm|syrkx), chore_potrf (hip|rocsolver), trsm_variant "
+    print("Key params: sched_workers, "
+          "sched (ws|fifo|lifo|spq|pbq|ip|rnd + reference-name aliases), "
+          "sched_bind (0|1|numa), gpu_exec_streams, gpu_max_inflight, "
+          "gpu_mem_percent, gpu_mem_limit_mb, chore_gemm (rocblas|hip), "
+          "chore_syrk (hip|dgemm|syrkx), chore_potrf (hip|rocsolver), "
+          "chore_qr (hand|rocsolver), qr_algo (house|bcgs), trsm_variant "
           "(invgemm|rocblas), dtd_window_size, comm_kind (tcp|rccl), "
-          "comm_base_port, profile_filename, profile_dot, stats")
+          "comm_max_inflight, comm_send_reserve, bcast_tree "
+          "(unicast|binomial), comm_base_port, profile_filename, "
+          "profile_dot, profile_roctx, stats, debug_history, "
+          "pins (task_profiler,print_steals,iterators_checker), "
+          "live_stats, live_stats_interval_ms, graph_debug")
+    print("\nTools: python -m parsec_amd.tools.trace2chrome | "
+          "trace2pandas | live_top ;  python -m parsec_amd.ptg file.jdf")
 
 
 if __name__ == "__main__":

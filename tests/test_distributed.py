@@ -105,7 +105,7 @@ rank = int(os.environ["RANK"]); world = int(os.environ["WORLD_SIZE"])
 pm.param_set("comm_base_port", os.environ["PORT"])
 ctx = pm.Context(nworkers=2, rank=rank, world=world, comm="tcp", gpu=-2)
 rng = random.Random(int(os.environ["SEED"]))  # identical on all ranks
-NT, nb = 10, 16
+NT, nb = int(os.environ.get("FUZZ_NT", "10")), 16
 A = pm.TiledMatrix(ctx, NT * nb, nb, nb, nb, world, 1)
 if rng.random() < 0.3:
     # ~1/3 of seeds run on a random rank table (two_dim_tabular analog):
